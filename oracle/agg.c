@@ -1,0 +1,692 @@
+/*
+ * agg.c — CPU restatement of openGemini's windowed reduction
+ * (engine/aggregate_cursor.go, engine/series_agg_func.gen.go,
+ *  engine/series_agg_reducer.gen.go, lib/record/column_util.go).
+ * TEST INFRASTRUCTURE ONLY — see oracle.h header note.
+ */
+#include "oracle.h"
+#include <string.h>
+#include <math.h>
+#include <stdlib.h>
+
+#define MAX_ROWS_PER_SEG 4096 /* reference caps segments at 1000 rows
+                                 (lib/util/util.go:72); slack for tests */
+
+/* influxql.MinTime/MaxTime (lib/util/lifted/influx/influxql/ast.go:92,102) */
+#define INFLUX_MIN_TIME (INT64_MIN + 2)
+#define INFLUX_MAX_TIME (INT64_MAX - 1)
+
+void orc_window(int64_t t, int64_t start_time, int64_t end_time, int64_t interval,
+                int64_t offset, int64_t *win_start, int64_t *win_end) {
+  /* lib/util/lifted/influx/query/select.go:579-656, no timezone */
+  if (interval == 0) {
+    *win_start = start_time;
+    *win_end = end_time + 1;
+    return;
+  }
+  t -= offset;
+  int64_t dt = t % interval;
+  if (dt < 0) dt += interval;
+  int64_t start;
+  if (INFLUX_MIN_TIME + dt >= t)
+    start = INFLUX_MIN_TIME;
+  else
+    start = t - dt;
+  start += offset;
+  int64_t d2 = interval - dt;
+  int64_t end;
+  if (INFLUX_MAX_TIME - d2 <= t)
+    end = INFLUX_MAX_TIME;
+  else
+    end = t + d2;
+  end += offset;
+  *win_start = start;
+  *win_end = end;
+}
+
+/* ---------------- ColVal helpers over a normalised (offset-0) bitmap ------ */
+
+static inline int bit_at(const uint8_t *bm, int i) {
+  return (bm[i >> 3] >> (i & 7)) & 1;
+}
+
+static int valid_count(const uint8_t *bm, int nilcount, int rows, int start, int end) {
+  /* lib/record/column.go:297-314 */
+  if (rows == 0 || nilcount == rows) return 0;
+  if (nilcount == 0) return end - start;
+  int c = 0;
+  for (int i = start; i < end; i++) c += bit_at(bm, i);
+  return c;
+}
+
+/* ---------------- per-op reduce over one segment group -------------------- */
+
+typedef struct {
+  int64_t index;
+  orc_val value;
+  int isnil;
+} redout;
+
+/* count: series_agg_func.gen.go:24-42 */
+static void reduce_count(const uint8_t *bm, int nilcount, int rows, int start, int end,
+                         redout *o) {
+  int64_t c = valid_count(bm, nilcount, rows, start, end);
+  o->index = start;
+  o->value.i = c;
+  o->isnil = (c == 0);
+}
+
+/* sum: series_agg_func.gen.go:48-78 (index = VALUE start, bug-compatible) */
+static void reduce_sum(int col_type, const orc_val *vals, const uint8_t *bm,
+                       int nilcount, int rows, int start, int end, redout *o) {
+  if (rows == 0) { /* cv.Length()+cv.NilCount==0 */
+    o->index = start;
+    o->value.i = 0;
+    o->isnil = 1;
+    return;
+  }
+  int vs = start, ve = end;
+  if (nilcount != 0) { /* GetValIndexRange, column.go:482-487 */
+    vs = valid_count(bm, nilcount, rows, 0, start);
+    ve = vs + valid_count(bm, nilcount, rows, start, end);
+  }
+  o->index = vs;
+  o->isnil = (ve == vs);
+  if (col_type == ORC_TYPE_FLOAT) {
+    double s = 0;
+    for (int i = vs; i < ve; i++) s += vals[i].f;
+    o->value.f = s;
+  } else {
+    int64_t s = 0;
+    for (int i = vs; i < ve; i++) s += vals[i].i;
+    o->value.i = s;
+  }
+}
+
+/* min/max: column_util.go:190-278 (first occurrence wins; Go NaN compare
+ * semantics carry over to C: NaN comparisons are false) */
+static void reduce_minmax(int col_type, int is_max, const orc_val *vals,
+                          const uint8_t *bm, int nilcount, int rows, int dense,
+                          int start, int end, redout *o) {
+  if (dense == 0) {
+    o->index = 0;
+    o->value.i = 0;
+    o->isnil = 1;
+    return;
+  }
+  int64_t row = -1;
+  orc_val best = {0};
+  if (nilcount == 0) {
+    best = vals[start];
+    row = start;
+    for (int i = start; i < end; i++) {
+      int better;
+      if (col_type == ORC_TYPE_FLOAT)
+        better = is_max ? (best.f < vals[i].f) : (best.f > vals[i].f);
+      else
+        better = is_max ? (best.i < vals[i].i) : (best.i > vals[i].i);
+      if (better) {
+        best = vals[i];
+        row = i;
+      }
+    }
+  } else {
+    int skip = valid_count(bm, nilcount, rows, 0, start);
+    int vIdx = skip;
+    for (int i = start; i < end && vIdx < dense; i++) {
+      if (!bit_at(bm, i)) continue;
+      int better;
+      if (vIdx == skip)
+        better = 1;
+      else if (col_type == ORC_TYPE_FLOAT)
+        better = is_max ? (best.f < vals[vIdx].f) : (best.f > vals[vIdx].f);
+      else
+        better = is_max ? (best.i < vals[vIdx].i) : (best.i > vals[vIdx].i);
+      if (better) {
+        best = vals[vIdx];
+        row = i;
+      }
+      vIdx++;
+    }
+  }
+  if (row == -1) { /* series_agg_func.gen.go:86-88 */
+    o->index = 0;
+    o->value.i = 0;
+    o->isnil = 1;
+  } else {
+    o->index = row;
+    o->value = best;
+    o->isnil = 0;
+  }
+}
+
+/* first/last: column_util.go:23-85 */
+static void reduce_first(const orc_val *vals, const uint8_t *bm, int nilcount,
+                         int rows, int dense, int start, int end, redout *o) {
+  if (dense == 0) {
+    o->index = 0;
+    o->value.i = 0;
+    o->isnil = 1;
+    return;
+  }
+  if (nilcount == 0) {
+    o->index = start;
+    o->value = vals[start];
+    o->isnil = 0;
+    return;
+  }
+  int vIdx = valid_count(bm, nilcount, rows, 0, start);
+  for (int i = start; i < end && vIdx < dense; i++) {
+    if (!bit_at(bm, i)) continue;
+    o->index = i;
+    o->value = vals[vIdx];
+    o->isnil = 0;
+    return;
+  }
+  o->index = 0;
+  o->value.i = 0;
+  o->isnil = 1;
+}
+
+static void reduce_last(const orc_val *vals, const uint8_t *bm, int nilcount, int rows,
+                        int dense, int start, int end, redout *o) {
+  if (dense == 0) {
+    o->index = 0;
+    o->value.i = 0;
+    o->isnil = 1;
+    return;
+  }
+  if (nilcount == 0) {
+    o->index = end - 1;
+    o->value = vals[end - 1];
+    o->isnil = 0;
+    return;
+  }
+  int64_t row = -1;
+  for (int i = end - 1; i >= start; i--) {
+    if (bit_at(bm, i)) {
+      row = i;
+      break;
+    }
+  }
+  if (row < start) {
+    o->index = 0;
+    o->value.i = 0;
+    o->isnil = 1;
+    return;
+  }
+  int vIdx = valid_count(bm, nilcount, rows, 0, (int)row);
+  o->index = row;
+  o->value = vals[vIdx];
+  o->isnil = 0;
+}
+
+/* ---------------- prevBuf-style accumulator + merge ----------------------- */
+/* one colBuf per op (series_agg_reducer.gen.go:70-120) */
+typedef struct {
+  int active; /* !isNil */
+  orc_val value;
+  int64_t time;
+  int64_t niltime; /* bug-compatible Time(index) of the last isNil group */
+} opacc;
+
+/* fv() merge semantics, series_agg_func.gen.go:44-274 */
+static void merge_op(int col_type, int op, opacc *a, const redout *r, int64_t rtime) {
+  if (r->isnil) {
+    a->niltime = rtime;
+    return;
+  }
+  if (!a->active) {
+    a->active = 1;
+    a->value = r->value;
+    a->time = rtime;
+    return;
+  }
+  switch (op) {
+  case ORC_AGG_COUNT: /* integerCountMerge: value += */
+    a->value.i += r->value.i;
+    break;
+  case ORC_AGG_SUM:
+    if (col_type == ORC_TYPE_FLOAT)
+      a->value.f += r->value.f;
+    else
+      a->value.i += r->value.i;
+    break;
+  case ORC_AGG_MIN: {
+    int repl = (col_type == ORC_TYPE_FLOAT) ? (r->value.f < a->value.f)
+                                            : (r->value.i < a->value.i);
+    if (repl) {
+      a->value = r->value;
+      a->time = rtime;
+    }
+    break;
+  }
+  case ORC_AGG_MAX: {
+    int repl = (col_type == ORC_TYPE_FLOAT) ? (r->value.f > a->value.f)
+                                            : (r->value.i > a->value.i);
+    if (repl) {
+      a->value = r->value;
+      a->time = rtime;
+    }
+    break;
+  }
+  case ORC_AGG_FIRST: /* floatFirstMerge: keep prev */
+    break;
+  case ORC_AGG_LAST: /* floatLastMerge: assign curr */
+    a->value = r->value;
+    a->time = rtime;
+    break;
+  }
+}
+
+/* ---------------- full scan-aggregate ------------------------------------- */
+
+typedef struct {
+  int64_t win_start;
+  int64_t first_row_time;
+  int has_win;
+  opacc acc[6]; /* count,sum,min,max,first,last */
+} winstate;
+
+static const int OPS[6] = {ORC_AGG_COUNT, ORC_AGG_SUM, ORC_AGG_MIN,
+                           ORC_AGG_MAX,   ORC_AGG_FIRST, ORC_AGG_LAST};
+
+static void flush_window(uint64_t sid, const winstate *w, orc_agg_row *out) {
+  memset(out, 0, sizeof(*out));
+  out->sid = sid;
+  out->win_start = w->win_start;
+  out->first_row_time = w->first_row_time;
+  const opacc *a = w->acc;
+  out->count = a[0].active ? a[0].value.i : 0;
+  out->count_time = a[0].active ? a[0].time : a[0].niltime;
+  out->sum = a[1].value;
+  out->sum_time = a[1].active ? a[1].time : a[1].niltime;
+  out->sum_isnil = !a[1].active;
+  out->minv = a[2].value;
+  out->min_time = a[2].active ? a[2].time : a[2].niltime;
+  out->min_isnil = !a[2].active;
+  out->maxv = a[3].value;
+  out->max_time = a[3].active ? a[3].time : a[3].niltime;
+  out->max_isnil = !a[3].active;
+  out->firstv = a[4].value;
+  out->first_time = a[4].active ? a[4].time : a[4].niltime;
+  out->first_isnil = !a[4].active;
+  out->lastv = a[5].value;
+  out->last_time = a[5].active ? a[5].time : a[5].niltime;
+  out->last_isnil = !a[5].active;
+}
+
+int64_t orc_scan_agg(const uint8_t *blob, int64_t blob_len, const orc_seg_desc *descs,
+                     int64_t nsegs, int col_type, int64_t start_time, int64_t end_time,
+                     int64_t interval, int64_t offset, orc_agg_row *out,
+                     int64_t out_cap) {
+  orc_val *vals = (orc_val *)malloc(MAX_ROWS_PER_SEG * sizeof(orc_val));
+  int64_t *times = (int64_t *)malloc(MAX_ROWS_PER_SEG * 8);
+  uint8_t *bm = (uint8_t *)malloc(MAX_ROWS_PER_SEG / 8 + 1);
+  if (!vals || !times || !bm) return -1;
+
+  int64_t nout = 0;
+  uint64_t cur_sid = 0;
+  int have_sid = 0;
+  winstate w;
+  memset(&w, 0, sizeof(w));
+  int64_t rc = -1;
+
+  for (int64_t s = 0; s < nsegs; s++) {
+    const orc_seg_desc *d = &descs[s];
+    if (d->data_offset + d->data_size > (uint64_t)blob_len ||
+        d->time_offset + d->time_size > (uint64_t)blob_len)
+      goto done;
+
+    if (!have_sid || d->sid != cur_sid) {
+      if (have_sid && w.has_win) {
+        if (nout >= out_cap) goto done;
+        flush_window(cur_sid, &w, &out[nout++]);
+      }
+      memset(&w, 0, sizeof(w));
+      cur_sid = d->sid;
+      have_sid = 1;
+    }
+
+    int rows = 0, nilcount = 0, trows = 0;
+    if (orc_decode_time_segment(blob + d->time_offset, d->time_size, times, &trows))
+      goto done;
+    if (orc_decode_data_segment(col_type, blob + d->data_offset, d->data_size, vals,
+                                bm, &rows, &nilcount))
+      goto done;
+    if (rows != trows || rows > MAX_ROWS_PER_SEG) goto done;
+    int dense = rows - nilcount;
+
+    /* intervalIndex over this segment (aggregate_cursor.go:343-356) */
+    int start = 0;
+    while (start < rows) {
+      int64_t ws, we;
+      orc_window(times[start], start_time, end_time, interval, offset, &ws, &we);
+      int end = start;
+      while (end < rows && times[end] >= ws && times[end] < we) end++;
+
+      if (w.has_win && w.win_start != ws) {
+        if (nout >= out_cap) goto done;
+        flush_window(cur_sid, &w, &out[nout++]);
+        memset(&w, 0, sizeof(w));
+      }
+      if (!w.has_win) {
+        w.has_win = 1;
+        w.win_start = ws;
+      }
+      /* every contributing record overwrites: deriveIntervalIndex reads the
+       * CURRENT record's times (aggregate_cursor.go:370-374) */
+      w.first_row_time = times[start];
+
+      for (int oi = 0; oi < 6; oi++) {
+        redout r;
+        int64_t rtime;
+        switch (OPS[oi]) {
+        case ORC_AGG_COUNT:
+          reduce_count(bm, nilcount, rows, start, end, &r);
+          rtime = times[r.index];
+          break;
+        case ORC_AGG_SUM:
+          reduce_sum(col_type, vals, bm, nilcount, rows, start, end, &r);
+          rtime = times[r.index < rows ? r.index : rows - 1]; /* bug-compat
+                     Time(valueIndex); clamp guards the all-valid edge */
+          break;
+        case ORC_AGG_MIN:
+          reduce_minmax(col_type, 0, vals, bm, nilcount, rows, dense, start, end, &r);
+          rtime = times[r.index];
+          break;
+        case ORC_AGG_MAX:
+          reduce_minmax(col_type, 1, vals, bm, nilcount, rows, dense, start, end, &r);
+          rtime = times[r.index];
+          break;
+        case ORC_AGG_FIRST:
+          reduce_first(vals, bm, nilcount, rows, dense, start, end, &r);
+          rtime = times[r.index];
+          break;
+        default:
+          reduce_last(vals, bm, nilcount, rows, dense, start, end, &r);
+          rtime = times[r.index];
+          break;
+        }
+        /* all-nil column: index := start (series_agg_reducer.gen.go:224-226) */
+        if (nilcount == rows && rows > 0) rtime = times[start];
+        merge_op(col_type, OPS[oi], &w.acc[oi], &r, rtime);
+      }
+      start = end;
+    }
+  }
+  if (have_sid && w.has_win) {
+    if (nout >= out_cap) goto done;
+    flush_window(cur_sid, &w, &out[nout++]);
+  }
+  rc = nout;
+done:
+  free(vals);
+  free(times);
+  free(bm);
+  return rc;
+}
+
+int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,
+                        const orc_seg_desc *descs, int64_t nsegs, int col_type,
+                        int64_t start_time, int64_t end_time, int64_t interval,
+                        int64_t offset, orc_agg_row *out, int64_t out_cap,
+                        int nthreads) {
+  /* split the descriptor list at sid boundaries into nthreads chunks;
+   * each series' windows land contiguously so outputs concatenate */
+  if (nsegs == 0) return 0;
+#ifdef _OPENMP
+  extern int omp_get_max_threads(void);
+  if (nthreads <= 0) nthreads = omp_get_max_threads();
+#else
+  nthreads = 1;
+#endif
+  if (nthreads > 64) nthreads = 64;
+
+  int64_t cuts[65];
+  cuts[0] = 0;
+  for (int t = 1; t < nthreads; t++) {
+    int64_t c = nsegs * t / nthreads;
+    /* advance to a sid boundary */
+    while (c < nsegs && c > 0 && descs[c].sid == descs[c - 1].sid) c++;
+    cuts[t] = c;
+  }
+  cuts[nthreads] = nsegs;
+
+  int64_t counts[64];
+  int64_t per_cap = out_cap; /* each chunk bounded by total cap */
+  orc_agg_row **bufs = (orc_agg_row **)malloc(sizeof(void *) * nthreads);
+  int fail = 0;
+#pragma omp parallel for num_threads(nthreads) schedule(static, 1)
+  for (int t = 0; t < nthreads; t++) {
+    int64_t lo = cuts[t], hi = cuts[t + 1];
+    if (lo >= hi) {
+      counts[t] = 0;
+      bufs[t] = 0;
+      continue;
+    }
+    /* worst case: one row per (seg,window-span) — bound loosely */
+    int64_t cap = per_cap;
+    bufs[t] = (orc_agg_row *)malloc(sizeof(orc_agg_row) * cap);
+    int64_t n = orc_scan_agg(blob, blob_len, descs + lo, hi - lo, col_type,
+                             start_time, end_time, interval, offset, bufs[t], cap);
+    if (n < 0) fail = 1;
+    counts[t] = n;
+  }
+  int64_t nout = 0;
+  for (int t = 0; t < nthreads; t++) {
+    if (!fail && counts[t] > 0) {
+      if (nout + counts[t] > out_cap)
+        fail = 1;
+      else {
+        memcpy(out + nout, bufs[t], sizeof(orc_agg_row) * counts[t]);
+        nout += counts[t];
+      }
+    }
+    free(bufs[t]);
+  }
+  free(bufs);
+  return fail ? -1 : nout;
+}
+
+/* ---------------- aggregateCursor record-stream emulation ----------------- */
+
+int64_t orc_agg_cursor(int col_type, uint32_t op, int multi_call,
+                       const void *vals_dense, const uint8_t *valid_bits,
+                       const int64_t *times, const int32_t *rec_rows, int nrecs,
+                       int64_t start_time, int64_t end_time, int64_t interval,
+                       int64_t offset, int max_record_size, int out_type,
+                       void *out_vals, uint8_t *out_nils, int64_t *out_times,
+                       int32_t *out_rec_rows, int *out_nrecs) {
+  /* aggregateCursor.Next (aggregate_cursor.go:267-385) driving ONE typed
+   * reducer (series_agg_reducer.gen.go:206-300) over a record stream.
+   * Records are slices of the concatenated arrays. out_type: value type of
+   * the output column (count → int). */
+  (void)out_type;
+  if (max_record_size <= 0) max_record_size = 1024;
+
+  int64_t val_pos = 0; /* dense value cursor over vals_dense */
+  int64_t row_pos = 0; /* row cursor over valid_bits/times */
+  int64_t nout = 0;
+  int cur_rec_rows = 0;
+  int nrecs_out = 0;
+
+  opacc acc;
+  memset(&acc, 0, sizeof(acc));
+
+  const double *fvals = (const double *)vals_dense;
+  const int64_t *ivals = (const int64_t *)vals_dense;
+  double *ofv = (double *)out_vals;
+  int64_t *oiv = (int64_t *)out_vals;
+
+  /* output append helper via macro to keep types straight */
+#define EMIT(valexpr_f, valexpr_i, isnil_, time_)                                     \
+  do {                                                                                \
+    if (op == ORC_AGG_COUNT || col_type == ORC_TYPE_INT)                              \
+      oiv[nout] = (isnil_) ? 0 : (valexpr_i);                                         \
+    else                                                                              \
+      ofv[nout] = (isnil_) ? 0 : (valexpr_f);                                         \
+    out_nils[nout] = (uint8_t)(isnil_);                                               \
+    out_times[nout] = (time_);                                                        \
+    nout++;                                                                           \
+    cur_rec_rows++;                                                                   \
+    if (cur_rec_rows >= max_record_size) {                                            \
+      out_rec_rows[nrecs_out++] = cur_rec_rows;                                       \
+      cur_rec_rows = 0;                                                               \
+    }                                                                                 \
+  } while (0)
+
+  for (int rec = 0; rec < nrecs; rec++) {
+    int rows = rec_rows[rec];
+    if (rows == 0) continue;
+    const int64_t *rtimes = times + row_pos;
+    const uint8_t *rbits = valid_bits; /* absolute row indexing below */
+
+    /* per-record dense values & nils */
+    int nilcount = 0;
+    for (int i = 0; i < rows; i++)
+      if (!((rbits[(row_pos + i) >> 3] >> ((row_pos + i) & 7)) & 1)) nilcount++;
+    int dense = rows - nilcount;
+
+    /* normalised record-local bitmap */
+    uint8_t lbm[MAX_ROWS_PER_SEG / 8 + 1];
+    memset(lbm, 0, sizeof(lbm));
+    for (int i = 0; i < rows; i++)
+      if ((rbits[(row_pos + i) >> 3] >> ((row_pos + i) & 7)) & 1)
+        lbm[i >> 3] |= (uint8_t)(1 << (i & 7));
+
+    const orc_val *rvals = (const orc_val *)(col_type == ORC_TYPE_FLOAT
+                                                 ? (const void *)(fvals + val_pos)
+                                                 : (const void *)(ivals + val_pos));
+
+    /* sameWindow: inNextWindow (aggregate_cursor.go:314-341) */
+    int same_window = 0;
+    {
+      /* find next non-empty record */
+      int has_next = 0;
+      int64_t next_first_time = 0;
+      int64_t rp = row_pos + rows;
+      for (int nr = rec + 1; nr < nrecs; nr++) {
+        if (rec_rows[nr] > 0) {
+          has_next = 1;
+          next_first_time = times[rp];
+          break;
+        }
+      }
+      /* a zero-row next record ⇒ inNextWin=true in Go; replicate: any
+       * following record with 0 rows and none after with >0 rows →
+       * nextRecord.RowNums()==0 → true. We check in order: */
+      int saw_empty_next = (rec + 1 < nrecs) && rec_rows[rec + 1] == 0;
+      if (saw_empty_next) {
+        same_window = 1;
+      } else if (has_next) {
+        if (interval == 0) {
+          same_window = 1; /* !HasInterval ⇒ inNextWin=true */
+        } else {
+          int64_t ws, we;
+          orc_window(next_first_time, start_time, end_time, interval, offset, &ws,
+                     &we);
+          int64_t last_t = rtimes[rows - 1];
+          same_window = (ws <= last_t && last_t < we);
+        }
+      }
+    }
+
+    /* intervalIndex (aggregate_cursor.go:343-356) */
+    int idx[MAX_ROWS_PER_SEG];
+    int nidx = 0;
+    if (interval == 0) {
+      idx[nidx++] = 0;
+    } else {
+      int64_t ws = 0, we = 0;
+      for (int i = 0; i < rows; i++) {
+        if (i == 0 || rtimes[i] >= we || rtimes[i] < ws) {
+          idx[nidx++] = i;
+          orc_window(rtimes[i], start_time, end_time, interval, offset, &ws, &we);
+        }
+      }
+    }
+
+    /* reducer.Aggregate (series_agg_reducer.gen.go:206-300) */
+    int first_index = 0, last_index = nidx - 1;
+    for (int gi = 0; gi < nidx; gi++) {
+      int gstart = idx[gi];
+      int gend = (gi < last_index) ? idx[gi + 1] : rows;
+
+      redout r;
+      switch (op) {
+      case ORC_AGG_COUNT:
+        reduce_count(lbm, nilcount, rows, gstart, gend, &r);
+        break;
+      case ORC_AGG_SUM:
+        reduce_sum(col_type, rvals, lbm, nilcount, rows, gstart, gend, &r);
+        break;
+      case ORC_AGG_MIN:
+        reduce_minmax(col_type, 0, rvals, lbm, nilcount, rows, dense, gstart, gend, &r);
+        break;
+      case ORC_AGG_MAX:
+        reduce_minmax(col_type, 1, rvals, lbm, nilcount, rows, dense, gstart, gend, &r);
+        break;
+      case ORC_AGG_FIRST:
+        reduce_first(rvals, lbm, nilcount, rows, dense, gstart, gend, &r);
+        break;
+      default:
+        reduce_last(rvals, lbm, nilcount, rows, dense, gstart, gend, &r);
+        break;
+      }
+      if (nilcount == rows) r.index = gstart; /* :224-226 */
+      int64_t rtime = rtimes[r.index < rows ? r.index : rows - 1];
+
+      /* multiCall output time: the CURRENT record's window-start row
+       * (deriveIntervalIndex, aggregate_cursor.go:358-375) */
+      int64_t mc_time = rtimes[gstart];
+
+      if (!r.isnil) {
+        if (gi == first_index && acc.active) {
+          /* A.1: merge into prevBuf */
+          merge_op(col_type, (int)op, &acc, &r, rtime);
+          if (first_index == last_index && same_window) {
+            /* A.1.1: stays pending */
+          } else {
+            /* A.1.2: flush prevBuf */
+            EMIT(acc.value.f, acc.value.i, 0, multi_call ? mc_time : acc.time);
+            memset(&acc, 0, sizeof(acc));
+          }
+          continue;
+        } else if (gi == last_index && same_window) {
+          /* A.2: becomes prevBuf */
+          memset(&acc, 0, sizeof(acc));
+          merge_op(col_type, (int)op, &acc, &r, rtime);
+          break;
+        }
+        /* A.3: complete window */
+        EMIT(r.value.f, r.value.i, 0, multi_call ? mc_time : rtime);
+      } else {
+        if (gi == first_index && acc.active &&
+            (first_index < last_index || !same_window)) {
+          /* B.1: flush prevBuf */
+          EMIT(acc.value.f, acc.value.i, 0, multi_call ? mc_time : acc.time);
+          memset(&acc, 0, sizeof(acc));
+          continue;
+        } else if (gi == last_index && same_window) {
+          break; /* B.2 */
+        }
+        /* B.3: null row */
+        EMIT(0, 0, 1, multi_call ? mc_time : rtime);
+      }
+    }
+
+    val_pos += dense;
+    row_pos += rows;
+  }
+  /* end of stream: Next() returns pending newRecord rows; prevBuf flushed by
+   * the final inNextWindow(nil)=false pass — the loop above already emitted
+   * everything except a still-pending acc (possible only if the last record
+   * ended at A.2/B.2 with same_window, which requires a next record; at true
+   * EOF same_window=false so nothing pends). */
+  (void)start_time;
+  if (cur_rec_rows > 0) out_rec_rows[nrecs_out++] = cur_rec_rows;
+  *out_nrecs = nrecs_out;
+  return nout;
+#undef EMIT
+}
