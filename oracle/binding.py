@@ -81,6 +81,10 @@ def load():
         C.POINTER(C.c_int64),
         C.POINTER(C.c_uint64),
     )
+    lib.orc_zigzag_encode.restype = C.c_uint64
+    lib.orc_zigzag_encode.argtypes = [C.c_int64]
+    lib.orc_zigzag_decode.restype = C.c_int64
+    lib.orc_zigzag_decode.argtypes = [C.c_uint64]
     lib.orc_gorilla_encode.restype = i64
     lib.orc_gorilla_encode.argtypes = [f64p, i64, u8p, i64]
     lib.orc_gorilla_decode.restype = i64

@@ -1,0 +1,304 @@
+"""Round-trip + edge-case tests for the CPU oracle codecs.
+
+Ports the shape of the reference's own codec tests:
+  lib/compress/float_test.go:67-153 (one/rand/small/same/int/smallDelta/RLE/
+  Snappy/NaN/Inf), lib/encoding/encoding_test.go:49-843 (int/timestamp round
+  trips). The reference cannot run here (no Go toolchain);
+  these ported tests + the transcribed golden aggregate cases pin the oracle.
+"""
+
+import numpy as np
+import pytest
+
+import binding as orc
+
+
+def rt_float(values):
+    v = np.asarray(values, dtype=np.float64)
+    b = orc.float_encode(v)
+    d = orc.float_decode(b)
+    assert len(d) == len(v)
+    nan = np.isnan(v)
+    assert np.array_equal(d[~nan], v[~nan])
+    assert np.isnan(d[nan]).all()
+    return b[0] >> 4 if len(b) else None
+
+
+def rt_int(values):
+    v = np.asarray(values, dtype=np.int64)
+    b = orc.int_encode(v)
+    assert np.array_equal(orc.int_decode(b), v)
+    return b[0] >> 4 if len(b) else None
+
+
+def rt_time(values):
+    v = np.asarray(values, dtype=np.int64)
+    b = orc.time_encode(v)
+    assert np.array_equal(orc.time_decode(b), v)
+    return b[0] >> 4 if len(b) else None
+
+
+class TestFloatCodec:
+    # mirrors lib/compress/float_test.go cases
+    def test_one(self):
+        rt_float([0.0])
+
+    def test_rand(self):
+        rng = np.random.default_rng(42)
+        tag = rt_float(rng.random(1000) * 1000)
+        assert tag in (0, 2, 3)
+
+    def test_small(self):
+        rng = np.random.default_rng(1)
+        assert rt_float((rng.integers(0, 10000, 4)) / 100) == 0  # null: n<=4
+
+    def test_same(self):
+        v = np.full(1000, 123.456)
+        assert rt_float(v) == 4
+        assert rt_float(np.zeros(1000)) == 4
+
+    def test_same_zero_is_two_bytes(self):
+        # compress.go:43-48: zero value stores no payload
+        b = orc.float_encode(np.zeros(1000))
+        assert len(b) == 3  # tag + u16 count
+
+    def test_int_values(self):
+        rng = np.random.default_rng(2)
+        tag = rt_float(rng.integers(0, 100, 1000).astype(np.float64))
+        assert tag == 3  # intOnly => gorilla, compresses well
+
+    def test_small_delta(self):
+        rng = np.random.default_rng(3)
+        rt_float(2 + 0.1 + rng.random(1000) / 10)
+
+    def test_rle(self):
+        v = (np.arange(1000) // 180).astype(np.float64)
+        assert rt_float(v) == 5
+
+    def test_rle_long_runs_split(self):
+        # run cap 1<<14 (compress.go:26)
+        v = np.concatenate([np.zeros(20000), np.full(20000, 7.5)])
+        assert rt_float(v) == 5
+
+    def test_nan_few_distinct_goes_rle(self):
+        v = (np.arange(1000) // 180).astype(np.float64)
+        v[1] = np.nan
+        assert rt_float(v) == 5
+
+    def test_nan_many_distinct(self):
+        rng = np.random.default_rng(4)
+        v = np.cumsum(rng.normal(0, 1, 1000))
+        v[7] = np.nan
+        tag = rt_float(v)
+        assert tag in (0, 2)  # snappy chosen; all-literal falls back to null
+
+    def test_all_nan_or_inf(self):
+        for x in (np.nan, np.inf, -np.inf):
+            rt_float(np.full(1000, x))
+
+    def test_inf_gorilla(self):
+        v = np.cumsum(np.ones(1000) * 0.5)
+        v[500] = np.inf
+        rt_float(v)
+
+    def test_gorilla_friendly_walk(self):
+        rng = np.random.default_rng(42)
+        v = np.cumsum(rng.integers(-256, 257, 1000) / 128.0)
+        b = orc.float_encode(v)
+        assert b[0] >> 4 == 3
+        assert len(b) / len(v) < 4.5  # the 2-4 B/pt regime SURVEY §8d names
+        assert np.array_equal(orc.float_decode(b), v)
+
+    def test_gorilla_empty_and_single(self):
+        assert len(orc.gorilla_decode(orc.gorilla_encode(np.zeros(0)))) == 0
+        v = np.array([3.5])
+        assert np.array_equal(orc.gorilla_decode(orc.gorilla_encode(v)), v)
+
+    def test_gorilla_nan_rejected(self):
+        with pytest.raises(ValueError):
+            orc.gorilla_encode(np.array([np.nan, 1.0]))
+
+    def test_gorilla_known_vector(self):
+        # hand-derived from batch_float.go: first value raw BE after tag byte
+        v = np.array([1.0, 1.0, 1.0])
+        b = orc.gorilla_encode(v)
+        assert b[0] == 0x10  # floatCompressedGorilla<<4
+        assert b[1:9] == np.float64(1.0).tobytes()[::-1]  # big-endian bits
+        # two repeats: control bits 0,0 then NaN sentinel delta
+        d = orc.gorilla_decode(b)
+        assert np.array_equal(d, v)
+
+
+class TestIntCodec:
+    def test_empty(self):
+        assert orc.int_encode(np.zeros(0, dtype=np.int64)) == b""
+
+    def test_one_two(self):
+        assert rt_int([5]) == 4
+        assert rt_int([5, -7]) == 4
+
+    def test_const_delta(self):
+        assert rt_int(np.arange(0, 10000, 10)) == 1
+        assert rt_int(np.full(1000, -42)) == 1
+        assert rt_int(np.arange(0, -5000, -5)) == 1
+
+    def test_simple8b(self):
+        rng = np.random.default_rng(5)
+        assert rt_int(rng.integers(0, 1000, 1000)) == 2
+
+    def test_zstd_or_uncompressed(self):
+        rng = np.random.default_rng(6)
+        tag = rt_int(rng.integers(-(2**62), 2**62, 1000))
+        assert tag in (3, 4)
+
+    def test_zstd_compressible(self):
+        # varying deltas beyond simple8b range but repetitive bytes -> zstd
+        v = np.where(np.arange(1000) % 2 == 0, 2**61, np.int64(7)).astype(np.int64)
+        tag = rt_int(v)
+        assert tag == 3
+
+    def test_extremes(self):
+        rt_int([np.iinfo(np.int64).min, 0, np.iinfo(np.int64).max, -1, 1])
+        rt_int(np.array([2**59, -(2**59)] * 100, dtype=np.int64))
+
+    def test_zigzag(self):
+        lib = orc.get()
+        for v in (0, 1, -1, 2**62, -(2**62), np.iinfo(np.int64).min):
+            assert lib.orc_zigzag_decode(lib.orc_zigzag_encode(v)) == v
+
+
+class TestTimeCodec:
+    def test_small(self):
+        assert rt_time([5]) == 4
+        assert rt_time([5, 9]) == 4
+
+    def test_const_delta(self):
+        assert rt_time(np.arange(0, 10**12, 10**9)) == 1
+
+    def test_simple8b_scale(self):
+        rng = np.random.default_rng(7)
+        v = np.cumsum(rng.integers(1, 5, 1000)) * 10**9
+        b = orc.time_encode(v)
+        assert b[0] >> 4 == 2
+        # scale factor is a power of 10 dividing all deltas
+        scale = int.from_bytes(b[1:9], "big")
+        assert scale == 10**9
+        assert np.array_equal(orc.time_decode(b), v)
+
+    def test_irregular(self):
+        rng = np.random.default_rng(8)
+        v = np.sort(rng.integers(0, 2**62, 500))
+        rt_time(v)
+
+    def test_simple8b_mixed_selectors(self):
+        # deltas forcing several simple8b selector classes in one block
+        deltas = np.concatenate(
+            [np.ones(300), np.full(100, 2**29), np.full(50, 2**55), np.ones(240)]
+        ).astype(np.int64)
+        v = np.cumsum(deltas)
+        rt_time(v)
+
+
+class TestSegment:
+    def test_full_roundtrip(self):
+        rng = np.random.default_rng(9)
+        v = np.cumsum(rng.normal(0, 1, 1000))
+        seg = orc.encode_data_segment(orc.ORC_TYPE_FLOAT, v, None, 1000, 0)
+        assert seg[0] == 30 + 3  # BlockFloat64Full
+        dv, bm, rows, nils = orc.decode_data_segment(orc.ORC_TYPE_FLOAT, seg)
+        assert rows == 1000 and nils == 0 and np.array_equal(dv, v)
+
+    def test_mixed_nulls(self):
+        rng = np.random.default_rng(10)
+        v = np.cumsum(rng.normal(0, 1, 1000))
+        valid = rng.random(1000) > 0.3
+        bm_in = np.packbits(valid.astype(np.uint8), bitorder="little")
+        dense = v[valid]
+        seg = orc.encode_data_segment(
+            orc.ORC_TYPE_FLOAT, dense, bm_in, 1000, int((~valid).sum())
+        )
+        assert seg[0] == 3  # raw type byte
+        dv, bm, rows, nils = orc.decode_data_segment(orc.ORC_TYPE_FLOAT, seg)
+        assert rows == 1000 and nils == int((~valid).sum())
+        assert np.array_equal(dv, dense) and np.array_equal(bm, bm_in)
+
+    def test_one_row(self):
+        seg = orc.encode_data_segment(orc.ORC_TYPE_INT, np.array([7], dtype=np.int64), None, 1, 0)
+        assert seg[0] == 16 + 1  # BlockIntegerOne
+        dv, _, rows, nils = orc.decode_data_segment(orc.ORC_TYPE_INT, seg)
+        assert rows == 1 and nils == 0 and dv[0] == 7
+
+    def test_empty(self):
+        seg = orc.encode_data_segment(
+            orc.ORC_TYPE_INT, np.zeros(0, dtype=np.int64), np.zeros(2, dtype=np.uint8), 10, 10
+        )
+        assert seg[0] == 40 + 1  # BlockIntegerEmpty
+        _, _, rows, nils = orc.decode_data_segment(orc.ORC_TYPE_INT, seg)
+        assert rows == 10 and nils == 10
+
+    def test_time_segment(self):
+        t = np.arange(1000, dtype=np.int64) * 10**9
+        seg = orc.encode_time_segment(t)
+        assert seg[0] == 30 + 1  # BlockIntegerFull
+        assert np.array_equal(orc.decode_time_segment(seg), t)
+        one = orc.encode_time_segment(t[:1])
+        assert one[0] == 16 + 1 and orc.decode_time_segment(one)[0] == 0
+
+
+class TestSnappy:
+    def test_roundtrip(self):
+        lib = orc.get()
+        rng = np.random.default_rng(11)
+        for n in (0, 1, 100, 70000):
+            src = rng.integers(0, 256, n).astype(np.uint8)
+            cap = int(lib.orc_snappy_max_encoded_len(n))
+            enc = np.zeros(cap, dtype=np.uint8)
+            el = lib.orc_snappy_encode(
+                src.ctypes.data_as(orc.C.POINTER(orc.C.c_uint8)), n,
+                enc.ctypes.data_as(orc.C.POINTER(orc.C.c_uint8)), cap,
+            )
+            assert el > 0 or n == 0 and el >= 0
+            dec = np.zeros(max(n, 1), dtype=np.uint8)
+            dl = lib.orc_snappy_decode(
+                enc.ctypes.data_as(orc.C.POINTER(orc.C.c_uint8)), el,
+                dec.ctypes.data_as(orc.C.POINTER(orc.C.c_uint8)), max(n, 1),
+            )
+            assert dl == n and np.array_equal(dec[:n], src)
+
+    def test_decode_with_copies(self):
+        # hand-built stream exercising copy1/copy2 tags (RLE-style overlap)
+        lib = orc.get()
+        stream = bytes([12, 3 << 2]) + b"abcd" + bytes([(4 << 2) | 1, 4])
+        # literal "abcd" then copy len=4+1=5? tag1: len=4+((tag>>2)&7)
+        # tag = (1<<2)|1 -> len 4+1=5, offset 4 -> "abcdabcda"... build simpler:
+        stream = bytes([8, 3 << 2]) + b"abcd" + bytes([(0 << 2) | 1, 4])
+        # copy1 len=4+0=4 offset=4 => "abcdabcd", total 8
+        dec = np.zeros(8, dtype=np.uint8)
+        dl = lib.orc_snappy_decode(
+            np.frombuffer(stream, dtype=np.uint8).ctypes.data_as(orc.C.POINTER(orc.C.c_uint8)),
+            len(stream),
+            dec.ctypes.data_as(orc.C.POINTER(orc.C.c_uint8)), 8,
+        )
+        assert dl == 8 and dec.tobytes() == b"abcdabcd"
+
+
+class TestWindow:
+    def test_basic(self):
+        m = 60 * 10**9
+        assert orc.window(125 * 10**9, 0, 10**15, m) == (120 * 10**9, 180 * 10**9)
+        assert orc.window(0, 0, 100, 60) == (0, 60)
+        assert orc.window(59, 0, 100, 60) == (0, 60)
+        assert orc.window(60, 0, 100, 60) == (60, 120)
+
+    def test_negative_times(self):
+        assert orc.window(-5, -100, 100, 60) == (-60, 0)
+        assert orc.window(-60, -100, 100, 60) == (-60, 0)
+        assert orc.window(-61, -100, 100, 60) == (-120, -60)
+
+    def test_offset(self):
+        # Interval.Offset shifts the grid (select.go:584,607,654)
+        assert orc.window(125, 0, 1000, 60, 10) == (70, 130)
+        assert orc.window(130, 0, 1000, 60, 10) == (130, 190)
+
+    def test_no_interval(self):
+        assert orc.window(55, 7, 1000, 0) == (7, 1001)
