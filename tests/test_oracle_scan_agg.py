@@ -80,9 +80,8 @@ def check(rows, exp, col_type):
             assert abs(r["sum"] - e["sum"]) <= 1e-9 * max(1.0, abs(e["sum"]))
             vget = lambda f: r[f]
         else:
-            iv = r[["sum", "min", "max", "first", "last"]].view(np.int64)
-            vget = lambda f: r[f].view(np.int64)
-            assert r["sum"].view(np.int64) == e["sum"]
+            vget = lambda f: int(np.array(r[f]).view(np.int64))
+            assert vget("sum") == e["sum"]
         for f in ("min", "max", "first", "last"):
             assert vget(f) == e[f], (r, e)
             assert r[f + "_time"] == e[f + "_time"]
