@@ -1,0 +1,248 @@
+#!/usr/bin/env python3
+"""bench.py — north-star benchmark of the MI355X scan-aggregate engine.
+
+Workload (BASELINE.json configs[1], the single-GPU headline config):
+  100k series × 1k float64 points (exactly one 1000-row TSSP segment per
+  series, lib/util/util.go:72), timestamps 1s step (const-delta),
+  Gorilla-friendly quantized random-walk values (~2 B/pt; --mode random
+  gives the ~8.5 B/pt worst case), query
+  `SELECT mean(value),min(value),max(value),count(value) GROUP BY time(1m)`
+  — mean is computed as sum+count (engine/executor/schema.go:376), all six
+  aggregate families are computed in the single fused pass.
+
+A "step" = one fused decode+aggregate pass over the whole shard (resident
+in HBM) plus, for N>1, the cross-shard RCCL GROUP BY merge.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--mode walk|random]
+For N>1 launch via torch.distributed.run (one rank per GPU).
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "oracle"))
+
+WINDOW_NS = 60 * 10**9  # GROUP BY time(1m)
+SEED = 42
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def cpu_baseline_leg(mode):
+    """Oracle (CPU restatement) timed on this host's cores — a REPORTED
+    baseline (kind='port'), never the product path. Bounded sample:
+    25k series × 1k pts (~10-20 core-seconds)."""
+    import binding as orc
+
+    nser = 25_000
+    t0 = time.time()
+    blob, descs = orc.gen_shard(SEED, nser, 1000, mode=mode)
+    gen_s = time.time() - t0
+    try:
+        import multiprocessing
+
+        cores = multiprocessing.cpu_count()
+    except Exception:
+        cores = 1
+    t0 = time.time()
+    rows = orc.scan_agg(blob, descs, orc.ORC_TYPE_FLOAT, 0, 2**62, WINDOW_NS,
+                        nthreads=cores)
+    wall = time.time() - t0
+    pts = nser * 1000
+    assert int(rows["count"].sum()) == pts
+    return {
+        "value": pts / wall,
+        "unit": "points/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{nser} series x 1k pts ({mode}), oracle scan_agg, "
+                  f"{wall:.2f}s wall on {cores} threads (gen {gen_s:.2f}s)",
+    }
+
+
+def read_traffic_file():
+    """PMC-measured HBM bytes per scan launch, if a committed measurement
+    exists (profiles/hbm_traffic.json, written from rocprofv3 --pmc runs —
+    see profiles/README)."""
+    p = os.path.join(REPO, "profiles", "hbm_traffic.json")
+    if os.path.exists(p):
+        try:
+            with open(p) as f:
+                return json.load(f)
+        except Exception:
+            return None
+    return None
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--mode", choices=["walk", "random"], default="walk")
+    ap.add_argument("--series", type=int, default=100_000)
+    ap.add_argument("--pts", type=int, default=1000)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import binding as orc  # oracle: data authoring + cpu_baseline only
+    import opengemini_amd as gx
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = world if world > 1 else args.gpus
+    if world == 1 and args.gpus > 1:
+        # driver contract: N>1 arrives via torch.distributed.run; a direct
+        # call with --gpus N>1 runs N independent shards on this one process?
+        # No — require torchrun so each rank owns one GPU.
+        log("WARNING: --gpus>1 without torchrun; running single-rank N=1")
+        n_gpus = 1
+
+    dist_on = world > 1
+    if dist_on:
+        import torch
+        import torch.distributed as dist
+
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+
+    gen_mode = orc.GEN_FLOAT_WALK if args.mode == "walk" else orc.GEN_FLOAT_RANDOM
+
+    # each rank authors its own shard: same shape, disjoint series
+    t0 = time.time()
+    blob, descs = orc.gen_shard(SEED + rank * 1_000_003, args.series, args.pts,
+                                mode=gen_mode)
+    log(f"[gen] {args.series}x{args.pts} pts in {time.time()-t0:.1f}s, "
+        f"{len(blob)/1e6:.0f} MB ({len(blob)/(args.series*args.pts):.2f} B/pt)")
+
+    t0 = time.time()
+    shard = gx.Shard(blob, descs, gx.engine.GEMX_TYPE_FLOAT, device=local_rank)
+    log(f"[attach] H2D resident in {time.time()-t0:.1f}s")
+
+    n_wins = (args.pts + 59) // 60 + 1
+    w0 = 0  # t0=0, windows start at ordinal 0
+
+    def step():
+        rows, stats = shard.scan_agg(0, 2**62, WINDOW_NS)
+        if dist_on:
+            from opengemini_amd.dist import window_partials, merge_across_shards
+
+            p = window_partials(rows, WINDOW_NS, 0, w0, n_wins)
+            merge_across_shards(p, device=f"cuda:{local_rank}")
+        return rows, stats
+
+    # warmup
+    for _ in range(args.warmup):
+        rows, stats = step()
+
+    # verification outside the timed region: count checksum
+    assert int(rows["count"].sum()) == args.series * args.pts
+
+    if dist_on:
+        import torch
+        import torch.distributed as dist
+
+        dist.barrier()
+        torch.cuda.synchronize()
+    t_start = time.time()
+    decode_ms_acc = 0.0
+    total_ms_acc = 0.0
+    for _ in range(args.steps):
+        rows, stats = step()
+        decode_ms_acc += stats["decode_ms"]
+        total_ms_acc += stats["total_ms"]
+    if dist_on:
+        import torch
+        import torch.distributed as dist
+
+        torch.cuda.synchronize()
+        dist.barrier()
+    elapsed = time.time() - t_start
+
+    if dist_on:
+        import torch
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64, device=f"cuda:{local_rank}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    pts_per_step = args.series * args.pts
+    total_pts = pts_per_step * n_gpus * args.steps
+    value = total_pts / elapsed
+    ms_per_step = elapsed / args.steps * 1000
+
+    if rank != 0:
+        return
+
+    # roofline: dominant kernel = fused decode+reduce (k_scan_fast).
+    # achieved = ALGORITHMIC bytes per launch (compressed data+time segment
+    # bytes — SURVEY.md §8d: B_alg/pt = C_ts + C_val; outputs negligible)
+    # ÷ average kernel duration from HIP events around the launch (the
+    # C-ABI records events on its own stream).
+    alg_bytes = shard.compressed_bytes
+    avg_decode_s = (decode_ms_acc / args.steps) / 1000.0
+    achieved = alg_bytes / avg_decode_s if avg_decode_s > 0 else 0.0
+    peak = 8.0e12  # MI355X HBM3E peak (MI355X_MICROARCH.md)
+    traffic = None
+    tf = read_traffic_file()
+    if tf and tf.get("mode") == args.mode and tf.get("series") == args.series:
+        traffic = tf.get("bytes_per_launch")
+
+    cpu_baseline = None
+    if not args.skip_cpu_baseline:
+        cpu_baseline = cpu_baseline_leg(gen_mode)
+
+    out = {
+        "metric": "points aggregated/sec",
+        "value": value,
+        "unit": "points/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": ms_per_step,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # no published number
+        "dtype": "f64",
+        "data": "synthetic",
+        "config": {
+            "workload": "100k series x 1k pts float64 Gorilla, mean/min/max/"
+                        "count GROUP BY time(1m), single TSSP file "
+                        f"(mode={args.mode}, {args.series}x{args.pts})",
+            "window": "1m",
+            "series_per_gpu": args.series,
+            "points_per_series": args.pts,
+            "parallelism": f"shard-per-gpu dp{n_gpus}",
+        },
+        "roofline": {
+            "bound": "hbm",
+            "achieved": achieved,
+            "peak": peak,
+            "unit": "GB/s",
+            "frac": achieved / peak,
+            "traffic": traffic,
+        },
+        "cpu_baseline": cpu_baseline,
+    }
+    # roofline units: report in GB/s for readability
+    out["roofline"]["achieved"] = achieved / 1e9
+    out["roofline"]["peak"] = peak / 1e9
+    if traffic is not None:
+        out["roofline"]["traffic"] = traffic
+    print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,137 @@
+/*
+ * gemx.h — C-ABI of the MI355X-native TSSP scan-and-aggregate engine
+ * (libgemx.so, built from opengemini_amd/csrc for gfx950 only).
+ *
+ * Drop-in boundary: these entry points are what a Go-side cgo cursor
+ * (`gpuAggregateCursor` implementing comm.KeyCursor,
+ *  /root/reference/engine/comm/cursor.go:46-56) would bind, substituted at
+ * the construction seam /root/reference/engine/iterators.go:919-929 where
+ * NewAggregateCursor is chosen. See INTEGRATION.md for the cgo stub.
+ *
+ * Interface mapping (reference → this ABI):
+ *   TSSPFile.ReadAt + readSegmentRecord      → shard blob + gemx_seg_desc[]
+ *     (engine/immutable/tssp_reader.go:586, tssp_file.go:369)
+ *   shard.CreateCursor / NewAggregateCursor  → gemx_scan_agg
+ *     (engine/iterators.go:130,921; engine/aggregate_cursor.go:90)
+ *   KeyCursor.NextAggData record pump        → gemx_agg_row retrieval
+ *     (engine/comm/cursor.go:51; host layer re-chunks into records)
+ *
+ * No torch types; plain pointers and sizes. All functions return 0 on
+ * success or a negative GEMX_E_* code; gemx_last_error() gives a message.
+ * The engine REQUIRES a GPU: there is no CPU fallback of any kind.
+ */
+#ifndef GEMX_H
+#define GEMX_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define GEMX_ABI_VERSION 1
+
+/* column data types; mirrors influx.Field_Type_* (lib/util/lifted/vm/
+ * protoparser/influx/parser.go:1363-1370) */
+#define GEMX_TYPE_INT 1
+#define GEMX_TYPE_FLOAT 3
+
+/* error codes */
+#define GEMX_OK 0
+#define GEMX_E_NOGPU -1        /* no HIP device — the engine never falls back */
+#define GEMX_E_INVALID -2      /* bad arguments / malformed descriptors */
+#define GEMX_E_HIP -3          /* HIP runtime failure */
+#define GEMX_E_UNSUPPORTED -4  /* codec not yet on-device (zstd, MLF) */
+#define GEMX_E_DECODE -5       /* corrupt segment detected on device */
+#define GEMX_E_CAP -6          /* output capacity too small */
+
+/* One segment descriptor: a ColumnMeta entry + its time-column twin
+ * (engine/immutable/tssp_file_meta.go:60 Segment{offset,size},
+ *  :145 ColumnMeta, :377 ChunkMeta). Offsets index the shard blob.
+ * Descriptors MUST be grouped by sid and time-ascending within sid
+ * (the order Location iteration produces, engine/immutable/location.go:261).
+ * Layout identical to the oracle's orc_seg_desc so harnesses can share
+ * numpy dtypes; the engine never links the oracle. */
+typedef struct {
+  uint64_t sid;
+  uint64_t data_offset;
+  uint32_t data_size;
+  uint32_t rows;
+  uint64_t time_offset;
+  uint32_t time_size;
+  uint32_t _pad;
+  int64_t min_time;
+  int64_t max_time;
+} gemx_seg_desc;
+
+typedef union {
+  double f;
+  int64_t i;
+} gemx_val;
+
+/* One (sid, window) aggregate row. Semantics documented field-by-field in
+ * oracle/oracle.h (orc_agg_row); layouts are kept identical so parity tests
+ * can compare buffers directly. */
+typedef struct {
+  uint64_t sid;
+  int64_t win_start;
+  int64_t first_row_time;
+  int64_t count;
+  int64_t count_time;
+  gemx_val sum;
+  int64_t sum_time;
+  gemx_val minv;
+  int64_t min_time;
+  gemx_val maxv;
+  int64_t max_time;
+  gemx_val firstv;
+  int64_t first_time;
+  gemx_val lastv;
+  int64_t last_time;
+  uint8_t min_isnil, max_isnil, first_isnil, last_isnil, sum_isnil;
+  uint8_t _pad[3];
+} gemx_agg_row;
+
+typedef struct {
+  double h2d_ms;        /* blob upload (attach time, not per query) */
+  double decode_ms;     /* fused decode+reduce kernel */
+  double merge_ms;      /* per-(sid,window) partial merge kernel */
+  double total_ms;      /* device wall for the query (events) */
+  uint64_t points;      /* rows scanned */
+  uint64_t compressed_bytes; /* data+time segment bytes (algorithmic reads) */
+  uint64_t n_rows;      /* output rows */
+} gemx_query_stats;
+
+typedef struct gemx_shard gemx_shard;
+
+/* library info */
+int gemx_abi_version(void);
+const char *gemx_last_error(void);
+
+/* device management: returns number of HIP devices (0 ⇒ engine unusable) */
+int gemx_device_count(void);
+
+/* Attach a shard: uploads the segment blob and descriptors to device HBM
+ * (resident thereafter — 288 GB HBM3E per MI355X holds full shards) and
+ * precomputes per-segment window spans + per-series output ranges for the
+ * given grouping. Replaces the readcache + Location list of
+ * engine/immutable/tssp_reader.go / location.go for this path. */
+int gemx_shard_attach(int device, const void *blob, uint64_t blob_bytes,
+                      const gemx_seg_desc *descs, uint64_t nsegs, int col_type,
+                      gemx_shard **out);
+int gemx_shard_close(gemx_shard *);
+
+/* Scan + GROUP BY time aggregate over the whole shard. Computes all six
+ * kernel families (count/sum/min/max/first/last — mean is sum+count per
+ * engine/executor/schema.go:376-388) in one fused pass.
+ * interval==0 ⇒ single window [start_time, end_time+1).
+ * Rows are returned grouped by sid (descriptor order), windows ascending.
+ * stats may be NULL. */
+int gemx_scan_agg(gemx_shard *, int64_t start_time, int64_t end_time,
+                  int64_t interval, int64_t offset, gemx_agg_row *out_host,
+                  uint64_t cap, uint64_t *n_out, gemx_query_stats *stats);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* GEMX_H */

@@ -1,0 +1,1559 @@
+/*
+ * gemx_engine.hip — MI355X-native (gfx950/CDNA4) TSSP scan-and-aggregate
+ * engine: fused segment decode + GROUP BY time reduction, written from
+ * scratch for CDNA4 (64-wide waves, one lane per compressed segment,
+ * HBM-bandwidth-bound — no MFMA on this path by design).
+ *
+ * Replaces, for this hot path, the reference's
+ *   engine/immutable column reader  (tssp_reader.go:586 ReadAt,
+ *                                    tssp_file.go:369 readSegmentRecord)
+ *   lib/encoding + lib/compress decoders (timestamp.go, int.go, float.go,
+ *                                    compress.go, tsm1/batch_float.go:278)
+ *   engine/aggregate_cursor + series_agg_reducer.gen window reduction
+ * behind the C-ABI in include/gemx.h. Semantics are specified by the CPU
+ * oracle (oracle/) which restates the reference line-by-line; this file
+ * cites the same reference lines where behaviour is subtle.
+ *
+ * Execution model:
+ *   k_scan_fast    one lane per nil-free segment, fully fused streaming
+ *                  decode+reduce, per-(segment,window) partials to HBM.
+ *   k_scan_general remaining segments (nil bitmaps, snappy, empty blocks):
+ *                  decode to per-lane HBM scratch, then reduce with the
+ *                  bug-compatible index semantics random access needs.
+ *   k_merge        one lane per (sid,window) output row: in-time-order merge
+ *                  of the contributing segments' partials with the exact
+ *                  fv() merge semantics (series_agg_func.gen.go:44-274).
+ * Determinism: float sums are serial within a segment and merged in segment
+ * time order — independent of launch geometry.
+ */
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+#include <stdio.h>
+#include <string.h>
+#include <stdlib.h>
+#include <math.h>
+#include <vector>
+#include <string>
+#include <algorithm>
+
+#include "../../include/gemx.h"
+
+#define UVNAN 0x7FF8000000000001ULL /* tsm1/float.go:17 */
+#define INFLUX_MIN_TIME (INT64_MIN + 2)
+#define INFLUX_MAX_TIME (INT64_MAX - 1)
+
+/* ---------------- device: common helpers ---------------- */
+
+__device__ __forceinline__ uint32_t d_u32be(const uint8_t *p) {
+  return ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) | ((uint32_t)p[2] << 8) |
+         p[3];
+}
+__device__ __forceinline__ uint64_t d_u64be(const uint8_t *p) {
+  uint64_t v = 0;
+#pragma unroll
+  for (int i = 0; i < 8; i++) v = (v << 8) | p[i];
+  return v;
+}
+__device__ __forceinline__ uint64_t d_u64le(const uint8_t *p) {
+  uint64_t v;
+  memcpy(&v, p, 8); /* device unaligned load is legal on CDNA */
+  return v;
+}
+__device__ __forceinline__ double d_f64le(const uint8_t *p) {
+  double d;
+  memcpy(&d, p, 8);
+  return d;
+}
+__device__ __forceinline__ int64_t d_zigzag_dec(uint64_t u) {
+  return (int64_t)((u >> 1) ^ (uint64_t)((int64_t)((u & 1) << 63) >> 63));
+}
+__device__ __forceinline__ int d_uvarint(const uint8_t *p, int64_t len, uint64_t *out) {
+  uint64_t v = 0;
+  int s = 0;
+  for (int i = 0; i < len && i < 10; i++) {
+    uint8_t b = p[i];
+    if (b < 0x80) {
+      *out = v | ((uint64_t)b << s);
+      return i + 1;
+    }
+    v |= (uint64_t)(b & 0x7f) << s;
+    s += 7;
+  }
+  return 0;
+}
+
+/* window ordinal + start (select.go:579-656, no timezone) */
+__device__ __host__ __forceinline__ int64_t win_floordiv(int64_t a, int64_t b) {
+  int64_t q = a / b;
+  if ((a % b) != 0 && ((a < 0) != (b < 0))) q--;
+  return q;
+}
+__device__ __host__ __forceinline__ int64_t win_ordinal(int64_t t, int64_t interval,
+                                                        int64_t offset) {
+  return win_floordiv(t - offset, interval);
+}
+__device__ __host__ __forceinline__ int64_t win_start_of(int64_t ord, int64_t interval,
+                                                         int64_t offset) {
+  int64_t s = ord * interval + offset;
+  return s; /* MinTime clamp (select.go:601-605) is unreachable for the sane
+               time ranges descriptors carry; host validates */
+}
+
+/* ---------------- device: bit reader (batch_float.go bit cursor) --------- */
+
+struct BitR {
+  const uint8_t *b;
+  int64_t len, pos;
+  uint64_t cur;
+  int nbits;
+  __device__ int refill() {
+    if (len - pos >= 8) {
+      cur = d_u64be(b + pos);
+      nbits = 64;
+      pos += 8;
+      return 0;
+    }
+    int64_t rem = len - pos;
+    if (rem <= 0) return -1;
+    uint64_t v = 0;
+    for (int64_t i = 0; i < rem; i++) v = (v << 8) | b[pos + i];
+    cur = v << (64 - rem * 8);
+    nbits = (int)(rem * 8);
+    pos = len;
+    return 0;
+  }
+  __device__ int read(int n, uint64_t *out) {
+    uint64_t v = 0;
+    int got = 0;
+    while (got < n) {
+      if (nbits == 0 && refill()) return -1;
+      int take = n - got;
+      if (take > nbits) take = nbits;
+      uint64_t chunk = cur >> (64 - take);
+      v = (take == 64) ? chunk : ((v << take) | chunk);
+      cur <<= take;
+      nbits -= take;
+      got += take;
+    }
+    *out = v;
+    return 0;
+  }
+};
+
+/* ---------------- device: value iterators ---------------- */
+
+/* float adaptive (lib/compress/float.go:139-161) streaming iterator,
+ * scratch-free paths: null / same / RLE / gorilla */
+struct FloatIter {
+  int kind; /* 0 null,3 gorilla,4 same,5 rle */
+  /* null */
+  const uint8_t *raw;
+  int64_t raw_n, raw_i;
+  /* same */
+  double same_v;
+  int64_t same_n, same_i;
+  /* rle */
+  const uint8_t *rle_p;
+  int64_t rle_len, rle_pos;
+  int64_t run_left;
+  double run_v;
+  /* gorilla */
+  BitR br;
+  uint64_t g_val;
+  uint8_t g_trail, g_mean;
+  int g_first, g_done;
+
+  __device__ int init(const uint8_t *enc, int64_t len) {
+    if (len < 1) return -1;
+    kind = enc[0] >> 4;
+    const uint8_t *in = enc + 1;
+    int64_t inlen = len - 1;
+    switch (kind) {
+    case 0:
+      raw = in;
+      raw_n = inlen / 8;
+      raw_i = 0;
+      return 0;
+    case 4: /* same (compress.go:51-66) */
+      if (inlen < 2) return -1;
+      same_n = ((int64_t)in[0] << 8) | in[1];
+      same_v = (inlen == 2) ? 0.0 : d_f64le(in + 2);
+      same_i = 0;
+      return 0;
+    case 5: /* RLE (compress.go:95-121) */
+      rle_p = in;
+      rle_len = inlen;
+      rle_pos = 0;
+      run_left = 0;
+      return 0;
+    case 3: /* gorilla (batch_float.go:278-514); stream has tsm1 tag byte */
+      if (inlen < 9) { g_done = 1; g_first = 0; return 0; }
+      g_val = d_u64be(in + 1);
+      g_done = (g_val == UVNAN);
+      g_first = 1;
+      g_trail = 0;
+      g_mean = 64;
+      br.b = in + 9;
+      br.len = inlen - 9;
+      br.pos = 0;
+      br.nbits = 0;
+      return 0;
+    default:
+      return -2; /* snappy handled by the general kernel; MLF unsupported */
+    }
+  }
+
+  __device__ int next(double *out) { /* 0 ok, -1 exhausted/corrupt */
+    switch (kind) {
+    case 0:
+      if (raw_i >= raw_n) return -1;
+      *out = d_f64le(raw + raw_i * 8);
+      raw_i++;
+      return 0;
+    case 4:
+      if (same_i >= same_n) return -1;
+      *out = same_v;
+      same_i++;
+      return 0;
+    case 5:
+      while (run_left == 0) {
+        if (rle_len - rle_pos < 2) return -1;
+        uint16_t m = (uint16_t)(((uint16_t)rle_p[rle_pos] << 8) | rle_p[rle_pos + 1]);
+        if (m >> 15) {
+          run_left = m & 0x7FFF;
+          run_v = 0.0;
+          rle_pos += 2;
+        } else {
+          if (rle_len - rle_pos < 10) return -1;
+          run_left = m;
+          run_v = d_f64le(rle_p + rle_pos + 2);
+          rle_pos += 10;
+        }
+      }
+      *out = run_v;
+      run_left--;
+      return 0;
+    case 3: {
+      if (g_first) {
+        if (g_done) return -1;
+        g_first = 0;
+        uint64_t u = g_val;
+        memcpy(out, &u, 8);
+        return 0;
+      }
+      if (g_done) return -1;
+      uint64_t bit;
+      if (br.read(1, &bit)) return -1;
+      if (bit) {
+        if (br.read(1, &bit)) return -1;
+        if (bit) {
+          uint64_t lm;
+          if (br.read(11, &lm)) return -1;
+          uint8_t lead = (uint8_t)((lm >> 6) & 0x1F);
+          g_mean = (uint8_t)(lm & 0x3F);
+          if (g_mean > 0) {
+            g_trail = (uint8_t)(64 - lead - g_mean);
+          } else {
+            g_trail = 0;
+            g_mean = 64;
+          }
+        }
+        uint64_t sbits;
+        if (br.read(g_mean, &sbits)) return -1;
+        g_val ^= sbits << (g_trail & 0x3F);
+        if (g_val == UVNAN) {
+          g_done = 1;
+          return -1;
+        }
+      }
+      uint64_t u = g_val;
+      memcpy(out, &u, 8);
+      return 0;
+    }
+    }
+    return -1;
+  }
+};
+
+/* int64 block iterator (lib/encoding/int.go:370-384), scratch-free paths:
+ * const-delta / simple8b / uncompressed (zstd → general/unsupported) */
+struct S8b {
+  uint64_t vals[0]; /* not used; decode word inline */
+};
+
+__device__ __forceinline__ void s8b_selinfo(int sel, int *n, int *bits) {
+  /* simple8b selector table (lib/util/lifted/encoding/simple8b/encoding.go:193) */
+  const int ns[16] = {240, 120, 60, 30, 20, 15, 12, 10, 8, 7, 6, 5, 4, 3, 2, 1};
+  const int bs[16] = {0, 0, 1, 2, 3, 4, 5, 6, 7, 8, 10, 12, 15, 20, 30, 60};
+  *n = ns[sel];
+  *bits = bs[sel];
+}
+
+struct IntIter {
+  int kind; /* 1 const,2 s8b,4 raw */
+  int64_t cur, delta;
+  int64_t left; /* const-delta values remaining (incl. current) */
+  /* s8b */
+  const uint8_t *words;
+  int64_t nwords, widx;
+  uint64_t w;
+  int w_n, w_bits, w_i;
+  int64_t remaining; /* srcCount remaining */
+  int first_pending;
+  /* raw */
+  const uint8_t *raw;
+  int64_t raw_n, raw_i;
+
+  __device__ int init(const uint8_t *enc, int64_t len) {
+    if (len < 5) return -1;
+    kind = enc[0] >> 4;
+    const uint8_t *in = enc + 1;
+    int64_t inlen = len - 1;
+    switch (kind) {
+    case 1: { /* const delta (int.go:214-254) */
+      if (inlen < 8) return -1;
+      cur = d_zigzag_dec(d_u64be(in));
+      in += 8;
+      inlen -= 8;
+      uint64_t zd, cnt;
+      int k = d_uvarint(in, inlen, &zd);
+      if (k <= 0) return -1;
+      in += k;
+      inlen -= k;
+      k = d_uvarint(in, inlen, &cnt);
+      if (k <= 0) return -1;
+      delta = d_zigzag_dec(zd);
+      left = (int64_t)cnt + 1;
+      return 0;
+    }
+    case 2: { /* simple8b (int.go:256-301) */
+      if (inlen < 16) return -1;
+      int64_t enc_count = (int64_t)d_u32be(in);
+      remaining = (int64_t)d_u32be(in + 4);
+      in += 8;
+      if (enc_count < 1) return -1;
+      cur = d_zigzag_dec(d_u64be(in));
+      words = in + 8;
+      nwords = enc_count - 1;
+      widx = 0;
+      w_i = 0;
+      w_n = 0;
+      first_pending = 1;
+      return 0;
+    }
+    case 4: { /* uncompressed (int.go:316-324): zigzag u64 BE each */
+      if (inlen < 4) return -1;
+      raw_n = (int64_t)d_u32be(in) / 8;
+      raw = in + 4;
+      raw_i = 0;
+      return 0;
+    }
+    default:
+      return -2; /* zstd: not on device yet */
+    }
+  }
+
+  __device__ int next(int64_t *out) {
+    switch (kind) {
+    case 1:
+      if (left <= 0) return -1;
+      *out = cur;
+      cur += delta;
+      left--;
+      return 0;
+    case 2:
+      if (remaining <= 0) return -1;
+      if (first_pending) {
+        first_pending = 0;
+        *out = cur;
+        remaining--;
+        return 0;
+      }
+      while (w_i >= w_n) {
+        if (widx >= nwords) return -1;
+        w = d_u64be(words + widx * 8);
+        widx++;
+        s8b_selinfo((int)(w >> 60), &w_n, &w_bits);
+        w_i = 0;
+      }
+      {
+        uint64_t v;
+        if (w_bits == 0)
+          v = 1;
+        else {
+          uint64_t mask = (w_bits == 60) ? ((1ULL << 60) - 1) : ((1ULL << w_bits) - 1);
+          v = (w >> (w_i * w_bits)) & mask;
+        }
+        w_i++;
+        cur = cur + d_zigzag_dec(v);
+        *out = cur;
+        remaining--;
+        return 0;
+      }
+    case 4:
+      if (raw_i >= raw_n) return -1;
+      *out = d_zigzag_dec(d_u64be(raw + raw_i * 8));
+      raw_i++;
+      return 0;
+    }
+    return -1;
+  }
+};
+
+/* timestamp iterator (lib/encoding/timestamp.go:310-324), scratch-free:
+ * const-delta / simple8b×scale / uncompressed (snappy → general) */
+struct TimeIter {
+  int kind;
+  int64_t cur, delta, left;
+  uint64_t scale;
+  const uint8_t *words;
+  int64_t nwords, widx;
+  uint64_t w;
+  int w_n, w_bits, w_i;
+  int64_t remaining;
+  int first_pending;
+  const uint8_t *raw;
+  int64_t raw_n, raw_i;
+  int raw_le; /* snappy-decoded scratch: raw little-endian (timestamp.go:274) */
+
+  __device__ int init(const uint8_t *enc, int64_t len) {
+    if (len < 5) return -1;
+    kind = enc[0] >> 4;
+    const uint8_t *in = enc + 1;
+    int64_t inlen = len - 1;
+    raw_le = 0;
+    switch (kind) {
+    case 1: { /* const delta (timestamp.go:190-225): first RAW u64 */
+      if (inlen < 8) return -1;
+      cur = (int64_t)d_u64be(in);
+      in += 8;
+      inlen -= 8;
+      uint64_t d, cnt;
+      int k = d_uvarint(in, inlen, &d);
+      if (k <= 0) return -1;
+      in += k;
+      inlen -= k;
+      k = d_uvarint(in, inlen, &cnt);
+      if (k <= 0) return -1;
+      delta = (int64_t)d;
+      left = (int64_t)cnt + 1;
+      return 0;
+    }
+    case 2: { /* simple8b × scale (timestamp.go:227-272) */
+      if (inlen < 24) return -1;
+      scale = d_u64be(in);
+      int64_t enc_count = (int64_t)d_u32be(in + 8);
+      remaining = (int64_t)d_u32be(in + 12);
+      in += 16;
+      if (enc_count < 1) return -1;
+      cur = (int64_t)d_u64be(in);
+      words = in + 8;
+      nwords = enc_count - 1;
+      widx = 0;
+      w_i = 0;
+      w_n = 0;
+      first_pending = 1;
+      return 0;
+    }
+    case 4: { /* uncompressed: zigzag u64be (timestamp.go:299-308) */
+      if (inlen < 4) return -1;
+      raw_n = (int64_t)d_u32be(in) / 8;
+      raw = in + 4;
+      raw_i = 0;
+      return 0;
+    }
+    default:
+      return -2; /* snappy handled via scratch in the general kernel */
+    }
+  }
+
+  /* random access for const-delta (the regular-timestamps fast path) */
+  __device__ __forceinline__ bool is_const() const { return kind == 1; }
+  __device__ __forceinline__ int64_t at_const(int64_t i) const {
+    return cur + delta * i; /* only valid before first next() */
+  }
+
+  __device__ int next(int64_t *out) {
+    switch (kind) {
+    case 1:
+      if (left <= 0) return -1;
+      *out = cur;
+      cur += delta;
+      left--;
+      return 0;
+    case 2:
+      if (remaining <= 0) return -1;
+      if (first_pending) {
+        first_pending = 0;
+        *out = cur;
+        remaining--;
+        return 0;
+      }
+      while (w_i >= w_n) {
+        if (widx >= nwords) return -1;
+        w = d_u64be(words + widx * 8);
+        widx++;
+        s8b_selinfo((int)(w >> 60), &w_n, &w_bits);
+        w_i = 0;
+      }
+      {
+        uint64_t v;
+        if (w_bits == 0)
+          v = 1;
+        else {
+          uint64_t mask = (w_bits == 60) ? ((1ULL << 60) - 1) : ((1ULL << w_bits) - 1);
+          v = (w >> (w_i * w_bits)) & mask;
+        }
+        w_i++;
+        cur = cur + (int64_t)(v * scale);
+        *out = cur;
+        remaining--;
+        return 0;
+      }
+    case 4:
+      if (raw_i >= raw_n) return -1;
+      if (raw_le)
+        *out = (int64_t)d_u64le(raw + raw_i * 8);
+      else
+        *out = d_zigzag_dec(d_u64be(raw + raw_i * 8));
+      raw_i++;
+      return 0;
+    }
+    return -1;
+  }
+};
+
+/* ---------------- device: segment header parse ---------------- */
+
+struct SegHeader {
+  int rows, nilcount;
+  const uint8_t *bitmap; /* NULL ⇒ all valid */
+  int64_t bm_off;
+  const uint8_t *enc;
+  int64_t enc_len;
+  int one_value; /* BlockOne: enc points at raw value bytes */
+};
+
+/* data segment (column_builder.go:446-487 + reader.go:674-717) */
+__device__ int parse_data_header(const uint8_t *seg, int64_t len, int col_type,
+                                 SegHeader *h) {
+  if (len < 1) return -1;
+  uint8_t typ = seg[0];
+  h->bitmap = nullptr;
+  h->bm_off = 0;
+  h->one_value = 0;
+  if (typ > 16 && typ < 21) { /* BlockOne */
+    h->rows = 1;
+    h->one_value = 1;
+    h->enc = seg + 1;
+    h->enc_len = len - 1;
+    h->nilcount = (len - 1 == 0) ? 1 : 0;
+    return 0;
+  }
+  if (typ >= 30 && typ < 35) { /* BlockFull */
+    if (len < 5) return -1;
+    h->rows = (int)d_u32be(seg + 1);
+    h->nilcount = 0;
+    h->enc = seg + 5;
+    h->enc_len = len - 5;
+    return 0;
+  }
+  if (typ >= 40 && typ < 45) { /* BlockEmpty */
+    if (len < 5) return -1;
+    h->rows = (int)d_u32be(seg + 1);
+    h->nilcount = h->rows;
+    h->enc = seg + 5;
+    h->enc_len = 0;
+    return 0;
+  }
+  if (typ != (uint8_t)col_type) return -1;
+  if (len < 13) return -1;
+  int64_t bmlen = (int64_t)d_u32be(seg + 1);
+  if (len < 13 + bmlen) return -1;
+  h->bitmap = seg + 5;
+  h->bm_off = (int64_t)d_u32be(seg + 5 + bmlen);
+  h->nilcount = (int)d_u32be(seg + 9 + bmlen);
+  h->enc = seg + 13 + bmlen;
+  h->enc_len = len - 13 - bmlen;
+  h->rows = -1; /* dense + nilcount, resolved by caller */
+  return 0;
+}
+
+__device__ __forceinline__ int bm_valid(const SegHeader *h, int i) {
+  if (!h->bitmap) return 1;
+  int64_t s = h->bm_off + i;
+  return (h->bitmap[s >> 3] >> (s & 7)) & 1;
+}
+
+/* ---------------- partials ---------------- */
+
+/* per-(segment,window) partial; op order: count,sum,min,max,first,last */
+struct Partial {
+  gemx_val v[6];
+  int64_t t[6];
+  int64_t first_row_time;
+  uint32_t nilmask;
+  uint32_t has_rows;
+};
+
+/* per-segment query-time metadata (host computed per query) */
+struct SegQ {
+  int64_t w_first;       /* window ordinal of min_time */
+  uint64_t partial_base; /* slot base */
+  uint32_t n_wins;
+  uint32_t series_idx;
+};
+
+/* per-series output metadata */
+struct SeriesQ {
+  uint64_t sid;
+  int64_t w_min;
+  uint64_t out_base;
+  uint32_t n_wins;
+  uint32_t seg_start, seg_count; /* into desc/segq arrays */
+  uint32_t _pad;
+};
+
+/* ---------------- fused scan kernels ---------------- */
+
+struct DevErr {
+  int code; /* first error wins */
+};
+
+__device__ __forceinline__ void set_err(DevErr *e, int code) {
+  atomicCAS(&e->code, 0, code);
+}
+
+/* accumulate one (t, value, valid) row stream into per-window partials.
+ * Shared by fast path (no nils) with streaming times/values. */
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_scan_fast(
+    const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
+    const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
+    uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
+    int64_t offset, int64_t q_start, int64_t q_end, DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (uint32_t li = gid; li < nseg_ids; li += gridDim.x * blockDim.x) {
+    uint32_t si = seg_ids[li];
+    const gemx_seg_desc d = descs[si];
+    const SegQ sq = segq[si];
+
+    /* time segment: [BlockIntegerFull][rows u32][Time enc] or
+     * [BlockIntegerOne][8B raw LE] (chunkdata_builder.go:91-95) */
+    TimeIter ti;
+    {
+      const uint8_t *tseg = blob + d.time_offset;
+      if (tseg[0] == 17) { /* BlockIntegerOne */
+        ti.kind = 1;
+        ti.cur = (int64_t)d_u64le(tseg + 1);
+        ti.delta = 0;
+        ti.left = 1;
+      } else if (tseg[0] == 31 && d.time_size > 5) {
+        if (ti.init(tseg + 5, d.time_size - 5)) {
+          set_err(err, GEMX_E_DECODE);
+          return;
+        }
+      } else {
+        set_err(err, GEMX_E_DECODE);
+        return;
+      }
+    }
+    SegHeader h;
+    if (parse_data_header(blob + d.data_offset, d.data_size, COLTYPE, &h)) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    int rows = (int)d.rows;
+
+    FloatIter fit;
+    IntIter iit;
+    int vrc;
+    if (h.one_value) {
+      /* single raw value; synthesize below */
+      vrc = 0;
+    } else if (COLTYPE == GEMX_TYPE_FLOAT) {
+      vrc = fit.init(h.enc, h.enc_len);
+    } else {
+      vrc = iit.init(h.enc, h.enc_len);
+    }
+    if (vrc) {
+      set_err(err, vrc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE);
+      return;
+    }
+
+    /* running group state (one window at a time; rows time-ascending) */
+    int64_t cur_ord = INT64_MIN;
+    int64_t cnt = 0;
+    double sumf = 0;
+    int64_t sumi = 0;
+    gemx_val minv = {0}, maxv = {0}, firstv = {0}, lastv = {0};
+    int64_t min_t = 0, max_t = 0, first_t = 0, last_t = 0, grp_start_t = 0;
+    Partial *base = partials + sq.partial_base;
+
+    for (int i = 0; i < rows; i++) {
+      int64_t t;
+      if (ti.next(&t)) { set_err(err, GEMX_E_DECODE); return; }
+      double fv = 0;
+      int64_t iv = 0;
+      if (h.one_value) {
+        if (COLTYPE == GEMX_TYPE_FLOAT)
+          fv = d_f64le(h.enc);
+        else
+          memcpy(&iv, h.enc, 8);
+      } else if (COLTYPE == GEMX_TYPE_FLOAT) {
+        if (fit.next(&fv)) { set_err(err, GEMX_E_DECODE); return; }
+      } else {
+        if (iit.next(&iv)) { set_err(err, GEMX_E_DECODE); return; }
+      }
+
+      int64_t ord = interval ? win_ordinal(t, interval, offset) : 0;
+      if (ord != cur_ord) {
+        if (cur_ord != INT64_MIN) {
+          /* flush group */
+          int64_t slot = cur_ord - sq.w_first;
+          Partial *p = base + slot;
+          p->v[0].i = cnt;
+          if (COLTYPE == GEMX_TYPE_FLOAT) p->v[1].f = sumf; else p->v[1].i = sumi;
+          p->v[2] = minv;
+          p->v[3] = maxv;
+          p->v[4] = firstv;
+          p->v[5] = lastv;
+          p->t[0] = grp_start_t;
+          p->t[1] = grp_start_t; /* no nils ⇒ valueIndex == row index */
+          p->t[2] = min_t;
+          p->t[3] = max_t;
+          p->t[4] = first_t;
+          p->t[5] = last_t;
+          p->first_row_time = grp_start_t;
+          p->nilmask = 0;
+          p->has_rows = 1;
+        }
+        cur_ord = ord;
+        if (ord < sq.w_first || ord >= sq.w_first + (int64_t)sq.n_wins) {
+          set_err(err, GEMX_E_INVALID); /* descriptor min/max_time lied */
+          return;
+        }
+        cnt = 0;
+        sumf = 0;
+        sumi = 0;
+        grp_start_t = t;
+        if (COLTYPE == GEMX_TYPE_FLOAT) {
+          minv.f = fv; maxv.f = fv; firstv.f = fv;
+        } else {
+          minv.i = iv; maxv.i = iv; firstv.i = iv;
+        }
+        min_t = t; max_t = t; first_t = t;
+      }
+      cnt++;
+      if (COLTYPE == GEMX_TYPE_FLOAT) {
+        sumf += fv;
+        /* first-occurrence-wins strict compares (column_util.go:204-209);
+         * NaN compares false on both → never replaces (Go parity) */
+        if (minv.f > fv) { minv.f = fv; min_t = t; }
+        if (maxv.f < fv) { maxv.f = fv; max_t = t; }
+        lastv.f = fv;
+      } else {
+        sumi += iv;
+        if (minv.i > iv) { minv.i = iv; min_t = t; }
+        if (maxv.i < iv) { maxv.i = iv; max_t = t; }
+        lastv.i = iv;
+      }
+      last_t = t;
+    }
+    if (cur_ord != INT64_MIN) {
+      int64_t slot = cur_ord - sq.w_first;
+      Partial *p = base + slot;
+      p->v[0].i = cnt;
+      if (COLTYPE == GEMX_TYPE_FLOAT) p->v[1].f = sumf; else p->v[1].i = sumi;
+      p->v[2] = minv;
+      p->v[3] = maxv;
+      p->v[4] = firstv;
+      p->v[5] = lastv;
+      p->t[0] = grp_start_t;
+      p->t[1] = grp_start_t;
+      p->t[2] = min_t;
+      p->t[3] = max_t;
+      p->t[4] = first_t;
+      p->t[5] = last_t;
+      p->first_row_time = grp_start_t;
+      p->nilmask = 0;
+      p->has_rows = 1;
+    }
+    (void)q_start;
+    (void)q_end;
+  }
+}
+
+/* snappy block decode, lane-serial (golang/snappy format;
+ * lib/compress/compress.go:132-144). Returns decoded length or -1. */
+__device__ int64_t d_snappy_decode(const uint8_t *src, int64_t len, uint8_t *dst,
+                                   int64_t cap) {
+  uint64_t dlen;
+  int hl = d_uvarint(src, len, &dlen);
+  if (hl <= 0 || (int64_t)dlen > cap) return -1;
+  int64_t s = hl, d = 0, n = (int64_t)dlen;
+  while (s < len) {
+    uint8_t tag = src[s];
+    int64_t length, off;
+    switch (tag & 3) {
+    case 0: {
+      int64_t l = tag >> 2;
+      s++;
+      if (l >= 60) {
+        int nb = (int)(l - 59);
+        if (s + nb > len) return -1;
+        l = 0;
+        for (int k = nb - 1; k >= 0; k--) l = (l << 8) | src[s + k];
+        s += nb;
+      }
+      length = l + 1;
+      if (s + length > len || d + length > n) return -1;
+      for (int64_t k = 0; k < length; k++) dst[d + k] = src[s + k];
+      s += length;
+      d += length;
+      continue;
+    }
+    case 1:
+      if (s + 2 > len) return -1;
+      length = 4 + ((tag >> 2) & 7);
+      off = ((int64_t)(tag >> 5) << 8) | src[s + 1];
+      s += 2;
+      break;
+    case 2:
+      if (s + 3 > len) return -1;
+      length = (tag >> 2) + 1;
+      off = (int64_t)src[s + 1] | ((int64_t)src[s + 2] << 8);
+      s += 3;
+      break;
+    default:
+      if (s + 5 > len) return -1;
+      length = (tag >> 2) + 1;
+      off = (int64_t)src[s + 1] | ((int64_t)src[s + 2] << 8) |
+            ((int64_t)src[s + 3] << 16) | ((int64_t)src[s + 4] << 24);
+      s += 5;
+      break;
+    }
+    if (off <= 0 || d < off || d + length > n) return -1;
+    for (int64_t k = 0; k < length; k++) dst[d + k] = dst[d + k - off];
+    d += length;
+  }
+  return d == n ? d : -1;
+}
+
+/* general kernel: segments with nil bitmaps / snappy / empty blocks.
+ * Decodes times+values into per-lane scratch then reduces with the exact
+ * bug-compatible index semantics (oracle/agg.c reduce_* functions). */
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_scan_general(
+    const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
+    const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
+    uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
+    int64_t offset, uint8_t *__restrict__ scratch, uint64_t scratch_per_lane,
+    uint32_t nlanes, DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  if (gid >= nlanes) return;
+  uint8_t *my = scratch + (uint64_t)gid * scratch_per_lane;
+  int64_t *tbuf = (int64_t *)my;                   /* 4096 × 8 */
+  int64_t *vbuf = (int64_t *)(my + 4096 * 8);      /* dense values */
+  uint8_t *sbuf = my + 2 * 4096 * 8;               /* snappy scratch 40KB */
+
+  for (uint32_t li = gid; li < nseg_ids; li += nlanes) {
+    uint32_t si = seg_ids[li];
+    const gemx_seg_desc d = descs[si];
+    const SegQ sq = segq[si];
+    int rows = (int)d.rows;
+    if (rows > 4096) { set_err(err, GEMX_E_INVALID); return; }
+
+    /* ---- decode times ---- */
+    {
+      const uint8_t *tseg = blob + d.time_offset;
+      int64_t tlen = d.time_size;
+      if (tlen < 5) { set_err(err, GEMX_E_DECODE); return; }
+      const uint8_t *enc = tseg + 5;
+      int64_t elen = tlen - 5;
+      if (tseg[0] == 17) { /* BlockIntegerOne */
+        memcpy(&tbuf[0], tseg + 1, 8);
+      } else {
+        int tag = enc[0] >> 4;
+        if (tag == 3) { /* snappy: raw LE bytes (timestamp.go:274-297) */
+          if (elen < 9) { set_err(err, GEMX_E_DECODE); return; }
+          int64_t comp_len = (int64_t)d_u32be(enc + 5);
+          int64_t dl = d_snappy_decode(enc + 9, comp_len, sbuf, 4096 * 8);
+          if (dl != rows * 8) { set_err(err, GEMX_E_DECODE); return; }
+          for (int i = 0; i < rows; i++) tbuf[i] = (int64_t)d_u64le(sbuf + i * 8);
+        } else {
+          TimeIter ti;
+          int rc = ti.init(enc, elen);
+          if (rc) { set_err(err, rc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+          for (int i = 0; i < rows; i++)
+            if (ti.next(&tbuf[i])) { set_err(err, GEMX_E_DECODE); return; }
+        }
+      }
+    }
+
+    /* ---- decode data (dense values + bitmap) ---- */
+    SegHeader h;
+    if (parse_data_header(blob + d.data_offset, d.data_size, COLTYPE, &h)) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    int nilcount;
+    int dense;
+    if (h.one_value) {
+      nilcount = h.nilcount;
+      dense = 1 - nilcount;
+      if (dense) memcpy(&vbuf[0], h.enc, 8);
+    } else if (h.enc_len == 0) { /* empty block */
+      nilcount = h.nilcount;
+      dense = 0;
+    } else if (COLTYPE == GEMX_TYPE_FLOAT) {
+      int tag = h.enc[0] >> 4;
+      if (tag == 2) { /* snappy floats (compress.go:81-84,149) */
+        int64_t dl = d_snappy_decode(h.enc + 1, h.enc_len - 1, sbuf, 4096 * 8);
+        if (dl < 0 || dl % 8) { set_err(err, GEMX_E_DECODE); return; }
+        dense = (int)(dl / 8);
+        for (int i = 0; i < dense; i++) vbuf[i] = (int64_t)d_u64le(sbuf + i * 8);
+      } else {
+        FloatIter fit;
+        int rc = fit.init(h.enc, h.enc_len);
+        if (rc) { set_err(err, rc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+        dense = 0;
+        double x;
+        while (dense < 4096 && fit.next(&x) == 0) {
+          memcpy(&vbuf[dense], &x, 8);
+          dense++;
+        }
+      }
+      nilcount = h.bitmap ? h.nilcount : 0;
+    } else {
+      IntIter iit;
+      int rc = iit.init(h.enc, h.enc_len);
+      if (rc) { set_err(err, rc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+      dense = 0;
+      int64_t x;
+      while (dense < 4096 && iit.next(&x) == 0) {
+        vbuf[dense] = x;
+        dense++;
+      }
+      nilcount = h.bitmap ? h.nilcount : 0;
+    }
+    if (h.bitmap == nullptr && !h.one_value && h.enc_len > 0 && dense != rows &&
+        h.nilcount == 0) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    if (h.bitmap && dense + nilcount != rows) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+
+    /* ---- per-window group reduce (oracle/agg.c semantics) ---- */
+    Partial *base = partials + sq.partial_base;
+    int start = 0;
+    while (start < rows) {
+      int64_t ord = interval ? win_ordinal(tbuf[start], interval, offset) : 0;
+      int end = start;
+      if (interval) {
+        while (end < rows && win_ordinal(tbuf[end], interval, offset) == ord) end++;
+      } else {
+        end = rows;
+      }
+      if (ord < sq.w_first || ord >= sq.w_first + (int64_t)sq.n_wins) {
+        set_err(err, GEMX_E_INVALID);
+        return;
+      }
+      Partial *p = base + (ord - sq.w_first);
+      p->has_rows = 1;
+      p->first_row_time = tbuf[start];
+      p->nilmask = 0;
+
+      /* valid prefix counts */
+      int v_before = 0;
+      for (int i = 0; i < start; i++) v_before += bm_valid(&h, i);
+      int v_in = 0;
+      for (int i = start; i < end; i++) v_in += bm_valid(&h, i);
+      int all_nil_col = (nilcount == rows);
+
+      /* count (series_agg_func.gen.go:24-42) */
+      p->v[0].i = v_in;
+      p->t[0] = tbuf[start];
+      if (v_in == 0) p->nilmask |= 1u << 0;
+      /* sum (:48-78; bug-compat time at value index) */
+      {
+        int vs = h.bitmap ? v_before : start;
+        int ve = vs + (h.bitmap ? v_in : (end - start));
+        if (COLTYPE == GEMX_TYPE_FLOAT) {
+          double s = 0;
+          for (int i = vs; i < ve; i++) { double x; memcpy(&x, &vbuf[i], 8); s += x; }
+          p->v[1].f = s;
+        } else {
+          int64_t s = 0;
+          for (int i = vs; i < ve; i++) s += vbuf[i];
+          p->v[1].i = s;
+        }
+        int idx = all_nil_col ? start : (vs < rows ? vs : rows - 1);
+        p->t[1] = tbuf[idx];
+        if (ve == vs) p->nilmask |= 1u << 1;
+      }
+      /* min/max (column_util.go:190-278) */
+      for (int m = 0; m < 2; m++) {
+        int op = 2 + m; /* 2 min, 3 max */
+        int row = -1;
+        int64_t bv = 0;
+        double bf = 0;
+        if (dense > 0) {
+          int vIdx = h.bitmap ? v_before : start;
+          int skip = vIdx;
+          for (int i = start; i < end && vIdx < dense; i++) {
+            if (!bm_valid(&h, i)) continue;
+            int better;
+            if (vIdx == skip)
+              better = 1;
+            else if (COLTYPE == GEMX_TYPE_FLOAT) {
+              double x;
+              memcpy(&x, &vbuf[vIdx], 8);
+              better = m ? (bf < x) : (bf > x);
+            } else
+              better = m ? (bv < vbuf[vIdx]) : (bv > vbuf[vIdx]);
+            if (better) {
+              if (COLTYPE == GEMX_TYPE_FLOAT) memcpy(&bf, &vbuf[vIdx], 8);
+              else bv = vbuf[vIdx];
+              row = i;
+            }
+            vIdx++;
+            if (!h.bitmap && i - start + 1 >= end - start) break;
+          }
+        }
+        if (row == -1) {
+          p->nilmask |= 1u << op;
+          p->t[op] = tbuf[all_nil_col ? start : 0]; /* fn returns index 0 */
+          p->v[op].i = 0;
+        } else {
+          if (COLTYPE == GEMX_TYPE_FLOAT) p->v[op].f = bf; else p->v[op].i = bv;
+          p->t[op] = tbuf[row];
+        }
+      }
+      /* first (column_util.go:23-55) */
+      {
+        int row = -1;
+        if (dense > 0) {
+          int vIdx = h.bitmap ? v_before : start;
+          for (int i = start; i < end && vIdx < dense; i++) {
+            if (!bm_valid(&h, i)) continue;
+            row = i;
+            break;
+          }
+          if (row >= 0) {
+            int vi = h.bitmap ? v_before : start;
+            p->v[4].i = vbuf[vi];
+          }
+        }
+        if (row == -1) {
+          p->nilmask |= 1u << 4;
+          p->t[4] = tbuf[all_nil_col ? start : 0];
+          p->v[4].i = 0;
+        } else
+          p->t[4] = tbuf[row];
+      }
+      /* last (column_util.go:57-85) */
+      {
+        int row = -1;
+        for (int i = end - 1; i >= start; i--) {
+          if (bm_valid(&h, i)) { row = i; break; }
+        }
+        if (row == -1 || dense == 0) {
+          p->nilmask |= 1u << 5;
+          p->t[5] = tbuf[all_nil_col ? start : 0];
+          p->v[5].i = 0;
+        } else {
+          int vIdx = 0;
+          if (h.bitmap) {
+            for (int i = 0; i < row; i++) vIdx += bm_valid(&h, i);
+          } else
+            vIdx = row;
+          p->v[5].i = vbuf[vIdx];
+          p->t[5] = tbuf[row];
+        }
+      }
+      start = end;
+    }
+  }
+}
+
+/* ---------------- merge kernel ---------------- */
+
+/* one lane per (sid, window): merge contributing segments' partials in time
+ * order with fv() semantics (series_agg_func.gen.go:44-274) */
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_merge(
+    const SeriesQ *__restrict__ series, uint32_t nseries,
+    const SegQ *__restrict__ segq, const Partial *__restrict__ partials,
+    gemx_agg_row *__restrict__ rows, uint64_t total_rows, int64_t interval,
+    int64_t offset, int64_t q_start) {
+  uint64_t gid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+  for (uint64_t r = gid; r < total_rows; r += gridDim.x * (uint64_t)blockDim.x) {
+    /* locate series by binary search on out_base */
+    uint32_t lo = 0, hi = nseries - 1;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi + 1) >> 1;
+      if (series[mid].out_base <= r) lo = mid;
+      else hi = mid - 1;
+    }
+    const SeriesQ s = series[lo];
+    int64_t w = s.w_min + (int64_t)(r - s.out_base);
+
+    gemx_agg_row out;
+    memset(&out, 0, sizeof(out));
+    out.sid = s.sid;
+    /* interval==0 ⇒ single window [start_time, end_time+1) (select.go:581) */
+    out.win_start = interval ? win_start_of(w, interval, offset) : q_start;
+
+    int active[6] = {0, 0, 0, 0, 0, 0};
+    gemx_val av[6];
+    int64_t at[6], nt[6];
+    int any = 0;
+    /* segments of this series whose span contains w: contiguous run; find
+     * first via binary search on w_first, then walk */
+    uint32_t a = s.seg_start, b = s.seg_start + s.seg_count;
+    /* first segment with w_first + n_wins > w */
+    uint32_t flo = a, fhi = b;
+    while (flo < fhi) {
+      uint32_t mid = (flo + fhi) >> 1;
+      if (segq[mid].w_first + (int64_t)segq[mid].n_wins > w) fhi = mid;
+      else flo = mid + 1;
+    }
+    for (uint32_t si = flo; si < b && segq[si].w_first <= w; si++) {
+      const SegQ q = segq[si];
+      if (w < q.w_first || w >= q.w_first + (int64_t)q.n_wins) continue;
+      const Partial p = partials[q.partial_base + (w - q.w_first)];
+      if (!p.has_rows) continue;
+      any = 1;
+      out.first_row_time = p.first_row_time;
+      for (int op = 0; op < 6; op++) {
+        if (p.nilmask & (1u << op)) {
+          nt[op] = p.t[op];
+          if (!active[op]) at[op] = p.t[op]; /* placeholder */
+          continue;
+        }
+        if (!active[op]) {
+          active[op] = 1;
+          av[op] = p.v[op];
+          at[op] = p.t[op];
+          continue;
+        }
+        switch (op) {
+        case 0: av[op].i += p.v[op].i; break; /* count merge */
+        case 1:
+          if (COLTYPE == GEMX_TYPE_FLOAT) av[op].f += p.v[op].f;
+          else av[op].i += p.v[op].i;
+          break;
+        case 2: {
+          int repl = (COLTYPE == GEMX_TYPE_FLOAT) ? (p.v[op].f < av[op].f)
+                                                  : (p.v[op].i < av[op].i);
+          if (repl) { av[op] = p.v[op]; at[op] = p.t[op]; }
+          break;
+        }
+        case 3: {
+          int repl = (COLTYPE == GEMX_TYPE_FLOAT) ? (p.v[op].f > av[op].f)
+                                                  : (p.v[op].i > av[op].i);
+          if (repl) { av[op] = p.v[op]; at[op] = p.t[op]; }
+          break;
+        }
+        case 4: break;                                  /* first: keep */
+        case 5: av[op] = p.v[op]; at[op] = p.t[op]; break; /* last: assign */
+        }
+      }
+    }
+
+    if (!any) {
+      out.count = -1; /* marks a gap row: host compacts it away */
+      rows[r] = out;
+      continue;
+    }
+    out.count = active[0] ? av[0].i : 0;
+    out.count_time = active[0] ? at[0] : nt[0];
+    out.sum = av[1];
+    out.sum_time = active[1] ? at[1] : nt[1];
+    out.sum_isnil = !active[1];
+    out.minv = av[2];
+    out.min_time = active[2] ? at[2] : nt[2];
+    out.min_isnil = !active[2];
+    out.maxv = av[3];
+    out.max_time = active[3] ? at[3] : nt[3];
+    out.max_isnil = !active[3];
+    out.firstv = av[4];
+    out.first_time = active[4] ? at[4] : nt[4];
+    out.first_isnil = !active[4];
+    out.lastv = av[5];
+    out.last_time = active[5] ? at[5] : nt[5];
+    out.last_isnil = !active[5];
+    if (!active[1]) out.sum.i = 0;
+    if (!active[2]) out.minv.i = 0;
+    if (!active[3]) out.maxv.i = 0;
+    if (!active[4]) out.firstv.i = 0;
+    if (!active[5]) out.lastv.i = 0;
+    rows[r] = out;
+  }
+}
+
+/* ---------------- host: engine ---------------- */
+
+static __thread char g_err[512];
+static void seterr(const char *msg) { snprintf(g_err, sizeof(g_err), "%s", msg); }
+
+static inline uint32_t h_u32be(const uint8_t *p) {
+  return ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) | ((uint32_t)p[2] << 8) |
+         p[3];
+}
+
+#define HIP_CHECK(x)                                                                   \
+  do {                                                                                 \
+    hipError_t _e = (x);                                                               \
+    if (_e != hipSuccess) {                                                            \
+      snprintf(g_err, sizeof(g_err), "HIP error %s at %s:%d", hipGetErrorString(_e),   \
+               __FILE__, __LINE__);                                                    \
+      return GEMX_E_HIP;                                                               \
+    }                                                                                  \
+  } while (0)
+
+struct gemx_shard {
+  int device;
+  int col_type;
+  uint64_t nsegs;
+  uint64_t blob_bytes;
+  uint8_t *d_blob;
+  gemx_seg_desc *d_descs;
+  std::vector<gemx_seg_desc> h_descs;
+  /* segment classification (attach-time) */
+  std::vector<uint32_t> fast_ids, general_ids;
+  uint32_t *d_fast_ids, *d_general_ids;
+  /* series grouping: ranges in desc order */
+  struct SeriesRange {
+    uint64_t sid;
+    uint32_t start, count;
+  };
+  std::vector<SeriesRange> series_ranges;
+  uint64_t total_rows_scanned; /* Σ rows */
+  hipStream_t stream;
+};
+
+extern "C" int gemx_abi_version(void) { return GEMX_ABI_VERSION; }
+extern "C" const char *gemx_last_error(void) { return g_err; }
+
+extern "C" int gemx_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+/* attach-time classification: peek headers in HOST memory */
+static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col_type,
+                            bool *fast) {
+  if (d.data_size < 1 || d.time_size < 1) return GEMX_E_INVALID;
+  const uint8_t *ds = blob + d.data_offset;
+  const uint8_t *ts = blob + d.time_offset;
+  *fast = true;
+  uint8_t dt = ds[0];
+  /* time: one-value or Full + {const-delta, simple8b, uncompressed} */
+  if (ts[0] == 17) {
+    /* one row: fine for fast path */
+  } else if (ts[0] == 31) {
+    if (d.time_size < 6) return GEMX_E_INVALID;
+    int ttag = ts[5] >> 4;
+    if (ttag == 3) *fast = false; /* snappy times need scratch */
+    else if (ttag != 1 && ttag != 2 && ttag != 4) return GEMX_E_INVALID;
+  } else
+    return GEMX_E_INVALID;
+  /* data */
+  if (dt > 16 && dt < 21) {
+    if (d.data_size <= 1) *fast = false; /* one-value null row: general */
+  } else if (dt >= 40 && dt < 45) {
+    *fast = false; /* empty block: general (all-nil bug-times) */
+  } else if (dt >= 30 && dt < 35) {
+    if (d.data_size < 6) return GEMX_E_INVALID;
+    int tag = ds[5] >> 4;
+    if (col_type == GEMX_TYPE_FLOAT) {
+      if (tag == 2) *fast = false;           /* snappy */
+      else if (tag == 6) return GEMX_E_UNSUPPORTED; /* MLF (config-gated off) */
+      else if (tag == 1) return GEMX_E_UNSUPPORTED; /* legacy gorilla */
+      else if (tag != 0 && tag != 3 && tag != 4 && tag != 5) return GEMX_E_INVALID;
+    } else {
+      if (tag == 3) return GEMX_E_UNSUPPORTED; /* zstd: not on device yet */
+      if (tag != 1 && tag != 2 && tag != 4) return GEMX_E_INVALID;
+    }
+  } else if (dt == (uint8_t)col_type) {
+    *fast = false; /* nil bitmap present */
+    /* still reject unsupported codecs */
+    uint32_t bmlen = h_u32be(ds + 1);
+    if (d.data_size < 13 + bmlen + 1) return GEMX_E_INVALID;
+    int tag = ds[13 + bmlen] >> 4;
+    if (col_type == GEMX_TYPE_FLOAT) {
+      if (tag == 6 || tag == 1) return GEMX_E_UNSUPPORTED;
+    } else {
+      if (tag == 3) return GEMX_E_UNSUPPORTED;
+    }
+  } else
+    return GEMX_E_INVALID;
+  return 0;
+}
+
+extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_bytes,
+                                 const gemx_seg_desc *descs, uint64_t nsegs,
+                                 int col_type, gemx_shard **out) {
+  if (gemx_device_count() == 0) {
+    seterr("no HIP device: the MI355X engine has no CPU fallback");
+    return GEMX_E_NOGPU;
+  }
+  if (col_type != GEMX_TYPE_FLOAT && col_type != GEMX_TYPE_INT) {
+    seterr("col_type must be GEMX_TYPE_FLOAT or GEMX_TYPE_INT");
+    return GEMX_E_INVALID;
+  }
+  HIP_CHECK(hipSetDevice(device));
+  gemx_shard *s = new gemx_shard();
+  s->device = device;
+  s->col_type = col_type;
+  s->nsegs = nsegs;
+  s->blob_bytes = blob_bytes;
+  s->h_descs.assign(descs, descs + nsegs);
+  s->total_rows_scanned = 0;
+
+  /* validate ordering + classify */
+  const uint8_t *hb = (const uint8_t *)blob;
+  for (uint64_t i = 0; i < nsegs; i++) {
+    const gemx_seg_desc &d = descs[i];
+    if (d.data_offset + d.data_size > blob_bytes ||
+        d.time_offset + d.time_size > blob_bytes || d.rows == 0 || d.rows > 4096) {
+      seterr("segment descriptor out of range");
+      delete s;
+      return GEMX_E_INVALID;
+    }
+    if (i > 0 && descs[i].sid == descs[i - 1].sid &&
+        descs[i].min_time < descs[i - 1].min_time) {
+      seterr("descriptors not time-ascending within sid");
+      delete s;
+      return GEMX_E_INVALID;
+    }
+    bool fast;
+    int rc = classify_segment(hb, d, col_type, &fast);
+    if (rc != 0) {
+      seterr(rc == GEMX_E_UNSUPPORTED
+                 ? "segment uses a codec not yet on-device (zstd/MLF/legacy)"
+                 : "malformed segment header");
+      delete s;
+      return rc;
+    }
+    if (fast)
+      s->fast_ids.push_back((uint32_t)i);
+    else
+      s->general_ids.push_back((uint32_t)i);
+    s->total_rows_scanned += d.rows;
+    /* series ranges */
+    if (s->series_ranges.empty() || s->series_ranges.back().sid != d.sid) {
+      s->series_ranges.push_back({d.sid, (uint32_t)i, 1});
+    } else {
+      s->series_ranges.back().count++;
+    }
+  }
+
+  HIP_CHECK(hipStreamCreate(&s->stream));
+  HIP_CHECK(hipMalloc(&s->d_blob, blob_bytes ? blob_bytes : 1));
+  HIP_CHECK(hipMemcpyAsync(s->d_blob, blob, blob_bytes, hipMemcpyHostToDevice,
+                           s->stream));
+  HIP_CHECK(hipMalloc(&s->d_descs, sizeof(gemx_seg_desc) * (nsegs ? nsegs : 1)));
+  HIP_CHECK(hipMemcpyAsync(s->d_descs, descs, sizeof(gemx_seg_desc) * nsegs,
+                           hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&s->d_fast_ids,
+                      sizeof(uint32_t) * (s->fast_ids.empty() ? 1 : s->fast_ids.size())));
+  if (!s->fast_ids.empty())
+    HIP_CHECK(hipMemcpyAsync(s->d_fast_ids, s->fast_ids.data(),
+                             sizeof(uint32_t) * s->fast_ids.size(),
+                             hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&s->d_general_ids,
+                      sizeof(uint32_t) *
+                          (s->general_ids.empty() ? 1 : s->general_ids.size())));
+  if (!s->general_ids.empty())
+    HIP_CHECK(hipMemcpyAsync(s->d_general_ids, s->general_ids.data(),
+                             sizeof(uint32_t) * s->general_ids.size(),
+                             hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipStreamSynchronize(s->stream));
+  *out = s;
+  return GEMX_OK;
+}
+
+extern "C" int gemx_shard_close(gemx_shard *s) {
+  if (!s) return GEMX_OK;
+  hipSetDevice(s->device);
+  hipFree(s->d_blob);
+  hipFree(s->d_descs);
+  hipFree(s->d_fast_ids);
+  hipFree(s->d_general_ids);
+  hipStreamDestroy(s->stream);
+  delete s;
+  return GEMX_OK;
+}
+
+extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time,
+                             int64_t interval, int64_t offset,
+                             gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                             gemx_query_stats *stats) {
+  if (!s) return GEMX_E_INVALID;
+  HIP_CHECK(hipSetDevice(s->device));
+  const uint64_t nsegs = s->nsegs;
+
+  /* host precompute: per-segment window spans, per-series output ranges */
+  std::vector<SegQ> segq(nsegs);
+  std::vector<SeriesQ> sq(s->series_ranges.size());
+  uint64_t partial_slots = 0, total_rows = 0;
+  for (size_t g = 0; g < s->series_ranges.size(); g++) {
+    auto &r = s->series_ranges[g];
+    int64_t wmin = INT64_MAX, wmax = INT64_MIN;
+    for (uint32_t i = r.start; i < r.start + r.count; i++) {
+      const gemx_seg_desc &d = s->h_descs[i];
+      int64_t w0 = interval ? win_ordinal(d.min_time, interval, offset) : 0;
+      int64_t w1 = interval ? win_ordinal(d.max_time, interval, offset) : 0;
+      segq[i].w_first = w0;
+      segq[i].n_wins = (uint32_t)(w1 - w0 + 1);
+      segq[i].partial_base = partial_slots;
+      segq[i].series_idx = (uint32_t)g;
+      partial_slots += segq[i].n_wins;
+      wmin = std::min(wmin, w0);
+      wmax = std::max(wmax, w1);
+    }
+    sq[g].sid = r.sid;
+    sq[g].w_min = wmin;
+    sq[g].out_base = total_rows;
+    sq[g].n_wins = (uint32_t)(wmax - wmin + 1);
+    sq[g].seg_start = r.start;
+    sq[g].seg_count = r.count;
+    total_rows += sq[g].n_wins;
+  }
+
+  /* device buffers */
+  SegQ *d_segq = nullptr;
+  SeriesQ *d_sq = nullptr;
+  Partial *d_part = nullptr;
+  gemx_agg_row *d_rows = nullptr;
+  DevErr *d_err = nullptr;
+  uint8_t *d_scratch = nullptr;
+  HIP_CHECK(hipMalloc(&d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
+  HIP_CHECK(hipMemcpyAsync(d_segq, segq.data(), sizeof(SegQ) * nsegs,
+                           hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&d_sq, sizeof(SeriesQ) * (sq.empty() ? 1 : sq.size())));
+  HIP_CHECK(hipMemcpyAsync(d_sq, sq.data(), sizeof(SeriesQ) * sq.size(),
+                           hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&d_part, sizeof(Partial) * (partial_slots ? partial_slots : 1)));
+  HIP_CHECK(hipMemsetAsync(d_part, 0, sizeof(Partial) * partial_slots, s->stream));
+  HIP_CHECK(hipMalloc(&d_rows, sizeof(gemx_agg_row) * (total_rows ? total_rows : 1)));
+  HIP_CHECK(hipMalloc(&d_err, sizeof(DevErr)));
+  HIP_CHECK(hipMemsetAsync(d_err, 0, sizeof(DevErr), s->stream));
+
+  /* general-kernel scratch: 4096×8 (times) + 4096×8 (values) + 40KB snappy */
+  const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960;
+  uint32_t gen_lanes = 0;
+  if (!s->general_ids.empty()) {
+    gen_lanes = (uint32_t)std::min<uint64_t>(s->general_ids.size(), 16384);
+    HIP_CHECK(hipMalloc(&d_scratch, scratch_per_lane * gen_lanes));
+  }
+
+  hipEvent_t ev0, ev1, ev2;
+  HIP_CHECK(hipEventCreate(&ev0));
+  HIP_CHECK(hipEventCreate(&ev1));
+  HIP_CHECK(hipEventCreate(&ev2));
+
+  HIP_CHECK(hipEventRecord(ev0, s->stream));
+  const int TPB = 256;
+  if (!s->fast_ids.empty()) {
+    uint32_t n = (uint32_t)s->fast_ids.size();
+    uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
+    if (s->col_type == GEMX_TYPE_FLOAT)
+      hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
+                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_fast_ids, n,
+                         d_part, interval, offset, start_time, end_time, d_err);
+    else
+      hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
+                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_fast_ids, n,
+                         d_part, interval, offset, start_time, end_time, d_err);
+  }
+  if (!s->general_ids.empty()) {
+    uint32_t n = (uint32_t)s->general_ids.size();
+    uint32_t blocks = (gen_lanes + TPB - 1) / TPB;
+    if (s->col_type == GEMX_TYPE_FLOAT)
+      hipLaunchKernelGGL((k_scan_general<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
+                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_general_ids, n,
+                         d_part, interval, offset, d_scratch, scratch_per_lane,
+                         gen_lanes, d_err);
+    else
+      hipLaunchKernelGGL((k_scan_general<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
+                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_general_ids, n,
+                         d_part, interval, offset, d_scratch, scratch_per_lane,
+                         gen_lanes, d_err);
+  }
+  HIP_CHECK(hipEventRecord(ev1, s->stream));
+  if (total_rows > 0) {
+    uint32_t blocks = (uint32_t)std::min<uint64_t>((total_rows + TPB - 1) / TPB, 65535);
+    if (s->col_type == GEMX_TYPE_FLOAT)
+      hipLaunchKernelGGL((k_merge<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part, d_rows,
+                         total_rows, interval, offset, start_time);
+    else
+      hipLaunchKernelGGL((k_merge<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part, d_rows,
+                         total_rows, interval, offset, start_time);
+  }
+  HIP_CHECK(hipEventRecord(ev2, s->stream));
+
+  /* check device error + fetch rows */
+  DevErr herr = {0};
+  HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
+                           s->stream));
+  std::vector<gemx_agg_row> hrows(total_rows);
+  HIP_CHECK(hipMemcpyAsync(hrows.data(), d_rows, sizeof(gemx_agg_row) * total_rows,
+                           hipMemcpyDeviceToHost, s->stream));
+  HIP_CHECK(hipStreamSynchronize(s->stream));
+
+  float ms_scan = 0, ms_merge = 0, ms_total = 0;
+  hipEventElapsedTime(&ms_scan, ev0, ev1);
+  hipEventElapsedTime(&ms_merge, ev1, ev2);
+  hipEventElapsedTime(&ms_total, ev0, ev2);
+  hipEventDestroy(ev0);
+  hipEventDestroy(ev1);
+  hipEventDestroy(ev2);
+
+  hipFree(d_segq);
+  hipFree(d_sq);
+  hipFree(d_part);
+  hipFree(d_rows);
+  hipFree(d_err);
+  if (d_scratch) hipFree(d_scratch);
+
+  if (herr.code != 0) {
+    seterr(herr.code == GEMX_E_UNSUPPORTED ? "unsupported codec on device"
+                                           : "segment decode failed on device");
+    return herr.code;
+  }
+
+  /* compact gap rows (count == -1) into caller buffer */
+  uint64_t n = 0;
+  for (uint64_t i = 0; i < total_rows; i++) {
+    if (hrows[i].count < 0) continue;
+    if (n >= cap) {
+      seterr("output capacity too small");
+      return GEMX_E_CAP;
+    }
+    out_host[n++] = hrows[i];
+  }
+  *n_out = n;
+
+  if (stats) {
+    stats->decode_ms = ms_scan;
+    stats->merge_ms = ms_merge;
+    stats->total_ms = ms_total;
+    stats->points = s->total_rows_scanned;
+    uint64_t cb = 0;
+    for (auto &d : s->h_descs) cb += d.data_size + d.time_size;
+    stats->compressed_bytes = cb;
+    stats->n_rows = n;
+    stats->h2d_ms = 0;
+  }
+  return GEMX_OK;
+}

@@ -123,6 +123,11 @@ def load():
     lib.orc_scan_agg.argtypes = [u8p, i64, C.c_void_p, i64, C.c_int, i64, i64, i64, i64, C.c_void_p, i64]
     lib.orc_scan_agg_mt.restype = i64
     lib.orc_scan_agg_mt.argtypes = [u8p, i64, C.c_void_p, i64, C.c_int, i64, i64, i64, i64, C.c_void_p, i64, C.c_int]
+    lib.orc_gen_shard.restype = i64
+    lib.orc_gen_shard.argtypes = [
+        C.c_uint64, C.c_uint64, C.c_uint64, C.c_uint32, i64, i64, C.c_int,
+        u8p, i64, C.c_void_p, i64, i64p,
+    ]
     lib.orc_agg_cursor.restype = i64
     lib.orc_agg_cursor.argtypes = [
         C.c_int, C.c_uint32, C.c_int,
@@ -317,6 +322,37 @@ def scan_agg(blob, descs, col_type, start_time, end_time, interval, offset=0, ou
     if n < 0:
         raise ValueError("scan_agg failed")
     return out[:n].copy()
+
+
+GEN_FLOAT_WALK = 0
+GEN_FLOAT_RANDOM = 1
+GEN_INT_SMALL = 2
+
+
+def gen_shard(seed, nseries, pts_per_series, seg_rows=1000, t0=0, step_ns=10**9,
+              mode=GEN_FLOAT_WALK):
+    """Bulk synthetic shard (OpenMP). Returns (blob_bytes, descs)."""
+    lib = get()
+    segs = nseries * ((pts_per_series + seg_rows - 1) // seg_rows)
+    descs = np.zeros(segs, dtype=SEG_DESC_DTYPE)
+    bytes_per_pt = 10 if mode == GEN_FLOAT_RANDOM else 6
+    cap = int(nseries * pts_per_series * bytes_per_pt + segs * 64 + 4096)
+    blob = np.zeros(cap, dtype=np.uint8)
+    n_out = C.c_int64(0)
+    rc = lib.orc_gen_shard(
+        seed, nseries, pts_per_series, seg_rows, t0, step_ns, mode,
+        _u8(blob), cap, descs.ctypes.data_as(C.c_void_p), segs, C.byref(n_out),
+    )
+    if rc == -2:
+        cap = int(nseries * pts_per_series * 12 + segs * 64 + 4096)
+        blob = np.zeros(cap, dtype=np.uint8)
+        rc = lib.orc_gen_shard(
+            seed, nseries, pts_per_series, seg_rows, t0, step_ns, mode,
+            _u8(blob), cap, descs.ctypes.data_as(C.c_void_p), segs, C.byref(n_out),
+        )
+    if rc < 0:
+        raise ValueError(f"gen_shard failed: {rc}")
+    return blob[:rc].tobytes(), descs[: n_out.value]
 
 
 def agg_cursor(col_type, op, multi_call, dense_vals, valid_bits, times, rec_rows,

@@ -218,6 +218,16 @@ int64_t orc_agg_cursor(int col_type, uint32_t op, int multi_call,
 /* xorshift64 PRNG (seeded; matches SURVEY.md §8d spec: seed=42 per run). */
 uint64_t orc_xorshift64(uint64_t *state);
 
+/* Bulk synthetic shard generator (bench harness; OpenMP). mode: 0 = float
+ * quantized random walk (Gorilla-friendly), 1 = float full-random bits
+ * (worst case), 2 = int64 uniform [0,1000) (simple8b). descs_out must hold
+ * nseries*ceil(pts/seg_rows) orc_seg_desc. Returns blob bytes written,
+ * -2 if blob_cap too small, -1 on error. */
+int64_t orc_gen_shard(uint64_t seed, uint64_t nseries, uint64_t pts_per_series,
+                      uint32_t seg_rows, int64_t t0, int64_t step_ns, int mode,
+                      uint8_t *blob, int64_t blob_cap, void *descs_out,
+                      int64_t desc_cap, int64_t *out_nsegs);
+
 #ifdef __cplusplus
 }
 #endif

@@ -1,0 +1,178 @@
+"""GPU ↔ oracle parity for the fused scan-aggregate engine.
+
+Bit-exact for count/min/max/first/last (values AND times), 1e-9 relative for
+sum — the tolerance BASELINE.json's north_star states for sum/mean (float
+sums are reassociated across segment boundaries; within a segment both sides
+sum serially in time order).
+"""
+
+import numpy as np
+import pytest
+
+import binding as orc
+from shard_helpers import INT, F, I, build_shard
+
+pytestmark = pytest.mark.gpu
+
+
+def gpu_shard(blob, descs, col_type):
+    import opengemini_amd as gx
+
+    return gx.Shard(blob, descs, col_type)
+
+
+def assert_parity(gpu_rows, orc_rows, col_type):
+    assert len(gpu_rows) == len(orc_rows), (len(gpu_rows), len(orc_rows))
+    # identical row identity + ordering
+    for f in ("sid", "win_start", "first_row_time", "count", "count_time",
+              "min_time", "max_time", "first_time", "last_time", "sum_time",
+              "min_isnil", "max_isnil", "first_isnil", "last_isnil", "sum_isnil"):
+        assert np.array_equal(gpu_rows[f], orc_rows[f]), f
+    if col_type == F:
+        for f in ("min", "max", "first", "last"):
+            # bit-exact incl. NaN payloads
+            assert np.array_equal(
+                gpu_rows[f].view(np.uint64), orc_rows[f].view(np.uint64)
+            ), f
+        s_g, s_o = gpu_rows["sum"], orc_rows["sum"]
+        tol = 1e-9 * np.maximum(1.0, np.abs(s_o))
+        assert np.all(np.abs(s_g - s_o) <= tol), "sum tolerance"
+    else:
+        for f in ("min", "max", "first", "last", "sum"):
+            assert np.array_equal(
+                gpu_rows[f].view(np.int64), orc_rows[f].view(np.int64)
+            ), f
+
+
+class TestScanAggParity:
+    def _run(self, blob, descs, col_type, interval=INT, start=0, end=2**62):
+        sh = gpu_shard(blob, descs, col_type)
+        try:
+            gpu_rows, stats = sh.scan_agg(start, end, interval)
+        finally:
+            sh.close()
+        orc_rows = orc.scan_agg(blob, descs, col_type, start, end, interval)
+        assert_parity(gpu_rows, orc_rows, col_type)
+        return gpu_rows
+
+    def test_float_no_nulls(self):
+        rng = np.random.default_rng(21)
+        blob, d, _ = build_shard(rng, F, [11, 22, 33], null_frac=0.0)
+        self._run(blob, d, F)
+
+    def test_float_with_nulls(self):
+        rng = np.random.default_rng(22)
+        blob, d, _ = build_shard(rng, F, [101, 202, 303])
+        self._run(blob, d, F)
+
+    def test_int_with_nulls(self):
+        rng = np.random.default_rng(23)
+        blob, d, _ = build_shard(rng, I, [1, 2, 3, 4])
+        self._run(blob, d, I)
+
+    def test_int_const_delta_values(self):
+        rng = np.random.default_rng(24)
+        blob, d, _ = build_shard(
+            rng, I, [7], null_frac=0.0,
+            value_fn=lambda r, n: (np.arange(n) * 5 + 100).astype(np.int64),
+        )
+        self._run(blob, d, I)
+
+    def test_float_rle_and_same(self):
+        rng = np.random.default_rng(25)
+        blob, d, _ = build_shard(
+            rng, F, [8], null_frac=0.0,
+            value_fn=lambda r, n: np.repeat(
+                np.round(r.normal(0, 10, max(1, n // 150 + 1)), 1), 150
+            )[:n],
+        )
+        self._run(blob, d, F)
+        blob, d, _ = build_shard(
+            rng, F, [9], null_frac=0.0, value_fn=lambda r, n: np.full(n, 3.25)
+        )
+        self._run(blob, d, F)
+
+    def test_float_null_codec(self):
+        # full-random mantissae force the null (uncompressed) fallback
+        rng = np.random.default_rng(26)
+        blob, d, _ = build_shard(
+            rng, F, [5], null_frac=0.0,
+            value_fn=lambda r, n: r.random(n) * 1e6 + r.random(n),
+        )
+        self._run(blob, d, F)
+
+    def test_float_snappy_nan(self):
+        rng = np.random.default_rng(27)
+
+        def vf(r, n):
+            v = np.cumsum(r.normal(0, 1, n))
+            if n > 8:
+                v[5] = np.nan  # extremeDataValues → snappy (general kernel)
+            return v
+
+        blob, d, _ = build_shard(rng, F, [6], null_frac=0.0, value_fn=vf)
+        self._run(blob, d, F)
+
+    def test_one_row_segments(self):
+        rng = np.random.default_rng(28)
+        blob, d, _ = build_shard(rng, F, [9], seg_range=(3, 4), row_range=(1, 2),
+                                 null_frac=0.0)
+        self._run(blob, d, F)
+
+    def test_all_null_segments(self):
+        rng = np.random.default_rng(29)
+        blob, d, _ = build_shard(rng, F, [4], null_frac=1.0)
+        self._run(blob, d, F)
+
+    def test_no_interval(self):
+        rng = np.random.default_rng(30)
+        blob, d, _ = build_shard(rng, F, [1, 2])
+        self._run(blob, d, F, interval=0, start=0, end=2**61)
+
+    def test_windows_span_segments(self):
+        # config#1 shape: one series, many segments, windows cross boundaries
+        rng = np.random.default_rng(31)
+        blob, d, _ = build_shard(rng, F, [77], seg_range=(8, 9),
+                                 row_range=(1000, 1001), null_frac=0.1)
+        self._run(blob, d, F)
+
+    def test_bulk_generated_shard(self):
+        # the bench generator path end-to-end (1k series × 1k pts)
+        blob, descs = orc.gen_shard(42, 1000, 1000)
+        self._run(blob, descs, F)
+
+    def test_bulk_int_shard(self):
+        blob, descs = orc.gen_shard(43, 500, 1000, mode=orc.GEN_INT_SMALL)
+        self._run(blob, descs, I)
+
+    def test_bulk_random_float(self):
+        blob, descs = orc.gen_shard(44, 500, 1000, mode=orc.GEN_FLOAT_RANDOM)
+        self._run(blob, descs, F)
+
+
+class TestEngineGuards:
+    def test_zstd_rejected_not_silently_decoded(self):
+        # int values with huge varying deltas → zstd; engine must refuse
+        # loudly (GEMX_E_UNSUPPORTED), never fall back to CPU
+        import opengemini_amd as gx
+
+        rng = np.random.default_rng(40)
+        blob, d, _ = build_shard(
+            rng, I, [1], null_frac=0.0,
+            value_fn=lambda r, n: np.where(
+                np.arange(n) % 2 == 0, np.int64(2**61), np.int64(7)
+            ),
+        )
+        with pytest.raises(gx.GemxError):
+            gx.Shard(blob, d, I)
+
+    def test_stats_sane(self):
+        blob, descs = orc.gen_shard(45, 100, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            rows, stats = sh.scan_agg(0, 2**62, INT)
+        finally:
+            sh.close()
+        assert stats["points"] == 100 * 1000
+        assert stats["decode_ms"] > 0
+        assert stats["compressed_bytes"] == len(blob)
