@@ -585,6 +585,10 @@ __device__ __forceinline__ int bm_valid(const SegHeader *h, int i) {
   return (h->bitmap[s >> 3] >> (s & 7)) & 1;
 }
 
+/* all-invalid bitmap for empty blocks / one-value nulls (they carry no
+ * bitmap bytes but every row is nil) */
+__device__ const uint8_t d_zero_bm[512] = {0};
+
 /* ---------------- partials ---------------- */
 
 /* per-(segment,window) partial; op order: count,sum,min,max,first,last */
@@ -946,6 +950,11 @@ __global__ void __launch_bounds__(256) k_scan_general(
       set_err(err, GEMX_E_DECODE);
       return;
     }
+    if (!h.bitmap && nilcount == rows && rows > 0) {
+      /* empty block / one-value null: no bitmap bytes, all rows nil */
+      h.bitmap = d_zero_bm;
+      h.bm_off = 0;
+    }
 
     /* ---- per-window group reduce (oracle/agg.c semantics) ---- */
     Partial *base = partials + sq.partial_base;
@@ -1216,6 +1225,24 @@ static inline uint32_t h_u32be(const uint8_t *p) {
     }                                                                                  \
   } while (0)
 
+/* cached per-query-shape device state (re-used across repeated queries —
+ * the cursor pattern re-issues the same window spec every NextAggData) */
+struct QueryPlan {
+  bool valid = false;
+  int64_t start = 0, end = 0, interval = 0, offset = 0;
+  std::vector<SegQ> segq;
+  std::vector<SeriesQ> sq;
+  uint64_t partial_slots = 0, total_rows = 0;
+  SegQ *d_segq = nullptr;
+  SeriesQ *d_sq = nullptr;
+  Partial *d_part = nullptr;
+  gemx_agg_row *d_rows = nullptr;
+  DevErr *d_err = nullptr;
+  gemx_agg_row *h_rows = nullptr; /* pinned staging */
+  uint8_t *d_scratch = nullptr;
+  uint32_t gen_lanes = 0;
+};
+
 struct gemx_shard {
   int device;
   int col_type;
@@ -1235,7 +1262,19 @@ struct gemx_shard {
   std::vector<SeriesRange> series_ranges;
   uint64_t total_rows_scanned; /* Σ rows */
   hipStream_t stream;
+  QueryPlan plan;
 };
+
+static void free_plan(QueryPlan &p) {
+  if (p.d_segq) hipFree(p.d_segq);
+  if (p.d_sq) hipFree(p.d_sq);
+  if (p.d_part) hipFree(p.d_part);
+  if (p.d_rows) hipFree(p.d_rows);
+  if (p.d_err) hipFree(p.d_err);
+  if (p.d_scratch) hipFree(p.d_scratch);
+  if (p.h_rows) hipHostFree(p.h_rows);
+  p = QueryPlan();
+}
 
 extern "C" int gemx_abi_version(void) { return GEMX_ABI_VERSION; }
 extern "C" const char *gemx_last_error(void) { return g_err; }
@@ -1383,6 +1422,7 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
 extern "C" int gemx_shard_close(gemx_shard *s) {
   if (!s) return GEMX_OK;
   hipSetDevice(s->device);
+  free_plan(s->plan);
   hipFree(s->d_blob);
   hipFree(s->d_descs);
   hipFree(s->d_fast_ids);
@@ -1399,61 +1439,77 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
   if (!s) return GEMX_E_INVALID;
   HIP_CHECK(hipSetDevice(s->device));
   const uint64_t nsegs = s->nsegs;
-
-  /* host precompute: per-segment window spans, per-series output ranges */
-  std::vector<SegQ> segq(nsegs);
-  std::vector<SeriesQ> sq(s->series_ranges.size());
-  uint64_t partial_slots = 0, total_rows = 0;
-  for (size_t g = 0; g < s->series_ranges.size(); g++) {
-    auto &r = s->series_ranges[g];
-    int64_t wmin = INT64_MAX, wmax = INT64_MIN;
-    for (uint32_t i = r.start; i < r.start + r.count; i++) {
-      const gemx_seg_desc &d = s->h_descs[i];
-      int64_t w0 = interval ? win_ordinal(d.min_time, interval, offset) : 0;
-      int64_t w1 = interval ? win_ordinal(d.max_time, interval, offset) : 0;
-      segq[i].w_first = w0;
-      segq[i].n_wins = (uint32_t)(w1 - w0 + 1);
-      segq[i].partial_base = partial_slots;
-      segq[i].series_idx = (uint32_t)g;
-      partial_slots += segq[i].n_wins;
-      wmin = std::min(wmin, w0);
-      wmax = std::max(wmax, w1);
-    }
-    sq[g].sid = r.sid;
-    sq[g].w_min = wmin;
-    sq[g].out_base = total_rows;
-    sq[g].n_wins = (uint32_t)(wmax - wmin + 1);
-    sq[g].seg_start = r.start;
-    sq[g].seg_count = r.count;
-    total_rows += sq[g].n_wins;
-  }
-
-  /* device buffers */
-  SegQ *d_segq = nullptr;
-  SeriesQ *d_sq = nullptr;
-  Partial *d_part = nullptr;
-  gemx_agg_row *d_rows = nullptr;
-  DevErr *d_err = nullptr;
-  uint8_t *d_scratch = nullptr;
-  HIP_CHECK(hipMalloc(&d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
-  HIP_CHECK(hipMemcpyAsync(d_segq, segq.data(), sizeof(SegQ) * nsegs,
-                           hipMemcpyHostToDevice, s->stream));
-  HIP_CHECK(hipMalloc(&d_sq, sizeof(SeriesQ) * (sq.empty() ? 1 : sq.size())));
-  HIP_CHECK(hipMemcpyAsync(d_sq, sq.data(), sizeof(SeriesQ) * sq.size(),
-                           hipMemcpyHostToDevice, s->stream));
-  HIP_CHECK(hipMalloc(&d_part, sizeof(Partial) * (partial_slots ? partial_slots : 1)));
-  HIP_CHECK(hipMemsetAsync(d_part, 0, sizeof(Partial) * partial_slots, s->stream));
-  HIP_CHECK(hipMalloc(&d_rows, sizeof(gemx_agg_row) * (total_rows ? total_rows : 1)));
-  HIP_CHECK(hipMalloc(&d_err, sizeof(DevErr)));
-  HIP_CHECK(hipMemsetAsync(d_err, 0, sizeof(DevErr), s->stream));
-
-  /* general-kernel scratch: 4096×8 (times) + 4096×8 (values) + 40KB snappy */
   const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960;
-  uint32_t gen_lanes = 0;
-  if (!s->general_ids.empty()) {
-    gen_lanes = (uint32_t)std::min<uint64_t>(s->general_ids.size(), 16384);
-    HIP_CHECK(hipMalloc(&d_scratch, scratch_per_lane * gen_lanes));
+
+  QueryPlan &P = s->plan;
+  if (!P.valid || P.start != start_time || P.end != end_time ||
+      P.interval != interval || P.offset != offset) {
+    free_plan(P);
+    /* host precompute: per-segment window spans, per-series output ranges */
+    P.segq.resize(nsegs);
+    P.sq.resize(s->series_ranges.size());
+    P.partial_slots = 0;
+    P.total_rows = 0;
+    for (size_t g = 0; g < s->series_ranges.size(); g++) {
+      auto &r = s->series_ranges[g];
+      int64_t wmin = INT64_MAX, wmax = INT64_MIN;
+      for (uint32_t i = r.start; i < r.start + r.count; i++) {
+        const gemx_seg_desc &d = s->h_descs[i];
+        int64_t w0 = interval ? win_ordinal(d.min_time, interval, offset) : 0;
+        int64_t w1 = interval ? win_ordinal(d.max_time, interval, offset) : 0;
+        P.segq[i].w_first = w0;
+        P.segq[i].n_wins = (uint32_t)(w1 - w0 + 1);
+        P.segq[i].partial_base = P.partial_slots;
+        P.segq[i].series_idx = (uint32_t)g;
+        P.partial_slots += P.segq[i].n_wins;
+        wmin = std::min(wmin, w0);
+        wmax = std::max(wmax, w1);
+      }
+      P.sq[g].sid = r.sid;
+      P.sq[g].w_min = wmin;
+      P.sq[g].out_base = P.total_rows;
+      P.sq[g].n_wins = (uint32_t)(wmax - wmin + 1);
+      P.sq[g].seg_start = r.start;
+      P.sq[g].seg_count = r.count;
+      P.total_rows += P.sq[g].n_wins;
+    }
+    HIP_CHECK(hipMalloc(&P.d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
+    HIP_CHECK(hipMemcpyAsync(P.d_segq, P.segq.data(), sizeof(SegQ) * nsegs,
+                             hipMemcpyHostToDevice, s->stream));
+    HIP_CHECK(hipMalloc(&P.d_sq,
+                        sizeof(SeriesQ) * (P.sq.empty() ? 1 : P.sq.size())));
+    HIP_CHECK(hipMemcpyAsync(P.d_sq, P.sq.data(), sizeof(SeriesQ) * P.sq.size(),
+                             hipMemcpyHostToDevice, s->stream));
+    HIP_CHECK(hipMalloc(&P.d_part,
+                        sizeof(Partial) * (P.partial_slots ? P.partial_slots : 1)));
+    HIP_CHECK(hipMalloc(&P.d_rows,
+                        sizeof(gemx_agg_row) * (P.total_rows ? P.total_rows : 1)));
+    HIP_CHECK(hipMalloc(&P.d_err, sizeof(DevErr)));
+    HIP_CHECK(hipHostMalloc(&P.h_rows,
+                            sizeof(gemx_agg_row) * (P.total_rows ? P.total_rows : 1)));
+    if (!s->general_ids.empty()) {
+      P.gen_lanes = (uint32_t)std::min<uint64_t>(s->general_ids.size(), 16384);
+      HIP_CHECK(hipMalloc(&P.d_scratch, scratch_per_lane * P.gen_lanes));
+    }
+    P.start = start_time;
+    P.end = end_time;
+    P.interval = interval;
+    P.offset = offset;
+    P.valid = true;
   }
+  SegQ *d_segq = P.d_segq;
+  SeriesQ *d_sq = P.d_sq;
+  Partial *d_part = P.d_part;
+  gemx_agg_row *d_rows = P.d_rows;
+  DevErr *d_err = P.d_err;
+  uint8_t *d_scratch = P.d_scratch;
+  uint32_t gen_lanes = P.gen_lanes;
+  const uint64_t partial_slots = P.partial_slots;
+  const uint64_t total_rows = P.total_rows;
+  std::vector<SeriesQ> &sq = P.sq;
+
+  HIP_CHECK(hipMemsetAsync(d_part, 0, sizeof(Partial) * partial_slots, s->stream));
+  HIP_CHECK(hipMemsetAsync(d_err, 0, sizeof(DevErr), s->stream));
 
   hipEvent_t ev0, ev1, ev2;
   HIP_CHECK(hipEventCreate(&ev0));
@@ -1502,12 +1558,12 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
-  /* check device error + fetch rows */
+  /* check device error + fetch rows (pinned staging) */
   DevErr herr = {0};
   HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
                            s->stream));
-  std::vector<gemx_agg_row> hrows(total_rows);
-  HIP_CHECK(hipMemcpyAsync(hrows.data(), d_rows, sizeof(gemx_agg_row) * total_rows,
+  gemx_agg_row *hrows = P.h_rows;
+  HIP_CHECK(hipMemcpyAsync(hrows, d_rows, sizeof(gemx_agg_row) * total_rows,
                            hipMemcpyDeviceToHost, s->stream));
   HIP_CHECK(hipStreamSynchronize(s->stream));
 
@@ -1518,13 +1574,6 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
   hipEventDestroy(ev0);
   hipEventDestroy(ev1);
   hipEventDestroy(ev2);
-
-  hipFree(d_segq);
-  hipFree(d_sq);
-  hipFree(d_part);
-  hipFree(d_rows);
-  hipFree(d_err);
-  if (d_scratch) hipFree(d_scratch);
 
   if (herr.code != 0) {
     seterr(herr.code == GEMX_E_UNSUPPORTED ? "unsupported codec on device"
