@@ -134,7 +134,7 @@ def main():
     w0 = 0  # t0=0, windows start at ordinal 0
 
     def step():
-        rows, stats = shard.scan_agg(0, 2**62, WINDOW_NS)
+        rows, stats = shard.scan_agg(0, 2**62, WINDOW_NS, group_all=True)
         if dist_on:
             from opengemini_amd.dist import window_partials, merge_across_shards
 

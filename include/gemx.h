@@ -131,6 +131,18 @@ int gemx_scan_agg(gemx_shard *, int64_t start_time, int64_t end_time,
                   int64_t interval, int64_t offset, gemx_agg_row *out_host,
                   uint64_t cap, uint64_t *n_out, gemx_query_stats *stats);
 
+/* Same scan, plus the on-device cross-series merge into the all-series
+ * `GROUP BY time` group (AggTagSetCursor.UpdateRec,
+ * engine/agg_tagset_cursor.go:1111-1122 + lib/record/reccord_functions.go):
+ * one output row per window (sid=0), min/max by value with earliest-time
+ * tie-break, first=min-time / last=max-time, sum/count accumulate.
+ * This is the north-star `SELECT mean(value) ... GROUP BY time(1m)` shape:
+ * only windows cross the boundary, not per-series partials. */
+int gemx_scan_agg_grouped(gemx_shard *, int64_t start_time, int64_t end_time,
+                          int64_t interval, int64_t offset,
+                          gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                          gemx_query_stats *stats);
+
 #ifdef __cplusplus
 }
 #endif

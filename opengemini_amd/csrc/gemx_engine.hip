@@ -1205,6 +1205,201 @@ __global__ void __launch_bounds__(256) k_merge(
   }
 }
 
+/* ---------------- cross-series group merge kernel ---------------- */
+/* AggTagSetCursor.UpdateRec for the all-series `GROUP BY time` group
+ * (engine/agg_tagset_cursor.go:1111-1122; lib/record/reccord_functions.go:
+ *  UpdateFloatMin:474 min by value, tie → smaller time; UpdateFloatMax:500;
+ *  first = min time / last = max time, ties keep first-processed :126-148;
+ *  sum/count accumulate :722-757). One block per window; 256 threads stride
+ *  the series deterministically, then an LDS tree with left-precedence
+ *  merges thread partials — bit-exact for count/min/max/first/last, fixed
+ *  reassociation for float sum (within the 1e-9 contract). */
+struct GAcc {
+  int64_t count;
+  gemx_val sum, minv, maxv, firstv, lastv;
+  int64_t min_t, max_t, first_t, last_t;
+  uint32_t active; /* bit1 sum, bit2 min, bit3 max, bit4 first, bit5 last */
+  uint32_t used;
+};
+
+template <int COLTYPE>
+__device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r) {
+  a->used = 1;
+  a->count += r->count;
+  if (!r->sum_isnil) {
+    if (!(a->active & 2)) {
+      a->sum = r->sum;
+      a->active |= 2;
+    } else if (COLTYPE == GEMX_TYPE_FLOAT)
+      a->sum.f += r->sum.f;
+    else
+      a->sum.i += r->sum.i;
+  }
+  if (!r->min_isnil) {
+    int take = !(a->active & 4);
+    if (!take) {
+      take = (COLTYPE == GEMX_TYPE_FLOAT)
+                 ? (r->minv.f < a->minv.f ||
+                    (r->minv.f == a->minv.f && r->min_time < a->min_t))
+                 : (r->minv.i < a->minv.i ||
+                    (r->minv.i == a->minv.i && r->min_time < a->min_t));
+    }
+    if (take) {
+      a->minv = r->minv;
+      a->min_t = r->min_time;
+      a->active |= 4;
+    }
+  }
+  if (!r->max_isnil) {
+    int take = !(a->active & 8);
+    if (!take) {
+      take = (COLTYPE == GEMX_TYPE_FLOAT)
+                 ? (r->maxv.f > a->maxv.f ||
+                    (r->maxv.f == a->maxv.f && r->max_time < a->max_t))
+                 : (r->maxv.i > a->maxv.i ||
+                    (r->maxv.i == a->maxv.i && r->max_time < a->max_t));
+    }
+    if (take) {
+      a->maxv = r->maxv;
+      a->max_t = r->max_time;
+      a->active |= 8;
+    }
+  }
+  if (!r->first_isnil) {
+    if (!(a->active & 16) || r->first_time < a->first_t) {
+      a->firstv = r->firstv;
+      a->first_t = r->first_time;
+      a->active |= 16;
+    }
+  }
+  if (!r->last_isnil) {
+    if (!(a->active & 32) || r->last_time > a->last_t) {
+      a->lastv = r->lastv;
+      a->last_t = r->last_time;
+      a->active |= 32;
+    }
+  }
+}
+
+template <int COLTYPE>
+__device__ __forceinline__ void gacc_merge(GAcc *l, const GAcc *r) {
+  /* left precedes right in series order; ties keep left */
+  if (!r->used) return;
+  l->used = 1;
+  l->count += r->count;
+  if (r->active & 2) {
+    if (!(l->active & 2)) {
+      l->sum = r->sum;
+      l->active |= 2;
+    } else if (COLTYPE == GEMX_TYPE_FLOAT)
+      l->sum.f += r->sum.f;
+    else
+      l->sum.i += r->sum.i;
+  }
+  if (r->active & 4) {
+    int take = !(l->active & 4);
+    if (!take)
+      take = (COLTYPE == GEMX_TYPE_FLOAT)
+                 ? (r->minv.f < l->minv.f ||
+                    (r->minv.f == l->minv.f && r->min_t < l->min_t))
+                 : (r->minv.i < l->minv.i ||
+                    (r->minv.i == l->minv.i && r->min_t < l->min_t));
+    if (take) {
+      l->minv = r->minv;
+      l->min_t = r->min_t;
+      l->active |= 4;
+    }
+  }
+  if (r->active & 8) {
+    int take = !(l->active & 8);
+    if (!take)
+      take = (COLTYPE == GEMX_TYPE_FLOAT)
+                 ? (r->maxv.f > l->maxv.f ||
+                    (r->maxv.f == l->maxv.f && r->max_t < l->max_t))
+                 : (r->maxv.i > l->maxv.i ||
+                    (r->maxv.i == l->maxv.i && r->max_t < l->max_t));
+    if (take) {
+      l->maxv = r->maxv;
+      l->max_t = r->max_t;
+      l->active |= 8;
+    }
+  }
+  if (r->active & 16) {
+    if (!(l->active & 16) || r->first_t < l->first_t) {
+      l->firstv = r->firstv;
+      l->first_t = r->first_t;
+      l->active |= 16;
+    }
+  }
+  if (r->active & 32) {
+    if (!(l->active & 32) || r->last_t > l->last_t) {
+      l->lastv = r->lastv;
+      l->last_t = r->last_t;
+      l->active |= 32;
+    }
+  }
+}
+
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_group(
+    const SeriesQ *__restrict__ series, uint32_t nseries,
+    const gemx_agg_row *__restrict__ rows, gemx_agg_row *__restrict__ out,
+    int64_t W0, uint32_t n_gwins, int64_t interval, int64_t offset,
+    int64_t q_start) {
+  __shared__ GAcc sh[256];
+  for (uint32_t wb = blockIdx.x; wb < n_gwins; wb += gridDim.x) {
+    int64_t w = W0 + (int64_t)wb;
+    GAcc a;
+    memset(&a, 0, sizeof(a));
+    for (uint32_t g = threadIdx.x; g < nseries; g += blockDim.x) {
+      int64_t local = w - series[g].w_min;
+      if (local < 0 || local >= (int64_t)series[g].n_wins) continue;
+      const gemx_agg_row *r = &rows[series[g].out_base + local];
+      if (r->count < 0) continue; /* gap marker */
+      gacc_row<COLTYPE>(&a, r);
+    }
+    sh[threadIdx.x] = a;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+      if (threadIdx.x < (uint32_t)s)
+        gacc_merge<COLTYPE>(&sh[threadIdx.x], &sh[threadIdx.x + s]);
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+      GAcc *g = &sh[0];
+      gemx_agg_row o;
+      memset(&o, 0, sizeof(o));
+      o.sid = 0;
+      int64_t ws = interval ? win_start_of(w, interval, offset) : q_start;
+      o.win_start = ws;
+      if (!g->used) {
+        o.count = -1; /* gap */
+      } else {
+        o.first_row_time = ws; /* BuildEmptyIntervalRec interval times */
+        o.count = g->count;
+        o.count_time = ws;
+        o.sum = g->sum;
+        o.sum_time = ws;
+        o.sum_isnil = !(g->active & 2);
+        o.minv = g->minv;
+        o.min_time = g->min_t;
+        o.min_isnil = !(g->active & 4);
+        o.maxv = g->maxv;
+        o.max_time = g->max_t;
+        o.max_isnil = !(g->active & 8);
+        o.firstv = g->firstv;
+        o.first_time = g->first_t;
+        o.first_isnil = !(g->active & 16);
+        o.lastv = g->lastv;
+        o.last_time = g->last_t;
+        o.last_isnil = !(g->active & 32);
+      }
+      out[wb] = o;
+    }
+    __syncthreads();
+  }
+}
+
 /* ---------------- host: engine ---------------- */
 
 static __thread char g_err[512];
@@ -1241,6 +1436,11 @@ struct QueryPlan {
   gemx_agg_row *h_rows = nullptr; /* pinned staging */
   uint8_t *d_scratch = nullptr;
   uint32_t gen_lanes = 0;
+  /* grouped output (all-series GROUP BY time) */
+  gemx_agg_row *d_grows = nullptr;
+  gemx_agg_row *h_grows = nullptr;
+  int64_t W0 = 0;
+  uint64_t n_gwins = 0;
 };
 
 struct gemx_shard {
@@ -1273,6 +1473,8 @@ static void free_plan(QueryPlan &p) {
   if (p.d_err) hipFree(p.d_err);
   if (p.d_scratch) hipFree(p.d_scratch);
   if (p.h_rows) hipHostFree(p.h_rows);
+  if (p.d_grows) hipFree(p.d_grows);
+  if (p.h_grows) hipHostFree(p.h_grows);
   p = QueryPlan();
 }
 
@@ -1432,10 +1634,10 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   return GEMX_OK;
 }
 
-extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time,
-                             int64_t interval, int64_t offset,
-                             gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
-                             gemx_query_stats *stats) {
+static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
+                     int64_t interval, int64_t offset, int group_all,
+                     gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                     gemx_query_stats *stats) {
   if (!s) return GEMX_E_INVALID;
   HIP_CHECK(hipSetDevice(s->device));
   const uint64_t nsegs = s->nsegs;
@@ -1490,6 +1692,20 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
     if (!s->general_ids.empty()) {
       P.gen_lanes = (uint32_t)std::min<uint64_t>(s->general_ids.size(), 16384);
       HIP_CHECK(hipMalloc(&P.d_scratch, scratch_per_lane * P.gen_lanes));
+    }
+    /* global window range for the grouped output */
+    {
+      int64_t W0 = INT64_MAX, W1 = INT64_MIN;
+      for (auto &g : P.sq) {
+        W0 = std::min(W0, g.w_min);
+        W1 = std::max(W1, g.w_min + (int64_t)g.n_wins - 1);
+      }
+      P.W0 = P.sq.empty() ? 0 : W0;
+      P.n_gwins = P.sq.empty() ? 0 : (uint64_t)(W1 - W0 + 1);
+      HIP_CHECK(hipMalloc(&P.d_grows,
+                          sizeof(gemx_agg_row) * (P.n_gwins ? P.n_gwins : 1)));
+      HIP_CHECK(hipHostMalloc(&P.h_grows,
+                              sizeof(gemx_agg_row) * (P.n_gwins ? P.n_gwins : 1)));
     }
     P.start = start_time;
     P.end = end_time;
@@ -1556,15 +1772,36 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part, d_rows,
                          total_rows, interval, offset, start_time);
   }
+  if (group_all && P.n_gwins > 0) {
+    uint32_t blocks = (uint32_t)std::min<uint64_t>(P.n_gwins, 65535);
+    if (s->col_type == GEMX_TYPE_FLOAT)
+      hipLaunchKernelGGL((k_group<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(256), 0,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_rows, P.d_grows,
+                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time);
+    else
+      hipLaunchKernelGGL((k_group<GEMX_TYPE_INT>), dim3(blocks), dim3(256), 0,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_rows, P.d_grows,
+                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time);
+  }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
   /* check device error + fetch rows (pinned staging) */
   DevErr herr = {0};
   HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
                            s->stream));
-  gemx_agg_row *hrows = P.h_rows;
-  HIP_CHECK(hipMemcpyAsync(hrows, d_rows, sizeof(gemx_agg_row) * total_rows,
-                           hipMemcpyDeviceToHost, s->stream));
+  gemx_agg_row *hrows;
+  uint64_t fetch_rows;
+  if (group_all) {
+    hrows = P.h_grows;
+    fetch_rows = P.n_gwins;
+    HIP_CHECK(hipMemcpyAsync(hrows, P.d_grows, sizeof(gemx_agg_row) * fetch_rows,
+                             hipMemcpyDeviceToHost, s->stream));
+  } else {
+    hrows = P.h_rows;
+    fetch_rows = total_rows;
+    HIP_CHECK(hipMemcpyAsync(hrows, d_rows, sizeof(gemx_agg_row) * fetch_rows,
+                             hipMemcpyDeviceToHost, s->stream));
+  }
   HIP_CHECK(hipStreamSynchronize(s->stream));
 
   float ms_scan = 0, ms_merge = 0, ms_total = 0;
@@ -1583,7 +1820,7 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
 
   /* compact gap rows (count == -1) into caller buffer */
   uint64_t n = 0;
-  for (uint64_t i = 0; i < total_rows; i++) {
+  for (uint64_t i = 0; i < fetch_rows; i++) {
     if (hrows[i].count < 0) continue;
     if (n >= cap) {
       seterr("output capacity too small");
@@ -1605,4 +1842,23 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
     stats->h2d_ms = 0;
   }
   return GEMX_OK;
+}
+
+extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time,
+                             int64_t interval, int64_t offset,
+                             gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                             gemx_query_stats *stats) {
+  return scan_impl(s, start_time, end_time, interval, offset, 0, out_host, cap,
+                   n_out, stats);
+}
+
+/* all-series GROUP BY time: adds the on-device AggTagSetCursor merge and
+ * returns one row per window (engine/agg_tagset_cursor.go:1111 UpdateRec) */
+extern "C" int gemx_scan_agg_grouped(gemx_shard *s, int64_t start_time,
+                                     int64_t end_time, int64_t interval,
+                                     int64_t offset, gemx_agg_row *out_host,
+                                     uint64_t cap, uint64_t *n_out,
+                                     gemx_query_stats *stats) {
+  return scan_impl(s, start_time, end_time, interval, offset, 1, out_host, cap,
+                   n_out, stats);
 }

@@ -120,11 +120,14 @@ def _load():
     ]
     lib.gemx_shard_close.restype = C.c_int
     lib.gemx_shard_close.argtypes = [C.c_void_p]
-    lib.gemx_scan_agg.restype = C.c_int
-    lib.gemx_scan_agg.argtypes = [
+    scan_sig = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_scan_agg.restype = C.c_int
+    lib.gemx_scan_agg.argtypes = scan_sig
+    lib.gemx_scan_agg_grouped.restype = C.c_int
+    lib.gemx_scan_agg_grouped.argtypes = scan_sig
     _lib = lib
     return lib
 
@@ -184,17 +187,38 @@ class Shard:
         except Exception:
             pass
 
-    def scan_agg(self, start_time, end_time, interval, offset=0, out_cap=None):
-        """One fused scan: all six aggregates per (sid, GROUP BY time window).
+    def _rows_bound(self, interval, offset, grouped):
+        """Exact upper bound on output rows from the descriptors."""
+        d = self._descs
+        if interval == 0:
+            n_sids = len(np.unique(d["sid"]))
+            return (1 if grouped else n_sids) + 4
+        w0 = (d["min_time"] - offset) // interval
+        w1 = (d["max_time"] - offset) // interval
+        if grouped:
+            return int(w1.max() - w0.min() + 1) + 4
+        # per-series window span: group consecutive equal sids
+        change = np.nonzero(np.diff(d["sid"]))[0] + 1
+        starts = np.concatenate([[0], change]).astype(np.int64)
+        w1max = np.maximum.reduceat(w1, starts)
+        w0min = np.minimum.reduceat(w0, starts)
+        return int((w1max - w0min + 1).sum()) + 4
+
+    def scan_agg(self, start_time, end_time, interval, offset=0, out_cap=None,
+                 group_all=False):
+        """One fused scan: all six aggregates per GROUP BY time window —
+        per-series rows, or (group_all=True) merged across all series
+        on-device (the AggTagSetCursor path, the north-star query shape).
 
         Returns (rows: np.ndarray[AGG_ROW_DTYPE], stats dict)."""
         lib = self._lib
         if out_cap is None:
-            out_cap = self.n_points + len(self._descs) + 16
-        out = np.zeros(out_cap, dtype=AGG_ROW_DTYPE)
+            out_cap = self._rows_bound(interval, offset, group_all)
+        out = np.empty(out_cap, dtype=AGG_ROW_DTYPE)
         n = C.c_uint64(0)
         st = _Stats()
-        rc = lib.gemx_scan_agg(
+        fn = lib.gemx_scan_agg_grouped if group_all else lib.gemx_scan_agg
+        rc = fn(
             self._h, start_time, end_time, interval, offset,
             out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
         )

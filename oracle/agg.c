@@ -690,3 +690,127 @@ int64_t orc_agg_cursor(int col_type, uint32_t op, int multi_call,
   return nout;
 #undef EMIT
 }
+
+/* ---------------- cross-series group merge ---------------- */
+/* AggTagSetCursor.UpdateRec semantics (see oracle.h declaration). */
+
+typedef struct {
+  int64_t win_start;
+  int used;
+  int64_t count;
+  int sum_active, min_active, max_active, first_active, last_active;
+  orc_val sum, minv, maxv, firstv, lastv;
+  int64_t min_t, max_t, first_t, last_t;
+} groupacc;
+
+static void group_update(groupacc *g, const orc_agg_row *r, int col_type) {
+  g->used = 1;
+  /* count: UpdateCount accumulates; nil contributes nothing (:740-745) */
+  g->count += r->count;
+  if (!r->sum_isnil) {
+    if (!g->sum_active) {
+      g->sum = r->sum;
+      g->sum_active = 1;
+    } else if (col_type == ORC_TYPE_FLOAT)
+      g->sum.f += r->sum.f;
+    else
+      g->sum.i += r->sum.i;
+  }
+  if (!r->min_isnil) {
+    int take;
+    if (!g->min_active)
+      take = 1;
+    else if (col_type == ORC_TYPE_FLOAT)
+      take = (r->minv.f < g->minv.f) ||
+             (r->minv.f == g->minv.f && r->min_time < g->min_t);
+    else
+      take = (r->minv.i < g->minv.i) ||
+             (r->minv.i == g->minv.i && r->min_time < g->min_t);
+    if (take) {
+      g->minv = r->minv;
+      g->min_t = r->min_time;
+      g->min_active = 1;
+    }
+  }
+  if (!r->max_isnil) {
+    int take;
+    if (!g->max_active)
+      take = 1;
+    else if (col_type == ORC_TYPE_FLOAT)
+      take = (r->maxv.f > g->maxv.f) ||
+             (r->maxv.f == g->maxv.f && r->max_time < g->max_t);
+    else
+      take = (r->maxv.i > g->maxv.i) ||
+             (r->maxv.i == g->maxv.i && r->max_time < g->max_t);
+    if (take) {
+      g->maxv = r->maxv;
+      g->max_t = r->max_time;
+      g->max_active = 1;
+    }
+  }
+  if (!r->first_isnil) {
+    if (!g->first_active || r->first_time < g->first_t) {
+      g->firstv = r->firstv;
+      g->first_t = r->first_time;
+      g->first_active = 1;
+    }
+  }
+  if (!r->last_isnil) {
+    if (!g->last_active || r->last_time > g->last_t) {
+      g->lastv = r->lastv;
+      g->last_t = r->last_time;
+      g->last_active = 1;
+    }
+  }
+}
+
+int64_t orc_group_merge(const orc_agg_row *rows, int64_t n, int col_type,
+                        int64_t interval, orc_agg_row *out, int64_t cap) {
+  if (n == 0) return 0;
+  /* global window range */
+  int64_t wmin = INT64_MAX, wmax = INT64_MIN;
+  for (int64_t i = 0; i < n; i++) {
+    if (rows[i].win_start < wmin) wmin = rows[i].win_start;
+    if (rows[i].win_start > wmax) wmax = rows[i].win_start;
+  }
+  int64_t n_wins = interval ? (wmax - wmin) / interval + 1 : 1;
+  groupacc *g = (groupacc *)calloc((size_t)n_wins, sizeof(groupacc));
+  if (!g) return -1;
+  for (int64_t i = 0; i < n; i++) {
+    int64_t idx = interval ? (rows[i].win_start - wmin) / interval : 0;
+    g[idx].win_start = rows[i].win_start;
+    group_update(&g[idx], &rows[i], col_type);
+  }
+  int64_t m = 0;
+  for (int64_t w = 0; w < n_wins; w++) {
+    if (!g[w].used) continue;
+    if (m >= cap) {
+      free(g);
+      return -1;
+    }
+    orc_agg_row *o = &out[m++];
+    memset(o, 0, sizeof(*o));
+    o->sid = 0;
+    o->win_start = g[w].win_start;
+    o->first_row_time = g[w].win_start;
+    o->count = g[w].count;
+    o->count_time = g[w].win_start;
+    o->sum = g[w].sum;
+    o->sum_time = g[w].win_start;
+    o->sum_isnil = !g[w].sum_active;
+    o->minv = g[w].minv;
+    o->min_time = g[w].min_t;
+    o->min_isnil = !g[w].min_active;
+    o->maxv = g[w].maxv;
+    o->max_time = g[w].max_t;
+    o->max_isnil = !g[w].max_active;
+    o->firstv = g[w].firstv;
+    o->first_time = g[w].first_t;
+    o->first_isnil = !g[w].first_active;
+    o->lastv = g[w].lastv;
+    o->last_time = g[w].last_t;
+    o->last_isnil = !g[w].last_active;
+  }
+  free(g);
+  return m;
+}

@@ -123,6 +123,8 @@ def load():
     lib.orc_scan_agg.argtypes = [u8p, i64, C.c_void_p, i64, C.c_int, i64, i64, i64, i64, C.c_void_p, i64]
     lib.orc_scan_agg_mt.restype = i64
     lib.orc_scan_agg_mt.argtypes = [u8p, i64, C.c_void_p, i64, C.c_int, i64, i64, i64, i64, C.c_void_p, i64, C.c_int]
+    lib.orc_group_merge.restype = i64
+    lib.orc_group_merge.argtypes = [C.c_void_p, i64, C.c_int, i64, C.c_void_p, i64]
     lib.orc_gen_shard.restype = i64
     lib.orc_gen_shard.argtypes = [
         C.c_uint64, C.c_uint64, C.c_uint64, C.c_uint32, i64, i64, C.c_int,
@@ -321,6 +323,21 @@ def scan_agg(blob, descs, col_type, start_time, end_time, interval, offset=0, ou
     )
     if n < 0:
         raise ValueError("scan_agg failed")
+    return out[:n].copy()
+
+
+def group_merge(rows, col_type, interval):
+    """AggTagSetCursor.UpdateRec group merge over per-(sid,window) rows."""
+    lib = get()
+    r = np.ascontiguousarray(rows, dtype=AGG_ROW_DTYPE)
+    cap = len(r) + 4
+    out = np.zeros(cap, dtype=AGG_ROW_DTYPE)
+    n = lib.orc_group_merge(
+        r.ctypes.data_as(C.c_void_p), len(r), col_type, interval,
+        out.ctypes.data_as(C.c_void_p), cap,
+    )
+    if n < 0:
+        raise ValueError("group_merge failed")
     return out[:n].copy()
 
 

@@ -186,6 +186,20 @@ int64_t orc_scan_agg(const uint8_t *blob, int64_t blob_len,
                      int64_t start_time, int64_t end_time, int64_t interval,
                      int64_t offset, orc_agg_row *out, int64_t out_cap);
 
+/* Merge per-(sid,window) rows into per-window group rows — the
+ * AggTagSetCursor.UpdateRec semantics for the all-series group of
+ * `GROUP BY time(w)` (engine/agg_tagset_cursor.go:1111-1122,
+ *  lib/record/reccord_functions.go: UpdateFloatMin:474 —
+ *  min by value, tie → smaller time, tie → first-processed;
+ *  UpdateFloatMax:500 symmetric; first = min time / last = max time,
+ *  ties keep first-processed (:126-148); sum/count accumulate (:722-757)).
+ * rows must be ordered by (sid, win_start) — scan_agg's output order = the
+ * reference's series iteration order. Output rows ascending by win_start,
+ * sid=0; count_time/sum_time/first_row_time = window start
+ * (BuildEmptyIntervalRec interval times). Returns out rows. */
+int64_t orc_group_merge(const orc_agg_row *rows, int64_t n, int col_type,
+                        int64_t interval, orc_agg_row *out, int64_t cap);
+
 /* Multi-threaded variant (OpenMP over series groups) — the bench.py
  * cpu_baseline leg. nthreads<=0 → all cores. */
 int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,

@@ -178,3 +178,60 @@ class TestEngineGuards:
         assert stats["points"] == 100 * 1000
         assert stats["decode_ms"] > 0
         assert stats["compressed_bytes"] == len(blob)
+
+
+class TestGroupedParity:
+    """gemx_scan_agg_grouped vs oracle scan_agg + orc_group_merge."""
+
+    def _run(self, blob, descs, col_type, interval=INT):
+        import opengemini_amd as gx
+
+        sh = gx.Shard(blob, descs, col_type)
+        try:
+            gpu_rows, stats = sh.scan_agg(0, 2**62, interval, group_all=True)
+        finally:
+            sh.close()
+        base = orc.scan_agg(blob, descs, col_type, 0, 2**62, interval)
+        ref = orc.group_merge(base, col_type, interval)
+        assert len(gpu_rows) == len(ref)
+        for f in ("win_start", "count", "min_time", "max_time", "first_time",
+                  "last_time", "min_isnil", "max_isnil", "first_isnil",
+                  "last_isnil", "sum_isnil"):
+            assert np.array_equal(gpu_rows[f], ref[f]), f
+        for f in ("min", "max", "first", "last"):
+            assert np.array_equal(
+                gpu_rows[f].view(np.uint64), ref[f].view(np.uint64)
+            ), f
+        if col_type == F:
+            tol = 1e-9 * np.maximum(1.0, np.abs(ref["sum"]))
+            both_nan = np.isnan(gpu_rows["sum"]) & np.isnan(ref["sum"])
+            assert np.all(both_nan | (np.abs(gpu_rows["sum"] - ref["sum"]) <= tol))
+        else:
+            assert np.array_equal(
+                gpu_rows["sum"].view(np.int64), ref["sum"].view(np.int64)
+            )
+
+    def test_grouped_float(self):
+        rng = np.random.default_rng(50)
+        blob, d, _ = build_shard(rng, F, list(range(1, 40)))
+        self._run(blob, d, F)
+
+    def test_grouped_int(self):
+        rng = np.random.default_rng(51)
+        blob, d, _ = build_shard(rng, I, list(range(1, 20)))
+        self._run(blob, d, I)
+
+    def test_grouped_bulk(self):
+        blob, descs = orc.gen_shard(46, 2000, 1000)
+        self._run(blob, descs, F)
+
+    def test_grouped_value_ties(self):
+        # equal values across series: earliest-time, then first-series wins
+        rng = np.random.default_rng(52)
+        blob, d, _ = build_shard(
+            rng, F, list(range(1, 30)), null_frac=0.0,
+            value_fn=lambda r, n: np.repeat(
+                r.integers(0, 3, max(1, n // 50 + 1)).astype(np.float64), 50
+            )[:n],
+        )
+        self._run(blob, d, F)
