@@ -1218,12 +1218,17 @@ struct GAcc {
   int64_t count;
   gemx_val sum, minv, maxv, firstv, lastv;
   int64_t min_t, max_t, first_t, last_t;
+  /* series-order index of each candidate: the reference processes series
+   * in sid order and ties keep the first-processed — the tree reduce
+   * resolves full ties by the smaller source index */
+  uint32_t min_src, max_src, first_src, last_src;
   uint32_t active; /* bit1 sum, bit2 min, bit3 max, bit4 first, bit5 last */
   uint32_t used;
 };
 
 template <int COLTYPE>
-__device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r) {
+__device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r,
+                                         uint32_t src) {
   a->used = 1;
   a->count += r->count;
   if (!r->sum_isnil) {
@@ -1247,6 +1252,7 @@ __device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r) {
     if (take) {
       a->minv = r->minv;
       a->min_t = r->min_time;
+      a->min_src = src;
       a->active |= 4;
     }
   }
@@ -1262,6 +1268,7 @@ __device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r) {
     if (take) {
       a->maxv = r->maxv;
       a->max_t = r->max_time;
+      a->max_src = src;
       a->active |= 8;
     }
   }
@@ -1269,6 +1276,7 @@ __device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r) {
     if (!(a->active & 16) || r->first_time < a->first_t) {
       a->firstv = r->firstv;
       a->first_t = r->first_time;
+      a->first_src = src;
       a->active |= 16;
     }
   }
@@ -1276,6 +1284,7 @@ __device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r) {
     if (!(a->active & 32) || r->last_time > a->last_t) {
       a->lastv = r->lastv;
       a->last_t = r->last_time;
+      a->last_src = src;
       a->active |= 32;
     }
   }
@@ -1283,7 +1292,9 @@ __device__ __forceinline__ void gacc_row(GAcc *a, const gemx_agg_row *r) {
 
 template <int COLTYPE>
 __device__ __forceinline__ void gacc_merge(GAcc *l, const GAcc *r) {
-  /* left precedes right in series order; ties keep left */
+  /* associative + commutative: full ties resolved by smaller source index
+   * (= the reference's first-processed-series-wins,
+   *  reccord_functions.go:489 `srcVal == v && t1 <= t2 → keep`) */
   if (!r->used) return;
   l->used = 1;
   l->count += r->count;
@@ -1301,12 +1312,17 @@ __device__ __forceinline__ void gacc_merge(GAcc *l, const GAcc *r) {
     if (!take)
       take = (COLTYPE == GEMX_TYPE_FLOAT)
                  ? (r->minv.f < l->minv.f ||
-                    (r->minv.f == l->minv.f && r->min_t < l->min_t))
+                    (r->minv.f == l->minv.f &&
+                     (r->min_t < l->min_t ||
+                      (r->min_t == l->min_t && r->min_src < l->min_src))))
                  : (r->minv.i < l->minv.i ||
-                    (r->minv.i == l->minv.i && r->min_t < l->min_t));
+                    (r->minv.i == l->minv.i &&
+                     (r->min_t < l->min_t ||
+                      (r->min_t == l->min_t && r->min_src < l->min_src))));
     if (take) {
       l->minv = r->minv;
       l->min_t = r->min_t;
+      l->min_src = r->min_src;
       l->active |= 4;
     }
   }
@@ -1315,26 +1331,35 @@ __device__ __forceinline__ void gacc_merge(GAcc *l, const GAcc *r) {
     if (!take)
       take = (COLTYPE == GEMX_TYPE_FLOAT)
                  ? (r->maxv.f > l->maxv.f ||
-                    (r->maxv.f == l->maxv.f && r->max_t < l->max_t))
+                    (r->maxv.f == l->maxv.f &&
+                     (r->max_t < l->max_t ||
+                      (r->max_t == l->max_t && r->max_src < l->max_src))))
                  : (r->maxv.i > l->maxv.i ||
-                    (r->maxv.i == l->maxv.i && r->max_t < l->max_t));
+                    (r->maxv.i == l->maxv.i &&
+                     (r->max_t < l->max_t ||
+                      (r->max_t == l->max_t && r->max_src < l->max_src))));
     if (take) {
       l->maxv = r->maxv;
       l->max_t = r->max_t;
+      l->max_src = r->max_src;
       l->active |= 8;
     }
   }
   if (r->active & 16) {
-    if (!(l->active & 16) || r->first_t < l->first_t) {
+    if (!(l->active & 16) || r->first_t < l->first_t ||
+        (r->first_t == l->first_t && r->first_src < l->first_src)) {
       l->firstv = r->firstv;
       l->first_t = r->first_t;
+      l->first_src = r->first_src;
       l->active |= 16;
     }
   }
   if (r->active & 32) {
-    if (!(l->active & 32) || r->last_t > l->last_t) {
+    if (!(l->active & 32) || r->last_t > l->last_t ||
+        (r->last_t == l->last_t && r->last_src < l->last_src)) {
       l->lastv = r->lastv;
       l->last_t = r->last_t;
+      l->last_src = r->last_src;
       l->active |= 32;
     }
   }
@@ -1356,7 +1381,7 @@ __global__ void __launch_bounds__(256) k_group(
       if (local < 0 || local >= (int64_t)series[g].n_wins) continue;
       const gemx_agg_row *r = &rows[series[g].out_base + local];
       if (r->count < 0) continue; /* gap marker */
-      gacc_row<COLTYPE>(&a, r);
+      gacc_row<COLTYPE>(&a, r, g);
     }
     sh[threadIdx.x] = a;
     __syncthreads();
