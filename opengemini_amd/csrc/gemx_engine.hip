@@ -49,10 +49,9 @@ __device__ __forceinline__ uint32_t d_u32be(const uint8_t *p) {
          p[3];
 }
 __device__ __forceinline__ uint64_t d_u64be(const uint8_t *p) {
-  uint64_t v = 0;
-#pragma unroll
-  for (int i = 0; i < 8; i++) v = (v << 8) | p[i];
-  return v;
+  uint64_t v;
+  memcpy(&v, p, 8); /* unaligned ok on CDNA */
+  return __builtin_bswap64(v);
 }
 __device__ __forceinline__ uint64_t d_u64le(const uint8_t *p) {
   uint64_t v;
@@ -101,40 +100,65 @@ __device__ __host__ __forceinline__ int64_t win_start_of(int64_t ord, int64_t in
 
 /* ---------------- device: bit reader (batch_float.go bit cursor) --------- */
 
+/* 128-bit-window MSB-first bit reader: valid bits are the top `have` bits
+ * of (hi,lo); a read of <=64 bits never straddles a refill boundary. The
+ * Gorilla inner loop (1+1+11+64 bits worst case per value) touches memory
+ * via one aligned-8B-swapped load per 64 consumed bits instead of
+ * byte-at-a-time (tsm1 batch_float.go reader, CDNA-friendly shape). */
 struct BitR {
   const uint8_t *b;
   int64_t len, pos;
-  uint64_t cur;
-  int nbits;
-  __device__ int refill() {
-    if (len - pos >= 8) {
-      cur = d_u64be(b + pos);
-      nbits = 64;
-      pos += 8;
-      return 0;
+  uint64_t hi, lo;
+  int have;
+
+  __device__ __forceinline__ void fill() {
+    while (have <= 64 && pos < len) {
+      uint64_t w;
+      int wbits;
+      if (len - pos >= 8) {
+        w = d_u64be(b + pos);
+        pos += 8;
+        wbits = 64;
+      } else {
+        int rem = (int)(len - pos);
+        w = 0;
+        for (int i = 0; i < rem; i++) w = (w << 8) | b[pos + i];
+        w <<= (8 - rem) * 8;
+        pos = len;
+        wbits = rem * 8;
+      }
+      if (have == 0) {
+        hi = w;
+        lo = 0;
+      } else if (have < 64) {
+        hi |= w >> have;
+        lo = w << (64 - have);
+      } else {
+        lo = w;
+      }
+      have += wbits;
     }
-    int64_t rem = len - pos;
-    if (rem <= 0) return -1;
-    uint64_t v = 0;
-    for (int64_t i = 0; i < rem; i++) v = (v << 8) | b[pos + i];
-    cur = v << (64 - rem * 8);
-    nbits = (int)(rem * 8);
-    pos = len;
-    return 0;
   }
-  __device__ int read(int n, uint64_t *out) {
-    uint64_t v = 0;
-    int got = 0;
-    while (got < n) {
-      if (nbits == 0 && refill()) return -1;
-      int take = n - got;
-      if (take > nbits) take = nbits;
-      uint64_t chunk = cur >> (64 - take);
-      v = (take == 64) ? chunk : ((v << take) | chunk);
-      cur <<= take;
-      nbits -= take;
-      got += take;
+
+  __device__ __forceinline__ void init(const uint8_t *p, int64_t n) {
+    b = p;
+    len = n;
+    pos = 0;
+    hi = lo = 0;
+    have = 0;
+    fill();
+  }
+
+  __device__ __forceinline__ int read(int n, uint64_t *out) {
+    if (have < n) {
+      fill();
+      if (have < n) return -1;
     }
+    uint64_t v = (n == 64) ? hi : (hi >> (64 - n));
+    hi = (n == 64) ? lo : ((hi << n) | (lo >> (64 - n)));
+    lo = (n == 64) ? 0 : (lo << n);
+    have -= n;
+    if (have <= 64) fill();
     *out = v;
     return 0;
   }
@@ -193,10 +217,7 @@ struct FloatIter {
       g_first = 1;
       g_trail = 0;
       g_mean = 64;
-      br.b = in + 9;
-      br.len = inlen - 9;
-      br.pos = 0;
-      br.nbits = 0;
+      br.init(in + 9, inlen - 9);
       return 0;
     default:
       return -2; /* snappy handled by the general kernel; MLF unsupported */
@@ -1366,9 +1387,41 @@ __device__ __forceinline__ void gacc_merge(GAcc *l, const GAcc *r) {
 }
 
 template <int COLTYPE>
-__global__ void __launch_bounds__(256) k_group(
+__global__ void __launch_bounds__(256) k_group_p1(
     const SeriesQ *__restrict__ series, uint32_t nseries,
-    const gemx_agg_row *__restrict__ rows, gemx_agg_row *__restrict__ out,
+    const gemx_agg_row *__restrict__ rows, GAcc *__restrict__ gtmp,
+    int64_t W0, uint32_t n_gwins, uint32_t split, uint32_t per_chunk) {
+  __shared__ GAcc sh[256];
+  for (uint32_t bb = blockIdx.x; bb < n_gwins * split; bb += gridDim.x) {
+    uint32_t wb = bb / split, c = bb % split;
+    int64_t w = W0 + (int64_t)wb;
+    uint32_t g0 = c * per_chunk;
+    uint32_t g1 = g0 + per_chunk;
+    if (g1 > nseries) g1 = nseries;
+    GAcc a;
+    memset(&a, 0, sizeof(a));
+    for (uint32_t g = g0 + threadIdx.x; g < g1; g += blockDim.x) {
+      int64_t local = w - series[g].w_min;
+      if (local < 0 || local >= (int64_t)series[g].n_wins) continue;
+      const gemx_agg_row *r = &rows[series[g].out_base + local];
+      if (r->count < 0) continue; /* gap marker */
+      gacc_row<COLTYPE>(&a, r, g);
+    }
+    sh[threadIdx.x] = a;
+    __syncthreads();
+    for (int s = 128; s > 0; s >>= 1) {
+      if (threadIdx.x < (uint32_t)s)
+        gacc_merge<COLTYPE>(&sh[threadIdx.x], &sh[threadIdx.x + s]);
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) gtmp[bb] = sh[0];
+    __syncthreads();
+  }
+}
+
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_group_p2(
+    const GAcc *__restrict__ gtmp, uint32_t split, gemx_agg_row *__restrict__ out,
     int64_t W0, uint32_t n_gwins, int64_t interval, int64_t offset,
     int64_t q_start) {
   __shared__ GAcc sh[256];
@@ -1376,13 +1429,8 @@ __global__ void __launch_bounds__(256) k_group(
     int64_t w = W0 + (int64_t)wb;
     GAcc a;
     memset(&a, 0, sizeof(a));
-    for (uint32_t g = threadIdx.x; g < nseries; g += blockDim.x) {
-      int64_t local = w - series[g].w_min;
-      if (local < 0 || local >= (int64_t)series[g].n_wins) continue;
-      const gemx_agg_row *r = &rows[series[g].out_base + local];
-      if (r->count < 0) continue; /* gap marker */
-      gacc_row<COLTYPE>(&a, r, g);
-    }
+    for (uint32_t c = threadIdx.x; c < split; c += blockDim.x)
+      gacc_merge<COLTYPE>(&a, &gtmp[wb * split + c]);
     sh[threadIdx.x] = a;
     __syncthreads();
     for (int s = 128; s > 0; s >>= 1) {
@@ -1464,6 +1512,8 @@ struct QueryPlan {
   /* grouped output (all-series GROUP BY time) */
   gemx_agg_row *d_grows = nullptr;
   gemx_agg_row *h_grows = nullptr;
+  void *d_gtmp = nullptr; /* GAcc[n_gwins × gsplit] */
+  uint32_t gsplit = 1, gper_chunk = 1;
   int64_t W0 = 0;
   uint64_t n_gwins = 0;
 };
@@ -1500,6 +1550,7 @@ static void free_plan(QueryPlan &p) {
   if (p.h_rows) hipHostFree(p.h_rows);
   if (p.d_grows) hipFree(p.d_grows);
   if (p.h_grows) hipHostFree(p.h_grows);
+  if (p.d_gtmp) hipFree(p.d_gtmp);
   p = QueryPlan();
 }
 
@@ -1731,6 +1782,17 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                           sizeof(gemx_agg_row) * (P.n_gwins ? P.n_gwins : 1)));
       HIP_CHECK(hipHostMalloc(&P.h_grows,
                               sizeof(gemx_agg_row) * (P.n_gwins ? P.n_gwins : 1)));
+      /* split the series dimension so the group stage fills the chip:
+       * aim for ≥2048 blocks total, ≤4096 series per chunk */
+      uint64_t nser = P.sq.size() ? P.sq.size() : 1;
+      uint64_t want = P.n_gwins ? (2048 + P.n_gwins - 1) / P.n_gwins : 1;
+      uint64_t per_chunk = (nser + want - 1) / want;
+      if (per_chunk < 256) per_chunk = 256;
+      if (per_chunk > nser) per_chunk = nser;
+      P.gper_chunk = (uint32_t)per_chunk;
+      P.gsplit = (uint32_t)((nser + per_chunk - 1) / per_chunk);
+      HIP_CHECK(hipMalloc(&P.d_gtmp, sizeof(GAcc) * (P.n_gwins ? P.n_gwins : 1) *
+                                         P.gsplit));
     }
     P.start = start_time;
     P.end = end_time;
@@ -1798,15 +1860,25 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          total_rows, interval, offset, start_time);
   }
   if (group_all && P.n_gwins > 0) {
-    uint32_t blocks = (uint32_t)std::min<uint64_t>(P.n_gwins, 65535);
-    if (s->col_type == GEMX_TYPE_FLOAT)
-      hipLaunchKernelGGL((k_group<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(256), 0,
-                         s->stream, d_sq, (uint32_t)sq.size(), d_rows, P.d_grows,
+    uint32_t b1 = (uint32_t)std::min<uint64_t>(P.n_gwins * P.gsplit, 65535);
+    uint32_t b2 = (uint32_t)std::min<uint64_t>(P.n_gwins, 65535);
+    if (s->col_type == GEMX_TYPE_FLOAT) {
+      hipLaunchKernelGGL((k_group_p1<GEMX_TYPE_FLOAT>), dim3(b1), dim3(256), 0,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_rows,
+                         (GAcc *)P.d_gtmp, P.W0, (uint32_t)P.n_gwins, P.gsplit,
+                         P.gper_chunk);
+      hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_FLOAT>), dim3(b2), dim3(256), 0,
+                         s->stream, (const GAcc *)P.d_gtmp, P.gsplit, P.d_grows,
                          P.W0, (uint32_t)P.n_gwins, interval, offset, start_time);
-    else
-      hipLaunchKernelGGL((k_group<GEMX_TYPE_INT>), dim3(blocks), dim3(256), 0,
-                         s->stream, d_sq, (uint32_t)sq.size(), d_rows, P.d_grows,
+    } else {
+      hipLaunchKernelGGL((k_group_p1<GEMX_TYPE_INT>), dim3(b1), dim3(256), 0,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_rows,
+                         (GAcc *)P.d_gtmp, P.W0, (uint32_t)P.n_gwins, P.gsplit,
+                         P.gper_chunk);
+      hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_INT>), dim3(b2), dim3(256), 0,
+                         s->stream, (const GAcc *)P.d_gtmp, P.gsplit, P.d_grows,
                          P.W0, (uint32_t)P.n_gwins, interval, offset, start_time);
+    }
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
