@@ -162,6 +162,23 @@ struct BitR {
     *out = v;
     return 0;
   }
+
+  /* consume up to 128 bits (the Gorilla worst case per value is 77) */
+  __device__ __forceinline__ void consume(int n) {
+    if (n >= 64) {
+      hi = lo;
+      lo = 0;
+      have -= 64;
+      n -= 64;
+      fill();
+    }
+    if (n) {
+      hi = (hi << n) | (lo >> (64 - n));
+      lo <<= n;
+      have -= n;
+    }
+    if (have <= 64) fill();
+  }
 };
 
 /* ---------------- device: value iterators ---------------- */
@@ -263,30 +280,35 @@ struct FloatIter {
         return 0;
       }
       if (g_done) return -1;
-      uint64_t bit;
-      if (br.read(1, &bit)) return -1;
-      if (bit) {
-        if (br.read(1, &bit)) return -1;
-        if (bit) {
-          uint64_t lm;
-          if (br.read(11, &lm)) return -1;
-          uint8_t lead = (uint8_t)((lm >> 6) & 0x1F);
-          g_mean = (uint8_t)(lm & 0x3F);
-          if (g_mean > 0) {
-            g_trail = (uint8_t)(64 - lead - g_mean);
-          } else {
-            g_trail = 0;
-            g_mean = 64;
-          }
-        }
-        uint64_t sbits;
-        if (br.read(g_mean, &sbits)) return -1;
+      /* branchless control decode: peek 13 bits (ctrl1+ctrl2+5 leading+
+       * 6 meaningful), then the significant bits at a known offset —
+       * one predicated path instead of three divergent ones
+       * (semantics identical to batch_float.go:384-505) */
+      uint32_t p13 = (uint32_t)(br.hi >> 51);
+      int ctrl1 = (int)(p13 >> 12);
+      int newwin = ctrl1 & (int)((p13 >> 11) & 1);
+      uint32_t lm = p13 & 0x7FF;
+      uint8_t lead = (uint8_t)((lm >> 6) & 0x1F);
+      uint8_t mean_raw = (uint8_t)(lm & 0x3F);
+      uint8_t mean_new = mean_raw ? mean_raw : 64;
+      uint8_t trail_new = mean_raw ? (uint8_t)(64 - lead - mean_raw) : 0;
+      if (newwin) {
+        g_mean = mean_new;
+        g_trail = trail_new;
+      }
+      int hdr = ctrl1 ? (newwin ? 13 : 2) : 1;
+      int nbits = hdr + (ctrl1 ? (int)g_mean : 0);
+      if (br.have < nbits) return -1; /* truncated stream */
+      if (ctrl1) {
+        uint64_t x = (br.hi << hdr) | (br.lo >> (64 - hdr));
+        uint64_t sbits = (g_mean == 64) ? x : (x >> (64 - g_mean));
         g_val ^= sbits << (g_trail & 0x3F);
         if (g_val == UVNAN) {
           g_done = 1;
           return -1;
         }
       }
+      br.consume(nbits);
       uint64_t u = g_val;
       memcpy(out, &u, 8);
       return 0;
