@@ -728,8 +728,12 @@ __global__ void __launch_bounds__(256) k_scan_fast(
       return;
     }
 
-    /* running group state (one window at a time; rows time-ascending) */
+    /* running group state (one window at a time; rows time-ascending).
+     * Window membership by range compare — the i64 floored division runs
+     * only at window changes (~1 in 60 rows), matching intervalIndex's
+     * "t >= endTime || t < startTime" test (aggregate_cursor.go:351) */
     int64_t cur_ord = INT64_MIN;
+    int64_t ws_cur = 1, we_cur = 0; /* empty range forces first window */
     int64_t cnt = 0;
     double sumf = 0;
     int64_t sumi = 0;
@@ -737,9 +741,21 @@ __global__ void __launch_bounds__(256) k_scan_fast(
     int64_t min_t = 0, max_t = 0, first_t = 0, last_t = 0, grp_start_t = 0;
     Partial *base = partials + sq.partial_base;
 
+    /* const-delta timestamps (the regular-grid case, timestamp.go:190):
+     * closed form, no per-row iterator dispatch */
+    const int t_const = (ti.kind == 1);
+    const int64_t t0c = t_const ? ti.cur : 0;
+    const int64_t dtc = t_const ? ti.delta : 0;
+    if (t_const && ti.left < rows) { set_err(err, GEMX_E_DECODE); return; }
+
     for (int i = 0; i < rows; i++) {
       int64_t t;
-      if (ti.next(&t)) { set_err(err, GEMX_E_DECODE); return; }
+      if (t_const) {
+        t = t0c + (int64_t)i * dtc;
+      } else if (ti.next(&t)) {
+        set_err(err, GEMX_E_DECODE);
+        return;
+      }
       double fv = 0;
       int64_t iv = 0;
       if (h.one_value) {
@@ -753,8 +769,7 @@ __global__ void __launch_bounds__(256) k_scan_fast(
         if (iit.next(&iv)) { set_err(err, GEMX_E_DECODE); return; }
       }
 
-      int64_t ord = interval ? win_ordinal(t, interval, offset) : 0;
-      if (ord != cur_ord) {
+      if (t >= we_cur || t < ws_cur) {
         if (cur_ord != INT64_MIN) {
           /* flush group */
           int64_t slot = cur_ord - sq.w_first;
@@ -775,8 +790,16 @@ __global__ void __launch_bounds__(256) k_scan_fast(
           p->nilmask = 0;
           p->has_rows = 1;
         }
-        cur_ord = ord;
-        if (ord < sq.w_first || ord >= sq.w_first + (int64_t)sq.n_wins) {
+        if (interval) {
+          cur_ord = win_ordinal(t, interval, offset);
+          ws_cur = cur_ord * interval + offset;
+          we_cur = ws_cur + interval;
+        } else {
+          cur_ord = 0;
+          ws_cur = INT64_MIN;
+          we_cur = INT64_MAX;
+        }
+        if (cur_ord < sq.w_first || cur_ord >= sq.w_first + (int64_t)sq.n_wins) {
           set_err(err, GEMX_E_INVALID); /* descriptor min/max_time lied */
           return;
         }
