@@ -1157,8 +1157,95 @@ __global__ void __launch_bounds__(256) k_scan_general(
 
 /* ---------------- merge kernel ---------------- */
 
-/* one lane per (sid, window): merge contributing segments' partials in time
- * order with fv() semantics (series_agg_func.gen.go:44-274) */
+/* per-(series,window) partial merge with fv() semantics
+ * (series_agg_func.gen.go:44-274); returns 0 when the window has no rows
+ * in this series. Shared by the per-series row kernel and the fused
+ * grouped path. */
+template <int COLTYPE>
+__device__ int merge_series_window(const SeriesQ &s, const SegQ *__restrict__ segq,
+                                   const Partial *__restrict__ partials, int64_t w,
+                                   gemx_agg_row *out) {
+  int active[6] = {0, 0, 0, 0, 0, 0};
+  gemx_val av[6];
+  int64_t at[6], nt[6];
+  int any = 0;
+  uint32_t a = s.seg_start, b = s.seg_start + s.seg_count;
+  /* first segment with w_first + n_wins > w (spans are time-sorted) */
+  uint32_t flo = a, fhi = b;
+  while (flo < fhi) {
+    uint32_t mid = (flo + fhi) >> 1;
+    if (segq[mid].w_first + (int64_t)segq[mid].n_wins > w) fhi = mid;
+    else flo = mid + 1;
+  }
+  for (uint32_t si = flo; si < b && segq[si].w_first <= w; si++) {
+    const SegQ q = segq[si];
+    if (w < q.w_first || w >= q.w_first + (int64_t)q.n_wins) continue;
+    const Partial p = partials[q.partial_base + (w - q.w_first)];
+    if (!p.has_rows) continue;
+    any = 1;
+    out->first_row_time = p.first_row_time;
+    for (int op = 0; op < 6; op++) {
+      if (p.nilmask & (1u << op)) {
+        nt[op] = p.t[op];
+        if (!active[op]) at[op] = p.t[op];
+        continue;
+      }
+      if (!active[op]) {
+        active[op] = 1;
+        av[op] = p.v[op];
+        at[op] = p.t[op];
+        continue;
+      }
+      switch (op) {
+      case 0: av[op].i += p.v[op].i; break; /* count merge */
+      case 1:
+        if (COLTYPE == GEMX_TYPE_FLOAT) av[op].f += p.v[op].f;
+        else av[op].i += p.v[op].i;
+        break;
+      case 2: {
+        int repl = (COLTYPE == GEMX_TYPE_FLOAT) ? (p.v[op].f < av[op].f)
+                                                : (p.v[op].i < av[op].i);
+        if (repl) { av[op] = p.v[op]; at[op] = p.t[op]; }
+        break;
+      }
+      case 3: {
+        int repl = (COLTYPE == GEMX_TYPE_FLOAT) ? (p.v[op].f > av[op].f)
+                                                : (p.v[op].i > av[op].i);
+        if (repl) { av[op] = p.v[op]; at[op] = p.t[op]; }
+        break;
+      }
+      case 4: break;                                    /* first: keep */
+      case 5: av[op] = p.v[op]; at[op] = p.t[op]; break; /* last: assign */
+      }
+    }
+  }
+  if (!any) return 0;
+  out->count = active[0] ? av[0].i : 0;
+  out->count_time = active[0] ? at[0] : nt[0];
+  out->sum = av[1];
+  out->sum_time = active[1] ? at[1] : nt[1];
+  out->sum_isnil = !active[1];
+  out->minv = av[2];
+  out->min_time = active[2] ? at[2] : nt[2];
+  out->min_isnil = !active[2];
+  out->maxv = av[3];
+  out->max_time = active[3] ? at[3] : nt[3];
+  out->max_isnil = !active[3];
+  out->firstv = av[4];
+  out->first_time = active[4] ? at[4] : nt[4];
+  out->first_isnil = !active[4];
+  out->lastv = av[5];
+  out->last_time = active[5] ? at[5] : nt[5];
+  out->last_isnil = !active[5];
+  if (!active[1]) out->sum.i = 0;
+  if (!active[2]) out->minv.i = 0;
+  if (!active[3]) out->maxv.i = 0;
+  if (!active[4]) out->firstv.i = 0;
+  if (!active[5]) out->lastv.i = 0;
+  return 1;
+}
+
+/* one lane per (sid, window) output row (per-series mode) */
 template <int COLTYPE>
 __global__ void __launch_bounds__(256) k_merge(
     const SeriesQ *__restrict__ series, uint32_t nseries,
@@ -1167,7 +1254,6 @@ __global__ void __launch_bounds__(256) k_merge(
     int64_t offset, int64_t q_start) {
   uint64_t gid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
   for (uint64_t r = gid; r < total_rows; r += gridDim.x * (uint64_t)blockDim.x) {
-    /* locate series by binary search on out_base */
     uint32_t lo = 0, hi = nseries - 1;
     while (lo < hi) {
       uint32_t mid = (lo + hi + 1) >> 1;
@@ -1176,97 +1262,12 @@ __global__ void __launch_bounds__(256) k_merge(
     }
     const SeriesQ s = series[lo];
     int64_t w = s.w_min + (int64_t)(r - s.out_base);
-
     gemx_agg_row out;
     memset(&out, 0, sizeof(out));
     out.sid = s.sid;
-    /* interval==0 ⇒ single window [start_time, end_time+1) (select.go:581) */
     out.win_start = interval ? win_start_of(w, interval, offset) : q_start;
-
-    int active[6] = {0, 0, 0, 0, 0, 0};
-    gemx_val av[6];
-    int64_t at[6], nt[6];
-    int any = 0;
-    /* segments of this series whose span contains w: contiguous run; find
-     * first via binary search on w_first, then walk */
-    uint32_t a = s.seg_start, b = s.seg_start + s.seg_count;
-    /* first segment with w_first + n_wins > w */
-    uint32_t flo = a, fhi = b;
-    while (flo < fhi) {
-      uint32_t mid = (flo + fhi) >> 1;
-      if (segq[mid].w_first + (int64_t)segq[mid].n_wins > w) fhi = mid;
-      else flo = mid + 1;
-    }
-    for (uint32_t si = flo; si < b && segq[si].w_first <= w; si++) {
-      const SegQ q = segq[si];
-      if (w < q.w_first || w >= q.w_first + (int64_t)q.n_wins) continue;
-      const Partial p = partials[q.partial_base + (w - q.w_first)];
-      if (!p.has_rows) continue;
-      any = 1;
-      out.first_row_time = p.first_row_time;
-      for (int op = 0; op < 6; op++) {
-        if (p.nilmask & (1u << op)) {
-          nt[op] = p.t[op];
-          if (!active[op]) at[op] = p.t[op]; /* placeholder */
-          continue;
-        }
-        if (!active[op]) {
-          active[op] = 1;
-          av[op] = p.v[op];
-          at[op] = p.t[op];
-          continue;
-        }
-        switch (op) {
-        case 0: av[op].i += p.v[op].i; break; /* count merge */
-        case 1:
-          if (COLTYPE == GEMX_TYPE_FLOAT) av[op].f += p.v[op].f;
-          else av[op].i += p.v[op].i;
-          break;
-        case 2: {
-          int repl = (COLTYPE == GEMX_TYPE_FLOAT) ? (p.v[op].f < av[op].f)
-                                                  : (p.v[op].i < av[op].i);
-          if (repl) { av[op] = p.v[op]; at[op] = p.t[op]; }
-          break;
-        }
-        case 3: {
-          int repl = (COLTYPE == GEMX_TYPE_FLOAT) ? (p.v[op].f > av[op].f)
-                                                  : (p.v[op].i > av[op].i);
-          if (repl) { av[op] = p.v[op]; at[op] = p.t[op]; }
-          break;
-        }
-        case 4: break;                                  /* first: keep */
-        case 5: av[op] = p.v[op]; at[op] = p.t[op]; break; /* last: assign */
-        }
-      }
-    }
-
-    if (!any) {
-      out.count = -1; /* marks a gap row: host compacts it away */
-      rows[r] = out;
-      continue;
-    }
-    out.count = active[0] ? av[0].i : 0;
-    out.count_time = active[0] ? at[0] : nt[0];
-    out.sum = av[1];
-    out.sum_time = active[1] ? at[1] : nt[1];
-    out.sum_isnil = !active[1];
-    out.minv = av[2];
-    out.min_time = active[2] ? at[2] : nt[2];
-    out.min_isnil = !active[2];
-    out.maxv = av[3];
-    out.max_time = active[3] ? at[3] : nt[3];
-    out.max_isnil = !active[3];
-    out.firstv = av[4];
-    out.first_time = active[4] ? at[4] : nt[4];
-    out.first_isnil = !active[4];
-    out.lastv = av[5];
-    out.last_time = active[5] ? at[5] : nt[5];
-    out.last_isnil = !active[5];
-    if (!active[1]) out.sum.i = 0;
-    if (!active[2]) out.minv.i = 0;
-    if (!active[3]) out.maxv.i = 0;
-    if (!active[4]) out.firstv.i = 0;
-    if (!active[5]) out.lastv.i = 0;
+    if (!merge_series_window<COLTYPE>(s, segq, partials, w, &out))
+      out.count = -1; /* gap row: host compacts it away */
     rows[r] = out;
   }
 }
@@ -1434,8 +1435,9 @@ __device__ __forceinline__ void gacc_merge(GAcc *l, const GAcc *r) {
 template <int COLTYPE>
 __global__ void __launch_bounds__(256) k_group_p1(
     const SeriesQ *__restrict__ series, uint32_t nseries,
-    const gemx_agg_row *__restrict__ rows, GAcc *__restrict__ gtmp,
-    int64_t W0, uint32_t n_gwins, uint32_t split, uint32_t per_chunk) {
+    const SegQ *__restrict__ segq, const Partial *__restrict__ partials,
+    GAcc *__restrict__ gtmp, int64_t W0, uint32_t n_gwins, uint32_t split,
+    uint32_t per_chunk) {
   __shared__ GAcc sh[256];
   for (uint32_t bb = blockIdx.x; bb < n_gwins * split; bb += gridDim.x) {
     uint32_t wb = bb / split, c = bb % split;
@@ -1446,17 +1448,19 @@ __global__ void __launch_bounds__(256) k_group_p1(
     GAcc a;
     memset(&a, 0, sizeof(a));
     for (uint32_t g = g0 + threadIdx.x; g < g1; g += blockDim.x) {
-      int64_t local = w - series[g].w_min;
-      if (local < 0 || local >= (int64_t)series[g].n_wins) continue;
-      const gemx_agg_row *r = &rows[series[g].out_base + local];
-      if (r->count < 0) continue; /* gap marker */
-      gacc_row<COLTYPE>(&a, r, g);
+      const SeriesQ s = series[g];
+      int64_t local = w - s.w_min;
+      if (local < 0 || local >= (int64_t)s.n_wins) continue;
+      gemx_agg_row r;
+      memset(&r, 0, sizeof(r));
+      if (!merge_series_window<COLTYPE>(s, segq, partials, w, &r)) continue;
+      gacc_row<COLTYPE>(&a, &r, g);
     }
     sh[threadIdx.x] = a;
     __syncthreads();
-    for (int s = 128; s > 0; s >>= 1) {
-      if (threadIdx.x < (uint32_t)s)
-        gacc_merge<COLTYPE>(&sh[threadIdx.x], &sh[threadIdx.x + s]);
+    for (int s2 = 128; s2 > 0; s2 >>= 1) {
+      if (threadIdx.x < (uint32_t)s2)
+        gacc_merge<COLTYPE>(&sh[threadIdx.x], &sh[threadIdx.x + s2]);
       __syncthreads();
     }
     if (threadIdx.x == 0) gtmp[bb] = sh[0];
@@ -1893,7 +1897,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          gen_lanes, d_err);
   }
   HIP_CHECK(hipEventRecord(ev1, s->stream));
-  if (total_rows > 0) {
+  if (!group_all && total_rows > 0) {
     uint32_t blocks = (uint32_t)std::min<uint64_t>((total_rows + TPB - 1) / TPB, 65535);
     if (s->col_type == GEMX_TYPE_FLOAT)
       hipLaunchKernelGGL((k_merge<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
@@ -1909,7 +1913,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     uint32_t b2 = (uint32_t)std::min<uint64_t>(P.n_gwins, 65535);
     if (s->col_type == GEMX_TYPE_FLOAT) {
       hipLaunchKernelGGL((k_group_p1<GEMX_TYPE_FLOAT>), dim3(b1), dim3(256), 0,
-                         s->stream, d_sq, (uint32_t)sq.size(), d_rows,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part,
                          (GAcc *)P.d_gtmp, P.W0, (uint32_t)P.n_gwins, P.gsplit,
                          P.gper_chunk);
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_FLOAT>), dim3(b2), dim3(256), 0,
@@ -1917,7 +1921,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          P.W0, (uint32_t)P.n_gwins, interval, offset, start_time);
     } else {
       hipLaunchKernelGGL((k_group_p1<GEMX_TYPE_INT>), dim3(b1), dim3(256), 0,
-                         s->stream, d_sq, (uint32_t)sq.size(), d_rows,
+                         s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part,
                          (GAcc *)P.d_gtmp, P.W0, (uint32_t)P.n_gwins, P.gsplit,
                          P.gper_chunk);
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_INT>), dim3(b2), dim3(256), 0,
