@@ -111,22 +111,31 @@ struct BitR {
   uint64_t hi, lo;
   int have;
 
+  uint64_t pw; /* prefetched word: its load issues one fill earlier than its
+                  consumption, giving ~64 decoded bits of latency cover */
+  int pwbits;
+
+  __device__ __forceinline__ void preload() {
+    if (len - pos >= 8) {
+      pw = d_u64be(b + pos);
+      pos += 8;
+      pwbits = 64;
+    } else if (pos < len) {
+      int rem = (int)(len - pos);
+      uint64_t w = 0;
+      for (int i = 0; i < rem; i++) w = (w << 8) | b[pos + i];
+      pw = w << ((8 - rem) * 8);
+      pos = len;
+      pwbits = rem * 8;
+    } else {
+      pwbits = 0;
+    }
+  }
+
   __device__ __forceinline__ void fill() {
-    while (have <= 64 && pos < len) {
-      uint64_t w;
-      int wbits;
-      if (len - pos >= 8) {
-        w = d_u64be(b + pos);
-        pos += 8;
-        wbits = 64;
-      } else {
-        int rem = (int)(len - pos);
-        w = 0;
-        for (int i = 0; i < rem; i++) w = (w << 8) | b[pos + i];
-        w <<= (8 - rem) * 8;
-        pos = len;
-        wbits = rem * 8;
-      }
+    while (have <= 64 && pwbits) {
+      uint64_t w = pw;
+      int wbits = pwbits;
       if (have == 0) {
         hi = w;
         lo = 0;
@@ -137,6 +146,7 @@ struct BitR {
         lo = w;
       }
       have += wbits;
+      preload();
     }
   }
 
@@ -146,6 +156,8 @@ struct BitR {
     pos = 0;
     hi = lo = 0;
     have = 0;
+    pwbits = 0;
+    preload();
     fill();
   }
 
@@ -733,6 +745,7 @@ __global__ void __launch_bounds__(256) k_scan_fast(
      * only at window changes (~1 in 60 rows), matching intervalIndex's
      * "t >= endTime || t < startTime" test (aggregate_cursor.go:351) */
     int64_t cur_ord = INT64_MIN;
+    int64_t prev_ord = (int64_t)sq.w_first - 1;
     int64_t ws_cur = 1, we_cur = 0; /* empty range forces first window */
     int64_t cnt = 0;
     double sumf = 0;
@@ -803,6 +816,12 @@ __global__ void __launch_bounds__(256) k_scan_fast(
           set_err(err, GEMX_E_INVALID); /* descriptor min/max_time lied */
           return;
         }
+        /* clear any skipped (gap) slots between the previous window and this
+         * one — each slot is owned by exactly this lane, so no memset of the
+         * whole partials buffer is needed */
+        for (int64_t gskip = prev_ord + 1; gskip < cur_ord; gskip++)
+          base[gskip - sq.w_first].has_rows = 0;
+        prev_ord = cur_ord;
         cnt = 0;
         sumf = 0;
         sumi = 0;
@@ -1025,6 +1044,7 @@ __global__ void __launch_bounds__(256) k_scan_general(
     /* ---- per-window group reduce (oracle/agg.c semantics) ---- */
     Partial *base = partials + sq.partial_base;
     int start = 0;
+    int64_t prev_ord = (int64_t)sq.w_first - 1;
     while (start < rows) {
       int64_t ord = interval ? win_ordinal(tbuf[start], interval, offset) : 0;
       int end = start;
@@ -1037,6 +1057,9 @@ __global__ void __launch_bounds__(256) k_scan_general(
         set_err(err, GEMX_E_INVALID);
         return;
       }
+      for (int64_t gskip = prev_ord + 1; gskip < ord; gskip++)
+        base[gskip - sq.w_first].has_rows = 0; /* lane-owned gap slots */
+      prev_ord = ord;
       Partial *p = base + (ord - sq.w_first);
       p->has_rows = 1;
       p->first_row_time = tbuf[start];
@@ -1860,7 +1883,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   const uint64_t total_rows = P.total_rows;
   std::vector<SeriesQ> &sq = P.sq;
 
-  HIP_CHECK(hipMemsetAsync(d_part, 0, sizeof(Partial) * partial_slots, s->stream));
+  /* partial slots need no global clear: every slot is owned by exactly one
+   * segment lane, which writes its windows and clears interior gap slots;
+   * span edges (min/max_time windows) always carry rows */
   HIP_CHECK(hipMemsetAsync(d_err, 0, sizeof(DevErr), s->stream));
 
   hipEvent_t ev0, ev1, ev2;
