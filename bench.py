@@ -38,20 +38,20 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
-def cpu_baseline_leg(mode):
+def cpu_baseline_leg(mode, mode_name):
     """Oracle (CPU restatement) timed on this host's cores — a REPORTED
     baseline (kind='port'), never the product path. Bounded sample:
     25k series × 1k pts (~10-20 core-seconds)."""
     import binding as orc
 
-    nser = 25_000
+    nser = 100_000  # full workload size: ~0.2-3 s wall = tens of core-seconds
     t0 = time.time()
     blob, descs = orc.gen_shard(SEED, nser, 1000, mode=mode)
     gen_s = time.time() - t0
     try:
         import multiprocessing
 
-        cores = multiprocessing.cpu_count()
+        cores = min(multiprocessing.cpu_count(), 256)  # oracle OpenMP cap
     except Exception:
         cores = 1
     t0 = time.time()
@@ -65,7 +65,7 @@ def cpu_baseline_leg(mode):
         "unit": "points/s",
         "cores": cores,
         "kind": "port",
-        "sample": f"{nser} series x 1k pts ({mode}), oracle scan_agg, "
+        "sample": f"{nser} series x 1k pts ({mode_name}), oracle scan_agg, "
                   f"{wall:.2f}s wall on {cores} threads (gen {gen_s:.2f}s)",
     }
 
@@ -202,7 +202,7 @@ def main():
 
     cpu_baseline = None
     if not args.skip_cpu_baseline:
-        cpu_baseline = cpu_baseline_leg(gen_mode)
+        cpu_baseline = cpu_baseline_leg(gen_mode, args.mode)
 
     out = {
         "metric": "points aggregated/sec",

@@ -441,9 +441,9 @@ int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,
 #else
   nthreads = 1;
 #endif
-  if (nthreads > 64) nthreads = 64;
+  if (nthreads > 256) nthreads = 256;
 
-  int64_t cuts[65];
+  int64_t cuts[257];
   cuts[0] = 0;
   for (int t = 1; t < nthreads; t++) {
     int64_t c = nsegs * t / nthreads;
@@ -453,7 +453,7 @@ int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,
   }
   cuts[nthreads] = nsegs;
 
-  int64_t counts[64];
+  int64_t counts[256];
   int64_t per_cap = out_cap; /* each chunk bounded by total cap */
   orc_agg_row **bufs = (orc_agg_row **)malloc(sizeof(void *) * nthreads);
   int fail = 0;
