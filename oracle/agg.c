@@ -814,3 +814,142 @@ int64_t orc_group_merge(const orc_agg_row *rows, int64_t n, int col_type,
   free(g);
   return m;
 }
+
+/* ---------------- PromQL rate over range vectors ---------------- */
+/* Restates RangeVectorCursor + rate_prom:
+ *  - sample steps: startSample = start + range; endSample = startSample +
+ *    floor((end-startSample)/step)*step (prom_range_vector_cursor.go:55-68)
+ *  - window per step ts: [ts-range, ts] via the dual-pointer bounds
+ *    times[i] >= ts-range, times[j] > ts (prom_range_vector_cursor.go:118-153)
+ *  - NaN points are dropped first (FilterRangeNANPoint,
+ *    prom_range_vector_cursor.go:88)
+ *  - value: counter-reset-adjusted delta + Prometheus extrapolation
+ *    (executor/agg_func_prom.go:218-252 CalcReduceResult;
+ *     prom_functions.go:114-160 floatPromRateMerge) — the clamp ORDER
+ *    (durationToZero before the threshold clamp) is load-bearing. */
+
+static double prom_rate_value(const int64_t *t, const double *v, int64_t n,
+                              int64_t ts, int64_t range_ns, int is_rate,
+                              int is_counter, int *isnil) {
+  if (n <= 1) {
+    *isnil = 1;
+    return 0;
+  }
+  int64_t first_time = t[0], last_time = t[n - 1];
+  double first_value = v[0], last_value = v[n - 1];
+  double reduce = last_value - first_value;
+  if (is_counter) {
+    double prev = first_value;
+    for (int64_t i = 0; i < n; i++) {
+      if (v[i] < prev) reduce += prev;
+      prev = v[i];
+    }
+  }
+  if (last_time == first_time || range_ns == 0) {
+    *isnil = 1;
+    return 0;
+  }
+  int64_t range_start = ts - range_ns, range_end = ts;
+  double dur_to_start = (double)(first_time - range_start) / 1e9;
+  double dur_to_end = (double)(range_end - last_time) / 1e9;
+  double sampled = (double)(last_time - first_time) / 1e9;
+  double avg_dur = sampled / (double)(n - 1);
+  if (is_counter && reduce > 0 && n > 0 && first_value >= 0) {
+    double dur_to_zero = sampled * (first_value / reduce);
+    if (dur_to_zero < dur_to_start) dur_to_start = dur_to_zero;
+  }
+  double thresh = avg_dur * 1.1;
+  double extrap = sampled;
+  if (dur_to_start >= thresh) dur_to_start = avg_dur / 2;
+  extrap += dur_to_start;
+  if (dur_to_end >= thresh) dur_to_end = avg_dur / 2;
+  extrap += dur_to_end;
+  double result = reduce * (extrap / sampled);
+  if (is_rate) result = result / ((double)range_ns / 1e9);
+  *isnil = 0;
+  return result;
+}
+
+int64_t orc_prom_rate(const uint8_t *blob, int64_t blob_len,
+                      const orc_seg_desc *descs, int64_t nsegs, int64_t start,
+                      int64_t end, int64_t range_ns, int64_t step_ns, int is_rate,
+                      int is_counter, orc_rate_row *out, int64_t cap) {
+  if (step_ns < 0 || range_ns <= 0) return -1;
+  int64_t start_sample = start + range_ns;
+  int64_t end_sample =
+      (step_ns == 0) ? start_sample
+                     : start_sample + (end - start_sample) / step_ns * step_ns;
+  if (end < start_sample) return 0;
+
+  orc_val *vals = (orc_val *)malloc(MAX_ROWS_PER_SEG * sizeof(orc_val));
+  int64_t *times = (int64_t *)malloc(MAX_ROWS_PER_SEG * 8);
+  uint8_t *bm = (uint8_t *)malloc(MAX_ROWS_PER_SEG / 8 + 1);
+  /* whole-series buffers (filtered points) */
+  int64_t sbuf_cap = 1 << 20;
+  int64_t *st = (int64_t *)malloc(sbuf_cap * 8);
+  double *sv = (double *)malloc(sbuf_cap * 8);
+  int64_t nout = 0, rc = -1;
+
+  int64_t i = 0;
+  while (i < nsegs) {
+    uint64_t sid = descs[i].sid;
+    int64_t npts = 0;
+    for (; i < nsegs && descs[i].sid == sid; i++) {
+      const orc_seg_desc *d = &descs[i];
+      if (d->data_offset + d->data_size > (uint64_t)blob_len ||
+          d->time_offset + d->time_size > (uint64_t)blob_len)
+        goto done;
+      int rows = 0, nil = 0, trows = 0;
+      if (orc_decode_time_segment(blob + d->time_offset, d->time_size, times,
+                                  &trows))
+        goto done;
+      if (orc_decode_data_segment(ORC_TYPE_FLOAT, blob + d->data_offset,
+                                  d->data_size, vals, bm, &rows, &nil))
+        goto done;
+      if (rows != trows) goto done;
+      int vIdx = 0;
+      for (int r = 0; r < rows; r++) {
+        if (nil > 0 && !((bm[r >> 3] >> (r & 7)) & 1)) continue;
+        double x = vals[vIdx++].f;
+        if (x != x) continue; /* FilterRangeNANPoint */
+        if (npts >= sbuf_cap) {
+          sbuf_cap *= 2;
+          st = (int64_t *)realloc(st, sbuf_cap * 8);
+          sv = (double *)realloc(sv, sbuf_cap * 8);
+        }
+        st[npts] = times[r];
+        sv[npts] = x;
+        npts++;
+      }
+    }
+    /* dual-pointer over steps (prom_range_vector_cursor.go:118-153) */
+    int64_t pi = 0, pj = 0;
+    for (int64_t ts = start_sample; ts <= end_sample;
+         ts += (step_ns ? step_ns : 1)) {
+      int64_t wstart = ts - range_ns;
+      while (pi < npts && st[pi] < wstart) pi++;
+      while (pj < npts && st[pj] <= ts) pj++;
+      int isnil;
+      double val = prom_rate_value(st + pi, sv + pi, pj - pi, ts, range_ns,
+                                   is_rate, is_counter, &isnil);
+      if (!isnil) {
+        if (nout >= cap) goto done;
+        out[nout].sid = sid;
+        out[nout].ts = ts;
+        out[nout].value = val;
+        out[nout].isnil = 0;
+        memset(out[nout]._pad, 0, sizeof(out[nout]._pad));
+        nout++;
+      }
+      if (step_ns == 0) break;
+    }
+  }
+  rc = nout;
+done:
+  free(vals);
+  free(times);
+  free(bm);
+  free(st);
+  free(sv);
+  return rc;
+}

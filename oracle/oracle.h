@@ -200,6 +200,26 @@ int64_t orc_scan_agg(const uint8_t *blob, int64_t blob_len,
 int64_t orc_group_merge(const orc_agg_row *rows, int64_t n, int col_type,
                         int64_t interval, orc_agg_row *out, int64_t cap);
 
+
+/* PromQL rate over range vectors (config #5): restates
+ * engine/prom_range_vector_cursor.go:49-153 (sample steps + dual-pointer
+ * [ts-range, ts] windows, NaN points dropped) and
+ * engine/prom_functions.go:107-160 + executor/agg_func_prom.go:218-252
+ * (counter resets + Prometheus extrapolation). is_rate/is_counter select
+ * rate (1,1), increase (0,1), delta (0,0). Emits only non-nil steps. */
+typedef struct {
+  uint64_t sid;
+  int64_t ts;
+  double value;
+  uint8_t isnil;
+  uint8_t _pad[7];
+} orc_rate_row;
+
+int64_t orc_prom_rate(const uint8_t *blob, int64_t blob_len,
+                      const orc_seg_desc *descs, int64_t nsegs, int64_t start,
+                      int64_t end, int64_t range_ns, int64_t step_ns, int is_rate,
+                      int is_counter, orc_rate_row *out, int64_t cap);
+
 /* Multi-threaded variant (OpenMP over series groups) — the bench.py
  * cpu_baseline leg. nthreads<=0 → all cores. */
 int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,
