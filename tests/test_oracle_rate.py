@@ -56,10 +56,11 @@ class TestRateSemantics:
         # (agg_func_prom.go:236-250)
         t = np.arange(10, dtype=np.int64) * S
         v = np.array([0, 1, 2, 3, 4, 0, 1, 2, 3, 4], dtype=np.float64)
-        got = run(0, 9 * S, 9 * S, 0, t=t, v=v)
+        blob, d = one_series_shard(t, v)
+        rows = orc.prom_rate(blob, d, 0, 9 * S, 9 * S, 0)
         # window [0,9]: delta=4-0 + reset add 4 = 8 over 9s, full coverage
         # extrapolation: durToStart=0, durToEnd=0 -> extrap=sampled -> 8/9
-        assert abs(got[9] - 8.0 / 9.0) < 1e-12
+        assert abs(float(rows["value"][0]) - 8.0 / 9.0) < 1e-12
 
     def test_increase_vs_rate(self):
         t = np.arange(10, dtype=np.int64) * S
