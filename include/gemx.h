@@ -92,6 +92,15 @@ typedef struct {
   uint8_t _pad[3];
 } gemx_agg_row;
 
+/* One (sid, sample-step) PromQL range-vector result row. */
+typedef struct {
+  uint64_t sid;
+  int64_t ts; /* sample time (startSample + k*step) */
+  double value;
+  uint8_t isnil;
+  uint8_t _pad[7];
+} gemx_rate_row;
+
 typedef struct {
   double h2d_ms;        /* blob upload (attach time, not per query) */
   double decode_ms;     /* fused decode+reduce kernel */
@@ -142,6 +151,19 @@ int gemx_scan_agg_grouped(gemx_shard *, int64_t start_time, int64_t end_time,
                           int64_t interval, int64_t offset,
                           gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                           gemx_query_stats *stats);
+
+/* PromQL rate()/increase()/delta() over range vectors — the
+ * RangeVectorCursor path (engine/prom_range_vector_cursor.go:49-153,
+ * prom_functions.go:107-160): sample steps ts from start+range to
+ * start+range+k*step ≤ end, window [ts-range, ts], NaN points dropped,
+ * counter resets + Prometheus extrapolation. is_rate/is_counter: rate=(1,1),
+ * increase=(0,1), delta=(0,0). Float columns only. Emits only non-nil rows,
+ * grouped by sid, steps ascending. Requires range/step+2 ≤ 8 open windows
+ * (GEMX_E_UNSUPPORTED otherwise this round). */
+int gemx_prom_rate(gemx_shard *, int64_t start_time, int64_t end_time,
+                   int64_t range_ns, int64_t step_ns, int is_rate,
+                   int is_counter, gemx_rate_row *out_host, uint64_t cap,
+                   uint64_t *n_out, gemx_query_stats *stats);
 
 #ifdef __cplusplus
 }

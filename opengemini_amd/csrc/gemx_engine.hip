@@ -1545,6 +1545,342 @@ __global__ void __launch_bounds__(256) k_group_p2(
   }
 }
 
+
+/* ================= PromQL rate over range vectors (config #5) =============
+ * Restates RangeVectorCursor + rate_prom (see oracle/agg.c orc_prom_rate for
+ * the line-cited CPU spec): sample steps ts = startSample + k*step, window
+ * [ts-range, ts] (dual-pointer bounds, prom_range_vector_cursor.go:118-153),
+ * NaN points dropped, counter resets + Prometheus extrapolation
+ * (agg_func_prom.go:218-252, prom_functions.go:114-160).
+ *
+ * Streaming design: a lane decodes its segment once; a ring of RATE_W
+ * statically-unrolled window slots holds the open sample windows (a point t
+ * belongs to steps ts ∈ [t, t+range] — at most range/step+1 of them, host
+ * rejects range/step+2 > RATE_W). Per-(segment, step) partials carry
+ * {first, last, count, resetAdj}; a merge kernel folds a series' segment
+ * partials in time order (the reset between adjacent segments falls out of
+ * first/last algebra) and applies the extrapolation. */
+
+#define RATE_W 8
+
+struct RatePartial {
+  int64_t first_t, last_t;
+  double first_v, last_v;
+  int64_t count;
+  double reset_adj; /* Σ pre-reset values within this (segment, window) */
+};
+
+struct RateSlot {
+  int64_t ts;     /* sample time; INT64_MIN = inactive */
+  int64_t ord;    /* step ordinal (global) */
+  RatePartial p;
+};
+
+struct RateSegQ {
+  int64_t s0;            /* first step ordinal this segment contributes to */
+  uint64_t partial_base; /* slot base */
+  uint32_t n_steps;
+  uint32_t series_idx;
+};
+
+struct RateSeriesQ {
+  uint64_t sid;
+  int64_t s_min;
+  uint64_t out_base;
+  uint32_t n_steps;
+  uint32_t seg_start, seg_count;
+  uint32_t _pad;
+};
+
+__device__ __forceinline__ void rate_slot_update(RateSlot *s, int64_t t, double v,
+                                                 int64_t range_ns) {
+  /* window [ts-range, ts] (closed both ends — :118-153's >= start, <= end) */
+  bool in = (s->ts != INT64_MIN) && t >= s->ts - range_ns && t <= s->ts;
+  if (!in) return;
+  if (s->p.count == 0) {
+    s->p.first_t = t;
+    s->p.first_v = v;
+  } else if (v < s->p.last_v) {
+    s->p.reset_adj += s->p.last_v; /* counter reset, agg_func_prom.go:236-250 */
+  }
+  s->p.last_t = t;
+  s->p.last_v = v;
+  s->p.count++;
+}
+
+template <int FAST>
+__global__ void __launch_bounds__(256) k_rate_scan(
+    const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
+    const RateSegQ *__restrict__ rsegq, const uint32_t *__restrict__ seg_ids,
+    uint32_t nseg_ids, RatePartial *__restrict__ partials, int64_t start_sample,
+    int64_t step_ns, int64_t range_ns, uint8_t *__restrict__ scratch,
+    uint64_t scratch_per_lane, uint32_t nlanes, DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t stride = FAST ? gridDim.x * blockDim.x : nlanes;
+  if (!FAST && gid >= nlanes) return;
+
+  for (uint32_t li = gid; li < nseg_ids; li += stride) {
+    uint32_t si = seg_ids[li];
+    const gemx_seg_desc d = descs[si];
+    const RateSegQ rq = rsegq[si];
+    int rows = (int)d.rows;
+    RatePartial *base = partials + rq.partial_base;
+
+    /* zero this segment's partial slots (lane-owned) */
+    for (uint32_t k = 0; k < rq.n_steps; k++) {
+      base[k].count = 0;
+      base[k].reset_adj = 0;
+    }
+    if (rq.n_steps == 0) continue;
+
+    /* ring init: slots j hold ordinals rq.s0+j */
+    RateSlot ring[RATE_W];
+#pragma unroll
+    for (int j = 0; j < RATE_W; j++) {
+      int64_t o = rq.s0 + j;
+      bool act = j < (int)rq.n_steps;
+      ring[j].ts = act ? (start_sample + o * step_ns) : INT64_MIN;
+      ring[j].ord = o;
+      ring[j].p.count = 0;
+      ring[j].p.reset_adj = 0;
+    }
+
+    /* value/time iterators (fast: streaming; general: via scratch) */
+    TimeIter ti;
+    FloatIter fit;
+    int64_t *tbuf = nullptr;
+    double *vbuf = nullptr;
+    uint8_t *bmv = nullptr;
+    int dense = 0, nilcount = 0;
+    SegHeader h;
+
+    if (FAST) {
+      const uint8_t *tseg = blob + d.time_offset;
+      if (tseg[0] == 17) {
+        ti.kind = 1;
+        ti.cur = (int64_t)d_u64le(tseg + 1);
+        ti.delta = 0;
+        ti.left = 1;
+      } else if (tseg[0] == 31 && d.time_size > 5) {
+        if (ti.init(tseg + 5, d.time_size - 5)) { set_err(err, GEMX_E_DECODE); return; }
+      } else { set_err(err, GEMX_E_DECODE); return; }
+      if (parse_data_header(blob + d.data_offset, d.data_size, GEMX_TYPE_FLOAT, &h)) {
+        set_err(err, GEMX_E_DECODE);
+        return;
+      }
+      if (!h.one_value) {
+        int vrc = fit.init(h.enc, h.enc_len);
+        if (vrc) { set_err(err, vrc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+      }
+    } else {
+      /* general: decode into per-lane scratch (same layout as k_scan_general) */
+      uint8_t *my = scratch + (uint64_t)gid * scratch_per_lane;
+      tbuf = (int64_t *)my;
+      vbuf = (double *)(my + 4096 * 8);
+      uint8_t *sbuf = my + 2 * 4096 * 8;
+      bmv = my + 2 * 4096 * 8 + 40960;
+      {
+        const uint8_t *tseg = blob + d.time_offset;
+        int64_t tlen = d.time_size;
+        if (tlen < 5) { set_err(err, GEMX_E_DECODE); return; }
+        if (tseg[0] == 17) {
+          memcpy(&tbuf[0], tseg + 1, 8);
+        } else {
+          const uint8_t *enc = tseg + 5;
+          int64_t elen = tlen - 5;
+          int tag = enc[0] >> 4;
+          if (tag == 3) {
+            if (elen < 9) { set_err(err, GEMX_E_DECODE); return; }
+            int64_t comp_len = (int64_t)d_u32be(enc + 5);
+            int64_t dl = d_snappy_decode(enc + 9, comp_len, sbuf, 4096 * 8);
+            if (dl != rows * 8) { set_err(err, GEMX_E_DECODE); return; }
+            for (int i = 0; i < rows; i++) tbuf[i] = (int64_t)d_u64le(sbuf + i * 8);
+          } else {
+            TimeIter t2;
+            int rc2 = t2.init(enc, elen);
+            if (rc2) { set_err(err, rc2 == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+            for (int i = 0; i < rows; i++)
+              if (t2.next(&tbuf[i])) { set_err(err, GEMX_E_DECODE); return; }
+          }
+        }
+      }
+      if (parse_data_header(blob + d.data_offset, d.data_size, GEMX_TYPE_FLOAT, &h)) {
+        set_err(err, GEMX_E_DECODE);
+        return;
+      }
+      nilcount = 0;
+      if (h.one_value) {
+        nilcount = h.nilcount;
+        dense = 1 - nilcount;
+        if (dense) memcpy(&vbuf[0], h.enc, 8);
+      } else if (h.enc_len == 0) {
+        nilcount = h.nilcount;
+        dense = 0;
+      } else {
+        int tag = h.enc[0] >> 4;
+        if (tag == 2) {
+          int64_t dl = d_snappy_decode(h.enc + 1, h.enc_len - 1, sbuf, 4096 * 8);
+          if (dl < 0 || dl % 8) { set_err(err, GEMX_E_DECODE); return; }
+          dense = (int)(dl / 8);
+          for (int i = 0; i < dense; i++) {
+            uint64_t u = d_u64le(sbuf + i * 8);
+            memcpy(&vbuf[i], &u, 8);
+          }
+        } else {
+          FloatIter f2;
+          int rc2 = f2.init(h.enc, h.enc_len);
+          if (rc2) { set_err(err, rc2 == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+          dense = 0;
+          double x;
+          while (dense < 4096 && f2.next(&x) == 0) vbuf[dense++] = x;
+        }
+        nilcount = h.bitmap ? h.nilcount : 0;
+      }
+      /* build a validity view for the streaming loop */
+      (void)bmv;
+    }
+
+    int64_t first_open = rq.s0; /* smallest possibly-active ordinal */
+    int vIdx = 0;
+    const int t_const = FAST && (ti.kind == 1);
+    const int64_t t0c = t_const ? ti.cur : 0;
+    const int64_t dtc = t_const ? ti.delta : 0;
+
+    for (int i = 0; i < rows; i++) {
+      int64_t t;
+      double fv;
+      int valid = 1;
+      if (FAST) {
+        if (t_const) t = t0c + (int64_t)i * dtc;
+        else if (ti.next(&t)) { set_err(err, GEMX_E_DECODE); return; }
+        if (h.one_value) {
+          fv = d_f64le(h.enc);
+        } else if (fit.next(&fv)) { set_err(err, GEMX_E_DECODE); return; }
+      } else {
+        t = tbuf[i];
+        if (h.bitmap) valid = bm_valid(&h, i);
+        else if (nilcount == rows && rows > 0) valid = 0;
+        if (valid) fv = vbuf[vIdx++];
+      }
+      if (!valid) continue;
+      if (fv != fv) continue; /* FilterRangeNANPoint */
+
+      /* flush ring slots whose ts < t (their window closed) */
+      if (step_ns > 0) {
+        int64_t want_first = first_open;
+        /* first still-open ordinal: smallest o with ts(o) >= t */
+        if (t > start_sample + first_open * step_ns) {
+          int64_t o = (t - start_sample + step_ns - 1) / step_ns; /* times>=start_sample here */
+          if (t <= start_sample) o = 0;
+          if (o > want_first) want_first = o;
+        }
+        while (first_open < want_first) {
+          int j = (int)((first_open - rq.s0) % RATE_W);
+          if (j < 0) j += RATE_W;
+          RateSlot *sl = &ring[j];
+          if (sl->ts != INT64_MIN && sl->ord == first_open) {
+            base[sl->ord - rq.s0] = sl->p;
+            int64_t no = sl->ord + RATE_W;
+            if (no < rq.s0 + (int64_t)rq.n_steps) {
+              sl->ord = no;
+              sl->ts = start_sample + no * step_ns;
+              sl->p.count = 0;
+              sl->p.reset_adj = 0;
+            } else {
+              sl->ts = INT64_MIN;
+            }
+          }
+          first_open++;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < RATE_W; j++) rate_slot_update(&ring[j], t, fv, range_ns);
+    }
+    /* flush the remaining active slots */
+#pragma unroll
+    for (int j = 0; j < RATE_W; j++) {
+      if (ring[j].ts != INT64_MIN) base[ring[j].ord - rq.s0] = ring[j].p;
+    }
+  }
+}
+
+/* per-(sid, step) merge of segment partials in time order + finalize */
+__global__ void __launch_bounds__(256) k_rate_merge(
+    const RateSeriesQ *__restrict__ series, uint32_t nseries,
+    const RateSegQ *__restrict__ rsegq, const RatePartial *__restrict__ partials,
+    gemx_rate_row *__restrict__ rows, uint64_t total_rows, int64_t start_sample,
+    int64_t step_ns, int64_t range_ns, int is_rate, int is_counter) {
+  uint64_t gid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+  for (uint64_t r = gid; r < total_rows; r += gridDim.x * (uint64_t)blockDim.x) {
+    uint32_t lo = 0, hi = nseries - 1;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi + 1) >> 1;
+      if (series[mid].out_base <= r) lo = mid;
+      else hi = mid - 1;
+    }
+    const RateSeriesQ s = series[lo];
+    int64_t o = s.s_min + (int64_t)(r - s.out_base);
+    int64_t ts = start_sample + o * step_ns;
+
+    RatePartial acc;
+    acc.count = 0;
+    acc.reset_adj = 0;
+    uint32_t a = s.seg_start, b = s.seg_start + s.seg_count;
+    uint32_t flo = a, fhi = b;
+    while (flo < fhi) {
+      uint32_t mid = (flo + fhi) >> 1;
+      if (rsegq[mid].s0 + (int64_t)rsegq[mid].n_steps > o) fhi = mid;
+      else flo = mid + 1;
+    }
+    for (uint32_t si = flo; si < b && rsegq[si].s0 <= o; si++) {
+      const RateSegQ q = rsegq[si];
+      if (o < q.s0 || o >= q.s0 + (int64_t)q.n_steps) continue;
+      RatePartial p = partials[q.partial_base + (o - q.s0)];
+      if (p.count == 0) continue;
+      if (acc.count == 0) {
+        acc = p;
+      } else {
+        /* reset across the segment boundary (within the window) */
+        if (p.first_v < acc.last_v) acc.reset_adj += acc.last_v;
+        acc.reset_adj += p.reset_adj;
+        acc.last_t = p.last_t;
+        acc.last_v = p.last_v;
+        acc.count += p.count;
+      }
+    }
+
+    gemx_rate_row out;
+    out.sid = s.sid;
+    out.ts = ts;
+    out.value = 0;
+    out.isnil = 1;
+    memset(out._pad, 0, sizeof(out._pad));
+    if (acc.count > 1 && acc.last_t != acc.first_t && range_ns != 0) {
+      double reduce = (acc.last_v - acc.first_v) + (is_counter ? acc.reset_adj : 0.0);
+      int64_t range_start = ts - range_ns;
+      double dur_to_start = (double)(acc.first_t - range_start) / 1e9;
+      double dur_to_end = (double)(ts - acc.last_t) / 1e9;
+      double sampled = (double)(acc.last_t - acc.first_t) / 1e9;
+      double avg_dur = sampled / (double)(acc.count - 1);
+      if (is_counter && reduce > 0 && acc.first_v >= 0) {
+        double dz = sampled * (acc.first_v / reduce);
+        if (dz < dur_to_start) dur_to_start = dz;
+      }
+      double thresh = avg_dur * 1.1;
+      double extrap = sampled;
+      if (dur_to_start >= thresh) dur_to_start = avg_dur / 2;
+      extrap += dur_to_start;
+      if (dur_to_end >= thresh) dur_to_end = avg_dur / 2;
+      extrap += dur_to_end;
+      double result = reduce * (extrap / sampled);
+      if (is_rate) result = result / ((double)range_ns / 1e9);
+      out.value = result;
+      out.isnil = 0;
+    }
+    rows[r] = out;
+  }
+}
+
 /* ---------------- host: engine ---------------- */
 
 static __thread char g_err[512];
@@ -1590,6 +1926,33 @@ struct QueryPlan {
   uint64_t n_gwins = 0;
 };
 
+/* cached rate-query device state */
+struct RatePlan {
+  bool valid = false;
+  int64_t start = 0, end = 0, range_ns = 0, step_ns = 0;
+  std::vector<RateSegQ> rsegq;
+  std::vector<RateSeriesQ> rsq;
+  uint64_t partial_slots = 0, total_rows = 0;
+  int64_t start_sample = 0, end_sample = 0;
+  RateSegQ *d_rsegq = nullptr;
+  RateSeriesQ *d_rsq = nullptr;
+  RatePartial *d_rpart = nullptr;
+  gemx_rate_row *d_rrows = nullptr;
+  gemx_rate_row *h_rrows = nullptr;
+  uint8_t *d_scratch = nullptr;
+  uint32_t gen_lanes = 0;
+};
+
+static void free_rate_plan(RatePlan &p) {
+  if (p.d_rsegq) hipFree(p.d_rsegq);
+  if (p.d_rsq) hipFree(p.d_rsq);
+  if (p.d_rpart) hipFree(p.d_rpart);
+  if (p.d_rrows) hipFree(p.d_rrows);
+  if (p.h_rrows) hipHostFree(p.h_rrows);
+  if (p.d_scratch) hipFree(p.d_scratch);
+  p = RatePlan();
+}
+
 struct gemx_shard {
   int device;
   int col_type;
@@ -1610,6 +1973,7 @@ struct gemx_shard {
   uint64_t total_rows_scanned; /* Σ rows */
   hipStream_t stream;
   QueryPlan plan;
+  RatePlan rate_plan;
 };
 
 static void free_plan(QueryPlan &p) {
@@ -1773,6 +2137,7 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   if (!s) return GEMX_OK;
   hipSetDevice(s->device);
   free_plan(s->plan);
+  free_rate_plan(s->rate_plan);
   hipFree(s->d_blob);
   hipFree(s->d_descs);
   hipFree(s->d_fast_ids);
@@ -2032,4 +2397,197 @@ extern "C" int gemx_scan_agg_grouped(gemx_shard *s, int64_t start_time,
                                      gemx_query_stats *stats) {
   return scan_impl(s, start_time, end_time, interval, offset, 1, out_host, cap,
                    n_out, stats);
+}
+
+extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,
+                              int64_t range_ns, int64_t step_ns, int is_rate,
+                              int is_counter, gemx_rate_row *out_host,
+                              uint64_t cap, uint64_t *n_out,
+                              gemx_query_stats *stats) {
+  if (!s) return GEMX_E_INVALID;
+  if (s->col_type != GEMX_TYPE_FLOAT) {
+    seterr("prom rate needs a float column");
+    return GEMX_E_INVALID;
+  }
+  if (range_ns <= 0 || step_ns < 0) {
+    seterr("invalid range/step");
+    return GEMX_E_INVALID;
+  }
+  if (step_ns > 0 && range_ns / step_ns + 2 > RATE_W) {
+    seterr("range/step+2 exceeds the open-window ring (8) — unsupported this round");
+    return GEMX_E_UNSUPPORTED;
+  }
+  HIP_CHECK(hipSetDevice(s->device));
+  const uint64_t nsegs = s->nsegs;
+  const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960 + 4096;
+
+  int64_t start_sample = start_time + range_ns;
+  if (end_time < start_sample) {
+    *n_out = 0;
+    return GEMX_OK;
+  }
+  int64_t end_sample = (step_ns == 0)
+                           ? start_sample
+                           : start_sample +
+                                 (end_time - start_sample) / step_ns * step_ns;
+  int64_t eff_step = step_ns ? step_ns : 1; /* single step when step==0 */
+  int64_t last_ord = (end_sample - start_sample) / eff_step;
+
+  RatePlan &P = s->rate_plan;
+  if (!P.valid || P.start != start_time || P.end != end_time ||
+      P.range_ns != range_ns || P.step_ns != step_ns) {
+    free_rate_plan(P);
+    P.rsegq.resize(nsegs);
+    P.rsq.resize(s->series_ranges.size());
+    P.partial_slots = 0;
+    P.total_rows = 0;
+    P.start_sample = start_sample;
+    P.end_sample = end_sample;
+    for (size_t g = 0; g < s->series_ranges.size(); g++) {
+      auto &r = s->series_ranges[g];
+      int64_t smin = INT64_MAX, smax = INT64_MIN;
+      for (uint32_t i = r.start; i < r.start + r.count; i++) {
+        const gemx_seg_desc &d = s->h_descs[i];
+        /* steps whose window [ts-range, ts] can contain a point of this
+         * segment: ts in [min_time, max_time + range], clamped to grid */
+        int64_t lo = d.min_time - start_sample; /* ceil to grid */
+        int64_t o0 = (lo <= 0) ? 0 : (lo + eff_step - 1) / eff_step;
+        int64_t hi = d.max_time + range_ns - start_sample; /* floor */
+        int64_t o1 = (hi < 0) ? -1 : hi / eff_step;
+        if (o1 > last_ord) o1 = last_ord;
+        if (o0 > last_ord || o1 < o0) {
+          P.rsegq[i].s0 = 0;
+          P.rsegq[i].n_steps = 0;
+          P.rsegq[i].partial_base = P.partial_slots;
+          P.rsegq[i].series_idx = (uint32_t)g;
+          continue;
+        }
+        P.rsegq[i].s0 = o0;
+        P.rsegq[i].n_steps = (uint32_t)(o1 - o0 + 1);
+        P.rsegq[i].partial_base = P.partial_slots;
+        P.rsegq[i].series_idx = (uint32_t)g;
+        P.partial_slots += P.rsegq[i].n_steps;
+        smin = std::min(smin, o0);
+        smax = std::max(smax, o1);
+      }
+      if (smin == INT64_MAX) { /* series entirely out of range */
+        smin = 0;
+        smax = -1;
+      }
+      P.rsq[g].sid = r.sid;
+      P.rsq[g].s_min = smin;
+      P.rsq[g].out_base = P.total_rows;
+      P.rsq[g].n_steps = (uint32_t)(smax - smin + 1);
+      P.rsq[g].seg_start = r.start;
+      P.rsq[g].seg_count = r.count;
+      P.total_rows += P.rsq[g].n_steps;
+    }
+    HIP_CHECK(hipMalloc(&P.d_rsegq, sizeof(RateSegQ) * (nsegs ? nsegs : 1)));
+    HIP_CHECK(hipMemcpyAsync(P.d_rsegq, P.rsegq.data(), sizeof(RateSegQ) * nsegs,
+                             hipMemcpyHostToDevice, s->stream));
+    HIP_CHECK(hipMalloc(&P.d_rsq,
+                        sizeof(RateSeriesQ) * (P.rsq.empty() ? 1 : P.rsq.size())));
+    HIP_CHECK(hipMemcpyAsync(P.d_rsq, P.rsq.data(),
+                             sizeof(RateSeriesQ) * P.rsq.size(),
+                             hipMemcpyHostToDevice, s->stream));
+    HIP_CHECK(hipMalloc(&P.d_rpart,
+                        sizeof(RatePartial) * (P.partial_slots ? P.partial_slots : 1)));
+    HIP_CHECK(hipMalloc(&P.d_rrows,
+                        sizeof(gemx_rate_row) * (P.total_rows ? P.total_rows : 1)));
+    HIP_CHECK(hipHostMalloc(&P.h_rrows,
+                            sizeof(gemx_rate_row) * (P.total_rows ? P.total_rows : 1)));
+    if (!s->general_ids.empty()) {
+      P.gen_lanes = (uint32_t)std::min<uint64_t>(s->general_ids.size(), 16384);
+      HIP_CHECK(hipMalloc(&P.d_scratch, scratch_per_lane * P.gen_lanes));
+    }
+    P.start = start_time;
+    P.end = end_time;
+    P.range_ns = range_ns;
+    P.step_ns = step_ns;
+    P.valid = true;
+  }
+
+  DevErr *d_err = nullptr;
+  HIP_CHECK(hipMalloc(&d_err, sizeof(DevErr)));
+  HIP_CHECK(hipMemsetAsync(d_err, 0, sizeof(DevErr), s->stream));
+
+  hipEvent_t ev0, ev1, ev2;
+  HIP_CHECK(hipEventCreate(&ev0));
+  HIP_CHECK(hipEventCreate(&ev1));
+  HIP_CHECK(hipEventCreate(&ev2));
+  HIP_CHECK(hipEventRecord(ev0, s->stream));
+  const int TPB = 256;
+  if (!s->fast_ids.empty()) {
+    uint32_t n = (uint32_t)s->fast_ids.size();
+    uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
+    hipLaunchKernelGGL((k_rate_scan<1>), dim3(blocks), dim3(TPB), 0, s->stream,
+                       s->d_blob, s->d_descs, P.d_rsegq, s->d_fast_ids, n,
+                       P.d_rpart, start_sample, eff_step, range_ns, nullptr, 0, 0,
+                       d_err);
+  }
+  if (!s->general_ids.empty()) {
+    uint32_t n = (uint32_t)s->general_ids.size();
+    uint32_t blocks = (P.gen_lanes + TPB - 1) / TPB;
+    hipLaunchKernelGGL((k_rate_scan<0>), dim3(blocks), dim3(TPB), 0, s->stream,
+                       s->d_blob, s->d_descs, P.d_rsegq, s->d_general_ids, n,
+                       P.d_rpart, start_sample, eff_step, range_ns, P.d_scratch,
+                       scratch_per_lane, P.gen_lanes, d_err);
+  }
+  HIP_CHECK(hipEventRecord(ev1, s->stream));
+  if (P.total_rows > 0) {
+    uint32_t blocks =
+        (uint32_t)std::min<uint64_t>((P.total_rows + TPB - 1) / TPB, 65535);
+    hipLaunchKernelGGL(k_rate_merge, dim3(blocks), dim3(TPB), 0, s->stream,
+                       P.d_rsq, (uint32_t)P.rsq.size(), P.d_rsegq, P.d_rpart,
+                       P.d_rrows, P.total_rows, start_sample, eff_step, range_ns,
+                       is_rate, is_counter);
+  }
+  HIP_CHECK(hipEventRecord(ev2, s->stream));
+
+  DevErr herr = {0};
+  HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
+                           s->stream));
+  HIP_CHECK(hipMemcpyAsync(P.h_rrows, P.d_rrows,
+                           sizeof(gemx_rate_row) * P.total_rows,
+                           hipMemcpyDeviceToHost, s->stream));
+  HIP_CHECK(hipStreamSynchronize(s->stream));
+
+  float ms_scan = 0, ms_merge = 0, ms_total = 0;
+  hipEventElapsedTime(&ms_scan, ev0, ev1);
+  hipEventElapsedTime(&ms_merge, ev1, ev2);
+  hipEventElapsedTime(&ms_total, ev0, ev2);
+  hipEventDestroy(ev0);
+  hipEventDestroy(ev1);
+  hipEventDestroy(ev2);
+  hipFree(d_err);
+  if (herr.code != 0) {
+    seterr(herr.code == GEMX_E_UNSUPPORTED ? "unsupported codec on device"
+                                           : "segment decode failed on device");
+    return herr.code;
+  }
+
+  /* compact: only non-nil rows leave (the reference appends only non-nil,
+   * prom reducer append path) */
+  uint64_t n = 0;
+  for (uint64_t i = 0; i < P.total_rows; i++) {
+    if (P.h_rrows[i].isnil) continue;
+    if (n >= cap) {
+      seterr("output capacity too small");
+      return GEMX_E_CAP;
+    }
+    out_host[n++] = P.h_rrows[i];
+  }
+  *n_out = n;
+  if (stats) {
+    stats->decode_ms = ms_scan;
+    stats->merge_ms = ms_merge;
+    stats->total_ms = ms_total;
+    stats->points = s->total_rows_scanned;
+    uint64_t cb = 0;
+    for (auto &d : s->h_descs) cb += d.data_size + d.time_size;
+    stats->compressed_bytes = cb;
+    stats->n_rows = n;
+    stats->h2d_ms = 0;
+  }
+  return GEMX_OK;
 }

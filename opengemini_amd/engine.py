@@ -70,6 +70,11 @@ AGG_ROW_DTYPE = np.dtype(
 )
 
 
+RATE_ROW_DTYPE = np.dtype(
+    [("sid", "<u8"), ("ts", "<i8"), ("value", "<f8"), ("isnil", "u1"), ("_pad", "u1", (7,))]
+)
+
+
 class _Stats(C.Structure):
     _fields_ = [
         ("h2d_ms", C.c_double),
@@ -128,6 +133,11 @@ def _load():
     lib.gemx_scan_agg.argtypes = scan_sig
     lib.gemx_scan_agg_grouped.restype = C.c_int
     lib.gemx_scan_agg_grouped.argtypes = scan_sig
+    lib.gemx_prom_rate.restype = C.c_int
+    lib.gemx_prom_rate.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int, C.c_int,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     _lib = lib
     return lib
 
@@ -227,6 +237,34 @@ class Shard:
             decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
             points=st.points, compressed_bytes=st.compressed_bytes,
             n_rows=st.n_rows,
+        )
+        return out[: n.value], stats
+
+
+    def prom_rate(self, start_time, end_time, range_ns, step_ns, is_rate=True,
+                  is_counter=True, out_cap=None):
+        """PromQL rate()/increase()/delta() over range vectors — the
+        RangeVectorCursor path (see include/gemx.h). Float columns only."""
+        lib = self._lib
+        if out_cap is None:
+            d = self._descs
+            nsteps = 1
+            if step_ns > 0 and end_time >= start_time + range_ns:
+                nsteps = int((end_time - (start_time + range_ns)) // step_ns) + 2
+            n_sids = len(np.unique(d["sid"]))
+            out_cap = nsteps * n_sids + 16
+        out = np.empty(out_cap, dtype=RATE_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_prom_rate(
+            self._h, start_time, end_time, range_ns, step_ns,
+            1 if is_rate else 0, 1 if is_counter else 0,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        stats = dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            points=st.points, compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
 

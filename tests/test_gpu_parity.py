@@ -235,3 +235,104 @@ class TestGroupedParity:
             )[:n],
         )
         self._run(blob, d, F)
+
+
+class TestRateParity:
+    """gemx_prom_rate vs orc_prom_rate: exact for reset-free counters,
+    1e-12 relative where counter-reset addition order differs."""
+
+    def _run(self, blob, descs, start, end, rng_ns, step, **kw):
+        import opengemini_amd as gx
+
+        sh = gx.Shard(blob, descs, F)
+        try:
+            gpu, stats = sh.prom_rate(start, end, rng_ns, step, **kw)
+        finally:
+            sh.close()
+        ref = orc.prom_rate(blob, descs, start, end, rng_ns, step, **kw)
+        assert len(gpu) == len(ref), (len(gpu), len(ref))
+        assert np.array_equal(gpu["sid"], ref["sid"])
+        assert np.array_equal(gpu["ts"], ref["ts"])
+        tol = 1e-12 * np.maximum(1.0, np.abs(ref["value"]))
+        assert np.all(np.abs(gpu["value"] - ref["value"]) <= tol)
+
+    def test_golden_shape(self):
+        # the reference test points (prom_range_vector_cursor_test.go srcRecs1)
+        S = 10**9
+        t = np.array([2, 3, 5, 9, 10, 11, 15], dtype=np.int64) * S
+        v = np.array([2, 3, 5, 9, 10, 11, 15], dtype=np.float64)
+        dseg = orc.encode_data_segment(F, v, None, len(v), 0)
+        tseg = orc.encode_time_segment(t)
+        d = np.zeros(1, dtype=orc.SEG_DESC_DTYPE)
+        d[0] = (7, 0, len(dseg), len(v), len(dseg), len(tseg), 0, t[0], t[-1])
+        self._run(dseg + tseg, d, -4 * S, 19 * S, 5 * S, 2 * S)
+
+    def test_counter_walk_multiseries(self):
+        # monotone counters, multi-segment series, 5m windows / 1m step
+        S = 10**9
+        blobs, descs = bytearray(), []
+        rng = np.random.default_rng(61)
+        for sid in range(1, 30):
+            t = np.arange(0, 3000, 1, dtype=np.int64) * S
+            v = np.cumsum(np.abs(rng.normal(1, 0.3, 3000)))
+            for lo in range(0, 3000, 1000):
+                hi = lo + 1000
+                ds = orc.encode_data_segment(F, v[lo:hi], None, 1000, 0)
+                ts = orc.encode_time_segment(t[lo:hi])
+                descs.append((sid, len(blobs), len(ds), 1000,
+                              len(blobs) + len(ds), len(ts), 0, t[lo], t[hi - 1]))
+                blobs += ds + ts
+        d = np.zeros(len(descs), dtype=orc.SEG_DESC_DTYPE)
+        for i, x in enumerate(descs):
+            d[i] = x
+        self._run(bytes(blobs), d, 0, 2999 * S, 300 * S, 60 * S)
+
+    def test_counter_resets(self):
+        S = 10**9
+        t = np.arange(0, 600, dtype=np.int64) * S
+        v = (np.arange(600, dtype=np.float64) % 97)  # frequent resets
+        dseg = orc.encode_data_segment(F, v, None, len(v), 0)
+        tseg = orc.encode_time_segment(t)
+        d = np.zeros(1, dtype=orc.SEG_DESC_DTYPE)
+        d[0] = (3, 0, len(dseg), len(v), len(dseg), len(tseg), 0, t[0], t[-1])
+        self._run(dseg + tseg, d, 0, 599 * S, 120 * S, 30 * S)
+
+    def test_with_nulls_and_nans(self):
+        S = 10**9
+        rng = np.random.default_rng(62)
+        n = 500
+        t = np.arange(n, dtype=np.int64) * S
+        v = np.cumsum(np.abs(rng.normal(1, 0.2, n)))
+        v[50] = np.nan  # NaN point: filtered (forces snappy/null + general)
+        valid = rng.random(n) > 0.1
+        bm = np.packbits(valid.astype(np.uint8), bitorder="little")
+        dense = v[valid]
+        dseg = orc.encode_data_segment(F, dense, bm, n, int((~valid).sum()))
+        tseg = orc.encode_time_segment(t)
+        d = np.zeros(1, dtype=orc.SEG_DESC_DTYPE)
+        d[0] = (9, 0, len(dseg), n, len(dseg), len(tseg), 0, t[0], t[-1])
+        self._run(dseg + tseg, d, 0, (n - 1) * S, 60 * S, 20 * S)
+
+    def test_increase_and_delta(self):
+        S = 10**9
+        t = np.arange(0, 400, dtype=np.int64) * S
+        v = np.cumsum(np.abs(np.random.default_rng(63).normal(1, 0.1, 400)))
+        dseg = orc.encode_data_segment(F, v, None, len(v), 0)
+        tseg = orc.encode_time_segment(t)
+        d = np.zeros(1, dtype=orc.SEG_DESC_DTYPE)
+        d[0] = (1, 0, len(dseg), len(v), len(dseg), len(tseg), 0, t[0], t[-1])
+        blob = dseg + tseg
+        self._run(blob, d, 0, 399 * S, 100 * S, 50 * S, is_rate=False)
+        self._run(blob, d, 0, 399 * S, 100 * S, 50 * S, is_rate=False, is_counter=False)
+
+    def test_ring_limit_rejected(self):
+        import opengemini_amd as gx
+
+        S = 10**9
+        blob, descs = orc.gen_shard(77, 10, 1000)
+        sh = gx.Shard(blob, descs, F)
+        try:
+            with pytest.raises(gx.GemxError):
+                sh.prom_rate(0, 999 * S, 600 * S, 10 * S)  # range/step=60 > ring
+        finally:
+            sh.close()
