@@ -132,21 +132,28 @@ struct BitR {
     }
   }
 
+  /* top up the 128-bit window to min(128, remaining): inserts PARTIAL
+   * words so a 77-bit Gorilla record (1+1+11+64) never starves while bits
+   * wait in the prefetch word. Invariant: (hi,lo) bits below offset
+   * `have` are zero. */
   __device__ __forceinline__ void fill() {
-    while (have <= 64 && pwbits) {
-      uint64_t w = pw;
-      int wbits = pwbits;
+    while (pwbits && have < 128) {
+      int take = pwbits < 128 - have ? pwbits : 128 - have;
       if (have == 0) {
-        hi = w;
+        hi = pw;
         lo = 0;
       } else if (have < 64) {
-        hi |= w >> have;
-        lo = w << (64 - have);
+        hi |= pw >> have;
+        lo |= pw << (64 - have);
+      } else if (have == 64) {
+        lo |= pw;
       } else {
-        lo = w;
+        lo |= pw >> (have - 64);
       }
-      have += wbits;
-      preload();
+      have += take;
+      pw = (take == 64) ? 0 : (pw << take);
+      pwbits -= take;
+      if (!pwbits) preload();
     }
   }
 
@@ -170,7 +177,7 @@ struct BitR {
     hi = (n == 64) ? lo : ((hi << n) | (lo >> (64 - n)));
     lo = (n == 64) ? 0 : (lo << n);
     have -= n;
-    if (have <= 64) fill();
+    fill();
     *out = v;
     return 0;
   }
@@ -189,7 +196,7 @@ struct BitR {
       lo <<= n;
       have -= n;
     }
-    if (have <= 64) fill();
+    fill();
   }
 };
 
