@@ -177,7 +177,7 @@ struct BitR {
     hi = (n == 64) ? lo : ((hi << n) | (lo >> (64 - n)));
     lo = (n == 64) ? 0 : (lo << n);
     have -= n;
-    fill();
+    if (have < 78) fill(); /* 77 = the worst-case Gorilla record */
     *out = v;
     return 0;
   }
@@ -189,14 +189,13 @@ struct BitR {
       lo = 0;
       have -= 64;
       n -= 64;
-      fill();
     }
     if (n) {
       hi = (hi << n) | (lo >> (64 - n));
       lo <<= n;
       have -= n;
     }
-    fill();
+    if (have < 78) fill(); /* keep >= one worst-case record buffered */
   }
 };
 
