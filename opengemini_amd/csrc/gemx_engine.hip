@@ -132,13 +132,14 @@ struct BitR {
     }
   }
 
-  /* top up the 128-bit window to min(128, remaining): inserts PARTIAL
-   * words so a 77-bit Gorilla record (1+1+11+64) never starves while bits
-   * wait in the prefetch word. Invariant: (hi,lo) bits below offset
-   * `have` are zero. */
+  /* whole-word fill: inserts the prefetched word when it fully fits
+   * (have + pwbits <= 128). With 64-bit words this is the have <= 64 fast
+   * case; tail words (<64 bits) slot in later too. `have` can therefore
+   * sit below a 77-bit record's need while bits remain — callers needing
+   * wide contiguous reads use the staged fallback (read()) when
+   * have < 77. Invariant: (hi,lo) bits below offset `have` are zero. */
   __device__ __forceinline__ void fill() {
-    while (pwbits && have < 128) {
-      int take = pwbits < 128 - have ? pwbits : 128 - have;
+    while (pwbits && have + pwbits <= 128) {
       if (have == 0) {
         hi = pw;
         lo = 0;
@@ -150,10 +151,8 @@ struct BitR {
       } else {
         lo |= pw >> (have - 64);
       }
-      have += take;
-      pw = (take == 64) ? 0 : (pw << take);
-      pwbits -= take;
-      if (!pwbits) preload();
+      have += pwbits;
+      preload();
     }
   }
 
@@ -177,7 +176,7 @@ struct BitR {
     hi = (n == 64) ? lo : ((hi << n) | (lo >> (64 - n)));
     lo = (n == 64) ? 0 : (lo << n);
     have -= n;
-    if (have < 78) fill(); /* 77 = the worst-case Gorilla record */
+    if (have <= 64) fill();
     *out = v;
     return 0;
   }
@@ -195,7 +194,7 @@ struct BitR {
       lo <<= n;
       have -= n;
     }
-    if (have < 78) fill(); /* keep >= one worst-case record buffered */
+    if (have <= 64) fill();
   }
 };
 
@@ -298,6 +297,37 @@ struct FloatIter {
         return 0;
       }
       if (g_done) return -1;
+      if (br.have < 77) {
+        /* near stream end (or tail-word fill): staged <=64-bit reads with
+         * refills between — never starves (batch_float.go:384-505 order) */
+        uint64_t bit;
+        if (br.read(1, &bit)) return -1;
+        if (bit) {
+          if (br.read(1, &bit)) return -1;
+          if (bit) {
+            uint64_t lm2;
+            if (br.read(11, &lm2)) return -1;
+            uint8_t lead2 = (uint8_t)((lm2 >> 6) & 0x1F);
+            g_mean = (uint8_t)(lm2 & 0x3F);
+            if (g_mean > 0) {
+              g_trail = (uint8_t)(64 - lead2 - g_mean);
+            } else {
+              g_trail = 0;
+              g_mean = 64;
+            }
+          }
+          uint64_t sb;
+          if (br.read(g_mean, &sb)) return -1;
+          g_val ^= sb << (g_trail & 0x3F);
+          if (g_val == UVNAN) {
+            g_done = 1;
+            return -1;
+          }
+        }
+        uint64_t u2 = g_val;
+        memcpy(out, &u2, 8);
+        return 0;
+      }
       /* branchless control decode: peek 13 bits (ctrl1+ctrl2+5 leading+
        * 6 meaningful), then the significant bits at a known offset —
        * one predicated path instead of three divergent ones
