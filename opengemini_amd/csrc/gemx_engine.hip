@@ -297,9 +297,47 @@ struct FloatIter {
         return 0;
       }
       if (g_done) return -1;
-      if (br.have < 77) {
-        /* near stream end (or tail-word fill): staged <=64-bit reads with
-         * refills between — never starves (batch_float.go:384-505 order) */
+      if (br.have >= 13) {
+        /* branchless control decode: peek 13 bits (ctrl1+ctrl2+5 leading+
+         * 6 meaningful), then the significant bits at a known offset —
+         * one predicated path instead of three divergent ones
+         * (semantics identical to batch_float.go:384-505) */
+        uint32_t p13 = (uint32_t)(br.hi >> 51);
+        int ctrl1 = (int)(p13 >> 12);
+        int newwin = ctrl1 & (int)((p13 >> 11) & 1);
+        uint32_t lm = p13 & 0x7FF;
+        uint8_t lead = (uint8_t)((lm >> 6) & 0x1F);
+        uint8_t mean_raw = (uint8_t)(lm & 0x3F);
+        uint8_t mean_new = mean_raw ? mean_raw : 64;
+        uint8_t trail_new = mean_raw ? (uint8_t)(64 - lead - mean_raw) : 0;
+        uint8_t eff_mean = newwin ? mean_new : g_mean;
+        uint8_t eff_trail = newwin ? trail_new : g_trail;
+        int hdr = ctrl1 ? (newwin ? 13 : 2) : 1;
+        int nbits = hdr + (ctrl1 ? (int)eff_mean : 0);
+        if (br.have >= nbits) {
+          if (newwin) {
+            g_mean = mean_new;
+            g_trail = trail_new;
+          }
+          if (ctrl1) {
+            uint64_t x = (br.hi << hdr) | (br.lo >> (64 - hdr));
+            uint64_t sbits = (eff_mean == 64) ? x : (x >> (64 - eff_mean));
+            g_val ^= sbits << (eff_trail & 0x3F);
+            if (g_val == UVNAN) {
+              g_done = 1;
+              return -1;
+            }
+          }
+          br.consume(nbits);
+          uint64_t u = g_val;
+          memcpy(out, &u, 8);
+          return 0;
+        }
+      }
+      {
+        /* rare: record wider than the buffered window (stream tails, very
+         * wide meaningful runs): staged <=64-bit reads with refills
+         * between — never starves (batch_float.go:384-505 order) */
         uint64_t bit;
         if (br.read(1, &bit)) return -1;
         if (bit) {
@@ -328,38 +366,6 @@ struct FloatIter {
         memcpy(out, &u2, 8);
         return 0;
       }
-      /* branchless control decode: peek 13 bits (ctrl1+ctrl2+5 leading+
-       * 6 meaningful), then the significant bits at a known offset —
-       * one predicated path instead of three divergent ones
-       * (semantics identical to batch_float.go:384-505) */
-      uint32_t p13 = (uint32_t)(br.hi >> 51);
-      int ctrl1 = (int)(p13 >> 12);
-      int newwin = ctrl1 & (int)((p13 >> 11) & 1);
-      uint32_t lm = p13 & 0x7FF;
-      uint8_t lead = (uint8_t)((lm >> 6) & 0x1F);
-      uint8_t mean_raw = (uint8_t)(lm & 0x3F);
-      uint8_t mean_new = mean_raw ? mean_raw : 64;
-      uint8_t trail_new = mean_raw ? (uint8_t)(64 - lead - mean_raw) : 0;
-      if (newwin) {
-        g_mean = mean_new;
-        g_trail = trail_new;
-      }
-      int hdr = ctrl1 ? (newwin ? 13 : 2) : 1;
-      int nbits = hdr + (ctrl1 ? (int)g_mean : 0);
-      if (br.have < nbits) return -1; /* truncated stream */
-      if (ctrl1) {
-        uint64_t x = (br.hi << hdr) | (br.lo >> (64 - hdr));
-        uint64_t sbits = (g_mean == 64) ? x : (x >> (64 - g_mean));
-        g_val ^= sbits << (g_trail & 0x3F);
-        if (g_val == UVNAN) {
-          g_done = 1;
-          return -1;
-        }
-      }
-      br.consume(nbits);
-      uint64_t u = g_val;
-      memcpy(out, &u, 8);
-      return 0;
     }
     }
     return -1;
