@@ -152,6 +152,24 @@ int gemx_scan_agg_grouped(gemx_shard *, int64_t start_time, int64_t end_time,
                           gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                           gemx_query_stats *stats);
 
+/* value-predicate pushdown (config #3): lib/binaryfilterfunc compare
+ * kernels (eval_generator.gen.go:31+) with FilterByField semantics
+ * (immutable/location.go:309) — rows failing the predicate (and nil rows)
+ * are removed before aggregation. filter_op: 0 none, 1 >, 2 >=, 3 <, 4 <=,
+ * 5 ==, 6 != against filter_f (float cols) / filter_i (int cols). */
+#define GEMX_FILTER_NONE 0
+#define GEMX_FILTER_GT 1
+#define GEMX_FILTER_GE 2
+#define GEMX_FILTER_LT 3
+#define GEMX_FILTER_LE 4
+#define GEMX_FILTER_EQ 5
+#define GEMX_FILTER_NEQ 6
+int gemx_scan_agg_ex(gemx_shard *, int64_t start_time, int64_t end_time,
+                     int64_t interval, int64_t offset, int group_all,
+                     int filter_op, double filter_f, int64_t filter_i,
+                     gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                     gemx_query_stats *stats);
+
 /* PromQL rate()/increase()/delta() over range vectors — the
  * RangeVectorCursor path (engine/prom_range_vector_cursor.go:49-153,
  * prom_functions.go:107-160): sample steps ts from start+range to

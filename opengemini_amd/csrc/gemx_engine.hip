@@ -727,12 +727,36 @@ __device__ __forceinline__ void set_err(DevErr *e, int code) {
 
 /* accumulate one (t, value, valid) row stream into per-window partials.
  * Shared by fast path (no nils) with streaming times/values. */
+__device__ __forceinline__ int d_filt_pass(int col_type, int filter_op, double ff,
+                                           int64_t fi, double xf, int64_t xi) {
+  /* lib/binaryfilterfunc compare kernels (eval_generator.gen.go:31+) */
+  if (col_type == GEMX_TYPE_FLOAT) {
+    switch (filter_op) {
+    case 1: return xf > ff;
+    case 2: return xf >= ff;
+    case 3: return xf < ff;
+    case 4: return xf <= ff;
+    case 5: return xf == ff;
+    default: return xf != ff;
+    }
+  }
+  switch (filter_op) {
+  case 1: return xi > fi;
+  case 2: return xi >= fi;
+  case 3: return xi < fi;
+  case 4: return xi <= fi;
+  case 5: return xi == fi;
+  default: return xi != fi;
+  }
+}
+
 template <int COLTYPE>
 __global__ void __launch_bounds__(256) k_scan_fast(
     const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
     const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
     uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
-    int64_t offset, int64_t q_start, int64_t q_end, DevErr *err) {
+    int64_t offset, int64_t q_start, int64_t q_end, int filter_op,
+    double filter_f, int64_t filter_i, DevErr *err) {
   uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
   for (uint32_t li = gid; li < nseg_ids; li += gridDim.x * blockDim.x) {
     uint32_t si = seg_ids[li];
@@ -823,6 +847,10 @@ __global__ void __launch_bounds__(256) k_scan_fast(
       } else {
         if (iit.next(&iv)) { set_err(err, GEMX_E_DECODE); return; }
       }
+
+      if (filter_op &&
+          !d_filt_pass(COLTYPE, filter_op, filter_f, filter_i, fv, iv))
+        continue; /* FilterByField drops the row before aggregation */
 
       if (t >= we_cur || t < ws_cur) {
         if (cur_ord != INT64_MIN) {
@@ -979,8 +1007,9 @@ __global__ void __launch_bounds__(256) k_scan_general(
     const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
     const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
     uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
-    int64_t offset, uint8_t *__restrict__ scratch, uint64_t scratch_per_lane,
-    uint32_t nlanes, DevErr *err) {
+    int64_t offset, int filter_op, double filter_f, int64_t filter_i,
+    uint8_t *__restrict__ scratch, uint64_t scratch_per_lane, uint32_t nlanes,
+    DevErr *err) {
   uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
   if (gid >= nlanes) return;
   uint8_t *my = scratch + (uint64_t)gid * scratch_per_lane;
@@ -1081,6 +1110,32 @@ __global__ void __launch_bounds__(256) k_scan_general(
       /* empty block / one-value null: no bitmap bytes, all rows nil */
       h.bitmap = d_zero_bm;
       h.bm_off = 0;
+    }
+
+    if (filter_op != 0) {
+      /* FilterByField: failing rows (incl. nils) removed before aggregation */
+      int w = 0, vi2 = 0;
+      for (int r2 = 0; r2 < rows; r2++) {
+        int valid2 = 1;
+        if (h.bitmap) valid2 = bm_valid(&h, r2);
+        else if (nilcount == rows && rows > 0) valid2 = 0;
+        if (!valid2) continue;
+        double xf = 0;
+        int64_t xi = 0;
+        if (COLTYPE == GEMX_TYPE_FLOAT) memcpy(&xf, &vbuf[vi2], 8);
+        else xi = vbuf[vi2];
+        if (d_filt_pass(COLTYPE, filter_op, filter_f, filter_i, xf, xi)) {
+          vbuf[w] = vbuf[vi2];
+          tbuf[w] = tbuf[r2];
+          w++;
+        }
+        vi2++;
+      }
+      rows = w;
+      dense = w;
+      nilcount = 0;
+      h.bitmap = nullptr;
+      if (rows == 0) continue;
     }
 
     /* ---- per-window group reduce (oracle/agg.c semantics) ---- */
@@ -2191,6 +2246,7 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
 
 static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                      int64_t interval, int64_t offset, int group_all,
+                     int filter_op, double filter_f, int64_t filter_i,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                      gemx_query_stats *stats) {
   if (!s) return GEMX_E_INVALID;
@@ -2308,11 +2364,13 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     if (s->col_type == GEMX_TYPE_FLOAT)
       hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
                          s->stream, s->d_blob, s->d_descs, d_segq, s->d_fast_ids, n,
-                         d_part, interval, offset, start_time, end_time, d_err);
+                         d_part, interval, offset, start_time, end_time,
+                         filter_op, filter_f, filter_i, d_err);
     else
       hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
                          s->stream, s->d_blob, s->d_descs, d_segq, s->d_fast_ids, n,
-                         d_part, interval, offset, start_time, end_time, d_err);
+                         d_part, interval, offset, start_time, end_time,
+                         filter_op, filter_f, filter_i, d_err);
   }
   if (!s->general_ids.empty()) {
     uint32_t n = (uint32_t)s->general_ids.size();
@@ -2320,13 +2378,13 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     if (s->col_type == GEMX_TYPE_FLOAT)
       hipLaunchKernelGGL((k_scan_general<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
                          s->stream, s->d_blob, s->d_descs, d_segq, s->d_general_ids, n,
-                         d_part, interval, offset, d_scratch, scratch_per_lane,
-                         gen_lanes, d_err);
+                         d_part, interval, offset, filter_op, filter_f, filter_i,
+                         d_scratch, scratch_per_lane, gen_lanes, d_err);
     else
       hipLaunchKernelGGL((k_scan_general<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
                          s->stream, s->d_blob, s->d_descs, d_segq, s->d_general_ids, n,
-                         d_part, interval, offset, d_scratch, scratch_per_lane,
-                         gen_lanes, d_err);
+                         d_part, interval, offset, filter_op, filter_f, filter_i,
+                         d_scratch, scratch_per_lane, gen_lanes, d_err);
   }
   HIP_CHECK(hipEventRecord(ev1, s->stream));
   if (!group_all && total_rows > 0) {
@@ -2426,8 +2484,8 @@ extern "C" int gemx_scan_agg(gemx_shard *s, int64_t start_time, int64_t end_time
                              int64_t interval, int64_t offset,
                              gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                              gemx_query_stats *stats) {
-  return scan_impl(s, start_time, end_time, interval, offset, 0, out_host, cap,
-                   n_out, stats);
+  return scan_impl(s, start_time, end_time, interval, offset, 0, 0, 0, 0,
+                   out_host, cap, n_out, stats);
 }
 
 /* all-series GROUP BY time: adds the on-device AggTagSetCursor merge and
@@ -2437,8 +2495,8 @@ extern "C" int gemx_scan_agg_grouped(gemx_shard *s, int64_t start_time,
                                      int64_t offset, gemx_agg_row *out_host,
                                      uint64_t cap, uint64_t *n_out,
                                      gemx_query_stats *stats) {
-  return scan_impl(s, start_time, end_time, interval, offset, 1, out_host, cap,
-                   n_out, stats);
+  return scan_impl(s, start_time, end_time, interval, offset, 1, 0, 0, 0,
+                   out_host, cap, n_out, stats);
 }
 
 extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,
@@ -2632,4 +2690,18 @@ extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_tim
     stats->h2d_ms = 0;
   }
   return GEMX_OK;
+}
+
+/* value-predicate pushdown (config #3 — binaryfilterfunc compare kernels,
+ * FilterByField semantics: failing/nil rows removed before aggregation).
+ * filter_op: 0 none, 1 >, 2 >=, 3 <, 4 <=, 5 ==, 6 != against filter_f
+ * (float columns) / filter_i (int columns). */
+extern "C" int gemx_scan_agg_ex(gemx_shard *s, int64_t start_time,
+                                int64_t end_time, int64_t interval,
+                                int64_t offset, int group_all, int filter_op,
+                                double filter_f, int64_t filter_i,
+                                gemx_agg_row *out_host, uint64_t cap,
+                                uint64_t *n_out, gemx_query_stats *stats) {
+  return scan_impl(s, start_time, end_time, interval, offset, group_all,
+                   filter_op, filter_f, filter_i, out_host, cap, n_out, stats);
 }

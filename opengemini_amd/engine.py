@@ -133,6 +133,12 @@ def _load():
     lib.gemx_scan_agg.argtypes = scan_sig
     lib.gemx_scan_agg_grouped.restype = C.c_int
     lib.gemx_scan_agg_grouped.argtypes = scan_sig
+    lib.gemx_scan_agg_ex.restype = C.c_int
+    lib.gemx_scan_agg_ex.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_int, C.c_double, C.c_int64,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     lib.gemx_prom_rate.restype = C.c_int
     lib.gemx_prom_rate.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int, C.c_int,
@@ -214,8 +220,10 @@ class Shard:
         w0min = np.minimum.reduceat(w0, starts)
         return int((w1max - w0min + 1).sum()) + 4
 
+    FILTER_OPS = {None: 0, "gt": 1, "ge": 2, "lt": 3, "le": 4, "eq": 5, "neq": 6}
+
     def scan_agg(self, start_time, end_time, interval, offset=0, out_cap=None,
-                 group_all=False):
+                 group_all=False, filter=None):
         """One fused scan: all six aggregates per GROUP BY time window —
         per-series rows, or (group_all=True) merged across all series
         on-device (the AggTagSetCursor path, the north-star query shape).
@@ -227,11 +235,22 @@ class Shard:
         out = np.empty(out_cap, dtype=AGG_ROW_DTYPE)
         n = C.c_uint64(0)
         st = _Stats()
-        fn = lib.gemx_scan_agg_grouped if group_all else lib.gemx_scan_agg
-        rc = fn(
-            self._h, start_time, end_time, interval, offset,
-            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
-        )
+        if filter is not None:
+            op_name, operand = filter
+            fop = self.FILTER_OPS[op_name]
+            ff = float(operand) if self.col_type == GEMX_TYPE_FLOAT else 0.0
+            fi = int(operand) if self.col_type == GEMX_TYPE_INT else 0
+            rc = lib.gemx_scan_agg_ex(
+                self._h, start_time, end_time, interval, offset,
+                1 if group_all else 0, fop, ff, fi,
+                out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+            )
+        else:
+            fn = lib.gemx_scan_agg_grouped if group_all else lib.gemx_scan_agg
+            rc = fn(
+                self._h, start_time, end_time, interval, offset,
+                out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+            )
         _check(rc, lib)
         stats = dict(
             decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,

@@ -316,10 +316,38 @@ static void flush_window(uint64_t sid, const winstate *w, orc_agg_row *out) {
   out->last_isnil = !a[5].active;
 }
 
-int64_t orc_scan_agg(const uint8_t *blob, int64_t blob_len, const orc_seg_desc *descs,
-                     int64_t nsegs, int col_type, int64_t start_time, int64_t end_time,
-                     int64_t interval, int64_t offset, orc_agg_row *out,
-                     int64_t out_cap) {
+static inline int filt_pass(int col_type, int filter_op, double ff, int64_t fi,
+                            const orc_val *v) {
+  /* lib/binaryfilterfunc compare kernels (eval_generator.gen.go:31+):
+   * GT/GE/LT/LE/EQ/NEQ on the scanned column; nil rows fail upstream */
+  if (filter_op == 0) return 1;
+  if (col_type == ORC_TYPE_FLOAT) {
+    double x = v->f;
+    switch (filter_op) {
+    case 1: return x > ff;
+    case 2: return x >= ff;
+    case 3: return x < ff;
+    case 4: return x <= ff;
+    case 5: return x == ff;
+    default: return x != ff;
+    }
+  }
+  int64_t x = v->i;
+  switch (filter_op) {
+  case 1: return x > fi;
+  case 2: return x >= fi;
+  case 3: return x < fi;
+  case 4: return x <= fi;
+  case 5: return x == fi;
+  default: return x != fi;
+  }
+}
+
+int64_t orc_scan_agg_f(const uint8_t *blob, int64_t blob_len,
+                       const orc_seg_desc *descs, int64_t nsegs, int col_type,
+                       int64_t start_time, int64_t end_time, int64_t interval,
+                       int64_t offset, int filter_op, double filter_f,
+                       int64_t filter_i, orc_agg_row *out, int64_t out_cap) {
   orc_val *vals = (orc_val *)malloc(MAX_ROWS_PER_SEG * sizeof(orc_val));
   int64_t *times = (int64_t *)malloc(MAX_ROWS_PER_SEG * 8);
   uint8_t *bm = (uint8_t *)malloc(MAX_ROWS_PER_SEG / 8 + 1);
@@ -355,6 +383,27 @@ int64_t orc_scan_agg(const uint8_t *blob, int64_t blob_len, const orc_seg_desc *
                                 bm, &rows, &nilcount))
       goto done;
     if (rows != trows || rows > MAX_ROWS_PER_SEG) goto done;
+
+    if (filter_op != 0) {
+      /* FilterByField semantics (immutable/location.go:309): failing rows
+       * (incl. nil rows) are removed from the record before aggregation */
+      int w = 0, vIdx = 0;
+      for (int r2 = 0; r2 < rows; r2++) {
+        int valid = (nilcount == 0) || ((nilcount < rows) && bit_at(bm, r2));
+        if (nilcount == rows) valid = 0;
+        if (!valid) continue;
+        if (filt_pass(col_type, filter_op, filter_f, filter_i, &vals[vIdx])) {
+          vals[w] = vals[vIdx];
+          times[w] = times[r2];
+          w++;
+        }
+        vIdx++;
+      }
+      rows = w;
+      nilcount = 0;
+      memset(bm, 0xFF, (size_t)((rows + 7) / 8));
+      if (rows == 0) continue;
+    }
     int dense = rows - nilcount;
 
     /* intervalIndex over this segment (aggregate_cursor.go:343-356) */
@@ -425,6 +474,16 @@ done:
   free(times);
   free(bm);
   return rc;
+}
+
+
+
+int64_t orc_scan_agg(const uint8_t *blob, int64_t blob_len, const orc_seg_desc *descs,
+                     int64_t nsegs, int col_type, int64_t start_time, int64_t end_time,
+                     int64_t interval, int64_t offset, orc_agg_row *out,
+                     int64_t out_cap) {
+  return orc_scan_agg_f(blob, blob_len, descs, nsegs, col_type, start_time,
+                        end_time, interval, offset, 0, 0, 0, out, out_cap);
 }
 
 int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,

@@ -121,6 +121,9 @@ def load():
     lib.orc_window.argtypes = [i64, i64, i64, i64, i64, i64p, i64p]
     lib.orc_scan_agg.restype = i64
     lib.orc_scan_agg.argtypes = [u8p, i64, C.c_void_p, i64, C.c_int, i64, i64, i64, i64, C.c_void_p, i64]
+    lib.orc_scan_agg_f.restype = i64
+    lib.orc_scan_agg_f.argtypes = [u8p, i64, C.c_void_p, i64, C.c_int, i64, i64, i64, i64,
+                                   C.c_int, C.c_double, i64, C.c_void_p, i64]
     lib.orc_scan_agg_mt.restype = i64
     lib.orc_scan_agg_mt.argtypes = [u8p, i64, C.c_void_p, i64, C.c_int, i64, i64, i64, i64, C.c_void_p, i64, C.c_int]
     lib.orc_group_merge.restype = i64
@@ -398,6 +401,30 @@ def gen_shard(seed, nseries, pts_per_series, seg_rows=1000, t0=0, step_ns=10**9,
     if rc < 0:
         raise ValueError(f"gen_shard failed: {rc}")
     return blob[:rc].tobytes(), descs[: n_out.value]
+
+
+FILTER_OPS = {None: 0, "gt": 1, "ge": 2, "lt": 3, "le": 4, "eq": 5, "neq": 6}
+
+
+def scan_agg_filtered(blob, descs, col_type, start_time, end_time, interval,
+                      filter_op, operand, offset=0, out_cap=None):
+    lib = get()
+    bts = np.frombuffer(blob, dtype=np.uint8)
+    d = np.ascontiguousarray(descs, dtype=SEG_DESC_DTYPE)
+    if out_cap is None:
+        out_cap = int(d["rows"].sum()) + len(d) + 16
+    out = np.zeros(out_cap, dtype=AGG_ROW_DTYPE)
+    fop = FILTER_OPS[filter_op]
+    ff = float(operand) if col_type == ORC_TYPE_FLOAT else 0.0
+    fi = int(operand) if col_type == ORC_TYPE_INT else 0
+    n = lib.orc_scan_agg_f(
+        _u8(bts), len(bts), d.ctypes.data_as(C.c_void_p), len(d), col_type,
+        start_time, end_time, interval, 0, fop, ff, fi,
+        out.ctypes.data_as(C.c_void_p), out_cap,
+    )
+    if n < 0:
+        raise ValueError("scan_agg_f failed")
+    return out[:n].copy()
 
 
 def agg_cursor(col_type, op, multi_call, dense_vals, valid_bits, times, rec_rows,

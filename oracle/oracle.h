@@ -186,6 +186,17 @@ int64_t orc_scan_agg(const uint8_t *blob, int64_t blob_len,
                      int64_t start_time, int64_t end_time, int64_t interval,
                      int64_t offset, orc_agg_row *out, int64_t out_cap);
 
+/* value-predicate pushdown (config #3 — lib/binaryfilterfunc compare
+ * kernels, eval_generator.gen.go:31+): rows failing the predicate (and nil
+ * rows) are removed before aggregation, as FilterByField does
+ * (immutable/location.go:309). filter_op: 0 none, 1 >, 2 >=, 3 <, 4 <=,
+ * 5 ==, 6 != against filter_f (float cols) / filter_i (int cols). */
+int64_t orc_scan_agg_f(const uint8_t *blob, int64_t blob_len,
+                       const orc_seg_desc *descs, int64_t nsegs, int col_type,
+                       int64_t start_time, int64_t end_time, int64_t interval,
+                       int64_t offset, int filter_op, double filter_f,
+                       int64_t filter_i, orc_agg_row *out, int64_t out_cap);
+
 /* Merge per-(sid,window) rows into per-window group rows — the
  * AggTagSetCursor.UpdateRec semantics for the all-series group of
  * `GROUP BY time(w)` (engine/agg_tagset_cursor.go:1111-1122,

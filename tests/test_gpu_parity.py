@@ -336,3 +336,64 @@ class TestRateParity:
                 sh.prom_rate(0, 999 * S, 600 * S, 10 * S)  # range/step=60 > ring
         finally:
             sh.close()
+
+
+class TestFilterParity:
+    """Value-predicate pushdown (config #3, binaryfilterfunc equivalents)."""
+
+    def _run(self, blob, descs, col_type, op, operand):
+        import opengemini_amd as gx
+
+        sh = gx.Shard(blob, descs, col_type)
+        try:
+            gpu, _ = sh.scan_agg(0, 2**62, INT, filter=(op, operand))
+        finally:
+            sh.close()
+        ref = orc.scan_agg_filtered(blob, descs, col_type, 0, 2**62, INT, op, operand)
+        assert_parity(gpu, ref, col_type)
+
+    def test_float_gt(self):
+        rng = np.random.default_rng(70)
+        blob, d, _ = build_shard(rng, F, [1, 2, 3], null_frac=0.0)
+        self._run(blob, d, F, "gt", 0.0)
+
+    def test_float_with_nulls_le(self):
+        rng = np.random.default_rng(71)
+        blob, d, _ = build_shard(rng, F, [4, 5])
+        self._run(blob, d, F, "le", 1.5)
+
+    def test_int_gt_500(self):
+        # the config #3 predicate shape: int64 values in [0,1000), value > 500
+        blob, descs = orc.gen_shard(47, 1000, 1000, mode=orc.GEN_INT_SMALL)
+        self._run(blob, descs, I, "gt", 500)
+
+    def test_filter_all_rows_out(self):
+        blob, descs = orc.gen_shard(48, 50, 1000, mode=orc.GEN_INT_SMALL)
+        self._run(blob, descs, I, "gt", 10**9)
+
+    def test_filter_eq(self):
+        blob, descs = orc.gen_shard(49, 100, 1000, mode=orc.GEN_INT_SMALL)
+        self._run(blob, descs, I, "eq", 7)
+
+    def test_filter_vs_numpy(self):
+        # independent truth: numpy filter + window aggregation
+        rng = np.random.default_rng(72)
+        from shard_helpers import expected_windows, check
+
+        blob, d, truth = build_shard(rng, F, [9, 10], null_frac=0.2)
+        rows = None
+        import opengemini_amd as gx
+
+        sh = gx.Shard(blob, d, F)
+        try:
+            rows, _ = sh.scan_agg(0, 2**62, INT, filter=("gt", 0.0))
+        finally:
+            sh.close()
+        # apply the same filter to the truth
+        ftruth = {}
+        for sid, (at, av, ax) in truth.items():
+            keep = ax & (av > 0.0)
+            ftruth[sid] = (at[keep], av[keep], np.ones(int(keep.sum()), dtype=bool))
+        exp = expected_windows(ftruth, INT)
+        exp = {k: v for k, v in exp.items() if v["count"] > 0}
+        check(rows, exp, F)
