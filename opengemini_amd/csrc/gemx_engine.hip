@@ -811,9 +811,14 @@ __global__ void __launch_bounds__(256) k_scan_fast(
      * only at window changes (~1 in 60 rows), matching intervalIndex's
      * "t >= endTime || t < startTime" test (aggregate_cursor.go:351) */
     int64_t cur_ord = INT64_MIN;
-    int64_t prev_ord = (int64_t)sq.w_first - 1;
     int64_t ws_cur = 1, we_cur = 0; /* empty range forces first window */
     int64_t cnt = 0;
+    /* clear this lane's partial slots: windows can be empty via time gaps
+     * or a predicate that drops every row */
+    {
+      Partial *pb = partials + sq.partial_base;
+      for (uint32_t k = 0; k < sq.n_wins; k++) pb[k].has_rows = 0;
+    }
     double sumf = 0;
     int64_t sumi = 0;
     gemx_val minv = {0}, maxv = {0}, firstv = {0}, lastv = {0};
@@ -886,12 +891,6 @@ __global__ void __launch_bounds__(256) k_scan_fast(
           set_err(err, GEMX_E_INVALID); /* descriptor min/max_time lied */
           return;
         }
-        /* clear any skipped (gap) slots between the previous window and this
-         * one — each slot is owned by exactly this lane, so no memset of the
-         * whole partials buffer is needed */
-        for (int64_t gskip = prev_ord + 1; gskip < cur_ord; gskip++)
-          base[gskip - sq.w_first].has_rows = 0;
-        prev_ord = cur_ord;
         cnt = 0;
         sumf = 0;
         sumi = 0;
@@ -1112,6 +1111,10 @@ __global__ void __launch_bounds__(256) k_scan_general(
       h.bm_off = 0;
     }
 
+    {
+      Partial *pb0 = partials + sq.partial_base;
+      for (uint32_t k = 0; k < sq.n_wins; k++) pb0[k].has_rows = 0;
+    }
     if (filter_op != 0) {
       /* FilterByField: failing rows (incl. nils) removed before aggregation */
       int w = 0, vi2 = 0;
@@ -1141,7 +1144,6 @@ __global__ void __launch_bounds__(256) k_scan_general(
     /* ---- per-window group reduce (oracle/agg.c semantics) ---- */
     Partial *base = partials + sq.partial_base;
     int start = 0;
-    int64_t prev_ord = (int64_t)sq.w_first - 1;
     while (start < rows) {
       int64_t ord = interval ? win_ordinal(tbuf[start], interval, offset) : 0;
       int end = start;
@@ -1154,9 +1156,6 @@ __global__ void __launch_bounds__(256) k_scan_general(
         set_err(err, GEMX_E_INVALID);
         return;
       }
-      for (int64_t gskip = prev_ord + 1; gskip < ord; gskip++)
-        base[gskip - sq.w_first].has_rows = 0; /* lane-owned gap slots */
-      prev_ord = ord;
       Partial *p = base + (ord - sq.w_first);
       p->has_rows = 1;
       p->first_row_time = tbuf[start];
