@@ -750,7 +750,7 @@ __device__ __forceinline__ int d_filt_pass(int col_type, int filter_op, double f
   }
 }
 
-template <int COLTYPE>
+template <int COLTYPE, int FILT>
 __global__ void __launch_bounds__(256) k_scan_fast(
     const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
     const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
@@ -853,7 +853,7 @@ __global__ void __launch_bounds__(256) k_scan_fast(
         if (iit.next(&iv)) { set_err(err, GEMX_E_DECODE); return; }
       }
 
-      if (filter_op &&
+      if (FILT &&
           !d_filt_pass(COLTYPE, filter_op, filter_f, filter_i, fv, iv))
         continue; /* FilterByField drops the row before aggregation */
 
@@ -2360,16 +2360,29 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   if (!s->fast_ids.empty()) {
     uint32_t n = (uint32_t)s->fast_ids.size();
     uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
-    if (s->col_type == GEMX_TYPE_FLOAT)
-      hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
-                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_fast_ids, n,
-                         d_part, interval, offset, start_time, end_time,
-                         filter_op, filter_f, filter_i, d_err);
-    else
-      hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
-                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_fast_ids, n,
-                         d_part, interval, offset, start_time, end_time,
-                         filter_op, filter_f, filter_i, d_err);
+    if (s->col_type == GEMX_TYPE_FLOAT) {
+      if (filter_op)
+        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT, 1>), dim3(blocks),
+                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
+                           s->d_fast_ids, n, d_part, interval, offset, start_time,
+                           end_time, filter_op, filter_f, filter_i, d_err);
+      else
+        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT, 0>), dim3(blocks),
+                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
+                           s->d_fast_ids, n, d_part, interval, offset, start_time,
+                           end_time, filter_op, filter_f, filter_i, d_err);
+    } else {
+      if (filter_op)
+        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT, 1>), dim3(blocks),
+                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
+                           s->d_fast_ids, n, d_part, interval, offset, start_time,
+                           end_time, filter_op, filter_f, filter_i, d_err);
+      else
+        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT, 0>), dim3(blocks),
+                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
+                           s->d_fast_ids, n, d_part, interval, offset, start_time,
+                           end_time, filter_op, filter_f, filter_i, d_err);
+    }
   }
   if (!s->general_ids.empty()) {
     uint32_t n = (uint32_t)s->general_ids.size();
