@@ -90,6 +90,11 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--mode", choices=["walk", "random"], default="walk")
+    ap.add_argument("--query", choices=["mean", "downsample", "rate"], default="mean",
+                    help="mean: grouped mean/min/max/count GROUP BY time(1m) "
+                         "(north star, config #2); downsample: per-series "
+                         "first/last/sum to 5m buckets (config #4 shape); "
+                         "rate: PromQL rate(value[5m]) step 1m (config #5 shape)")
     ap.add_argument("--series", type=int, default=100_000)
     ap.add_argument("--pts", type=int, default=1000)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
@@ -132,22 +137,47 @@ def main():
 
     n_wins = (args.pts + 59) // 60 + 1
     w0 = 0  # t0=0, windows start at ordinal 0
+    RANGE_NS = 300 * 10**9  # rate(value[5m])
+    DS_NS = 300 * 10**9     # downsample to 5m buckets
 
     def step():
-        rows, stats = shard.scan_agg(0, 2**62, WINDOW_NS, group_all=True)
-        if dist_on:
-            from opengemini_amd.dist import window_partials, merge_across_shards
+        if args.query == "mean":
+            rows, stats = shard.scan_agg(0, 2**62, WINDOW_NS, group_all=True)
+            if dist_on:
+                from opengemini_amd.dist import window_partials, merge_across_shards
 
-            p = window_partials(rows, WINDOW_NS, 0, w0, n_wins)
-            merge_across_shards(p, device=f"cuda:{local_rank}")
+                p = window_partials(rows, WINDOW_NS, 0, w0, n_wins)
+                merge_across_shards(p, device=f"cuda:{local_rank}")
+        elif args.query == "downsample":
+            # per-series first/last/sum partials — the FileSequenceAggregator
+            # reduce shape (engine/record_plan.go:1184); output stays
+            # per-series (the downsample writer consumes it), no collective
+            rows, stats = shard.scan_agg(0, 2**62, DS_NS, group_all=False)
+        else:  # rate
+            rows, stats = shard.prom_rate(0, (args.pts - 1) * 10**9, RANGE_NS,
+                                          WINDOW_NS)
+            if dist_on:
+                # cross-shard partial-sum of sum(rate()) per step over RCCL
+                import torch
+                import torch.distributed as dist
+
+                nsteps = max(1, (args.pts * 10**9 - RANGE_NS) // WINDOW_NS + 1)
+                sums = np.zeros(int(nsteps))
+                idx = ((rows["ts"] - RANGE_NS) // WINDOW_NS).astype(np.int64)
+                np.add.at(sums, np.clip(idx, 0, int(nsteps) - 1), rows["value"])
+                t = torch.from_numpy(sums).to(f"cuda:{local_rank}")
+                dist.all_reduce(t, op=dist.ReduceOp.SUM)
         return rows, stats
 
     # warmup
     for _ in range(args.warmup):
         rows, stats = step()
 
-    # verification outside the timed region: count checksum
-    assert int(rows["count"].sum()) == args.series * args.pts
+    # verification outside the timed region
+    if args.query in ("mean", "downsample"):
+        assert int(rows["count"].sum()) == args.series * args.pts
+    else:
+        assert len(rows) > 0
 
     if dist_on:
         import torch
@@ -218,9 +248,13 @@ def main():
         "dtype": "f64",
         "data": "synthetic",
         "config": {
-            "workload": "100k series x 1k pts float64 Gorilla, mean/min/max/"
-                        "count GROUP BY time(1m), single TSSP file "
-                        f"(mode={args.mode}, {args.series}x{args.pts})",
+            "workload": {
+                "mean": "100k series x 1k pts float64 Gorilla, mean/min/max/"
+                        "count GROUP BY time(1m), single TSSP file",
+                "downsample": "per-series first/last/sum to 5m buckets "
+                              "(downsample pipeline reduce shape)",
+                "rate": "PromQL rate(value[5m]) step 1m over range vectors",
+            }[args.query] + f" (mode={args.mode}, {args.series}x{args.pts})",
             "window": "1m",
             "series_per_gpu": args.series,
             "points_per_series": args.pts,
