@@ -111,15 +111,25 @@ struct BitR {
   uint64_t hi, lo;
   int have;
 
-  uint64_t pw; /* prefetched word: its load issues one fill earlier than its
-                  consumption, giving ~64 decoded bits of latency cover */
-  int pwbits;
+  uint64_t pw, pw1; /* two prefetched words: one 16-byte load covers 128
+                       decoded bits, halving VMEM instructions on the
+                       stream and keeping a load in flight ahead of use */
+  int pwbits, pw1bits;
 
   __device__ __forceinline__ void preload() {
-    if (len - pos >= 8) {
+    if (len - pos >= 16) {
+      uint64_t two[2];
+      memcpy(two, b + pos, 16); /* unaligned 16B load, legal on CDNA */
+      pw = __builtin_bswap64(two[0]);
+      pw1 = __builtin_bswap64(two[1]);
+      pos += 16;
+      pwbits = 64;
+      pw1bits = 64;
+    } else if (len - pos >= 8) {
       pw = d_u64be(b + pos);
       pos += 8;
       pwbits = 64;
+      pw1bits = 0;
     } else if (pos < len) {
       int rem = (int)(len - pos);
       uint64_t w = 0;
@@ -127,8 +137,10 @@ struct BitR {
       pw = w << ((8 - rem) * 8);
       pos = len;
       pwbits = rem * 8;
+      pw1bits = 0;
     } else {
       pwbits = 0;
+      pw1bits = 0;
     }
   }
 
@@ -152,7 +164,10 @@ struct BitR {
         lo |= pw >> (have - 64);
       }
       have += pwbits;
-      preload();
+      pw = pw1;
+      pwbits = pw1bits;
+      pw1bits = 0;
+      if (!pwbits) preload();
     }
   }
 
@@ -163,6 +178,7 @@ struct BitR {
     hi = lo = 0;
     have = 0;
     pwbits = 0;
+    pw1bits = 0;
     preload();
     fill();
   }
