@@ -231,7 +231,8 @@ def main():
         traffic = tf.get("bytes_per_launch")
 
     cpu_baseline = None
-    if not args.skip_cpu_baseline:
+    if not args.skip_cpu_baseline and args.query == "mean":
+        # oracle leg matches the headline query; other shapes report GPU only
         cpu_baseline = cpu_baseline_leg(gen_mode, args.mode)
 
     out = {

@@ -397,3 +397,46 @@ class TestFilterParity:
         exp = expected_windows(ftruth, INT)
         exp = {k: v for k, v in exp.items() if v["count"] > 0}
         check(rows, exp, F)
+
+
+class TestGroupedFilter:
+    def test_grouped_with_filter(self):
+        import opengemini_amd as gx
+
+        blob, descs = orc.gen_shard(53, 500, 1000, mode=orc.GEN_INT_SMALL)
+        sh = gx.Shard(blob, descs, I)
+        try:
+            gpu, _ = sh.scan_agg(0, 2**62, INT, group_all=True, filter=("gt", 500))
+        finally:
+            sh.close()
+        base = orc.scan_agg_filtered(blob, descs, I, 0, 2**62, INT, "gt", 500)
+        ref = orc.group_merge(base, I, INT)
+        assert len(gpu) == len(ref)
+        assert np.array_equal(gpu["count"], ref["count"])
+        assert np.array_equal(gpu["min"].view(np.int64), ref["min"].view(np.int64))
+        assert np.array_equal(gpu["sum"].view(np.int64), ref["sum"].view(np.int64))
+        assert int(gpu["min"].view(np.int64).min()) >= 501
+
+
+class TestAggCursorSurface:
+    def test_next_agg_batches(self):
+        import opengemini_amd as gx
+
+        blob, descs = orc.gen_shard(54, 100, 1000)
+        sh = gx.Shard(blob, descs, F)
+        try:
+            cur = gx.AggCursor(sh, 0, 2**62, INT, chunk_size=1024)
+            total = 0
+            nbatches = 0
+            while True:
+                batch = cur.next_agg()
+                if batch is None:
+                    break
+                assert len(batch) <= 1024
+                total += len(batch)
+                nbatches += 1
+            ref = orc.scan_agg(blob, descs, F, 0, 2**62, INT)
+            assert total == len(ref)
+            assert nbatches == (total + 1023) // 1024
+        finally:
+            sh.close()
