@@ -1853,6 +1853,7 @@ __global__ void __launch_bounds__(256) k_rate_scan(
     }
 
     int64_t first_open = rq.s0; /* smallest possibly-active ordinal */
+    int64_t first_open_ts = start_sample + rq.s0 * step_ns;
     int vIdx = 0;
     const int t_const = FAST && (ti.kind == 1);
     const int64_t t0c = t_const ? ti.cur : 0;
@@ -1877,16 +1878,12 @@ __global__ void __launch_bounds__(256) k_rate_scan(
       if (!valid) continue;
       if (fv != fv) continue; /* FilterRangeNANPoint */
 
-      /* flush ring slots whose ts < t (their window closed) */
+      /* flush ring slots whose ts < t (their window closed) — compare
+       * against the running first-open sample time, no per-point division */
       if (step_ns > 0) {
-        int64_t want_first = first_open;
-        /* first still-open ordinal: smallest o with ts(o) >= t */
-        if (t > start_sample + first_open * step_ns) {
-          int64_t o = (t - start_sample + step_ns - 1) / step_ns; /* times>=start_sample here */
-          if (t <= start_sample) o = 0;
-          if (o > want_first) want_first = o;
-        }
-        while (first_open < want_first) {
+        while (t > first_open_ts) {
+          int64_t want_first = first_open + 1;
+          first_open_ts += step_ns;
           int j = (int)((first_open - rq.s0) % RATE_W);
           if (j < 0) j += RATE_W;
           RateSlot *sl = &ring[j];
@@ -1902,7 +1899,7 @@ __global__ void __launch_bounds__(256) k_rate_scan(
               sl->ts = INT64_MIN;
             }
           }
-          first_open++;
+          first_open = want_first;
         }
       }
 #pragma unroll
