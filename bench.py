@@ -23,6 +23,17 @@ import os
 import sys
 import time
 
+# divide host cores across ranks before OpenMP initialises (the shard
+# generator is OpenMP-parallel; 8 ranks x all-cores would thrash)
+_world = int(os.environ.get("WORLD_SIZE", "1"))
+if _world > 1 and "OMP_NUM_THREADS" not in os.environ:
+    try:
+        import multiprocessing as _mp
+
+        os.environ["OMP_NUM_THREADS"] = str(max(1, _mp.cpu_count() // _world))
+    except Exception:
+        pass
+
 import numpy as np
 
 REPO = os.path.dirname(os.path.abspath(__file__))
@@ -231,8 +242,8 @@ def main():
         traffic = tf.get("bytes_per_launch")
 
     cpu_baseline = None
-    if not args.skip_cpu_baseline and args.query == "mean":
-        # oracle leg matches the headline query; other shapes report GPU only
+    if not args.skip_cpu_baseline and args.query == "mean" and n_gpus == 1:
+        # contract: cpu_baseline on rank 0 at N=1 only
         cpu_baseline = cpu_baseline_leg(gen_mode, args.mode)
 
     out = {

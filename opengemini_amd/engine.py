@@ -254,8 +254,8 @@ class Shard:
         _check(rc, lib)
         stats = dict(
             decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
-            points=st.points, compressed_bytes=st.compressed_bytes,
-            n_rows=st.n_rows,
+            host_ms=st.h2d_ms, points=st.points,
+            compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
 
