@@ -112,25 +112,15 @@ struct BitR {
   uint64_t hi, lo;
   int have;
 
-  uint64_t pw, pw1; /* two prefetched words: one 16-byte load covers 128
-                       decoded bits, halving VMEM instructions on the
-                       stream and keeping a load in flight ahead of use */
-  int pwbits, pw1bits;
+  uint64_t pw; /* prefetched word: its load issues one fill earlier than
+                  its consumption, covering part of the load latency */
+  int pwbits;
 
   __device__ __forceinline__ void preload() {
-    if (len - pos >= 16) {
-      uint64_t two[2];
-      memcpy(two, b + pos, 16); /* unaligned 16B load, legal on CDNA */
-      pw = __builtin_bswap64(two[0]);
-      pw1 = __builtin_bswap64(two[1]);
-      pos += 16;
-      pwbits = 64;
-      pw1bits = 64;
-    } else if (len - pos >= 8) {
+    if (len - pos >= 8) {
       pw = d_u64be(b + pos);
       pos += 8;
       pwbits = 64;
-      pw1bits = 0;
     } else if (pos < len) {
       int rem = (int)(len - pos);
       uint64_t w = 0;
@@ -138,10 +128,8 @@ struct BitR {
       pw = w << ((8 - rem) * 8);
       pos = len;
       pwbits = rem * 8;
-      pw1bits = 0;
     } else {
       pwbits = 0;
-      pw1bits = 0;
     }
   }
 
@@ -165,10 +153,7 @@ struct BitR {
         lo |= pw >> (have - 64);
       }
       have += pwbits;
-      pw = pw1;
-      pwbits = pw1bits;
-      pw1bits = 0;
-      if (!pwbits) preload();
+      preload();
     }
   }
 
@@ -179,7 +164,6 @@ struct BitR {
     hi = lo = 0;
     have = 0;
     pwbits = 0;
-    pw1bits = 0;
     preload();
     fill();
   }

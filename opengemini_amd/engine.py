@@ -176,6 +176,8 @@ class Shard:
         self._descs = np.ascontiguousarray(descs, dtype=SEG_DESC_DTYPE)
         self.col_type = col_type
         self.n_points = int(self._descs["rows"].sum())
+        self._bound_cache = {}
+        self._n_sids = None
         self.compressed_bytes = int(
             self._descs["data_size"].sum() + self._descs["time_size"].sum()
         )
@@ -203,12 +205,25 @@ class Shard:
         except Exception:
             pass
 
+    def _sid_count(self):
+        if self._n_sids is None:
+            d = self._descs["sid"]
+            self._n_sids = int((np.diff(d) != 0).sum()) + 1 if len(d) else 0
+        return self._n_sids
+
     def _rows_bound(self, interval, offset, grouped):
-        """Exact upper bound on output rows from the descriptors."""
+        """Exact upper bound on output rows from the descriptors (cached)."""
+        key = (interval, offset, grouped)
+        if key in self._bound_cache:
+            return self._bound_cache[key]
+        r = self._rows_bound_compute(interval, offset, grouped)
+        self._bound_cache[key] = r
+        return r
+
+    def _rows_bound_compute(self, interval, offset, grouped):
         d = self._descs
         if interval == 0:
-            n_sids = len(np.unique(d["sid"]))
-            return (1 if grouped else n_sids) + 4
+            return (1 if grouped else self._sid_count()) + 4
         w0 = (d["min_time"] - offset) // interval
         w1 = (d["max_time"] - offset) // interval
         if grouped:
@@ -270,8 +285,7 @@ class Shard:
             nsteps = 1
             if step_ns > 0 and end_time >= start_time + range_ns:
                 nsteps = int((end_time - (start_time + range_ns)) // step_ns) + 2
-            n_sids = len(np.unique(d["sid"]))
-            out_cap = nsteps * n_sids + 16
+            out_cap = nsteps * self._sid_count() + 16
         out = np.empty(out_cap, dtype=RATE_ROW_DTYPE)
         n = C.c_uint64(0)
         st = _Stats()
