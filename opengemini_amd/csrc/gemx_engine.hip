@@ -2431,23 +2431,20 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
-  /* check device error + fetch rows (pinned staging) */
+  /* check device error + fetch rows straight into the caller's buffer
+   * (gap rows compacted in place afterwards — one copy, no pinned bounce) */
   DevErr herr = {0};
   HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
                            s->stream));
-  gemx_agg_row *hrows;
-  uint64_t fetch_rows;
-  if (group_all) {
-    hrows = P.h_grows;
-    fetch_rows = P.n_gwins;
-    HIP_CHECK(hipMemcpyAsync(hrows, P.d_grows, sizeof(gemx_agg_row) * fetch_rows,
-                             hipMemcpyDeviceToHost, s->stream));
-  } else {
-    hrows = P.h_rows;
-    fetch_rows = total_rows;
-    HIP_CHECK(hipMemcpyAsync(hrows, d_rows, sizeof(gemx_agg_row) * fetch_rows,
-                             hipMemcpyDeviceToHost, s->stream));
+  uint64_t fetch_rows = group_all ? P.n_gwins : total_rows;
+  if (fetch_rows > cap) {
+    seterr("output capacity too small");
+    return GEMX_E_CAP;
   }
+  gemx_agg_row *hrows = out_host;
+  HIP_CHECK(hipMemcpyAsync(hrows, group_all ? P.d_grows : d_rows,
+                           sizeof(gemx_agg_row) * fetch_rows,
+                           hipMemcpyDeviceToHost, s->stream));
   auto t_sync0 = std::chrono::steady_clock::now();
   HIP_CHECK(hipStreamSynchronize(s->stream));
   auto t_sync1 = std::chrono::steady_clock::now();
@@ -2466,15 +2463,16 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     return herr.code;
   }
 
-  /* compact gap rows (count == -1) into caller buffer */
+  /* in-place gap compaction (count == -1); no copy until the first gap */
   uint64_t n = 0;
-  for (uint64_t i = 0; i < fetch_rows; i++) {
-    if (hrows[i].count < 0) continue;
-    if (n >= cap) {
-      seterr("output capacity too small");
-      return GEMX_E_CAP;
+  {
+    uint64_t i = 0;
+    while (i < fetch_rows && hrows[i].count >= 0) i++;
+    n = i;
+    for (; i < fetch_rows; i++) {
+      if (hrows[i].count < 0) continue;
+      out_host[n++] = hrows[i];
     }
-    out_host[n++] = hrows[i];
   }
   *n_out = n;
   auto t_comp = std::chrono::steady_clock::now();
@@ -2662,7 +2660,11 @@ extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_tim
   DevErr herr = {0};
   HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
                            s->stream));
-  HIP_CHECK(hipMemcpyAsync(P.h_rrows, P.d_rrows,
+  if (P.total_rows > cap) {
+    seterr("output capacity too small");
+    return GEMX_E_CAP;
+  }
+  HIP_CHECK(hipMemcpyAsync(out_host, P.d_rrows,
                            sizeof(gemx_rate_row) * P.total_rows,
                            hipMemcpyDeviceToHost, s->stream));
   HIP_CHECK(hipStreamSynchronize(s->stream));
@@ -2681,16 +2683,18 @@ extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_tim
     return herr.code;
   }
 
-  /* compact: only non-nil rows leave (the reference appends only non-nil,
-   * prom reducer append path) */
+  /* in-place compact: only non-nil rows leave (the reference appends only
+   * non-nil, prom reducer append path) */
   uint64_t n = 0;
-  for (uint64_t i = 0; i < P.total_rows; i++) {
-    if (P.h_rrows[i].isnil) continue;
-    if (n >= cap) {
-      seterr("output capacity too small");
-      return GEMX_E_CAP;
+  {
+    uint64_t i = 0;
+    while (i < P.total_rows && !out_host[i].isnil) i++;
+    n = i;
+    for (; i < P.total_rows; i++) {
+      if (out_host[i].isnil) continue;
+      out_host[n] = out_host[i];
+      n++;
     }
-    out_host[n++] = P.h_rrows[i];
   }
   *n_out = n;
   if (stats) {
