@@ -177,7 +177,9 @@ class Shard:
         self.col_type = col_type
         self.n_points = int(self._descs["rows"].sum())
         self._bound_cache = {}
-        self._n_sids = None
+        self._out_cache = {}  # pooled output buffers (CircularRecordPool role:
+        self._n_sids = None   # rows are valid until the next query, as the
+                              # reference's pooled records are)
         self.compressed_bytes = int(
             self._descs["data_size"].sum() + self._descs["time_size"].sum()
         )
@@ -204,6 +206,17 @@ class Shard:
             self.close()
         except Exception:
             pass
+
+    def _pooled_out(self, kind, cap, dtype):
+        """Reusable output buffer (warm pages; rows are valid until the next
+        query on this shard — the CircularRecordPool contract,
+        engine/aggregate_cursor.go:100)."""
+        cur = self._out_cache.get(kind)
+        if cur is None or len(cur) < cap or cur.dtype != dtype:
+            cur = np.empty(cap, dtype=dtype)
+            cur[:] = 0  # touch pages once
+            self._out_cache[kind] = cur
+        return cur
 
     def _sid_count(self):
         if self._n_sids is None:
@@ -247,7 +260,7 @@ class Shard:
         lib = self._lib
         if out_cap is None:
             out_cap = self._rows_bound(interval, offset, group_all)
-        out = np.empty(out_cap, dtype=AGG_ROW_DTYPE)
+        out = self._pooled_out("agg", out_cap, AGG_ROW_DTYPE)
         n = C.c_uint64(0)
         st = _Stats()
         if filter is not None:
@@ -286,7 +299,7 @@ class Shard:
             if step_ns > 0 and end_time >= start_time + range_ns:
                 nsteps = int((end_time - (start_time + range_ns)) // step_ns) + 2
             out_cap = nsteps * self._sid_count() + 16
-        out = np.empty(out_cap, dtype=RATE_ROW_DTYPE)
+        out = self._pooled_out("rate", out_cap, RATE_ROW_DTYPE)
         n = C.c_uint64(0)
         st = _Stats()
         rc = lib.gemx_prom_rate(
