@@ -183,6 +183,13 @@ int gemx_prom_rate(gemx_shard *, int64_t start_time, int64_t end_time,
                    int is_counter, gemx_rate_row *out_host, uint64_t cap,
                    uint64_t *n_out, gemx_query_stats *stats);
 
+/* irate (is_rate=1) / idelta (is_rate=0): instantaneous rate from the
+ * window's last two points (prom_functions.go:469-514). */
+int gemx_prom_irate(gemx_shard *, int64_t start_time, int64_t end_time,
+                    int64_t range_ns, int64_t step_ns, int is_rate,
+                    gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,
+                    gemx_query_stats *stats);
+
 #ifdef __cplusplus
 }
 #endif

@@ -1663,6 +1663,8 @@ __global__ void __launch_bounds__(256) k_group_p2(
 struct RatePartial {
   int64_t first_t, last_t;
   double first_v, last_v;
+  int64_t prev_t; /* second-to-last point (irate, prom_functions.go:469) */
+  double prev_v;
   int64_t count;
   double reset_adj; /* Σ pre-reset values within this (segment, window) */
 };
@@ -1697,8 +1699,11 @@ __device__ __forceinline__ void rate_slot_update(RateSlot *s, int64_t t, double 
   if (s->p.count == 0) {
     s->p.first_t = t;
     s->p.first_v = v;
-  } else if (v < s->p.last_v) {
-    s->p.reset_adj += s->p.last_v; /* counter reset, agg_func_prom.go:236-250 */
+  } else {
+    if (v < s->p.last_v)
+      s->p.reset_adj += s->p.last_v; /* counter reset, agg_func_prom.go:236-250 */
+    s->p.prev_t = s->p.last_t;
+    s->p.prev_v = s->p.last_v;
   }
   s->p.last_t = t;
   s->p.last_v = v;
@@ -1903,7 +1908,7 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     const RateSeriesQ *__restrict__ series, uint32_t nseries,
     const RateSegQ *__restrict__ rsegq, const RatePartial *__restrict__ partials,
     gemx_rate_row *__restrict__ rows, uint64_t total_rows, int64_t start_sample,
-    int64_t step_ns, int64_t range_ns, int is_rate, int is_counter) {
+    int64_t step_ns, int64_t range_ns, int is_rate, int is_counter, int func) {
   uint64_t gid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
   for (uint64_t r = gid; r < total_rows; r += gridDim.x * (uint64_t)blockDim.x) {
     uint32_t lo = 0, hi = nseries - 1;
@@ -1937,6 +1942,13 @@ __global__ void __launch_bounds__(256) k_rate_merge(
         /* reset across the segment boundary (within the window) */
         if (p.first_v < acc.last_v) acc.reset_adj += acc.last_v;
         acc.reset_adj += p.reset_adj;
+        if (p.count >= 2) {
+          acc.prev_t = p.prev_t;
+          acc.prev_v = p.prev_v;
+        } else { /* p.count == 1: previous-of-last is acc's last */
+          acc.prev_t = acc.last_t;
+          acc.prev_v = acc.last_v;
+        }
         acc.last_t = p.last_t;
         acc.last_v = p.last_v;
         acc.count += p.count;
@@ -1949,6 +1961,21 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     out.value = 0;
     out.isnil = 1;
     memset(out._pad, 0, sizeof(out._pad));
+    if (func == 1) {
+      /* irate/idelta: last two points (prom_functions.go:479-506) */
+      if (acc.count >= 2 && acc.last_t != acc.prev_t && range_ns != 0) {
+        double rv;
+        if (is_rate && acc.last_v < acc.prev_v)
+          rv = acc.last_v;
+        else
+          rv = acc.last_v - acc.prev_v;
+        if (is_rate) rv /= (double)(acc.last_t - acc.prev_t) / 1e9;
+        out.value = rv;
+        out.isnil = 0;
+      }
+      rows[r] = out;
+      continue;
+    }
     if (acc.count > 1 && acc.last_t != acc.first_t && range_ns != 0) {
       double reduce = (acc.last_v - acc.first_v) + (is_counter ? acc.reset_adj : 0.0);
       int64_t range_start = ts - range_ns;
@@ -2512,11 +2539,11 @@ extern "C" int gemx_scan_agg_grouped(gemx_shard *s, int64_t start_time,
                    out_host, cap, n_out, stats);
 }
 
-extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,
-                              int64_t range_ns, int64_t step_ns, int is_rate,
-                              int is_counter, gemx_rate_row *out_host,
-                              uint64_t cap, uint64_t *n_out,
-                              gemx_query_stats *stats) {
+static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
+                          int64_t range_ns, int64_t step_ns, int is_rate,
+                          int is_counter, int func, gemx_rate_row *out_host,
+                          uint64_t cap, uint64_t *n_out,
+                          gemx_query_stats *stats) {
   if (!s) return GEMX_E_INVALID;
   if (s->col_type != GEMX_TYPE_FLOAT) {
     seterr("prom rate needs a float column");
@@ -2653,7 +2680,7 @@ extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_tim
     hipLaunchKernelGGL(k_rate_merge, dim3(blocks), dim3(TPB), 0, s->stream,
                        P.d_rsq, (uint32_t)P.rsq.size(), P.d_rsegq, P.d_rpart,
                        P.d_rrows, P.total_rows, start_sample, eff_step, range_ns,
-                       is_rate, is_counter);
+                       is_rate, is_counter, func);
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
@@ -2723,4 +2750,24 @@ extern "C" int gemx_scan_agg_ex(gemx_shard *s, int64_t start_time,
                                 uint64_t *n_out, gemx_query_stats *stats) {
   return scan_impl(s, start_time, end_time, interval, offset, group_all,
                    filter_op, filter_f, filter_i, out_host, cap, n_out, stats);
+}
+
+extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,
+                              int64_t range_ns, int64_t step_ns, int is_rate,
+                              int is_counter, gemx_rate_row *out_host,
+                              uint64_t cap, uint64_t *n_out,
+                              gemx_query_stats *stats) {
+  return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, is_rate,
+                        is_counter, 0, out_host, cap, n_out, stats);
+}
+
+/* irate (is_rate=1) / idelta (is_rate=0): instantaneous rate from the
+ * window's last two points (prom_functions.go:469-514). */
+extern "C" int gemx_prom_irate(gemx_shard *s, int64_t start_time,
+                               int64_t end_time, int64_t range_ns,
+                               int64_t step_ns, int is_rate,
+                               gemx_rate_row *out_host, uint64_t cap,
+                               uint64_t *n_out, gemx_query_stats *stats) {
+  return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, is_rate, 0,
+                        1, out_host, cap, n_out, stats);
 }

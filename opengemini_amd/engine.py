@@ -144,6 +144,11 @@ def _load():
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int, C.c_int,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_prom_irate.restype = C.c_int
+    lib.gemx_prom_irate.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     _lib = lib
     return lib
 
@@ -313,6 +318,31 @@ class Shard:
             points=st.points, compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
+
+    def prom_irate(self, start_time, end_time, range_ns, step_ns, is_rate=True,
+                   out_cap=None):
+        """irate()/idelta() — instantaneous rate from the window's last two
+        points (prom_functions.go:469-514)."""
+        lib = self._lib
+        if out_cap is None:
+            nsteps = 1
+            if step_ns > 0 and end_time >= start_time + range_ns:
+                nsteps = int((end_time - (start_time + range_ns)) // step_ns) + 2
+            out_cap = nsteps * self._sid_count() + 16
+        out = self._pooled_out("rate", out_cap, RATE_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_prom_irate(
+            self._h, start_time, end_time, range_ns, step_ns,
+            1 if is_rate else 0,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        return out[: n.value], dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            points=st.points, compressed_bytes=st.compressed_bytes,
+            n_rows=st.n_rows,
+        )
 
 
 class AggCursor:

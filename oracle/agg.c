@@ -1012,3 +1012,94 @@ done:
   free(sv);
   return rc;
 }
+
+/* irate/idelta: last two points of the window
+ * (prom_functions.go:469-514 floatIRateReduce/Merge) */
+int64_t orc_prom_irate(const uint8_t *blob, int64_t blob_len,
+                       const orc_seg_desc *descs, int64_t nsegs, int64_t start,
+                       int64_t end, int64_t range_ns, int64_t step_ns,
+                       int is_rate, orc_rate_row *out, int64_t cap) {
+  if (step_ns < 0 || range_ns <= 0) return -1;
+  int64_t start_sample = start + range_ns;
+  int64_t end_sample =
+      (step_ns == 0) ? start_sample
+                     : start_sample + (end - start_sample) / step_ns * step_ns;
+  if (end < start_sample) return 0;
+
+  orc_val *vals = (orc_val *)malloc(MAX_ROWS_PER_SEG * sizeof(orc_val));
+  int64_t *times = (int64_t *)malloc(MAX_ROWS_PER_SEG * 8);
+  uint8_t *bm = (uint8_t *)malloc(MAX_ROWS_PER_SEG / 8 + 1);
+  int64_t sbuf_cap = 1 << 20;
+  int64_t *st = (int64_t *)malloc(sbuf_cap * 8);
+  double *sv = (double *)malloc(sbuf_cap * 8);
+  int64_t nout = 0, rc = -1;
+
+  int64_t i = 0;
+  while (i < nsegs) {
+    uint64_t sid = descs[i].sid;
+    int64_t npts = 0;
+    for (; i < nsegs && descs[i].sid == sid; i++) {
+      const orc_seg_desc *d = &descs[i];
+      if (d->data_offset + d->data_size > (uint64_t)blob_len ||
+          d->time_offset + d->time_size > (uint64_t)blob_len)
+        goto done;
+      int rows = 0, nil = 0, trows = 0;
+      if (orc_decode_time_segment(blob + d->time_offset, d->time_size, times, &trows))
+        goto done;
+      if (orc_decode_data_segment(ORC_TYPE_FLOAT, blob + d->data_offset,
+                                  d->data_size, vals, bm, &rows, &nil))
+        goto done;
+      if (rows != trows) goto done;
+      int vIdx = 0;
+      for (int r = 0; r < rows; r++) {
+        if (nil > 0 && !((bm[r >> 3] >> (r & 7)) & 1)) continue;
+        double x = vals[vIdx++].f;
+        if (x != x) continue;
+        if (npts >= sbuf_cap) {
+          sbuf_cap *= 2;
+          st = (int64_t *)realloc(st, sbuf_cap * 8);
+          sv = (double *)realloc(sv, sbuf_cap * 8);
+        }
+        st[npts] = times[r];
+        sv[npts] = x;
+        npts++;
+      }
+    }
+    int64_t pi = 0, pj = 0;
+    for (int64_t ts = start_sample; ts <= end_sample;
+         ts += (step_ns ? step_ns : 1)) {
+      int64_t wstart = ts - range_ns;
+      while (pi < npts && st[pi] < wstart) pi++;
+      while (pj < npts && st[pj] <= ts) pj++;
+      int64_t n = pj - pi;
+      if (n >= 2) {
+        int64_t prev_t = st[pj - 2], last_t = st[pj - 1];
+        double prev_v = sv[pj - 2], last_v = sv[pj - 1];
+        if (last_t != prev_t) {
+          double val;
+          if (is_rate && last_v < prev_v)
+            val = last_v; /* counter reset, prom_functions.go:487-488 */
+          else
+            val = last_v - prev_v;
+          if (is_rate) val /= (double)(last_t - prev_t) / 1e9;
+          if (nout >= cap) goto done;
+          out[nout].sid = sid;
+          out[nout].ts = ts;
+          out[nout].value = val;
+          out[nout].isnil = 0;
+          memset(out[nout]._pad, 0, sizeof(out[nout]._pad));
+          nout++;
+        }
+      }
+      if (step_ns == 0) break;
+    }
+  }
+  rc = nout;
+done:
+  free(vals);
+  free(times);
+  free(bm);
+  free(st);
+  free(sv);
+  return rc;
+}

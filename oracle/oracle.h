@@ -231,6 +231,14 @@ int64_t orc_prom_rate(const uint8_t *blob, int64_t blob_len,
                       int64_t end, int64_t range_ns, int64_t step_ns, int is_rate,
                       int is_counter, orc_rate_row *out, int64_t cap);
 
+/* irate/idelta: instantaneous rate from the window's last two points
+ * (prom_functions.go:469-514). is_rate=1 → irate (per-second, counter
+ * reset → lastValue); 0 → idelta. */
+int64_t orc_prom_irate(const uint8_t *blob, int64_t blob_len,
+                       const orc_seg_desc *descs, int64_t nsegs, int64_t start,
+                       int64_t end, int64_t range_ns, int64_t step_ns,
+                       int is_rate, orc_rate_row *out, int64_t cap);
+
 /* Multi-threaded variant (OpenMP over series groups) — the bench.py
  * cpu_baseline leg. nthreads<=0 → all cores. */
 int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,

@@ -440,3 +440,26 @@ class TestAggCursorSurface:
             assert nbatches == (total + 1023) // 1024
         finally:
             sh.close()
+
+
+class TestIrateParity:
+    def test_irate_matches_oracle(self):
+        import opengemini_amd as gx
+
+        S = 10**9
+        blob, descs = orc.gen_shard(81, 2000, 1000)
+        sh = gx.Shard(blob, descs, F)
+        try:
+            gpu, _ = sh.prom_irate(0, 999 * S, 300 * S, 60 * S)
+            gpu = gpu.copy()  # pooled buffer: rows valid until the next query
+            gpu_d, _ = sh.prom_irate(0, 999 * S, 300 * S, 60 * S, is_rate=False)
+            gpu_d = gpu_d.copy()
+        finally:
+            sh.close()
+        ref = orc.prom_irate(blob, descs, 0, 999 * S, 300 * S, 60 * S)
+        assert len(gpu) == len(ref)
+        assert np.array_equal(gpu["sid"], ref["sid"])
+        assert np.array_equal(gpu["ts"], ref["ts"])
+        assert np.array_equal(gpu["value"], ref["value"])  # bit-exact: 2 points
+        ref_d = orc.prom_irate(blob, descs, 0, 999 * S, 300 * S, 60 * S, is_rate=False)
+        assert np.array_equal(gpu_d["value"], ref_d["value"])

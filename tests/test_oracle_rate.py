@@ -107,3 +107,39 @@ class TestRateSemantics:
         a = orc.prom_rate(blob1, d1, 0, 99 * S, 10 * S, 5 * S)
         b = orc.prom_rate(bytes(blobs), d3, 0, 99 * S, 10 * S, 5 * S)
         assert np.array_equal(a.tobytes(), b.tobytes())
+
+
+class TestIrateGolden:
+    """TestIrateFunctions transcribed (prom_range_vector_cursor_test.go:342+):
+    srcRecs1 points v=t so every instantaneous slope is 1."""
+
+    def test_irate_1(self):  # irate(value[5]) start=-3 end=18 step=2 (opt1)
+        blob, d = one_series_shard(SRC_T, SRC_V)
+        rows = orc.prom_irate(blob, d, -3 * S, 18 * S, 5 * S, 2 * S)
+        got = {int(r["ts"] // S): round(float(r["value"]), 10) for r in rows}
+        assert got == {4: 1.0, 6: 1.0, 8: 1.0, 10: 1.0, 12: 1.0, 14: 1.0, 16: 1.0}
+
+    def test_irate_2(self):  # irate(value[3]) start=-1 end=18 step=2 (opt2)
+        blob, d = one_series_shard(SRC_T, SRC_V)
+        rows = orc.prom_irate(blob, d, -1 * S, 18 * S, 3 * S, 2 * S)
+        got = {int(r["ts"] // S): round(float(r["value"]), 10) for r in rows}
+        assert got == {4: 1.0, 6: 1.0, 10: 1.0, 12: 1.0}
+
+    def test_irate_counter_reset(self):
+        t = np.array([1, 2, 3], dtype=np.int64) * S
+        v = np.array([10.0, 2.0, 3.0])
+        blob, d = one_series_shard(t, v)
+        rows = orc.prom_irate(blob, d, 0, 3 * S, 3 * S, 0)
+        # last two: (2,2),(3,3): no reset between them -> 1/s; window [0,3]
+        assert abs(float(rows["value"][0]) - 1.0) < 1e-12
+        v2 = np.array([1.0, 10.0, 3.0])
+        blob2, d2 = one_series_shard(t, v2)
+        rows2 = orc.prom_irate(blob2, d2, 0, 3 * S, 3 * S, 0)
+        # reset: lastValue < prevValue -> resultValue = lastValue = 3 over 1s
+        assert abs(float(rows2["value"][0]) - 3.0) < 1e-12
+
+    def test_idelta(self):
+        blob, d = one_series_shard(SRC_T, SRC_V)
+        r1 = orc.prom_irate(blob, d, -3 * S, 18 * S, 5 * S, 2 * S, is_rate=False)
+        # idelta = lastValue - prevValue without per-second conversion
+        assert all(v in (1.0, 2.0, 4.0) for v in np.round(r1["value"], 9))
