@@ -1691,6 +1691,62 @@ struct RateSeriesQ {
   uint32_t _pad;
 };
 
+
+/* ---- *_over_time family (prom_functions.go:172-342) ----
+ * Reuses the RatePartial slots with a field re-mapping:
+ *   first_v → running value (sum / Kahan mean / min / max / last)
+ *   last_v  → Kahan compensation (sum/avg)
+ *   count   → point count
+ * FUNC codes for the ring/merge templates: */
+#define GEMX_PF_RATE 0
+#define GEMX_PF_IRATE 1
+#define GEMX_PF_SUM_OT 2
+#define GEMX_PF_COUNT_OT 3
+#define GEMX_PF_AVG_OT 4
+#define GEMX_PF_MIN_OT 5
+#define GEMX_PF_MAX_OT 6
+#define GEMX_PF_LAST_OT 7
+
+__device__ __forceinline__ void d_kahan_inc(double inc, double &sum, double &c) {
+  /* executor.KahanSumInc */
+  double t = sum + inc;
+  if (fabs(sum) >= fabs(inc))
+    c += (sum - t) + inc;
+  else
+    c += (inc - t) + sum;
+  sum = t;
+}
+
+template <int FUNC>
+__device__ __forceinline__ void ot_slot_update(RateSlot *s, int64_t t, double v,
+                                               int64_t range_ns) {
+  bool in = (s->ts != INT64_MIN) && t >= s->ts - range_ns && t <= s->ts;
+  if (!in) return;
+  RatePartial *p = &s->p;
+  if (FUNC == GEMX_PF_SUM_OT) {
+    d_kahan_inc(v, p->first_v, p->last_v); /* floatPromSumReduce Kahan */
+  } else if (FUNC == GEMX_PF_COUNT_OT) {
+    /* count only */
+  } else if (FUNC == GEMX_PF_AVG_OT) {
+    /* floatAvgReduce: streaming Kahan mean with Inf carve-outs */
+    double count = (double)(p->count + 1);
+    double mean = p->first_v;
+    bool skip = false;
+    if (isinf(mean)) {
+      if (isinf(v) && (mean > 0) == (v > 0)) skip = true;
+      else if (!isinf(v) && !isnan(v)) skip = true;
+    }
+    if (!skip) d_kahan_inc(v / count - mean / count, p->first_v, p->last_v);
+  } else if (FUNC == GEMX_PF_MIN_OT) {
+    if (p->count == 0 || v < p->first_v || isnan(p->first_v)) p->first_v = v;
+  } else if (FUNC == GEMX_PF_MAX_OT) {
+    if (p->count == 0 || v > p->first_v || isnan(p->first_v)) p->first_v = v;
+  } else if (FUNC == GEMX_PF_LAST_OT) {
+    p->first_v = v;
+  }
+  p->count++;
+}
+
 __device__ __forceinline__ void rate_slot_update(RateSlot *s, int64_t t, double v,
                                                  int64_t range_ns) {
   /* window [ts-range, ts] (closed both ends — :118-153's >= start, <= end) */
@@ -1710,7 +1766,7 @@ __device__ __forceinline__ void rate_slot_update(RateSlot *s, int64_t t, double 
   s->p.count++;
 }
 
-template <int FAST>
+template <int FAST, int FUNC>
 __global__ void __launch_bounds__(256) k_rate_scan(
     const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
     const RateSegQ *__restrict__ rsegq, const uint32_t *__restrict__ seg_ids,
@@ -1745,6 +1801,8 @@ __global__ void __launch_bounds__(256) k_rate_scan(
       ring[j].ord = o;
       ring[j].p.count = 0;
       ring[j].p.reset_adj = 0;
+      ring[j].p.first_v = 0;
+      ring[j].p.last_v = 0;
     }
 
     /* value/time iterators (fast: streaming; general: via scratch) */
@@ -1885,6 +1943,8 @@ __global__ void __launch_bounds__(256) k_rate_scan(
               sl->ts = start_sample + no * step_ns;
               sl->p.count = 0;
               sl->p.reset_adj = 0;
+              sl->p.first_v = 0;
+              sl->p.last_v = 0;
             } else {
               sl->ts = INT64_MIN;
             }
@@ -1893,7 +1953,12 @@ __global__ void __launch_bounds__(256) k_rate_scan(
         }
       }
 #pragma unroll
-      for (int j = 0; j < RATE_W; j++) rate_slot_update(&ring[j], t, fv, range_ns);
+      for (int j = 0; j < RATE_W; j++) {
+        if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE)
+          rate_slot_update(&ring[j], t, fv, range_ns);
+        else
+          ot_slot_update<FUNC>(&ring[j], t, fv, range_ns);
+      }
     }
     /* flush the remaining active slots */
 #pragma unroll
@@ -1924,6 +1989,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     RatePartial acc;
     acc.count = 0;
     acc.reset_adj = 0;
+    acc.first_v = 0;
+    acc.last_v = 0;
     uint32_t a = s.seg_start, b = s.seg_start + s.seg_count;
     uint32_t flo = a, fhi = b;
     while (flo < fhi) {
@@ -1936,6 +2003,42 @@ __global__ void __launch_bounds__(256) k_rate_merge(
       if (o < q.s0 || o >= q.s0 + (int64_t)q.n_steps) continue;
       RatePartial p = partials[q.partial_base + (o - q.s0)];
       if (p.count == 0) continue;
+      if (func >= GEMX_PF_SUM_OT) {
+        /* per-record tail first (reduce returns sum+c / mean+c, then
+         * the merge funcs combine those values — prom_functions.go) */
+        double pv = p.first_v;
+        if ((func == GEMX_PF_SUM_OT || func == GEMX_PF_AVG_OT) && !isinf(pv))
+          pv += p.last_v;
+        if (acc.count == 0) {
+          acc.first_v = pv;
+          acc.count = p.count;
+        } else {
+          switch (func) {
+          case GEMX_PF_SUM_OT:
+          case GEMX_PF_COUNT_OT:
+            acc.first_v += pv;
+            break;
+          case GEMX_PF_AVG_OT: {
+            double pc = (double)acc.count, cc = (double)p.count;
+            acc.first_v = (acc.first_v * pc + pv * cc) / (pc + cc);
+            break;
+          }
+          case GEMX_PF_MIN_OT:
+            if (isnan(acc.first_v) || (!isnan(pv) && pv < acc.first_v))
+              acc.first_v = pv;
+            break;
+          case GEMX_PF_MAX_OT:
+            if (isnan(acc.first_v) || (!isnan(pv) && pv > acc.first_v))
+              acc.first_v = pv;
+            break;
+          case GEMX_PF_LAST_OT:
+            acc.first_v = pv;
+            break;
+          }
+          acc.count += p.count;
+        }
+        continue;
+      }
       if (acc.count == 0) {
         acc = p;
       } else {
@@ -1961,6 +2064,15 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     out.value = 0;
     out.isnil = 1;
     memset(out._pad, 0, sizeof(out._pad));
+    if (func >= GEMX_PF_SUM_OT) {
+      if (acc.count > 0) {
+        /* tails were applied per partial at merge time */
+        out.value = (func == GEMX_PF_COUNT_OT) ? (double)acc.count : acc.first_v;
+        out.isnil = 0;
+      }
+      rows[r] = out;
+      continue;
+    }
     if (func == 1) {
       /* irate/idelta: last two points (prom_functions.go:479-506) */
       if (acc.count >= 2 && acc.last_t != acc.prev_t && range_ns != 0) {
@@ -2660,18 +2772,41 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   if (!s->fast_ids.empty()) {
     uint32_t n = (uint32_t)s->fast_ids.size();
     uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
-    hipLaunchKernelGGL((k_rate_scan<1>), dim3(blocks), dim3(TPB), 0, s->stream,
-                       s->d_blob, s->d_descs, P.d_rsegq, s->d_fast_ids, n,
-                       P.d_rpart, start_sample, eff_step, range_ns, nullptr, 0, 0,
-                       d_err);
+#define LAUNCH_RATE_FAST(F)                                                     \
+    hipLaunchKernelGGL((k_rate_scan<1, F>), dim3(blocks), dim3(TPB), 0,         \
+                       s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
+                       s->d_fast_ids, n, P.d_rpart, start_sample, eff_step,     \
+                       range_ns, nullptr, 0, 0, d_err)
+    switch (func) {
+    case GEMX_PF_SUM_OT: LAUNCH_RATE_FAST(GEMX_PF_SUM_OT); break;
+    case GEMX_PF_COUNT_OT: LAUNCH_RATE_FAST(GEMX_PF_COUNT_OT); break;
+    case GEMX_PF_AVG_OT: LAUNCH_RATE_FAST(GEMX_PF_AVG_OT); break;
+    case GEMX_PF_MIN_OT: LAUNCH_RATE_FAST(GEMX_PF_MIN_OT); break;
+    case GEMX_PF_MAX_OT: LAUNCH_RATE_FAST(GEMX_PF_MAX_OT); break;
+    case GEMX_PF_LAST_OT: LAUNCH_RATE_FAST(GEMX_PF_LAST_OT); break;
+    default: LAUNCH_RATE_FAST(GEMX_PF_RATE); break;
+    }
+#undef LAUNCH_RATE_FAST
   }
   if (!s->general_ids.empty()) {
     uint32_t n = (uint32_t)s->general_ids.size();
     uint32_t blocks = (P.gen_lanes + TPB - 1) / TPB;
-    hipLaunchKernelGGL((k_rate_scan<0>), dim3(blocks), dim3(TPB), 0, s->stream,
-                       s->d_blob, s->d_descs, P.d_rsegq, s->d_general_ids, n,
-                       P.d_rpart, start_sample, eff_step, range_ns, P.d_scratch,
-                       scratch_per_lane, P.gen_lanes, d_err);
+#define LAUNCH_RATE_GEN(F)                                                      \
+    hipLaunchKernelGGL((k_rate_scan<0, F>), dim3(blocks), dim3(TPB), 0,         \
+                       s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
+                       s->d_general_ids, n, P.d_rpart, start_sample, eff_step,  \
+                       range_ns, P.d_scratch, scratch_per_lane, P.gen_lanes,    \
+                       d_err)
+    switch (func) {
+    case GEMX_PF_SUM_OT: LAUNCH_RATE_GEN(GEMX_PF_SUM_OT); break;
+    case GEMX_PF_COUNT_OT: LAUNCH_RATE_GEN(GEMX_PF_COUNT_OT); break;
+    case GEMX_PF_AVG_OT: LAUNCH_RATE_GEN(GEMX_PF_AVG_OT); break;
+    case GEMX_PF_MIN_OT: LAUNCH_RATE_GEN(GEMX_PF_MIN_OT); break;
+    case GEMX_PF_MAX_OT: LAUNCH_RATE_GEN(GEMX_PF_MAX_OT); break;
+    case GEMX_PF_LAST_OT: LAUNCH_RATE_GEN(GEMX_PF_LAST_OT); break;
+    default: LAUNCH_RATE_GEN(GEMX_PF_RATE); break;
+    }
+#undef LAUNCH_RATE_GEN
   }
   HIP_CHECK(hipEventRecord(ev1, s->stream));
   if (P.total_rows > 0) {
@@ -2770,4 +2905,19 @@ extern "C" int gemx_prom_irate(gemx_shard *s, int64_t start_time,
                                uint64_t *n_out, gemx_query_stats *stats) {
   return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, is_rate, 0,
                         1, out_host, cap, n_out, stats);
+}
+
+/* *_over_time (prom_functions.go:172-342): func selects
+ * sum(2)/count(3)/avg(4)/min(5)/max(6)/last(7) over [ts-range, ts]. */
+extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
+                                   int64_t end_time, int64_t range_ns,
+                                   int64_t step_ns, int func,
+                                   gemx_rate_row *out_host, uint64_t cap,
+                                   uint64_t *n_out, gemx_query_stats *stats) {
+  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_LAST_OT) {
+    seterr("unknown over_time func");
+    return GEMX_E_INVALID;
+  }
+  return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, 0, 0, func,
+                        out_host, cap, n_out, stats);
 }

@@ -149,6 +149,11 @@ def _load():
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_prom_over_time.restype = C.c_int
+    lib.gemx_prom_over_time.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     _lib = lib
     return lib
 
@@ -343,6 +348,9 @@ class Shard:
             points=st.points, compressed_bytes=st.compressed_bytes,
             n_rows=st.n_rows,
         )
+
+
+OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6, "last": 7}
 
 
 class AggCursor:

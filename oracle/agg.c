@@ -1103,3 +1103,138 @@ done:
   free(sv);
   return rc;
 }
+
+/* *_over_time (prom_functions.go:172-342): whole-window reduce per the
+ * reference's reduce functions (Kahan sum/mean with Inf carve-outs,
+ * NaN-aware min/max). Computed over the full window slice — the engine
+ * computes per-segment partials then the reference's merge functions, so
+ * float sum/avg parity is within reassociation tolerance (1e-9), the rest
+ * exact. func: 2 sum, 3 count, 4 avg, 5 min, 6 max, 7 last. */
+static void kahan_inc(double inc, double *sum, double *c) {
+  double t = *sum + inc;
+  if (fabs(*sum) >= fabs(inc))
+    *c += (*sum - t) + inc;
+  else
+    *c += (inc - t) + *sum;
+  *sum = t;
+}
+
+int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
+                           const orc_seg_desc *descs, int64_t nsegs,
+                           int64_t start, int64_t end, int64_t range_ns,
+                           int64_t step_ns, int func, orc_rate_row *out,
+                           int64_t cap) {
+  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 7) return -1;
+  int64_t start_sample = start + range_ns;
+  int64_t end_sample =
+      (step_ns == 0) ? start_sample
+                     : start_sample + (end - start_sample) / step_ns * step_ns;
+  if (end < start_sample) return 0;
+
+  orc_val *vals = (orc_val *)malloc(MAX_ROWS_PER_SEG * sizeof(orc_val));
+  int64_t *times = (int64_t *)malloc(MAX_ROWS_PER_SEG * 8);
+  uint8_t *bm = (uint8_t *)malloc(MAX_ROWS_PER_SEG / 8 + 1);
+  int64_t sbuf_cap = 1 << 20;
+  int64_t *st = (int64_t *)malloc(sbuf_cap * 8);
+  double *sv = (double *)malloc(sbuf_cap * 8);
+  int64_t nout = 0, rc = -1;
+
+  int64_t i = 0;
+  while (i < nsegs) {
+    uint64_t sid = descs[i].sid;
+    int64_t npts = 0;
+    for (; i < nsegs && descs[i].sid == sid; i++) {
+      const orc_seg_desc *d = &descs[i];
+      if (d->data_offset + d->data_size > (uint64_t)blob_len ||
+          d->time_offset + d->time_size > (uint64_t)blob_len)
+        goto done;
+      int rows = 0, nil = 0, trows = 0;
+      if (orc_decode_time_segment(blob + d->time_offset, d->time_size, times, &trows))
+        goto done;
+      if (orc_decode_data_segment(ORC_TYPE_FLOAT, blob + d->data_offset,
+                                  d->data_size, vals, bm, &rows, &nil))
+        goto done;
+      if (rows != trows) goto done;
+      int vIdx = 0;
+      for (int r = 0; r < rows; r++) {
+        if (nil > 0 && !((bm[r >> 3] >> (r & 7)) & 1)) continue;
+        double x = vals[vIdx++].f;
+        if (x != x) continue;
+        if (npts >= sbuf_cap) {
+          sbuf_cap *= 2;
+          st = (int64_t *)realloc(st, sbuf_cap * 8);
+          sv = (double *)realloc(sv, sbuf_cap * 8);
+        }
+        st[npts] = times[r];
+        sv[npts] = x;
+        npts++;
+      }
+    }
+    int64_t pi = 0, pj = 0;
+    for (int64_t ts = start_sample; ts <= end_sample;
+         ts += (step_ns ? step_ns : 1)) {
+      int64_t wstart = ts - range_ns;
+      while (pi < npts && st[pi] < wstart) pi++;
+      while (pj < npts && st[pj] <= ts) pj++;
+      int64_t n = pj - pi;
+      if (n >= 1) {
+        double v = 0;
+        switch (func) {
+        case 2: { /* sum_over_time */
+          double s = 0, cc = 0;
+          for (int64_t k = pi; k < pj; k++) kahan_inc(sv[k], &s, &cc);
+          v = isinf(s) ? s : s + cc;
+          break;
+        }
+        case 3:
+          v = (double)n;
+          break;
+        case 4: { /* avg_over_time, floatAvgReduce */
+          double mean = 0, cc = 0, count = 0;
+          for (int64_t k = pi; k < pj; k++) {
+            count++;
+            if (isinf(mean)) {
+              if (isinf(sv[k]) && (mean > 0) == (sv[k] > 0)) continue;
+              if (!isinf(sv[k]) && !isnan(sv[k])) continue;
+            }
+            kahan_inc(sv[k] / count - mean / count, &mean, &cc);
+          }
+          v = isinf(mean) ? mean : mean + cc;
+          break;
+        }
+        case 5: {
+          v = sv[pi];
+          for (int64_t k = pi + 1; k < pj; k++)
+            if (sv[k] < v || isnan(v)) v = sv[k];
+          break;
+        }
+        case 6: {
+          v = sv[pi];
+          for (int64_t k = pi + 1; k < pj; k++)
+            if (sv[k] > v || isnan(v)) v = sv[k];
+          break;
+        }
+        default:
+          v = sv[pj - 1];
+          break;
+        }
+        if (nout >= cap) goto done;
+        out[nout].sid = sid;
+        out[nout].ts = ts;
+        out[nout].value = v;
+        out[nout].isnil = 0;
+        memset(out[nout]._pad, 0, sizeof(out[nout]._pad));
+        nout++;
+      }
+      if (step_ns == 0) break;
+    }
+  }
+  rc = nout;
+done:
+  free(vals);
+  free(times);
+  free(bm);
+  free(st);
+  free(sv);
+  return rc;
+}

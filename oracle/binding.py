@@ -133,6 +133,9 @@ def load():
         C.c_uint64, C.c_uint64, C.c_uint64, C.c_uint32, i64, i64, C.c_int,
         u8p, i64, C.c_void_p, i64, i64p,
     ]
+    lib.orc_prom_over_time.restype = i64
+    lib.orc_prom_over_time.argtypes = [u8p, i64, C.c_void_p, i64, i64, i64, i64, i64,
+                                       C.c_int, C.c_void_p, i64]
     lib.orc_prom_irate.restype = i64
     lib.orc_prom_irate.argtypes = [u8p, i64, C.c_void_p, i64, i64, i64, i64, i64,
                                    C.c_int, C.c_void_p, i64]
@@ -372,6 +375,27 @@ def prom_rate(blob, descs, start, end, range_ns, step_ns, is_rate=True, is_count
     )
     if n < 0:
         raise ValueError("prom_rate failed")
+    return out[:n].copy()
+
+
+OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6, "last": 7}
+
+
+def prom_over_time(blob, descs, start, end, range_ns, step_ns, func, cap=None):
+    lib = get()
+    bts = np.frombuffer(blob, dtype=np.uint8)
+    d = np.ascontiguousarray(descs, dtype=SEG_DESC_DTYPE)
+    if cap is None:
+        nsteps = 1 if step_ns == 0 else int((end - (start + range_ns)) // step_ns + 2) if end >= start + range_ns else 1
+        cap = nsteps * len(np.unique(d["sid"])) + 16
+    out = np.zeros(cap, dtype=RATE_ROW_DTYPE)
+    n = lib.orc_prom_over_time(
+        _u8(bts), len(bts), d.ctypes.data_as(C.c_void_p), len(d),
+        start, end, range_ns, step_ns, OT_FUNCS[func],
+        out.ctypes.data_as(C.c_void_p), cap,
+    )
+    if n < 0:
+        raise ValueError("prom_over_time failed")
     return out[:n].copy()
 
 

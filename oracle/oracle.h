@@ -239,6 +239,14 @@ int64_t orc_prom_irate(const uint8_t *blob, int64_t blob_len,
                        int64_t end, int64_t range_ns, int64_t step_ns,
                        int is_rate, orc_rate_row *out, int64_t cap);
 
+/* *_over_time family (prom_functions.go:172-342): func 2 sum, 3 count,
+ * 4 avg, 5 min, 6 max, 7 last over [ts-range, ts] windows. */
+int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
+                           const orc_seg_desc *descs, int64_t nsegs,
+                           int64_t start, int64_t end, int64_t range_ns,
+                           int64_t step_ns, int func, orc_rate_row *out,
+                           int64_t cap);
+
 /* Multi-threaded variant (OpenMP over series groups) — the bench.py
  * cpu_baseline leg. nthreads<=0 → all cores. */
 int64_t orc_scan_agg_mt(const uint8_t *blob, int64_t blob_len,

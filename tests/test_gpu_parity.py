@@ -463,3 +463,27 @@ class TestIrateParity:
         assert np.array_equal(gpu["value"], ref["value"])  # bit-exact: 2 points
         ref_d = orc.prom_irate(blob, descs, 0, 999 * S, 300 * S, 60 * S, is_rate=False)
         assert np.array_equal(gpu_d["value"], ref_d["value"])
+
+
+class TestOverTimeParity:
+    def test_all_funcs(self):
+        import opengemini_amd as gx
+
+        S = 10**9
+        blob, descs = orc.gen_shard(82, 1000, 1000)
+        sh = gx.Shard(blob, descs, F)
+        try:
+            for func in ("sum", "count", "avg", "min", "max", "last"):
+                gpu, _ = sh.prom_over_time(0, 999 * S, 300 * S, 60 * S, func)
+                gpu = gpu.copy()
+                ref = orc.prom_over_time(blob, descs, 0, 999 * S, 300 * S, 60 * S, func)
+                assert len(gpu) == len(ref), func
+                assert np.array_equal(gpu["sid"], ref["sid"]), func
+                assert np.array_equal(gpu["ts"], ref["ts"]), func
+                if func in ("sum", "avg"):
+                    tol = 1e-9 * np.maximum(1.0, np.abs(ref["value"]))
+                    assert np.all(np.abs(gpu["value"] - ref["value"]) <= tol), func
+                else:
+                    assert np.array_equal(gpu["value"], ref["value"]), func
+        finally:
+            sh.close()

@@ -143,3 +143,32 @@ class TestIrateGolden:
         r1 = orc.prom_irate(blob, d, -3 * S, 18 * S, 5 * S, 2 * S, is_rate=False)
         # idelta = lastValue - prevValue without per-second conversion
         assert all(v in (1.0, 2.0, 4.0) for v in np.round(r1["value"], 9))
+
+
+class TestOverTime:
+    """sum/count/avg/min/max/last_over_time vs numpy over the same windows."""
+
+    def _numpy_windows(self, t, v, start, end, rng_ns, step):
+        ss = start + rng_ns
+        es = ss + (end - ss) // step * step
+        out = {}
+        for ts in range(ss, es + 1, step):
+            m = (t >= ts - rng_ns) & (t <= ts)
+            if m.sum():
+                out[ts] = v[m]
+        return out
+
+    def test_all_funcs(self):
+        rng = np.random.default_rng(90)
+        t = (np.arange(200, dtype=np.int64) * 3 + 1) * S
+        v = rng.normal(10, 4, 200)
+        blob, d = one_series_shard(t, v)
+        wins = self._numpy_windows(t, v, 0, int(t[-1]), 60 * S, 20 * S)
+        for func, npf in (("sum", np.sum), ("count", len), ("avg", np.mean),
+                          ("min", np.min), ("max", np.max), ("last", lambda x: x[-1])):
+            rows = orc.prom_over_time(blob, d, 0, int(t[-1]), 60 * S, 20 * S, func)
+            got = {int(r["ts"]): float(r["value"]) for r in rows}
+            assert set(got) == set(wins), func
+            for ts, w in wins.items():
+                ref = float(npf(w))
+                assert abs(got[ts] - ref) <= 1e-9 * max(1.0, abs(ref)), (func, ts)
