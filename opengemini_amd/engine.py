@@ -349,6 +349,30 @@ class Shard:
             n_rows=st.n_rows,
         )
 
+    def prom_over_time(self, start_time, end_time, range_ns, step_ns, func,
+                       out_cap=None):
+        """sum/count/avg/min/max/last_over_time (prom_functions.go:172-342)."""
+        lib = self._lib
+        if out_cap is None:
+            nsteps = 1
+            if step_ns > 0 and end_time >= start_time + range_ns:
+                nsteps = int((end_time - (start_time + range_ns)) // step_ns) + 2
+            out_cap = nsteps * self._sid_count() + 16
+        out = self._pooled_out("rate", out_cap, RATE_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_prom_over_time(
+            self._h, start_time, end_time, range_ns, step_ns, OT_FUNCS[func],
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        return out[: n.value], dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            points=st.points, compressed_bytes=st.compressed_bytes,
+            n_rows=st.n_rows,
+        )
+
+
 
 OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6, "last": 7}
 
