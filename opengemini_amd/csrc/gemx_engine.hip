@@ -763,6 +763,7 @@ __global__ void __launch_bounds__(256) k_scan_fast(
     uint32_t si = seg_ids[li];
     const gemx_seg_desc d = descs[si];
     const SegQ sq = segq[si];
+    if (sq.n_wins == 0) continue; /* segment outside the query range */
 
     /* time segment: [BlockIntegerFull][rows u32][Time enc] or
      * [BlockIntegerOne][8B raw LE] (chunkdata_builder.go:91-95) */
@@ -1007,9 +1008,9 @@ __global__ void __launch_bounds__(256) k_scan_general(
     const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
     const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
     uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
-    int64_t offset, int filter_op, double filter_f, int64_t filter_i,
-    uint8_t *__restrict__ scratch, uint64_t scratch_per_lane, uint32_t nlanes,
-    DevErr *err) {
+    int64_t offset, int64_t q_start, int64_t q_end, int filter_op,
+    double filter_f, int64_t filter_i, uint8_t *__restrict__ scratch,
+    uint64_t scratch_per_lane, uint32_t nlanes, DevErr *err) {
   uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
   if (gid >= nlanes) return;
   uint8_t *my = scratch + (uint64_t)gid * scratch_per_lane;
@@ -1021,6 +1022,7 @@ __global__ void __launch_bounds__(256) k_scan_general(
     uint32_t si = seg_ids[li];
     const gemx_seg_desc d = descs[si];
     const SegQ sq = segq[si];
+    if (sq.n_wins == 0) continue; /* segment outside the query range */
     int rows = (int)d.rows;
     if (rows > 4096) { set_err(err, GEMX_E_INVALID); return; }
 
@@ -1116,13 +1118,50 @@ __global__ void __launch_bounds__(256) k_scan_general(
       Partial *pb0 = partials + sq.partial_base;
       for (uint32_t k = 0; k < sq.n_wins; k++) pb0[k].has_rows = 0;
     }
+    if (filter_op == 0 && (d.min_time < q_start || d.max_time > q_end)) {
+      /* time slicing KEEPS nil rows inside the range (record slicing,
+       * immutable/location.go) — rebuild a compact bitmap in scratch */
+      uint8_t *nb = my + 2 * 4096 * 8 + 40960; /* 512B spare */
+      for (int k2 = 0; k2 < 512; k2++) nb[k2] = 0;
+      int w_rows = 0, w_vals = 0, vi3 = 0;
+      for (int r2 = 0; r2 < rows; r2++) {
+        int valid3 = 1;
+        if (h.bitmap) valid3 = bm_valid(&h, r2);
+        else if (nilcount == rows && rows > 0) valid3 = 0;
+        if (tbuf[r2] < q_start || tbuf[r2] > q_end) {
+          if (valid3) vi3++;
+          continue;
+        }
+        if (valid3) {
+          vbuf[w_vals++] = vbuf[vi3++];
+          nb[w_rows >> 3] |= (uint8_t)(1u << (w_rows & 7));
+        }
+        tbuf[w_rows] = tbuf[r2];
+        w_rows++;
+      }
+      rows = w_rows;
+      dense = w_vals;
+      nilcount = w_rows - w_vals;
+      if (nilcount > 0) {
+        h.bitmap = nb;
+        h.bm_off = 0;
+      } else {
+        h.bitmap = nullptr;
+      }
+      if (rows == 0) continue;
+    }
     if (filter_op != 0) {
-      /* FilterByField: failing rows (incl. nils) removed before aggregation */
+      /* FilterByField: failing rows (incl. nils) removed before aggregation;
+       * out-of-range rows pruned in the same pass */
       int w = 0, vi2 = 0;
       for (int r2 = 0; r2 < rows; r2++) {
         int valid2 = 1;
         if (h.bitmap) valid2 = bm_valid(&h, r2);
         else if (nilcount == rows && rows > 0) valid2 = 0;
+        if (tbuf[r2] < q_start || tbuf[r2] > q_end) {
+          if (valid2) vi2++;
+          continue;
+        }
         if (!valid2) continue;
         double xf = 0;
         int64_t xi = 0;
@@ -2153,6 +2192,11 @@ struct QueryPlan {
   /* grouped output (all-series GROUP BY time) */
   gemx_agg_row *d_grows = nullptr;
   gemx_agg_row *h_grows = nullptr;
+  /* query-scoped segment routing when the time range clips segments:
+   * boundary-crossing segments go through the general (slicing) kernel */
+  bool clipped = false;
+  uint32_t *d_fast_q = nullptr, *d_gen_q = nullptr;
+  uint32_t n_fast_q = 0, n_gen_q = 0;
   void *d_gtmp = nullptr; /* GAcc[n_gwins × gsplit] */
   uint32_t gsplit = 1, gper_chunk = 1;
   int64_t W0 = 0;
@@ -2220,6 +2264,8 @@ static void free_plan(QueryPlan &p) {
   if (p.d_grows) hipFree(p.d_grows);
   if (p.h_grows) hipHostFree(p.h_grows);
   if (p.d_gtmp) hipFree(p.d_gtmp);
+  if (p.d_fast_q) hipFree(p.d_fast_q);
+  if (p.d_gen_q) hipFree(p.d_gen_q);
   p = QueryPlan();
 }
 
@@ -2399,28 +2445,71 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.sq.resize(s->series_ranges.size());
     P.partial_slots = 0;
     P.total_rows = 0;
+    std::vector<uint32_t> fast_q, gen_q;
+    std::vector<char> is_gen(nsegs, 0);
+    for (auto id : s->general_ids) is_gen[id] = 1;
+    bool any_clip = false;
     for (size_t g = 0; g < s->series_ranges.size(); g++) {
       auto &r = s->series_ranges[g];
       int64_t wmin = INT64_MAX, wmax = INT64_MIN;
       for (uint32_t i = r.start; i < r.start + r.count; i++) {
         const gemx_seg_desc &d = s->h_descs[i];
-        int64_t w0 = interval ? win_ordinal(d.min_time, interval, offset) : 0;
-        int64_t w1 = interval ? win_ordinal(d.max_time, interval, offset) : 0;
-        P.segq[i].w_first = w0;
-        P.segq[i].n_wins = (uint32_t)(w1 - w0 + 1);
         P.segq[i].partial_base = P.partial_slots;
         P.segq[i].series_idx = (uint32_t)g;
+        if (d.max_time < start_time || d.min_time > end_time) {
+          /* fully outside the query range: skipped */
+          P.segq[i].w_first = 0;
+          P.segq[i].n_wins = 0;
+          any_clip = true;
+          continue;
+        }
+        int64_t mt = std::max(d.min_time, start_time);
+        int64_t xt = std::min(d.max_time, end_time);
+        bool clip = (d.min_time < start_time || d.max_time > end_time);
+        int64_t w0 = interval ? win_ordinal(mt, interval, offset) : 0;
+        int64_t w1 = interval ? win_ordinal(xt, interval, offset) : 0;
+        P.segq[i].w_first = w0;
+        P.segq[i].n_wins = (uint32_t)(w1 - w0 + 1);
         P.partial_slots += P.segq[i].n_wins;
         wmin = std::min(wmin, w0);
         wmax = std::max(wmax, w1);
+        if (clip) {
+          any_clip = true;
+          gen_q.push_back(i); /* boundary rows sliced in the general kernel */
+        } else if (is_gen[i]) {
+          gen_q.push_back(i);
+        } else {
+          fast_q.push_back(i);
+        }
       }
       P.sq[g].sid = r.sid;
+      if (wmin == INT64_MAX) { /* series entirely out of range */
+        wmin = 0;
+        wmax = -1;
+      }
       P.sq[g].w_min = wmin;
       P.sq[g].out_base = P.total_rows;
       P.sq[g].n_wins = (uint32_t)(wmax - wmin + 1);
       P.sq[g].seg_start = r.start;
       P.sq[g].seg_count = r.count;
       P.total_rows += P.sq[g].n_wins;
+    }
+    P.clipped = any_clip;
+    if (any_clip) {
+      P.n_fast_q = (uint32_t)fast_q.size();
+      P.n_gen_q = (uint32_t)gen_q.size();
+      HIP_CHECK(hipMalloc(&P.d_fast_q,
+                          sizeof(uint32_t) * (fast_q.empty() ? 1 : fast_q.size())));
+      if (!fast_q.empty())
+        HIP_CHECK(hipMemcpyAsync(P.d_fast_q, fast_q.data(),
+                                 sizeof(uint32_t) * fast_q.size(),
+                                 hipMemcpyHostToDevice, s->stream));
+      HIP_CHECK(hipMalloc(&P.d_gen_q,
+                          sizeof(uint32_t) * (gen_q.empty() ? 1 : gen_q.size())));
+      if (!gen_q.empty())
+        HIP_CHECK(hipMemcpyAsync(P.d_gen_q, gen_q.data(),
+                                 sizeof(uint32_t) * gen_q.size(),
+                                 hipMemcpyHostToDevice, s->stream));
     }
     HIP_CHECK(hipMalloc(&P.d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
     HIP_CHECK(hipMemcpyAsync(P.d_segq, P.segq.data(), sizeof(SegQ) * nsegs,
@@ -2494,45 +2583,68 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
 
   HIP_CHECK(hipEventRecord(ev0, s->stream));
   const int TPB = 256;
-  if (!s->fast_ids.empty()) {
-    uint32_t n = (uint32_t)s->fast_ids.size();
+  const uint32_t *fast_list = s->d_fast_ids;
+  uint32_t fast_n = (uint32_t)s->fast_ids.size();
+  const uint32_t *gen_list = s->d_general_ids;
+  uint32_t gen_n = (uint32_t)s->general_ids.size();
+  if (P.clipped) {
+    fast_list = P.d_fast_q;
+    fast_n = P.n_fast_q;
+    gen_list = P.d_gen_q;
+    gen_n = P.n_gen_q;
+  }
+  if (fast_n > 0) {
+    uint32_t n = fast_n;
     uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
     if (s->col_type == GEMX_TYPE_FLOAT) {
       if (filter_op)
         hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT, 1>), dim3(blocks),
                            dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           s->d_fast_ids, n, d_part, interval, offset, start_time,
-                           end_time, filter_op, filter_f, filter_i, d_err);
+                           (const uint32_t *)fast_list, n, d_part, interval,
+                           offset, start_time, end_time, filter_op, filter_f,
+                           filter_i, d_err);
       else
         hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT, 0>), dim3(blocks),
                            dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           s->d_fast_ids, n, d_part, interval, offset, start_time,
-                           end_time, filter_op, filter_f, filter_i, d_err);
+                           (const uint32_t *)fast_list, n, d_part, interval,
+                           offset, start_time, end_time, filter_op, filter_f,
+                           filter_i, d_err);
     } else {
       if (filter_op)
         hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT, 1>), dim3(blocks),
                            dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           s->d_fast_ids, n, d_part, interval, offset, start_time,
-                           end_time, filter_op, filter_f, filter_i, d_err);
+                           (const uint32_t *)fast_list, n, d_part, interval,
+                           offset, start_time, end_time, filter_op, filter_f,
+                           filter_i, d_err);
       else
         hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT, 0>), dim3(blocks),
                            dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           s->d_fast_ids, n, d_part, interval, offset, start_time,
-                           end_time, filter_op, filter_f, filter_i, d_err);
+                           (const uint32_t *)fast_list, n, d_part, interval,
+                           offset, start_time, end_time, filter_op, filter_f,
+                           filter_i, d_err);
     }
   }
-  if (!s->general_ids.empty()) {
-    uint32_t n = (uint32_t)s->general_ids.size();
+  if (gen_n > 0) {
+    if (!d_scratch) {
+      /* clipping can route fast segments here on shards that had none */
+      P.gen_lanes = (uint32_t)std::min<uint32_t>(gen_n, 16384);
+      HIP_CHECK(hipMalloc(&P.d_scratch, scratch_per_lane * P.gen_lanes));
+      d_scratch = P.d_scratch;
+      gen_lanes = P.gen_lanes;
+    }
+    uint32_t n = gen_n;
     uint32_t blocks = (gen_lanes + TPB - 1) / TPB;
     if (s->col_type == GEMX_TYPE_FLOAT)
       hipLaunchKernelGGL((k_scan_general<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
-                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_general_ids, n,
-                         d_part, interval, offset, filter_op, filter_f, filter_i,
+                         s->stream, s->d_blob, s->d_descs, d_segq,
+                         (const uint32_t *)gen_list, n, d_part, interval, offset,
+                         start_time, end_time, filter_op, filter_f, filter_i,
                          d_scratch, scratch_per_lane, gen_lanes, d_err);
     else
       hipLaunchKernelGGL((k_scan_general<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
-                         s->stream, s->d_blob, s->d_descs, d_segq, s->d_general_ids, n,
-                         d_part, interval, offset, filter_op, filter_f, filter_i,
+                         s->stream, s->d_blob, s->d_descs, d_segq,
+                         (const uint32_t *)gen_list, n, d_part, interval, offset,
+                         start_time, end_time, filter_op, filter_f, filter_i,
                          d_scratch, scratch_per_lane, gen_lanes, d_err);
   }
   HIP_CHECK(hipEventRecord(ev1, s->stream));

@@ -384,13 +384,23 @@ int64_t orc_scan_agg_f(const uint8_t *blob, int64_t blob_len,
       goto done;
     if (rows != trows || rows > MAX_ROWS_PER_SEG) goto done;
 
+    /* query time range: rows outside [start_time, end_time] never reach the
+     * cursor (Location pruning + record slicing, immutable/location.go) */
+    int clip = (d->min_time < start_time || d->max_time > end_time);
+    if (d->max_time < start_time || d->min_time > end_time) continue;
+
     if (filter_op != 0) {
       /* FilterByField semantics (immutable/location.go:309): failing rows
-       * (incl. nil rows) are removed from the record before aggregation */
+       * (incl. nil rows) are removed from the record before aggregation;
+       * out-of-range rows pruned in the same pass */
       int w = 0, vIdx = 0;
       for (int r2 = 0; r2 < rows; r2++) {
         int valid = (nilcount == 0) || ((nilcount < rows) && bit_at(bm, r2));
         if (nilcount == rows) valid = 0;
+        if (times[r2] < start_time || times[r2] > end_time) {
+          if (valid) vIdx++;
+          continue;
+        }
         if (!valid) continue;
         if (filt_pass(col_type, filter_op, filter_f, filter_i, &vals[vIdx])) {
           vals[w] = vals[vIdx];
@@ -402,6 +412,30 @@ int64_t orc_scan_agg_f(const uint8_t *blob, int64_t blob_len,
       rows = w;
       nilcount = 0;
       memset(bm, 0xFF, (size_t)((rows + 7) / 8));
+      if (rows == 0) continue;
+    } else if (clip) {
+      /* time slicing alone KEEPS nil rows inside the range (the record is
+       * sliced, not value-filtered) */
+      int w_rows = 0, w_vals = 0, vIdx = 0;
+      uint8_t nb[MAX_ROWS_PER_SEG / 8 + 1];
+      memset(nb, 0, sizeof(nb));
+      for (int r2 = 0; r2 < rows; r2++) {
+        int valid = (nilcount == 0) || ((nilcount < rows) && bit_at(bm, r2));
+        if (nilcount == rows) valid = 0;
+        if (times[r2] < start_time || times[r2] > end_time) {
+          if (valid) vIdx++;
+          continue;
+        }
+        if (valid) {
+          vals[w_vals++] = vals[vIdx++];
+          nb[w_rows >> 3] |= (uint8_t)(1u << (w_rows & 7));
+        }
+        times[w_rows] = times[r2];
+        w_rows++;
+      }
+      rows = w_rows;
+      nilcount = w_rows - w_vals;
+      memcpy(bm, nb, (size_t)((w_rows + 7) / 8) + 1);
       if (rows == 0) continue;
     }
     int dense = rows - nilcount;

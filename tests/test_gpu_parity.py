@@ -487,3 +487,63 @@ class TestOverTimeParity:
                     assert np.array_equal(gpu["value"], ref["value"]), func
         finally:
             sh.close()
+
+
+class TestTimeRangeClipping:
+    """Query [start,end] bounds: rows outside never reach the aggregation
+    (Location pruning + record slicing semantics)."""
+
+    def _run(self, blob, descs, col_type, start, end, interval=INT):
+        import opengemini_amd as gx
+
+        sh = gx.Shard(blob, descs, col_type)
+        try:
+            gpu, _ = sh.scan_agg(start, end, interval)
+            gpu = gpu.copy()
+        finally:
+            sh.close()
+        ref = orc.scan_agg(blob, descs, col_type, start, end, interval)
+        assert_parity(gpu, ref, col_type)
+
+    def test_mid_segment_cut(self):
+        S = 10**9
+        blob, descs = orc.gen_shard(91, 50, 1000)
+        # cut inside segments: [250s, 750s]
+        self._run(blob, descs, F, 250 * S, 750 * S)
+
+    def test_cut_with_nulls(self):
+        rng = np.random.default_rng(92)
+        blob, d, _ = build_shard(rng, F, [1, 2, 3])
+        tmin = int(d["min_time"].min())
+        tmax = int(d["max_time"].max())
+        span = tmax - tmin
+        self._run(blob, d, F, tmin + span // 4, tmax - span // 4)
+
+    def test_everything_outside(self):
+        blob, descs = orc.gen_shard(93, 20, 1000)
+        import opengemini_amd as gx
+
+        sh = gx.Shard(blob, descs, F)
+        try:
+            gpu, _ = sh.scan_agg(10**15, 2 * 10**15, INT)
+        finally:
+            sh.close()
+        assert len(gpu) == 0
+
+    def test_grouped_clipped(self):
+        import opengemini_amd as gx
+
+        S = 10**9
+        blob, descs = orc.gen_shard(94, 200, 1000)
+        sh = gx.Shard(blob, descs, F)
+        try:
+            gpu, _ = sh.scan_agg(100 * S, 899 * S, INT, group_all=True)
+            gpu = gpu.copy()
+        finally:
+            sh.close()
+        base = orc.scan_agg(blob, descs, F, 100 * S, 899 * S, INT)
+        ref = orc.group_merge(base, F, INT)
+        assert len(gpu) == len(ref)
+        assert np.array_equal(gpu["count"], ref["count"])
+        assert np.array_equal(gpu["min"], ref["min"])
+        assert int(gpu["count"].sum()) == 200 * 800  # rows 100..899 per series
