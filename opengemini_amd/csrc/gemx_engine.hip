@@ -2434,7 +2434,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   if (!s) return GEMX_E_INVALID;
   HIP_CHECK(hipSetDevice(s->device));
   const uint64_t nsegs = s->nsegs;
-  const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960;
+  const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960 + 512; /* +512: clip bitmap */
 
   QueryPlan &P = s->plan;
   if (!P.valid || P.start != start_time || P.end != end_time ||
