@@ -1714,6 +1714,17 @@ struct RateSlot {
   RatePartial p;
 };
 
+/* rate/irate window aggregates are prefix differences over the time-sorted
+ * stream: count_o = C(last)-C(first-1), resetAdj_o = R(last)-R(first),
+ * first = the first point >= ts(o)-range, last/prev = shared stream tail.
+ * A window only needs a SNAPSHOT at its first in-window point. */
+struct RateSnap {
+  int64_t first_t;
+  double first_v;
+  int64_t count_at;  /* stream count AFTER the first point */
+  double reset_at;   /* stream resetAdj AFTER the first point's pair */
+};
+
 struct RateSegQ {
   int64_t s0;            /* first step ordinal this segment contributes to */
   uint64_t partial_base; /* slot base */
@@ -1941,6 +1952,18 @@ __global__ void __launch_bounds__(256) k_rate_scan(
 
     int64_t first_open = rq.s0; /* smallest possibly-active ordinal */
     int64_t first_open_ts = start_sample + rq.s0 * step_ns;
+    /* shared stream state + snapshot ring (RATE/IRATE fast path) */
+    RateSnap snap[RATE_W];
+#pragma unroll
+    for (int j = 0; j < RATE_W; j++) snap[j].count_at = -1; /* unopened */
+    int64_t sh_count = 0;
+    double sh_reset = 0;
+    int64_t sh_last_t = 0, sh_prev_t = 0;
+    double sh_last_v = 0, sh_prev_v = 0;
+    int64_t hi_open = rq.s0 - 1; /* highest ordinal whose window has begun */
+    int64_t next_hi_start =
+        start_sample + rq.s0 * step_ns - range_ns; /* ts(hi_open+1)-range */
+    const int64_t last_ord_seg = rq.s0 + (int64_t)rq.n_steps - 1;
     int vIdx = 0;
     const int t_const = FAST && (ti.kind == 1);
     const int64_t t0c = t_const ? ti.cur : 0;
@@ -1965,44 +1988,99 @@ __global__ void __launch_bounds__(256) k_rate_scan(
       if (!valid) continue;
       if (fv != fv) continue; /* FilterRangeNANPoint */
 
-      /* flush ring slots whose ts < t (their window closed) — compare
-       * against the running first-open sample time, no per-point division */
-      if (step_ns > 0) {
-        while (t > first_open_ts) {
-          int64_t want_first = first_open + 1;
-          first_open_ts += step_ns;
+      if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE) {
+        /* close windows whose ts < t: emit from shared − snapshot */
+        while (t > first_open_ts && first_open <= last_ord_seg) {
           int j = (int)((first_open - rq.s0) % RATE_W);
           if (j < 0) j += RATE_W;
-          RateSlot *sl = &ring[j];
-          if (sl->ts != INT64_MIN && sl->ord == first_open) {
-            base[sl->ord - rq.s0] = sl->p;
-            int64_t no = sl->ord + RATE_W;
-            if (no < rq.s0 + (int64_t)rq.n_steps) {
-              sl->ord = no;
-              sl->ts = start_sample + no * step_ns;
-              sl->p.count = 0;
-              sl->p.reset_adj = 0;
-              sl->p.first_v = 0;
-              sl->p.last_v = 0;
-            } else {
-              sl->ts = INT64_MIN;
-            }
-          }
-          first_open = want_first;
+          RatePartial *p = &base[first_open - rq.s0];
+          if (snap[j].count_at >= 0) {
+            p->first_t = snap[j].first_t;
+            p->first_v = snap[j].first_v;
+            p->prev_t = sh_prev_t;
+            p->prev_v = sh_prev_v;
+            p->last_t = sh_last_t;
+            p->last_v = sh_last_v;
+            p->count = sh_count - snap[j].count_at + 1;
+            p->reset_adj = sh_reset - snap[j].reset_at;
+            snap[j].count_at = -1;
+          } /* else: window had no points — slot stays count 0 */
+          first_open++;
+          first_open_ts += step_ns;
         }
-      }
+        /* stream update (one per point, all open windows share it) */
+        if (sh_count > 0 && fv < sh_last_v) sh_reset += sh_last_v;
+        sh_prev_t = sh_last_t;
+        sh_prev_v = sh_last_v;
+        sh_last_t = t;
+        sh_last_v = fv;
+        sh_count++;
+        /* open windows whose start (ts-range) has been reached: this point
+         * is their first in-window point (amortised: steps per segment) */
+        while (hi_open < last_ord_seg && t >= next_hi_start) {
+          hi_open++;
+          next_hi_start += step_ns;
+          if (hi_open < first_open) continue; /* already closed (empty) */
+          int j = (int)((hi_open - rq.s0) % RATE_W);
+          if (j < 0) j += RATE_W;
+          snap[j].first_t = t;
+          snap[j].first_v = fv;
+          snap[j].count_at = sh_count; /* count AFTER this point */
+          snap[j].reset_at = sh_reset; /* pair (prev,this) is not in-window */
+        }
+      } else {
+        /* over_time family keeps the slot ring (no prefix structure) */
+        if (step_ns > 0) {
+          while (t > first_open_ts) {
+            int64_t want_first = first_open + 1;
+            first_open_ts += step_ns;
+            int j = (int)((first_open - rq.s0) % RATE_W);
+            if (j < 0) j += RATE_W;
+            RateSlot *sl = &ring[j];
+            if (sl->ts != INT64_MIN && sl->ord == first_open) {
+              base[sl->ord - rq.s0] = sl->p;
+              int64_t no = sl->ord + RATE_W;
+              if (no < rq.s0 + (int64_t)rq.n_steps) {
+                sl->ord = no;
+                sl->ts = start_sample + no * step_ns;
+                sl->p.count = 0;
+                sl->p.reset_adj = 0;
+                sl->p.first_v = 0;
+                sl->p.last_v = 0;
+              } else {
+                sl->ts = INT64_MIN;
+              }
+            }
+            first_open = want_first;
+          }
+        }
 #pragma unroll
-      for (int j = 0; j < RATE_W; j++) {
-        if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE)
-          rate_slot_update(&ring[j], t, fv, range_ns);
-        else
-          ot_slot_update<FUNC>(&ring[j], t, fv, range_ns);
+        for (int j = 0; j < RATE_W; j++) ot_slot_update<FUNC>(&ring[j], t, fv, range_ns);
       }
     }
-    /* flush the remaining active slots */
+    /* flush the remaining open windows */
+    if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE) {
+      for (int64_t o = first_open; o <= last_ord_seg; o++) {
+        int j = (int)((o - rq.s0) % RATE_W);
+        if (j < 0) j += RATE_W;
+        RatePartial *p = &base[o - rq.s0];
+        if (snap[j].count_at >= 0 && o >= hi_open - RATE_W + 1 && o <= hi_open) {
+          p->first_t = snap[j].first_t;
+          p->first_v = snap[j].first_v;
+          p->prev_t = sh_prev_t;
+          p->prev_v = sh_prev_v;
+          p->last_t = sh_last_t;
+          p->last_v = sh_last_v;
+          p->count = sh_count - snap[j].count_at + 1;
+          p->reset_adj = sh_reset - snap[j].reset_at;
+          snap[j].count_at = -1;
+        }
+      }
+    } else {
 #pragma unroll
-    for (int j = 0; j < RATE_W; j++) {
-      if (ring[j].ts != INT64_MIN) base[ring[j].ord - rq.s0] = ring[j].p;
+      for (int j = 0; j < RATE_W; j++) {
+        if (ring[j].ts != INT64_MIN) base[ring[j].ord - rq.s0] = ring[j].p;
+      }
     }
   }
 }
