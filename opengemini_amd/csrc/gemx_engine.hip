@@ -834,6 +834,90 @@ __global__ void __launch_bounds__(256) k_scan_fast(
     const int64_t dtc = t_const ? ti.delta : 0;
     if (t_const && ti.left < rows) { set_err(err, GEMX_E_DECODE); return; }
 
+#define GEMX_DECODE_ONE(fv, iv)                                                   \
+      do {                                                                         \
+        if (h.one_value) {                                                         \
+          if (COLTYPE == GEMX_TYPE_FLOAT)                                          \
+            fv = d_f64le(h.enc);                                                   \
+          else                                                                     \
+            memcpy(&iv, h.enc, 8);                                                 \
+        } else if (COLTYPE == GEMX_TYPE_FLOAT) {                                   \
+          if (fit.next(&fv)) { set_err(err, GEMX_E_DECODE); return; }              \
+        } else {                                                                   \
+          if (iit.next(&iv)) { set_err(err, GEMX_E_DECODE); return; }              \
+        }                                                                          \
+      } while (0)
+
+    if (t_const && dtc >= 0 && !FILT && interval) {
+      /* two-level loop for the regular-grid case: the inner loop runs a
+       * whole window's rows with no time/window arithmetic — min/max track
+       * ROW indices, times reconstruct at flush (t = t0 + row*dt) */
+      int i = 0;
+      while (i < rows) {
+        int64_t t_i = t0c + (int64_t)i * dtc;
+        int64_t ord = win_ordinal(t_i, interval, offset);
+        if (ord < sq.w_first || ord >= sq.w_first + (int64_t)sq.n_wins) {
+          set_err(err, GEMX_E_INVALID);
+          return;
+        }
+        int64_t we = ord * interval + offset + interval;
+        int gend;
+        if (dtc == 0) {
+          gend = rows;
+        } else {
+          int64_t n_in = (we - 1 - t_i) / dtc + 1;
+          gend = (n_in >= (int64_t)(rows - i)) ? rows : i + (int)n_in;
+        }
+        /* first row of the group initialises every accumulator */
+        double fv = 0;
+        int64_t iv = 0;
+        GEMX_DECODE_ONE(fv, iv);
+        double sf = (COLTYPE == GEMX_TYPE_FLOAT) ? fv : 0.0;
+        int64_t si2 = (COLTYPE == GEMX_TYPE_FLOAT) ? 0 : iv;
+        gemx_val mn, mx, fvv, lvv;
+        if (COLTYPE == GEMX_TYPE_FLOAT) {
+          mn.f = mx.f = fvv.f = lvv.f = fv;
+        } else {
+          mn.i = mx.i = fvv.i = lvv.i = iv;
+        }
+        int min_row = i, max_row = i;
+        for (int k = i + 1; k < gend; k++) {
+          GEMX_DECODE_ONE(fv, iv);
+          if (COLTYPE == GEMX_TYPE_FLOAT) {
+            sf += fv;
+            /* first-occurrence-wins strict compares (column_util.go:204-209);
+             * NaN compares false -> never replaces (Go parity) */
+            if (mn.f > fv) { mn.f = fv; min_row = k; }
+            if (mx.f < fv) { mx.f = fv; max_row = k; }
+            lvv.f = fv;
+          } else {
+            si2 += iv;
+            if (mn.i > iv) { mn.i = iv; min_row = k; }
+            if (mx.i < iv) { mx.i = iv; max_row = k; }
+            lvv.i = iv;
+          }
+        }
+        Partial *p = base + (ord - sq.w_first);
+        p->v[0].i = gend - i;
+        if (COLTYPE == GEMX_TYPE_FLOAT) p->v[1].f = sf; else p->v[1].i = si2;
+        p->v[2] = mn;
+        p->v[3] = mx;
+        p->v[4] = fvv;
+        p->v[5] = lvv;
+        p->t[0] = t_i;
+        p->t[1] = t_i; /* no nils ⇒ valueIndex == row index */
+        p->t[2] = t0c + (int64_t)min_row * dtc;
+        p->t[3] = t0c + (int64_t)max_row * dtc;
+        p->t[4] = t_i;
+        p->t[5] = t0c + (int64_t)(gend - 1) * dtc;
+        p->first_row_time = t_i;
+        p->nilmask = 0;
+        p->has_rows = 1;
+        i = gend;
+      }
+      continue;
+    }
+
     for (int i = 0; i < rows; i++) {
       int64_t t;
       if (t_const) {
@@ -942,6 +1026,7 @@ __global__ void __launch_bounds__(256) k_scan_fast(
     (void)q_start;
     (void)q_end;
   }
+#undef GEMX_DECODE_ONE
 }
 
 /* snappy block decode, lane-serial (golang/snappy format;
