@@ -170,6 +170,31 @@ int gemx_scan_agg_ex(gemx_shard *, int64_t start_time, int64_t end_time,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                      gemx_query_stats *stats);
 
+/* Pre-aggregation metadata (the matchPreAgg path,
+ * engine/iterators_helper.go:90): a query with only pre-computable calls
+ * (count/sum/min/max/first/last), NO interval and NO field condition is
+ * served from per-chunk pre-agg metadata when every row of the chunk lies
+ * inside the query time range (reader.go:1256 `cm.allRowsInRange(ctx.tr)`),
+ * and decodes only boundary chunks. The reference writes FloatPreAgg /
+ * IntegerPreAgg into ColumnMeta at flush time (pre_aggregation.go:410,:330);
+ * this engine computes the same per-series whole-shard aggregates once on
+ * device and caches them on the shard handle.
+ *
+ * gemx_preagg_build: compute + cache (idempotent; gemx_scan_preagg builds
+ * lazily if it was not called). One fused scan over the whole shard. */
+int gemx_preagg_build(gemx_shard *);
+
+/* Serve a calls-only / no-interval / no-predicate query:
+ * per-series rows (win_start = start_time), grouped by sid in descriptor
+ * order, series fully inside [start_time,end_time] copied from the cache
+ * (no GPU work), series partially covered re-scanned on device, disjoint
+ * series omitted. *n_meta_out (may be NULL) returns how many rows were
+ * served from metadata alone. Results are identical to
+ * gemx_scan_agg(start,end,interval=0). */
+int gemx_scan_preagg(gemx_shard *, int64_t start_time, int64_t end_time,
+                     gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                     uint64_t *n_meta_out, gemx_query_stats *stats);
+
 /* PromQL rate()/increase()/delta() over range vectors — the
  * RangeVectorCursor path (engine/prom_range_vector_cursor.go:49-153,
  * prom_functions.go:107-160): sample steps ts from start+range to
@@ -189,6 +214,21 @@ int gemx_prom_irate(gemx_shard *, int64_t start_time, int64_t end_time,
                     int64_t range_ns, int64_t step_ns, int is_rate,
                     gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,
                     gemx_query_stats *stats);
+
+/* xxx_over_time family (engine/prom_functions.go: sum_over_time:232,
+ * count_over_time:222, avg_over_time:341 Kahan streaming mean,
+ * min/max_over_time:300-339 NaN-aware, last_over_time:528): func selects
+ * the reducer. Same sampling grid and window semantics as gemx_prom_rate. */
+#define GEMX_PF_SUM_OT 2
+#define GEMX_PF_COUNT_OT 3
+#define GEMX_PF_AVG_OT 4
+#define GEMX_PF_MIN_OT 5
+#define GEMX_PF_MAX_OT 6
+#define GEMX_PF_LAST_OT 7
+int gemx_prom_over_time(gemx_shard *, int64_t start_time, int64_t end_time,
+                        int64_t range_ns, int64_t step_ns, int func,
+                        gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,
+                        gemx_query_stats *stats);
 
 #ifdef __cplusplus
 }

@@ -2414,6 +2414,15 @@ struct gemx_shard {
   hipStream_t stream;
   QueryPlan plan;
   RatePlan rate_plan;
+  /* pre-aggregation metadata (pre_aggregation.go FloatPreAgg role): one
+   * whole-range aggregate row per series, computed on device once and
+   * served for matchPreAgg-shaped queries whose range covers the series
+   * (reader.go:1256 allRowsInRange branch) */
+  QueryPlan sub_plan; /* plan cache for preagg boundary re-scans */
+  std::vector<gemx_agg_row> preagg;
+  std::vector<int64_t> ser_min_t, ser_max_t; /* per-series time bounds */
+  int64_t shard_min_t = 0, shard_max_t = 0;
+  bool preagg_valid = false;
 };
 
 static void free_plan(QueryPlan &p) {
@@ -2542,11 +2551,22 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     else
       s->general_ids.push_back((uint32_t)i);
     s->total_rows_scanned += d.rows;
-    /* series ranges */
+    /* series ranges + per-series/shard time bounds (for preagg coverage) */
     if (s->series_ranges.empty() || s->series_ranges.back().sid != d.sid) {
       s->series_ranges.push_back({d.sid, (uint32_t)i, 1});
+      s->ser_min_t.push_back(d.min_time);
+      s->ser_max_t.push_back(d.max_time);
     } else {
       s->series_ranges.back().count++;
+      s->ser_min_t.back() = std::min(s->ser_min_t.back(), d.min_time);
+      s->ser_max_t.back() = std::max(s->ser_max_t.back(), d.max_time);
+    }
+    if (i == 0) {
+      s->shard_min_t = d.min_time;
+      s->shard_max_t = d.max_time;
+    } else {
+      s->shard_min_t = std::min(s->shard_min_t, d.min_time);
+      s->shard_max_t = std::max(s->shard_max_t, d.max_time);
     }
   }
 
@@ -2579,6 +2599,7 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   if (!s) return GEMX_OK;
   hipSetDevice(s->device);
   free_plan(s->plan);
+  free_plan(s->sub_plan);
   free_rate_plan(s->rate_plan);
   hipFree(s->d_blob);
   hipFree(s->d_descs);
@@ -2589,17 +2610,23 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   return GEMX_OK;
 }
 
+/* skip_series (nullable, one char per series): series marked 1 are excluded
+ * from the scan — zero segments decoded, zero output rows. Used by the
+ * preagg path to re-scan only range-boundary series (the complement of
+ * reader.go:1256's allRowsInRange fast branch). The subset is a pure
+ * function of (start,end), so the dedicated sub_plan cache keyed on the
+ * range stays coherent. */
 static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                      int64_t interval, int64_t offset, int group_all,
                      int filter_op, double filter_f, int64_t filter_i,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
-                     gemx_query_stats *stats) {
+                     gemx_query_stats *stats, const char *skip_series = nullptr) {
   if (!s) return GEMX_E_INVALID;
   HIP_CHECK(hipSetDevice(s->device));
   const uint64_t nsegs = s->nsegs;
   const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960 + 512; /* +512: clip bitmap */
 
-  QueryPlan &P = s->plan;
+  QueryPlan &P = skip_series ? s->sub_plan : s->plan;
   if (!P.valid || P.start != start_time || P.end != end_time ||
       P.interval != interval || P.offset != offset) {
     free_plan(P);
@@ -2612,14 +2639,16 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     std::vector<char> is_gen(nsegs, 0);
     for (auto id : s->general_ids) is_gen[id] = 1;
     bool any_clip = false;
+    if (skip_series) any_clip = true; /* subset launches use the queues */
     for (size_t g = 0; g < s->series_ranges.size(); g++) {
       auto &r = s->series_ranges[g];
       int64_t wmin = INT64_MAX, wmax = INT64_MIN;
+      const bool skip_g = skip_series && skip_series[g];
       for (uint32_t i = r.start; i < r.start + r.count; i++) {
         const gemx_seg_desc &d = s->h_descs[i];
         P.segq[i].partial_base = P.partial_slots;
         P.segq[i].series_idx = (uint32_t)g;
-        if (d.max_time < start_time || d.min_time > end_time) {
+        if (skip_g || d.max_time < start_time || d.min_time > end_time) {
           /* fully outside the query range: skipped */
           P.segq[i].w_first = 0;
           P.segq[i].n_wins = 0;
@@ -3160,6 +3189,96 @@ extern "C" int gemx_scan_agg_ex(gemx_shard *s, int64_t start_time,
                                 uint64_t *n_out, gemx_query_stats *stats) {
   return scan_impl(s, start_time, end_time, interval, offset, group_all,
                    filter_op, filter_f, filter_i, out_host, cap, n_out, stats);
+}
+
+/* ---------------- pre-aggregation metadata ---------------- */
+/* The reference stores per-chunk FloatPreAgg/IntegerPreAgg in ColumnMeta at
+ * flush time (pre_aggregation.go:410,:330) and serves matchPreAgg queries
+ * (iterators_helper.go:90: calls only, no interval, no field condition)
+ * from it whenever the chunk lies fully inside the query range
+ * (reader.go:1256 allRowsInRange). Here the same metadata is one
+ * whole-range aggregate row per series, computed by the scan kernels once
+ * per attached shard and cached host-side; for interval==0 every row field
+ * except win_start is range-independent, so a covering query is a pure
+ * memcpy with win_start rewritten. */
+extern "C" int gemx_preagg_build(gemx_shard *s) {
+  if (!s) return GEMX_E_INVALID;
+  if (s->preagg_valid) return GEMX_OK;
+  const size_t nser = s->series_ranges.size();
+  if (nser == 0) {
+    s->preagg_valid = true;
+    return GEMX_OK;
+  }
+  std::vector<gemx_agg_row> rows(nser);
+  uint64_t n = 0;
+  int rc = scan_impl(s, s->shard_min_t, s->shard_max_t, 0, 0, 0, 0, 0, 0,
+                     rows.data(), nser, &n, nullptr);
+  if (rc != 0) return rc;
+  if (n != nser) { /* every series has ≥1 row, so this cannot drop rows */
+    seterr("preagg build produced an unexpected row count");
+    return GEMX_E_INVALID;
+  }
+  s->preagg = std::move(rows);
+  s->preagg_valid = true;
+  return GEMX_OK;
+}
+
+extern "C" int gemx_scan_preagg(gemx_shard *s, int64_t start_time,
+                                int64_t end_time, gemx_agg_row *out_host,
+                                uint64_t cap, uint64_t *n_out,
+                                uint64_t *n_meta_out, gemx_query_stats *stats) {
+  if (!s || !out_host || !n_out) return GEMX_E_INVALID;
+  int rc = gemx_preagg_build(s);
+  if (rc != 0) return rc;
+  const size_t nser = s->series_ranges.size();
+  /* classify each series against the query range (reader.go:1256) */
+  std::vector<char> skip(nser, 0); /* 1 = do not scan (covered or disjoint) */
+  std::vector<char> covered(nser, 0);
+  size_t n_cov = 0, n_partial = 0;
+  for (size_t g = 0; g < nser; g++) {
+    if (s->ser_max_t[g] < start_time || s->ser_min_t[g] > end_time) {
+      skip[g] = 1; /* disjoint: no output row */
+    } else if (s->ser_min_t[g] >= start_time && s->ser_max_t[g] <= end_time) {
+      skip[g] = 1;
+      covered[g] = 1;
+      n_cov++;
+    } else {
+      n_partial++;
+    }
+  }
+  uint64_t n_scan = 0;
+  std::vector<gemx_agg_row> scanned;
+  if (n_partial) {
+    scanned.resize(n_partial);
+    rc = scan_impl(s, start_time, end_time, 0, 0, 0, 0, 0, 0, scanned.data(),
+                   n_partial, &n_scan, stats, skip.data());
+    if (rc != 0) return rc;
+  } else if (stats) {
+    memset(stats, 0, sizeof(*stats));
+    stats->points = s->total_rows_scanned;
+  }
+  if (n_cov + n_scan > cap) {
+    seterr("output capacity too small");
+    return GEMX_E_CAP;
+  }
+  /* emit in series (descriptor) order: cached rows for covered series,
+   * scanned rows for boundary series (scan output is already in series
+   * order; a boundary series with no row actually in range emits nothing) */
+  uint64_t n = 0, sp = 0;
+  for (size_t g = 0; g < nser; g++) {
+    if (covered[g]) {
+      out_host[n] = s->preagg[g];
+      out_host[n].win_start = start_time; /* k_merge: q_start when interval==0 */
+      n++;
+    } else if (!skip[g] && sp < n_scan &&
+               scanned[sp].sid == s->series_ranges[g].sid) {
+      out_host[n++] = scanned[sp++];
+    }
+  }
+  *n_out = n;
+  if (n_meta_out) *n_meta_out = n_cov;
+  if (stats) stats->n_rows = n;
+  return GEMX_OK;
 }
 
 extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,

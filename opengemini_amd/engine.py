@@ -139,6 +139,14 @@ def _load():
         C.c_int, C.c_double, C.c_int64,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_preagg_build.restype = C.c_int
+    lib.gemx_preagg_build.argtypes = [C.c_void_p]
+    lib.gemx_scan_preagg.restype = C.c_int
+    lib.gemx_scan_preagg.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
+        C.POINTER(_Stats),
+    ]
     lib.gemx_prom_rate.restype = C.c_int
     lib.gemx_prom_rate.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int, C.c_int,
@@ -297,6 +305,44 @@ class Shard:
         )
         return out[: n.value], stats
 
+
+    def preagg_build(self):
+        """Compute + cache per-series whole-shard pre-agg rows on the handle
+        (the ColumnMeta FloatPreAgg/IntegerPreAgg role,
+        engine/immutable/pre_aggregation.go:410,:330). Idempotent; done
+        lazily by scan_preagg otherwise."""
+        _check(self._lib.gemx_preagg_build(self._h), self._lib)
+
+    def scan_preagg(self, start_time, end_time, out_cap=None):
+        """matchPreAgg-shaped query (iterators_helper.go:90: calls only, no
+        interval, no field condition): per-series whole-range aggregates.
+        Series fully inside [start,end] are served from the cached pre-agg
+        metadata without touching the GPU (reader.go:1256 allRowsInRange
+        branch); boundary series re-scan on device. Results are identical
+        to scan_agg(start, end, interval=0).
+
+        Returns (rows, stats); stats['meta_rows'] = rows served from
+        metadata alone."""
+        lib = self._lib
+        if out_cap is None:
+            out_cap = self._rows_bound(0, 0, False)
+        out = self._pooled_out("agg", out_cap, AGG_ROW_DTYPE)
+        n = C.c_uint64(0)
+        nm = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_scan_preagg(
+            self._h, start_time, end_time,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(nm),
+            C.byref(st),
+        )
+        _check(rc, lib)
+        stats = dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            host_ms=st.h2d_ms, points=st.points,
+            compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
+            meta_rows=int(nm.value),
+        )
+        return out[: n.value], stats
 
     def prom_rate(self, start_time, end_time, range_ns, step_ns, is_rate=True,
                   is_counter=True, out_cap=None):

@@ -547,3 +547,90 @@ class TestTimeRangeClipping:
         assert np.array_equal(gpu["count"], ref["count"])
         assert np.array_equal(gpu["min"], ref["min"])
         assert int(gpu["count"].sum()) == 200 * 800  # rows 100..899 per series
+
+
+class TestPreAggParity:
+    """gemx_scan_preagg — the matchPreAgg path (iterators_helper.go:90):
+    series fully inside the range served from cached per-series pre-agg
+    metadata (reader.go:1256 allRowsInRange), boundary series re-scanned.
+    Must be identical to a plain interval=0 scan of the same range."""
+
+    def _compare(self, sh, blob, descs, col_type, start, end):
+        pre, pstats = sh.scan_preagg(start, end)
+        pre = pre.copy()  # pooled buffer: copy before the next query
+        ref, _ = sh.scan_agg(start, end, 0)
+        assert_parity(pre, ref.copy(), col_type)
+        orc_rows = orc.scan_agg(blob, descs, col_type, start, end, 0)
+        assert_parity(pre, orc_rows, col_type)
+        return pre, pstats
+
+    def test_mixed_coverage_float(self):
+        rng = np.random.default_rng(1101)
+        blob, descs, _ = build_shard(rng, F, range(1, 121))
+        sh = gpu_shard(blob, descs, F)
+        try:
+            S = 10**9
+            # series start in [0,100)s and span ~100-1500s: this range makes
+            # a mix of covered, boundary and disjoint series
+            pre, st = self._compare(sh, blob, descs, F, 120 * S, 700 * S)
+            assert 0 < st["meta_rows"] < len(descs)
+            # and a second, different range reuses the cache correctly
+            self._compare(sh, blob, descs, F, 0, 10**15)
+        finally:
+            sh.close()
+
+    def test_mixed_coverage_int(self):
+        rng = np.random.default_rng(1102)
+        blob, descs, _ = build_shard(rng, I, range(1, 61))
+        sh = gpu_shard(blob, descs, I)
+        try:
+            S = 10**9
+            self._compare(sh, blob, descs, I, 90 * S, 400 * S)
+        finally:
+            sh.close()
+
+    def test_full_coverage_is_pure_metadata(self):
+        blob, descs = orc.gen_shard(1103, 300, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            sh.preagg_build()
+            pre, st = sh.scan_preagg(-10**15, 10**15)
+            pre = pre.copy()
+            # every series served from metadata, no kernel launched
+            assert st["meta_rows"] == 300
+            assert st["decode_ms"] == 0.0
+            assert len(pre) == 300
+            ref = orc.scan_agg(blob, descs, F, -10**15, 10**15, 0)
+            assert_parity(pre, ref, F)
+        finally:
+            sh.close()
+
+    def test_all_disjoint(self):
+        blob, descs = orc.gen_shard(1104, 20, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            pre, st = sh.scan_preagg(10**15, 2 * 10**15)
+            assert len(pre) == 0 and st["meta_rows"] == 0
+        finally:
+            sh.close()
+
+    def test_exact_boundaries_cover(self):
+        # range == [shard min_time, shard max_time] exactly: all covered
+        blob, descs = orc.gen_shard(1105, 50, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            lo = int(descs["min_time"].min())
+            hi = int(descs["max_time"].max())
+            pre, st = sh.scan_preagg(lo, hi)
+            pre = pre.copy()
+            assert st["meta_rows"] == 50
+            ref = orc.scan_agg(blob, descs, F, lo, hi, 0)
+            assert_parity(pre, ref, F)
+            # one nanosecond narrower: every series becomes a boundary scan
+            pre2, st2 = sh.scan_preagg(lo + 1, hi - 1)
+            pre2 = pre2.copy()
+            assert st2["meta_rows"] == 0
+            ref2 = orc.scan_agg(blob, descs, F, lo + 1, hi - 1, 0)
+            assert_parity(pre2, ref2, F)
+        finally:
+            sh.close()
