@@ -2644,14 +2644,28 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       auto &r = s->series_ranges[g];
       int64_t wmin = INT64_MAX, wmax = INT64_MIN;
       const bool skip_g = skip_series && skip_series[g];
+      /* running max of (w_first + n_wins): merge_series_window's binary
+       * search requires it non-decreasing across the series' segments, so
+       * excluded segments (n_wins == 0) get a sentinel w_first clamped to
+       * keep the sequence monotone */
+      int64_t run = INT64_MIN;
       for (uint32_t i = r.start; i < r.start + r.count; i++) {
         const gemx_seg_desc &d = s->h_descs[i];
         P.segq[i].partial_base = P.partial_slots;
         P.segq[i].series_idx = (uint32_t)g;
         if (skip_g || d.max_time < start_time || d.min_time > end_time) {
-          /* fully outside the query range: skipped */
-          P.segq[i].w_first = 0;
+          /* fully outside the query range (or series excluded): skipped */
+          int64_t cand;
+          if (interval) {
+            int64_t ct = std::min(std::max(d.min_time, start_time), end_time);
+            cand = win_ordinal(ct, interval, offset) +
+                   (d.min_time > end_time ? 1 : 0);
+          } else {
+            cand = (d.min_time > end_time) ? 1 : 0;
+          }
+          P.segq[i].w_first = std::max(cand, run);
           P.segq[i].n_wins = 0;
+          run = P.segq[i].w_first;
           any_clip = true;
           continue;
         }
@@ -2662,6 +2676,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
         int64_t w1 = interval ? win_ordinal(xt, interval, offset) : 0;
         P.segq[i].w_first = w0;
         P.segq[i].n_wins = (uint32_t)(w1 - w0 + 1);
+        run = std::max(run, w1 + 1);
         P.partial_slots += P.segq[i].n_wins;
         wmin = std::min(wmin, w0);
         wmax = std::max(wmax, w1);
@@ -3002,6 +3017,10 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     for (size_t g = 0; g < s->series_ranges.size(); g++) {
       auto &r = s->series_ranges[g];
       int64_t smin = INT64_MAX, smax = INT64_MIN;
+      /* running max of (s0 + n_steps): k_rate_merge's binary search needs
+       * it non-decreasing, so grid-excluded segments get a clamped
+       * sentinel s0 (same invariant as the agg plan's w_first) */
+      int64_t run = INT64_MIN;
       for (uint32_t i = r.start; i < r.start + r.count; i++) {
         const gemx_seg_desc &d = s->h_descs[i];
         /* steps whose window [ts-range, ts] can contain a point of this
@@ -3012,10 +3031,12 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
         int64_t o1 = (hi < 0) ? -1 : hi / eff_step;
         if (o1 > last_ord) o1 = last_ord;
         if (o0 > last_ord || o1 < o0) {
-          P.rsegq[i].s0 = 0;
+          int64_t cand = std::min(o0, last_ord + 1);
+          P.rsegq[i].s0 = std::max(cand, run);
           P.rsegq[i].n_steps = 0;
           P.rsegq[i].partial_base = P.partial_slots;
           P.rsegq[i].series_idx = (uint32_t)g;
+          run = P.rsegq[i].s0;
           continue;
         }
         P.rsegq[i].s0 = o0;
@@ -3023,6 +3044,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
         P.rsegq[i].partial_base = P.partial_slots;
         P.rsegq[i].series_idx = (uint32_t)g;
         P.partial_slots += P.rsegq[i].n_steps;
+        run = std::max(run, o1 + 1);
         smin = std::min(smin, o0);
         smax = std::max(smax, o1);
       }

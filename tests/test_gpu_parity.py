@@ -570,10 +570,12 @@ class TestPreAggParity:
         sh = gpu_shard(blob, descs, F)
         try:
             S = 10**9
-            # series start in [0,100)s and span ~100-1500s: this range makes
-            # a mix of covered, boundary and disjoint series
-            pre, st = self._compare(sh, blob, descs, F, 120 * S, 700 * S)
-            assert 0 < st["meta_rows"] < len(descs)
+            # series start in [0,100)s and span ~100-1500s: a range open on
+            # the left and cut at 450s leaves short series fully covered and
+            # long series as boundary re-scans
+            pre, st = self._compare(sh, blob, descs, F, -10**15, 450 * S)
+            n_series = len(np.unique(descs["sid"]))
+            assert 0 < st["meta_rows"] < n_series
             # and a second, different range reuses the cache correctly
             self._compare(sh, blob, descs, F, 0, 10**15)
         finally:
@@ -585,7 +587,8 @@ class TestPreAggParity:
         sh = gpu_shard(blob, descs, I)
         try:
             S = 10**9
-            self._compare(sh, blob, descs, I, 90 * S, 400 * S)
+            pre, st = self._compare(sh, blob, descs, I, -10**15, 300 * S)
+            assert st["meta_rows"] > 0
         finally:
             sh.close()
 
@@ -632,5 +635,44 @@ class TestPreAggParity:
             assert st2["meta_rows"] == 0
             ref2 = orc.scan_agg(blob, descs, F, lo + 1, hi - 1, 0)
             assert_parity(pre2, ref2, F)
+        finally:
+            sh.close()
+
+
+class TestVariedSpanClipping:
+    """Series with different time spans + a clipping range: segments fully
+    outside the range interleave with in-range ones inside a series, which
+    must not break the merge search (regression: sentinel window ordinals)."""
+
+    def test_interval_clip_varied_spans(self):
+        rng = np.random.default_rng(1201)
+        blob, descs, _ = build_shard(rng, F, range(1, 101))
+        sh = gpu_shard(blob, descs, F)
+        try:
+            S = 10**9
+            for (a, b) in [(-10**15, 450 * S), (120 * S, 700 * S),
+                           (0, 10**15)]:
+                gpu, _ = sh.scan_agg(a, b, INT)
+                ref = orc.scan_agg(blob, descs, F, a, b, INT)
+                assert_parity(gpu.copy(), ref, F)
+        finally:
+            sh.close()
+
+    def test_rate_varied_spans(self):
+        rng = np.random.default_rng(1202)
+        blob, descs, _ = build_shard(rng, F, range(1, 61), null_frac=0.0)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            S = 10**9
+            start, end = 50 * S, 600 * S
+            rng_ns, step = 300 * S, 60 * S
+            gpu, _ = sh.prom_rate(start, end, rng_ns, step)
+            ref = orc.prom_rate(blob, descs, start, end, rng_ns, step,
+                                is_rate=True, is_counter=True)
+            assert len(gpu) == len(ref), (len(gpu), len(ref))
+            assert np.array_equal(gpu["sid"], ref["sid"])
+            assert np.array_equal(gpu["ts"], ref["ts"])
+            ok = np.isclose(gpu["value"], ref["value"], rtol=1e-12, atol=0)
+            assert np.all(ok)
         finally:
             sh.close()
