@@ -230,6 +230,60 @@ int gemx_prom_over_time(gemx_shard *, int64_t start_time, int64_t end_time,
                         gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,
                         gemx_query_stats *stats);
 
+/* ---------------- write side (downsample output) ----------------
+ * The downsample service rewrites TSSP files with aggregated columns
+ * (engine/engine_downsample.go via WriteIntoStorageTransform,
+ * engine/executor/record_plan.go:494; segment layout from
+ * engine/immutable/column_builder.go:428-501 and chunkdata_builder.go:91-95;
+ * block codecs lib/encoding/{int,timestamp}.go + lib/compress/float.go).
+ * gemx_encode_shard is that writer's column encoder for this path: it emits
+ * segments this engine AND the reference's readers decode. Codec selection
+ * follows the reference except that branches this engine cannot decode
+ * on-device are replaced by always-valid uncompressed forms: int zstd →
+ * uncompressed (int.go:168), time snappy → uncompressed (timestamp.go:85),
+ * float snappy/RLE → compressedNull raw (float.go case 0). */
+
+/* Rows must be grouped by sid with times ascending within each sid (the
+ * order ChunkMeta/WriteRecord requires). values holds one element per row
+ * (float64 for GEMX_TYPE_FLOAT, int64 for GEMX_TYPE_INT); rows whose
+ * valid[i]==0 are nil (valid==NULL ⇒ all rows valid). Segments are cut at
+ * sid changes and every seg_rows rows (≤1000 in the reference,
+ * config.maxRowsPerSegment). Writes the blob and one gemx_seg_desc per
+ * segment; *blob_bytes_out / *n_segs_out return the used sizes. */
+int gemx_encode_shard(int col_type, const uint64_t *sids, const int64_t *times,
+                      const void *values, const uint8_t *valid,
+                      uint64_t n_rows, uint32_t seg_rows, uint8_t *blob_out,
+                      uint64_t blob_cap, gemx_seg_desc *descs_out,
+                      uint64_t descs_cap, uint64_t *n_segs_out,
+                      uint64_t *blob_bytes_out);
+
+/* Conservative capacity bound for gemx_encode_shard outputs. */
+int gemx_encode_bound(int col_type, uint64_t n_rows, uint32_t seg_rows,
+                      uint64_t *blob_bound, uint64_t *descs_bound);
+
+/* aggregate column selector for gemx_downsample_write */
+#define GEMX_OP_COUNT 0
+#define GEMX_OP_SUM 1
+#define GEMX_OP_MIN 2
+#define GEMX_OP_MAX 3
+#define GEMX_OP_FIRST 4
+#define GEMX_OP_LAST 5
+
+/* Downsample end-to-end: scan + GROUP BY time aggregate on device (the
+ * per-series path), then encode the chosen aggregate column as a new
+ * shard: one row per (sid, window) with data, time = the window's first
+ * row time (the multiCall time-column semantics of the aggregate cursor,
+ * see oracle/oracle.h), nil aggregates written as nil rows.
+ * GEMX_OP_COUNT always writes a GEMX_TYPE_INT column; other ops keep the
+ * source column type. The output attaches with gemx_shard_attach and is
+ * readable by the reference's segment readers. */
+int gemx_downsample_write(gemx_shard *, int64_t start_time, int64_t end_time,
+                          int64_t interval, int64_t offset, int op,
+                          uint32_t seg_rows, uint8_t *blob_out,
+                          uint64_t blob_cap, gemx_seg_desc *descs_out,
+                          uint64_t descs_cap, uint64_t *n_segs_out,
+                          uint64_t *blob_bytes_out);
+
 #ifdef __cplusplus
 }
 #endif

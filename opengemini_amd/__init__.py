@@ -16,4 +16,5 @@ from .engine import (  # noqa: F401
     device_count,
     abi_version,
     build_extension,
+    encode_shard,
 )

@@ -3337,3 +3337,6 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
   return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, 0, 0, func,
                         out_host, cap, n_out, stats);
 }
+
+/* ---------------- write side (downsample output) ---------------- */
+#include "gemx_writer.hpp"

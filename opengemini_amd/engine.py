@@ -147,6 +147,23 @@ def _load():
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
         C.POINTER(_Stats),
     ]
+    lib.gemx_encode_shard.restype = C.c_int
+    lib.gemx_encode_shard.argtypes = [
+        C.c_int, C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p, C.c_uint64,
+        C.c_uint32, C.c_void_p, C.c_uint64, C.c_void_p, C.c_uint64,
+        C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
+    ]
+    lib.gemx_encode_bound.restype = C.c_int
+    lib.gemx_encode_bound.argtypes = [
+        C.c_int, C.c_uint64, C.c_uint32,
+        C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
+    ]
+    lib.gemx_downsample_write.restype = C.c_int
+    lib.gemx_downsample_write.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_uint32, C.c_void_p, C.c_uint64, C.c_void_p, C.c_uint64,
+        C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
+    ]
     lib.gemx_prom_rate.restype = C.c_int
     lib.gemx_prom_rate.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int, C.c_int,
@@ -164,6 +181,47 @@ def _load():
     ]
     _lib = lib
     return lib
+
+
+DOWNSAMPLE_OPS = {"count": 0, "sum": 1, "min": 2, "max": 3, "first": 4,
+                  "last": 5}
+
+
+def encode_shard(col_type, sids, times, values, valid=None, seg_rows=1000):
+    """TSSP segment writer (host-side; no GPU needed). Rows must be grouped
+    by sid, times ascending within sid. Returns (blob: bytes, descs:
+    np.ndarray[SEG_DESC_DTYPE]) attachable by Shard and readable by the
+    reference's segment readers. See include/gemx.h for the codec
+    selection (column_builder.go / lib/encoding / lib/compress)."""
+    lib = _load()
+    sids = np.ascontiguousarray(sids, dtype=np.uint64)
+    times = np.ascontiguousarray(times, dtype=np.int64)
+    n = len(sids)
+    if col_type == GEMX_TYPE_FLOAT:
+        values = np.ascontiguousarray(values, dtype=np.float64)
+    else:
+        values = np.ascontiguousarray(values, dtype=np.int64)
+    vptr = None
+    if valid is not None:
+        valid = np.ascontiguousarray(valid, dtype=np.uint8)
+        vptr = valid.ctypes.data_as(C.c_void_p)
+    bb = C.c_uint64(0)
+    db = C.c_uint64(0)
+    _check(lib.gemx_encode_bound(col_type, n, seg_rows, C.byref(bb),
+                                 C.byref(db)), lib)
+    blob = np.zeros(bb.value, dtype=np.uint8)
+    descs = np.zeros(db.value, dtype=SEG_DESC_DTYPE)
+    nseg = C.c_uint64(0)
+    used = C.c_uint64(0)
+    rc = lib.gemx_encode_shard(
+        col_type, sids.ctypes.data_as(C.c_void_p),
+        times.ctypes.data_as(C.c_void_p), values.ctypes.data_as(C.c_void_p),
+        vptr, n, seg_rows, blob.ctypes.data_as(C.c_void_p), bb.value,
+        descs.ctypes.data_as(C.c_void_p), db.value, C.byref(nseg),
+        C.byref(used),
+    )
+    _check(rc, lib)
+    return blob[: used.value].tobytes(), descs[: nseg.value].copy()
 
 
 def abi_version():
@@ -343,6 +401,34 @@ class Shard:
             meta_rows=int(nm.value),
         )
         return out[: n.value], stats
+
+    def downsample_write(self, start_time, end_time, interval, offset=0,
+                         op="sum", seg_rows=1000):
+        """Downsample end-to-end (config #4 write side): device scan +
+        GROUP BY time aggregate, then re-encode the chosen aggregate
+        column as a new TSSP shard (WriteIntoStorageTransform role,
+        executor/record_plan.go:494). Returns (blob, descs) — attachable
+        and re-queryable. op: count/sum/min/max/first/last; count yields
+        an int64 column."""
+        lib = self._lib
+        opc = DOWNSAMPLE_OPS[op]
+        n_rows = self._rows_bound(interval, offset, False)
+        bb = C.c_uint64(0)
+        db = C.c_uint64(0)
+        _check(lib.gemx_encode_bound(self.col_type, n_rows, seg_rows,
+                                     C.byref(bb), C.byref(db)), lib)
+        blob = np.zeros(bb.value, dtype=np.uint8)
+        descs = np.zeros(db.value, dtype=SEG_DESC_DTYPE)
+        nseg = C.c_uint64(0)
+        used = C.c_uint64(0)
+        rc = lib.gemx_downsample_write(
+            self._h, start_time, end_time, interval, offset, opc, seg_rows,
+            blob.ctypes.data_as(C.c_void_p), bb.value,
+            descs.ctypes.data_as(C.c_void_p), db.value, C.byref(nseg),
+            C.byref(used),
+        )
+        _check(rc, lib)
+        return blob[: used.value].tobytes(), descs[: nseg.value].copy()
 
     def prom_rate(self, start_time, end_time, range_ns, step_ns, is_rate=True,
                   is_counter=True, out_cap=None):
