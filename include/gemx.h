@@ -170,6 +170,21 @@ int gemx_scan_agg_ex(gemx_shard *, int64_t start_time, int64_t end_time,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                      gemx_query_stats *stats);
 
+/* hash GROUP BY tag (engine/executor/hash_agg_transform.go): the executor
+ * hashes each series' tag set into a group; series_group passes that
+ * sid→group mapping (one uint32 per series, descriptor order, values
+ * < n_groups) and the engine aggregates per (group, window) on device —
+ * only groups × windows rows cross the boundary. Output rows carry the
+ * group id in the sid field, grouped by group, windows ascending; within
+ * a group, series merge in descriptor order, so value/time ties resolve
+ * to the first-processed series exactly as AggTagSetCursor.UpdateRec
+ * does (engine/agg_tagset_cursor.go:1111). */
+int gemx_scan_agg_tags(gemx_shard *, const uint32_t *series_group,
+                       uint32_t n_groups, int64_t start_time,
+                       int64_t end_time, int64_t interval, int64_t offset,
+                       gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                       gemx_query_stats *stats);
+
 /* Pre-aggregation metadata (the matchPreAgg path,
  * engine/iterators_helper.go:90): a query with only pre-computable calls
  * (count/sum/min/max/first/last), NO interval and NO field condition is

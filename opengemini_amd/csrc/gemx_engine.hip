@@ -1767,6 +1767,106 @@ __global__ void __launch_bounds__(256) k_group_p2(
 }
 
 
+/* ---------------- hash GROUP BY tag (executor/hash_agg_transform.go) ----
+ * The executor's hash-agg step maps each series to a group via the tag-set
+ * hash dictionary; here the caller passes that sid→group mapping (one
+ * uint32 per series, in descriptor order) and the engine produces one row
+ * per (group, window) on device — only groups × windows cross PCIe, not
+ * series × windows. Within a group, series are processed in descriptor
+ * (sid) order, the same order AggTagSetCursor sees them, so ties resolve
+ * to the first-processed series exactly as UpdateRec does
+ * (engine/agg_tagset_cursor.go:1111, lib/record/reccord_functions.go). */
+struct TagChunk {
+  uint32_t ser_off;    /* offset into the order[] permutation */
+  uint32_t count;      /* series in this chunk */
+  uint32_t off_in_grp; /* position of ser_off within its group */
+  uint32_t _pad;
+};
+
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_tag_p1(
+    const SeriesQ *__restrict__ series, const uint32_t *__restrict__ order,
+    const TagChunk *__restrict__ chunks, uint32_t n_chunks,
+    const SegQ *__restrict__ segq, const Partial *__restrict__ partials,
+    GAcc *__restrict__ gtmp, int64_t W0, uint32_t n_gwins) {
+  __shared__ GAcc sh[256];
+  for (uint32_t bb = blockIdx.x; bb < n_gwins * n_chunks; bb += gridDim.x) {
+    uint32_t wb = bb / n_chunks, c = bb % n_chunks;
+    int64_t w = W0 + (int64_t)wb;
+    const TagChunk ch = chunks[c];
+    GAcc a;
+    memset(&a, 0, sizeof(a));
+    for (uint32_t i = threadIdx.x; i < ch.count; i += blockDim.x) {
+      const uint32_t g = order[ch.ser_off + i];
+      const SeriesQ s = series[g];
+      int64_t local = w - s.w_min;
+      if (local < 0 || local >= (int64_t)s.n_wins) continue;
+      gemx_agg_row r;
+      memset(&r, 0, sizeof(r));
+      if (!merge_series_window<COLTYPE>(s, segq, partials, w, &r)) continue;
+      gacc_row<COLTYPE>(&a, &r, ch.off_in_grp + i);
+    }
+    sh[threadIdx.x] = a;
+    __syncthreads();
+    for (int s2 = 128; s2 > 0; s2 >>= 1) {
+      if (threadIdx.x < (uint32_t)s2)
+        gacc_merge<COLTYPE>(&sh[threadIdx.x], &sh[threadIdx.x + s2]);
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) gtmp[bb] = sh[0];
+    __syncthreads();
+  }
+}
+
+/* one lane per (group, window) output row; chunks of a group merge in
+ * ascending src order (left precedence, as the tagset cursor processes) */
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_tag_p2(
+    const GAcc *__restrict__ gtmp, const uint32_t *__restrict__ cstart,
+    const uint32_t *__restrict__ ccount, uint32_t n_chunks,
+    gemx_agg_row *__restrict__ out, uint32_t n_groups, int64_t W0,
+    uint32_t n_gwins, int64_t interval, int64_t offset, int64_t q_start) {
+  uint64_t total = (uint64_t)n_groups * n_gwins;
+  for (uint64_t r = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; r < total;
+       r += gridDim.x * (uint64_t)blockDim.x) {
+    uint32_t grp = (uint32_t)(r / n_gwins);
+    uint32_t wb = (uint32_t)(r % n_gwins);
+    int64_t w = W0 + (int64_t)wb;
+    GAcc a;
+    memset(&a, 0, sizeof(a));
+    for (uint32_t j = 0; j < ccount[grp]; j++)
+      gacc_merge<COLTYPE>(&a, &gtmp[(uint64_t)wb * n_chunks + cstart[grp] + j]);
+    gemx_agg_row o;
+    memset(&o, 0, sizeof(o));
+    o.sid = grp; /* group id in the sid slot */
+    int64_t ws = interval ? win_start_of(w, interval, offset) : q_start;
+    o.win_start = ws;
+    if (!a.used) {
+      o.count = -1; /* gap: host compacts */
+    } else {
+      o.first_row_time = ws;
+      o.count = a.count;
+      o.count_time = ws;
+      o.sum = a.sum;
+      o.sum_time = ws;
+      o.sum_isnil = !(a.active & 2);
+      o.minv = a.minv;
+      o.min_time = a.min_t;
+      o.min_isnil = !(a.active & 4);
+      o.maxv = a.maxv;
+      o.max_time = a.max_t;
+      o.max_isnil = !(a.active & 8);
+      o.firstv = a.firstv;
+      o.first_time = a.first_t;
+      o.first_isnil = !(a.active & 16);
+      o.lastv = a.lastv;
+      o.last_time = a.last_t;
+      o.last_isnil = !(a.active & 32);
+    }
+    out[r] = o;
+  }
+}
+
 /* ================= PromQL rate over range vectors (config #5) =============
  * Restates RangeVectorCursor + rate_prom (see oracle/agg.c orc_prom_rate for
  * the line-cited CPU spec): sample steps ts = startSample + k*step, window
@@ -2366,6 +2466,32 @@ struct QueryPlan {
   uint64_t n_gwins = 0;
 };
 
+/* cached GROUP BY tag device state (sid→group permutation + chunk table;
+ * group structure depends only on the mapping, gtmp/rows also on the
+ * query's window count) */
+struct TagPlan {
+  bool valid = false;
+  std::vector<uint32_t> group_ids; /* cached copy for key comparison */
+  uint32_t n_groups = 0;
+  uint32_t n_chunks = 0;
+  uint64_t n_gwins = 0; /* sizing key from the owning QueryPlan */
+  uint32_t *d_order = nullptr;
+  TagChunk *d_chunks = nullptr;
+  uint32_t *d_cstart = nullptr, *d_ccount = nullptr;
+  GAcc *d_gtmp = nullptr;
+  gemx_agg_row *d_rows = nullptr;
+};
+
+static void free_tag_plan(TagPlan &p) {
+  if (p.d_order) hipFree(p.d_order);
+  if (p.d_chunks) hipFree(p.d_chunks);
+  if (p.d_cstart) hipFree(p.d_cstart);
+  if (p.d_ccount) hipFree(p.d_ccount);
+  if (p.d_gtmp) hipFree(p.d_gtmp);
+  if (p.d_rows) hipFree(p.d_rows);
+  p = TagPlan();
+}
+
 /* cached rate-query device state */
 struct RatePlan {
   bool valid = false;
@@ -2419,6 +2545,7 @@ struct gemx_shard {
    * served for matchPreAgg-shaped queries whose range covers the series
    * (reader.go:1256 allRowsInRange branch) */
   QueryPlan sub_plan; /* plan cache for preagg boundary re-scans */
+  TagPlan tag_plan;   /* GROUP BY tag permutation + buffers */
   std::vector<gemx_agg_row> preagg;
   std::vector<int64_t> ser_min_t, ser_max_t; /* per-series time bounds */
   int64_t shard_min_t = 0, shard_max_t = 0;
@@ -2600,6 +2727,7 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   hipSetDevice(s->device);
   free_plan(s->plan);
   free_plan(s->sub_plan);
+  free_tag_plan(s->tag_plan);
   free_rate_plan(s->rate_plan);
   hipFree(s->d_blob);
   hipFree(s->d_descs);
@@ -2616,11 +2744,17 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
  * reader.go:1256's allRowsInRange fast branch). The subset is a pure
  * function of (start,end), so the dedicated sub_plan cache keyed on the
  * range stays coherent. */
+struct TagQuery {
+  const uint32_t *groups; /* per series, descriptor order */
+  uint32_t n_groups;
+};
+
 static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                      int64_t interval, int64_t offset, int group_all,
                      int filter_op, double filter_f, int64_t filter_i,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
-                     gemx_query_stats *stats, const char *skip_series = nullptr) {
+                     gemx_query_stats *stats, const char *skip_series = nullptr,
+                     const TagQuery *tagq = nullptr) {
   if (!s) return GEMX_E_INVALID;
   HIP_CHECK(hipSetDevice(s->device));
   const uint64_t nsegs = s->nsegs;
@@ -2855,7 +2989,97 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          d_scratch, scratch_per_lane, gen_lanes, d_err);
   }
   HIP_CHECK(hipEventRecord(ev1, s->stream));
-  if (!group_all && total_rows > 0) {
+  /* GROUP BY tag stage: ensure the permutation/chunk tables and buffers */
+  TagPlan &T = s->tag_plan;
+  if (tagq) {
+    const uint32_t nser = (uint32_t)s->series_ranges.size();
+    bool same_map = T.valid && T.n_groups == tagq->n_groups &&
+                    T.group_ids.size() == nser &&
+                    memcmp(T.group_ids.data(), tagq->groups,
+                           sizeof(uint32_t) * nser) == 0;
+    if (!same_map || T.n_gwins != P.n_gwins) {
+      free_tag_plan(T);
+      T.group_ids.assign(tagq->groups, tagq->groups + nser);
+      T.n_groups = tagq->n_groups;
+      T.n_gwins = P.n_gwins;
+      /* stable sort series by group: counting sort keeps sid order */
+      std::vector<uint32_t> gcnt(T.n_groups + 1, 0);
+      for (uint32_t i = 0; i < nser; i++) gcnt[T.group_ids[i] + 1]++;
+      for (uint32_t g2 = 1; g2 <= T.n_groups; g2++) gcnt[g2] += gcnt[g2 - 1];
+      std::vector<uint32_t> order(nser ? nser : 1);
+      std::vector<uint32_t> pos(gcnt.begin(), gcnt.end() - 1);
+      for (uint32_t i = 0; i < nser; i++) order[pos[T.group_ids[i]]++] = i;
+      /* chunk each group (<=256 series per chunk) */
+      std::vector<TagChunk> chunks;
+      std::vector<uint32_t> cstart(T.n_groups), ccount(T.n_groups);
+      const uint32_t PER = 256;
+      for (uint32_t g2 = 0; g2 < T.n_groups; g2++) {
+        uint32_t a = gcnt[g2], b = gcnt[g2 + 1];
+        cstart[g2] = (uint32_t)chunks.size();
+        for (uint32_t o = a; o < b; o += PER)
+          chunks.push_back({o, std::min(PER, b - o), o - a, 0});
+        ccount[g2] = (uint32_t)chunks.size() - cstart[g2];
+      }
+      T.n_chunks = (uint32_t)chunks.size();
+      if (T.n_chunks == 0) { /* no groups with series: still allocate */
+        chunks.push_back({0, 0, 0, 0});
+      }
+      HIP_CHECK(hipMalloc(&T.d_order, sizeof(uint32_t) * order.size()));
+      HIP_CHECK(hipMemcpyAsync(T.d_order, order.data(),
+                               sizeof(uint32_t) * order.size(),
+                               hipMemcpyHostToDevice, s->stream));
+      HIP_CHECK(hipMalloc(&T.d_chunks, sizeof(TagChunk) * chunks.size()));
+      HIP_CHECK(hipMemcpyAsync(T.d_chunks, chunks.data(),
+                               sizeof(TagChunk) * chunks.size(),
+                               hipMemcpyHostToDevice, s->stream));
+      HIP_CHECK(hipMalloc(&T.d_cstart,
+                          sizeof(uint32_t) * (T.n_groups ? T.n_groups : 1)));
+      HIP_CHECK(hipMalloc(&T.d_ccount,
+                          sizeof(uint32_t) * (T.n_groups ? T.n_groups : 1)));
+      if (T.n_groups) {
+        HIP_CHECK(hipMemcpyAsync(T.d_cstart, cstart.data(),
+                                 sizeof(uint32_t) * T.n_groups,
+                                 hipMemcpyHostToDevice, s->stream));
+        HIP_CHECK(hipMemcpyAsync(T.d_ccount, ccount.data(),
+                                 sizeof(uint32_t) * T.n_groups,
+                                 hipMemcpyHostToDevice, s->stream));
+      }
+      uint64_t gelems = (uint64_t)(P.n_gwins ? P.n_gwins : 1) *
+                        (T.n_chunks ? T.n_chunks : 1);
+      HIP_CHECK(hipMalloc(&T.d_gtmp, sizeof(GAcc) * gelems));
+      uint64_t relems = (uint64_t)(T.n_groups ? T.n_groups : 1) *
+                        (P.n_gwins ? P.n_gwins : 1);
+      HIP_CHECK(hipMalloc(&T.d_rows, sizeof(gemx_agg_row) * relems));
+      T.valid = true;
+    }
+    if (P.n_gwins > 0 && T.n_chunks > 0) {
+      uint32_t b1 = (uint32_t)std::min<uint64_t>(
+          (uint64_t)P.n_gwins * T.n_chunks, 65535);
+      uint64_t total = (uint64_t)T.n_groups * P.n_gwins;
+      uint32_t b2 =
+          (uint32_t)std::min<uint64_t>((total + 255) / 256, 65535);
+      if (s->col_type == GEMX_TYPE_FLOAT) {
+        hipLaunchKernelGGL((k_tag_p1<GEMX_TYPE_FLOAT>), dim3(b1), dim3(256),
+                           0, s->stream, d_sq, T.d_order, T.d_chunks,
+                           T.n_chunks, d_segq, d_part, T.d_gtmp, P.W0,
+                           (uint32_t)P.n_gwins);
+        hipLaunchKernelGGL((k_tag_p2<GEMX_TYPE_FLOAT>), dim3(b2), dim3(256),
+                           0, s->stream, (const GAcc *)T.d_gtmp, T.d_cstart,
+                           T.d_ccount, T.n_chunks, T.d_rows, T.n_groups, P.W0,
+                           (uint32_t)P.n_gwins, interval, offset, start_time);
+      } else {
+        hipLaunchKernelGGL((k_tag_p1<GEMX_TYPE_INT>), dim3(b1), dim3(256), 0,
+                           s->stream, d_sq, T.d_order, T.d_chunks, T.n_chunks,
+                           d_segq, d_part, T.d_gtmp, P.W0,
+                           (uint32_t)P.n_gwins);
+        hipLaunchKernelGGL((k_tag_p2<GEMX_TYPE_INT>), dim3(b2), dim3(256), 0,
+                           s->stream, (const GAcc *)T.d_gtmp, T.d_cstart,
+                           T.d_ccount, T.n_chunks, T.d_rows, T.n_groups, P.W0,
+                           (uint32_t)P.n_gwins, interval, offset, start_time);
+      }
+    }
+  }
+  if (!tagq && !group_all && total_rows > 0) {
     uint32_t blocks = (uint32_t)std::min<uint64_t>((total_rows + TPB - 1) / TPB, 65535);
     if (s->col_type == GEMX_TYPE_FLOAT)
       hipLaunchKernelGGL((k_merge<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
@@ -2866,7 +3090,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part, d_rows,
                          total_rows, interval, offset, start_time);
   }
-  if (group_all && P.n_gwins > 0) {
+  if (!tagq && group_all && P.n_gwins > 0) {
     uint32_t b1 = (uint32_t)std::min<uint64_t>(P.n_gwins * P.gsplit, 65535);
     uint32_t b2 = (uint32_t)std::min<uint64_t>(P.n_gwins, 65535);
     if (s->col_type == GEMX_TYPE_FLOAT) {
@@ -2894,15 +3118,17 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   DevErr herr = {0};
   HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
                            s->stream));
-  uint64_t fetch_rows = group_all ? P.n_gwins : total_rows;
+  uint64_t fetch_rows = tagq ? (uint64_t)s->tag_plan.n_groups * P.n_gwins
+                             : (group_all ? P.n_gwins : total_rows);
   if (fetch_rows > cap) {
     seterr("output capacity too small");
     return GEMX_E_CAP;
   }
   gemx_agg_row *hrows = out_host;
-  HIP_CHECK(hipMemcpyAsync(hrows, group_all ? P.d_grows : d_rows,
-                           sizeof(gemx_agg_row) * fetch_rows,
-                           hipMemcpyDeviceToHost, s->stream));
+  HIP_CHECK(hipMemcpyAsync(
+      hrows,
+      tagq ? s->tag_plan.d_rows : (group_all ? P.d_grows : d_rows),
+      sizeof(gemx_agg_row) * fetch_rows, hipMemcpyDeviceToHost, s->stream));
   auto t_sync0 = std::chrono::steady_clock::now();
   HIP_CHECK(hipStreamSynchronize(s->stream));
   auto t_sync1 = std::chrono::steady_clock::now();
@@ -3211,6 +3437,31 @@ extern "C" int gemx_scan_agg_ex(gemx_shard *s, int64_t start_time,
                                 uint64_t *n_out, gemx_query_stats *stats) {
   return scan_impl(s, start_time, end_time, interval, offset, group_all,
                    filter_op, filter_f, filter_i, out_host, cap, n_out, stats);
+}
+
+/* hash GROUP BY tag: series_group maps each series (descriptor order) to
+ * a group id < n_groups — the executor's tag-set hash dictionary role
+ * (engine/executor/hash_agg_transform.go); the engine returns one row per
+ * (group, window) with the group id in the sid slot. */
+extern "C" int gemx_scan_agg_tags(gemx_shard *s, const uint32_t *series_group,
+                                  uint32_t n_groups, int64_t start_time,
+                                  int64_t end_time, int64_t interval,
+                                  int64_t offset, gemx_agg_row *out_host,
+                                  uint64_t cap, uint64_t *n_out,
+                                  gemx_query_stats *stats) {
+  if (!s || !series_group || n_groups == 0) {
+    seterr("scan_agg_tags: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  for (size_t i = 0; i < s->series_ranges.size(); i++) {
+    if (series_group[i] >= n_groups) {
+      seterr("series_group value out of range");
+      return GEMX_E_INVALID;
+    }
+  }
+  TagQuery tq{series_group, n_groups};
+  return scan_impl(s, start_time, end_time, interval, offset, 0, 0, 0, 0,
+                   out_host, cap, n_out, stats, nullptr, &tq);
 }
 
 /* ---------------- pre-aggregation metadata ---------------- */

@@ -147,6 +147,12 @@ def _load():
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
         C.POINTER(_Stats),
     ]
+    lib.gemx_scan_agg_tags.restype = C.c_int
+    lib.gemx_scan_agg_tags.argtypes = [
+        C.c_void_p, C.c_void_p, C.c_uint32, C.c_int64, C.c_int64, C.c_int64,
+        C.c_int64, C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64),
+        C.POINTER(_Stats),
+    ]
     lib.gemx_encode_shard.restype = C.c_int
     lib.gemx_encode_shard.argtypes = [
         C.c_int, C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p, C.c_uint64,
@@ -399,6 +405,34 @@ class Shard:
             host_ms=st.h2d_ms, points=st.points,
             compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
             meta_rows=int(nm.value),
+        )
+        return out[: n.value], stats
+
+    def scan_agg_tags(self, series_group, n_groups, start_time, end_time,
+                      interval, offset=0, out_cap=None):
+        """Hash GROUP BY tag (executor/hash_agg_transform.go): series_group
+        maps each series (descriptor order) to a group id < n_groups — the
+        executor's tag-set hash dictionary. One row per (group, window),
+        merged on device; rows carry the group id in the sid field."""
+        lib = self._lib
+        series_group = np.ascontiguousarray(series_group, dtype=np.uint32)
+        if len(series_group) != self._sid_count():
+            raise GemxError("series_group must have one entry per series")
+        if out_cap is None:
+            out_cap = int(n_groups) * self._rows_bound(interval, offset, True)
+        out = self._pooled_out("agg", out_cap, AGG_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_scan_agg_tags(
+            self._h, series_group.ctypes.data_as(C.c_void_p), int(n_groups),
+            start_time, end_time, interval, offset,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        stats = dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            host_ms=st.h2d_ms, points=st.points,
+            compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
 
