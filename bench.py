@@ -101,11 +101,17 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--mode", choices=["walk", "random"], default="walk")
-    ap.add_argument("--query", choices=["mean", "downsample", "rate"], default="mean",
+    ap.add_argument("--query",
+                    choices=["mean", "downsample", "rate", "tags", "preagg"],
+                    default="mean",
                     help="mean: grouped mean/min/max/count GROUP BY time(1m) "
                          "(north star, config #2); downsample: per-series "
                          "first/last/sum to 5m buckets (config #4 shape); "
-                         "rate: PromQL rate(value[5m]) step 1m (config #5 shape)")
+                         "rate: PromQL rate(value[5m]) step 1m (config #5 "
+                         "shape); tags: hash GROUP BY tag, 1000 groups "
+                         "(config #3 high-cardinality shape); preagg: "
+                         "calls-only no-interval query served from pre-agg "
+                         "metadata (SURVEY.md 3d)")
     ap.add_argument("--series", type=int, default=100_000)
     ap.add_argument("--pts", type=int, default=1000)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
@@ -146,6 +152,12 @@ def main():
     shard = gx.Shard(blob, descs, gx.engine.GEMX_TYPE_FLOAT, device=local_rank)
     log(f"[attach] H2D resident in {time.time()-t0:.1f}s")
 
+    NGROUPS = 1000
+    sids_u = descs["sid"]
+    keep = np.ones(len(sids_u), dtype=bool)
+    keep[1:] = sids_u[1:] != sids_u[:-1]
+    gmap = (sids_u[keep] % NGROUPS).astype(np.uint32)
+
     n_wins = (args.pts + 59) // 60 + 1
     w0 = 0  # t0=0, windows start at ordinal 0
     RANGE_NS = 300 * 10**9  # rate(value[5m])
@@ -164,6 +176,25 @@ def main():
             # reduce shape (engine/record_plan.go:1184); output stays
             # per-series (the downsample writer consumes it), no collective
             rows, stats = shard.scan_agg(0, 2**62, DS_NS, group_all=False)
+        elif args.query == "tags":
+            rows, stats = shard.scan_agg_tags(gmap, NGROUPS, 0, 2**62,
+                                              WINDOW_NS)
+            if dist_on:
+                # cross-shard merge per (group, window): count+sum all-reduce
+                import torch
+                import torch.distributed as dist
+
+                acc = np.zeros((NGROUPS, n_wins, 2))
+                gi = rows["sid"].astype(np.int64)
+                wi = (rows["win_start"] // WINDOW_NS).astype(np.int64)
+                np.add.at(acc, (gi, np.clip(wi, 0, n_wins - 1), 0),
+                          rows["count"])
+                np.add.at(acc, (gi, np.clip(wi, 0, n_wins - 1), 1),
+                          rows["sum"])
+                t = torch.from_numpy(acc.reshape(-1)).to(f"cuda:{local_rank}")
+                dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        elif args.query == "preagg":
+            rows, stats = shard.scan_preagg(-2**62, 2**62)
         else:  # rate
             rows, stats = shard.prom_rate(0, (args.pts - 1) * 10**9, RANGE_NS,
                                           WINDOW_NS)
@@ -185,7 +216,7 @@ def main():
         rows, stats = step()
 
     # verification outside the timed region
-    if args.query in ("mean", "downsample"):
+    if args.query in ("mean", "downsample", "tags", "preagg"):
         assert int(rows["count"].sum()) == args.series * args.pts
     else:
         assert len(rows) > 0
@@ -266,6 +297,10 @@ def main():
                 "downsample": "per-series first/last/sum to 5m buckets "
                               "(downsample pipeline reduce shape)",
                 "rate": "PromQL rate(value[5m]) step 1m over range vectors",
+                "tags": "hash GROUP BY tag, 1000 groups x 1m windows merged "
+                        "on device (config #3 high-cardinality shape)",
+                "preagg": "calls-only no-interval query served from pre-agg "
+                          "metadata (after first scan caches it)",
             }[args.query] + f" (mode={args.mode}, {args.series}x{args.pts})",
             "window": "1m",
             "series_per_gpu": args.series,
