@@ -751,7 +751,15 @@ __device__ __forceinline__ int d_filt_pass(int col_type, int filter_op, double f
   }
 }
 
-template <int COLTYPE, int FILT>
+/* GRIDP selects the compiled path: 1 = regular-grid segments only
+ * (const-delta timestamps, non-negative delta, no predicate, interval>0 —
+ * the host routes only such segments here), 0 = the streaming path for
+ * everything else the fast kernel handles. Splitting the two paths into
+ * separate instantiations drops the grid kernel from 157 VGPRs /
+ * 3 waves/SIMD (with the streaming state live) to 116 VGPRs /
+ * 4 waves/SIMD with no VGPR spills — more resident waves to hide the
+ * serial 8-byte blob-refill latency this kernel is bound by. */
+template <int COLTYPE, int FILT, int GRIDP>
 __global__ void __launch_bounds__(256) k_scan_fast(
     const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
     const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
@@ -848,7 +856,11 @@ __global__ void __launch_bounds__(256) k_scan_fast(
         }                                                                          \
       } while (0)
 
-    if (t_const && dtc >= 0 && !FILT && interval) {
+    if (GRIDP && !(t_const && dtc >= 0 && !FILT && interval)) {
+      set_err(err, GEMX_E_INVALID); /* host routing invariant violated */
+      return;
+    }
+    if (GRIDP) {
       /* two-level loop for the regular-grid case: the inner loop runs a
        * whole window's rows with no time/window arithmetic — min/max track
        * ROW indices, times reconstruct at flush (t = t0 + row*dt) */
@@ -918,7 +930,8 @@ __global__ void __launch_bounds__(256) k_scan_fast(
       continue;
     }
 
-    for (int i = 0; i < rows; i++) {
+    if (GRIDP) continue; /* unreachable; keeps the stream path compiled out */
+    for (int i = 0; GRIDP == 0 && i < rows; i++) {
       int64_t t;
       if (t_const) {
         t = t0c + (int64_t)i * dtc;
@@ -2460,6 +2473,8 @@ struct QueryPlan {
   bool clipped = false;
   uint32_t *d_fast_q = nullptr, *d_gen_q = nullptr;
   uint32_t n_fast_q = 0, n_gen_q = 0;
+  uint32_t *d_fastg_q = nullptr, *d_fasts_q = nullptr;
+  uint32_t n_fastg_q = 0, n_fasts_q = 0;
   void *d_gtmp = nullptr; /* GAcc[n_gwins × gsplit] */
   uint32_t gsplit = 1, gper_chunk = 1;
   int64_t W0 = 0;
@@ -2530,6 +2545,10 @@ struct gemx_shard {
   /* segment classification (attach-time) */
   std::vector<uint32_t> fast_ids, general_ids;
   uint32_t *d_fast_ids, *d_general_ids;
+  /* fast split by time codec: grid (const-delta, dt>=0) vs streaming */
+  std::vector<uint32_t> fast_grid_ids, fast_stream_ids;
+  uint32_t *d_fast_grid_ids = nullptr, *d_fast_stream_ids = nullptr;
+  std::vector<char> is_grid; /* per segment */
   /* series grouping: ranges in desc order */
   struct SeriesRange {
     uint64_t sid;
@@ -2565,6 +2584,8 @@ static void free_plan(QueryPlan &p) {
   if (p.d_gtmp) hipFree(p.d_gtmp);
   if (p.d_fast_q) hipFree(p.d_fast_q);
   if (p.d_gen_q) hipFree(p.d_gen_q);
+  if (p.d_fastg_q) hipFree(p.d_fastg_q);
+  if (p.d_fasts_q) hipFree(p.d_fasts_q);
   p = QueryPlan();
 }
 
@@ -2577,22 +2598,43 @@ extern "C" int gemx_device_count(void) {
   return n;
 }
 
-/* attach-time classification: peek headers in HOST memory */
+static int h_uvarint(const uint8_t *p, int64_t len, uint64_t *out) {
+  uint64_t v = 0;
+  int sh = 0, i = 0;
+  while (i < len && i < 10) {
+    v |= (uint64_t)(p[i] & 0x7F) << sh;
+    if (!(p[i] & 0x80)) { *out = v; return i + 1; }
+    sh += 7;
+    i++;
+  }
+  return -1;
+}
+
+/* attach-time classification: peek headers in HOST memory.
+ * *grid = const-delta timestamps with non-negative delta — segments the
+ * leaner GRIDP=1 instantiation of k_scan_fast can take. */
 static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col_type,
-                            bool *fast) {
+                            bool *fast, bool *grid) {
   if (d.data_size < 1 || d.time_size < 1) return GEMX_E_INVALID;
   const uint8_t *ds = blob + d.data_offset;
   const uint8_t *ts = blob + d.time_offset;
   *fast = true;
+  *grid = false;
   uint8_t dt = ds[0];
   /* time: one-value or Full + {const-delta, simple8b, uncompressed} */
   if (ts[0] == 17) {
-    /* one row: fine for fast path */
+    *grid = true; /* single row: delta 0 */
   } else if (ts[0] == 31) {
     if (d.time_size < 6) return GEMX_E_INVALID;
     int ttag = ts[5] >> 4;
     if (ttag == 3) *fast = false; /* snappy times need scratch */
     else if (ttag != 1 && ttag != 2 && ttag != 4) return GEMX_E_INVALID;
+    if (ttag == 1 && d.time_size >= 15) {
+      uint64_t delta = 0;
+      if (h_uvarint(ts + 6 + 8, d.time_size - 14, &delta) > 0 &&
+          (int64_t)delta >= 0)
+        *grid = true;
+    }
   } else
     return GEMX_E_INVALID;
   /* data */
@@ -2664,8 +2706,8 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       delete s;
       return GEMX_E_INVALID;
     }
-    bool fast;
-    int rc = classify_segment(hb, d, col_type, &fast);
+    bool fast, grid;
+    int rc = classify_segment(hb, d, col_type, &fast, &grid);
     if (rc != 0) {
       seterr(rc == GEMX_E_UNSUPPORTED
                  ? "segment uses a codec not yet on-device (zstd/MLF/legacy)"
@@ -2673,10 +2715,17 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       delete s;
       return rc;
     }
-    if (fast)
+    if (fast) {
       s->fast_ids.push_back((uint32_t)i);
-    else
+      if (grid)
+        s->fast_grid_ids.push_back((uint32_t)i);
+      else
+        s->fast_stream_ids.push_back((uint32_t)i);
+    } else {
       s->general_ids.push_back((uint32_t)i);
+    }
+    if (s->is_grid.size() < i + 1) s->is_grid.resize(nsegs, 0);
+    s->is_grid[i] = fast && grid;
     s->total_rows_scanned += d.rows;
     /* series ranges + per-series/shard time bounds (for preagg coverage) */
     if (s->series_ranges.empty() || s->series_ranges.back().sid != d.sid) {
@@ -2710,6 +2759,20 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     HIP_CHECK(hipMemcpyAsync(s->d_fast_ids, s->fast_ids.data(),
                              sizeof(uint32_t) * s->fast_ids.size(),
                              hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&s->d_fast_grid_ids,
+                      sizeof(uint32_t) *
+                          (s->fast_grid_ids.empty() ? 1 : s->fast_grid_ids.size())));
+  if (!s->fast_grid_ids.empty())
+    HIP_CHECK(hipMemcpyAsync(s->d_fast_grid_ids, s->fast_grid_ids.data(),
+                             sizeof(uint32_t) * s->fast_grid_ids.size(),
+                             hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&s->d_fast_stream_ids,
+                      sizeof(uint32_t) *
+                          (s->fast_stream_ids.empty() ? 1 : s->fast_stream_ids.size())));
+  if (!s->fast_stream_ids.empty())
+    HIP_CHECK(hipMemcpyAsync(s->d_fast_stream_ids, s->fast_stream_ids.data(),
+                             sizeof(uint32_t) * s->fast_stream_ids.size(),
+                             hipMemcpyHostToDevice, s->stream));
   HIP_CHECK(hipMalloc(&s->d_general_ids,
                       sizeof(uint32_t) *
                           (s->general_ids.empty() ? 1 : s->general_ids.size())));
@@ -2732,6 +2795,8 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   hipFree(s->d_blob);
   hipFree(s->d_descs);
   hipFree(s->d_fast_ids);
+  hipFree(s->d_fast_grid_ids);
+  hipFree(s->d_fast_stream_ids);
   hipFree(s->d_general_ids);
   hipStreamDestroy(s->stream);
   delete s;
@@ -2769,7 +2834,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.sq.resize(s->series_ranges.size());
     P.partial_slots = 0;
     P.total_rows = 0;
-    std::vector<uint32_t> fast_q, gen_q;
+    std::vector<uint32_t> fast_q, gen_q, fast_gq, fast_sq;
     std::vector<char> is_gen(nsegs, 0);
     for (auto id : s->general_ids) is_gen[id] = 1;
     bool any_clip = false;
@@ -2821,6 +2886,10 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
           gen_q.push_back(i);
         } else {
           fast_q.push_back(i);
+          if (s->is_grid[i])
+            fast_gq.push_back(i);
+          else
+            fast_sq.push_back(i);
         }
       }
       P.sq[g].sid = r.sid;
@@ -2850,6 +2919,20 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       if (!gen_q.empty())
         HIP_CHECK(hipMemcpyAsync(P.d_gen_q, gen_q.data(),
                                  sizeof(uint32_t) * gen_q.size(),
+                                 hipMemcpyHostToDevice, s->stream));
+      P.n_fastg_q = (uint32_t)fast_gq.size();
+      P.n_fasts_q = (uint32_t)fast_sq.size();
+      HIP_CHECK(hipMalloc(&P.d_fastg_q,
+                          sizeof(uint32_t) * (fast_gq.empty() ? 1 : fast_gq.size())));
+      if (!fast_gq.empty())
+        HIP_CHECK(hipMemcpyAsync(P.d_fastg_q, fast_gq.data(),
+                                 sizeof(uint32_t) * fast_gq.size(),
+                                 hipMemcpyHostToDevice, s->stream));
+      HIP_CHECK(hipMalloc(&P.d_fasts_q,
+                          sizeof(uint32_t) * (fast_sq.empty() ? 1 : fast_sq.size())));
+      if (!fast_sq.empty())
+        HIP_CHECK(hipMemcpyAsync(P.d_fasts_q, fast_sq.data(),
+                                 sizeof(uint32_t) * fast_sq.size(),
                                  hipMemcpyHostToDevice, s->stream));
     }
     HIP_CHECK(hipMalloc(&P.d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
@@ -2926,44 +3009,63 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   const int TPB = 256;
   const uint32_t *fast_list = s->d_fast_ids;
   uint32_t fast_n = (uint32_t)s->fast_ids.size();
+  const uint32_t *grid_list = s->d_fast_grid_ids;
+  uint32_t grid_n = (uint32_t)s->fast_grid_ids.size();
+  const uint32_t *strm_list = s->d_fast_stream_ids;
+  uint32_t strm_n = (uint32_t)s->fast_stream_ids.size();
   const uint32_t *gen_list = s->d_general_ids;
   uint32_t gen_n = (uint32_t)s->general_ids.size();
   if (P.clipped) {
     fast_list = P.d_fast_q;
     fast_n = P.n_fast_q;
+    grid_list = P.d_fastg_q;
+    grid_n = P.n_fastg_q;
+    strm_list = P.d_fasts_q;
+    strm_n = P.n_fasts_q;
     gen_list = P.d_gen_q;
     gen_n = P.n_gen_q;
   }
-  if (fast_n > 0) {
-    uint32_t n = fast_n;
+  /* the lean GRIDP=1 instantiation takes const-delta-time segments when
+   * the query has an interval and no predicate; everything else fast goes
+   * through the streaming instantiation */
+  const bool use_grid = (interval != 0) && (filter_op == 0);
+  struct FastLaunch {
+    const uint32_t *list;
+    uint32_t n;
+    int gridp;
+  } launches[2];
+  int n_launches = 0;
+  if (use_grid) {
+    if (grid_n) launches[n_launches++] = {grid_list, grid_n, 1};
+    if (strm_n) launches[n_launches++] = {strm_list, strm_n, 0};
+  } else if (fast_n) {
+    launches[n_launches++] = {fast_list, fast_n, 0};
+  }
+  for (int li = 0; li < n_launches; li++) {
+    uint32_t n = launches[li].n;
     uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
+    const uint32_t *lst = launches[li].list;
+#define GEMX_LAUNCH_FAST(CT, FLT, GP)                                          \
+    hipLaunchKernelGGL((k_scan_fast<CT, FLT, GP>), dim3(blocks), dim3(TPB), 0, \
+                       s->stream, s->d_blob, s->d_descs, d_segq, lst, n,       \
+                       d_part, interval, offset, start_time, end_time,         \
+                       filter_op, filter_f, filter_i, d_err)
     if (s->col_type == GEMX_TYPE_FLOAT) {
-      if (filter_op)
-        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT, 1>), dim3(blocks),
-                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           (const uint32_t *)fast_list, n, d_part, interval,
-                           offset, start_time, end_time, filter_op, filter_f,
-                           filter_i, d_err);
+      if (launches[li].gridp)
+        GEMX_LAUNCH_FAST(GEMX_TYPE_FLOAT, 0, 1);
+      else if (filter_op)
+        GEMX_LAUNCH_FAST(GEMX_TYPE_FLOAT, 1, 0);
       else
-        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_FLOAT, 0>), dim3(blocks),
-                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           (const uint32_t *)fast_list, n, d_part, interval,
-                           offset, start_time, end_time, filter_op, filter_f,
-                           filter_i, d_err);
+        GEMX_LAUNCH_FAST(GEMX_TYPE_FLOAT, 0, 0);
     } else {
-      if (filter_op)
-        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT, 1>), dim3(blocks),
-                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           (const uint32_t *)fast_list, n, d_part, interval,
-                           offset, start_time, end_time, filter_op, filter_f,
-                           filter_i, d_err);
+      if (launches[li].gridp)
+        GEMX_LAUNCH_FAST(GEMX_TYPE_INT, 0, 1);
+      else if (filter_op)
+        GEMX_LAUNCH_FAST(GEMX_TYPE_INT, 1, 0);
       else
-        hipLaunchKernelGGL((k_scan_fast<GEMX_TYPE_INT, 0>), dim3(blocks),
-                           dim3(TPB), 0, s->stream, s->d_blob, s->d_descs, d_segq,
-                           (const uint32_t *)fast_list, n, d_part, interval,
-                           offset, start_time, end_time, filter_op, filter_f,
-                           filter_i, d_err);
+        GEMX_LAUNCH_FAST(GEMX_TYPE_INT, 0, 0);
     }
+#undef GEMX_LAUNCH_FAST
   }
   if (gen_n > 0) {
     if (!d_scratch) {
