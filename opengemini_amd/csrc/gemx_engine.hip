@@ -129,8 +129,24 @@ struct BitR {
       pos = len;
       pwbits = rem * 8;
     } else {
+      pw = 0; /* fill_one ORs pw unconditionally — keep it zero */
       pwbits = 0;
     }
+  }
+
+  /* branchless single insert for the hot consume path: precondition
+   * have <= 64, pw left-aligned with zero low bits (pw == 0 when the
+   * stream is exhausted). One unconditional 128-bit funnel OR replaces
+   * the 4-way shift branch + while loop of fill() — the decode loop is
+   * issue-bound (SQ: ~150% port demand at 4 waves), so every removed
+   * branch/VALU slot is wall time. Availability floor matches fill():
+   * have' = have + 64 >= 64 on full words. */
+  __device__ __forceinline__ void fill_one() {
+    unsigned __int128 ins = ((unsigned __int128)pw << 64) >> have;
+    hi |= (uint64_t)(ins >> 64);
+    lo |= (uint64_t)ins;
+    have += pwbits;
+    preload();
   }
 
   /* whole-word fill: inserts the prefetched word when it fully fits
@@ -184,18 +200,12 @@ struct BitR {
 
   /* consume up to 128 bits (the Gorilla worst case per value is 77) */
   __device__ __forceinline__ void consume(int n) {
-    if (n >= 64) {
-      hi = lo;
-      lo = 0;
-      have -= 64;
-      n -= 64;
-    }
-    if (n) {
-      hi = (hi << n) | (lo >> (64 - n));
-      lo <<= n;
-      have -= n;
-    }
-    if (have <= 64) fill();
+    /* n <= 77 < 128: one u128 shift, no n>=64 branch */
+    unsigned __int128 w = (((unsigned __int128)hi << 64) | lo) << n;
+    hi = (uint64_t)(w >> 64);
+    lo = (uint64_t)w;
+    have -= n;
+    if (have <= 64) fill_one();
   }
 };
 
