@@ -308,11 +308,14 @@ struct FloatIter {
         return 0;
       }
       if (g_done) return -1;
-      if (br.have >= 13) {
+      {
         /* branchless control decode: peek 13 bits (ctrl1+ctrl2+5 leading+
          * 6 meaningful), then the significant bits at a known offset —
          * one predicated path instead of three divergent ones
-         * (semantics identical to batch_float.go:384-505) */
+         * (semantics identical to batch_float.go:384-505). The peek is
+         * pure ALU on the (zero-padded) window, so it runs before the
+         * availability check; ONE branch guards both the 13-bit header
+         * and the record body (issue-bound loop: branches are the cost) */
         uint32_t p13 = (uint32_t)(br.hi >> 51);
         int ctrl1 = (int)(p13 >> 12);
         int newwin = ctrl1 & (int)((p13 >> 11) & 1);
@@ -325,19 +328,17 @@ struct FloatIter {
         uint8_t eff_trail = newwin ? trail_new : g_trail;
         int hdr = ctrl1 ? (newwin ? 13 : 2) : 1;
         int nbits = hdr + (ctrl1 ? (int)eff_mean : 0);
-        if (br.have >= nbits) {
-          if (newwin) {
-            g_mean = mean_new;
-            g_trail = trail_new;
-          }
-          if (ctrl1) {
-            uint64_t x = (br.hi << hdr) | (br.lo >> (64 - hdr));
-            uint64_t sbits = (eff_mean == 64) ? x : (x >> (64 - eff_mean));
-            g_val ^= sbits << (eff_trail & 0x3F);
-            if (g_val == UVNAN) {
-              g_done = 1;
-              return -1;
-            }
+        int need = nbits < 13 ? 13 : nbits;
+        if (br.have >= need) {
+          g_mean = newwin ? mean_new : g_mean;
+          g_trail = newwin ? trail_new : g_trail;
+          uint64_t x = (br.hi << hdr) | (br.lo >> (64 - hdr));
+          uint64_t sbits = (eff_mean == 64) ? x : (x >> (64 - eff_mean));
+          uint64_t mask = (uint64_t)0 - (uint64_t)ctrl1;
+          g_val ^= (sbits << (eff_trail & 0x3F)) & mask;
+          if (g_val == UVNAN) {
+            g_done = 1;
+            return -1;
           }
           br.consume(nbits);
           uint64_t u = g_val;
