@@ -170,6 +170,12 @@ int gemx_scan_agg_ex(gemx_shard *, int64_t start_time, int64_t end_time,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                      gemx_query_stats *stats);
 
+/* Pin a caller-owned output buffer (hipHostRegister) so row fetches into
+ * it run at pinned-DMA speed; optional — unregistered buffers work too.
+ * Mirrors why the reference pools records (aggregate_cursor.go:100). */
+int gemx_host_register(void *p, uint64_t bytes);
+int gemx_host_unregister(void *p);
+
 /* hash GROUP BY tag (engine/executor/hash_agg_transform.go): the executor
  * hashes each series' tag set into a group; series_group passes that
  * sid→group mapping (one uint32 per series, descriptor order, values

@@ -3577,6 +3577,27 @@ extern "C" int gemx_scan_agg_tags(gemx_shard *s, const uint32_t *series_group,
                    out_host, cap, n_out, stats, nullptr, &tq);
 }
 
+/* Pin caller-owned host buffers (hipHostRegister) so D2H row fetches run
+ * at pinned-copy speed. The Python host layer registers its pooled output
+ * buffers once at allocation — the reference's CircularRecordPool keeps
+ * records pooled for the same reason (engine/aggregate_cursor.go:100).
+ * Registration is an optimization only: unregistered buffers still work. */
+extern "C" int gemx_host_register(void *p, uint64_t bytes) {
+  if (!p || !bytes) return GEMX_E_INVALID;
+  hipError_t e = hipHostRegister(p, bytes, hipHostRegisterPortable);
+  if (e != hipSuccess && e != hipErrorHostMemoryAlreadyRegistered) {
+    seterr("hipHostRegister failed");
+    return GEMX_E_HIP;
+  }
+  return GEMX_OK;
+}
+
+extern "C" int gemx_host_unregister(void *p) {
+  if (!p) return GEMX_E_INVALID;
+  hipHostUnregister(p);
+  return GEMX_OK;
+}
+
 /* ---------------- pre-aggregation metadata ---------------- */
 /* The reference stores per-chunk FloatPreAgg/IntegerPreAgg in ColumnMeta at
  * flush time (pre_aggregation.go:410,:330) and serves matchPreAgg queries

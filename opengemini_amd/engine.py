@@ -147,6 +147,10 @@ def _load():
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
         C.POINTER(_Stats),
     ]
+    lib.gemx_host_register.restype = C.c_int
+    lib.gemx_host_register.argtypes = [C.c_void_p, C.c_uint64]
+    lib.gemx_host_unregister.restype = C.c_int
+    lib.gemx_host_unregister.argtypes = [C.c_void_p]
     lib.gemx_scan_agg_tags.restype = C.c_int
     lib.gemx_scan_agg_tags.argtypes = [
         C.c_void_p, C.c_void_p, C.c_uint32, C.c_int64, C.c_int64, C.c_int64,
@@ -282,6 +286,14 @@ class Shard:
         if getattr(self, "_h", None):
             self._lib.gemx_shard_close(self._h)
             self._h = None
+        # unregister pinned pooled buffers BEFORE they are garbage-collected:
+        # a stale hipHostRegister range over freed pages breaks later copies
+        for arr in getattr(self, "_out_cache", {}).values():
+            try:
+                self._lib.gemx_host_unregister(arr.ctypes.data_as(C.c_void_p))
+            except Exception:
+                pass
+        self._out_cache = {}
 
     def __del__(self):
         try:
@@ -295,8 +307,13 @@ class Shard:
         engine/aggregate_cursor.go:100)."""
         cur = self._out_cache.get(kind)
         if cur is None or len(cur) < cap or cur.dtype != dtype:
+            if cur is not None:
+                self._lib.gemx_host_unregister(cur.ctypes.data_as(C.c_void_p))
             cur = np.empty(cap, dtype=dtype)
             cur[:] = 0  # touch pages once
+            # pin the pooled buffer: D2H row fetches then run at DMA speed
+            self._lib.gemx_host_register(
+                cur.ctypes.data_as(C.c_void_p), cur.nbytes)
             self._out_cache[kind] = cur
         return cur
 
