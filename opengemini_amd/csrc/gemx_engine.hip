@@ -731,6 +731,9 @@ struct SeriesQ {
 
 struct DevErr {
   int code; /* first error wins */
+  unsigned int _pad;
+  unsigned long long gaps; /* gap rows written by the merge kernels; the
+                              host skips its compaction scan when zero */
 };
 
 __device__ __forceinline__ void set_err(DevErr *e, int code) {
@@ -1519,7 +1522,7 @@ __global__ void __launch_bounds__(256) k_merge(
     const SeriesQ *__restrict__ series, uint32_t nseries,
     const SegQ *__restrict__ segq, const Partial *__restrict__ partials,
     gemx_agg_row *__restrict__ rows, uint64_t total_rows, int64_t interval,
-    int64_t offset, int64_t q_start) {
+    int64_t offset, int64_t q_start, DevErr *__restrict__ err) {
   uint64_t gid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
   for (uint64_t r = gid; r < total_rows; r += gridDim.x * (uint64_t)blockDim.x) {
     uint32_t lo = 0, hi = nseries - 1;
@@ -1534,8 +1537,11 @@ __global__ void __launch_bounds__(256) k_merge(
     memset(&out, 0, sizeof(out));
     out.sid = s.sid;
     out.win_start = interval ? win_start_of(w, interval, offset) : q_start;
-    if (!merge_series_window<COLTYPE>(s, segq, partials, w, &out))
+    if (!merge_series_window<COLTYPE>(s, segq, partials, w, &out)) {
       out.count = -1; /* gap row: host compacts it away */
+      atomicAdd(&err->gaps, 1ull); /* gaps are rare: the host skips its
+                                      whole compaction scan when zero */
+    }
     rows[r] = out;
   }
 }
@@ -1740,7 +1746,7 @@ template <int COLTYPE>
 __global__ void __launch_bounds__(256) k_group_p2(
     const GAcc *__restrict__ gtmp, uint32_t split, gemx_agg_row *__restrict__ out,
     int64_t W0, uint32_t n_gwins, int64_t interval, int64_t offset,
-    int64_t q_start) {
+    int64_t q_start, DevErr *__restrict__ err) {
   __shared__ GAcc sh[256];
   for (uint32_t wb = blockIdx.x; wb < n_gwins; wb += gridDim.x) {
     int64_t w = W0 + (int64_t)wb;
@@ -1764,6 +1770,7 @@ __global__ void __launch_bounds__(256) k_group_p2(
       o.win_start = ws;
       if (!g->used) {
         o.count = -1; /* gap */
+        atomicAdd(&err->gaps, 1ull);
       } else {
         o.first_row_time = ws; /* BuildEmptyIntervalRec interval times */
         o.count = g->count;
@@ -1849,7 +1856,8 @@ __global__ void __launch_bounds__(256) k_tag_p2(
     const GAcc *__restrict__ gtmp, const uint32_t *__restrict__ cstart,
     const uint32_t *__restrict__ ccount, uint32_t n_chunks,
     gemx_agg_row *__restrict__ out, uint32_t n_groups, int64_t W0,
-    uint32_t n_gwins, int64_t interval, int64_t offset, int64_t q_start) {
+    uint32_t n_gwins, int64_t interval, int64_t offset, int64_t q_start,
+    DevErr *__restrict__ err) {
   uint64_t total = (uint64_t)n_groups * n_gwins;
   for (uint64_t r = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x; r < total;
        r += gridDim.x * (uint64_t)blockDim.x) {
@@ -1867,6 +1875,7 @@ __global__ void __launch_bounds__(256) k_tag_p2(
     o.win_start = ws;
     if (!a.used) {
       o.count = -1; /* gap: host compacts */
+      atomicAdd(&err->gaps, 1ull);
     } else {
       o.first_row_time = ws;
       o.count = a.count;
@@ -2299,7 +2308,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     const RateSeriesQ *__restrict__ series, uint32_t nseries,
     const RateSegQ *__restrict__ rsegq, const RatePartial *__restrict__ partials,
     gemx_rate_row *__restrict__ rows, uint64_t total_rows, int64_t start_sample,
-    int64_t step_ns, int64_t range_ns, int is_rate, int is_counter, int func) {
+    int64_t step_ns, int64_t range_ns, int is_rate, int is_counter, int func,
+    DevErr *__restrict__ err) {
   uint64_t gid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
   for (uint64_t r = gid; r < total_rows; r += gridDim.x * (uint64_t)blockDim.x) {
     uint32_t lo = 0, hi = nseries - 1;
@@ -2396,7 +2406,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
         out.value = (func == GEMX_PF_COUNT_OT) ? (double)acc.count : acc.first_v;
         out.isnil = 0;
       }
-      rows[r] = out;
+      if (out.isnil) atomicAdd(&err->gaps, 1ull);
+    rows[r] = out;
       continue;
     }
     if (func == 1) {
@@ -2411,7 +2422,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
         out.value = rv;
         out.isnil = 0;
       }
-      rows[r] = out;
+      if (out.isnil) atomicAdd(&err->gaps, 1ull);
+    rows[r] = out;
       continue;
     }
     if (acc.count > 1 && acc.last_t != acc.first_t && range_ns != 0) {
@@ -2436,6 +2448,7 @@ __global__ void __launch_bounds__(256) k_rate_merge(
       out.value = result;
       out.isnil = 0;
     }
+    if (out.isnil) atomicAdd(&err->gaps, 1ull);
     rows[r] = out;
   }
 }
@@ -3179,7 +3192,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
         hipLaunchKernelGGL((k_tag_p2<GEMX_TYPE_FLOAT>), dim3(b2), dim3(256),
                            0, s->stream, (const GAcc *)T.d_gtmp, T.d_cstart,
                            T.d_ccount, T.n_chunks, T.d_rows, T.n_groups, P.W0,
-                           (uint32_t)P.n_gwins, interval, offset, start_time);
+                           (uint32_t)P.n_gwins, interval, offset, start_time,
+                           d_err);
       } else {
         hipLaunchKernelGGL((k_tag_p1<GEMX_TYPE_INT>), dim3(b1), dim3(256), 0,
                            s->stream, d_sq, T.d_order, T.d_chunks, T.n_chunks,
@@ -3188,7 +3202,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
         hipLaunchKernelGGL((k_tag_p2<GEMX_TYPE_INT>), dim3(b2), dim3(256), 0,
                            s->stream, (const GAcc *)T.d_gtmp, T.d_cstart,
                            T.d_ccount, T.n_chunks, T.d_rows, T.n_groups, P.W0,
-                           (uint32_t)P.n_gwins, interval, offset, start_time);
+                           (uint32_t)P.n_gwins, interval, offset, start_time,
+                           d_err);
       }
     }
   }
@@ -3197,11 +3212,11 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     if (s->col_type == GEMX_TYPE_FLOAT)
       hipLaunchKernelGGL((k_merge<GEMX_TYPE_FLOAT>), dim3(blocks), dim3(TPB), 0,
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part, d_rows,
-                         total_rows, interval, offset, start_time);
+                         total_rows, interval, offset, start_time, d_err);
     else
       hipLaunchKernelGGL((k_merge<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part, d_rows,
-                         total_rows, interval, offset, start_time);
+                         total_rows, interval, offset, start_time, d_err);
   }
   if (!tagq && group_all && P.n_gwins > 0) {
     uint32_t b1 = (uint32_t)std::min<uint64_t>(P.n_gwins * P.gsplit, 65535);
@@ -3213,7 +3228,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          P.gper_chunk);
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_FLOAT>), dim3(b2), dim3(256), 0,
                          s->stream, (const GAcc *)P.d_gtmp, P.gsplit, P.d_grows,
-                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time);
+                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time,
+                         d_err);
     } else {
       hipLaunchKernelGGL((k_group_p1<GEMX_TYPE_INT>), dim3(b1), dim3(256), 0,
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part,
@@ -3221,7 +3237,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          P.gper_chunk);
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_INT>), dim3(b2), dim3(256), 0,
                          s->stream, (const GAcc *)P.d_gtmp, P.gsplit, P.d_grows,
-                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time);
+                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time,
+                         d_err);
     }
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
@@ -3260,9 +3277,14 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     return herr.code;
   }
 
-  /* in-place gap compaction (count == -1); no copy until the first gap */
+  /* in-place gap compaction (count == -1). The merge kernels counted the
+   * gap rows on device, so the common all-populated case skips the host
+   * scan entirely (reading one field of every row still streams half the
+   * buffer through the cores). */
   uint64_t n = 0;
-  {
+  if (herr.gaps == 0) {
+    n = fetch_rows;
+  } else {
     uint64_t i = 0;
     while (i < fetch_rows && hrows[i].count >= 0) i++;
     n = i;
@@ -3480,7 +3502,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     hipLaunchKernelGGL(k_rate_merge, dim3(blocks), dim3(TPB), 0, s->stream,
                        P.d_rsq, (uint32_t)P.rsq.size(), P.d_rsegq, P.d_rpart,
                        P.d_rrows, P.total_rows, start_sample, eff_step, range_ns,
-                       is_rate, is_counter, func);
+                       is_rate, is_counter, func, d_err);
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
@@ -3513,7 +3535,9 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   /* in-place compact: only non-nil rows leave (the reference appends only
    * non-nil, prom reducer append path) */
   uint64_t n = 0;
-  {
+  if (herr.gaps == 0) {
+    n = P.total_rows;
+  } else {
     uint64_t i = 0;
     while (i < P.total_rows && !out_host[i].isnil) i++;
     n = i;
