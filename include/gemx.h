@@ -170,6 +170,22 @@ int gemx_scan_agg_ex(gemx_shard *, int64_t start_time, int64_t end_time,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                      gemx_query_stats *stats);
 
+/* Async query pipeline — the cursor read-ahead model (the reference's
+ * cursor pump also reads ahead of the consumer): gemx_scan_agg_begin
+ * enqueues a plain (group_all=0) or grouped (group_all=1) scan and
+ * returns immediately; gemx_scan_agg_finish waits for the OLDEST
+ * in-flight query, validates and compacts its rows, and fills stats.
+ * Up to two queries may be in flight (double-buffered device rows); the
+ * same (start,end,interval,offset) must be used while queries are in
+ * flight (the plan cannot rebuild under them). The caller's out_host
+ * must stay untouched until the matching finish; register it with
+ * gemx_host_register for full overlap. */
+int gemx_scan_agg_begin(gemx_shard *, int64_t start_time, int64_t end_time,
+                        int64_t interval, int64_t offset, int group_all,
+                        gemx_agg_row *out_host, uint64_t cap);
+int gemx_scan_agg_finish(gemx_shard *, uint64_t *n_out,
+                         gemx_query_stats *stats);
+
 /* Pin a caller-owned output buffer (hipHostRegister) so row fetches into
  * it run at pinned-DMA speed; optional — unregistered buffers work too.
  * Mirrors why the reference pools records (aggregate_cursor.go:100). */

@@ -2484,13 +2484,14 @@ struct QueryPlan {
   SegQ *d_segq = nullptr;
   SeriesQ *d_sq = nullptr;
   Partial *d_part = nullptr;
-  gemx_agg_row *d_rows = nullptr;
-  DevErr *d_err = nullptr;
+  gemx_agg_row *d_rows2[2] = {nullptr, nullptr}; /* double-buffered: the
+      copy of slot A's rows overlaps slot B's kernels (async API) */
+  DevErr *d_err2[2] = {nullptr, nullptr};
   gemx_agg_row *h_rows = nullptr; /* pinned staging */
   uint8_t *d_scratch = nullptr;
   uint32_t gen_lanes = 0;
   /* grouped output (all-series GROUP BY time) */
-  gemx_agg_row *d_grows = nullptr;
+  gemx_agg_row *d_grows2[2] = {nullptr, nullptr};
   gemx_agg_row *h_grows = nullptr;
   /* query-scoped segment routing when the time range clips segments:
    * boundary-crossing segments go through the general (slicing) kernel */
@@ -2581,6 +2582,22 @@ struct gemx_shard {
   std::vector<SeriesRange> series_ranges;
   uint64_t total_rows_scanned; /* Σ rows */
   hipStream_t stream;
+  /* async pipeline: kernels run on `stream`, row D2H on `copy_stream`
+   * gated by an event — the copy of query i overlaps the kernels of
+   * query i+1 (double-buffered row buffers in the plan). The cursor
+   * surface exposes this as begin/finish (the reference's cursor pump
+   * reads ahead the same way). */
+  hipStream_t copy_stream = nullptr;
+  hipEvent_t ev_q[2][3];        /* per-slot decode start/mid/end */
+  hipEvent_t ev_copy[2];        /* per-slot copy-done */
+  DevErr *h_err2[2] = {nullptr, nullptr}; /* pinned per-slot error+gaps */
+  struct PendingScan {
+    bool active = false;
+    int slot = 0;
+    uint64_t fetch_rows = 0;
+    gemx_agg_row *out = nullptr;
+  } pend[2];
+  int pend_head = 0, pend_count = 0, q_slot = 0;
   QueryPlan plan;
   RatePlan rate_plan;
   /* pre-aggregation metadata (pre_aggregation.go FloatPreAgg role): one
@@ -2599,11 +2616,13 @@ static void free_plan(QueryPlan &p) {
   if (p.d_segq) hipFree(p.d_segq);
   if (p.d_sq) hipFree(p.d_sq);
   if (p.d_part) hipFree(p.d_part);
-  if (p.d_rows) hipFree(p.d_rows);
-  if (p.d_err) hipFree(p.d_err);
+  for (int i = 0; i < 2; i++) {
+    if (p.d_rows2[i]) hipFree(p.d_rows2[i]);
+    if (p.d_err2[i]) hipFree(p.d_err2[i]);
+    if (p.d_grows2[i]) hipFree(p.d_grows2[i]);
+  }
   if (p.d_scratch) hipFree(p.d_scratch);
   if (p.h_rows) hipHostFree(p.h_rows);
-  if (p.d_grows) hipFree(p.d_grows);
   if (p.h_grows) hipHostFree(p.h_grows);
   if (p.d_gtmp) hipFree(p.d_gtmp);
   if (p.d_fast_q) hipFree(p.d_fast_q);
@@ -2771,6 +2790,12 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
   }
 
   HIP_CHECK(hipStreamCreate(&s->stream));
+  HIP_CHECK(hipStreamCreate(&s->copy_stream));
+  for (int sl = 0; sl < 2; sl++) {
+    for (int e = 0; e < 3; e++) HIP_CHECK(hipEventCreate(&s->ev_q[sl][e]));
+    HIP_CHECK(hipEventCreate(&s->ev_copy[sl]));
+    HIP_CHECK(hipHostMalloc(&s->h_err2[sl], sizeof(DevErr)));
+  }
   HIP_CHECK(hipMalloc(&s->d_blob, blob_bytes ? blob_bytes : 1));
   HIP_CHECK(hipMemcpyAsync(s->d_blob, blob, blob_bytes, hipMemcpyHostToDevice,
                            s->stream));
@@ -2823,6 +2848,12 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   hipFree(s->d_fast_stream_ids);
   hipFree(s->d_general_ids);
   hipStreamDestroy(s->stream);
+  if (s->copy_stream) hipStreamDestroy(s->copy_stream);
+  for (int sl = 0; sl < 2; sl++) {
+    for (int e = 0; e < 3; e++) hipEventDestroy(s->ev_q[sl][e]);
+    hipEventDestroy(s->ev_copy[sl]);
+    if (s->h_err2[sl]) hipHostFree(s->h_err2[sl]);
+  }
   delete s;
   return GEMX_OK;
 }
@@ -2838,13 +2869,29 @@ struct TagQuery {
   uint32_t n_groups;
 };
 
+static int scan_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
+                        gemx_agg_row *out_host, uint64_t *n_out,
+                        gemx_query_stats *stats);
+
 static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                      int64_t interval, int64_t offset, int group_all,
                      int filter_op, double filter_f, int64_t filter_i,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                      gemx_query_stats *stats, const char *skip_series = nullptr,
-                     const TagQuery *tagq = nullptr) {
+                     const TagQuery *tagq = nullptr, int async_begin = 0) {
   if (!s) return GEMX_E_INVALID;
+  if (async_begin && (tagq || skip_series)) {
+    seterr("async begin supports plain/grouped scans only");
+    return GEMX_E_INVALID;
+  }
+  if (async_begin && s->pend_count >= 2) {
+    seterr("two queries already in flight: call gemx_scan_agg_finish");
+    return GEMX_E_INVALID;
+  }
+  if (!async_begin && s->pend_count > 0) {
+    seterr("async queries in flight: call gemx_scan_agg_finish first");
+    return GEMX_E_INVALID;
+  }
   HIP_CHECK(hipSetDevice(s->device));
   const uint64_t nsegs = s->nsegs;
   const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960 + 512; /* +512: clip bitmap */
@@ -2852,6 +2899,10 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   QueryPlan &P = skip_series ? s->sub_plan : s->plan;
   if (!P.valid || P.start != start_time || P.end != end_time ||
       P.interval != interval || P.offset != offset) {
+    if (s->pend_count > 0) {
+      seterr("cannot rebuild the query plan with queries in flight");
+      return GEMX_E_INVALID;
+    }
     free_plan(P);
     /* host precompute: per-segment window spans, per-series output ranges */
     P.segq.resize(nsegs);
@@ -2968,9 +3019,11 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                              hipMemcpyHostToDevice, s->stream));
     HIP_CHECK(hipMalloc(&P.d_part,
                         sizeof(Partial) * (P.partial_slots ? P.partial_slots : 1)));
-    HIP_CHECK(hipMalloc(&P.d_rows,
-                        sizeof(gemx_agg_row) * (P.total_rows ? P.total_rows : 1)));
-    HIP_CHECK(hipMalloc(&P.d_err, sizeof(DevErr)));
+    for (int sl = 0; sl < 2; sl++) {
+      HIP_CHECK(hipMalloc(&P.d_rows2[sl],
+                          sizeof(gemx_agg_row) * (P.total_rows ? P.total_rows : 1)));
+      HIP_CHECK(hipMalloc(&P.d_err2[sl], sizeof(DevErr)));
+    }
     HIP_CHECK(hipHostMalloc(&P.h_rows,
                             sizeof(gemx_agg_row) * (P.total_rows ? P.total_rows : 1)));
     if (!s->general_ids.empty()) {
@@ -2986,8 +3039,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       }
       P.W0 = P.sq.empty() ? 0 : W0;
       P.n_gwins = P.sq.empty() ? 0 : (uint64_t)(W1 - W0 + 1);
-      HIP_CHECK(hipMalloc(&P.d_grows,
-                          sizeof(gemx_agg_row) * (P.n_gwins ? P.n_gwins : 1)));
+      for (int sl = 0; sl < 2; sl++)
+        HIP_CHECK(hipMalloc(&P.d_grows2[sl],
+                            sizeof(gemx_agg_row) * (P.n_gwins ? P.n_gwins : 1)));
       HIP_CHECK(hipHostMalloc(&P.h_grows,
                               sizeof(gemx_agg_row) * (P.n_gwins ? P.n_gwins : 1)));
       /* split the series dimension so the group stage fills the chip:
@@ -3008,11 +3062,13 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.offset = offset;
     P.valid = true;
   }
+  const int slot = s->q_slot;
+  s->q_slot ^= 1;
   SegQ *d_segq = P.d_segq;
   SeriesQ *d_sq = P.d_sq;
   Partial *d_part = P.d_part;
-  gemx_agg_row *d_rows = P.d_rows;
-  DevErr *d_err = P.d_err;
+  gemx_agg_row *d_rows = P.d_rows2[slot];
+  DevErr *d_err = P.d_err2[slot];
   uint8_t *d_scratch = P.d_scratch;
   uint32_t gen_lanes = P.gen_lanes;
   const uint64_t partial_slots = P.partial_slots;
@@ -3024,11 +3080,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
    * span edges (min/max_time windows) always carry rows */
   HIP_CHECK(hipMemsetAsync(d_err, 0, sizeof(DevErr), s->stream));
 
-  hipEvent_t ev0, ev1, ev2;
-  HIP_CHECK(hipEventCreate(&ev0));
-  HIP_CHECK(hipEventCreate(&ev1));
-  HIP_CHECK(hipEventCreate(&ev2));
-
+  hipEvent_t ev0 = s->ev_q[slot][0], ev1 = s->ev_q[slot][1],
+             ev2 = s->ev_q[slot][2];
   HIP_CHECK(hipEventRecord(ev0, s->stream));
   const int TPB = 256;
   const uint32_t *fast_list = s->d_fast_ids;
@@ -3227,50 +3280,69 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          (GAcc *)P.d_gtmp, P.W0, (uint32_t)P.n_gwins, P.gsplit,
                          P.gper_chunk);
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_FLOAT>), dim3(b2), dim3(256), 0,
-                         s->stream, (const GAcc *)P.d_gtmp, P.gsplit, P.d_grows,
-                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time,
-                         d_err);
+                         s->stream, (const GAcc *)P.d_gtmp, P.gsplit,
+                         P.d_grows2[slot], P.W0, (uint32_t)P.n_gwins, interval,
+                         offset, start_time, d_err);
     } else {
       hipLaunchKernelGGL((k_group_p1<GEMX_TYPE_INT>), dim3(b1), dim3(256), 0,
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part,
                          (GAcc *)P.d_gtmp, P.W0, (uint32_t)P.n_gwins, P.gsplit,
                          P.gper_chunk);
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_INT>), dim3(b2), dim3(256), 0,
-                         s->stream, (const GAcc *)P.d_gtmp, P.gsplit, P.d_grows,
-                         P.W0, (uint32_t)P.n_gwins, interval, offset, start_time,
-                         d_err);
+                         s->stream, (const GAcc *)P.d_gtmp, P.gsplit,
+                         P.d_grows2[slot], P.W0, (uint32_t)P.n_gwins, interval,
+                         offset, start_time, d_err);
     }
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
-  /* check device error + fetch rows straight into the caller's buffer
-   * (gap rows compacted in place afterwards — one copy, no pinned bounce) */
-  DevErr herr = {0};
-  HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
-                           s->stream));
+  /* hand the row fetch to the copy stream so it can overlap the NEXT
+   * query's kernels (double-buffered d_rows): copy waits on ev2, rows
+   * land straight in the caller's (pinned-registered) buffer. */
   uint64_t fetch_rows = tagq ? (uint64_t)s->tag_plan.n_groups * P.n_gwins
                              : (group_all ? P.n_gwins : total_rows);
   if (fetch_rows > cap) {
     seterr("output capacity too small");
     return GEMX_E_CAP;
   }
-  gemx_agg_row *hrows = out_host;
+  HIP_CHECK(hipStreamWaitEvent(s->copy_stream, ev2, 0));
+  HIP_CHECK(hipMemcpyAsync(s->h_err2[slot], d_err, sizeof(DevErr),
+                           hipMemcpyDeviceToHost, s->copy_stream));
   HIP_CHECK(hipMemcpyAsync(
-      hrows,
-      tagq ? s->tag_plan.d_rows : (group_all ? P.d_grows : d_rows),
-      sizeof(gemx_agg_row) * fetch_rows, hipMemcpyDeviceToHost, s->stream));
+      out_host,
+      tagq ? s->tag_plan.d_rows : (group_all ? P.d_grows2[slot] : d_rows),
+      sizeof(gemx_agg_row) * fetch_rows, hipMemcpyDeviceToHost,
+      s->copy_stream));
+  HIP_CHECK(hipEventRecord(s->ev_copy[slot], s->copy_stream));
+
+  if (async_begin) {
+    auto &pe = s->pend[(s->pend_head + s->pend_count) & 1];
+    pe.active = true;
+    pe.slot = slot;
+    pe.fetch_rows = fetch_rows;
+    pe.out = out_host;
+    s->pend_count++;
+    if (n_out) *n_out = 0;
+    return GEMX_OK;
+  }
+  return scan_deliver(s, slot, fetch_rows, out_host, n_out, stats);
+}
+
+/* completion half: wait for the slot's copy, validate, compact, fill
+ * stats from the persistent per-slot events */
+static int scan_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
+                        gemx_agg_row *out_host, uint64_t *n_out,
+                        gemx_query_stats *stats) {
   auto t_sync0 = std::chrono::steady_clock::now();
-  HIP_CHECK(hipStreamSynchronize(s->stream));
+  HIP_CHECK(hipEventSynchronize(s->ev_copy[slot]));
   auto t_sync1 = std::chrono::steady_clock::now();
 
   float ms_scan = 0, ms_merge = 0, ms_total = 0;
-  hipEventElapsedTime(&ms_scan, ev0, ev1);
-  hipEventElapsedTime(&ms_merge, ev1, ev2);
-  hipEventElapsedTime(&ms_total, ev0, ev2);
-  hipEventDestroy(ev0);
-  hipEventDestroy(ev1);
-  hipEventDestroy(ev2);
+  hipEventElapsedTime(&ms_scan, s->ev_q[slot][0], s->ev_q[slot][1]);
+  hipEventElapsedTime(&ms_merge, s->ev_q[slot][1], s->ev_q[slot][2]);
+  hipEventElapsedTime(&ms_total, s->ev_q[slot][0], s->ev_q[slot][2]);
 
+  const DevErr herr = *s->h_err2[slot];
   if (herr.code != 0) {
     seterr(herr.code == GEMX_E_UNSUPPORTED ? "unsupported codec on device"
                                            : "segment decode failed on device");
@@ -3286,11 +3358,12 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     n = fetch_rows;
   } else {
     uint64_t i = 0;
-    while (i < fetch_rows && hrows[i].count >= 0) i++;
+    while (i < fetch_rows && out_host[i].count >= 0) i++;
     n = i;
     for (; i < fetch_rows; i++) {
-      if (hrows[i].count < 0) continue;
-      out_host[n++] = hrows[i];
+      if (out_host[i].count < 0) continue;
+      out_host[n] = out_host[i];
+      n++;
     }
   }
   *n_out = n;
@@ -3574,6 +3647,40 @@ extern "C" int gemx_scan_agg_ex(gemx_shard *s, int64_t start_time,
                                 uint64_t *n_out, gemx_query_stats *stats) {
   return scan_impl(s, start_time, end_time, interval, offset, group_all,
                    filter_op, filter_f, filter_i, out_host, cap, n_out, stats);
+}
+
+/* Async pipeline (the cursor read-ahead model): begin enqueues the whole
+ * query — kernels on the shard's compute stream, the row fetch on its
+ * copy stream gated by an event — and returns immediately; finish waits
+ * for the OLDEST in-flight query and compacts/validates its rows. Up to
+ * two queries may be in flight (double-buffered row buffers), so
+ * begin(i+1) before finish(i) overlaps i's PCIe fetch with i+1's decode.
+ * Rows land in the caller's buffer, which must stay untouched until its
+ * finish. Plain and grouped scans only; same (start,end,interval,offset)
+ * while in flight (the cached-plan shapes the reference cursor pump
+ * also reuses). */
+extern "C" int gemx_scan_agg_begin(gemx_shard *s, int64_t start_time,
+                                   int64_t end_time, int64_t interval,
+                                   int64_t offset, int group_all,
+                                   gemx_agg_row *out_host, uint64_t cap) {
+  uint64_t n = 0;
+  return scan_impl(s, start_time, end_time, interval, offset, group_all, 0,
+                   0, 0, out_host, cap, &n, nullptr, nullptr, nullptr, 1);
+}
+
+extern "C" int gemx_scan_agg_finish(gemx_shard *s, uint64_t *n_out,
+                                    gemx_query_stats *stats) {
+  if (!s || !n_out) return GEMX_E_INVALID;
+  if (s->pend_count == 0) {
+    seterr("no query in flight");
+    return GEMX_E_INVALID;
+  }
+  auto &pe = s->pend[s->pend_head & 1];
+  s->pend_head++;
+  s->pend_count--;
+  pe.active = false;
+  HIP_CHECK(hipSetDevice(s->device));
+  return scan_deliver(s, pe.slot, pe.fetch_rows, pe.out, n_out, stats);
 }
 
 /* hash GROUP BY tag: series_group maps each series (descriptor order) to

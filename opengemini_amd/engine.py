@@ -147,6 +147,15 @@ def _load():
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
         C.POINTER(_Stats),
     ]
+    lib.gemx_scan_agg_begin.restype = C.c_int
+    lib.gemx_scan_agg_begin.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_void_p, C.c_uint64,
+    ]
+    lib.gemx_scan_agg_finish.restype = C.c_int
+    lib.gemx_scan_agg_finish.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     lib.gemx_host_register.restype = C.c_int
     lib.gemx_host_register.argtypes = [C.c_void_p, C.c_uint64]
     lib.gemx_host_unregister.restype = C.c_int
@@ -386,6 +395,39 @@ class Shard:
         )
         return out[: n.value], stats
 
+
+    def scan_agg_begin(self, start_time, end_time, interval, offset=0,
+                       group_all=False, out_cap=None, buf_id=0):
+        """Enqueue a scan without waiting (cursor read-ahead: up to two in
+        flight, double-buffered). Returns the output buffer; rows are
+        valid only after the matching scan_agg_finish. buf_id selects one
+        of two pooled buffers so two in-flight queries don't collide."""
+        lib = self._lib
+        if out_cap is None:
+            out_cap = self._rows_bound(interval, offset, group_all)
+        out = self._pooled_out(f"agg{buf_id}", out_cap, AGG_ROW_DTYPE)
+        rc = lib.gemx_scan_agg_begin(
+            self._h, start_time, end_time, interval, offset,
+            1 if group_all else 0,
+            out.ctypes.data_as(C.c_void_p), out_cap,
+        )
+        _check(rc, lib)
+        return out
+
+    def scan_agg_finish(self, out):
+        """Complete the oldest in-flight scan_agg_begin; `out` is the
+        buffer that begin returned. Returns (rows, stats)."""
+        lib = self._lib
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_scan_agg_finish(self._h, C.byref(n), C.byref(st))
+        _check(rc, lib)
+        stats = dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            host_ms=st.h2d_ms, points=st.points,
+            compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
+        )
+        return out[: n.value], stats
 
     def preagg_build(self):
         """Compute + cache per-series whole-shard pre-agg rows on the handle
