@@ -207,9 +207,17 @@ def main():
                 dist.all_reduce(t, op=dist.ReduceOp.SUM)
         elif args.query == "preagg":
             rows, stats = shard.scan_preagg(-2**62, 2**62)
-        else:  # rate
-            rows, stats = shard.prom_rate(0, (args.pts - 1) * 10**9, RANGE_NS,
-                                          WINDOW_NS)
+        else:  # rate — pipelined like downsample (one query in flight)
+            st9 = step.__dict__
+            if st9.get("rpend") is None:
+                st9["rpend"] = shard.prom_rate_begin(
+                    0, (args.pts - 1) * 10**9, RANGE_NS, WINDOW_NS, buf_id=0)
+                st9["rbuf"] = 1
+            rout = shard.prom_rate_begin(0, (args.pts - 1) * 10**9, RANGE_NS,
+                                         WINDOW_NS, buf_id=st9["rbuf"])
+            st9["rbuf"] = 1 - st9["rbuf"]
+            rows, stats = shard.prom_rate_finish(st9["rpend"])
+            st9["rpend"] = rout
             if dist_on:
                 # cross-shard partial-sum of sum(rate()) per step over RCCL
                 import torch

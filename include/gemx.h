@@ -186,6 +186,16 @@ int gemx_scan_agg_begin(gemx_shard *, int64_t start_time, int64_t end_time,
 int gemx_scan_agg_finish(gemx_shard *, uint64_t *n_out,
                          gemx_query_stats *stats);
 
+/* Async begin/finish for the rate family (rate/irate/over_time) — same
+ * pipeline contract as gemx_scan_agg_begin/finish. func: a GEMX_PF_*_OT
+ * code for the over_time reducers, 0 for rate()/increase()/delta() with
+ * is_rate/is_counter as in gemx_prom_rate, 1 for irate/idelta. */
+int gemx_prom_begin(gemx_shard *, int64_t start_time, int64_t end_time,
+                    int64_t range_ns, int64_t step_ns, int is_rate,
+                    int is_counter, int func, gemx_rate_row *out_host,
+                    uint64_t cap);
+int gemx_prom_finish(gemx_shard *, uint64_t *n_out, gemx_query_stats *stats);
+
 /* Pin a caller-owned output buffer (hipHostRegister) so row fetches into
  * it run at pinned-DMA speed; optional — unregistered buffers work too.
  * Mirrors why the reference pools records (aggregate_cursor.go:100). */

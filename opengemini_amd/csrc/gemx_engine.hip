@@ -1900,6 +1900,10 @@ __global__ void __launch_bounds__(256) k_tag_p2(
   }
 }
 
+static int rate_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
+                        gemx_rate_row *out_host, uint64_t *n_out,
+                        gemx_query_stats *stats);
+
 /* ================= PromQL rate over range vectors (config #5) =============
  * Restates RangeVectorCursor + rate_prom (see oracle/agg.c orc_prom_rate for
  * the line-cited CPU spec): sample steps ts = startSample + k*step, window
@@ -2543,7 +2547,8 @@ struct RatePlan {
   RateSegQ *d_rsegq = nullptr;
   RateSeriesQ *d_rsq = nullptr;
   RatePartial *d_rpart = nullptr;
-  gemx_rate_row *d_rrows = nullptr;
+  gemx_rate_row *d_rrows2[2] = {nullptr, nullptr}; /* double-buffered */
+  DevErr *d_err2[2] = {nullptr, nullptr};
   gemx_rate_row *h_rrows = nullptr;
   uint8_t *d_scratch = nullptr;
   uint32_t gen_lanes = 0;
@@ -2553,7 +2558,10 @@ static void free_rate_plan(RatePlan &p) {
   if (p.d_rsegq) hipFree(p.d_rsegq);
   if (p.d_rsq) hipFree(p.d_rsq);
   if (p.d_rpart) hipFree(p.d_rpart);
-  if (p.d_rrows) hipFree(p.d_rrows);
+  for (int i = 0; i < 2; i++) {
+    if (p.d_rrows2[i]) hipFree(p.d_rrows2[i]);
+    if (p.d_err2[i]) hipFree(p.d_err2[i]);
+  }
   if (p.h_rrows) hipHostFree(p.h_rrows);
   if (p.d_scratch) hipFree(p.d_scratch);
   p = RatePlan();
@@ -2598,6 +2606,17 @@ struct gemx_shard {
     gemx_agg_row *out = nullptr;
   } pend[2];
   int pend_head = 0, pend_count = 0, q_slot = 0;
+  /* rate family pipeline (separate slots/events from the agg path) */
+  hipEvent_t ev_r[2][3];
+  hipEvent_t ev_rcopy[2];
+  DevErr *h_rerr2[2] = {nullptr, nullptr};
+  struct PendingRate {
+    bool active = false;
+    int slot = 0;
+    uint64_t fetch_rows = 0;
+    gemx_rate_row *out = nullptr;
+  } rpend[2];
+  int rpend_head = 0, rpend_count = 0, r_slot = 0;
   QueryPlan plan;
   RatePlan rate_plan;
   /* pre-aggregation metadata (pre_aggregation.go FloatPreAgg role): one
@@ -2792,9 +2811,14 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
   HIP_CHECK(hipStreamCreate(&s->stream));
   HIP_CHECK(hipStreamCreate(&s->copy_stream));
   for (int sl = 0; sl < 2; sl++) {
-    for (int e = 0; e < 3; e++) HIP_CHECK(hipEventCreate(&s->ev_q[sl][e]));
+    for (int e = 0; e < 3; e++) {
+      HIP_CHECK(hipEventCreate(&s->ev_q[sl][e]));
+      HIP_CHECK(hipEventCreate(&s->ev_r[sl][e]));
+    }
     HIP_CHECK(hipEventCreate(&s->ev_copy[sl]));
+    HIP_CHECK(hipEventCreate(&s->ev_rcopy[sl]));
     HIP_CHECK(hipHostMalloc(&s->h_err2[sl], sizeof(DevErr)));
+    HIP_CHECK(hipHostMalloc(&s->h_rerr2[sl], sizeof(DevErr)));
   }
   HIP_CHECK(hipMalloc(&s->d_blob, blob_bytes ? blob_bytes : 1));
   HIP_CHECK(hipMemcpyAsync(s->d_blob, blob, blob_bytes, hipMemcpyHostToDevice,
@@ -2850,9 +2874,14 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   hipStreamDestroy(s->stream);
   if (s->copy_stream) hipStreamDestroy(s->copy_stream);
   for (int sl = 0; sl < 2; sl++) {
-    for (int e = 0; e < 3; e++) hipEventDestroy(s->ev_q[sl][e]);
+    for (int e = 0; e < 3; e++) {
+      hipEventDestroy(s->ev_q[sl][e]);
+      hipEventDestroy(s->ev_r[sl][e]);
+    }
     hipEventDestroy(s->ev_copy[sl]);
+    hipEventDestroy(s->ev_rcopy[sl]);
     if (s->h_err2[sl]) hipHostFree(s->h_err2[sl]);
+    if (s->h_rerr2[sl]) hipHostFree(s->h_rerr2[sl]);
   }
   delete s;
   return GEMX_OK;
@@ -3408,8 +3437,16 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                           int64_t range_ns, int64_t step_ns, int is_rate,
                           int is_counter, int func, gemx_rate_row *out_host,
                           uint64_t cap, uint64_t *n_out,
-                          gemx_query_stats *stats) {
+                          gemx_query_stats *stats, int async_begin = 0) {
   if (!s) return GEMX_E_INVALID;
+  if (async_begin && s->rpend_count >= 2) {
+    seterr("two rate queries already in flight: call gemx_prom_finish");
+    return GEMX_E_INVALID;
+  }
+  if (!async_begin && s->rpend_count > 0) {
+    seterr("async rate queries in flight: call gemx_prom_finish first");
+    return GEMX_E_INVALID;
+  }
   if (s->col_type != GEMX_TYPE_FLOAT) {
     seterr("prom rate needs a float column");
     return GEMX_E_INVALID;
@@ -3441,6 +3478,10 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   RatePlan &P = s->rate_plan;
   if (!P.valid || P.start != start_time || P.end != end_time ||
       P.range_ns != range_ns || P.step_ns != step_ns) {
+    if (s->rpend_count > 0) {
+      seterr("cannot rebuild the rate plan with queries in flight");
+      return GEMX_E_INVALID;
+    }
     free_rate_plan(P);
     P.rsegq.resize(nsegs);
     P.rsq.resize(s->series_ranges.size());
@@ -3504,8 +3545,11 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                              hipMemcpyHostToDevice, s->stream));
     HIP_CHECK(hipMalloc(&P.d_rpart,
                         sizeof(RatePartial) * (P.partial_slots ? P.partial_slots : 1)));
-    HIP_CHECK(hipMalloc(&P.d_rrows,
-                        sizeof(gemx_rate_row) * (P.total_rows ? P.total_rows : 1)));
+    for (int sl = 0; sl < 2; sl++) {
+      HIP_CHECK(hipMalloc(&P.d_rrows2[sl],
+                          sizeof(gemx_rate_row) * (P.total_rows ? P.total_rows : 1)));
+      HIP_CHECK(hipMalloc(&P.d_err2[sl], sizeof(DevErr)));
+    }
     HIP_CHECK(hipHostMalloc(&P.h_rrows,
                             sizeof(gemx_rate_row) * (P.total_rows ? P.total_rows : 1)));
     if (!s->general_ids.empty()) {
@@ -3519,14 +3563,13 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.valid = true;
   }
 
-  DevErr *d_err = nullptr;
-  HIP_CHECK(hipMalloc(&d_err, sizeof(DevErr)));
+  const int slot = s->r_slot;
+  s->r_slot ^= 1;
+  DevErr *d_err = P.d_err2[slot];
   HIP_CHECK(hipMemsetAsync(d_err, 0, sizeof(DevErr), s->stream));
 
-  hipEvent_t ev0, ev1, ev2;
-  HIP_CHECK(hipEventCreate(&ev0));
-  HIP_CHECK(hipEventCreate(&ev1));
-  HIP_CHECK(hipEventCreate(&ev2));
+  hipEvent_t ev0 = s->ev_r[slot][0], ev1 = s->ev_r[slot][1],
+             ev2 = s->ev_r[slot][2];
   HIP_CHECK(hipEventRecord(ev0, s->stream));
   const int TPB = 256;
   if (!s->fast_ids.empty()) {
@@ -3574,31 +3617,45 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
         (uint32_t)std::min<uint64_t>((P.total_rows + TPB - 1) / TPB, 65535);
     hipLaunchKernelGGL(k_rate_merge, dim3(blocks), dim3(TPB), 0, s->stream,
                        P.d_rsq, (uint32_t)P.rsq.size(), P.d_rsegq, P.d_rpart,
-                       P.d_rrows, P.total_rows, start_sample, eff_step, range_ns,
-                       is_rate, is_counter, func, d_err);
+                       P.d_rrows2[slot], P.total_rows, start_sample, eff_step,
+                       range_ns, is_rate, is_counter, func, d_err);
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
-  DevErr herr = {0};
-  HIP_CHECK(hipMemcpyAsync(&herr, d_err, sizeof(DevErr), hipMemcpyDeviceToHost,
-                           s->stream));
   if (P.total_rows > cap) {
     seterr("output capacity too small");
     return GEMX_E_CAP;
   }
-  HIP_CHECK(hipMemcpyAsync(out_host, P.d_rrows,
+  HIP_CHECK(hipStreamWaitEvent(s->copy_stream, ev2, 0));
+  HIP_CHECK(hipMemcpyAsync(s->h_rerr2[slot], d_err, sizeof(DevErr),
+                           hipMemcpyDeviceToHost, s->copy_stream));
+  HIP_CHECK(hipMemcpyAsync(out_host, P.d_rrows2[slot],
                            sizeof(gemx_rate_row) * P.total_rows,
-                           hipMemcpyDeviceToHost, s->stream));
-  HIP_CHECK(hipStreamSynchronize(s->stream));
+                           hipMemcpyDeviceToHost, s->copy_stream));
+  HIP_CHECK(hipEventRecord(s->ev_rcopy[slot], s->copy_stream));
 
+  if (async_begin) {
+    auto &pe = s->rpend[(s->rpend_head + s->rpend_count) & 1];
+    pe.active = true;
+    pe.slot = slot;
+    pe.fetch_rows = P.total_rows;
+    pe.out = out_host;
+    s->rpend_count++;
+    if (n_out) *n_out = 0;
+    return GEMX_OK;
+  }
+  return rate_deliver(s, slot, P.total_rows, out_host, n_out, stats);
+}
+
+static int rate_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
+                        gemx_rate_row *out_host, uint64_t *n_out,
+                        gemx_query_stats *stats) {
+  HIP_CHECK(hipEventSynchronize(s->ev_rcopy[slot]));
   float ms_scan = 0, ms_merge = 0, ms_total = 0;
-  hipEventElapsedTime(&ms_scan, ev0, ev1);
-  hipEventElapsedTime(&ms_merge, ev1, ev2);
-  hipEventElapsedTime(&ms_total, ev0, ev2);
-  hipEventDestroy(ev0);
-  hipEventDestroy(ev1);
-  hipEventDestroy(ev2);
-  hipFree(d_err);
+  hipEventElapsedTime(&ms_scan, s->ev_r[slot][0], s->ev_r[slot][1]);
+  hipEventElapsedTime(&ms_merge, s->ev_r[slot][1], s->ev_r[slot][2]);
+  hipEventElapsedTime(&ms_total, s->ev_r[slot][0], s->ev_r[slot][2]);
+  const DevErr herr = *s->h_rerr2[slot];
   if (herr.code != 0) {
     seterr(herr.code == GEMX_E_UNSUPPORTED ? "unsupported codec on device"
                                            : "segment decode failed on device");
@@ -3606,15 +3663,16 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   }
 
   /* in-place compact: only non-nil rows leave (the reference appends only
-   * non-nil, prom reducer append path) */
+   * non-nil, prom reducer append path); skipped when the merge counted
+   * zero nil rows */
   uint64_t n = 0;
   if (herr.gaps == 0) {
-    n = P.total_rows;
+    n = fetch_rows;
   } else {
     uint64_t i = 0;
-    while (i < P.total_rows && !out_host[i].isnil) i++;
+    while (i < fetch_rows && !out_host[i].isnil) i++;
     n = i;
-    for (; i < P.total_rows; i++) {
+    for (; i < fetch_rows; i++) {
       if (out_host[i].isnil) continue;
       out_host[n] = out_host[i];
       n++;
@@ -3817,6 +3875,39 @@ extern "C" int gemx_scan_preagg(gemx_shard *s, int64_t start_time,
   if (n_meta_out) *n_meta_out = n_cov;
   if (stats) stats->n_rows = n;
   return GEMX_OK;
+}
+
+/* Async begin/finish for the whole rate family (rate/irate/over_time;
+ * same pipeline contract as gemx_scan_agg_begin/finish). func:
+ * GEMX_PF_* over_time codes, or for rate/irate pass func=0/1 with
+ * is_rate/is_counter as in the sync entry points. */
+extern "C" int gemx_prom_begin(gemx_shard *s, int64_t start_time,
+                               int64_t end_time, int64_t range_ns,
+                               int64_t step_ns, int is_rate, int is_counter,
+                               int func, gemx_rate_row *out_host,
+                               uint64_t cap) {
+  if (s && end_time < start_time + range_ns) {
+    seterr("empty sample grid: nothing to enqueue");
+    return GEMX_E_INVALID;
+  }
+  uint64_t n = 0;
+  return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, is_rate,
+                        is_counter, func, out_host, cap, &n, nullptr, 1);
+}
+
+extern "C" int gemx_prom_finish(gemx_shard *s, uint64_t *n_out,
+                                gemx_query_stats *stats) {
+  if (!s || !n_out) return GEMX_E_INVALID;
+  if (s->rpend_count == 0) {
+    seterr("no rate query in flight");
+    return GEMX_E_INVALID;
+  }
+  auto &pe = s->rpend[s->rpend_head & 1];
+  s->rpend_head++;
+  s->rpend_count--;
+  pe.active = false;
+  HIP_CHECK(hipSetDevice(s->device));
+  return rate_deliver(s, pe.slot, pe.fetch_rows, pe.out, n_out, stats);
 }
 
 extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,

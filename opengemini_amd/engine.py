@@ -156,6 +156,15 @@ def _load():
     lib.gemx_scan_agg_finish.argtypes = [
         C.c_void_p, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_prom_begin.restype = C.c_int
+    lib.gemx_prom_begin.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_int, C.c_int, C.c_void_p, C.c_uint64,
+    ]
+    lib.gemx_prom_finish.restype = C.c_int
+    lib.gemx_prom_finish.argtypes = [
+        C.c_void_p, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     lib.gemx_host_register.restype = C.c_int
     lib.gemx_host_register.argtypes = [C.c_void_p, C.c_uint64]
     lib.gemx_host_unregister.restype = C.c_int
@@ -428,6 +437,42 @@ class Shard:
             compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
+
+    def _rate_cap(self, start_time, end_time, range_ns, step_ns):
+        nsteps = 1
+        if step_ns > 0 and end_time >= start_time + range_ns:
+            nsteps = int((end_time - (start_time + range_ns)) // step_ns) + 2
+        return nsteps * self._sid_count() + 16
+
+    def prom_rate_begin(self, start_time, end_time, range_ns, step_ns,
+                        is_rate=True, is_counter=True, func=0, out_cap=None,
+                        buf_id=0):
+        """Enqueue a rate-family query without waiting (pipeline contract
+        as scan_agg_begin). func: 0 rate/increase/delta, 1 irate/idelta,
+        or an over_time code from OT_FUNCS values."""
+        lib = self._lib
+        if out_cap is None:
+            out_cap = self._rate_cap(start_time, end_time, range_ns, step_ns)
+        out = self._pooled_out(f"rate{buf_id}", out_cap, RATE_ROW_DTYPE)
+        rc = lib.gemx_prom_begin(
+            self._h, start_time, end_time, range_ns, step_ns,
+            1 if is_rate else 0, 1 if is_counter else 0, func,
+            out.ctypes.data_as(C.c_void_p), out_cap,
+        )
+        _check(rc, lib)
+        return out
+
+    def prom_rate_finish(self, out):
+        lib = self._lib
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_prom_finish(self._h, C.byref(n), C.byref(st))
+        _check(rc, lib)
+        return out[: n.value], dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            points=st.points, compressed_bytes=st.compressed_bytes,
+            n_rows=st.n_rows,
+        )
 
     def preagg_build(self):
         """Compute + cache per-series whole-shard pre-agg rows on the handle

@@ -949,3 +949,28 @@ class TestAsyncPipeline:
             assert len(rows) > 0
         finally:
             sh.close()
+
+    def test_rate_pipelined_equals_sync(self):
+        blob, descs = orc.gen_shard(1504, 300, 1000)
+        import opengemini_amd as gx
+
+        sh = gx.Shard(blob, descs, F)
+        S = 10**9
+        try:
+            ref, _ = sh.prom_rate(0, 999 * S, 300 * S, 60 * S)
+            ref = ref.copy()
+            b0 = sh.prom_rate_begin(0, 999 * S, 300 * S, 60 * S, buf_id=0)
+            b1 = sh.prom_rate_begin(0, 999 * S, 300 * S, 60 * S, buf_id=1)
+            r0, st = sh.prom_rate_finish(b0)
+            r1, _ = sh.prom_rate_finish(b1)
+            for r in (r0, r1):
+                assert len(r) == len(ref)
+                assert np.array_equal(r["sid"], ref["sid"])
+                assert np.array_equal(r["ts"], ref["ts"])
+                assert np.array_equal(r["value"].view(np.uint64),
+                                      ref["value"].view(np.uint64))
+            assert st["decode_ms"] > 0
+            with pytest.raises(gx.GemxError):
+                sh.prom_rate_finish(b0)  # none left in flight
+        finally:
+            sh.close()
