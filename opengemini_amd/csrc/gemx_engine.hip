@@ -923,22 +923,26 @@ __global__ void __launch_bounds__(256) k_scan_fast(
             lvv.i = iv;
           }
         }
-        Partial *p = base + (ord - sq.w_first);
-        p->v[0].i = gend - i;
-        if (COLTYPE == GEMX_TYPE_FLOAT) p->v[1].f = sf; else p->v[1].i = si2;
-        p->v[2] = mn;
-        p->v[3] = mx;
-        p->v[4] = fvv;
-        p->v[5] = lvv;
-        p->t[0] = t_i;
-        p->t[1] = t_i; /* no nils ⇒ valueIndex == row index */
-        p->t[2] = t0c + (int64_t)min_row * dtc;
-        p->t[3] = t0c + (int64_t)max_row * dtc;
-        p->t[4] = t_i;
-        p->t[5] = t0c + (int64_t)(gend - 1) * dtc;
-        p->first_row_time = t_i;
-        p->nilmask = 0;
-        p->has_rows = 1;
+        /* assemble in registers, store as one struct: the compiler emits
+         * wide (dwordx4) stores instead of 17 scalar ones — the flush is
+         * the hot write path at one per (segment, window) */
+        Partial tmp;
+        tmp.v[0].i = gend - i;
+        if (COLTYPE == GEMX_TYPE_FLOAT) tmp.v[1].f = sf; else tmp.v[1].i = si2;
+        tmp.v[2] = mn;
+        tmp.v[3] = mx;
+        tmp.v[4] = fvv;
+        tmp.v[5] = lvv;
+        tmp.t[0] = t_i;
+        tmp.t[1] = t_i; /* no nils ⇒ valueIndex == row index */
+        tmp.t[2] = t0c + (int64_t)min_row * dtc;
+        tmp.t[3] = t0c + (int64_t)max_row * dtc;
+        tmp.t[4] = t_i;
+        tmp.t[5] = t0c + (int64_t)(gend - 1) * dtc;
+        tmp.first_row_time = t_i;
+        tmp.nilmask = 0;
+        tmp.has_rows = 1;
+        base[ord - sq.w_first] = tmp;
         i = gend;
       }
       continue;
