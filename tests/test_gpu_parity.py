@@ -974,3 +974,32 @@ class TestAsyncPipeline:
                 sh.prom_rate_finish(b0)  # none left in flight
         finally:
             sh.close()
+
+
+class TestSingleBigSeries:
+    """config #1 shape: one series spanning many segments (10M pts in the
+    reference's config; 200k here for test speed). Exercises the
+    single-series group stage and the many-segments-per-series merge."""
+
+    def test_one_series_many_segments(self):
+        blob, descs = orc.gen_shard(1601, 1, 200_000)
+        assert len(descs) == 200  # 1000-row segments, one sid
+        sh = gpu_shard(blob, descs, F)
+        try:
+            per, _ = sh.scan_agg(0, 2**62, INT)
+            per = per.copy()
+            grp, _ = sh.scan_agg(0, 2**62, INT, group_all=True)
+            grp = grp.copy()
+        finally:
+            sh.close()
+        ref = orc.scan_agg(blob, descs, F, 0, 2**62, INT)
+        assert_parity(per, ref, F)
+        # single series: grouped == per-series except sid=0 and the
+        # interval time-column fixups (count_time/sum_time = win_start)
+        assert len(grp) == len(per)
+        assert np.array_equal(grp["count"], per["count"])
+        for f in ("min", "max", "first", "last"):
+            assert np.array_equal(grp[f].view(np.uint64),
+                                  per[f].view(np.uint64)), f
+        assert np.array_equal(grp["min_time"], per["min_time"])
+        assert int(per["count"].sum()) == 200_000
