@@ -196,6 +196,19 @@ int gemx_prom_begin(gemx_shard *, int64_t start_time, int64_t end_time,
                     uint64_t cap);
 int gemx_prom_finish(gemx_shard *, uint64_t *n_out, gemx_query_stats *stats);
 
+/* Series-subset scan — the tag-predicate seam for the column-store path
+ * (config #3): the executor evaluates tag conditions against its index
+ * (tsi scan / lib/binaryfilterfunc on tag columns) and passes the
+ * qualifying series as a byte mask (one per series, descriptor order,
+ * 1 = include). Excluded series decode nothing; the optional value
+ * predicate composes; group_all=1 merges the included series only. */
+int gemx_scan_agg_series(gemx_shard *, const uint8_t *series_mask,
+                         int64_t start_time, int64_t end_time,
+                         int64_t interval, int64_t offset, int group_all,
+                         int filter_op, double filter_f, int64_t filter_i,
+                         gemx_agg_row *out_host, uint64_t cap,
+                         uint64_t *n_out, gemx_query_stats *stats);
+
 /* Pin a caller-owned output buffer (hipHostRegister) so row fetches into
  * it run at pinned-DMA speed; optional — unregistered buffers work too.
  * Mirrors why the reference pools records (aggregate_cursor.go:100). */

@@ -2486,6 +2486,7 @@ static inline uint32_t h_u32be(const uint8_t *p) {
 struct QueryPlan {
   bool valid = false;
   int64_t start = 0, end = 0, interval = 0, offset = 0;
+  uint64_t skip_hash = 0; /* FNV of the series-exclusion set (0 = none) */
   std::vector<SegQ> segq;
   std::vector<SeriesQ> sq;
   uint64_t partial_slots = 0, total_rows = 0;
@@ -2865,6 +2866,11 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
 extern "C" int gemx_shard_close(gemx_shard *s) {
   if (!s) return GEMX_OK;
   hipSetDevice(s->device);
+  /* drain in-flight async queries so frees don't race the copy stream */
+  hipStreamSynchronize(s->stream);
+  if (s->copy_stream) hipStreamSynchronize(s->copy_stream);
+  s->pend_count = 0;
+  s->rpend_count = 0;
   free_plan(s->plan);
   free_plan(s->sub_plan);
   free_tag_plan(s->tag_plan);
@@ -2930,8 +2936,18 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960 + 512; /* +512: clip bitmap */
 
   QueryPlan &P = skip_series ? s->sub_plan : s->plan;
+  uint64_t skip_hash = 0;
+  if (skip_series) {
+    skip_hash = 1469598103934665603ull; /* FNV-1a over the mask bytes */
+    for (size_t g = 0; g < s->series_ranges.size(); g++) {
+      skip_hash ^= (uint8_t)skip_series[g];
+      skip_hash *= 1099511628211ull;
+    }
+    if (!skip_hash) skip_hash = 1;
+  }
   if (!P.valid || P.start != start_time || P.end != end_time ||
-      P.interval != interval || P.offset != offset) {
+      P.interval != interval || P.offset != offset ||
+      P.skip_hash != skip_hash) {
     if (s->pend_count > 0) {
       seterr("cannot rebuild the query plan with queries in flight");
       return GEMX_E_INVALID;
@@ -3093,6 +3109,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.end = end_time;
     P.interval = interval;
     P.offset = offset;
+    P.skip_hash = skip_hash;
     P.valid = true;
   }
   const int slot = s->q_slot;
@@ -3709,6 +3726,32 @@ extern "C" int gemx_scan_agg_ex(gemx_shard *s, int64_t start_time,
                                 uint64_t *n_out, gemx_query_stats *stats) {
   return scan_impl(s, start_time, end_time, interval, offset, group_all,
                    filter_op, filter_f, filter_i, out_host, cap, n_out, stats);
+}
+
+/* Series-subset scan — the tag-predicate seam for the column-store path
+ * (config #3): the executor evaluates the tag condition against its
+ * index (lib/binaryfilterfunc on tag columns / tsi index scan) and
+ * passes the qualifying series as a mask (one byte per series in
+ * descriptor order, 1 = include). Excluded series decode nothing and
+ * emit nothing; an optional value predicate composes (filter_op as in
+ * gemx_scan_agg_ex). group_all merges the included series only. */
+extern "C" int gemx_scan_agg_series(gemx_shard *s, const uint8_t *series_mask,
+                                    int64_t start_time, int64_t end_time,
+                                    int64_t interval, int64_t offset,
+                                    int group_all, int filter_op,
+                                    double filter_f, int64_t filter_i,
+                                    gemx_agg_row *out_host, uint64_t cap,
+                                    uint64_t *n_out, gemx_query_stats *stats) {
+  if (!s || !series_mask) {
+    seterr("scan_agg_series: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  const size_t nser = s->series_ranges.size();
+  std::vector<char> skip(nser);
+  for (size_t g = 0; g < nser; g++) skip[g] = series_mask[g] ? 0 : 1;
+  return scan_impl(s, start_time, end_time, interval, offset, group_all,
+                   filter_op, filter_f, filter_i, out_host, cap, n_out, stats,
+                   skip.data());
 }
 
 /* Async pipeline (the cursor read-ahead model): begin enqueues the whole

@@ -156,6 +156,12 @@ def _load():
     lib.gemx_scan_agg_finish.argtypes = [
         C.c_void_p, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_scan_agg_series.restype = C.c_int
+    lib.gemx_scan_agg_series.argtypes = [
+        C.c_void_p, C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64,
+        C.c_int, C.c_int, C.c_double, C.c_int64,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     lib.gemx_prom_begin.restype = C.c_int
     lib.gemx_prom_begin.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
@@ -509,6 +515,41 @@ class Shard:
             host_ms=st.h2d_ms, points=st.points,
             compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
             meta_rows=int(nm.value),
+        )
+        return out[: n.value], stats
+
+    def scan_agg_series(self, series_mask, start_time, end_time, interval,
+                        offset=0, group_all=False, filter=None, out_cap=None):
+        """Series-subset scan (tag-predicate seam, config #3): series_mask
+        has one truthy entry per included series (descriptor order) — the
+        executor's index/tag filter decides membership. Optional value
+        predicate composes like scan_agg(filter=...)."""
+        lib = self._lib
+        series_mask = np.ascontiguousarray(series_mask, dtype=np.uint8)
+        if len(series_mask) != self._sid_count():
+            raise GemxError("series_mask must have one entry per series")
+        if out_cap is None:
+            out_cap = self._rows_bound(interval, offset, group_all)
+        out = self._pooled_out("agg", out_cap, AGG_ROW_DTYPE)
+        fop, ff, fi = 0, 0.0, 0
+        if filter is not None:
+            op_name, operand = filter
+            fop = self.FILTER_OPS[op_name]
+            ff = float(operand) if self.col_type == GEMX_TYPE_FLOAT else 0.0
+            fi = int(operand) if self.col_type == GEMX_TYPE_INT else 0
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_scan_agg_series(
+            self._h, series_mask.ctypes.data_as(C.c_void_p),
+            start_time, end_time, interval, offset, 1 if group_all else 0,
+            fop, ff, fi,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        stats = dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            host_ms=st.h2d_ms, points=st.points,
+            compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
 

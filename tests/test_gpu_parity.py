@@ -1003,3 +1003,62 @@ class TestSingleBigSeries:
                                   per[f].view(np.uint64)), f
         assert np.array_equal(grp["min_time"], per["min_time"])
         assert int(per["count"].sum()) == 200_000
+
+
+class TestSeriesMask:
+    """gemx_scan_agg_series: tag-predicate series selection. Oracle = scan
+    of the same blob restricted to the selected series' rows."""
+
+    def _series_order(self, descs):
+        sids = descs["sid"]
+        keep = np.ones(len(sids), dtype=bool)
+        keep[1:] = sids[1:] != sids[:-1]
+        return sids[keep]
+
+    def test_mask_parity(self):
+        blob, descs = orc.gen_shard(1701, 300, 1000)
+        sh = gpu_shard(blob, descs, F)
+        order = self._series_order(descs)
+        rng = np.random.default_rng(1701)
+        mask = (rng.random(len(order)) < 0.4).astype(np.uint8)
+        sel = set(order[mask == 1].tolist())
+        try:
+            rows, _ = sh.scan_agg_series(mask, 0, 2**62, INT)
+            rows = rows.copy()
+            # grouped over the masked subset
+            grows, _ = sh.scan_agg_series(mask, 0, 2**62, INT, group_all=True)
+            grows = grows.copy()
+            # different mask over the same range: plan must re-key
+            mask2 = 1 - mask
+            rows2, _ = sh.scan_agg_series(mask2, 0, 2**62, INT)
+            rows2 = rows2.copy()
+        finally:
+            sh.close()
+        base = orc.scan_agg(blob, descs, F, 0, 2**62, INT)
+        ref = base[np.isin(base["sid"], list(sel))]
+        assert_parity(rows, ref, F)
+        ref2 = base[~np.isin(base["sid"], list(sel))]
+        assert_parity(rows2, ref2, F)
+        gref = orc.group_merge(ref, F, INT)
+        assert len(grows) == len(gref)
+        for f in ("win_start", "count", "min_time", "max_time"):
+            assert np.array_equal(grows[f], gref[f]), f
+        assert np.allclose(grows["sum"], gref["sum"], rtol=1e-9)
+
+    def test_mask_with_value_filter(self):
+        blob, descs = orc.gen_shard(1702, 200, 1000)
+        sh = gpu_shard(blob, descs, F)
+        order = self._series_order(descs)
+        mask = np.zeros(len(order), dtype=np.uint8)
+        mask[::3] = 1
+        sel = order[mask == 1]
+        try:
+            rows, _ = sh.scan_agg_series(mask, 0, 2**62, INT,
+                                         filter=("gt", 0.0))
+            rows = rows.copy()
+        finally:
+            sh.close()
+        ref = orc.scan_agg_filtered(blob, descs, F, 0, 2**62, INT,
+                                    "gt", 0.0)
+        ref = ref[np.isin(ref["sid"], sel)]
+        assert_parity(rows, ref, F)
