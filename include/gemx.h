@@ -196,6 +196,22 @@ int gemx_prom_begin(gemx_shard *, int64_t start_time, int64_t end_time,
                     uint64_t cap);
 int gemx_prom_finish(gemx_shard *, uint64_t *n_out, gemx_query_stats *stats);
 
+/* Cross-field predicate scan (config #3: SELECT agg(value) WHERE
+ * other_field <op> x): filter_shard holds the predicate column of the
+ * same measurement — same device, same (sid, rows) segment sequence.
+ * The predicate evaluates on device over the filter column
+ * (lib/binaryfilterfunc compare kernels; nil rows fail) into a row
+ * bitmap cached on value_shard, which the scan applies before
+ * aggregation (FilterByField, immutable/location.go:309). filter_op /
+ * filter_f / filter_i as in gemx_scan_agg_ex, typed by the FILTER
+ * shard's column type. */
+int gemx_scan_agg_xfield(gemx_shard *value_shard, gemx_shard *filter_shard,
+                         int filter_op, double filter_f, int64_t filter_i,
+                         int64_t start_time, int64_t end_time,
+                         int64_t interval, int64_t offset, int group_all,
+                         gemx_agg_row *out_host, uint64_t cap,
+                         uint64_t *n_out, gemx_query_stats *stats);
+
 /* Series-subset scan — the tag-predicate seam for the column-store path
  * (config #3): the executor evaluates tag conditions against its index
  * (tsi scan / lib/binaryfilterfunc on tag columns) and passes the

@@ -1060,6 +1060,111 @@ __global__ void __launch_bounds__(256) k_scan_fast(
 #undef GEMX_DECODE_ONE
 }
 
+/* cross-field predicate evaluation (config #3: binaryfilterfunc compare
+ * kernels, lib/binaryfilterfunc/eval_generator.gen.go:31+, applied as
+ * FilterByField over a condition on a DIFFERENT field): one lane per
+ * segment of the FILTER shard decodes its values and writes one pass/
+ * fail bit per row (nil rows fail, as the reference's condition
+ * evaluation does). The bitmap then drives the value shard's general
+ * scan. */
+__device__ int64_t d_snappy_decode(const uint8_t *src, int64_t len,
+                                   uint8_t *dst, int64_t cap);
+
+template <int FCOLTYPE>
+__global__ void __launch_bounds__(256) k_eval_filter(
+    const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
+    uint32_t nsegs, const uint64_t *__restrict__ row_base, int filter_op,
+    double filter_f, int64_t filter_i, uint8_t *__restrict__ out_bm,
+    uint8_t *__restrict__ scratch, uint64_t scratch_per_lane, uint32_t nlanes,
+    DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  if (gid >= nlanes) return;
+  uint8_t *my = scratch + (uint64_t)gid * scratch_per_lane;
+  int64_t *vbuf = (int64_t *)my;       /* dense values: 4096 × 8 */
+  uint8_t *sbuf = my + 4096 * 8;       /* snappy scratch 40KB */
+
+  for (uint32_t si = gid; si < nsegs; si += nlanes) {
+    const gemx_seg_desc d = descs[si];
+    int rows = (int)d.rows;
+    if (rows > 4096) { set_err(err, GEMX_E_INVALID); return; }
+    SegHeader h;
+    if (parse_data_header(blob + d.data_offset, d.data_size, FCOLTYPE, &h)) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    int nilcount;
+    int dense;
+    if (h.one_value) {
+      nilcount = h.nilcount;
+      dense = 1 - nilcount;
+      if (dense) memcpy(&vbuf[0], h.enc, 8);
+    } else if (h.enc_len == 0) {
+      nilcount = h.nilcount;
+      dense = 0;
+    } else if (FCOLTYPE == GEMX_TYPE_FLOAT) {
+      int tag = h.enc[0] >> 4;
+      if (tag == 2) {
+        int64_t dl = d_snappy_decode(h.enc + 1, h.enc_len - 1, sbuf, 4096 * 8);
+        if (dl < 0 || dl % 8) { set_err(err, GEMX_E_DECODE); return; }
+        dense = (int)(dl / 8);
+        for (int i = 0; i < dense; i++) vbuf[i] = (int64_t)d_u64le(sbuf + i * 8);
+      } else {
+        FloatIter fit;
+        int rc = fit.init(h.enc, h.enc_len);
+        if (rc) { set_err(err, rc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+        dense = 0;
+        double x;
+        while (dense < 4096 && fit.next(&x) == 0) {
+          memcpy(&vbuf[dense], &x, 8);
+          dense++;
+        }
+      }
+      nilcount = h.bitmap ? h.nilcount : 0;
+    } else {
+      IntIter iit;
+      int rc = iit.init(h.enc, h.enc_len);
+      if (rc) { set_err(err, rc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
+      dense = 0;
+      int64_t x;
+      while (dense < 4096 && iit.next(&x) == 0) {
+        vbuf[dense] = x;
+        dense++;
+      }
+      nilcount = h.bitmap ? h.nilcount : 0;
+    }
+    if (h.bitmap && dense + nilcount != rows) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    if (!h.bitmap && nilcount == rows && rows > 0) {
+      h.bitmap = d_zero_bm;
+      h.bm_off = 0;
+    }
+    const uint64_t rb = row_base[si];
+    int vi = 0;
+    for (int r = 0; r < rows; r++) {
+      int valid = 1;
+      if (h.bitmap) valid = bm_valid(&h, r);
+      else if (nilcount == rows && rows > 0) valid = 0;
+      int pass = 0;
+      if (valid) {
+        double xf = 0;
+        int64_t xi = 0;
+        if (FCOLTYPE == GEMX_TYPE_FLOAT) memcpy(&xf, &vbuf[vi], 8);
+        else xi = vbuf[vi];
+        vi++;
+        pass = d_filt_pass(FCOLTYPE, filter_op, filter_f, filter_i, xf, xi);
+      }
+      const uint64_t bit = rb + (uint64_t)r;
+      /* lanes own whole segments; segments need not start byte-aligned in
+       * the bitmap, so neighbouring lanes can share an edge byte */
+      if (pass)
+        atomicOr((unsigned int *)(out_bm + ((bit >> 5) << 2)),
+                 1u << (bit & 31));
+    }
+  }
+}
+
 /* snappy block decode, lane-serial (golang/snappy format;
  * lib/compress/compress.go:132-144). Returns decoded length or -1. */
 __device__ int64_t d_snappy_decode(const uint8_t *src, int64_t len, uint8_t *dst,
@@ -1126,7 +1231,12 @@ __global__ void __launch_bounds__(256) k_scan_general(
     uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
     int64_t offset, int64_t q_start, int64_t q_end, int filter_op,
     double filter_f, int64_t filter_i, uint8_t *__restrict__ scratch,
-    uint64_t scratch_per_lane, uint32_t nlanes, DevErr *err) {
+    uint64_t scratch_per_lane, uint32_t nlanes,
+    const uint8_t *__restrict__ xrow_bm, /* cross-field row predicate bits
+        (one per row over the whole shard, segment base = xrow_base[si]);
+        rows with a 0 bit are removed before aggregation (FilterByField
+        with a condition on ANOTHER field, location.go:309) */
+    const uint64_t *__restrict__ xrow_base, DevErr *err) {
   uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
   if (gid >= nlanes) return;
   uint8_t *my = scratch + (uint64_t)gid * scratch_per_lane;
@@ -1233,6 +1343,41 @@ __global__ void __launch_bounds__(256) k_scan_general(
     {
       Partial *pb0 = partials + sq.partial_base;
       for (uint32_t k = 0; k < sq.n_wins; k++) pb0[k].has_rows = 0;
+    }
+    if (xrow_bm) {
+      /* cross-field predicate: keep rows whose bit is set, preserving the
+       * value column's nil structure (its own scratch slice — the clip
+       * stage below rebuilds into a different one) */
+      uint8_t *xnb = my + 2 * 4096 * 8 + 40960 + 512;
+      for (int k2 = 0; k2 < 512; k2++) xnb[k2] = 0;
+      const uint64_t rb = xrow_base[si];
+      int w_rows = 0, w_vals = 0, vi4 = 0;
+      for (int r2 = 0; r2 < rows; r2++) {
+        int valid4 = 1;
+        if (h.bitmap) valid4 = bm_valid(&h, r2);
+        else if (nilcount == rows && rows > 0) valid4 = 0;
+        uint64_t bit = rb + (uint64_t)r2;
+        if (!((xrow_bm[bit >> 3] >> (bit & 7)) & 1)) {
+          if (valid4) vi4++;
+          continue;
+        }
+        if (valid4) {
+          vbuf[w_vals++] = vbuf[vi4++];
+          xnb[w_rows >> 3] |= (uint8_t)(1u << (w_rows & 7));
+        }
+        tbuf[w_rows] = tbuf[r2];
+        w_rows++;
+      }
+      rows = w_rows;
+      dense = w_vals;
+      nilcount = w_rows - w_vals;
+      if (nilcount > 0) {
+        h.bitmap = xnb;
+        h.bm_off = 0;
+      } else {
+        h.bitmap = nullptr;
+      }
+      if (rows == 0) continue;
     }
     if (filter_op == 0 && (d.min_time < q_start || d.max_time > q_end)) {
       /* time slicing KEEPS nil rows inside the range (record slicing,
@@ -2622,6 +2767,21 @@ struct gemx_shard {
     gemx_rate_row *out = nullptr;
   } rpend[2];
   int rpend_head = 0, rpend_count = 0, r_slot = 0;
+  /* cross-field predicate state: row-base prefix sums + cached pass
+   * bitmap for the last (filter shard, op, operand) */
+  uint64_t *d_row_base = nullptr;
+  uint8_t *d_xbm = nullptr;
+  uint64_t xbm_bytes = 0;
+  uint8_t *d_xscratch = nullptr;
+  uint32_t xlanes = 0;
+  struct XKey {
+    const gemx_shard *fs = nullptr;
+    int op = 0;
+    double f = 0;
+    int64_t i = 0;
+    bool valid = false;
+  } xkey;
+  const gemx_shard *x_checked = nullptr; /* alignment validated against */
   QueryPlan plan;
   RatePlan rate_plan;
   /* pre-aggregation metadata (pre_aggregation.go FloatPreAgg role): one
@@ -2881,6 +3041,9 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   hipFree(s->d_fast_grid_ids);
   hipFree(s->d_fast_stream_ids);
   hipFree(s->d_general_ids);
+  if (s->d_row_base) hipFree(s->d_row_base);
+  if (s->d_xbm) hipFree(s->d_xbm);
+  if (s->d_xscratch) hipFree(s->d_xscratch);
   hipStreamDestroy(s->stream);
   if (s->copy_stream) hipStreamDestroy(s->copy_stream);
   for (int sl = 0; sl < 2; sl++) {
@@ -2917,7 +3080,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                      int filter_op, double filter_f, int64_t filter_i,
                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
                      gemx_query_stats *stats, const char *skip_series = nullptr,
-                     const TagQuery *tagq = nullptr, int async_begin = 0) {
+                     const TagQuery *tagq = nullptr, int async_begin = 0,
+                     const uint8_t *xbm = nullptr,
+                     const uint64_t *xbase = nullptr) {
   if (!s) return GEMX_E_INVALID;
   if (async_begin && (tagq || skip_series)) {
     seterr("async begin supports plain/grouped scans only");
@@ -2933,7 +3098,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   }
   HIP_CHECK(hipSetDevice(s->device));
   const uint64_t nsegs = s->nsegs;
-  const uint64_t scratch_per_lane = 4096 * 8 * 2 + 40960 + 512; /* +512: clip bitmap */
+  const uint64_t scratch_per_lane =
+      4096 * 8 * 2 + 40960 + 1024; /* +512 clip bitmap, +512 xfield bitmap */
 
   QueryPlan &P = skip_series ? s->sub_plan : s->plan;
   uint64_t skip_hash = 0;
@@ -2943,8 +3109,10 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       skip_hash ^= (uint8_t)skip_series[g];
       skip_hash *= 1099511628211ull;
     }
-    if (!skip_hash) skip_hash = 1;
   }
+  if (xbm) /* xfield routes everything general — distinct plan key */
+    skip_hash ^= 0x9e3779b97f4a7c15ull;
+  if ((skip_series || xbm) && !skip_hash) skip_hash = 1;
   if (!P.valid || P.start != start_time || P.end != end_time ||
       P.interval != interval || P.offset != offset ||
       P.skip_hash != skip_hash) {
@@ -2962,7 +3130,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     std::vector<char> is_gen(nsegs, 0);
     for (auto id : s->general_ids) is_gen[id] = 1;
     bool any_clip = false;
-    if (skip_series) any_clip = true; /* subset launches use the queues */
+    const bool all_gen = (xbm != nullptr); /* fast kernels cannot apply the
+        cross-field bitmap: route everything through the general kernel */
+    if (skip_series || all_gen) any_clip = true; /* launches use the queues */
     for (size_t g = 0; g < s->series_ranges.size(); g++) {
       auto &r = s->series_ranges[g];
       int64_t wmin = INT64_MAX, wmax = INT64_MIN;
@@ -3006,7 +3176,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
         if (clip) {
           any_clip = true;
           gen_q.push_back(i); /* boundary rows sliced in the general kernel */
-        } else if (is_gen[i]) {
+        } else if (all_gen || is_gen[i]) {
           gen_q.push_back(i);
         } else {
           fast_q.push_back(i);
@@ -3209,13 +3379,15 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          s->stream, s->d_blob, s->d_descs, d_segq,
                          (const uint32_t *)gen_list, n, d_part, interval, offset,
                          start_time, end_time, filter_op, filter_f, filter_i,
-                         d_scratch, scratch_per_lane, gen_lanes, d_err);
+                         d_scratch, scratch_per_lane, gen_lanes, xbm, xbase,
+                         d_err);
     else
       hipLaunchKernelGGL((k_scan_general<GEMX_TYPE_INT>), dim3(blocks), dim3(TPB), 0,
                          s->stream, s->d_blob, s->d_descs, d_segq,
                          (const uint32_t *)gen_list, n, d_part, interval, offset,
                          start_time, end_time, filter_op, filter_f, filter_i,
-                         d_scratch, scratch_per_lane, gen_lanes, d_err);
+                         d_scratch, scratch_per_lane, gen_lanes, xbm, xbase,
+                         d_err);
   }
   HIP_CHECK(hipEventRecord(ev1, s->stream));
   /* GROUP BY tag stage: ensure the permutation/chunk tables and buffers */
@@ -3752,6 +3924,110 @@ extern "C" int gemx_scan_agg_series(gemx_shard *s, const uint8_t *series_mask,
   return scan_impl(s, start_time, end_time, interval, offset, group_all,
                    filter_op, filter_f, filter_i, out_host, cap, n_out, stats,
                    skip.data());
+}
+
+/* Cross-field predicate scan (config #3: `SELECT agg(value) WHERE
+ * other_field <op> x`): filter_shard holds the predicate column of the
+ * SAME measurement — same device, same (sid, rows) segment sequence as
+ * value_shard (the row-group alignment ChunkMeta guarantees,
+ * tssp_file_meta.go). A one-lane-per-segment kernel evaluates the
+ * predicate over the filter column (binaryfilterfunc compare kernels;
+ * nil rows fail) into a row bitmap, which the value shard's scan then
+ * applies before aggregation (FilterByField, location.go:309). The
+ * bitmap caches on value_shard keyed by (filter_shard, op, operand). */
+extern "C" int gemx_scan_agg_xfield(gemx_shard *vs, gemx_shard *fs,
+                                    int filter_op, double filter_f,
+                                    int64_t filter_i, int64_t start_time,
+                                    int64_t end_time, int64_t interval,
+                                    int64_t offset, int group_all,
+                                    gemx_agg_row *out_host, uint64_t cap,
+                                    uint64_t *n_out, gemx_query_stats *stats) {
+  if (!vs || !fs || filter_op < 1 || filter_op > 6) {
+    seterr("scan_agg_xfield: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  if (vs->device != fs->device) {
+    seterr("value and filter shards must be on the same device");
+    return GEMX_E_INVALID;
+  }
+  if (vs->x_checked != fs) {
+    if (vs->nsegs != fs->nsegs) {
+      seterr("filter shard segment count differs");
+      return GEMX_E_INVALID;
+    }
+    for (uint64_t i = 0; i < vs->nsegs; i++) {
+      if (vs->h_descs[i].sid != fs->h_descs[i].sid ||
+          vs->h_descs[i].rows != fs->h_descs[i].rows) {
+        seterr("filter shard segments not row-aligned with value shard");
+        return GEMX_E_INVALID;
+      }
+    }
+    vs->x_checked = fs;
+    vs->xkey.valid = false;
+  }
+  HIP_CHECK(hipSetDevice(vs->device));
+  if (!vs->d_row_base) {
+    std::vector<uint64_t> base(vs->nsegs ? vs->nsegs : 1);
+    uint64_t acc = 0;
+    for (uint64_t i = 0; i < vs->nsegs; i++) {
+      base[i] = acc;
+      acc += vs->h_descs[i].rows;
+    }
+    HIP_CHECK(hipMalloc(&vs->d_row_base, sizeof(uint64_t) * base.size()));
+    HIP_CHECK(hipMemcpyAsync(vs->d_row_base, base.data(),
+                             sizeof(uint64_t) * base.size(),
+                             hipMemcpyHostToDevice, vs->stream));
+    vs->xbm_bytes = (vs->total_rows_scanned + 63) / 64 * 8; /* word pad */
+    HIP_CHECK(hipMalloc(&vs->d_xbm, vs->xbm_bytes ? vs->xbm_bytes : 8));
+  }
+  const bool same = vs->xkey.valid && vs->xkey.fs == fs &&
+                    vs->xkey.op == filter_op && vs->xkey.f == filter_f &&
+                    vs->xkey.i == filter_i;
+  if (!same) {
+    if (vs->pend_count > 0 || vs->rpend_count > 0) {
+      seterr("cannot re-evaluate the predicate with queries in flight");
+      return GEMX_E_INVALID;
+    }
+    if (!vs->d_xscratch) {
+      vs->xlanes = (uint32_t)std::min<uint64_t>(vs->nsegs, 16384);
+      HIP_CHECK(hipMalloc(&vs->d_xscratch,
+                          (uint64_t)(4096 * 8 + 40960) * vs->xlanes));
+    }
+    HIP_CHECK(hipMemsetAsync(vs->d_xbm, 0, vs->xbm_bytes, vs->stream));
+    DevErr *e = vs->plan.valid ? vs->plan.d_err2[0] : nullptr;
+    DevErr htmp = {0};
+    DevErr *d_etmp = nullptr;
+    if (!e) {
+      HIP_CHECK(hipMalloc(&d_etmp, sizeof(DevErr)));
+      e = d_etmp;
+    }
+    HIP_CHECK(hipMemsetAsync(e, 0, sizeof(DevErr), vs->stream));
+    uint32_t blocks = (vs->xlanes + 255) / 256;
+    if (fs->col_type == GEMX_TYPE_FLOAT)
+      hipLaunchKernelGGL((k_eval_filter<GEMX_TYPE_FLOAT>), dim3(blocks),
+                         dim3(256), 0, vs->stream, fs->d_blob, fs->d_descs,
+                         (uint32_t)fs->nsegs, vs->d_row_base, filter_op,
+                         filter_f, filter_i, vs->d_xbm, vs->d_xscratch,
+                         (uint64_t)(4096 * 8 + 40960), vs->xlanes, e);
+    else
+      hipLaunchKernelGGL((k_eval_filter<GEMX_TYPE_INT>), dim3(blocks),
+                         dim3(256), 0, vs->stream, fs->d_blob, fs->d_descs,
+                         (uint32_t)fs->nsegs, vs->d_row_base, filter_op,
+                         filter_f, filter_i, vs->d_xbm, vs->d_xscratch,
+                         (uint64_t)(4096 * 8 + 40960), vs->xlanes, e);
+    HIP_CHECK(hipMemcpyAsync(&htmp, e, sizeof(DevErr), hipMemcpyDeviceToHost,
+                             vs->stream));
+    HIP_CHECK(hipStreamSynchronize(vs->stream));
+    if (d_etmp) hipFree(d_etmp);
+    if (htmp.code != 0) {
+      seterr("filter column decode failed on device");
+      return htmp.code;
+    }
+    vs->xkey = gemx_shard::XKey{fs, filter_op, filter_f, filter_i, true};
+  }
+  return scan_impl(vs, start_time, end_time, interval, offset, group_all, 0,
+                   0, 0, out_host, cap, n_out, stats, nullptr, nullptr, 0,
+                   vs->d_xbm, vs->d_row_base);
 }
 
 /* Async pipeline (the cursor read-ahead model): begin enqueues the whole

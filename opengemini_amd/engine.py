@@ -156,6 +156,12 @@ def _load():
     lib.gemx_scan_agg_finish.argtypes = [
         C.c_void_p, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_scan_agg_xfield.restype = C.c_int
+    lib.gemx_scan_agg_xfield.argtypes = [
+        C.c_void_p, C.c_void_p, C.c_int, C.c_double, C.c_int64,
+        C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     lib.gemx_scan_agg_series.restype = C.c_int
     lib.gemx_scan_agg_series.argtypes = [
         C.c_void_p, C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64,
@@ -515,6 +521,35 @@ class Shard:
             host_ms=st.h2d_ms, points=st.points,
             compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
             meta_rows=int(nm.value),
+        )
+        return out[: n.value], stats
+
+    def scan_agg_xfield(self, filter_shard, filter, start_time, end_time,
+                        interval, offset=0, group_all=False, out_cap=None):
+        """Cross-field predicate (config #3): aggregate THIS shard's
+        column over rows where filter_shard's column passes the
+        predicate. filter = (op_name, operand) typed by the filter
+        shard's column type."""
+        lib = self._lib
+        op_name, operand = filter
+        fop = self.FILTER_OPS[op_name]
+        ff = float(operand) if filter_shard.col_type == GEMX_TYPE_FLOAT else 0.0
+        fi = int(operand) if filter_shard.col_type == GEMX_TYPE_INT else 0
+        if out_cap is None:
+            out_cap = self._rows_bound(interval, offset, group_all)
+        out = self._pooled_out("agg", out_cap, AGG_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_scan_agg_xfield(
+            self._h, filter_shard._h, fop, ff, fi,
+            start_time, end_time, interval, offset, 1 if group_all else 0,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        stats = dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            host_ms=st.h2d_ms, points=st.points,
+            compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
 
