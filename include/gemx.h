@@ -212,6 +212,29 @@ int gemx_scan_agg_xfield(gemx_shard *value_shard, gemx_shard *filter_shard,
                          gemx_agg_row *out_host, uint64_t cap,
                          uint64_t *n_out, gemx_query_stats *stats);
 
+/* One compare condition for CNF composition (gemx_scan_agg_cnf):
+ * conditions with equal `group` OR together; groups AND together.
+ * filter_shard NULL = the value shard's own column. op/f/i typed by the
+ * filter shard's column type, as in gemx_scan_agg_xfield. */
+typedef struct {
+  gemx_shard *filter_shard;
+  int op;
+  double f;
+  int64_t i;
+  uint32_t group;
+  uint32_t _pad;
+} gemx_cond;
+
+/* CNF predicate scan: the AND-of-ORs condition tree lib/binaryfilterfunc
+ * normalizes to, evaluated across any mix of row-aligned field columns.
+ * Not cached across calls (single conditions should use
+ * gemx_scan_agg_xfield, which caches its bitmap). */
+int gemx_scan_agg_cnf(gemx_shard *value_shard, const gemx_cond *conds,
+                      uint32_t n_conds, int64_t start_time, int64_t end_time,
+                      int64_t interval, int64_t offset, int group_all,
+                      gemx_agg_row *out_host, uint64_t cap, uint64_t *n_out,
+                      gemx_query_stats *stats);
+
 /* Series-subset scan — the tag-predicate seam for the column-store path
  * (config #3): the executor evaluates tag conditions against its index
  * (tsi scan / lib/binaryfilterfunc on tag columns) and passes the

@@ -1165,6 +1165,14 @@ __global__ void __launch_bounds__(256) k_eval_filter(
   }
 }
 
+/* elementwise AND of two row bitmaps (CNF group combination) */
+__global__ void __launch_bounds__(256) k_bm_and(uint64_t *__restrict__ dst,
+                                                const uint64_t *__restrict__ src,
+                                                uint64_t nwords) {
+  uint64_t i = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
+  for (; i < nwords; i += gridDim.x * (uint64_t)blockDim.x) dst[i] &= src[i];
+}
+
 /* snappy block decode, lane-serial (golang/snappy format;
  * lib/compress/compress.go:132-144). Returns decoded length or -1. */
 __device__ int64_t d_snappy_decode(const uint8_t *src, int64_t len, uint8_t *dst,
@@ -3935,6 +3943,68 @@ extern "C" int gemx_scan_agg_series(gemx_shard *s, const uint8_t *series_mask,
  * nil rows fail) into a row bitmap, which the value shard's scan then
  * applies before aggregation (FilterByField, location.go:309). The
  * bitmap caches on value_shard keyed by (filter_shard, op, operand). */
+/* validate that fs is row-aligned with vs (cached per pair) */
+static int xfield_check_aligned(gemx_shard *vs, gemx_shard *fs) {
+  if (vs->device != fs->device) {
+    seterr("value and filter shards must be on the same device");
+    return GEMX_E_INVALID;
+  }
+  if (vs->nsegs != fs->nsegs) {
+    seterr("filter shard segment count differs");
+    return GEMX_E_INVALID;
+  }
+  for (uint64_t i = 0; i < vs->nsegs; i++) {
+    if (vs->h_descs[i].sid != fs->h_descs[i].sid ||
+        vs->h_descs[i].rows != fs->h_descs[i].rows) {
+      seterr("filter shard segments not row-aligned with value shard");
+      return GEMX_E_INVALID;
+    }
+  }
+  return GEMX_OK;
+}
+
+static int xfield_ensure_base(gemx_shard *vs) {
+  if (vs->d_row_base) return GEMX_OK;
+  std::vector<uint64_t> base(vs->nsegs ? vs->nsegs : 1);
+  uint64_t acc = 0;
+  for (uint64_t i = 0; i < vs->nsegs; i++) {
+    base[i] = acc;
+    acc += vs->h_descs[i].rows;
+  }
+  HIP_CHECK(hipMalloc(&vs->d_row_base, sizeof(uint64_t) * base.size()));
+  HIP_CHECK(hipMemcpyAsync(vs->d_row_base, base.data(),
+                           sizeof(uint64_t) * base.size(),
+                           hipMemcpyHostToDevice, vs->stream));
+  vs->xbm_bytes = (vs->total_rows_scanned + 63) / 64 * 8; /* word pad */
+  HIP_CHECK(hipMalloc(&vs->d_xbm, vs->xbm_bytes ? vs->xbm_bytes : 8));
+  return GEMX_OK;
+}
+
+/* evaluate one compare over fs's column, OR-ing pass bits into bm */
+static int xfield_eval_into(gemx_shard *vs, gemx_shard *fs, int filter_op,
+                            double filter_f, int64_t filter_i, uint8_t *bm,
+                            DevErr *e) {
+  if (!vs->d_xscratch) {
+    vs->xlanes = (uint32_t)std::min<uint64_t>(vs->nsegs, 16384);
+    HIP_CHECK(hipMalloc(&vs->d_xscratch,
+                        (uint64_t)(4096 * 8 + 40960) * vs->xlanes));
+  }
+  uint32_t blocks = (vs->xlanes + 255) / 256;
+  if (fs->col_type == GEMX_TYPE_FLOAT)
+    hipLaunchKernelGGL((k_eval_filter<GEMX_TYPE_FLOAT>), dim3(blocks),
+                       dim3(256), 0, vs->stream, fs->d_blob, fs->d_descs,
+                       (uint32_t)fs->nsegs, vs->d_row_base, filter_op,
+                       filter_f, filter_i, bm, vs->d_xscratch,
+                       (uint64_t)(4096 * 8 + 40960), vs->xlanes, e);
+  else
+    hipLaunchKernelGGL((k_eval_filter<GEMX_TYPE_INT>), dim3(blocks),
+                       dim3(256), 0, vs->stream, fs->d_blob, fs->d_descs,
+                       (uint32_t)fs->nsegs, vs->d_row_base, filter_op,
+                       filter_f, filter_i, bm, vs->d_xscratch,
+                       (uint64_t)(4096 * 8 + 40960), vs->xlanes, e);
+  return GEMX_OK;
+}
+
 extern "C" int gemx_scan_agg_xfield(gemx_shard *vs, gemx_shard *fs,
                                     int filter_op, double filter_f,
                                     int64_t filter_i, int64_t start_time,
@@ -3946,40 +4016,15 @@ extern "C" int gemx_scan_agg_xfield(gemx_shard *vs, gemx_shard *fs,
     seterr("scan_agg_xfield: bad arguments");
     return GEMX_E_INVALID;
   }
-  if (vs->device != fs->device) {
-    seterr("value and filter shards must be on the same device");
-    return GEMX_E_INVALID;
-  }
   if (vs->x_checked != fs) {
-    if (vs->nsegs != fs->nsegs) {
-      seterr("filter shard segment count differs");
-      return GEMX_E_INVALID;
-    }
-    for (uint64_t i = 0; i < vs->nsegs; i++) {
-      if (vs->h_descs[i].sid != fs->h_descs[i].sid ||
-          vs->h_descs[i].rows != fs->h_descs[i].rows) {
-        seterr("filter shard segments not row-aligned with value shard");
-        return GEMX_E_INVALID;
-      }
-    }
+    int rc = xfield_check_aligned(vs, fs);
+    if (rc) return rc;
     vs->x_checked = fs;
     vs->xkey.valid = false;
   }
   HIP_CHECK(hipSetDevice(vs->device));
-  if (!vs->d_row_base) {
-    std::vector<uint64_t> base(vs->nsegs ? vs->nsegs : 1);
-    uint64_t acc = 0;
-    for (uint64_t i = 0; i < vs->nsegs; i++) {
-      base[i] = acc;
-      acc += vs->h_descs[i].rows;
-    }
-    HIP_CHECK(hipMalloc(&vs->d_row_base, sizeof(uint64_t) * base.size()));
-    HIP_CHECK(hipMemcpyAsync(vs->d_row_base, base.data(),
-                             sizeof(uint64_t) * base.size(),
-                             hipMemcpyHostToDevice, vs->stream));
-    vs->xbm_bytes = (vs->total_rows_scanned + 63) / 64 * 8; /* word pad */
-    HIP_CHECK(hipMalloc(&vs->d_xbm, vs->xbm_bytes ? vs->xbm_bytes : 8));
-  }
+  int rc = xfield_ensure_base(vs);
+  if (rc) return rc;
   const bool same = vs->xkey.valid && vs->xkey.fs == fs &&
                     vs->xkey.op == filter_op && vs->xkey.f == filter_f &&
                     vs->xkey.i == filter_i;
@@ -3988,43 +4033,115 @@ extern "C" int gemx_scan_agg_xfield(gemx_shard *vs, gemx_shard *fs,
       seterr("cannot re-evaluate the predicate with queries in flight");
       return GEMX_E_INVALID;
     }
-    if (!vs->d_xscratch) {
-      vs->xlanes = (uint32_t)std::min<uint64_t>(vs->nsegs, 16384);
-      HIP_CHECK(hipMalloc(&vs->d_xscratch,
-                          (uint64_t)(4096 * 8 + 40960) * vs->xlanes));
-    }
     HIP_CHECK(hipMemsetAsync(vs->d_xbm, 0, vs->xbm_bytes, vs->stream));
-    DevErr *e = vs->plan.valid ? vs->plan.d_err2[0] : nullptr;
     DevErr htmp = {0};
     DevErr *d_etmp = nullptr;
-    if (!e) {
-      HIP_CHECK(hipMalloc(&d_etmp, sizeof(DevErr)));
-      e = d_etmp;
+    HIP_CHECK(hipMalloc(&d_etmp, sizeof(DevErr)));
+    HIP_CHECK(hipMemsetAsync(d_etmp, 0, sizeof(DevErr), vs->stream));
+    rc = xfield_eval_into(vs, fs, filter_op, filter_f, filter_i, vs->d_xbm,
+                          d_etmp);
+    if (rc) {
+      hipFree(d_etmp);
+      return rc;
     }
-    HIP_CHECK(hipMemsetAsync(e, 0, sizeof(DevErr), vs->stream));
-    uint32_t blocks = (vs->xlanes + 255) / 256;
-    if (fs->col_type == GEMX_TYPE_FLOAT)
-      hipLaunchKernelGGL((k_eval_filter<GEMX_TYPE_FLOAT>), dim3(blocks),
-                         dim3(256), 0, vs->stream, fs->d_blob, fs->d_descs,
-                         (uint32_t)fs->nsegs, vs->d_row_base, filter_op,
-                         filter_f, filter_i, vs->d_xbm, vs->d_xscratch,
-                         (uint64_t)(4096 * 8 + 40960), vs->xlanes, e);
-    else
-      hipLaunchKernelGGL((k_eval_filter<GEMX_TYPE_INT>), dim3(blocks),
-                         dim3(256), 0, vs->stream, fs->d_blob, fs->d_descs,
-                         (uint32_t)fs->nsegs, vs->d_row_base, filter_op,
-                         filter_f, filter_i, vs->d_xbm, vs->d_xscratch,
-                         (uint64_t)(4096 * 8 + 40960), vs->xlanes, e);
-    HIP_CHECK(hipMemcpyAsync(&htmp, e, sizeof(DevErr), hipMemcpyDeviceToHost,
-                             vs->stream));
+    HIP_CHECK(hipMemcpyAsync(&htmp, d_etmp, sizeof(DevErr),
+                             hipMemcpyDeviceToHost, vs->stream));
     HIP_CHECK(hipStreamSynchronize(vs->stream));
-    if (d_etmp) hipFree(d_etmp);
+    hipFree(d_etmp);
     if (htmp.code != 0) {
       seterr("filter column decode failed on device");
       return htmp.code;
     }
     vs->xkey = gemx_shard::XKey{fs, filter_op, filter_f, filter_i, true};
   }
+  return scan_impl(vs, start_time, end_time, interval, offset, group_all, 0,
+                   0, 0, out_host, cap, n_out, stats, nullptr, nullptr, 0,
+                   vs->d_xbm, vs->d_row_base);
+}
+
+/* CNF composition of compare conditions across fields (the condition
+ * tree lib/binaryfilterfunc normalizes to): conditions sharing a group
+ * id OR together (they evaluate into one zeroed bitmap — the eval
+ * kernel's atomicOr makes OR free), and the groups AND together.
+ * cond.filter_shard NULL means the value shard itself. Results are not
+ * cached across calls (each call re-evaluates; single-condition callers
+ * should prefer gemx_scan_agg_xfield, which caches). */
+extern "C" int gemx_scan_agg_cnf(gemx_shard *vs, const gemx_cond *conds,
+                                 uint32_t n_conds, int64_t start_time,
+                                 int64_t end_time, int64_t interval,
+                                 int64_t offset, int group_all,
+                                 gemx_agg_row *out_host, uint64_t cap,
+                                 uint64_t *n_out, gemx_query_stats *stats) {
+  if (!vs || !conds || n_conds == 0 || n_conds > 64) {
+    seterr("scan_agg_cnf: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  if (vs->pend_count > 0 || vs->rpend_count > 0) {
+    seterr("cannot evaluate predicates with queries in flight");
+    return GEMX_E_INVALID;
+  }
+  HIP_CHECK(hipSetDevice(vs->device));
+  int rc = xfield_ensure_base(vs);
+  if (rc) return rc;
+  for (uint32_t c = 0; c < n_conds; c++) {
+    gemx_shard *fs = conds[c].filter_shard ? conds[c].filter_shard : vs;
+    if (conds[c].op < 1 || conds[c].op > 6) {
+      seterr("scan_agg_cnf: bad compare op");
+      return GEMX_E_INVALID;
+    }
+    if (fs != vs) {
+      rc = xfield_check_aligned(vs, fs);
+      if (rc) return rc;
+    }
+  }
+  /* group-ordered evaluation: group bitmap accumulates ORs, then ANDs
+   * into the final bitmap */
+  uint8_t *d_grp = nullptr;
+  HIP_CHECK(hipMalloc(&d_grp, vs->xbm_bytes ? vs->xbm_bytes : 8));
+  DevErr htmp = {0};
+  DevErr *d_etmp = nullptr;
+  HIP_CHECK(hipMalloc(&d_etmp, sizeof(DevErr)));
+  HIP_CHECK(hipMemsetAsync(d_etmp, 0, sizeof(DevErr), vs->stream));
+  const uint64_t nwords = vs->xbm_bytes / 8;
+  uint32_t bm_blocks =
+      (uint32_t)std::min<uint64_t>((nwords + 255) / 256, 65535);
+  bool first_group = true;
+  uint32_t c = 0;
+  int err_rc = GEMX_OK;
+  while (c < n_conds && err_rc == GEMX_OK) {
+    uint32_t g = conds[c].group;
+    HIP_CHECK(hipMemsetAsync(d_grp, 0, vs->xbm_bytes, vs->stream));
+    while (c < n_conds && conds[c].group == g) {
+      gemx_shard *fs = conds[c].filter_shard ? conds[c].filter_shard : vs;
+      err_rc = xfield_eval_into(vs, fs, conds[c].op, conds[c].f, conds[c].i,
+                                d_grp, d_etmp);
+      if (err_rc) break;
+      c++;
+    }
+    if (err_rc) break;
+    if (first_group) {
+      HIP_CHECK(hipMemcpyAsync(vs->d_xbm, d_grp, vs->xbm_bytes,
+                               hipMemcpyDeviceToDevice, vs->stream));
+      first_group = false;
+    } else if (nwords) {
+      hipLaunchKernelGGL(k_bm_and, dim3(bm_blocks), dim3(256), 0, vs->stream,
+                         (uint64_t *)vs->d_xbm, (const uint64_t *)d_grp,
+                         nwords);
+    }
+  }
+  if (err_rc == GEMX_OK) {
+    HIP_CHECK(hipMemcpyAsync(&htmp, d_etmp, sizeof(DevErr),
+                             hipMemcpyDeviceToHost, vs->stream));
+    HIP_CHECK(hipStreamSynchronize(vs->stream));
+  }
+  hipFree(d_grp);
+  hipFree(d_etmp);
+  if (err_rc) return err_rc;
+  if (htmp.code != 0) {
+    seterr("filter column decode failed on device");
+    return htmp.code;
+  }
+  vs->xkey.valid = false; /* the cached single-cond bitmap is overwritten */
   return scan_impl(vs, start_time, end_time, interval, offset, group_all, 0,
                    0, 0, out_host, cap, n_out, stats, nullptr, nullptr, 0,
                    vs->d_xbm, vs->d_row_base);

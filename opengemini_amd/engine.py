@@ -162,6 +162,12 @@ def _load():
         C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_scan_agg_cnf.restype = C.c_int
+    lib.gemx_scan_agg_cnf.argtypes = [
+        C.c_void_p, C.c_void_p, C.c_uint32,
+        C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
+    ]
     lib.gemx_scan_agg_series.restype = C.c_int
     lib.gemx_scan_agg_series.argtypes = [
         C.c_void_p, C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64,
@@ -542,6 +548,46 @@ class Shard:
         st = _Stats()
         rc = lib.gemx_scan_agg_xfield(
             self._h, filter_shard._h, fop, ff, fi,
+            start_time, end_time, interval, offset, 1 if group_all else 0,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        stats = dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms, total_ms=st.total_ms,
+            host_ms=st.h2d_ms, points=st.points,
+            compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
+        )
+        return out[: n.value], stats
+
+    def scan_agg_cnf(self, conds, start_time, end_time, interval, offset=0,
+                     group_all=False, out_cap=None):
+        """CNF predicate scan: conds is a list of
+        (filter_shard_or_None, op_name, operand, group) — same group ORs,
+        groups AND. None uses this shard's own column."""
+        lib = self._lib
+
+        class _Cond(C.Structure):
+            _fields_ = [("fs", C.c_void_p), ("op", C.c_int),
+                        ("f", C.c_double), ("i", C.c_int64),
+                        ("group", C.c_uint32), ("_pad", C.c_uint32)]
+
+        arr = (_Cond * len(conds))()
+        for k, (fsh, op_name, operand, grp) in enumerate(conds):
+            tgt = fsh if fsh is not None else self
+            arr[k].fs = tgt._h if fsh is not None else None
+            arr[k].op = self.FILTER_OPS[op_name]
+            arr[k].f = (float(operand)
+                        if tgt.col_type == GEMX_TYPE_FLOAT else 0.0)
+            arr[k].i = (int(operand)
+                        if tgt.col_type == GEMX_TYPE_INT else 0)
+            arr[k].group = grp
+        if out_cap is None:
+            out_cap = self._rows_bound(interval, offset, group_all)
+        out = self._pooled_out("agg", out_cap, AGG_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_scan_agg_cnf(
+            self._h, C.byref(arr), len(conds),
             start_time, end_time, interval, offset, 1 if group_all else 0,
             out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
         )

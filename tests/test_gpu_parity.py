@@ -1179,3 +1179,51 @@ class TestCrossFieldPredicate:
         finally:
             vsh.close()
             fsh.close()
+
+
+class TestCNFPredicates:
+    """gemx_scan_agg_cnf: AND-of-OR condition trees across fields."""
+
+    def test_and_of_ors(self):
+        import opengemini_amd as gx
+        rng = np.random.default_rng(1901)
+        ns, pts = 30, 400
+        n = ns * pts
+        sids = np.repeat(np.arange(1, ns + 1, dtype=np.uint64), pts)
+        times = np.tile(np.arange(pts, dtype=np.int64) * 10**9, ns)
+        vals = np.round(np.cumsum(rng.normal(0, 1, n)) * 128) / 128
+        fa = rng.normal(0, 10, n)           # float field A
+        fb = rng.integers(0, 100, n).astype(np.int64)  # int field B
+        vblob, vd = gx.encode_shard(F, sids, times, vals)
+        ablob, ad = gx.encode_shard(F, sids, times, fa)
+        bblob, bd = gx.encode_shard(I, sids, times, fb)
+        vsh = gx.Shard(vblob, np.ascontiguousarray(vd), F)
+        ash = gx.Shard(ablob, np.ascontiguousarray(ad), F)
+        bsh = gx.Shard(bblob, np.ascontiguousarray(bd), I)
+        try:
+            # (fa > 5 OR fa < -5) AND (fb >= 20) AND (value > 0)
+            conds = [(ash, "gt", 5.0, 0), (ash, "lt", -5.0, 0),
+                     (bsh, "ge", 20, 1), (None, "gt", 0.0, 2)]
+            rows, _ = vsh.scan_agg_cnf(conds, 0, 2**62, INT)
+            rows = rows.copy()
+        finally:
+            vsh.close(); ash.close(); bsh.close()
+        passm = ((fa > 5.0) | (fa < -5.0)) & (fb >= 20) & (vals > 0.0)
+        # expected per (sid, window)
+        got_count = 0
+        idx = 0
+        for sid in range(1, ns + 1):
+            m = (sids == sid) & passm
+            st, sv = times[m], vals[m]
+            wins = (st // INT) * INT
+            for w in np.unique(wins):
+                wm = wins == w
+                vv = sv[wm]
+                r = rows[idx]; idx += 1
+                assert int(r["sid"]) == sid and int(r["win_start"]) == w
+                assert int(r["count"]) == len(vv)
+                assert abs(r["sum"] - vv.sum()) <= 1e-9 * max(1, abs(vv.sum()))
+                assert r["min"] == vv.min() and r["max"] == vv.max()
+                got_count += len(vv)
+        assert idx == len(rows)
+        assert got_count == int(passm.sum())
