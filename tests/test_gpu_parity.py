@@ -1296,3 +1296,71 @@ class TestOffsetsAndNegativeTimes:
                 assert_parity(gpu, ref, F)
         finally:
             sh.close()
+
+
+class TestRateFuzzAndFilterEdges:
+    def test_rate_param_fuzz(self):
+        """Random (range, step) combos within the 8-slot window ring,
+        each vs the oracle."""
+        rng = np.random.default_rng(2101)
+        blob, descs = orc.gen_shard(2101, 80, 1000)
+        sh = gpu_shard(blob, descs, F)
+        S = 10**9
+        try:
+            for _ in range(8):
+                step = int(rng.integers(10, 120)) * S
+                k = int(rng.integers(1, 7))  # range/step + 2 <= 8
+                range_ns = k * step - int(rng.integers(0, step // S)) * S
+                if range_ns <= 0:
+                    range_ns = step
+                end = int(rng.integers(500, 999)) * S
+                gpu, _ = sh.prom_rate(0, end, range_ns, step)
+                gpu = gpu.copy()
+                ref = orc.prom_rate(blob, descs, 0, end, range_ns, step,
+                                    is_rate=True, is_counter=True)
+                assert len(gpu) == len(ref), (range_ns, step, end)
+                assert np.array_equal(gpu["sid"], ref["sid"])
+                assert np.array_equal(gpu["ts"], ref["ts"])
+                assert np.allclose(gpu["value"], ref["value"], rtol=1e-12)
+            # over the ring bound: loud refusal, not wrong answers
+            import opengemini_amd as gx
+            with pytest.raises(gx.GemxError):
+                sh.prom_rate(0, 999 * S, 1000 * S, 10 * S)
+        finally:
+            sh.close()
+
+    def test_filter_nan_operand_drops_everything(self):
+        blob, descs = orc.gen_shard(2102, 30, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            rows, _ = sh.scan_agg(0, 2**62, INT, filter=("gt", float("nan")))
+            assert len(rows) == 0  # x > NaN is false: every row dropped
+            # x != NaN is TRUE in IEEE (Go and C++ agree): rows survive —
+            # parity with the oracle is the contract
+            rows2, _ = sh.scan_agg(0, 2**62, INT, filter=("neq", float("nan")))
+            rows2 = rows2.copy()
+        finally:
+            sh.close()
+        ref = orc.scan_agg_filtered(blob, descs, F, 0, 2**62, INT,
+                                    "neq", float("nan"))
+        assert_parity(rows2, ref, F)
+
+    def test_int_extreme_values(self):
+        import opengemini_amd as gx
+        vals = np.array([0, 2**63 - 1, -2**63, 1, -1, 2**62, -2**62] * 20,
+                        dtype=np.int64)
+        n = len(vals)
+        sids = np.full(n, 1, dtype=np.uint64)
+        times = np.arange(n, dtype=np.int64) * 10**9
+        blob, descs = gx.encode_shard(I, sids, times, vals)
+        descs = np.ascontiguousarray(descs)
+        sh = gpu_shard(blob, descs, I)
+        try:
+            gpu, _ = sh.scan_agg(0, 2**62, INT)
+            gpu = gpu.copy()
+        finally:
+            sh.close()
+        ref = orc.scan_agg(blob, descs, I, 0, 2**62, INT)
+        assert_parity(gpu, ref, I)
+        assert int(np.array(gpu[0]["min"]).view(np.int64)) == -2**63
+        assert int(np.array(gpu[0]["max"]).view(np.int64)) == 2**63 - 1
