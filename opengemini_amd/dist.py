@@ -4,55 +4,109 @@ ts-sql-side StreamAggregateTransform re-aggregation
 
 Shards partition one-per-GPU (one process per GPU over torch.distributed;
 backend "nccl" IS RCCL on ROCm). The only inter-GPU traffic on this path is
-the final GROUP BY time merge of per-shard partials: windows × ~32 B —
-sub-millisecond over xGMI, done as one collective per reduction kind rather
-than a ring over a big tensor (prefer one-shot reduce; ring
-all-reduce is per-link bound and irrelevant at these sizes).
+the final GROUP BY time merge of per-shard partials: windows × 96 B —
+sub-millisecond over xGMI. Per SURVEY.md §8e the min/max/first/last
+families carry timestamps and tie-break on them (then on processing
+order), which no builtin collective op expresses — so the merge is ONE
+all_gather of the partial tensor followed by a deterministic fold in rank
+order on every rank (the "gather + merge kernel" shape; symmetric
+all_gather because the tensor is tiny and every rank may serve results).
+
+Fold semantics per window = AggTagSetCursor.UpdateRec applied shard after
+shard in rank order (engine/agg_tagset_cursor.go:1111;
+lib/record/reccord_functions.go): count/sum accumulate; min/max by value
+with earlier-time tie-break, full ties keep the first-processed (lower
+rank); first = smallest time (ties keep first-processed), last = largest
+time (ties keep first-processed).
 """
 
 import numpy as np
 import torch
 import torch.distributed as dist
 
+# partial layout per window (float64 slots; times bit-cast from int64,
+# exact — float64 slots hold the BITS via .view, not a lossy cast)
+NCOLS = 12
+(C_COUNT, C_SUM, C_MIN, C_MINT, C_MAX, C_MAXT, C_FIRST, C_FIRSTT, C_LAST,
+ C_LASTT, C_HASMM, C_HASFL) = range(NCOLS)
+
 
 def window_partials(rows, interval, offset, w0, n_wins):
-    """Collapse per-(sid,window) rows into per-window group partials for the
-    count/sum/min/max families (AggTagSetCursor.UpdateRec role,
-    engine/agg_tagset_cursor.go:1111, for the all-series group of
-    `GROUP BY time(w)` with no tag dimensions).
-
-    Returns float64 tensor [n_wins, 4]: count, sum, min, max
-    (count stored as float64 — exact for counts < 2^53)."""
+    """Collapse per-(sid,window) GROUPED rows (sh.scan_agg(group_all=True)
+    output, one row per window) into the [n_wins, 12] partial tensor for
+    the cross-shard merge. Missing windows stay marked empty."""
+    out = np.zeros((n_wins, NCOLS), dtype=np.float64)
+    out[:, C_MIN] = np.inf
+    out[:, C_MAX] = -np.inf
+    ti = out.view(np.int64)  # time slots carried bit-exact
+    ti[:, C_MINT] = 2**63 - 1
+    ti[:, C_MAXT] = 2**63 - 1
+    ti[:, C_FIRSTT] = 2**63 - 1
+    ti[:, C_LASTT] = -(2**63)
     idx = ((rows["win_start"] - offset) // interval - w0).astype(np.int64)
-    out = np.zeros((n_wins, 4), dtype=np.float64)
-    np.add.at(out[:, 0], idx, rows["count"].astype(np.float64))
-    np.add.at(out[:, 1], idx, np.where(rows["sum_isnil"] == 1, 0.0, rows["sum"]))
-    out[:, 2] = np.inf
-    out[:, 3] = -np.inf
-    np.minimum.at(out[:, 2], idx, np.where(rows["min_isnil"] == 1, np.inf, rows["min"]))
-    np.maximum.at(out[:, 3], idx, np.where(rows["max_isnil"] == 1, -np.inf, rows["max"]))
+    out[idx, C_COUNT] = rows["count"].astype(np.float64)
+    out[idx, C_SUM] = np.where(rows["sum_isnil"] == 1, 0.0, rows["sum"])
+    valid = rows["min_isnil"] == 0
+    out[idx, C_MIN] = np.where(valid, rows["min"], np.inf)
+    ti[idx, C_MINT] = np.where(valid, rows["min_time"], 2**63 - 1)
+    out[idx, C_MAX] = np.where(valid, rows["max"], -np.inf)
+    ti[idx, C_MAXT] = np.where(valid, rows["max_time"], 2**63 - 1)
+    out[idx, C_HASMM] = valid.astype(np.float64)
+    fvalid = rows["first_isnil"] == 0
+    out[idx, C_FIRST] = np.where(fvalid, rows["first"], 0.0)
+    ti[idx, C_FIRSTT] = np.where(fvalid, rows["first_time"], 2**63 - 1)
+    out[idx, C_LAST] = np.where(fvalid, rows["last"], 0.0)
+    ti[idx, C_LASTT] = np.where(fvalid, rows["last_time"], -(2**63))
+    out[idx, C_HASFL] = fvalid.astype(np.float64)
     return out
+
+
+def _fold(acc, nxt):
+    """UpdateRec fold of one shard's partials into the accumulator
+    (vectorized over windows; nxt is the LATER-processed shard, so full
+    ties keep acc)."""
+    ai = acc.view(np.int64)
+    ni = nxt.view(np.int64)
+    acc[:, C_COUNT] += nxt[:, C_COUNT]
+    acc[:, C_SUM] += nxt[:, C_SUM]
+    a_has = acc[:, C_HASMM] > 0
+    n_has = nxt[:, C_HASMM] > 0
+    # min by value, tie -> smaller time, full tie -> keep acc (first-
+    # processed); NaN-safe: comparisons with NaN are false -> keep acc
+    take = n_has & (~a_has | (nxt[:, C_MIN] < acc[:, C_MIN]) |
+                    ((nxt[:, C_MIN] == acc[:, C_MIN]) &
+                     (ni[:, C_MINT] < ai[:, C_MINT])))
+    acc[take, C_MIN] = nxt[take, C_MIN]
+    ai[take, C_MINT] = ni[take, C_MINT]
+    take = n_has & (~a_has | (nxt[:, C_MAX] > acc[:, C_MAX]) |
+                    ((nxt[:, C_MAX] == acc[:, C_MAX]) &
+                     (ni[:, C_MAXT] < ai[:, C_MAXT])))
+    acc[take, C_MAX] = nxt[take, C_MAX]
+    ai[take, C_MAXT] = ni[take, C_MAXT]
+    acc[:, C_HASMM] = np.maximum(acc[:, C_HASMM], nxt[:, C_HASMM])
+    a_f = acc[:, C_HASFL] > 0
+    n_f = nxt[:, C_HASFL] > 0
+    take = n_f & (~a_f | (ni[:, C_FIRSTT] < ai[:, C_FIRSTT]))
+    acc[take, C_FIRST] = nxt[take, C_FIRST]
+    ai[take, C_FIRSTT] = ni[take, C_FIRSTT]
+    take = n_f & (~a_f | (ni[:, C_LASTT] > ai[:, C_LASTT]))
+    acc[take, C_LAST] = nxt[take, C_LAST]
+    ai[take, C_LASTT] = ni[take, C_LASTT]
+    acc[:, C_HASFL] = np.maximum(acc[:, C_HASFL], nxt[:, C_HASFL])
+    return acc
 
 
 def merge_across_shards(partials, device=None, group=None):
-    """One-shot cross-shard reduce of per-window group partials.
-
-    partials: np.ndarray [n_wins, 4] (count,sum,min,max) for THIS rank's
-    shard. Returns the reduced [n_wins, 4] on every rank (all_reduce keeps
-    it simple and the tensor is tiny; the reference's merge lands on ts-sql
-    exactly once — agg_transform.go:34)."""
-    t_add = torch.from_numpy(partials[:, :2].copy())
-    t_min = torch.from_numpy(partials[:, 2].copy())
-    t_max = torch.from_numpy(partials[:, 3].copy())
+    """One all_gather of this rank's [n_wins, 12] partials, then the
+    deterministic rank-ordered fold on every rank. Returns the merged
+    [n_wins, 12]."""
+    world = dist.get_world_size(group=group)
+    t = torch.from_numpy(np.ascontiguousarray(partials))
     if device is not None:
-        t_add = t_add.to(device)
-        t_min = t_min.to(device)
-        t_max = t_max.to(device)
-    dist.all_reduce(t_add, op=dist.ReduceOp.SUM, group=group)
-    dist.all_reduce(t_min, op=dist.ReduceOp.MIN, group=group)
-    dist.all_reduce(t_max, op=dist.ReduceOp.MAX, group=group)
-    out = np.empty_like(partials)
-    out[:, :2] = t_add.cpu().numpy()
-    out[:, 2] = t_min.cpu().numpy()
-    out[:, 3] = t_max.cpu().numpy()
-    return out
+        t = t.to(device)
+    gathered = [torch.empty_like(t) for _ in range(world)]
+    dist.all_gather(gathered, t, group=group)
+    acc = gathered[0].cpu().numpy().copy()
+    for r in range(1, world):
+        acc = _fold(acc, gathered[r].cpu().numpy())
+    return acc
