@@ -212,6 +212,16 @@ int gemx_scan_agg_xfield(gemx_shard *value_shard, gemx_shard *filter_shard,
                          gemx_agg_row *out_host, uint64_t cap,
                          uint64_t *n_out, gemx_query_stats *stats);
 
+/* Grouped scan that also emits EMPTY windows (count 0, every aggregate
+ * nil, times = window start) instead of dropping them — the
+ * BuildEmptyIntervalRec shape fill() consumes; the fill policy itself
+ * stays executor-side as in the reference. */
+int gemx_scan_agg_grouped_fill(gemx_shard *, int64_t start_time,
+                               int64_t end_time, int64_t interval,
+                               int64_t offset, gemx_agg_row *out_host,
+                               uint64_t cap, uint64_t *n_out,
+                               gemx_query_stats *stats);
+
 /* One compare condition for CNF composition (gemx_scan_agg_cnf):
  * conditions with equal `group` OR together; groups AND together.
  * filter_shard NULL = the value shard's own column. op/f/i typed by the

@@ -1903,7 +1903,7 @@ template <int COLTYPE>
 __global__ void __launch_bounds__(256) k_group_p2(
     const GAcc *__restrict__ gtmp, uint32_t split, gemx_agg_row *__restrict__ out,
     int64_t W0, uint32_t n_gwins, int64_t interval, int64_t offset,
-    int64_t q_start, DevErr *__restrict__ err) {
+    int64_t q_start, int keep_empty, DevErr *__restrict__ err) {
   __shared__ GAcc sh[256];
   for (uint32_t wb = blockIdx.x; wb < n_gwins; wb += gridDim.x) {
     int64_t w = W0 + (int64_t)wb;
@@ -1926,8 +1926,19 @@ __global__ void __launch_bounds__(256) k_group_p2(
       int64_t ws = interval ? win_start_of(w, interval, offset) : q_start;
       o.win_start = ws;
       if (!g->used) {
-        o.count = -1; /* gap */
-        atomicAdd(&err->gaps, 1ull);
+        if (keep_empty) {
+          /* BuildEmptyIntervalRec: emit the window with zero count and
+           * every aggregate nil (the fill transform's input shape) */
+          o.count = 0;
+          o.first_row_time = ws;
+          o.count_time = ws;
+          o.sum_time = ws;
+          o.min_isnil = o.max_isnil = o.first_isnil = o.last_isnil =
+              o.sum_isnil = 1;
+        } else {
+          o.count = -1; /* gap */
+          atomicAdd(&err->gaps, 1ull);
+        }
       } else {
         o.first_row_time = ws; /* BuildEmptyIntervalRec interval times */
         o.count = g->count;
@@ -3090,7 +3101,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                      gemx_query_stats *stats, const char *skip_series = nullptr,
                      const TagQuery *tagq = nullptr, int async_begin = 0,
                      const uint8_t *xbm = nullptr,
-                     const uint64_t *xbase = nullptr) {
+                     const uint64_t *xbase = nullptr, int keep_empty = 0) {
   if (!s) return GEMX_E_INVALID;
   if (async_begin && (tagq || skip_series)) {
     seterr("async begin supports plain/grouped scans only");
@@ -3348,10 +3359,14 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   }
   for (int li = 0; li < n_launches; li++) {
     uint32_t n = launches[li].n;
-    uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
+    /* small grids (config #1: one deep series = few segments) spread over
+     * more CUs with smaller blocks — 10k lanes in 256-thread blocks land
+     * on only ~40 of 256 CUs */
+    uint32_t tpb = (n < 64 * 1024) ? 64 : (uint32_t)TPB;
+    uint32_t blocks = std::min<uint32_t>((n + tpb - 1) / tpb, 65535);
     const uint32_t *lst = launches[li].list;
 #define GEMX_LAUNCH_FAST(CT, FLT, GP)                                          \
-    hipLaunchKernelGGL((k_scan_fast<CT, FLT, GP>), dim3(blocks), dim3(TPB), 0, \
+    hipLaunchKernelGGL((k_scan_fast<CT, FLT, GP>), dim3(blocks), dim3(tpb), 0, \
                        s->stream, s->d_blob, s->d_descs, d_segq, lst, n,       \
                        d_part, interval, offset, start_time, end_time,         \
                        filter_op, filter_f, filter_i, d_err)
@@ -3512,7 +3527,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_FLOAT>), dim3(b2), dim3(256), 0,
                          s->stream, (const GAcc *)P.d_gtmp, P.gsplit,
                          P.d_grows2[slot], P.W0, (uint32_t)P.n_gwins, interval,
-                         offset, start_time, d_err);
+                         offset, start_time, keep_empty, d_err);
     } else {
       hipLaunchKernelGGL((k_group_p1<GEMX_TYPE_INT>), dim3(b1), dim3(256), 0,
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part,
@@ -3521,7 +3536,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       hipLaunchKernelGGL((k_group_p2<GEMX_TYPE_INT>), dim3(b2), dim3(256), 0,
                          s->stream, (const GAcc *)P.d_gtmp, P.gsplit,
                          P.d_grows2[slot], P.W0, (uint32_t)P.n_gwins, interval,
-                         offset, start_time, d_err);
+                         offset, start_time, keep_empty, d_err);
     }
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
@@ -3932,6 +3947,21 @@ extern "C" int gemx_scan_agg_series(gemx_shard *s, const uint8_t *series_mask,
   return scan_impl(s, start_time, end_time, interval, offset, group_all,
                    filter_op, filter_f, filter_i, out_host, cap, n_out, stats,
                    skip.data());
+}
+
+/* Grouped scan that also emits EMPTY windows (count 0, all aggregates
+ * nil) instead of dropping them — the BuildEmptyIntervalRec shape the
+ * fill() transform consumes (engine/agg_tagset_cursor.go interval
+ * records; fill itself stays executor-side as in the reference). */
+extern "C" int gemx_scan_agg_grouped_fill(gemx_shard *s, int64_t start_time,
+                                          int64_t end_time, int64_t interval,
+                                          int64_t offset,
+                                          gemx_agg_row *out_host, uint64_t cap,
+                                          uint64_t *n_out,
+                                          gemx_query_stats *stats) {
+  return scan_impl(s, start_time, end_time, interval, offset, 1, 0, 0, 0,
+                   out_host, cap, n_out, stats, nullptr, nullptr, 0, nullptr,
+                   nullptr, 1);
 }
 
 /* Cross-field predicate scan (config #3: `SELECT agg(value) WHERE

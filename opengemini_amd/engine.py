@@ -162,6 +162,8 @@ def _load():
         C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_scan_agg_grouped_fill.restype = C.c_int
+    lib.gemx_scan_agg_grouped_fill.argtypes = scan_sig
     lib.gemx_scan_agg_cnf.restype = C.c_int
     lib.gemx_scan_agg_cnf.argtypes = [
         C.c_void_p, C.c_void_p, C.c_uint32,
@@ -558,6 +560,25 @@ class Shard:
             compressed_bytes=st.compressed_bytes, n_rows=st.n_rows,
         )
         return out[: n.value], stats
+
+    def scan_agg_grouped_fill(self, start_time, end_time, interval,
+                              offset=0, out_cap=None):
+        """Grouped scan emitting empty windows too (count 0, aggregates
+        nil) — the interval-record shape fill() consumes."""
+        lib = self._lib
+        if out_cap is None:
+            out_cap = self._rows_bound(interval, offset, True)
+        out = self._pooled_out("agg", out_cap, AGG_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_scan_agg_grouped_fill(
+            self._h, start_time, end_time, interval, offset,
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        return out[: n.value], dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms,
+            total_ms=st.total_ms, n_rows=st.n_rows)
 
     def scan_agg_cnf(self, conds, start_time, end_time, interval, offset=0,
                      group_all=False, out_cap=None):

@@ -1364,3 +1364,40 @@ class TestRateFuzzAndFilterEdges:
         assert_parity(gpu, ref, I)
         assert int(np.array(gpu[0]["min"]).view(np.int64)) == -2**63
         assert int(np.array(gpu[0]["max"]).view(np.int64)) == 2**63 - 1
+
+
+class TestGroupedFill:
+    def test_empty_windows_emitted(self):
+        import opengemini_amd as gx
+        # two series with a time GAP: rows at 0..99s and 300..399s
+        rng = np.random.default_rng(2201)
+        S = 10**9
+        sids = np.repeat([1, 2], 200).astype(np.uint64)
+        t1 = np.concatenate([np.arange(100), np.arange(300, 400)]).astype(
+            np.int64) * S
+        times = np.tile(t1, 2)
+        vals = rng.normal(0, 1, 400)
+        blob, descs = gx.encode_shard(F, sids, times, vals)
+        descs = np.ascontiguousarray(descs)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            dense, _ = sh.scan_agg_grouped_fill(0, 2**62, INT)
+            dense = dense.copy()
+            compact, _ = sh.scan_agg(0, 2**62, INT, group_all=True)
+            compact = compact.copy()
+        finally:
+            sh.close()
+        # dense covers every window ordinal from 0..399s: 7 windows;
+        # compact drops the empty ones (100..299s = ordinals 2,3,4)
+        assert len(dense) == 7
+        assert len(compact) == 4
+        empt = dense[dense["count"] == 0]
+        assert len(empt) == 3
+        assert np.all(empt["min_isnil"] == 1)
+        assert np.all(empt["sum_isnil"] == 1)
+        assert np.array_equal(empt["win_start"],
+                              np.array([2, 3, 4]) * INT)
+        # populated rows identical between the two modes
+        pop = dense[dense["count"] > 0]
+        assert np.array_equal(pop.view(np.uint8).reshape(len(pop), -1),
+                              compact.view(np.uint8).reshape(len(compact), -1))
