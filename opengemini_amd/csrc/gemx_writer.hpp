@@ -322,9 +322,46 @@ static int64_t w_time_encode(const int64_t *src, int64_t n, uint8_t *dst,
   return w_time_raw(src, n, dst, cap);
 }
 
-/* float adaptive codec (float.go:164-254). Emits same-value / gorilla /
- * compressedNull; snappy+RLE branches and the MLF path are replaced by
- * compressedNull (raw), which every reader accepts (float.go:139 case 0). */
+/* RLE runs (compress.go:68-121): [5<<4] then per run either
+ * [0x8000|count u16] for zero runs or [count u16][value f64]. */
+static int64_t w_float_rle(const double *src, int64_t n, uint8_t *dst,
+                           int64_t cap) {
+  int64_t p = 1;
+  if (cap < 1) return -1;
+  dst[0] = 5 << 4;
+  int64_t i = 0;
+  while (i < n) {
+    int64_t j = i;
+    uint64_t b0;
+    memcpy(&b0, &src[i], 8);
+    while (j < n && j - i < 0x7FFF) {
+      uint64_t bj;
+      memcpy(&bj, &src[j], 8);
+      if (bj != b0) break;
+      j++;
+    }
+    uint16_t cnt = (uint16_t)(j - i);
+    if (src[i] == 0.0 && b0 == 0) { /* +0.0 zero-run form */
+      if (p + 2 > cap) return -1;
+      dst[p] = (uint8_t)(0x80 | (cnt >> 8));
+      dst[p + 1] = (uint8_t)cnt;
+      p += 2;
+    } else {
+      if (p + 10 > cap) return -1;
+      dst[p] = (uint8_t)(cnt >> 8);
+      dst[p + 1] = (uint8_t)cnt;
+      memcpy(dst + p + 2, &src[i], 8);
+      p += 10;
+    }
+    i = j;
+  }
+  return p;
+}
+
+/* float adaptive codec (float.go:164-254). Emits same-value / RLE /
+ * gorilla / compressedNull; the snappy branch and the MLF path are
+ * replaced by compressedNull (raw), which every reader accepts
+ * (float.go:139 case 0). */
 static int64_t w_float_encode(const double *src, int64_t n, uint8_t *dst,
                               int64_t cap) {
   int64_t in_bytes = n * 8;
@@ -336,12 +373,22 @@ static int64_t w_float_encode(const double *src, int64_t n, uint8_t *dst,
   };
   if (n <= 4) return raw_out(); /* float.go:168-171 */
   bool has_nan = false;
-  bool all_same = true;
+  int64_t distinct = 1;
+  uint64_t seen[8];
+  memcpy(&seen[0], &src[0], 8);
   for (int64_t i = 0; i < n; i++) {
     if (std::isnan(src[i])) has_nan = true;
-    if (i && src[i] != src[0]) all_same = false;
+    uint64_t b;
+    memcpy(&b, &src[i], 8);
+    bool found = false;
+    for (int64_t k = 0; k < distinct && k < 8; k++)
+      if (seen[k] == b) { found = true; break; }
+    if (!found && distinct < 9) {
+      if (distinct < 8) seen[distinct] = b;
+      distinct++;
+    }
   }
-  if (all_same && !has_nan) {
+  if (distinct == 1 && !has_nan) {
     /* same-value (compress.go:51-66): [4<<4][count u16][value f64] */
     if (n > 0xFFFF || cap < 1 + 2 + 8) return raw_out();
     dst[0] = 4 << 4;
@@ -349,6 +396,11 @@ static int64_t w_float_encode(const double *src, int64_t n, uint8_t *dst,
     dst[2] = (uint8_t)n;
     memcpy(dst + 3, &src[0], 8);
     return 11;
+  }
+  if (distinct <= 8 && !has_nan) { /* float.go:176-179 → RLE */
+    int64_t r = w_float_rle(src, n, dst, cap);
+    if (r > 0 && r <= in_bytes * 90 / 100) return r;
+    /* unprofitable RLE (no runs): fall through to gorilla/raw */
   }
   if (has_nan) return raw_out(); /* extremeData → snappy in the reference */
   /* clamp the zero-filled window: gorilla worst case ≈ 10 B/value + 64 */

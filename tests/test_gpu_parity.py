@@ -1401,3 +1401,29 @@ class TestGroupedFill:
         pop = dense[dense["count"] > 0]
         assert np.array_equal(pop.view(np.uint8).reshape(len(pop), -1),
                               compact.view(np.uint8).reshape(len(compact), -1))
+
+
+class TestOverTimeAsync:
+    def test_over_time_pipelined_equals_sync(self):
+        blob, descs = orc.gen_shard(2301, 200, 1000)
+        import opengemini_amd as gx
+        from opengemini_amd.engine import OT_FUNCS
+
+        sh = gx.Shard(blob, descs, F)
+        S = 10**9
+        try:
+            for fn in ("sum", "avg", "max"):
+                ref, _ = sh.prom_over_time(0, 999 * S, 300 * S, 60 * S, fn)
+                ref = ref.copy()
+                b0 = sh.prom_rate_begin(0, 999 * S, 300 * S, 60 * S,
+                                        func=OT_FUNCS[fn], buf_id=0)
+                b1 = sh.prom_rate_begin(0, 999 * S, 300 * S, 60 * S,
+                                        func=OT_FUNCS[fn], buf_id=1)
+                r0, _ = sh.prom_rate_finish(b0)
+                r1, _ = sh.prom_rate_finish(b1)
+                for r in (r0, r1):
+                    assert len(r) == len(ref)
+                    assert np.array_equal(r["value"].view(np.uint64),
+                                          ref["value"].view(np.uint64)), fn
+        finally:
+            sh.close()

@@ -310,3 +310,45 @@ class TestCrossEncoderParity:
             assert np.array_equal(a[f].view(np.uint64),
                                   b[f].view(np.uint64)), f
         assert np.allclose(a["sum"], b["sum"], rtol=1e-12, atol=0)
+
+
+class TestRLEEmission:
+    def test_piecewise_constant_rle(self):
+        # few distinct values in runs → RLE (tag 5), as the reference
+        # selects for distinct <= 8 (float.go:176)
+        vals = np.repeat([1.5, 0.0, -2.25, 0.0], [200, 150, 100, 50])
+        n = len(vals)
+        sids = np.full(n, 1, dtype=np.uint64)
+        times = np.arange(n, dtype=np.int64) * 10 ** 9
+        blob, descs, rows = roundtrip(F, sids, times, vals)
+        expect_check(rows, sids, times, vals, None, F)
+        assert seg_tags(blob, descs)[0] == (33, 5)
+
+    def test_negative_zero_bits_survive(self):
+        vals = np.array([0.0, -0.0] * 100 + [1.0] * 300)
+        n = len(vals)
+        sids = np.full(n, 1, dtype=np.uint64)
+        times = np.arange(n, dtype=np.int64) * 10 ** 9
+        from opengemini_amd import engine as gxe2
+        blob, descs = gxe2.encode_shard(F, sids, times, vals)
+        import binding as orc2
+        # decode through the oracle's segment decoder and compare BITS
+        rows = orc2.scan_agg(blob, np.ascontiguousarray(descs), F,
+                             MIN_I, MAX_I, 0)
+        assert rows[0]["count"] == n
+        # first value must still be +0.0 and the min -0.0-aware compare
+        # unchanged; stronger: full first/last bit checks
+        assert np.array(rows[0]["first"]).view(np.uint64) == 0
+        assert rows[0]["last"] == 1.0
+
+    def test_few_distinct_no_runs_falls_back(self):
+        # 8 distinct values alternating every row: RLE would be 10 B/run
+        # of length 1 → unprofitable → gorilla/raw fallback, still decodes
+        rng = np.random.default_rng(21)
+        pal = rng.normal(0, 1, 8)
+        vals = pal[np.arange(600) % 8]
+        sids = np.full(600, 1, dtype=np.uint64)
+        times = np.arange(600, dtype=np.int64) * 10 ** 9
+        blob, descs, rows = roundtrip(F, sids, times, vals)
+        expect_check(rows, sids, times, vals, None, F)
+        assert seg_tags(blob, descs)[0][1] != 5
