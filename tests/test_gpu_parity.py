@@ -1427,3 +1427,35 @@ class TestOverTimeAsync:
                                           ref["value"].view(np.uint64)), fn
         finally:
             sh.close()
+
+
+class TestDegenerateShards:
+    def test_single_row_shard(self):
+        import opengemini_amd as gx
+        blob, descs = gx.encode_shard(
+            F, np.array([7], dtype=np.uint64),
+            np.array([5 * 10**9], dtype=np.int64), np.array([2.5]))
+        sh = gpu_shard(blob, np.ascontiguousarray(descs), F)
+        try:
+            rows, _ = sh.scan_agg(0, 2**62, INT)
+            assert len(rows) == 1
+            assert rows[0]["count"] == 1 and rows[0]["first"] == 2.5
+            g, _ = sh.scan_agg(0, 2**62, INT, group_all=True)
+            assert len(g) == 1 and g[0]["count"] == 1
+            pre, st = sh.scan_preagg(0, 2**62)
+            assert len(pre) == 1 and st["meta_rows"] == 1
+        finally:
+            sh.close()
+
+    def test_range_excludes_everything(self):
+        blob, descs = orc.gen_shard(2401, 10, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            for group_all in (False, True):
+                rows, _ = sh.scan_agg(-10**18, -10**17, INT,
+                                      group_all=group_all)
+                assert len(rows) == 0
+            r, _ = sh.prom_rate(-10**18, -10**17, 300 * 10**9, 60 * 10**9)
+            assert len(r) == 0
+        finally:
+            sh.close()
