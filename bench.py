@@ -100,7 +100,8 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--mode", choices=["walk", "random"], default="walk")
+    ap.add_argument("--mode", choices=["walk", "random", "int"],
+                    default="walk")
     ap.add_argument("--query",
                     choices=["mean", "downsample", "rate", "tags", "preagg"],
                     default="mean",
@@ -139,7 +140,13 @@ def main():
         torch.cuda.set_device(local_rank)
         dist.init_process_group("nccl")
 
-    gen_mode = orc.GEN_FLOAT_WALK if args.mode == "walk" else orc.GEN_FLOAT_RANDOM
+    if args.mode == "int" and args.query == "rate":
+        log("rate needs a float column (prom path); use --mode walk/random")
+        sys.exit(2)
+    gen_mode = {"walk": orc.GEN_FLOAT_WALK, "random": orc.GEN_FLOAT_RANDOM,
+                "int": orc.GEN_INT_SMALL}[args.mode]
+    col_type = (gx.engine.GEMX_TYPE_INT if args.mode == "int"
+                else gx.engine.GEMX_TYPE_FLOAT)
 
     # each rank authors its own shard: same shape, disjoint series
     t0 = time.time()
@@ -149,7 +156,7 @@ def main():
         f"{len(blob)/1e6:.0f} MB ({len(blob)/(args.series*args.pts):.2f} B/pt)")
 
     t0 = time.time()
-    shard = gx.Shard(blob, descs, gx.engine.GEMX_TYPE_FLOAT, device=local_rank)
+    shard = gx.Shard(blob, descs, col_type, device=local_rank)
     log(f"[attach] H2D resident in {time.time()-t0:.1f}s")
 
     NGROUPS = 1000
@@ -293,7 +300,8 @@ def main():
         traffic = tf.get("bytes_per_launch")
 
     cpu_baseline = None
-    if not args.skip_cpu_baseline and args.query == "mean" and n_gpus == 1:
+    if (not args.skip_cpu_baseline and args.query == "mean" and n_gpus == 1
+            and args.mode != "int"):
         # contract: cpu_baseline on rank 0 at N=1 only
         cpu_baseline = cpu_baseline_leg(gen_mode, args.mode)
 
@@ -308,7 +316,7 @@ def main():
         "higher_is_better": True,
         "scaling": "weak",
         "vs_baseline": None,  # no published number
-        "dtype": "f64",
+        "dtype": "i64" if args.mode == "int" else "f64",
         "data": "synthetic",
         "config": {
             "workload": {
