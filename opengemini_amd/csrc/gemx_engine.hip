@@ -2696,12 +2696,12 @@ struct TagPlan {
 };
 
 static void free_tag_plan(TagPlan &p) {
-  if (p.d_order) hipFree(p.d_order);
-  if (p.d_chunks) hipFree(p.d_chunks);
-  if (p.d_cstart) hipFree(p.d_cstart);
-  if (p.d_ccount) hipFree(p.d_ccount);
-  if (p.d_gtmp) hipFree(p.d_gtmp);
-  if (p.d_rows) hipFree(p.d_rows);
+  if (p.d_order) (void)hipFree(p.d_order);
+  if (p.d_chunks) (void)hipFree(p.d_chunks);
+  if (p.d_cstart) (void)hipFree(p.d_cstart);
+  if (p.d_ccount) (void)hipFree(p.d_ccount);
+  if (p.d_gtmp) (void)hipFree(p.d_gtmp);
+  if (p.d_rows) (void)hipFree(p.d_rows);
   p = TagPlan();
 }
 
@@ -2724,15 +2724,15 @@ struct RatePlan {
 };
 
 static void free_rate_plan(RatePlan &p) {
-  if (p.d_rsegq) hipFree(p.d_rsegq);
-  if (p.d_rsq) hipFree(p.d_rsq);
-  if (p.d_rpart) hipFree(p.d_rpart);
+  if (p.d_rsegq) (void)hipFree(p.d_rsegq);
+  if (p.d_rsq) (void)hipFree(p.d_rsq);
+  if (p.d_rpart) (void)hipFree(p.d_rpart);
   for (int i = 0; i < 2; i++) {
-    if (p.d_rrows2[i]) hipFree(p.d_rrows2[i]);
-    if (p.d_err2[i]) hipFree(p.d_err2[i]);
+    if (p.d_rrows2[i]) (void)hipFree(p.d_rrows2[i]);
+    if (p.d_err2[i]) (void)hipFree(p.d_err2[i]);
   }
-  if (p.h_rrows) hipHostFree(p.h_rrows);
-  if (p.d_scratch) hipFree(p.d_scratch);
+  if (p.h_rrows) (void)hipHostFree(p.h_rrows);
+  if (p.d_scratch) (void)hipFree(p.d_scratch);
   p = RatePlan();
 }
 
@@ -2816,22 +2816,22 @@ struct gemx_shard {
 };
 
 static void free_plan(QueryPlan &p) {
-  if (p.d_segq) hipFree(p.d_segq);
-  if (p.d_sq) hipFree(p.d_sq);
-  if (p.d_part) hipFree(p.d_part);
+  if (p.d_segq) (void)hipFree(p.d_segq);
+  if (p.d_sq) (void)hipFree(p.d_sq);
+  if (p.d_part) (void)hipFree(p.d_part);
   for (int i = 0; i < 2; i++) {
-    if (p.d_rows2[i]) hipFree(p.d_rows2[i]);
-    if (p.d_err2[i]) hipFree(p.d_err2[i]);
-    if (p.d_grows2[i]) hipFree(p.d_grows2[i]);
+    if (p.d_rows2[i]) (void)hipFree(p.d_rows2[i]);
+    if (p.d_err2[i]) (void)hipFree(p.d_err2[i]);
+    if (p.d_grows2[i]) (void)hipFree(p.d_grows2[i]);
   }
-  if (p.d_scratch) hipFree(p.d_scratch);
-  if (p.h_rows) hipHostFree(p.h_rows);
-  if (p.h_grows) hipHostFree(p.h_grows);
-  if (p.d_gtmp) hipFree(p.d_gtmp);
-  if (p.d_fast_q) hipFree(p.d_fast_q);
-  if (p.d_gen_q) hipFree(p.d_gen_q);
-  if (p.d_fastg_q) hipFree(p.d_fastg_q);
-  if (p.d_fasts_q) hipFree(p.d_fasts_q);
+  if (p.d_scratch) (void)hipFree(p.d_scratch);
+  if (p.h_rows) (void)hipHostFree(p.h_rows);
+  if (p.h_grows) (void)hipHostFree(p.h_grows);
+  if (p.d_gtmp) (void)hipFree(p.d_gtmp);
+  if (p.d_fast_q) (void)hipFree(p.d_fast_q);
+  if (p.d_gen_q) (void)hipFree(p.d_gen_q);
+  if (p.d_fastg_q) (void)hipFree(p.d_fastg_q);
+  if (p.d_fasts_q) (void)hipFree(p.d_fasts_q);
   p = QueryPlan();
 }
 
@@ -3044,36 +3044,36 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
 
 extern "C" int gemx_shard_close(gemx_shard *s) {
   if (!s) return GEMX_OK;
-  hipSetDevice(s->device);
+  (void)hipSetDevice(s->device);
   /* drain in-flight async queries so frees don't race the copy stream */
-  hipStreamSynchronize(s->stream);
-  if (s->copy_stream) hipStreamSynchronize(s->copy_stream);
+  (void)hipStreamSynchronize(s->stream);
+  if (s->copy_stream) (void)hipStreamSynchronize(s->copy_stream);
   s->pend_count = 0;
   s->rpend_count = 0;
   free_plan(s->plan);
   free_plan(s->sub_plan);
   free_tag_plan(s->tag_plan);
   free_rate_plan(s->rate_plan);
-  hipFree(s->d_blob);
-  hipFree(s->d_descs);
-  hipFree(s->d_fast_ids);
-  hipFree(s->d_fast_grid_ids);
-  hipFree(s->d_fast_stream_ids);
-  hipFree(s->d_general_ids);
-  if (s->d_row_base) hipFree(s->d_row_base);
-  if (s->d_xbm) hipFree(s->d_xbm);
-  if (s->d_xscratch) hipFree(s->d_xscratch);
-  hipStreamDestroy(s->stream);
-  if (s->copy_stream) hipStreamDestroy(s->copy_stream);
+  (void)hipFree(s->d_blob);
+  (void)hipFree(s->d_descs);
+  (void)hipFree(s->d_fast_ids);
+  (void)hipFree(s->d_fast_grid_ids);
+  (void)hipFree(s->d_fast_stream_ids);
+  (void)hipFree(s->d_general_ids);
+  if (s->d_row_base) (void)hipFree(s->d_row_base);
+  if (s->d_xbm) (void)hipFree(s->d_xbm);
+  if (s->d_xscratch) (void)hipFree(s->d_xscratch);
+  (void)hipStreamDestroy(s->stream);
+  if (s->copy_stream) (void)hipStreamDestroy(s->copy_stream);
   for (int sl = 0; sl < 2; sl++) {
     for (int e = 0; e < 3; e++) {
-      hipEventDestroy(s->ev_q[sl][e]);
-      hipEventDestroy(s->ev_r[sl][e]);
+      (void)hipEventDestroy(s->ev_q[sl][e]);
+      (void)hipEventDestroy(s->ev_r[sl][e]);
     }
-    hipEventDestroy(s->ev_copy[sl]);
-    hipEventDestroy(s->ev_rcopy[sl]);
-    if (s->h_err2[sl]) hipHostFree(s->h_err2[sl]);
-    if (s->h_rerr2[sl]) hipHostFree(s->h_rerr2[sl]);
+    (void)hipEventDestroy(s->ev_copy[sl]);
+    (void)hipEventDestroy(s->ev_rcopy[sl]);
+    if (s->h_err2[sl]) (void)hipHostFree(s->h_err2[sl]);
+    if (s->h_rerr2[sl]) (void)hipHostFree(s->h_rerr2[sl]);
   }
   delete s;
   return GEMX_OK;
@@ -3583,9 +3583,9 @@ static int scan_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
   auto t_sync1 = std::chrono::steady_clock::now();
 
   float ms_scan = 0, ms_merge = 0, ms_total = 0;
-  hipEventElapsedTime(&ms_scan, s->ev_q[slot][0], s->ev_q[slot][1]);
-  hipEventElapsedTime(&ms_merge, s->ev_q[slot][1], s->ev_q[slot][2]);
-  hipEventElapsedTime(&ms_total, s->ev_q[slot][0], s->ev_q[slot][2]);
+  (void)hipEventElapsedTime(&ms_scan, s->ev_q[slot][0], s->ev_q[slot][1]);
+  (void)hipEventElapsedTime(&ms_merge, s->ev_q[slot][1], s->ev_q[slot][2]);
+  (void)hipEventElapsedTime(&ms_total, s->ev_q[slot][0], s->ev_q[slot][2]);
 
   const DevErr herr = *s->h_err2[slot];
   if (herr.code != 0) {
@@ -3868,9 +3868,9 @@ static int rate_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
                         gemx_query_stats *stats) {
   HIP_CHECK(hipEventSynchronize(s->ev_rcopy[slot]));
   float ms_scan = 0, ms_merge = 0, ms_total = 0;
-  hipEventElapsedTime(&ms_scan, s->ev_r[slot][0], s->ev_r[slot][1]);
-  hipEventElapsedTime(&ms_merge, s->ev_r[slot][1], s->ev_r[slot][2]);
-  hipEventElapsedTime(&ms_total, s->ev_r[slot][0], s->ev_r[slot][2]);
+  (void)hipEventElapsedTime(&ms_scan, s->ev_r[slot][0], s->ev_r[slot][1]);
+  (void)hipEventElapsedTime(&ms_merge, s->ev_r[slot][1], s->ev_r[slot][2]);
+  (void)hipEventElapsedTime(&ms_total, s->ev_r[slot][0], s->ev_r[slot][2]);
   const DevErr herr = *s->h_rerr2[slot];
   if (herr.code != 0) {
     seterr(herr.code == GEMX_E_UNSUPPORTED ? "unsupported codec on device"
@@ -4071,13 +4071,13 @@ extern "C" int gemx_scan_agg_xfield(gemx_shard *vs, gemx_shard *fs,
     rc = xfield_eval_into(vs, fs, filter_op, filter_f, filter_i, vs->d_xbm,
                           d_etmp);
     if (rc) {
-      hipFree(d_etmp);
+      (void)hipFree(d_etmp);
       return rc;
     }
     HIP_CHECK(hipMemcpyAsync(&htmp, d_etmp, sizeof(DevErr),
                              hipMemcpyDeviceToHost, vs->stream));
     HIP_CHECK(hipStreamSynchronize(vs->stream));
-    hipFree(d_etmp);
+    (void)hipFree(d_etmp);
     if (htmp.code != 0) {
       seterr("filter column decode failed on device");
       return htmp.code;
@@ -4164,8 +4164,8 @@ extern "C" int gemx_scan_agg_cnf(gemx_shard *vs, const gemx_cond *conds,
                              hipMemcpyDeviceToHost, vs->stream));
     HIP_CHECK(hipStreamSynchronize(vs->stream));
   }
-  hipFree(d_grp);
-  hipFree(d_etmp);
+  (void)hipFree(d_grp);
+  (void)hipFree(d_etmp);
   if (err_rc) return err_rc;
   if (htmp.code != 0) {
     seterr("filter column decode failed on device");
@@ -4253,7 +4253,7 @@ extern "C" int gemx_host_register(void *p, uint64_t bytes) {
 
 extern "C" int gemx_host_unregister(void *p) {
   if (!p) return GEMX_E_INVALID;
-  hipHostUnregister(p);
+  (void)hipHostUnregister(p);
   return GEMX_OK;
 }
 
