@@ -326,14 +326,22 @@ int gemx_prom_irate(gemx_shard *, int64_t start_time, int64_t end_time,
 
 /* xxx_over_time family (engine/prom_functions.go: sum_over_time:232,
  * count_over_time:222, avg_over_time:341 Kahan streaming mean,
- * min/max_over_time:300-339 NaN-aware, last_over_time:528): func selects
- * the reducer. Same sampling grid and window semantics as gemx_prom_rate. */
+ * min/max_over_time:300-339 NaN-aware, last_over_time:528,
+ * stdvar/stddev_over_time:516-573 sequential Kahan-Welford,
+ * present_over_time:577): func selects the reducer. Same sampling grid
+ * and window semantics as gemx_prom_rate. Windows spanning a segment
+ * boundary combine stdvar states with Chan's parallel formula (1e-9
+ * relative vs the reference's sequential stream; single-segment windows
+ * are bit-exact). */
 #define GEMX_PF_SUM_OT 2
 #define GEMX_PF_COUNT_OT 3
 #define GEMX_PF_AVG_OT 4
 #define GEMX_PF_MIN_OT 5
 #define GEMX_PF_MAX_OT 6
 #define GEMX_PF_LAST_OT 7
+#define GEMX_PF_STDVAR_OT 8
+#define GEMX_PF_STDDEV_OT 9
+#define GEMX_PF_PRESENT_OT 10
 int gemx_prom_over_time(gemx_shard *, int64_t start_time, int64_t end_time,
                         int64_t range_ns, int64_t step_ns, int func,
                         gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,

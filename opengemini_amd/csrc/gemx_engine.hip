@@ -2146,6 +2146,9 @@ struct RateSeriesQ {
 #define GEMX_PF_MIN_OT 5
 #define GEMX_PF_MAX_OT 6
 #define GEMX_PF_LAST_OT 7
+#define GEMX_PF_STDVAR_OT 8
+#define GEMX_PF_STDDEV_OT 9
+#define GEMX_PF_PRESENT_OT 10
 
 __device__ __forceinline__ void d_kahan_inc(double inc, double &sum, double &c) {
   /* executor.KahanSumInc */
@@ -2183,7 +2186,16 @@ __device__ __forceinline__ void ot_slot_update(RateSlot *s, int64_t t, double v,
     if (p->count == 0 || v > p->first_v || isnan(p->first_v)) p->first_v = v;
   } else if (FUNC == GEMX_PF_LAST_OT) {
     p->first_v = v;
-  }
+  } else if (FUNC == GEMX_PF_STDVAR_OT || FUNC == GEMX_PF_STDDEV_OT) {
+    /* sequential Kahan-Welford exactly as floatStdVarOverTimeMerger
+     * (prom_functions.go:530-556): mean in (first_v,last_v), M2 aux in
+     * (prev_v, reset_adj) */
+    double c = (double)(p->count + 1);
+    double delta = v - (p->first_v + p->last_v);
+    d_kahan_inc(delta / c, p->first_v, p->last_v);
+    d_kahan_inc(delta * (v - (p->first_v + p->last_v)), p->prev_v,
+                p->reset_adj);
+  } /* PRESENT: count alone */
   p->count++;
 }
 
@@ -2243,6 +2255,7 @@ __global__ void __launch_bounds__(256) k_rate_scan(
       ring[j].p.reset_adj = 0;
       ring[j].p.first_v = 0;
       ring[j].p.last_v = 0;
+      ring[j].p.prev_v = 0; /* stdvar aux (M2) accumulator */
     }
 
     /* value/time iterators (fast: streaming; general: via scratch) */
@@ -2437,6 +2450,8 @@ __global__ void __launch_bounds__(256) k_rate_scan(
                 sl->p.reset_adj = 0;
                 sl->p.first_v = 0;
                 sl->p.last_v = 0;
+                sl->p.prev_v = 0; /* stdvar M2 — stale state here doubles
+                                     later windows' variance */
               } else {
                 sl->ts = INT64_MIN;
               }
@@ -2511,6 +2526,28 @@ __global__ void __launch_bounds__(256) k_rate_merge(
       if (o < q.s0 || o >= q.s0 + (int64_t)q.n_steps) continue;
       RatePartial p = partials[q.partial_base + (o - q.s0)];
       if (p.count == 0) continue;
+      if (func == GEMX_PF_STDVAR_OT || func == GEMX_PF_STDDEV_OT) {
+        /* windows inside one segment reproduce the reference's
+         * sequential stream bit-exactly; across segment boundaries the
+         * Welford states combine with Chan's parallel formula (documented
+         * 1e-9-relative deviation — the sequential order cannot be
+         * reconstructed from per-segment states) */
+        double pm = p.first_v + p.last_v;
+        double pM2 = p.prev_v + p.reset_adj;
+        if (acc.count == 0) {
+          acc.first_v = pm;
+          acc.prev_v = pM2;
+          acc.count = p.count;
+        } else {
+          double na = (double)acc.count, nb = (double)p.count;
+          double n = na + nb;
+          double d2 = pm - acc.first_v;
+          acc.prev_v += pM2 + d2 * d2 * na * nb / n;
+          acc.first_v += d2 * nb / n;
+          acc.count += p.count;
+        }
+        continue;
+      }
       if (func >= GEMX_PF_SUM_OT) {
         /* per-record tail first (reduce returns sum+c / mean+c, then
          * the merge funcs combine those values — prom_functions.go) */
@@ -2575,7 +2612,16 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     if (func >= GEMX_PF_SUM_OT) {
       if (acc.count > 0) {
         /* tails were applied per partial at merge time */
-        out.value = (func == GEMX_PF_COUNT_OT) ? (double)acc.count : acc.first_v;
+        if (func == GEMX_PF_COUNT_OT)
+          out.value = (double)acc.count;
+        else if (func == GEMX_PF_PRESENT_OT)
+          out.value = 1.0;
+        else if (func == GEMX_PF_STDVAR_OT)
+          out.value = acc.prev_v / (double)acc.count;
+        else if (func == GEMX_PF_STDDEV_OT)
+          out.value = sqrt(acc.prev_v / (double)acc.count);
+        else
+          out.value = acc.first_v;
         out.isnil = 0;
       }
       if (out.isnil) atomicAdd(&err->gaps, 1ull);
@@ -3803,6 +3849,9 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_MIN_OT: LAUNCH_RATE_FAST(GEMX_PF_MIN_OT); break;
     case GEMX_PF_MAX_OT: LAUNCH_RATE_FAST(GEMX_PF_MAX_OT); break;
     case GEMX_PF_LAST_OT: LAUNCH_RATE_FAST(GEMX_PF_LAST_OT); break;
+    case GEMX_PF_STDVAR_OT: LAUNCH_RATE_FAST(GEMX_PF_STDVAR_OT); break;
+    case GEMX_PF_STDDEV_OT: LAUNCH_RATE_FAST(GEMX_PF_STDDEV_OT); break;
+    case GEMX_PF_PRESENT_OT: LAUNCH_RATE_FAST(GEMX_PF_PRESENT_OT); break;
     default: LAUNCH_RATE_FAST(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_FAST
@@ -3823,6 +3872,9 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_MIN_OT: LAUNCH_RATE_GEN(GEMX_PF_MIN_OT); break;
     case GEMX_PF_MAX_OT: LAUNCH_RATE_GEN(GEMX_PF_MAX_OT); break;
     case GEMX_PF_LAST_OT: LAUNCH_RATE_GEN(GEMX_PF_LAST_OT); break;
+    case GEMX_PF_STDVAR_OT: LAUNCH_RATE_GEN(GEMX_PF_STDVAR_OT); break;
+    case GEMX_PF_STDDEV_OT: LAUNCH_RATE_GEN(GEMX_PF_STDDEV_OT); break;
+    case GEMX_PF_PRESENT_OT: LAUNCH_RATE_GEN(GEMX_PF_PRESENT_OT); break;
     default: LAUNCH_RATE_GEN(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_GEN
@@ -4407,7 +4459,7 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
                                    int64_t step_ns, int func,
                                    gemx_rate_row *out_host, uint64_t cap,
                                    uint64_t *n_out, gemx_query_stats *stats) {
-  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_LAST_OT) {
+  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_PRESENT_OT) {
     seterr("unknown over_time func");
     return GEMX_E_INVALID;
   }

@@ -1158,7 +1158,7 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
                            int64_t start, int64_t end, int64_t range_ns,
                            int64_t step_ns, int func, orc_rate_row *out,
                            int64_t cap) {
-  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 7) return -1;
+  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 10) return -1;
   int64_t start_sample = start + range_ns;
   int64_t end_sample =
       (step_ns == 0) ? start_sample
@@ -1248,6 +1248,23 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
             if (sv[k] > v || isnan(v)) v = sv[k];
           break;
         }
+        case 8:   /* stdvar_over_time (prom_functions.go:523-564) */
+        case 9: { /* stddev_over_time: sqrt of the same */
+          double count = 0, mean = 0, cMean = 0, aux = 0, cAux = 0;
+          for (int64_t k = pi; k < pj; k++) {
+            double fv = sv[k];
+            count++;
+            double delta = fv - (mean + cMean);
+            kahan_inc(delta / count, &mean, &cMean);
+            kahan_inc(delta * (fv - (mean + cMean)), &aux, &cAux);
+          }
+          v = (aux + cAux) / count;
+          if (func == 9) v = sqrt(v);
+          break;
+        }
+        case 10: /* present_over_time (intervalExistMark) */
+          v = 1.0;
+          break;
         default:
           v = sv[pj - 1];
           break;

@@ -378,7 +378,8 @@ def prom_rate(blob, descs, start, end, range_ns, step_ns, is_rate=True, is_count
     return out[:n].copy()
 
 
-OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6, "last": 7}
+OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6,
+            "last": 7, "stdvar": 8, "stddev": 9, "present": 10}
 
 
 def prom_over_time(blob, descs, start, end, range_ns, step_ns, func, cap=None):
@@ -391,7 +392,7 @@ def prom_over_time(blob, descs, start, end, range_ns, step_ns, func, cap=None):
     out = np.zeros(cap, dtype=RATE_ROW_DTYPE)
     n = lib.orc_prom_over_time(
         _u8(bts), len(bts), d.ctypes.data_as(C.c_void_p), len(d),
-        start, end, range_ns, step_ns, OT_FUNCS[func],
+        start, end, range_ns, step_ns, OT_FUNCS[func] if isinstance(func, str) else int(func),
         out.ctypes.data_as(C.c_void_p), cap,
     )
     if n < 0:

@@ -239,7 +239,9 @@ int64_t orc_prom_irate(const uint8_t *blob, int64_t blob_len,
                        int64_t end, int64_t range_ns, int64_t step_ns,
                        int is_rate, orc_rate_row *out, int64_t cap);
 
-/* *_over_time family (prom_functions.go:172-342): func 2 sum, 3 count,
+/* *_over_time family (prom_functions.go:172-342,516-600): func 2 sum, 3
+ * count, 8 stdvar, 9 stddev, 10 present (sequential Kahan-Welford for
+ * stdvar/stddev),
  * 4 avg, 5 min, 6 max, 7 last over [ts-range, ts] windows. */
 int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
                            const orc_seg_desc *descs, int64_t nsegs,

@@ -1491,3 +1491,44 @@ class TestFullScaleParity:
                                   gref[f].view(np.uint64)), f
         tol = 1e-9 * np.maximum(1.0, np.abs(gref["sum"]))
         assert np.all(np.abs(grp["sum"] - gref["sum"]) <= tol)
+
+
+class TestStdVarOverTimeGPU:
+    def test_stdvar_stddev_present_parity(self):
+        """Single-segment windows are bit-exact (same sequential
+        Kahan-Welford); the multi-segment build_shard case checks the
+        documented 1e-9 tolerance of the Chan state combination."""
+        S = 10**9
+        # uniform shard: every series one segment -> bit-exact
+        blob, descs = orc.gen_shard(2501, 150, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            for fn in ("stdvar", "stddev", "present"):
+                gpu, _ = sh.prom_over_time(0, 999 * S, 300 * S, 60 * S, fn)
+                gpu = gpu.copy()
+                ref = orc.prom_over_time(blob, descs, 0, 999 * S, 300 * S,
+                                         60 * S, fn)
+                assert len(gpu) == len(ref), fn
+                assert np.array_equal(gpu["sid"], ref["sid"])
+                assert np.array_equal(gpu["ts"], ref["ts"])
+                assert np.array_equal(gpu["value"].view(np.uint64),
+                                      ref["value"].view(np.uint64)), fn
+        finally:
+            sh.close()
+        # multi-segment series: windows span segment boundaries
+        rng = np.random.default_rng(2502)
+        blob2, descs2, _ = build_shard(rng, F, range(1, 41), null_frac=0.05)
+        sh2 = gpu_shard(blob2, descs2, F)
+        try:
+            for fn in ("stdvar", "stddev"):
+                gpu, _ = sh2.prom_over_time(0, 800 * S, 240 * S, 60 * S, fn)
+                gpu = gpu.copy()
+                ref = orc.prom_over_time(blob2, descs2, 0, 800 * S, 240 * S,
+                                         60 * S, fn)
+                assert len(gpu) == len(ref), fn
+                assert np.array_equal(gpu["ts"], ref["ts"])
+                ok = np.isclose(gpu["value"], ref["value"], rtol=1e-9,
+                                atol=1e-12)
+                assert np.all(ok), fn
+        finally:
+            sh2.close()

@@ -172,3 +172,41 @@ class TestOverTime:
             for ts, w in wins.items():
                 ref = float(npf(w))
                 assert abs(got[ts] - ref) <= 1e-9 * max(1.0, abs(ref)), (func, ts)
+
+
+class TestStdVarPresentOverTime:
+    """stdvar/stddev (sequential Kahan-Welford, prom_functions.go:516-573)
+    and present_over_time (:577) against an independent numpy restatement."""
+
+    def test_stdvar_matches_numpy(self):
+        import shard_helpers as sh
+        rng = np.random.default_rng(31)
+        blob, descs, truth = sh.build_shard(rng, sh.F, [1, 2, 3],
+                                            null_frac=0.1)
+        S = 10**9
+        start, end, rng_ns, step = 0, 900 * S, 300 * S, 60 * S
+        for func, code in (("stdvar", 8), ("stddev", 9)):
+            rows = orc.prom_over_time(blob, descs, start, end, rng_ns,
+                                          step, code)
+            for r in rows:
+                sid = int(r["sid"])
+                at, av, ax = truth[sid]
+                m = (at >= r["ts"] - rng_ns) & (at <= r["ts"]) & ax
+                vv = av[m]
+                vv = vv[~np.isnan(vv)]
+                assert len(vv) >= 1
+                # population variance (the reference divides by n)
+                exp = float(np.mean((vv - vv.mean()) ** 2))
+                if func == "stddev":
+                    exp = exp ** 0.5
+                assert abs(r["value"] - exp) <= 1e-9 * max(1.0, abs(exp))
+
+    def test_present_is_one_for_any_sample(self):
+        import shard_helpers as sh
+        rng = np.random.default_rng(32)
+        blob, descs, truth = sh.build_shard(rng, sh.F, [5], null_frac=0.0)
+        S = 10**9
+        rows = orc.prom_over_time(blob, descs, 0, 800 * S, 120 * S,
+                                      60 * S, 10)
+        assert len(rows) > 0
+        assert np.all(rows["value"] == 1.0)
