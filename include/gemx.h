@@ -342,6 +342,8 @@ int gemx_prom_irate(gemx_shard *, int64_t start_time, int64_t end_time,
 #define GEMX_PF_STDVAR_OT 8
 #define GEMX_PF_STDDEV_OT 9
 #define GEMX_PF_PRESENT_OT 10
+#define GEMX_PF_CHANGES_OT 11  /* changes_prom (CalcChange) */
+#define GEMX_PF_RESETS_OT 12   /* resets_prom (CalcResets) */
 int gemx_prom_over_time(gemx_shard *, int64_t start_time, int64_t end_time,
                         int64_t range_ns, int64_t step_ns, int func,
                         gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,

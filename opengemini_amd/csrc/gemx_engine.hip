@@ -2149,6 +2149,8 @@ struct RateSeriesQ {
 #define GEMX_PF_STDVAR_OT 8
 #define GEMX_PF_STDDEV_OT 9
 #define GEMX_PF_PRESENT_OT 10
+#define GEMX_PF_CHANGES_OT 11
+#define GEMX_PF_RESETS_OT 12
 
 __device__ __forceinline__ void d_kahan_inc(double inc, double &sum, double &c) {
   /* executor.KahanSumInc */
@@ -2195,6 +2197,18 @@ __device__ __forceinline__ void ot_slot_update(RateSlot *s, int64_t t, double v,
     d_kahan_inc(delta / c, p->first_v, p->last_v);
     d_kahan_inc(delta * (v - (p->first_v + p->last_v)), p->prev_v,
                 p->reset_adj);
+  } else if (FUNC == GEMX_PF_CHANGES_OT || FUNC == GEMX_PF_RESETS_OT) {
+    /* executor.CalcChange / CalcResets streamed: counter in first_v,
+     * window-partial's first value in prev_v, last value in reset_adj */
+    if (p->count == 0) {
+      p->prev_v = v;
+    } else {
+      bool hit = (FUNC == GEMX_PF_CHANGES_OT)
+                     ? (v != p->reset_adj && !(isnan(v) && isnan(p->reset_adj)))
+                     : (v < p->reset_adj);
+      if (hit) p->first_v += 1.0;
+    }
+    p->reset_adj = v;
   } /* PRESENT: count alone */
   p->count++;
 }
@@ -2526,6 +2540,25 @@ __global__ void __launch_bounds__(256) k_rate_merge(
       if (o < q.s0 || o >= q.s0 + (int64_t)q.n_steps) continue;
       RatePartial p = partials[q.partial_base + (o - q.s0)];
       if (p.count == 0) continue;
+      if (func == GEMX_PF_CHANGES_OT || func == GEMX_PF_RESETS_OT) {
+        if (acc.count == 0) {
+          acc.first_v = p.first_v;  /* counter */
+          acc.prev_v = p.prev_v;    /* first value */
+          acc.reset_adj = p.reset_adj; /* last value */
+          acc.count = p.count;
+        } else {
+          /* the boundary pair (A's last, B's first) counts by the same
+           * rule, then B's internal counter adds on */
+          double a = acc.reset_adj, bb = p.prev_v;
+          bool hit = (func == GEMX_PF_CHANGES_OT)
+                         ? (bb != a && !(isnan(a) && isnan(bb)))
+                         : (bb < a);
+          acc.first_v += p.first_v + (hit ? 1.0 : 0.0);
+          acc.reset_adj = p.reset_adj;
+          acc.count += p.count;
+        }
+        continue;
+      }
       if (func == GEMX_PF_STDVAR_OT || func == GEMX_PF_STDDEV_OT) {
         /* windows inside one segment reproduce the reference's
          * sequential stream bit-exactly; across segment boundaries the
@@ -2616,6 +2649,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
           out.value = (double)acc.count;
         else if (func == GEMX_PF_PRESENT_OT)
           out.value = 1.0;
+        else if (func == GEMX_PF_CHANGES_OT || func == GEMX_PF_RESETS_OT)
+          out.value = acc.first_v;
         else if (func == GEMX_PF_STDVAR_OT)
           out.value = acc.prev_v / (double)acc.count;
         else if (func == GEMX_PF_STDDEV_OT)
@@ -3852,6 +3887,8 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_STDVAR_OT: LAUNCH_RATE_FAST(GEMX_PF_STDVAR_OT); break;
     case GEMX_PF_STDDEV_OT: LAUNCH_RATE_FAST(GEMX_PF_STDDEV_OT); break;
     case GEMX_PF_PRESENT_OT: LAUNCH_RATE_FAST(GEMX_PF_PRESENT_OT); break;
+    case GEMX_PF_CHANGES_OT: LAUNCH_RATE_FAST(GEMX_PF_CHANGES_OT); break;
+    case GEMX_PF_RESETS_OT: LAUNCH_RATE_FAST(GEMX_PF_RESETS_OT); break;
     default: LAUNCH_RATE_FAST(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_FAST
@@ -3875,6 +3912,8 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_STDVAR_OT: LAUNCH_RATE_GEN(GEMX_PF_STDVAR_OT); break;
     case GEMX_PF_STDDEV_OT: LAUNCH_RATE_GEN(GEMX_PF_STDDEV_OT); break;
     case GEMX_PF_PRESENT_OT: LAUNCH_RATE_GEN(GEMX_PF_PRESENT_OT); break;
+    case GEMX_PF_CHANGES_OT: LAUNCH_RATE_GEN(GEMX_PF_CHANGES_OT); break;
+    case GEMX_PF_RESETS_OT: LAUNCH_RATE_GEN(GEMX_PF_RESETS_OT); break;
     default: LAUNCH_RATE_GEN(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_GEN
@@ -4459,7 +4498,7 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
                                    int64_t step_ns, int func,
                                    gemx_rate_row *out_host, uint64_t cap,
                                    uint64_t *n_out, gemx_query_stats *stats) {
-  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_PRESENT_OT) {
+  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_RESETS_OT) {
     seterr("unknown over_time func");
     return GEMX_E_INVALID;
   }

@@ -1158,7 +1158,7 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
                            int64_t start, int64_t end, int64_t range_ns,
                            int64_t step_ns, int func, orc_rate_row *out,
                            int64_t cap) {
-  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 10) return -1;
+  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 12) return -1;
   int64_t start_sample = start + range_ns;
   int64_t end_sample =
       (step_ns == 0) ? start_sample
@@ -1265,6 +1265,23 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
         case 10: /* present_over_time (intervalExistMark) */
           v = 1.0;
           break;
+        case 11: { /* changes (executor.CalcChange: consecutive pairs;
+                      a NaN->NaN pair is not a change) */
+          int64_t cc = 0;
+          for (int64_t k = pi + 1; k < pj; k++) {
+            double a = sv[k - 1], bb = sv[k];
+            if (bb != a && !(a != a && bb != bb)) cc++;
+          }
+          v = (double)cc;
+          break;
+        }
+        case 12: { /* resets (executor.CalcResets: count decreases) */
+          int64_t cc = 0;
+          for (int64_t k = pi + 1; k < pj; k++)
+            if (sv[k] < sv[k - 1]) cc++;
+          v = (double)cc;
+          break;
+        }
         default:
           v = sv[pj - 1];
           break;

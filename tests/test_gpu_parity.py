@@ -1532,3 +1532,34 @@ class TestStdVarOverTimeGPU:
                 assert np.all(ok), fn
         finally:
             sh2.close()
+
+    def test_changes_resets_parity(self):
+        S = 10**9
+        blob, descs = orc.gen_shard(2503, 150, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            for fn in ("changes", "resets"):
+                gpu, _ = sh.prom_over_time(0, 999 * S, 300 * S, 60 * S, fn)
+                gpu = gpu.copy()
+                ref = orc.prom_over_time(blob, descs, 0, 999 * S, 300 * S,
+                                         60 * S, fn)
+                assert len(gpu) == len(ref), fn
+                assert np.array_equal(gpu["value"], ref["value"]), fn
+        finally:
+            sh.close()
+        # multi-segment: the boundary pair between partials must count
+        rng = np.random.default_rng(2504)
+        blob2, descs2, _ = build_shard(
+            rng, F, range(1, 31), null_frac=0.1,
+            value_fn=lambda r, n: r.integers(0, 3, n).astype(float))
+        sh2 = gpu_shard(blob2, descs2, F)
+        try:
+            for fn in ("changes", "resets"):
+                gpu, _ = sh2.prom_over_time(0, 800 * S, 240 * S, 60 * S, fn)
+                gpu = gpu.copy()
+                ref = orc.prom_over_time(blob2, descs2, 0, 800 * S, 240 * S,
+                                         60 * S, fn)
+                assert len(gpu) == len(ref), fn
+                assert np.array_equal(gpu["value"], ref["value"]), fn
+        finally:
+            sh2.close()

@@ -210,3 +210,26 @@ class TestStdVarPresentOverTime:
                                       60 * S, 10)
         assert len(rows) > 0
         assert np.all(rows["value"] == 1.0)
+
+    def test_changes_resets_match_numpy(self):
+        import shard_helpers as sh
+        rng = np.random.default_rng(33)
+        # integer-ish values so repeats and decreases are common
+        blob, descs, truth = sh.build_shard(
+            rng, sh.F, [1, 2], null_frac=0.1,
+            value_fn=lambda r, n: r.integers(0, 4, n).astype(float))
+        S = 10**9
+        for func, code in (("changes", 11), ("resets", 12)):
+            rows = orc.prom_over_time(blob, descs, 0, 700 * S, 180 * S,
+                                      60 * S, code)
+            assert len(rows) > 0
+            for r in rows:
+                at, av, ax = truth[int(r["sid"])]
+                m = (at >= r["ts"] - 180 * S) & (at <= r["ts"]) & ax
+                vv = av[m]
+                vv = vv[~np.isnan(vv)]
+                if func == "changes":
+                    exp = float(np.count_nonzero(vv[1:] != vv[:-1]))
+                else:
+                    exp = float(np.count_nonzero(vv[1:] < vv[:-1]))
+                assert r["value"] == exp, (func, r["ts"])
