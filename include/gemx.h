@@ -344,6 +344,17 @@ int gemx_prom_irate(gemx_shard *, int64_t start_time, int64_t end_time,
 #define GEMX_PF_PRESENT_OT 10
 #define GEMX_PF_CHANGES_OT 11  /* changes_prom (CalcChange) */
 #define GEMX_PF_RESETS_OT 12   /* resets_prom (CalcResets) */
+#define GEMX_PF_DERIV 13
+#define GEMX_PF_PREDICT 14
+
+/* deriv / predict_linear (prom_functions.go:358-436): Kahan-compensated
+ * least squares over the window's points with x relative to the sample
+ * time, the constY fast path, and for predict_linear value =
+ * slope*scalar + intercept (scalar = prediction horizon, seconds). */
+int gemx_prom_linear(gemx_shard *, int64_t start_time, int64_t end_time,
+                     int64_t range_ns, int64_t step_ns, int is_predict,
+                     double scalar, gemx_rate_row *out_host, uint64_t cap,
+                     uint64_t *n_out, gemx_query_stats *stats);
 int gemx_prom_over_time(gemx_shard *, int64_t start_time, int64_t end_time,
                         int64_t range_ns, int64_t step_ns, int func,
                         gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,

@@ -2096,7 +2096,24 @@ struct RatePartial {
   double prev_v;
   int64_t count;
   double reset_adj; /* Σ pre-reset values within this (segment, window) */
+  /* extra state for the linear-regression family (deriv/predict_linear,
+   * prom_functions.go:369-436): the OT funcs repurpose fields freely —
+   * linear uses first_v/last_v = sumX+c, prev_v/reset_adj = sumY+c,
+   * first_t/last_t bits = sumXY+c, prev_t bits = sumX2, aux0 = cX2,
+   * aux1 = fv (the window's first value), aux2 = constY flag */
+  double aux0, aux1, aux2;
 };
+
+__device__ __forceinline__ double d_bits_f(int64_t b) {
+  double x;
+  memcpy(&x, &b, 8);
+  return x;
+}
+__device__ __forceinline__ int64_t d_f_bits(double x) {
+  int64_t b;
+  memcpy(&b, &x, 8);
+  return b;
+}
 
 struct RateSlot {
   int64_t ts;     /* sample time; INT64_MIN = inactive */
@@ -2151,6 +2168,8 @@ struct RateSeriesQ {
 #define GEMX_PF_PRESENT_OT 10
 #define GEMX_PF_CHANGES_OT 11
 #define GEMX_PF_RESETS_OT 12
+#define GEMX_PF_DERIV 13
+#define GEMX_PF_PREDICT 14
 
 __device__ __forceinline__ void d_kahan_inc(double inc, double &sum, double &c) {
   /* executor.KahanSumInc */
@@ -2209,6 +2228,29 @@ __device__ __forceinline__ void ot_slot_update(RateSlot *s, int64_t t, double v,
       if (hit) p->first_v += 1.0;
     }
     p->reset_adj = v;
+  } else if (FUNC == GEMX_PF_DERIV || FUNC == GEMX_PF_PREDICT) {
+    /* linearMergeFunc (prom_functions.go:369-410): Kahan sums of
+     * x=(t-ts)/1e9, v, x*v, x*x; fv = first value; constY tracked.
+     * fp contract off: the muls must not fuse into the Kahan adds or
+     * the compensation drifts a ulp from the reference's x86 stream */
+#pragma clang fp contract(off)
+    if (p->count == 0) {
+      p->aux1 = v;
+      p->aux2 = 1.0;
+    } else if (v != p->aux1) {
+      p->aux2 = 0.0;
+    }
+    double x = (double)(t - s->ts) / 1e9;
+    d_kahan_inc(x, p->first_v, p->last_v);
+    d_kahan_inc(v, p->prev_v, p->reset_adj);
+    double sxy = d_bits_f(p->first_t), cxy = d_bits_f(p->last_t);
+    d_kahan_inc(x * v, sxy, cxy);
+    p->first_t = d_f_bits(sxy);
+    p->last_t = d_f_bits(cxy);
+    double sx2 = d_bits_f(p->prev_t), cx2 = p->aux0;
+    d_kahan_inc(x * x, sx2, cx2);
+    p->prev_t = d_f_bits(sx2);
+    p->aux0 = cx2;
   } /* PRESENT: count alone */
   p->count++;
 }
@@ -2464,8 +2506,13 @@ __global__ void __launch_bounds__(256) k_rate_scan(
                 sl->p.reset_adj = 0;
                 sl->p.first_v = 0;
                 sl->p.last_v = 0;
-                sl->p.prev_v = 0; /* stdvar M2 — stale state here doubles
-                                     later windows' variance */
+                sl->p.prev_v = 0; /* stdvar M2 etc. */
+                sl->p.first_t = 0;
+                sl->p.last_t = 0;
+                sl->p.prev_t = 0;
+                sl->p.aux0 = 0;
+                sl->p.aux1 = 0;
+                sl->p.aux2 = 0;
               } else {
                 sl->ts = INT64_MIN;
               }
@@ -2510,7 +2557,7 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     const RateSegQ *__restrict__ rsegq, const RatePartial *__restrict__ partials,
     gemx_rate_row *__restrict__ rows, uint64_t total_rows, int64_t start_sample,
     int64_t step_ns, int64_t range_ns, int is_rate, int is_counter, int func,
-    DevErr *__restrict__ err) {
+    double scalar, DevErr *__restrict__ err) {
   uint64_t gid = blockIdx.x * (uint64_t)blockDim.x + threadIdx.x;
   for (uint64_t r = gid; r < total_rows; r += gridDim.x * (uint64_t)blockDim.x) {
     uint32_t lo = 0, hi = nseries - 1;
@@ -2540,6 +2587,29 @@ __global__ void __launch_bounds__(256) k_rate_merge(
       if (o < q.s0 || o >= q.s0 + (int64_t)q.n_steps) continue;
       RatePartial p = partials[q.partial_base + (o - q.s0)];
       if (p.count == 0) continue;
+      if (func == GEMX_PF_DERIV || func == GEMX_PF_PREDICT) {
+        double sx = p.first_v + p.last_v;
+        double sy = p.prev_v + p.reset_adj;
+        double sxy = d_bits_f(p.first_t) + d_bits_f(p.last_t);
+        double sx2 = d_bits_f(p.prev_t) + p.aux0;
+        if (acc.count == 0) {
+          acc.first_v = sx;
+          acc.prev_v = sy;
+          acc.first_t = d_f_bits(sxy);
+          acc.prev_t = d_f_bits(sx2);
+          acc.aux1 = p.aux1;
+          acc.aux2 = p.aux2;
+          acc.count = p.count;
+        } else {
+          acc.first_v += sx;
+          acc.prev_v += sy;
+          acc.first_t = d_f_bits(d_bits_f(acc.first_t) + sxy);
+          acc.prev_t = d_f_bits(d_bits_f(acc.prev_t) + sx2);
+          if (!(p.aux2 > 0.0) || p.aux1 != acc.aux1) acc.aux2 = 0.0;
+          acc.count += p.count;
+        }
+        continue;
+      }
       if (func == GEMX_PF_CHANGES_OT || func == GEMX_PF_RESETS_OT) {
         if (acc.count == 0) {
           acc.first_v = p.first_v;  /* counter */
@@ -2643,6 +2713,7 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     out.isnil = 1;
     memset(out._pad, 0, sizeof(out._pad));
     if (func >= GEMX_PF_SUM_OT) {
+      bool lin_nil = false;
       if (acc.count > 0) {
         /* tails were applied per partial at merge time */
         if (func == GEMX_PF_COUNT_OT)
@@ -2651,6 +2722,30 @@ __global__ void __launch_bounds__(256) k_rate_merge(
           out.value = 1.0;
         else if (func == GEMX_PF_CHANGES_OT || func == GEMX_PF_RESETS_OT)
           out.value = acc.first_v;
+        else if (func == GEMX_PF_DERIV || func == GEMX_PF_PREDICT) {
+#pragma clang fp contract(off)
+          if (acc.count <= 1) {
+            lin_nil = true; /* pointCount <= 1: no slope */
+          } else if (acc.aux2 > 0.0) {
+            /* constY fast path (prom_functions.go:411-420) */
+            if (isinf(acc.aux1))
+              out.value = nan("");
+            else
+              out.value = (func == GEMX_PF_DERIV) ? 0.0 : acc.aux1;
+          } else {
+            double n = (double)acc.count;
+            double sumX = acc.first_v, sumY = acc.prev_v;
+            double sumXY = d_bits_f(acc.first_t);
+            double sumX2 = d_bits_f(acc.prev_t);
+            double covXY = sumXY - sumX * sumY / n;
+            double varX = sumX2 - sumX * sumX / n;
+            double dv = covXY / varX;
+            if (func == GEMX_PF_DERIV)
+              out.value = dv;
+            else
+              out.value = dv * scalar + (sumY / n - dv * sumX / n);
+          }
+        }
         else if (func == GEMX_PF_STDVAR_OT)
           out.value = acc.prev_v / (double)acc.count;
         else if (func == GEMX_PF_STDDEV_OT)
@@ -3734,7 +3829,8 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                           int64_t range_ns, int64_t step_ns, int is_rate,
                           int is_counter, int func, gemx_rate_row *out_host,
                           uint64_t cap, uint64_t *n_out,
-                          gemx_query_stats *stats, int async_begin = 0) {
+                          gemx_query_stats *stats, int async_begin = 0,
+                          double scalar = 0.0) {
   if (!s) return GEMX_E_INVALID;
   if (async_begin && s->rpend_count >= 2) {
     seterr("two rate queries already in flight: call gemx_prom_finish");
@@ -3889,6 +3985,8 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_PRESENT_OT: LAUNCH_RATE_FAST(GEMX_PF_PRESENT_OT); break;
     case GEMX_PF_CHANGES_OT: LAUNCH_RATE_FAST(GEMX_PF_CHANGES_OT); break;
     case GEMX_PF_RESETS_OT: LAUNCH_RATE_FAST(GEMX_PF_RESETS_OT); break;
+    case GEMX_PF_DERIV: LAUNCH_RATE_FAST(GEMX_PF_DERIV); break;
+    case GEMX_PF_PREDICT: LAUNCH_RATE_FAST(GEMX_PF_PREDICT); break;
     default: LAUNCH_RATE_FAST(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_FAST
@@ -3914,6 +4012,8 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_PRESENT_OT: LAUNCH_RATE_GEN(GEMX_PF_PRESENT_OT); break;
     case GEMX_PF_CHANGES_OT: LAUNCH_RATE_GEN(GEMX_PF_CHANGES_OT); break;
     case GEMX_PF_RESETS_OT: LAUNCH_RATE_GEN(GEMX_PF_RESETS_OT); break;
+    case GEMX_PF_DERIV: LAUNCH_RATE_GEN(GEMX_PF_DERIV); break;
+    case GEMX_PF_PREDICT: LAUNCH_RATE_GEN(GEMX_PF_PREDICT); break;
     default: LAUNCH_RATE_GEN(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_GEN
@@ -3925,7 +4025,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     hipLaunchKernelGGL(k_rate_merge, dim3(blocks), dim3(TPB), 0, s->stream,
                        P.d_rsq, (uint32_t)P.rsq.size(), P.d_rsegq, P.d_rpart,
                        P.d_rrows2[slot], P.total_rows, start_sample, eff_step,
-                       range_ns, is_rate, is_counter, func, d_err);
+                       range_ns, is_rate, is_counter, func, scalar, d_err);
   }
   HIP_CHECK(hipEventRecord(ev2, s->stream));
 
@@ -4471,6 +4571,22 @@ extern "C" int gemx_prom_finish(gemx_shard *s, uint64_t *n_out,
   return rate_deliver(s, pe.slot, pe.fetch_rows, pe.out, n_out, stats);
 }
 
+/* deriv / predict_linear (prom_functions.go:358-436): least-squares
+ * slope of the window's points with x relative to the sample time,
+ * Kahan-compensated sums, the constY fast path, and for predict_linear
+ * value = slope*scalar + intercept (scalar = the prediction horizon in
+ * seconds, args[1] of the call). */
+extern "C" int gemx_prom_linear(gemx_shard *s, int64_t start_time,
+                                int64_t end_time, int64_t range_ns,
+                                int64_t step_ns, int is_predict,
+                                double scalar, gemx_rate_row *out_host,
+                                uint64_t cap, uint64_t *n_out,
+                                gemx_query_stats *stats) {
+  return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, 0, 0,
+                        is_predict ? GEMX_PF_PREDICT : GEMX_PF_DERIV,
+                        out_host, cap, n_out, stats, 0, scalar);
+}
+
 extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,
                               int64_t range_ns, int64_t step_ns, int is_rate,
                               int is_counter, gemx_rate_row *out_host,
@@ -4498,7 +4614,7 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
                                    int64_t step_ns, int func,
                                    gemx_rate_row *out_host, uint64_t cap,
                                    uint64_t *n_out, gemx_query_stats *stats) {
-  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_RESETS_OT) {
+  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_PREDICT) {
     seterr("unknown over_time func");
     return GEMX_E_INVALID;
   }

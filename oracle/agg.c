@@ -1034,6 +1034,7 @@ int64_t orc_prom_rate(const uint8_t *blob, int64_t blob_len,
         memset(out[nout]._pad, 0, sizeof(out[nout]._pad));
         nout++;
       }
+    skip_emit:;
       if (step_ns == 0) break;
     }
   }
@@ -1158,7 +1159,16 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
                            int64_t start, int64_t end, int64_t range_ns,
                            int64_t step_ns, int func, orc_rate_row *out,
                            int64_t cap) {
-  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 12) return -1;
+  return orc_prom_over_time_s(blob, blob_len, descs, nsegs, start, end,
+                              range_ns, step_ns, func, 0.0, out, cap);
+}
+
+int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
+                             const orc_seg_desc *descs, int64_t nsegs,
+                             int64_t start, int64_t end, int64_t range_ns,
+                             int64_t step_ns, int func, double scalar,
+                             orc_rate_row *out, int64_t cap) {
+  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 14) return -1;
   int64_t start_sample = start + range_ns;
   int64_t end_sample =
       (step_ns == 0) ? start_sample
@@ -1282,6 +1292,34 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
           v = (double)cc;
           break;
         }
+        case 13:   /* deriv (linearMergeFunc, prom_functions.go:369) */
+        case 14: { /* predict_linear */
+          if (n <= 1) { v = 0; goto skip_emit; }
+          double fv0 = sv[pi];
+          int constY = 1;
+          double cnt = 0, sX = 0, cX = 0, sY = 0, cY = 0, sXY = 0, cXY = 0,
+                 sX2 = 0, cX2 = 0;
+          for (int64_t k = pi; k < pj; k++) {
+            if (constY && sv[k] != fv0) constY = 0;
+            cnt += 1.0;
+            double x = (double)(st[k] - ts) / 1e9;
+            kahan_inc(x, &sX, &cX);
+            kahan_inc(sv[k], &sY, &cY);
+            kahan_inc(x * sv[k], &sXY, &cXY);
+            kahan_inc(x * x, &sX2, &cX2);
+          }
+          if (constY) {
+            if (isinf(fv0)) v = 0.0 / 0.0;
+            else v = (func == 13) ? 0.0 : fv0;
+            break;
+          }
+          sX += cX; sY += cY; sXY += cXY; sX2 += cX2;
+          double covXY = sXY - sX * sY / cnt;
+          double varX = sX2 - sX * sX / cnt;
+          double dv = covXY / varX;
+          v = (func == 13) ? dv : (dv * scalar + (sY / cnt - dv * sX / cnt));
+          break;
+        }
         default:
           v = sv[pj - 1];
           break;
@@ -1294,6 +1332,7 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
         memset(out[nout]._pad, 0, sizeof(out[nout]._pad));
         nout++;
       }
+    skip_emit:;
       if (step_ns == 0) break;
     }
   }

@@ -133,6 +133,10 @@ def load():
         C.c_uint64, C.c_uint64, C.c_uint64, C.c_uint32, i64, i64, C.c_int,
         u8p, i64, C.c_void_p, i64, i64p,
     ]
+    lib.orc_prom_over_time_s.restype = i64
+    lib.orc_prom_over_time_s.argtypes = [u8p, i64, C.c_void_p, i64, i64, i64,
+                                         i64, i64, C.c_int, C.c_double,
+                                         C.c_void_p, i64]
     lib.orc_prom_over_time.restype = i64
     lib.orc_prom_over_time.argtypes = [u8p, i64, C.c_void_p, i64, i64, i64, i64, i64,
                                        C.c_int, C.c_void_p, i64]
@@ -381,6 +385,27 @@ def prom_rate(blob, descs, start, end, range_ns, step_ns, is_rate=True, is_count
 OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6,
             "last": 7, "stdvar": 8, "stddev": 9, "present": 10, "changes": 11,
             "resets": 12}
+
+
+def prom_linear(blob, descs, start, end, range_ns, step_ns, is_predict=False,
+                scalar=0.0, cap=None):
+    """deriv / predict_linear (prom_functions.go:358-436)."""
+    lib = get()
+    bts = np.frombuffer(blob, dtype=np.uint8)
+    d = np.ascontiguousarray(descs, dtype=SEG_DESC_DTYPE)
+    if cap is None:
+        nsteps = 1 if step_ns == 0 else (
+            int((end - (start + range_ns)) // step_ns + 2)
+            if end >= start + range_ns else 1)
+        cap = nsteps * len(np.unique(d["sid"])) + 16
+    out = np.zeros(cap, dtype=RATE_ROW_DTYPE)
+    n = lib.orc_prom_over_time_s(
+        _u8(bts), len(bts), d.ctypes.data_as(C.c_void_p), len(d),
+        start, end, range_ns, step_ns, 14 if is_predict else 13,
+        float(scalar), out.ctypes.data_as(C.c_void_p), cap,
+    )
+    assert n >= 0, "oracle prom_linear failed"
+    return out[:n]
 
 
 def prom_over_time(blob, descs, start, end, range_ns, step_ns, func, cap=None):

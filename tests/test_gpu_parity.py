@@ -1563,3 +1563,34 @@ class TestStdVarOverTimeGPU:
                 assert np.array_equal(gpu["value"], ref["value"]), fn
         finally:
             sh2.close()
+
+    def test_deriv_predict_parity(self):
+        S = 10**9
+        blob, descs = orc.gen_shard(2505, 150, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            for pred, sc in ((False, 0.0), (True, 600.0)):
+                gpu, _ = sh.prom_linear(0, 999 * S, 300 * S, 60 * S,
+                                        is_predict=pred, scalar=sc)
+                gpu = gpu.copy()
+                ref = orc.prom_linear(blob, descs, 0, 999 * S, 300 * S,
+                                      60 * S, is_predict=pred, scalar=sc)
+                assert len(gpu) == len(ref), pred
+                assert np.array_equal(gpu["ts"], ref["ts"])
+                assert np.array_equal(gpu["value"].view(np.uint64),
+                                      ref["value"].view(np.uint64)), pred
+        finally:
+            sh.close()
+        # multi-segment windows: 1e-9 (Kahan tails fold per partial)
+        rng = np.random.default_rng(2506)
+        blob2, descs2, _ = build_shard(rng, F, range(1, 31), null_frac=0.05)
+        sh2 = gpu_shard(blob2, descs2, F)
+        try:
+            gpu, _ = sh2.prom_linear(0, 800 * S, 240 * S, 60 * S)
+            gpu = gpu.copy()
+            ref = orc.prom_linear(blob2, descs2, 0, 800 * S, 240 * S, 60 * S)
+            assert len(gpu) == len(ref)
+            ok = np.isclose(gpu["value"], ref["value"], rtol=1e-9, atol=1e-12)
+            assert np.all(ok)
+        finally:
+            sh2.close()

@@ -233,3 +233,42 @@ class TestStdVarPresentOverTime:
                 else:
                     exp = float(np.count_nonzero(vv[1:] < vv[:-1]))
                 assert r["value"] == exp, (func, r["ts"])
+
+    def test_deriv_predict_match_numpy(self):
+        import shard_helpers as sh
+        rng = np.random.default_rng(34)
+        blob, descs, truth = sh.build_shard(rng, sh.F, [1, 2], null_frac=0.1)
+        S = 10**9
+        rows = orc.prom_linear(blob, descs, 0, 700 * S, 180 * S, 60 * S)
+        prows = orc.prom_linear(blob, descs, 0, 700 * S, 180 * S, 60 * S,
+                                is_predict=True, scalar=300.0)
+        assert len(rows) > 0 and len(prows) == len(rows)
+        for r, p in zip(rows, prows):
+            at, av, ax = truth[int(r["sid"])]
+            m = (at >= r["ts"] - 180 * S) & (at <= r["ts"]) & ax
+            tt, vv = at[m], av[m]
+            keep = ~np.isnan(vv)
+            tt, vv = tt[keep], vv[keep]
+            assert len(vv) >= 2
+            x = (tt - r["ts"]) / 1e9
+            # least squares slope/intercept
+            n = len(x)
+            vx = np.sum(x * x) - np.sum(x) ** 2 / n
+            cov = np.sum(x * vv) - np.sum(x) * np.sum(vv) / n
+            slope = cov / vx
+            inter = np.mean(vv) - slope * np.mean(x)
+            assert abs(r["value"] - slope) <= 1e-6 * max(1.0, abs(slope))
+            exp_p = slope * 300.0 + inter
+            assert abs(p["value"] - exp_p) <= 1e-6 * max(1.0, abs(exp_p))
+
+    def test_deriv_single_point_emits_nothing(self):
+        import opengemini_amd.engine as gxe  # noqa: F401 (dtype import path)
+        # one point per window: deriv must skip those sample steps
+        S = 10**9
+        vals = np.array([1.0, 5.0])
+        sids = np.array([1, 1], dtype=np.uint64)
+        times = np.array([0, 500 * S], dtype=np.int64)
+        blob, descs = gxe.encode_shard(3, sids, times, vals)
+        rows = orc.prom_linear(bytes(blob), np.ascontiguousarray(descs),
+                               0, 600 * S, 60 * S, 60 * S)
+        assert len(rows) == 0  # never two points inside one 60s window
