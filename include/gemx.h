@@ -342,6 +342,7 @@ int gemx_prom_irate(gemx_shard *, int64_t start_time, int64_t end_time,
 #define GEMX_PF_STDVAR_OT 8
 #define GEMX_PF_STDDEV_OT 9
 #define GEMX_PF_PRESENT_OT 10
+#define GEMX_PF_ABSENT_OT 15  /* absent_over_time: 1 for EMPTY windows */
 #define GEMX_PF_CHANGES_OT 11  /* changes_prom (CalcChange) */
 #define GEMX_PF_RESETS_OT 12   /* resets_prom (CalcResets) */
 #define GEMX_PF_DERIV 13

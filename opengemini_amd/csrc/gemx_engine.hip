@@ -2170,6 +2170,7 @@ struct RateSeriesQ {
 #define GEMX_PF_RESETS_OT 12
 #define GEMX_PF_DERIV 13
 #define GEMX_PF_PREDICT 14
+#define GEMX_PF_ABSENT_OT 15
 
 __device__ __forceinline__ void d_kahan_inc(double inc, double &sum, double &c) {
   /* executor.KahanSumInc */
@@ -2713,6 +2714,16 @@ __global__ void __launch_bounds__(256) k_rate_merge(
     out.isnil = 1;
     memset(out._pad, 0, sizeof(out._pad));
     if (func >= GEMX_PF_SUM_OT) {
+      if (func == GEMX_PF_ABSENT_OT) {
+        /* inverse emit: 1 for windows with no samples */
+        if (acc.count == 0) {
+          out.value = 1.0;
+          out.isnil = 0;
+        }
+        if (out.isnil) atomicAdd(&err->gaps, 1ull);
+        rows[r] = out;
+        continue;
+      }
       bool lin_nil = false;
       if (acc.count > 0) {
         /* tails were applied per partial at merge time */
@@ -3987,6 +3998,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_RESETS_OT: LAUNCH_RATE_FAST(GEMX_PF_RESETS_OT); break;
     case GEMX_PF_DERIV: LAUNCH_RATE_FAST(GEMX_PF_DERIV); break;
     case GEMX_PF_PREDICT: LAUNCH_RATE_FAST(GEMX_PF_PREDICT); break;
+    case GEMX_PF_ABSENT_OT: LAUNCH_RATE_FAST(GEMX_PF_COUNT_OT); break;
     default: LAUNCH_RATE_FAST(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_FAST
@@ -4014,6 +4026,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     case GEMX_PF_RESETS_OT: LAUNCH_RATE_GEN(GEMX_PF_RESETS_OT); break;
     case GEMX_PF_DERIV: LAUNCH_RATE_GEN(GEMX_PF_DERIV); break;
     case GEMX_PF_PREDICT: LAUNCH_RATE_GEN(GEMX_PF_PREDICT); break;
+    case GEMX_PF_ABSENT_OT: LAUNCH_RATE_GEN(GEMX_PF_COUNT_OT); break;
     default: LAUNCH_RATE_GEN(GEMX_PF_RATE); break;
     }
 #undef LAUNCH_RATE_GEN
@@ -4614,7 +4627,7 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
                                    int64_t step_ns, int func,
                                    gemx_rate_row *out_host, uint64_t cap,
                                    uint64_t *n_out, gemx_query_stats *stats) {
-  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_PREDICT) {
+  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_ABSENT_OT) {
     seterr("unknown over_time func");
     return GEMX_E_INVALID;
   }

@@ -814,7 +814,7 @@ class Shard:
 
 OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6, "last": 7,
             "stdvar": 8, "stddev": 9, "present": 10, "changes": 11,
-            "resets": 12}
+            "resets": 12, "absent": 15}
 
 
 class AggCursor:

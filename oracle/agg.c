@@ -1168,7 +1168,7 @@ int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
                              int64_t start, int64_t end, int64_t range_ns,
                              int64_t step_ns, int func, double scalar,
                              orc_rate_row *out, int64_t cap) {
-  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 14) return -1;
+  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 15) return -1;
   int64_t start_sample = start + range_ns;
   int64_t end_sample =
       (step_ns == 0) ? start_sample
@@ -1221,6 +1221,19 @@ int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
       while (pi < npts && st[pi] < wstart) pi++;
       while (pj < npts && st[pj] <= ts) pj++;
       int64_t n = pj - pi;
+      if (func == 15) { /* absent_over_time: 1 for EMPTY windows only */
+        if (n == 0) {
+          if (nout >= cap) goto done;
+          out[nout].sid = sid;
+          out[nout].ts = ts;
+          out[nout].value = 1.0;
+          out[nout].isnil = 0;
+          memset(out[nout]._pad, 0, sizeof(out[nout]._pad));
+          nout++;
+        }
+        if (step_ns == 0) break;
+        continue;
+      }
       if (n >= 1) {
         double v = 0;
         switch (func) {

@@ -384,7 +384,7 @@ def prom_rate(blob, descs, start, end, range_ns, step_ns, is_rate=True, is_count
 
 OT_FUNCS = {"sum": 2, "count": 3, "avg": 4, "min": 5, "max": 6,
             "last": 7, "stdvar": 8, "stddev": 9, "present": 10, "changes": 11,
-            "resets": 12}
+            "resets": 12, "absent": 15}
 
 
 def prom_linear(blob, descs, start, end, range_ns, step_ns, is_predict=False,

@@ -1594,3 +1594,31 @@ class TestStdVarOverTimeGPU:
             assert np.all(ok)
         finally:
             sh2.close()
+
+    def test_absent_over_time(self):
+        import opengemini_amd as gx
+        S = 10**9
+        # series with a 200s gap: absent emits 1 only inside the gap
+        rng = np.random.default_rng(2507)
+        sids = np.repeat([1, 2], 300).astype(np.uint64)
+        t1 = np.concatenate([np.arange(100), np.arange(300, 500)]).astype(
+            np.int64) * S
+        times = np.tile(t1, 2)
+        blob, descs = gx.encode_shard(F, sids, times,
+                                      rng.normal(0, 1, 600))
+        descs = np.ascontiguousarray(descs)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            gpu, _ = sh.prom_over_time(0, 499 * S, 60 * S, 30 * S, "absent")
+            gpu = gpu.copy()
+        finally:
+            sh.close()
+        ref = orc.prom_over_time(blob, descs, 0, 499 * S, 60 * S, 30 * S, 15)
+        assert len(gpu) == len(ref) > 0
+        assert np.array_equal(gpu["sid"], ref["sid"])
+        assert np.array_equal(gpu["ts"], ref["ts"])
+        assert np.all(gpu["value"] == 1.0)
+        # every emitted step's window truly has no points
+        for r in gpu[gpu["sid"] == 1]:
+            m = (t1 >= r["ts"] - 60 * S) & (t1 <= r["ts"])
+            assert not m.any()
