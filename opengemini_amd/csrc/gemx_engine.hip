@@ -2293,10 +2293,14 @@ __global__ void __launch_bounds__(256) k_rate_scan(
     int rows = (int)d.rows;
     RatePartial *base = partials + rq.partial_base;
 
-    /* zero this segment's partial slots (lane-owned) */
+    /* zero this segment's partial slots (lane-owned, FULL struct: the
+     * over_time ring only re-writes slots it opens, and a recycled plan
+     * buffer can hold a previous query's slots — count-only clearing
+     * left stale aux state behind for the wider reducers) */
     for (uint32_t k = 0; k < rq.n_steps; k++) {
-      base[k].count = 0;
-      base[k].reset_adj = 0;
+      RatePartial z;
+      memset(&z, 0, sizeof(z));
+      base[k] = z;
     }
     if (rq.n_steps == 0) continue;
 
@@ -2308,11 +2312,13 @@ __global__ void __launch_bounds__(256) k_rate_scan(
       bool act = j < (int)rq.n_steps;
       ring[j].ts = act ? (start_sample + o * step_ns) : INT64_MIN;
       ring[j].ord = o;
-      ring[j].p.count = 0;
-      ring[j].p.reset_adj = 0;
-      ring[j].p.first_v = 0;
-      ring[j].p.last_v = 0;
-      ring[j].p.prev_v = 0; /* stdvar aux (M2) accumulator */
+      /* full zero: the wider reducers (stdvar M2, linear-regression
+       * sums in the time slots and aux fields) read-accumulate every
+       * field, and scratch-resident ring memory carries residue from
+       * earlier launches on the same stream */
+      RatePartial z;
+      memset(&z, 0, sizeof(z));
+      ring[j].p = z;
     }
 
     /* value/time iterators (fast: streaming; general: via scratch) */
@@ -3880,6 +3886,9 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   int64_t last_ord = (end_sample - start_sample) / eff_step;
 
   RatePlan &P = s->rate_plan;
+#ifdef GEMX_RATE_NOCACHE
+  P.valid = false;
+#endif
   if (!P.valid || P.start != start_time || P.end != end_time ||
       P.range_ns != range_ns || P.step_ns != step_ns) {
     if (s->rpend_count > 0) {
@@ -4046,6 +4055,9 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     seterr("output capacity too small");
     return GEMX_E_CAP;
   }
+#ifdef GEMX_RATE_SYNC_DEBUG
+  HIP_CHECK(hipStreamSynchronize(s->stream));
+#endif
   HIP_CHECK(hipStreamWaitEvent(s->copy_stream, ev2, 0));
   HIP_CHECK(hipMemcpyAsync(s->h_rerr2[slot], d_err, sizeof(DevErr),
                            hipMemcpyDeviceToHost, s->copy_stream));

@@ -1622,3 +1622,47 @@ class TestStdVarOverTimeGPU:
         for r in gpu[gpu["sid"] == 1]:
             m = (t1 >= r["ts"] - 60 * S) & (t1 <= r["ts"])
             assert not m.any()
+
+
+class TestPromFamilyFuzz:
+    def test_all_functions_random_grids(self):
+        """Every range-vector function over random (range, step) grids,
+        GPU vs oracle."""
+        S = 10**9
+        rng = np.random.default_rng(2601)
+        blob, descs = orc.gen_shard(2601, 60, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            for trial in range(5):
+                step = int(rng.integers(20, 150)) * S
+                k = int(rng.integers(1, 6))
+                range_ns = k * step
+                end = int(rng.integers(600, 999)) * S
+                for fn in ("sum", "count", "avg", "min", "max", "last",
+                           "stdvar", "stddev", "present", "changes",
+                           "resets"):
+                    gpu, _ = sh.prom_over_time(0, end, range_ns, step, fn)
+                    gpu = gpu.copy()
+                    ref = orc.prom_over_time(blob, descs, 0, end, range_ns,
+                                             step, fn)
+                    assert len(gpu) == len(ref), (fn, trial)
+                    ok = np.isclose(gpu["value"], ref["value"], rtol=1e-9,
+                                    atol=1e-12)
+                    both_nan = np.isnan(gpu["value"]) & np.isnan(ref["value"])
+                    assert np.all(ok | both_nan), (fn, trial)
+                # rate family + linear
+                gpu, _ = sh.prom_rate(0, end, range_ns, step)
+                gpu = gpu.copy()
+                ref = orc.prom_rate(blob, descs, 0, end, range_ns, step,
+                                    is_rate=True, is_counter=True)
+                assert len(gpu) == len(ref), ("rate", trial)
+                assert np.allclose(gpu["value"], ref["value"], rtol=1e-12)
+                gpu, _ = sh.prom_linear(0, end, range_ns, step)
+                gpu = gpu.copy()
+                ref = orc.prom_linear(blob, descs, 0, end, range_ns, step)
+                assert len(gpu) == len(ref), ("deriv", trial)
+                ok = np.isclose(gpu["value"], ref["value"], rtol=1e-9,
+                                atol=1e-12)
+                assert np.all(ok), ("deriv", trial)
+        finally:
+            sh.close()
