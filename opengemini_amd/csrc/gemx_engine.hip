@@ -2293,21 +2293,29 @@ __global__ void __launch_bounds__(256) k_rate_scan(
     int rows = (int)d.rows;
     RatePartial *base = partials + rq.partial_base;
 
-    /* zero this segment's partial slots (lane-owned, FULL struct: the
-     * over_time ring only re-writes slots it opens, and a recycled plan
-     * buffer can hold a previous query's slots — count-only clearing
-     * left stale aux state behind for the wider reducers) */
-    for (uint32_t k = 0; k < rq.n_steps; k++) {
-      RatePartial z;
-      memset(&z, 0, sizeof(z));
-      base[k] = z;
+    /* zero this segment's partial slots (lane-owned). The rate/irate
+     * prefix path writes every field it reads at flush, so count+reset
+     * suffice; the wider reducers read-accumulate everything and need
+     * the full struct cleared. */
+    if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE) {
+      for (uint32_t k = 0; k < rq.n_steps; k++) {
+        base[k].count = 0;
+        base[k].reset_adj = 0;
+      }
+    } else {
+      for (uint32_t k = 0; k < rq.n_steps; k++) {
+        RatePartial z;
+        memset(&z, 0, sizeof(z));
+        base[k] = z;
+      }
     }
     if (rq.n_steps == 0) continue;
 
-    /* ring init: slots j hold ordinals rq.s0+j */
+    /* ring init: slots j hold ordinals rq.s0+j (the rate/irate prefix
+     * path never touches the ring — skip its scratch traffic there) */
     RateSlot ring[RATE_W];
 #pragma unroll
-    for (int j = 0; j < RATE_W; j++) {
+    for (int j = 0; FUNC > GEMX_PF_IRATE && j < RATE_W; j++) {
       int64_t o = rq.s0 + j;
       bool act = j < (int)rq.n_steps;
       ring[j].ts = act ? (start_sample + o * step_ns) : INT64_MIN;
