@@ -1696,7 +1696,8 @@ __global__ void __launch_bounds__(256) k_merge(
     out.win_start = interval ? win_start_of(w, interval, offset) : q_start;
     if (!merge_series_window<COLTYPE>(s, segq, partials, w, &out)) {
       out.count = -1; /* gap row: host compacts it away */
-      atomicAdd(&err->gaps, 1ull); /* gaps are rare: the host skips its
+      __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM); /* gaps are rare: the host skips its
                                       whole compaction scan when zero */
     }
     rows[r] = out;
@@ -1937,7 +1938,8 @@ __global__ void __launch_bounds__(256) k_group_p2(
               o.sum_isnil = 1;
         } else {
           o.count = -1; /* gap */
-          atomicAdd(&err->gaps, 1ull);
+          __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM);
         }
       } else {
         o.first_row_time = ws; /* BuildEmptyIntervalRec interval times */
@@ -2043,7 +2045,8 @@ __global__ void __launch_bounds__(256) k_tag_p2(
     o.win_start = ws;
     if (!a.used) {
       o.count = -1; /* gap: host compacts */
-      atomicAdd(&err->gaps, 1ull);
+      __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM);
     } else {
       o.first_row_time = ws;
       o.count = a.count;
@@ -2734,7 +2737,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
           out.value = 1.0;
           out.isnil = 0;
         }
-        if (out.isnil) atomicAdd(&err->gaps, 1ull);
+        if (out.isnil) __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM);
         rows[r] = out;
         continue;
       }
@@ -2779,7 +2783,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
           out.value = acc.first_v;
         out.isnil = 0;
       }
-      if (out.isnil) atomicAdd(&err->gaps, 1ull);
+      if (out.isnil) __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM);
     rows[r] = out;
       continue;
     }
@@ -2795,7 +2800,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
         out.value = rv;
         out.isnil = 0;
       }
-      if (out.isnil) atomicAdd(&err->gaps, 1ull);
+      if (out.isnil) __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM);
     rows[r] = out;
       continue;
     }
@@ -2821,7 +2827,8 @@ __global__ void __launch_bounds__(256) k_rate_merge(
       out.value = result;
       out.isnil = 0;
     }
-    if (out.isnil) atomicAdd(&err->gaps, 1ull);
+    if (out.isnil) __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_SYSTEM);
     rows[r] = out;
   }
 }
