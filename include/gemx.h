@@ -356,6 +356,17 @@ int gemx_prom_linear(gemx_shard *, int64_t start_time, int64_t end_time,
                      int64_t range_ns, int64_t step_ns, int is_predict,
                      double scalar, gemx_rate_row *out_host, uint64_t cap,
                      uint64_t *n_out, gemx_query_stats *stats);
+
+/* quantile_over_time / mad_over_time (executor/agg_func_prom.go:626-695):
+ * per-window collect, LDS sort, CalcQuantile's rank = q*(n-1) linear
+ * interpolation; mad = median of |v - median|. q outside [0,1] yields
+ * -Inf/+Inf as the reference does; windows of more than 4096 points are
+ * refused with GEMX_E_UNSUPPORTED (the LDS sort capacity, like the rate
+ * window-ring bound). Synchronous (a host prefix-sum phase). */
+int gemx_prom_quantile(gemx_shard *, int64_t start_time, int64_t end_time,
+                       int64_t range_ns, int64_t step_ns, int is_mad,
+                       double q, gemx_rate_row *out_host, uint64_t cap,
+                       uint64_t *n_out, gemx_query_stats *stats);
 int gemx_prom_over_time(gemx_shard *, int64_t start_time, int64_t end_time,
                         int64_t range_ns, int64_t step_ns, int func,
                         gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,

@@ -2174,6 +2174,8 @@ struct RateSeriesQ {
 #define GEMX_PF_DERIV 13
 #define GEMX_PF_PREDICT 14
 #define GEMX_PF_ABSENT_OT 15
+#define GEMX_PF_QUANTILE 16
+#define GEMX_PF_MAD 17
 
 __device__ __forceinline__ void d_kahan_inc(double inc, double &sum, double &c) {
   /* executor.KahanSumInc */
@@ -2284,7 +2286,12 @@ __global__ void __launch_bounds__(256) k_rate_scan(
     const RateSegQ *__restrict__ rsegq, const uint32_t *__restrict__ seg_ids,
     uint32_t nseg_ids, RatePartial *__restrict__ partials, int64_t start_sample,
     int64_t step_ns, int64_t range_ns, uint8_t *__restrict__ scratch,
-    uint64_t scratch_per_lane, uint32_t nlanes, DevErr *err) {
+    uint64_t scratch_per_lane, uint32_t nlanes,
+    /* quantile/mad collect phases (FUNC 16/17): per-(series,step) bucket
+     * counters and, in the fill phase, value scatter via qoff offsets */
+    const RateSeriesQ *__restrict__ rsq_dev, uint32_t *__restrict__ qcnt,
+    const uint64_t *__restrict__ qoff, double *__restrict__ qvals,
+    DevErr *err) {
   uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
   uint32_t stride = FAST ? gridDim.x * blockDim.x : nlanes;
   if (!FAST && gid >= nlanes) return;
@@ -2300,7 +2307,9 @@ __global__ void __launch_bounds__(256) k_rate_scan(
      * prefix path writes every field it reads at flush, so count+reset
      * suffice; the wider reducers read-accumulate everything and need
      * the full struct cleared. */
-    if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE) {
+    if (FUNC == GEMX_PF_QUANTILE || FUNC == GEMX_PF_MAD) {
+      /* bucket collect: no partial slots used */
+    } else if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE) {
       for (uint32_t k = 0; k < rq.n_steps; k++) {
         base[k].count = 0;
         base[k].reset_adj = 0;
@@ -2318,7 +2327,8 @@ __global__ void __launch_bounds__(256) k_rate_scan(
      * path never touches the ring — skip its scratch traffic there) */
     RateSlot ring[RATE_W];
 #pragma unroll
-    for (int j = 0; FUNC > GEMX_PF_IRATE && j < RATE_W; j++) {
+    for (int j = 0; FUNC > GEMX_PF_IRATE && FUNC < GEMX_PF_QUANTILE &&
+                    j < RATE_W; j++) {
       int64_t o = rq.s0 + j;
       bool act = j < (int)rq.n_steps;
       ring[j].ts = act ? (start_sample + o * step_ns) : INT64_MIN;
@@ -2465,6 +2475,25 @@ __global__ void __launch_bounds__(256) k_rate_scan(
       if (!valid) continue;
       if (fv != fv) continue; /* FilterRangeNANPoint */
 
+      if (FUNC == GEMX_PF_QUANTILE || FUNC == GEMX_PF_MAD) {
+        /* scatter this point into every sample-step bucket whose window
+         * [ts-range, ts] contains it (≤ range/step+1 ≤ ring-bound steps) */
+        const RateSeriesQ sq2 = rsq_dev[rq.series_idx];
+        int64_t o_lo = t - start_sample; /* ceil to grid */
+        o_lo = (o_lo <= 0) ? 0 : (o_lo + step_ns - 1) / step_ns;
+        int64_t o_hi = (t + range_ns - start_sample) / step_ns;
+        if (o_lo < rq.s0) o_lo = rq.s0;
+        int64_t last2 = rq.s0 + (int64_t)rq.n_steps - 1;
+        if (o_hi > last2) o_hi = last2;
+        for (int64_t o = o_lo; o <= o_hi; o++) {
+          int64_t ts2 = start_sample + o * step_ns;
+          if (t < ts2 - range_ns || t > ts2) continue;
+          uint64_t bucket = sq2.out_base + (uint64_t)(o - sq2.s_min);
+          uint32_t idx = atomicAdd(&qcnt[bucket], 1u);
+          if (qvals) qvals[qoff[bucket] + idx] = fv;
+        }
+        continue;
+      }
       if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE) {
         /* close windows whose ts < t: emit from shared − snapshot */
         while (t > first_open_ts && first_open <= last_ord_seg) {
@@ -2560,12 +2589,117 @@ __global__ void __launch_bounds__(256) k_rate_scan(
           snap[j].count_at = -1;
         }
       }
-    } else {
+    } else if (FUNC < GEMX_PF_QUANTILE) {
 #pragma unroll
       for (int j = 0; j < RATE_W; j++) {
         if (ring[j].ts != INT64_MIN) base[ring[j].ord - rq.s0] = ring[j].p;
       }
     }
+  }
+}
+
+/* quantile/mad finalize: one block per (series, step) bucket — load the
+ * collected values into LDS, odd-even sort, then CalcQuantile's linear
+ * interpolation (executor/agg_func_prom.go:651,670: rank = q*(n-1);
+ * mad = median of |v - median|). Capacity 4096 values per window
+ * (enforced host-side with a loud GEMX_E_UNSUPPORTED, like the rate
+ * window-ring bound). */
+__global__ void __launch_bounds__(256) k_quantile_final(
+    const RateSeriesQ *__restrict__ series, uint32_t nseries,
+    const uint32_t *__restrict__ qcnt, const uint64_t *__restrict__ qoff,
+    const double *__restrict__ qvals, gemx_rate_row *__restrict__ rows,
+    uint64_t total_rows, int64_t start_sample, int64_t step_ns, int is_mad,
+    double q, DevErr *__restrict__ err) {
+  /* fp contract off: the interpolation v[lo]*(1-w) + v[hi]*w must not
+   * fuse or it drifts a ulp from the reference's x86 arithmetic */
+#pragma clang fp contract(off)
+  __shared__ double sv[4096];
+  for (uint64_t b = blockIdx.x; b < total_rows; b += gridDim.x) {
+    /* series lookup for the sid/ts labels */
+    uint32_t lo = 0, hi = nseries - 1;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi + 1) >> 1;
+      if (series[mid].out_base <= b) lo = mid;
+      else hi = mid - 1;
+    }
+    const RateSeriesQ s = series[lo];
+    int64_t o = s.s_min + (int64_t)(b - s.out_base);
+    gemx_rate_row out;
+    out.sid = s.sid;
+    out.ts = start_sample + o * step_ns;
+    out.value = 0;
+    out.isnil = 1;
+    memset(out._pad, 0, sizeof(out._pad));
+    uint32_t n = qcnt[b];
+    if (n == 0 || n > 4096) {
+      if (threadIdx.x == 0) {
+        __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+        rows[b] = out;
+      }
+      __syncthreads();
+      continue;
+    }
+    const double *src = qvals + qoff[b];
+    for (uint32_t i = threadIdx.x; i < n; i += blockDim.x) sv[i] = src[i];
+    __syncthreads();
+    /* odd-even transposition sort: n rounds over n elements */
+    for (uint32_t round = 0; round < n; round++) {
+      uint32_t par = round & 1;
+      for (uint32_t i = threadIdx.x; 2 * i + 1 + par < n; i += blockDim.x) {
+        uint32_t a = 2 * i + par, c = a + 1;
+        double x = sv[a], y = sv[c];
+        if (x > y) {
+          sv[a] = y;
+          sv[c] = x;
+        }
+      }
+      __syncthreads();
+    }
+    const double qe = is_mad ? 0.5 : q; /* CalcMad's first pass is the
+                                           median regardless of q */
+    double res;
+    if (qe != qe) {
+      res = nan("");
+    } else if (qe < 0) {
+      res = -INFINITY;
+    } else if (qe > 1) {
+      res = INFINITY;
+    } else {
+      double rank = qe * ((double)n - 1.0);
+      uint32_t l2 = (uint32_t)rank;
+      uint32_t h2 = (l2 + 1 < n) ? l2 + 1 : n - 1;
+      double w = rank - floor(rank);
+      res = sv[l2] * (1 - w) + sv[h2] * w;
+    }
+    if (is_mad) {
+      double med = res;
+      __syncthreads();
+      for (uint32_t i = threadIdx.x; i < n; i += blockDim.x)
+        sv[i] = fabs(sv[i] - med);
+      __syncthreads();
+      for (uint32_t round = 0; round < n; round++) {
+        uint32_t par = round & 1;
+        for (uint32_t i = threadIdx.x; 2 * i + 1 + par < n; i += blockDim.x) {
+          uint32_t a = 2 * i + par, c = a + 1;
+          double x = sv[a], y = sv[c];
+          if (x > y) {
+            sv[a] = y;
+            sv[c] = x;
+          }
+        }
+        __syncthreads();
+      }
+      double rank = 0.5 * ((double)n - 1.0);
+      uint32_t l2 = (uint32_t)rank;
+      uint32_t h2 = (l2 + 1 < n) ? l2 + 1 : n - 1;
+      double w = rank - floor(rank);
+      res = sv[l2] * (1 - w) + sv[h2] * w;
+    }
+    out.value = res;
+    out.isnil = 0;
+    if (threadIdx.x == 0) rows[b] = out;
+    __syncthreads();
   }
 }
 
@@ -3864,6 +3998,10 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                           gemx_query_stats *stats, int async_begin = 0,
                           double scalar = 0.0) {
   if (!s) return GEMX_E_INVALID;
+  if (async_begin && func >= GEMX_PF_QUANTILE) {
+    seterr("quantile/mad are synchronous (host prefix-sum phase)");
+    return GEMX_E_INVALID;
+  }
   if (async_begin && s->rpend_count >= 2) {
     seterr("two rate queries already in flight: call gemx_prom_finish");
     return GEMX_E_INVALID;
@@ -4000,14 +4138,84 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
              ev2 = s->ev_r[slot][2];
   HIP_CHECK(hipEventRecord(ev0, s->stream));
   const int TPB = 256;
-  if (!s->fast_ids.empty()) {
+  const bool qmode = (func == GEMX_PF_QUANTILE || func == GEMX_PF_MAD);
+  uint32_t *d_qcnt = nullptr;
+  uint64_t *d_qoff = nullptr;
+  double *d_qvals = nullptr;
+  if (qmode) {
+    /* two collect phases: count per (series, step) bucket, then value
+     * scatter via host prefix-sum offsets; the finalize kernel sorts
+     * each bucket in LDS (CalcQuantile/CalcMad) */
+    const uint64_t nb = P.total_rows ? P.total_rows : 1;
+    HIP_CHECK(hipMalloc(&d_qcnt, sizeof(uint32_t) * nb));
+    HIP_CHECK(hipMalloc(&d_qoff, sizeof(uint64_t) * nb));
+    HIP_CHECK(hipMemsetAsync(d_qcnt, 0, sizeof(uint32_t) * nb, s->stream));
+#define LAUNCH_QPHASE(FASTV, LIST, NSEG, SCR, SPL, LANES)                      \
+    hipLaunchKernelGGL((k_rate_scan<FASTV, GEMX_PF_QUANTILE>), dim3(blocks),   \
+                       dim3(TPB), 0, s->stream, s->d_blob, s->d_descs,         \
+                       P.d_rsegq, LIST, NSEG, P.d_rpart, start_sample,         \
+                       eff_step, range_ns, SCR, SPL, LANES, P.d_rsq, d_qcnt,   \
+                       d_qoff, d_qvals, d_err)
+    if (!s->fast_ids.empty()) {
+      uint32_t n = (uint32_t)s->fast_ids.size();
+      uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
+      LAUNCH_QPHASE(1, s->d_fast_ids, n, nullptr, 0, 0);
+    }
+    if (!s->general_ids.empty()) {
+      if (!P.d_scratch) {
+        P.gen_lanes = (uint32_t)std::min<uint64_t>(s->general_ids.size(), 16384);
+        HIP_CHECK(hipMalloc(&P.d_scratch, scratch_per_lane * P.gen_lanes));
+      }
+      uint32_t n = (uint32_t)s->general_ids.size();
+      uint32_t blocks = (P.gen_lanes + TPB - 1) / TPB;
+      LAUNCH_QPHASE(0, s->d_general_ids, n, P.d_scratch, scratch_per_lane,
+                    P.gen_lanes);
+    }
+    std::vector<uint32_t> hcnt(nb);
+    HIP_CHECK(hipMemcpyAsync(hcnt.data(), d_qcnt, sizeof(uint32_t) * nb,
+                             hipMemcpyDeviceToHost, s->stream));
+    HIP_CHECK(hipStreamSynchronize(s->stream));
+    std::vector<uint64_t> hoff(nb);
+    uint64_t acc2 = 0;
+    uint32_t maxc = 0;
+    for (uint64_t i = 0; i < nb; i++) {
+      hoff[i] = acc2;
+      acc2 += hcnt[i];
+      if (hcnt[i] > maxc) maxc = hcnt[i];
+    }
+    if (maxc > 4096) {
+      hipFree(d_qcnt);
+      hipFree(d_qoff);
+      seterr("quantile window holds >4096 points — beyond the LDS sort "
+             "capacity this round");
+      return GEMX_E_UNSUPPORTED;
+    }
+    HIP_CHECK(hipMemcpyAsync(d_qoff, hoff.data(), sizeof(uint64_t) * nb,
+                             hipMemcpyHostToDevice, s->stream));
+    HIP_CHECK(hipMalloc(&d_qvals, sizeof(double) * (acc2 ? acc2 : 1)));
+    HIP_CHECK(hipMemsetAsync(d_qcnt, 0, sizeof(uint32_t) * nb, s->stream));
+    if (!s->fast_ids.empty()) {
+      uint32_t n = (uint32_t)s->fast_ids.size();
+      uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
+      LAUNCH_QPHASE(1, s->d_fast_ids, n, nullptr, 0, 0);
+    }
+    if (!s->general_ids.empty()) {
+      uint32_t n = (uint32_t)s->general_ids.size();
+      uint32_t blocks = (P.gen_lanes + TPB - 1) / TPB;
+      LAUNCH_QPHASE(0, s->d_general_ids, n, P.d_scratch, scratch_per_lane,
+                    P.gen_lanes);
+    }
+#undef LAUNCH_QPHASE
+  }
+  if (!qmode && !s->fast_ids.empty()) {
     uint32_t n = (uint32_t)s->fast_ids.size();
     uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
 #define LAUNCH_RATE_FAST(F)                                                     \
     hipLaunchKernelGGL((k_rate_scan<1, F>), dim3(blocks), dim3(TPB), 0,         \
                        s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
                        s->d_fast_ids, n, P.d_rpart, start_sample, eff_step,     \
-                       range_ns, nullptr, 0, 0, d_err)
+                       range_ns, nullptr, 0, 0, nullptr, nullptr, nullptr,      \
+                       nullptr, d_err)
     switch (func) {
     case GEMX_PF_SUM_OT: LAUNCH_RATE_FAST(GEMX_PF_SUM_OT); break;
     case GEMX_PF_COUNT_OT: LAUNCH_RATE_FAST(GEMX_PF_COUNT_OT); break;
@@ -4035,7 +4243,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                        s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
                        s->d_general_ids, n, P.d_rpart, start_sample, eff_step,  \
                        range_ns, P.d_scratch, scratch_per_lane, P.gen_lanes,    \
-                       d_err)
+                       nullptr, nullptr, nullptr, nullptr, d_err)
     switch (func) {
     case GEMX_PF_SUM_OT: LAUNCH_RATE_GEN(GEMX_PF_SUM_OT); break;
     case GEMX_PF_COUNT_OT: LAUNCH_RATE_GEN(GEMX_PF_COUNT_OT); break;
@@ -4056,7 +4264,14 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
 #undef LAUNCH_RATE_GEN
   }
   HIP_CHECK(hipEventRecord(ev1, s->stream));
-  if (P.total_rows > 0) {
+  if (qmode && P.total_rows > 0) {
+    uint32_t blocks = (uint32_t)std::min<uint64_t>(P.total_rows, 65535);
+    hipLaunchKernelGGL(k_quantile_final, dim3(blocks), dim3(256), 0,
+                       s->stream, P.d_rsq, (uint32_t)P.rsq.size(), d_qcnt,
+                       d_qoff, d_qvals, P.d_rrows2[slot], P.total_rows,
+                       start_sample, eff_step, func == GEMX_PF_MAD ? 1 : 0,
+                       scalar, d_err);
+  } else if (P.total_rows > 0) {
     uint32_t blocks =
         (uint32_t)std::min<uint64_t>((P.total_rows + TPB - 1) / TPB, 65535);
     hipLaunchKernelGGL(k_rate_merge, dim3(blocks), dim3(TPB), 0, s->stream,
@@ -4081,6 +4296,15 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                            hipMemcpyDeviceToHost, s->copy_stream));
   HIP_CHECK(hipEventRecord(s->ev_rcopy[slot], s->copy_stream));
 
+  if (qmode) {
+    /* synchronous by construction (host prefix-sum already synced the
+     * stream once); finish the copy, then free the collect buffers */
+    int rcq = rate_deliver(s, slot, P.total_rows, out_host, n_out, stats);
+    hipFree(d_qcnt);
+    hipFree(d_qoff);
+    if (d_qvals) hipFree(d_qvals);
+    return rcq;
+  }
   if (async_begin) {
     auto &pe = s->rpend[(s->rpend_head + s->rpend_count) & 1];
     pe.active = true;
@@ -4627,6 +4851,20 @@ extern "C" int gemx_prom_linear(gemx_shard *s, int64_t start_time,
                         out_host, cap, n_out, stats, 0, scalar);
 }
 
+/* quantile_over_time / mad_over_time (executor/agg_func_prom.go:626-695):
+ * per-window collect + LDS sort + CalcQuantile's linear interpolation;
+ * mad = median of |v - median|. q outside [0,1] follows the reference
+ * (-Inf / +Inf); windows of more than 4096 points are refused loudly. */
+extern "C" int gemx_prom_quantile(gemx_shard *s, int64_t start_time,
+                                  int64_t end_time, int64_t range_ns,
+                                  int64_t step_ns, int is_mad, double q,
+                                  gemx_rate_row *out_host, uint64_t cap,
+                                  uint64_t *n_out, gemx_query_stats *stats) {
+  return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, 0, 0,
+                        is_mad ? GEMX_PF_MAD : GEMX_PF_QUANTILE, out_host,
+                        cap, n_out, stats, 0, q);
+}
+
 extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,
                               int64_t range_ns, int64_t step_ns, int is_rate,
                               int is_counter, gemx_rate_row *out_host,
@@ -4654,7 +4892,7 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
                                    int64_t step_ns, int func,
                                    gemx_rate_row *out_host, uint64_t cap,
                                    uint64_t *n_out, gemx_query_stats *stats) {
-  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_ABSENT_OT) {
+  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_MAD) {
     seterr("unknown over_time func");
     return GEMX_E_INVALID;
   }

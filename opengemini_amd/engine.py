@@ -222,6 +222,12 @@ def _load():
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_prom_quantile.restype = C.c_int
+    lib.gemx_prom_quantile.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_double, C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64),
+        C.POINTER(_Stats),
+    ]
     lib.gemx_prom_linear.restype = C.c_int
     lib.gemx_prom_linear.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
@@ -780,6 +786,26 @@ class Shard:
         rc = lib.gemx_prom_linear(
             self._h, start_time, end_time, range_ns, step_ns,
             1 if is_predict else 0, float(scalar),
+            out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        return out[: n.value], dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms,
+            total_ms=st.total_ms, n_rows=st.n_rows)
+
+    def prom_quantile(self, start_time, end_time, range_ns, step_ns,
+                      q=0.5, is_mad=False, out_cap=None):
+        """quantile_over_time(q, ...) / mad_over_time over range vectors
+        (<=4096 points per window; loud refusal beyond)."""
+        lib = self._lib
+        if out_cap is None:
+            out_cap = self._rate_cap(start_time, end_time, range_ns, step_ns)
+        out = self._pooled_out("rate", out_cap, RATE_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_prom_quantile(
+            self._h, start_time, end_time, range_ns, step_ns,
+            1 if is_mad else 0, float(q),
             out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
         )
         _check(rc, lib)

@@ -1154,6 +1154,11 @@ static void kahan_inc(double inc, double *sum, double *c) {
   *sum = t;
 }
 
+int orc_dcmp(const void *a, const void *b) {
+  double x = *(const double *)a, y = *(const double *)b;
+  return (x > y) - (x < y);
+}
+
 int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
                            const orc_seg_desc *descs, int64_t nsegs,
                            int64_t start, int64_t end, int64_t range_ns,
@@ -1168,7 +1173,7 @@ int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
                              int64_t start, int64_t end, int64_t range_ns,
                              int64_t step_ns, int func, double scalar,
                              orc_rate_row *out, int64_t cap) {
-  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 15) return -1;
+  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 17) return -1;
   int64_t start_sample = start + range_ns;
   int64_t end_sample =
       (step_ns == 0) ? start_sample
@@ -1303,6 +1308,39 @@ int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
           for (int64_t k = pi + 1; k < pj; k++)
             if (sv[k] < sv[k - 1]) cc++;
           v = (double)cc;
+          break;
+        }
+        case 16:   /* quantile_over_time (executor.CalcQuantile:
+                      sort, rank = q*(n-1), linear interpolation) */
+        case 17: { /* mad_over_time (CalcMad: median of |v - median|) */
+          double *w = (double *)malloc((size_t)n * 8);
+          for (int64_t k = 0; k < n; k++) w[k] = sv[pi + k];
+          qsort(w, (size_t)n, 8, orc_dcmp);
+          double q = (func == 16) ? scalar : 0.5;
+          double res;
+          if (q != q) res = 0.0 / 0.0;
+          else if (q < 0) res = -1.0 / 0.0;
+          else if (q > 1) res = 1.0 / 0.0;
+          else {
+            double rank = q * ((double)n - 1.0);
+            int64_t lo2 = (int64_t)floor(rank);
+            if (lo2 < 0) lo2 = 0;
+            int64_t hi2 = lo2 + 1 < n ? lo2 + 1 : n - 1;
+            double wgt = rank - floor(rank);
+            res = w[lo2] * (1 - wgt) + w[hi2] * wgt;
+          }
+          if (func == 17) {
+            double med = res;
+            for (int64_t k = 0; k < n; k++) w[k] = fabs(w[k] - med);
+            qsort(w, (size_t)n, 8, orc_dcmp);
+            double rank = 0.5 * ((double)n - 1.0);
+            int64_t lo2 = (int64_t)floor(rank);
+            int64_t hi2 = lo2 + 1 < n ? lo2 + 1 : n - 1;
+            double wgt = rank - floor(rank);
+            res = w[lo2] * (1 - wgt) + w[hi2] * wgt;
+          }
+          free(w);
+          v = res;
           break;
         }
         case 13:   /* deriv (linearMergeFunc, prom_functions.go:369) */

@@ -408,6 +408,27 @@ def prom_linear(blob, descs, start, end, range_ns, step_ns, is_predict=False,
     return out[:n]
 
 
+def prom_quantile(blob, descs, start, end, range_ns, step_ns, q=0.5,
+                  is_mad=False, cap=None):
+    """quantile_over_time / mad_over_time (CalcQuantile/CalcMad)."""
+    lib = get()
+    bts = np.frombuffer(blob, dtype=np.uint8)
+    d = np.ascontiguousarray(descs, dtype=SEG_DESC_DTYPE)
+    if cap is None:
+        nsteps = 1 if step_ns == 0 else (
+            int((end - (start + range_ns)) // step_ns + 2)
+            if end >= start + range_ns else 1)
+        cap = nsteps * len(np.unique(d["sid"])) + 16
+    out = np.zeros(cap, dtype=RATE_ROW_DTYPE)
+    n = lib.orc_prom_over_time_s(
+        _u8(bts), len(bts), d.ctypes.data_as(C.c_void_p), len(d),
+        start, end, range_ns, step_ns, 17 if is_mad else 16, float(q),
+        out.ctypes.data_as(C.c_void_p), cap,
+    )
+    assert n >= 0, "oracle prom_quantile failed"
+    return out[:n]
+
+
 def prom_over_time(blob, descs, start, end, range_ns, step_ns, func, cap=None):
     lib = get()
     bts = np.frombuffer(blob, dtype=np.uint8)

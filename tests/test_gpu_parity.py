@@ -1666,3 +1666,40 @@ class TestPromFamilyFuzz:
                 assert np.all(ok), ("deriv", trial)
         finally:
             sh.close()
+
+    def test_quantile_mad_parity(self):
+        S = 10**9
+        blob, descs = orc.gen_shard(2508, 150, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            for q in (0.5, 0.95, 0.0, 1.5, -0.5):
+                gpu, _ = sh.prom_quantile(0, 999 * S, 300 * S, 60 * S, q=q)
+                gpu = gpu.copy()
+                ref = orc.prom_quantile(blob, descs, 0, 999 * S, 300 * S,
+                                        60 * S, q=q)
+                assert len(gpu) == len(ref), q
+                assert np.array_equal(gpu["value"].view(np.uint64),
+                                      ref["value"].view(np.uint64)), q
+            gpu, _ = sh.prom_quantile(0, 999 * S, 300 * S, 60 * S,
+                                      is_mad=True)
+            gpu = gpu.copy()
+            ref = orc.prom_quantile(blob, descs, 0, 999 * S, 300 * S, 60 * S,
+                                    is_mad=True)
+            assert np.array_equal(gpu["value"].view(np.uint64),
+                                  ref["value"].view(np.uint64))
+        finally:
+            sh.close()
+        # multi-segment windows + nulls (general path)
+        rng = np.random.default_rng(2509)
+        blob2, descs2, _ = build_shard(rng, F, range(1, 31), null_frac=0.1)
+        sh2 = gpu_shard(blob2, descs2, F)
+        try:
+            gpu, _ = sh2.prom_quantile(0, 800 * S, 240 * S, 60 * S, q=0.9)
+            gpu = gpu.copy()
+            ref = orc.prom_quantile(blob2, descs2, 0, 800 * S, 240 * S,
+                                    60 * S, q=0.9)
+            assert len(gpu) == len(ref)
+            assert np.array_equal(gpu["value"].view(np.uint64),
+                                  ref["value"].view(np.uint64))
+        finally:
+            sh2.close()

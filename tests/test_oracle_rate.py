@@ -272,3 +272,36 @@ class TestStdVarPresentOverTime:
         rows = orc.prom_linear(bytes(blob), np.ascontiguousarray(descs),
                                0, 600 * S, 60 * S, 60 * S)
         assert len(rows) == 0  # never two points inside one 60s window
+
+    def test_quantile_mad_match_numpy(self):
+        import shard_helpers as sh
+        rng = np.random.default_rng(35)
+        blob, descs, truth = sh.build_shard(rng, sh.F, [1, 2], null_frac=0.1)
+        S = 10**9
+        for q in (0.0, 0.25, 0.5, 0.9, 1.0):
+            rows = orc.prom_quantile(blob, descs, 0, 700 * S, 180 * S,
+                                     60 * S, q=q)
+            for r in rows[:20]:
+                at, av, ax = truth[int(r["sid"])]
+                m = (at >= r["ts"] - 180 * S) & (at <= r["ts"]) & ax
+                vv = np.sort(av[m][~np.isnan(av[m])])
+                n = len(vv)
+                rank = q * (n - 1)
+                lo = int(np.floor(rank)); hi = min(n - 1, lo + 1)
+                w = rank - np.floor(rank)
+                exp = vv[lo] * (1 - w) + vv[hi] * w
+                assert abs(r["value"] - exp) <= 1e-12 * max(1, abs(exp)), q
+        mrows = orc.prom_quantile(blob, descs, 0, 700 * S, 180 * S, 60 * S,
+                                  is_mad=True)
+        for r in mrows[:10]:
+            at, av, ax = truth[int(r["sid"])]
+            m = (at >= r["ts"] - 180 * S) & (at <= r["ts"]) & ax
+            vv = av[m][~np.isnan(av[m])]
+            def pq(x, qq):
+                x = np.sort(x); n = len(x)
+                rank = qq * (n - 1)
+                lo = int(np.floor(rank)); hi = min(n - 1, lo + 1)
+                w = rank - np.floor(rank)
+                return x[lo] * (1 - w) + x[hi] * w
+            exp = pq(np.abs(vv - pq(vv, 0.5)), 0.5)
+            assert abs(r["value"] - exp) <= 1e-12 * max(1, abs(exp))
