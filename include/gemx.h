@@ -367,6 +367,16 @@ int gemx_prom_quantile(gemx_shard *, int64_t start_time, int64_t end_time,
                        int64_t range_ns, int64_t step_ns, int is_mad,
                        double q, gemx_rate_row *out_host, uint64_t cap,
                        uint64_t *n_out, gemx_query_stats *stats);
+
+/* holt_winters (executor/agg_func_prom.go:700-760): double-exponential
+ * smoothing over the window's time-ordered points (sf/tf in [0,1];
+ * <2 points emits nothing; NaN/Inf anywhere -> NaN). Same window cap
+ * and synchronous contract as gemx_prom_quantile. Points sharing one
+ * timestamp within a window have ambiguous order (documented). */
+int gemx_prom_holt(gemx_shard *, int64_t start_time, int64_t end_time,
+                   int64_t range_ns, int64_t step_ns, double sf, double tf,
+                   gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,
+                   gemx_query_stats *stats);
 int gemx_prom_over_time(gemx_shard *, int64_t start_time, int64_t end_time,
                         int64_t range_ns, int64_t step_ns, int func,
                         gemx_rate_row *out_host, uint64_t cap, uint64_t *n_out,

@@ -2176,6 +2176,7 @@ struct RateSeriesQ {
 #define GEMX_PF_ABSENT_OT 15
 #define GEMX_PF_QUANTILE 16
 #define GEMX_PF_MAD 17
+#define GEMX_PF_HOLT 18
 
 __device__ __forceinline__ void d_kahan_inc(double inc, double &sum, double &c) {
   /* executor.KahanSumInc */
@@ -2291,7 +2292,7 @@ __global__ void __launch_bounds__(256) k_rate_scan(
      * counters and, in the fill phase, value scatter via qoff offsets */
     const RateSeriesQ *__restrict__ rsq_dev, uint32_t *__restrict__ qcnt,
     const uint64_t *__restrict__ qoff, double *__restrict__ qvals,
-    DevErr *err) {
+    int64_t *__restrict__ qts, DevErr *err) {
   uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
   uint32_t stride = FAST ? gridDim.x * blockDim.x : nlanes;
   if (!FAST && gid >= nlanes) return;
@@ -2307,7 +2308,7 @@ __global__ void __launch_bounds__(256) k_rate_scan(
      * prefix path writes every field it reads at flush, so count+reset
      * suffice; the wider reducers read-accumulate everything and need
      * the full struct cleared. */
-    if (FUNC == GEMX_PF_QUANTILE || FUNC == GEMX_PF_MAD) {
+    if (FUNC >= GEMX_PF_QUANTILE) {
       /* bucket collect: no partial slots used */
     } else if (FUNC == GEMX_PF_RATE || FUNC == GEMX_PF_IRATE) {
       for (uint32_t k = 0; k < rq.n_steps; k++) {
@@ -2475,7 +2476,7 @@ __global__ void __launch_bounds__(256) k_rate_scan(
       if (!valid) continue;
       if (fv != fv) continue; /* FilterRangeNANPoint */
 
-      if (FUNC == GEMX_PF_QUANTILE || FUNC == GEMX_PF_MAD) {
+      if (FUNC >= GEMX_PF_QUANTILE) {
         /* scatter this point into every sample-step bucket whose window
          * [ts-range, ts] contains it (≤ range/step+1 ≤ ring-bound steps) */
         const RateSeriesQ sq2 = rsq_dev[rq.series_idx];
@@ -2490,7 +2491,10 @@ __global__ void __launch_bounds__(256) k_rate_scan(
           if (t < ts2 - range_ns || t > ts2) continue;
           uint64_t bucket = sq2.out_base + (uint64_t)(o - sq2.s_min);
           uint32_t idx = atomicAdd(&qcnt[bucket], 1u);
-          if (qvals) qvals[qoff[bucket] + idx] = fv;
+          if (qvals) {
+            qvals[qoff[bucket] + idx] = fv;
+            if (qts) qts[qoff[bucket] + idx] = t;
+          }
         }
         continue;
       }
@@ -2699,6 +2703,93 @@ __global__ void __launch_bounds__(256) k_quantile_final(
     out.value = res;
     out.isnil = 0;
     if (threadIdx.x == 0) rows[b] = out;
+    __syncthreads();
+  }
+}
+
+/* holt_winters finalize (CalcHoltWinters + calcTrendValue,
+ * executor/agg_func_prom.go:700-760): one block per bucket sorts the
+ * collected (t, v) pairs by time in LDS, then thread 0 runs the
+ * sequential double-exponential smoothing. <2 points emits nothing;
+ * any NaN/Inf in the window -> NaN. Points sharing one timestamp within
+ * a window have ambiguous order (the reference's order is its stream
+ * order, unrecoverable from an unordered collect) — documented. */
+__global__ void __launch_bounds__(256) k_holt_final(
+    const RateSeriesQ *__restrict__ series, uint32_t nseries,
+    const uint32_t *__restrict__ qcnt, const uint64_t *__restrict__ qoff,
+    const double *__restrict__ qvals, const int64_t *__restrict__ qts,
+    gemx_rate_row *__restrict__ rows, uint64_t total_rows,
+    int64_t start_sample, int64_t step_ns, double sf, double tf,
+    DevErr *__restrict__ err) {
+#pragma clang fp contract(off)
+  __shared__ double sv[4096];
+  __shared__ int64_t st[4096];
+  for (uint64_t b = blockIdx.x; b < total_rows; b += gridDim.x) {
+    uint32_t lo = 0, hi = nseries - 1;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi + 1) >> 1;
+      if (series[mid].out_base <= b) lo = mid;
+      else hi = mid - 1;
+    }
+    const RateSeriesQ s = series[lo];
+    int64_t o = s.s_min + (int64_t)(b - s.out_base);
+    gemx_rate_row out;
+    out.sid = s.sid;
+    out.ts = start_sample + o * step_ns;
+    out.value = 0;
+    out.isnil = 1;
+    memset(out._pad, 0, sizeof(out._pad));
+    uint32_t n = qcnt[b];
+    if (n < 2 || n > 4096) {
+      if (threadIdx.x == 0) {
+        __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+        rows[b] = out;
+      }
+      __syncthreads();
+      continue;
+    }
+    const uint64_t off = qoff[b];
+    for (uint32_t i = threadIdx.x; i < n; i += blockDim.x) {
+      sv[i] = qvals[off + i];
+      st[i] = qts[off + i];
+    }
+    __syncthreads();
+    for (uint32_t round = 0; round < n; round++) {
+      uint32_t par = round & 1;
+      for (uint32_t i = threadIdx.x; 2 * i + 1 + par < n; i += blockDim.x) {
+        uint32_t a = 2 * i + par, c = a + 1;
+        if (st[a] > st[c]) {
+          int64_t tt = st[a];
+          st[a] = st[c];
+          st[c] = tt;
+          double x = sv[a];
+          sv[a] = sv[c];
+          sv[c] = x;
+        }
+      }
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+      int bad = 0;
+      for (uint32_t i = 0; i < n; i++)
+        if (isnan(sv[i]) || isinf(sv[i])) bad = 1;
+      if (bad) {
+        out.value = nan("");
+      } else {
+        double s0h = 0, s1h = sv[0], bh = sv[1] - sv[0];
+        for (uint32_t i = 1; i < n; i++) {
+          double x = sf * sv[i];
+          if (i - 1 != 0) bh = tf * (s1h - s0h) + (1 - tf) * bh;
+          double y = (1 - sf) * (s1h + bh);
+          s0h = s1h;
+          s1h = x + y;
+        }
+        out.value = s1h;
+      }
+      out.isnil = 0;
+      rows[b] = out;
+    }
     __syncthreads();
   }
 }
@@ -3996,7 +4087,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                           int is_counter, int func, gemx_rate_row *out_host,
                           uint64_t cap, uint64_t *n_out,
                           gemx_query_stats *stats, int async_begin = 0,
-                          double scalar = 0.0) {
+                          double scalar = 0.0, double scalar2 = 0.0) {
   if (!s) return GEMX_E_INVALID;
   if (async_begin && func >= GEMX_PF_QUANTILE) {
     seterr("quantile/mad are synchronous (host prefix-sum phase)");
@@ -4138,10 +4229,11 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
              ev2 = s->ev_r[slot][2];
   HIP_CHECK(hipEventRecord(ev0, s->stream));
   const int TPB = 256;
-  const bool qmode = (func == GEMX_PF_QUANTILE || func == GEMX_PF_MAD);
+  const bool qmode = (func >= GEMX_PF_QUANTILE);
   uint32_t *d_qcnt = nullptr;
   uint64_t *d_qoff = nullptr;
   double *d_qvals = nullptr;
+  int64_t *d_qts = nullptr;
   if (qmode) {
     /* two collect phases: count per (series, step) bucket, then value
      * scatter via host prefix-sum offsets; the finalize kernel sorts
@@ -4155,7 +4247,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                        dim3(TPB), 0, s->stream, s->d_blob, s->d_descs,         \
                        P.d_rsegq, LIST, NSEG, P.d_rpart, start_sample,         \
                        eff_step, range_ns, SCR, SPL, LANES, P.d_rsq, d_qcnt,   \
-                       d_qoff, d_qvals, d_err)
+                       d_qoff, d_qvals, d_qts, d_err)
     if (!s->fast_ids.empty()) {
       uint32_t n = (uint32_t)s->fast_ids.size();
       uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
@@ -4193,6 +4285,8 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     HIP_CHECK(hipMemcpyAsync(d_qoff, hoff.data(), sizeof(uint64_t) * nb,
                              hipMemcpyHostToDevice, s->stream));
     HIP_CHECK(hipMalloc(&d_qvals, sizeof(double) * (acc2 ? acc2 : 1)));
+    if (func == GEMX_PF_HOLT)
+      HIP_CHECK(hipMalloc(&d_qts, sizeof(int64_t) * (acc2 ? acc2 : 1)));
     HIP_CHECK(hipMemsetAsync(d_qcnt, 0, sizeof(uint32_t) * nb, s->stream));
     if (!s->fast_ids.empty()) {
       uint32_t n = (uint32_t)s->fast_ids.size();
@@ -4215,7 +4309,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                        s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
                        s->d_fast_ids, n, P.d_rpart, start_sample, eff_step,     \
                        range_ns, nullptr, 0, 0, nullptr, nullptr, nullptr,      \
-                       nullptr, d_err)
+                       nullptr, nullptr, d_err)
     switch (func) {
     case GEMX_PF_SUM_OT: LAUNCH_RATE_FAST(GEMX_PF_SUM_OT); break;
     case GEMX_PF_COUNT_OT: LAUNCH_RATE_FAST(GEMX_PF_COUNT_OT); break;
@@ -4243,7 +4337,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                        s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
                        s->d_general_ids, n, P.d_rpart, start_sample, eff_step,  \
                        range_ns, P.d_scratch, scratch_per_lane, P.gen_lanes,    \
-                       nullptr, nullptr, nullptr, nullptr, d_err)
+                       nullptr, nullptr, nullptr, nullptr, nullptr, d_err)
     switch (func) {
     case GEMX_PF_SUM_OT: LAUNCH_RATE_GEN(GEMX_PF_SUM_OT); break;
     case GEMX_PF_COUNT_OT: LAUNCH_RATE_GEN(GEMX_PF_COUNT_OT); break;
@@ -4266,11 +4360,17 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   HIP_CHECK(hipEventRecord(ev1, s->stream));
   if (qmode && P.total_rows > 0) {
     uint32_t blocks = (uint32_t)std::min<uint64_t>(P.total_rows, 65535);
-    hipLaunchKernelGGL(k_quantile_final, dim3(blocks), dim3(256), 0,
-                       s->stream, P.d_rsq, (uint32_t)P.rsq.size(), d_qcnt,
-                       d_qoff, d_qvals, P.d_rrows2[slot], P.total_rows,
-                       start_sample, eff_step, func == GEMX_PF_MAD ? 1 : 0,
-                       scalar, d_err);
+    if (func == GEMX_PF_HOLT)
+      hipLaunchKernelGGL(k_holt_final, dim3(blocks), dim3(256), 0, s->stream,
+                         P.d_rsq, (uint32_t)P.rsq.size(), d_qcnt, d_qoff,
+                         d_qvals, d_qts, P.d_rrows2[slot], P.total_rows,
+                         start_sample, eff_step, scalar, scalar2, d_err);
+    else
+      hipLaunchKernelGGL(k_quantile_final, dim3(blocks), dim3(256), 0,
+                         s->stream, P.d_rsq, (uint32_t)P.rsq.size(), d_qcnt,
+                         d_qoff, d_qvals, P.d_rrows2[slot], P.total_rows,
+                         start_sample, eff_step, func == GEMX_PF_MAD ? 1 : 0,
+                         scalar, d_err);
   } else if (P.total_rows > 0) {
     uint32_t blocks =
         (uint32_t)std::min<uint64_t>((P.total_rows + TPB - 1) / TPB, 65535);
@@ -4303,6 +4403,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     hipFree(d_qcnt);
     hipFree(d_qoff);
     if (d_qvals) hipFree(d_qvals);
+    if (d_qts) hipFree(d_qts);
     return rcq;
   }
   if (async_begin) {
@@ -4865,6 +4966,23 @@ extern "C" int gemx_prom_quantile(gemx_shard *s, int64_t start_time,
                         cap, n_out, stats, 0, q);
 }
 
+/* holt_winters (executor/agg_func_prom.go:700-760): double-exponential
+ * smoothing over the window's time-ordered points; sf/tf in [0,1];
+ * <2 points emits nothing, NaN/Inf anywhere -> NaN. Same 4096-point
+ * window cap and synchronous contract as gemx_prom_quantile. */
+extern "C" int gemx_prom_holt(gemx_shard *s, int64_t start_time,
+                              int64_t end_time, int64_t range_ns,
+                              int64_t step_ns, double sf, double tf,
+                              gemx_rate_row *out_host, uint64_t cap,
+                              uint64_t *n_out, gemx_query_stats *stats) {
+  if (sf < 0 || sf > 1 || tf < 0 || tf > 1) {
+    seterr("holt_winters factors must be within [0, 1]");
+    return GEMX_E_INVALID;
+  }
+  return prom_rate_impl(s, start_time, end_time, range_ns, step_ns, 0, 0,
+                        GEMX_PF_HOLT, out_host, cap, n_out, stats, 0, sf, tf);
+}
+
 extern "C" int gemx_prom_rate(gemx_shard *s, int64_t start_time, int64_t end_time,
                               int64_t range_ns, int64_t step_ns, int is_rate,
                               int is_counter, gemx_rate_row *out_host,
@@ -4892,7 +5010,7 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
                                    int64_t step_ns, int func,
                                    gemx_rate_row *out_host, uint64_t cap,
                                    uint64_t *n_out, gemx_query_stats *stats) {
-  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_MAD) {
+  if (func < GEMX_PF_SUM_OT || func > GEMX_PF_HOLT) {
     seterr("unknown over_time func");
     return GEMX_E_INVALID;
   }

@@ -222,6 +222,12 @@ def _load():
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
         C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(_Stats),
     ]
+    lib.gemx_prom_holt.restype = C.c_int
+    lib.gemx_prom_holt.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_double,
+        C.c_double, C.c_void_p, C.c_uint64, C.POINTER(C.c_uint64),
+        C.POINTER(_Stats),
+    ]
     lib.gemx_prom_quantile.restype = C.c_int
     lib.gemx_prom_quantile.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
@@ -807,6 +813,25 @@ class Shard:
             self._h, start_time, end_time, range_ns, step_ns,
             1 if is_mad else 0, float(q),
             out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n), C.byref(st),
+        )
+        _check(rc, lib)
+        return out[: n.value], dict(
+            decode_ms=st.decode_ms, merge_ms=st.merge_ms,
+            total_ms=st.total_ms, n_rows=st.n_rows)
+
+    def prom_holt(self, start_time, end_time, range_ns, step_ns, sf, tf,
+                  out_cap=None):
+        """holt_winters over range vectors."""
+        lib = self._lib
+        if out_cap is None:
+            out_cap = self._rate_cap(start_time, end_time, range_ns, step_ns)
+        out = self._pooled_out("rate", out_cap, RATE_ROW_DTYPE)
+        n = C.c_uint64(0)
+        st = _Stats()
+        rc = lib.gemx_prom_holt(
+            self._h, start_time, end_time, range_ns, step_ns, float(sf),
+            float(tf), out.ctypes.data_as(C.c_void_p), out_cap, C.byref(n),
+            C.byref(st),
         )
         _check(rc, lib)
         return out[: n.value], dict(

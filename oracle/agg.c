@@ -1164,8 +1164,8 @@ int64_t orc_prom_over_time(const uint8_t *blob, int64_t blob_len,
                            int64_t start, int64_t end, int64_t range_ns,
                            int64_t step_ns, int func, orc_rate_row *out,
                            int64_t cap) {
-  return orc_prom_over_time_s(blob, blob_len, descs, nsegs, start, end,
-                              range_ns, step_ns, func, 0.0, out, cap);
+  return orc_prom_over_time_s2(blob, blob_len, descs, nsegs, start, end,
+                               range_ns, step_ns, func, 0.0, 0.0, out, cap);
 }
 
 int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
@@ -1173,7 +1173,18 @@ int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
                              int64_t start, int64_t end, int64_t range_ns,
                              int64_t step_ns, int func, double scalar,
                              orc_rate_row *out, int64_t cap) {
-  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 17) return -1;
+  return orc_prom_over_time_s2(blob, blob_len, descs, nsegs, start, end,
+                               range_ns, step_ns, func, scalar, 0.0, out,
+                               cap);
+}
+
+int64_t orc_prom_over_time_s2(const uint8_t *blob, int64_t blob_len,
+                              const orc_seg_desc *descs, int64_t nsegs,
+                              int64_t start, int64_t end, int64_t range_ns,
+                              int64_t step_ns, int func, double scalar,
+                              double scalar2, orc_rate_row *out,
+                              int64_t cap) {
+  if (step_ns < 0 || range_ns <= 0 || func < 2 || func > 18) return -1;
   int64_t start_sample = start + range_ns;
   int64_t end_sample =
       (step_ns == 0) ? start_sample
@@ -1308,6 +1319,27 @@ int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
           for (int64_t k = pi + 1; k < pj; k++)
             if (sv[k] < sv[k - 1]) cc++;
           v = (double)cc;
+          break;
+        }
+        case 18: { /* holt_winters (CalcHoltWinters + calcTrendValue,
+                      agg_func_prom.go:700-760): sf=scalar, tf=scalar2;
+                      sequential over the window's time-ordered values;
+                      <2 points emits nothing; NaN/Inf anywhere -> NaN */
+          if (n < 2) goto skip_emit;
+          double sf = scalar, tf = scalar2;
+          int badv = 0;
+          for (int64_t k2 = pi; k2 < pj; k2++)
+            if (sv[k2] != sv[k2] || isinf(sv[k2])) badv = 1;
+          if (badv) { v = 0.0 / 0.0; break; }
+          double s0h = 0, s1h = sv[pi], bh = sv[pi + 1] - sv[pi];
+          for (int64_t k2 = 1; k2 < n; k2++) {
+            double x = sf * sv[pi + k2];
+            if (k2 - 1 != 0) bh = tf * (s1h - s0h) + (1 - tf) * bh;
+            double y = (1 - sf) * (s1h + bh);
+            s0h = s1h;
+            s1h = x + y;
+          }
+          v = s1h;
           break;
         }
         case 16:   /* quantile_over_time (executor.CalcQuantile:

@@ -137,6 +137,10 @@ def load():
     lib.orc_prom_over_time_s.argtypes = [u8p, i64, C.c_void_p, i64, i64, i64,
                                          i64, i64, C.c_int, C.c_double,
                                          C.c_void_p, i64]
+    lib.orc_prom_over_time_s2.restype = i64
+    lib.orc_prom_over_time_s2.argtypes = [u8p, i64, C.c_void_p, i64, i64, i64,
+                                          i64, i64, C.c_int, C.c_double,
+                                          C.c_double, C.c_void_p, i64]
     lib.orc_prom_over_time.restype = i64
     lib.orc_prom_over_time.argtypes = [u8p, i64, C.c_void_p, i64, i64, i64, i64, i64,
                                        C.c_int, C.c_void_p, i64]
@@ -426,6 +430,26 @@ def prom_quantile(blob, descs, start, end, range_ns, step_ns, q=0.5,
         out.ctypes.data_as(C.c_void_p), cap,
     )
     assert n >= 0, "oracle prom_quantile failed"
+    return out[:n]
+
+
+def prom_holt(blob, descs, start, end, range_ns, step_ns, sf, tf, cap=None):
+    """holt_winters (CalcHoltWinters)."""
+    lib = get()
+    bts = np.frombuffer(blob, dtype=np.uint8)
+    d = np.ascontiguousarray(descs, dtype=SEG_DESC_DTYPE)
+    if cap is None:
+        nsteps = 1 if step_ns == 0 else (
+            int((end - (start + range_ns)) // step_ns + 2)
+            if end >= start + range_ns else 1)
+        cap = nsteps * len(np.unique(d["sid"])) + 16
+    out = np.zeros(cap, dtype=RATE_ROW_DTYPE)
+    n = lib.orc_prom_over_time_s2(
+        _u8(bts), len(bts), d.ctypes.data_as(C.c_void_p), len(d),
+        start, end, range_ns, step_ns, 18, float(sf), float(tf),
+        out.ctypes.data_as(C.c_void_p), cap,
+    )
+    assert n >= 0, "oracle prom_holt failed"
     return out[:n]
 
 

@@ -243,6 +243,11 @@ int64_t orc_prom_irate(const uint8_t *blob, int64_t blob_len,
  * count, 8 stdvar, 9 stddev, 10 present (sequential Kahan-Welford for
  * stdvar/stddev),
  * 4 avg, 5 min, 6 max, 7 last over [ts-range, ts] windows. */
+int64_t orc_prom_over_time_s2(const uint8_t *blob, int64_t blob_len,
+                              const orc_seg_desc *descs, int64_t nsegs,
+                              int64_t start, int64_t end, int64_t range_ns,
+                              int64_t step_ns, int func, double scalar,
+                              double scalar2, orc_rate_row *out, int64_t cap);
 int64_t orc_prom_over_time_s(const uint8_t *blob, int64_t blob_len,
                              const orc_seg_desc *descs, int64_t nsegs,
                              int64_t start, int64_t end, int64_t range_ns,

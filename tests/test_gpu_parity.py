@@ -1703,3 +1703,33 @@ class TestPromFamilyFuzz:
                                   ref["value"].view(np.uint64))
         finally:
             sh2.close()
+
+    def test_holt_winters_parity(self):
+        S = 10**9
+        blob, descs = orc.gen_shard(2510, 150, 1000)
+        sh = gpu_shard(blob, descs, F)
+        try:
+            gpu, _ = sh.prom_holt(0, 999 * S, 300 * S, 60 * S, 0.3, 0.6)
+            gpu = gpu.copy()
+        finally:
+            sh.close()
+        ref = orc.prom_holt(blob, descs, 0, 999 * S, 300 * S, 60 * S,
+                            0.3, 0.6)
+        assert len(gpu) == len(ref)
+        assert np.array_equal(gpu["ts"], ref["ts"])
+        assert np.array_equal(gpu["value"].view(np.uint64),
+                              ref["value"].view(np.uint64))
+        # multi-segment windows (distinct timestamps: bit-exact order)
+        rng = np.random.default_rng(2511)
+        blob2, descs2, _ = build_shard(rng, F, range(1, 31), null_frac=0.05)
+        sh2 = gpu_shard(blob2, descs2, F)
+        try:
+            gpu, _ = sh2.prom_holt(0, 800 * S, 240 * S, 60 * S, 0.5, 0.2)
+            gpu = gpu.copy()
+        finally:
+            sh2.close()
+        ref = orc.prom_holt(blob2, descs2, 0, 800 * S, 240 * S, 60 * S,
+                            0.5, 0.2)
+        assert len(gpu) == len(ref)
+        assert np.array_equal(gpu["value"].view(np.uint64),
+                              ref["value"].view(np.uint64))

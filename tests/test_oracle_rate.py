@@ -305,3 +305,25 @@ class TestStdVarPresentOverTime:
                 return x[lo] * (1 - w) + x[hi] * w
             exp = pq(np.abs(vv - pq(vv, 0.5)), 0.5)
             assert abs(r["value"] - exp) <= 1e-12 * max(1, abs(exp))
+
+    def test_holt_winters_matches_numpy(self):
+        import shard_helpers as sh
+        rng = np.random.default_rng(36)
+        blob, descs, truth = sh.build_shard(rng, sh.F, [1], null_frac=0.0)
+        S = 10**9
+        rows = orc.prom_holt(blob, descs, 0, 700 * S, 180 * S, 60 * S,
+                             0.3, 0.6)
+        assert len(rows) > 0
+        for r in rows[:8]:
+            at, av, ax = truth[1]
+            m = (at >= r["ts"] - 180 * S) & (at <= r["ts"]) & ax
+            vv = av[m]
+            assert len(vv) >= 2
+            s0, s1, b = 0.0, vv[0], vv[1] - vv[0]
+            for i in range(1, len(vv)):
+                x = 0.3 * vv[i]
+                if i - 1 != 0:
+                    b = 0.6 * (s1 - s0) + 0.4 * b
+                y = 0.7 * (s1 + b)
+                s0, s1 = s1, x + y
+            assert abs(r["value"] - s1) <= 1e-9 * max(1, abs(s1))
