@@ -1733,3 +1733,37 @@ class TestPromFamilyFuzz:
         assert len(gpu) == len(ref)
         assert np.array_equal(gpu["value"].view(np.uint64),
                               ref["value"].view(np.uint64))
+
+
+class TestConcurrentHandles:
+    def test_two_shards_two_threads(self):
+        """Two shard handles queried concurrently from two host threads
+        (ctypes releases the GIL): per-handle streams must isolate."""
+        import threading
+
+        import opengemini_amd as gx
+
+        blob1, descs1 = orc.gen_shard(2701, 400, 1000)
+        blob2, descs2 = orc.gen_shard(2702, 300, 1000)
+        sh1 = gx.Shard(blob1, descs1, F)
+        sh2 = gx.Shard(blob2, descs2, F)
+        ref1 = orc.scan_agg(blob1, descs1, F, 0, 2**62, INT)
+        ref2 = orc.scan_agg(blob2, descs2, F, 0, 2**62, INT)
+        errs = []
+
+        def worker(sh, ref):
+            try:
+                for _ in range(20):
+                    rows, _ = sh.scan_agg(0, 2**62, INT)
+                    assert len(rows) == len(ref)
+                    assert np.array_equal(rows["count"], ref["count"])
+                    assert np.array_equal(rows["min"].view(np.uint64),
+                                          ref["min"].view(np.uint64))
+            except Exception as e:  # surface across the thread boundary
+                errs.append(e)
+
+        t1 = threading.Thread(target=worker, args=(sh1, ref1))
+        t2 = threading.Thread(target=worker, args=(sh2, ref2))
+        t1.start(); t2.start(); t1.join(); t2.join()
+        sh1.close(); sh2.close()
+        assert not errs, errs
