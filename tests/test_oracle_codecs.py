@@ -302,3 +302,47 @@ class TestWindow:
 
     def test_no_interval(self):
         assert orc.window(55, 7, 1000, 0) == (7, 1001)
+
+
+class TestCodecFuzz:
+    def test_gorilla_roundtrip_fuzz(self):
+        rng = np.random.default_rng(555)
+        for trial in range(40):
+            n = int(rng.integers(1, 1200))
+            mode = trial % 4
+            if mode == 0:
+                vals = np.round(np.cumsum(rng.normal(0, 1, n)) * 128) / 128
+            elif mode == 1:
+                vals = rng.random(n) * 10 ** rng.integers(-6, 9)
+            elif mode == 2:
+                vals = np.full(n, float(rng.normal()))
+            else:
+                vals = np.repeat(rng.normal(0, 5, max(1, n // 10)), 10)[:n]
+            enc = orc.gorilla_encode(vals)
+            dec = orc.gorilla_decode(bytes(enc), n)
+            assert np.array_equal(np.asarray(dec), vals), trial
+
+    def test_int_time_roundtrip_fuzz(self):
+        rng = np.random.default_rng(556)
+        for trial in range(40):
+            n = int(rng.integers(1, 1200))
+            hi = int(rng.choice([5, 1000, 2**30, 2**59, 2**62]))
+            vals = rng.integers(-hi, hi, n).astype(np.int64)
+            if trial % 3 == 0:
+                vals = np.arange(n, dtype=np.int64) * int(
+                    rng.integers(1, 1000)) + int(rng.integers(-10**6, 10**6))
+            seg = orc.encode_data_segment(orc.ORC_TYPE_INT, vals, None, n, 0)
+            got, bm, rows, nil = orc.decode_data_segment(
+                orc.ORC_TYPE_INT, bytes(seg), max(n, 1))
+            assert rows == n and nil == 0
+            assert np.array_equal(
+                np.asarray(got)[:n].view(np.int64), vals), trial
+            # times: mixture of regular and irregular grids
+            if trial % 2 == 0:
+                tt = np.arange(n, dtype=np.int64) * int(
+                    rng.integers(1, 10**10)) + int(rng.integers(0, 10**15))
+            else:
+                tt = np.cumsum(rng.integers(1, 10**9, n)).astype(np.int64)
+            tseg = orc.encode_time_segment(tt)
+            tdec = orc.decode_time_segment(bytes(tseg), max(n, 1))
+            assert np.array_equal(np.asarray(tdec), tt), trial
