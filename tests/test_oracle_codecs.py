@@ -313,7 +313,7 @@ class TestCodecFuzz:
             if mode == 0:
                 vals = np.round(np.cumsum(rng.normal(0, 1, n)) * 128) / 128
             elif mode == 1:
-                vals = rng.random(n) * 10 ** rng.integers(-6, 9)
+                vals = rng.random(n) * 10.0 ** float(rng.integers(-6, 9))
             elif mode == 2:
                 vals = np.full(n, float(rng.normal()))
             else:
