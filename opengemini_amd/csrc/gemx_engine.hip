@@ -792,12 +792,12 @@ __global__ void __launch_bounds__(256) k_scan_fast(
     TimeIter ti;
     {
       const uint8_t *tseg = blob + d.time_offset;
-      if (tseg[0] == 17) { /* BlockIntegerOne */
+      if (tseg[0] == 18) { /* BlockIntegerOne */
         ti.kind = 1;
         ti.cur = (int64_t)d_u64le(tseg + 1);
         ti.delta = 0;
         ti.left = 1;
-      } else if (tseg[0] == 31 && d.time_size > 5) {
+      } else if (tseg[0] == 32 && d.time_size > 5) {
         if (ti.init(tseg + 5, d.time_size - 5)) {
           set_err(err, GEMX_E_DECODE);
           return;
@@ -1267,7 +1267,7 @@ __global__ void __launch_bounds__(256) k_scan_general(
       if (tlen < 5) { set_err(err, GEMX_E_DECODE); return; }
       const uint8_t *enc = tseg + 5;
       int64_t elen = tlen - 5;
-      if (tseg[0] == 17) { /* BlockIntegerOne */
+      if (tseg[0] == 18) { /* BlockIntegerOne */
         memcpy(&tbuf[0], tseg + 1, 8);
       } else {
         int tag = enc[0] >> 4;
@@ -2354,12 +2354,12 @@ __global__ void __launch_bounds__(256) k_rate_scan(
 
     if (FAST) {
       const uint8_t *tseg = blob + d.time_offset;
-      if (tseg[0] == 17) {
+      if (tseg[0] == 18) {
         ti.kind = 1;
         ti.cur = (int64_t)d_u64le(tseg + 1);
         ti.delta = 0;
         ti.left = 1;
-      } else if (tseg[0] == 31 && d.time_size > 5) {
+      } else if (tseg[0] == 32 && d.time_size > 5) {
         if (ti.init(tseg + 5, d.time_size - 5)) { set_err(err, GEMX_E_DECODE); return; }
       } else { set_err(err, GEMX_E_DECODE); return; }
       if (parse_data_header(blob + d.data_offset, d.data_size, GEMX_TYPE_FLOAT, &h)) {
@@ -2381,7 +2381,7 @@ __global__ void __launch_bounds__(256) k_rate_scan(
         const uint8_t *tseg = blob + d.time_offset;
         int64_t tlen = d.time_size;
         if (tlen < 5) { set_err(err, GEMX_E_DECODE); return; }
-        if (tseg[0] == 17) {
+        if (tseg[0] == 18) {
           memcpy(&tbuf[0], tseg + 1, 8);
         } else {
           const uint8_t *enc = tseg + 5;
@@ -3301,9 +3301,9 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
   *grid = false;
   uint8_t dt = ds[0];
   /* time: one-value or Full + {const-delta, simple8b, uncompressed} */
-  if (ts[0] == 17) {
+  if (ts[0] == 18) {
     *grid = true; /* single row: delta 0 */
-  } else if (ts[0] == 31) {
+  } else if (ts[0] == 32) {
     if (d.time_size < 6) return GEMX_E_INVALID;
     int ttag = ts[5] >> 4;
     if (ttag == 3) *fast = false; /* snappy times need scratch */

@@ -32,6 +32,17 @@
 
 namespace gemxw {
 
+/* Block band tags (lib/encoding/encoding.go:46-65): One/Full/Empty tags
+ * are sequential ordinals Float=1, Int=2, Bool=3, String=4 within each
+ * band (Float64One=17, IntegerOne=18, ... IntegerFull=32, Float64Empty=41),
+ * NOT base + influx field type. */
+static inline uint8_t w_blk_ord(int col_type) {
+  return col_type == GEMX_TYPE_FLOAT ? 1 : col_type == GEMX_TYPE_INT ? 2 : 3;
+}
+static inline uint8_t w_blk_one(int col_type) { return 16 + w_blk_ord(col_type); }
+static inline uint8_t w_blk_full(int col_type) { return 30 + w_blk_ord(col_type); }
+static inline uint8_t w_blk_empty(int col_type) { return 40 + w_blk_ord(col_type); }
+
 static inline void wput_u32be(uint8_t *p, uint32_t v) {
   p[0] = (uint8_t)(v >> 24);
   p[1] = (uint8_t)(v >> 16);
@@ -424,19 +435,19 @@ static int64_t w_data_segment(int col_type, const void *vals,
   if (col_type != GEMX_TYPE_INT && col_type != GEMX_TYPE_FLOAT) return -1;
   if (rows == 1 && val_bytes > 0) { /* one-row (column_builder.go:489-491) */
     if (cap < 1 + val_bytes) return -1;
-    dst[0] = (uint8_t)(16 + col_type);
+    dst[0] = w_blk_one(col_type);
     memcpy(dst + 1, vals, (size_t)val_bytes);
     return 1 + val_bytes;
   }
   int64_t p = 0;
   if (nil_count == 0) { /* full */
     if (cap < 5) return -1;
-    dst[0] = (uint8_t)(30 + col_type);
+    dst[0] = w_blk_full(col_type);
     wput_u32be(dst + 1, (uint32_t)rows);
     p = 5;
   } else if (nil_count == rows) { /* empty */
     if (cap < 5) return -1;
-    dst[0] = (uint8_t)(40 + col_type);
+    dst[0] = w_blk_empty(col_type);
     wput_u32be(dst + 1, (uint32_t)rows);
     return 5;
   } else { /* mixed: [type][bmLen][bitmap][bmOffset=0][nilCount] */
@@ -459,14 +470,14 @@ static int64_t w_data_segment(int col_type, const void *vals,
 
 static int64_t w_time_segment(const int64_t *times, int rows, uint8_t *dst,
                               int64_t cap) {
-  if (rows == 1) { /* BlockIntegerOne (chunkdata_builder.go:91) */
+  if (rows == 1) { /* BlockIntegerOne = 18 (chunkdata_builder.go:91) */
     if (cap < 9) return -1;
-    dst[0] = 16 + GEMX_TYPE_INT;
+    dst[0] = w_blk_one(GEMX_TYPE_INT);
     memcpy(dst + 1, times, 8);
     return 9;
   }
   if (cap < 5) return -1;
-  dst[0] = 30 + GEMX_TYPE_INT;
+  dst[0] = w_blk_full(GEMX_TYPE_INT); /* BlockIntegerFull = 32 */
   wput_u32be(dst + 1, (uint32_t)rows);
   int64_t enc = w_time_encode(times, rows, dst + 5, cap - 5);
   if (enc < 0) return -1;

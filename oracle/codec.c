@@ -983,7 +983,7 @@ int64_t orc_encode_data_segment(int col_type, const void *vals, const uint8_t *b
   /* one-row fast path: column_builder.go:489-491 + :226-228 */
   if (rows == 1 && val_bytes > 0 && val_bytes < 16) {
     if (cap < 1 + val_bytes) return -1;
-    dst[0] = (uint8_t)(ORC_BLOCK_ONE_BASE + col_type);
+    dst[0] = ORC_BLOCK_ONE(col_type);
     memcpy(dst + 1, vals, (size_t)val_bytes);
     return 1 + val_bytes;
   }
@@ -992,12 +992,12 @@ int64_t orc_encode_data_segment(int col_type, const void *vals, const uint8_t *b
   if (nil_count == 0) {
     /* full: EncodeColumnHeader rewrite, column_builder.go:428-436, :493-501 */
     if (cap < 5) return -1;
-    dst[0] = (uint8_t)(ORC_BLOCK_FULL_BASE + col_type);
+    dst[0] = ORC_BLOCK_FULL(col_type);
     put_u32be(dst + 1, (uint32_t)rows);
     p = 5;
   } else if (nil_count == rows) {
     if (cap < 5) return -1;
-    dst[0] = (uint8_t)(ORC_BLOCK_EMPTY_BASE + col_type);
+    dst[0] = ORC_BLOCK_EMPTY(col_type);
     put_u32be(dst + 1, (uint32_t)rows);
     return 5; /* empty: no data encoded (Encoding of len 0 in = no-op) */
   } else {
@@ -1026,12 +1026,12 @@ int64_t orc_encode_time_segment(const int64_t *times, int rows, uint8_t *dst,
   /* chunkdata_builder.go:91-95 */
   if (rows == 1) {
     if (cap < 9) return -1;
-    dst[0] = ORC_BLOCK_ONE_BASE + ORC_TYPE_INT;
+    dst[0] = ORC_BLOCK_ONE(ORC_TYPE_INT); /* BlockIntegerOne = 18 */
     memcpy(dst + 1, times, 8);
     return 9;
   }
   if (cap < 5) return -1;
-  dst[0] = ORC_BLOCK_FULL_BASE + ORC_TYPE_INT;
+  dst[0] = ORC_BLOCK_FULL(ORC_TYPE_INT); /* BlockIntegerFull = 32 */
   put_u32be(dst + 1, (uint32_t)rows);
   int64_t enc = orc_time_encode(times, rows, dst + 5, cap - 5);
   if (enc < 0) return -1;
@@ -1115,13 +1115,13 @@ int orc_decode_data_segment(int col_type, const uint8_t *seg, int64_t len, void 
 int orc_decode_time_segment(const uint8_t *seg, int64_t len, int64_t *times, int *rows) {
   /* reader.go:638-672 appendTimeColumnData */
   if (len < 1) return -1;
-  if (seg[0] == ORC_BLOCK_ONE_BASE + ORC_TYPE_INT) {
+  if (seg[0] == ORC_BLOCK_ONE(ORC_TYPE_INT)) {
     if (len < 9) return -1;
     memcpy(times, seg + 1, 8);
     *rows = 1;
     return 0;
   }
-  if (seg[0] != ORC_BLOCK_FULL_BASE + ORC_TYPE_INT) return -1;
+  if (seg[0] != ORC_BLOCK_FULL(ORC_TYPE_INT)) return -1;
   if (len < 5) return -1;
   int nrows = (int)get_u32be(seg + 1);
   int64_t got = orc_time_decode(seg + 5, len - 5, times, nrows);

@@ -31,10 +31,25 @@ extern "C" {
 #define ORC_TYPE_FLOAT 3
 #define ORC_TYPE_BOOLEAN 5
 
-/* ---- block type tags (lib/encoding/encoding.go:29-65) ---- */
-#define ORC_BLOCK_ONE_BASE 16   /* BlockOneBegin; One = base+type */
+/* ---- block type tags (lib/encoding/encoding.go:29-65) ----
+ * One/Full/Empty band tags are SEQUENTIAL ORDINALS, not base+field-type:
+ * Float64One=17, IntegerOne=18, BooleanOne=19, StringOne=20; Full band
+ * starts at 31, Empty at 41, in the same Float,Integer,Boolean,String
+ * order. Band membership checks are open intervals (encoding.go:67-77). */
+#define ORC_BLOCK_ONE_BASE 16   /* BlockOneBegin */
 #define ORC_BLOCK_FULL_BASE 30  /* BlockFullBegin */
 #define ORC_BLOCK_EMPTY_BASE 40 /* BlockEmptyBegin */
+
+/* band ordinal for a field type: Float=1, Int=2, Bool=3, String=4 */
+static inline int orc_blk_ord(int col_type) {
+  return col_type == ORC_TYPE_FLOAT     ? 1
+         : col_type == ORC_TYPE_INT     ? 2
+         : col_type == ORC_TYPE_BOOLEAN ? 3
+                                        : 4;
+}
+#define ORC_BLOCK_ONE(t) ((uint8_t)(ORC_BLOCK_ONE_BASE + orc_blk_ord(t)))
+#define ORC_BLOCK_FULL(t) ((uint8_t)(ORC_BLOCK_FULL_BASE + orc_blk_ord(t)))
+#define ORC_BLOCK_EMPTY(t) ((uint8_t)(ORC_BLOCK_EMPTY_BASE + orc_blk_ord(t)))
 
 /* ============== low-level codecs ============== */
 

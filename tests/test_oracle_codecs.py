@@ -204,7 +204,7 @@ class TestSegment:
         rng = np.random.default_rng(9)
         v = np.cumsum(rng.normal(0, 1, 1000))
         seg = orc.encode_data_segment(orc.ORC_TYPE_FLOAT, v, None, 1000, 0)
-        assert seg[0] == 30 + 3  # BlockFloat64Full
+        assert seg[0] == 31  # BlockFloat64Full (encoding.go:53)
         dv, bm, rows, nils = orc.decode_data_segment(orc.ORC_TYPE_FLOAT, seg)
         assert rows == 1000 and nils == 0 and np.array_equal(dv, v)
 
@@ -224,7 +224,7 @@ class TestSegment:
 
     def test_one_row(self):
         seg = orc.encode_data_segment(orc.ORC_TYPE_INT, np.array([7], dtype=np.int64), None, 1, 0)
-        assert seg[0] == 16 + 1  # BlockIntegerOne
+        assert seg[0] == 18  # BlockIntegerOne (encoding.go:47)
         dv, _, rows, nils = orc.decode_data_segment(orc.ORC_TYPE_INT, seg)
         assert rows == 1 and nils == 0 and dv[0] == 7
 
@@ -232,17 +232,17 @@ class TestSegment:
         seg = orc.encode_data_segment(
             orc.ORC_TYPE_INT, np.zeros(0, dtype=np.int64), np.zeros(2, dtype=np.uint8), 10, 10
         )
-        assert seg[0] == 40 + 1  # BlockIntegerEmpty
+        assert seg[0] == 42  # BlockIntegerEmpty (encoding.go:61)
         _, _, rows, nils = orc.decode_data_segment(orc.ORC_TYPE_INT, seg)
         assert rows == 10 and nils == 10
 
     def test_time_segment(self):
         t = np.arange(1000, dtype=np.int64) * 10**9
         seg = orc.encode_time_segment(t)
-        assert seg[0] == 30 + 1  # BlockIntegerFull
+        assert seg[0] == 32  # BlockIntegerFull (encoding.go:54)
         assert np.array_equal(orc.decode_time_segment(seg), t)
         one = orc.encode_time_segment(t[:1])
-        assert one[0] == 16 + 1 and orc.decode_time_segment(one)[0] == 0
+        assert one[0] == 18 and orc.decode_time_segment(one)[0] == 0
 
 
 class TestSnappy:

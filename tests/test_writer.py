@@ -91,7 +91,7 @@ class TestFloatRoundTrip:
         expect_check(rows, sids, times, vals, None, F)
         # walk data → gorilla (tag 3) full segments, const-delta times
         tags = seg_tags(blob, descs)
-        assert all(t == 33 and c == 3 for t, c in tags)
+        assert all(t == 31 and c == 3 for t, c in tags)  # BlockFloat64Full
 
     def test_mixed_nils(self):
         rng = np.random.default_rng(8)
@@ -111,7 +111,7 @@ class TestFloatRoundTrip:
         valid = np.zeros(10, dtype=np.uint8)
         blob, descs, rows = roundtrip(F, sids, times, vals, valid)
         assert len(rows) == 1 and rows[0]["count"] == 0
-        assert blob[descs[0]["data_offset"]] == 40 + 3  # empty block
+        assert blob[descs[0]["data_offset"]] == 41  # BlockFloat64Empty
 
     def test_single_row_series(self):
         sids = np.array([1, 2, 2], dtype=np.uint64)
@@ -119,7 +119,7 @@ class TestFloatRoundTrip:
         vals = np.array([1.5, -2.25, 4.0])
         blob, descs, rows = roundtrip(F, sids, times, vals)
         expect_check(rows, sids, times, vals, None, F)
-        assert blob[descs[0]["data_offset"]] == 16 + 3  # one-value block
+        assert blob[descs[0]["data_offset"]] == 17  # BlockFloat64One
 
     def test_same_value(self):
         sids = np.full(500, 3, dtype=np.uint64)
@@ -127,7 +127,7 @@ class TestFloatRoundTrip:
         vals = np.full(500, 2.5)
         blob, descs, rows = roundtrip(F, sids, times, vals)
         expect_check(rows, sids, times, vals, None, F)
-        assert seg_tags(blob, descs)[0] == (33, 4)  # same-value codec
+        assert seg_tags(blob, descs)[0] == (31, 4)  # same-value codec
 
     def test_nan_values_raw(self):
         # NaN collides with the gorilla terminator → writer must fall back
@@ -138,7 +138,7 @@ class TestFloatRoundTrip:
         times = np.arange(300, dtype=np.int64) * 10 ** 9
         blob, descs, rows = roundtrip(F, sids, times, vals)
         expect_check(rows, sids, times, vals, None, F)
-        assert seg_tags(blob, descs)[0] == (33, 0)  # compressedNull raw
+        assert seg_tags(blob, descs)[0] == (31, 0)  # compressedNull raw
 
     def test_inf_values_gorilla(self):
         vals = np.linspace(-5, 5, 200)
@@ -164,7 +164,7 @@ class TestIntRoundTrip:
         vals = np.arange(800, dtype=np.int64) * 7 + 3
         blob, descs, rows = roundtrip(I, sids, times, vals)
         expect_check(rows, sids, times, vals, None, I)
-        assert seg_tags(blob, descs)[0] == (31, 1)  # const-delta
+        assert seg_tags(blob, descs)[0] == (32, 1)  # BlockIntegerFull, const-delta
 
     def test_simple8b(self):
         rng = np.random.default_rng(11)
@@ -173,7 +173,7 @@ class TestIntRoundTrip:
         times = np.arange(900, dtype=np.int64) * 10 ** 9
         blob, descs, rows = roundtrip(I, sids, times, vals)
         expect_check(rows, sids, times, vals, None, I)
-        assert seg_tags(blob, descs)[0] == (31, 2)  # simple8b
+        assert seg_tags(blob, descs)[0] == (32, 2)  # BlockIntegerFull, simple8b
 
     def test_huge_deltas_raw(self):
         rng = np.random.default_rng(12)
@@ -182,7 +182,7 @@ class TestIntRoundTrip:
         times = np.arange(500, dtype=np.int64) * 10 ** 9
         blob, descs, rows = roundtrip(I, sids, times, vals)
         expect_check(rows, sids, times, vals, None, I)
-        assert seg_tags(blob, descs)[0] == (31, 4)  # uncompressed
+        assert seg_tags(blob, descs)[0] == (32, 4)  # BlockIntegerFull, uncompressed
 
     def test_int_nils(self):
         rng = np.random.default_rng(13)
@@ -199,14 +199,14 @@ class TestTimeCodecs:
     def _ttag(self, blob, d):
         off = int(d["time_offset"])
         t = blob[off]
-        return (t, blob[off + 5] >> 4 if t == 31 else None)
+        return (t, blob[off + 5] >> 4 if t == 32 else None)
 
     def test_const_delta_times(self):
         sids = np.full(100, 1, dtype=np.uint64)
         times = np.arange(100, dtype=np.int64) * 60 * 10 ** 9 + 17
         vals = np.arange(100, dtype=np.float64)
         blob, descs, rows = roundtrip(F, sids, times, vals)
-        assert self._ttag(blob, descs[0]) == (31, 1)
+        assert self._ttag(blob, descs[0]) == (32, 1)
         assert int(rows[0]["first_time"]) == 17
 
     def test_irregular_times_s8b(self):
@@ -216,7 +216,7 @@ class TestTimeCodecs:
         sids = np.full(400, 1, dtype=np.uint64)
         vals = rng.normal(0, 1, 400)
         blob, descs, rows = roundtrip(F, sids, times, vals)
-        assert self._ttag(blob, descs[0]) == (31, 2)
+        assert self._ttag(blob, descs[0]) == (32, 2)
         expect_check(rows, sids, times, vals, None, F)
 
     def test_wild_times_raw(self):
@@ -228,7 +228,7 @@ class TestTimeCodecs:
         sids = np.full(50, 1, dtype=np.uint64)
         vals = rng.normal(0, 1, 50)
         blob, descs, rows = roundtrip(F, sids, times, vals)
-        assert self._ttag(blob, descs[0]) == (31, 4)
+        assert self._ttag(blob, descs[0]) == (32, 4)
         expect_check(rows, sids, times, vals, None, F)
 
 
@@ -322,7 +322,7 @@ class TestRLEEmission:
         times = np.arange(n, dtype=np.int64) * 10 ** 9
         blob, descs, rows = roundtrip(F, sids, times, vals)
         expect_check(rows, sids, times, vals, None, F)
-        assert seg_tags(blob, descs)[0] == (33, 5)
+        assert seg_tags(blob, descs)[0] == (31, 5)
 
     def test_negative_zero_bits_survive(self):
         vals = np.array([0.0, -0.0] * 100 + [1.0] * 300)
