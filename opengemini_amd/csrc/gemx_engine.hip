@@ -3349,6 +3349,115 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
   return 0;
 }
 
+/* minimal libzstd prototypes (container ships libzstd.so.1 without headers) */
+extern "C" size_t ZSTD_decompress(void *dst, size_t dstCap, const void *src,
+                                  size_t srcSize);
+extern "C" unsigned ZSTD_isError(size_t code);
+
+/* Host-side transcode of zstd int blocks at attach.
+ *
+ * The reference selects zstd for int64 blocks that are neither const-delta
+ * nor simple8b-packable (lib/encoding/int.go:199-201; block format
+ * [3<<4][srcLen u32be][compLen u32be][zstd frame of raw LE int64s],
+ * int.go:136-166/:303-314). zstd frames are not decodable on-device, so a
+ * reference-written shard containing one such block would be unqueryable.
+ * Instead of rejecting the shard, attach rewrites each zstd block ONCE into
+ * the reference's own always-valid uncompressed form
+ * [4<<4][srcLen u32be][zigzag u64be × n] (int.go:168-177), which every
+ * kernel already decodes. The rewrite preserves segment headers and time
+ * segments byte-for-byte; only encData changes, and only for zstd blocks.
+ * Cost: one host pass over the affected segments at attach, nothing at
+ * query time. Returns 0; *rewrote says whether new_blob/new_descs are in
+ * use. GEMX_E_DECODE on a malformed frame. */
+static int transcode_zstd_segments(const uint8_t *blob, uint64_t blob_bytes,
+                                   const gemx_seg_desc *descs, uint64_t nsegs,
+                                   int col_type, std::vector<uint8_t> &new_blob,
+                                   std::vector<gemx_seg_desc> &new_descs,
+                                   bool *rewrote) {
+  *rewrote = false;
+  if (col_type != GEMX_TYPE_INT) return 0;
+  /* pass 1: is there anything to do? (common case: no — zero overhead) */
+  auto enc_of = [&](const gemx_seg_desc &d) -> const uint8_t * {
+    if (d.data_offset + d.data_size > blob_bytes || d.data_size < 6)
+      return nullptr;
+    const uint8_t *ds = blob + d.data_offset;
+    uint8_t dt = ds[0];
+    if (dt >= 30 && dt < 35) return ds + 5; /* BlockFull */
+    if (dt == (uint8_t)col_type) {          /* bitmap layout */
+      uint32_t bmlen = h_u32be(ds + 1);
+      if (d.data_size < (uint64_t)13 + bmlen + 1) return nullptr;
+      return ds + 13 + bmlen;
+    }
+    return nullptr; /* One/Empty blocks never carry zstd */
+  };
+  bool any = false;
+  for (uint64_t i = 0; i < nsegs && !any; i++) {
+    const uint8_t *enc = enc_of(descs[i]);
+    if (enc && (enc[0] >> 4) == 3) any = true;
+  }
+  if (!any) return 0;
+
+  new_blob.reserve(blob_bytes + blob_bytes / 4);
+  new_descs.assign(descs, descs + nsegs);
+  std::vector<int64_t> tmp;
+  for (uint64_t i = 0; i < nsegs; i++) {
+    const gemx_seg_desc &d = descs[i];
+    gemx_seg_desc &nd = new_descs[i];
+    /* data segment */
+    nd.data_offset = new_blob.size();
+    const uint8_t *ds = blob + d.data_offset;
+    const uint8_t *enc = enc_of(d);
+    if (enc && (enc[0] >> 4) == 3) {
+      int64_t hdr = enc - ds;
+      int64_t elen = (int64_t)d.data_size - hdr;
+      if (elen < 9) {
+        seterr("zstd int block: truncated header");
+        return GEMX_E_DECODE;
+      }
+      uint32_t src_len = h_u32be(enc + 1);
+      uint32_t comp_len = h_u32be(enc + 5);
+      if ((int64_t)comp_len + 9 > elen || (src_len & 7) != 0) {
+        seterr("zstd int block: bad srcLen/compLen");
+        return GEMX_E_DECODE;
+      }
+      tmp.resize(src_len / 8);
+      size_t dl = ZSTD_decompress(tmp.data(), src_len, enc + 9, comp_len);
+      if (ZSTD_isError(dl) || dl != src_len) {
+        seterr("zstd int block: frame decode failed");
+        return GEMX_E_DECODE;
+      }
+      /* header bytes verbatim, then the uncompressed form */
+      new_blob.insert(new_blob.end(), ds, ds + hdr);
+      size_t p = new_blob.size();
+      new_blob.resize(p + 5 + src_len);
+      uint8_t *w = new_blob.data() + p;
+      w[0] = 4 << 4;
+      w[1] = (uint8_t)(src_len >> 24);
+      w[2] = (uint8_t)(src_len >> 16);
+      w[3] = (uint8_t)(src_len >> 8);
+      w[4] = (uint8_t)src_len;
+      for (size_t k = 0; k < src_len / 8; k++) {
+        int64_t v;
+        memcpy(&v, &tmp[k], 8); /* frame holds raw LE int64s */
+        uint64_t zz = ((uint64_t)v << 1) ^ (uint64_t)(v >> 63);
+        for (int b = 0; b < 8; b++)
+          w[5 + k * 8 + b] = (uint8_t)(zz >> (56 - 8 * b));
+      }
+      nd.data_size = hdr + 5 + src_len;
+    } else {
+      new_blob.insert(new_blob.end(), ds, ds + d.data_size);
+      nd.data_size = d.data_size;
+    }
+    /* time segment verbatim */
+    nd.time_offset = new_blob.size();
+    new_blob.insert(new_blob.end(), blob + d.time_offset,
+                    blob + d.time_offset + d.time_size);
+    nd.time_size = d.time_size;
+  }
+  *rewrote = true;
+  return 0;
+}
+
 extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_bytes,
                                  const gemx_seg_desc *descs, uint64_t nsegs,
                                  int col_type, gemx_shard **out) {
@@ -3361,6 +3470,22 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     return GEMX_E_INVALID;
   }
   HIP_CHECK(hipSetDevice(device));
+  /* rewrite zstd int blocks (reference-valid, not on-device) into the
+   * uncompressed form before anything else sees the blob; no-op and
+   * zero-copy when the shard has none */
+  std::vector<uint8_t> tblob;
+  std::vector<gemx_seg_desc> tdescs;
+  bool rewrote = false;
+  {
+    int trc = transcode_zstd_segments((const uint8_t *)blob, blob_bytes, descs,
+                                      nsegs, col_type, tblob, tdescs, &rewrote);
+    if (trc != 0) return trc;
+    if (rewrote) {
+      blob = tblob.data();
+      blob_bytes = tblob.size();
+      descs = tdescs.data();
+    }
+  }
   gemx_shard *s = new gemx_shard();
   s->device = device;
   s->col_type = col_type;
@@ -3389,7 +3514,7 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     int rc = classify_segment(hb, d, col_type, &fast, &grid);
     if (rc != 0) {
       seterr(rc == GEMX_E_UNSUPPORTED
-                 ? "segment uses a codec not yet on-device (zstd/MLF/legacy)"
+                 ? "segment uses a codec not yet on-device (MLF/legacy gorilla)"
                  : "malformed segment header");
       delete s;
       return rc;

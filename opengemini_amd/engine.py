@@ -95,7 +95,7 @@ def build_extension(verbose=False):
     """Compile libgemx.so for gfx950 (hipcc cross-compiles without a GPU)."""
     cmd = [
         "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
-        "-shared", _SRC, "-o", _SO,
+        "-shared", _SRC, "-o", _SO, "-l:libzstd.so.1",
     ]
     r = subprocess.run(cmd, capture_output=True, text=True)
     if r.returncode != 0:
