@@ -1060,6 +1060,239 @@ __global__ void __launch_bounds__(256) k_scan_fast(
 #undef GEMX_DECODE_ONE
 }
 
+/* ---------------- branchless Gorilla grid kernel ----------------
+ *
+ * The headline walk/random path: one lane per nil-free, const-delta-time,
+ * gorilla-coded full segment (host routes exactly those here). Semantics
+ * identical to k_scan_fast<FLOAT,0,1> over the same segments
+ * (batch_float.go:278-514 decode; intervalIndex windowing
+ * aggregate_cursor.go:343; reduce semantics series_agg_func.gen.go).
+ *
+ * Why a separate kernel: the r1 PMC profile shows the grid kernel
+ * issue-port-bound (~140% aggregate issue demand, 55% of wave time
+ * parked) with ~3x the essential instruction count — the cost is
+ * wave-divergent dual-path execution around the data-dependent refill
+ * (`have<=64`), the availability check (`have>=need`), its staged
+ * fallback, and the per-record uvnan exit. With 64 independent segments
+ * per wave those branches diverge nearly every record, so both sides
+ * issue. This kernel removes every data-dependent branch from the
+ * record decode:
+ *  - refill is one UNCONDITIONAL 8-byte load + predicated 128-bit
+ *    funnel insert (cndmask, no branch), run twice per record (before
+ *    the 13-bit header peek and before the significant bits), which
+ *    guarantees have>=65 at each read point (max header 13, max sbits
+ *    64) with no availability check at all;
+ *  - the load needs no length check because d_blob is over-allocated by
+ *    16 zeroed bytes and reads past a segment's logical end land in the
+ *    following segment's bytes (never consumed on a well-formed stream:
+ *    decode stops after `rows` records + terminator);
+ *  - uvnan is OR-folded into a flag checked once per segment: a
+ *    mid-stream uvnan (corrupt) or a missing terminator raises
+ *    GEMX_E_DECODE after the fact instead of branching per record.
+ */
+struct GorW {
+  const uint8_t *p;      /* next unloaded word (stream is linear: the load
+                            address chain never depends on decoded bits) */
+  const uint8_t *pclamp; /* last in-bounds 8-byte load (blob + bytes + 8,
+                            inside the 16-byte pad); a corrupt stream can
+                            over-consume far past its segment, so the load
+                            address is clamped — decoded garbage is then
+                            caught by the terminator check */
+  uint64_t w0, w1, w2, w3; /* 256-bit window, bit cursor bp inside w0:
+                              any <=77-bit record at bp<=63 needs <=140
+                              bits = always inside w0..w2; w3+L are the
+                              refill pipeline */
+  uint64_t L;              /* preloaded word at p: issued ~4 records
+                              before its consumption on walk data */
+  int bp;                  /* 0..63 */
+
+  __device__ __forceinline__ uint64_t ld() {
+    uint64_t w = d_u64be(p < pclamp ? p : pclamp);
+    p += 8;
+    return w;
+  }
+  __device__ __forceinline__ void init(const uint8_t *stream,
+                                       const uint8_t *clamp) {
+    p = stream;
+    pclamp = clamp;
+    w0 = ld();
+    w1 = ld();
+    w2 = ld();
+    w3 = ld();
+    L = ld();
+    p -= 8; /* p tracks the address OF L until L is consumed */
+    bp = 0;
+  }
+  /* 64 bits starting at bit bp of (a,b); bp in [0,63] — the (>>1) split
+   * keeps the shift amount in range without a select */
+  static __device__ __forceinline__ uint64_t fun(uint64_t a, uint64_t b,
+                                                 int bp) {
+    return (a << bp) | ((b >> (63 - bp)) >> 1);
+  }
+};
+
+__global__ void __launch_bounds__(256) k_scan_grid_gor(
+    const uint8_t *__restrict__ blob, uint64_t blob_bytes,
+    const gemx_seg_desc *__restrict__ descs,
+    const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
+    uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
+    int64_t offset, DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (uint32_t li = gid; li < nseg_ids; li += gridDim.x * blockDim.x) {
+    uint32_t si = seg_ids[li];
+    const gemx_seg_desc d = descs[si];
+    const SegQ sq = segq[si];
+    if (sq.n_wins == 0) continue;
+
+    /* const-delta time (routing guarantee; timestamp.go:190) */
+    int64_t t0c, dtc;
+    {
+      const uint8_t *tseg = blob + d.time_offset;
+      if (tseg[0] == 18) {
+        t0c = (int64_t)d_u64le(tseg + 1);
+        dtc = 0;
+      } else {
+        TimeIter ti;
+        if (ti.init(tseg + 5, d.time_size - 5) || ti.kind != 1 ||
+            ti.left < (int64_t)d.rows) {
+          set_err(err, GEMX_E_DECODE);
+          return;
+        }
+        t0c = ti.cur;
+        dtc = ti.delta;
+      }
+    }
+    /* gorilla full data block (routing guarantee) */
+    SegHeader h;
+    if (parse_data_header(blob + d.data_offset, d.data_size, GEMX_TYPE_FLOAT,
+                          &h) ||
+        h.one_value || h.enc_len < 10 || (h.enc[0] >> 4) != 3) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    const int rows = (int)d.rows;
+    const uint8_t *in = h.enc + 1; /* strip the adaptive tag (float.go:89) */
+    uint64_t g_val = d_u64be(in + 1);
+    if (g_val == UVNAN) { /* empty stream but rows > 0 */
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    GorW br;
+    br.init(in + 9, blob + blob_bytes + 8);
+    uint32_t g_mean = 64, g_trail = 0;
+    uint64_t bad = 0; /* count of uvnan hits; exactly 1 (terminator) is legal */
+
+    Partial *base = partials + sq.partial_base;
+    for (uint32_t k = 0; k < sq.n_wins; k++) base[k].has_rows = 0;
+
+/* one record, straight-line: funnel-extract the 13-bit header
+ * (ctrl0+ctrl1+5 lead+6 meaningful) and the significant bits from the
+ * 256-bit window, advance the bit cursor, then shift the window by
+ * 0/1 words (predicated moves) with one predicated reload. The only
+ * branch is the rare two-word advance (record wider than 64 bits
+ * crossing a word boundary). */
+#define GOR_NEXT()                                                             \
+    do {                                                                       \
+      uint64_t A = GorW::fun(br.w0, br.w1, br.bp);                             \
+      uint32_t p13 = (uint32_t)(A >> 51);                                      \
+      uint32_t ctrl1 = p13 >> 12;                                              \
+      uint32_t neww = ctrl1 & ((p13 >> 11) & 1);                               \
+      uint32_t mr = p13 & 0x3F;                                                \
+      g_mean = neww ? (mr ? mr : 64u) : g_mean;                                \
+      g_trail = neww ? (mr ? (64u - ((p13 >> 6) & 0x1F) - mr) : 0u) : g_trail; \
+      uint32_t hdr = 1 + ctrl1 + (neww ? 11u : 0u);                            \
+      uint64_t Cc = GorW::fun(br.w1, br.w2, br.bp);                            \
+      uint64_t B = (A << hdr) | (Cc >> (64 - hdr)); /* hdr >= 1 */             \
+      uint64_t sb = (g_mean == 64) ? B : (B >> (64 - g_mean));                 \
+      g_val ^= ctrl1 ? (sb << (g_trail & 63)) : 0;                             \
+      bad += (uint64_t)(g_val == UVNAN);                                       \
+      uint32_t np = (uint32_t)br.bp + hdr + (ctrl1 ? g_mean : 0);              \
+      uint32_t adv = np >> 6;                                                  \
+      br.bp = (int)(np & 63);                                                  \
+      if (__builtin_expect(adv >= 2, 0)) { /* rare: wide record */             \
+        br.w0 = br.w2;                                                         \
+        br.w1 = br.w3;                                                         \
+        br.w2 = br.L;                                                          \
+        br.p += 8;                                                             \
+        br.w3 = br.ld();                                                       \
+        br.L = br.ld();                                                        \
+        br.p -= 8;                                                             \
+      } else {                                                                 \
+        const int c1 = (int)adv;                                               \
+        br.w0 = c1 ? br.w1 : br.w0;                                            \
+        br.w1 = c1 ? br.w2 : br.w1;                                            \
+        br.w2 = c1 ? br.w3 : br.w2;                                            \
+        br.w3 = c1 ? br.L : br.w3;                                             \
+        br.p += c1 ? 8 : 0;                                                    \
+        br.L = d_u64be(br.p < br.pclamp ? br.p : br.pclamp);                   \
+      }                                                                        \
+    } while (0)
+
+    int first_pending = 1;
+    int i = 0;
+    while (i < rows) {
+      int64_t t_i = t0c + (int64_t)i * dtc;
+      int64_t ord = win_ordinal(t_i, interval, offset);
+      if (ord < sq.w_first || ord >= sq.w_first + (int64_t)sq.n_wins) {
+        set_err(err, GEMX_E_INVALID);
+        return;
+      }
+      int64_t we = ord * interval + offset + interval;
+      int gend;
+      if (dtc == 0) {
+        gend = rows;
+      } else {
+        int64_t n_in = (we - 1 - t_i) / dtc + 1;
+        gend = (n_in >= (int64_t)(rows - i)) ? rows : i + (int)n_in;
+      }
+      if (!first_pending) GOR_NEXT();
+      first_pending = 0;
+      double fv;
+      memcpy(&fv, &g_val, 8);
+      double sf = fv, mn = fv, mx = fv, lastv = fv;
+      const double firstv = fv;
+      int min_row = i, max_row = i;
+      for (int k = i + 1; k < gend; k++) {
+        GOR_NEXT();
+        double v;
+        memcpy(&v, &g_val, 8);
+        sf += v;
+        /* first-occurrence-wins strict compares (column_util.go:204-209);
+         * NaN compares false -> never replaces (Go parity) */
+        if (mn > v) { mn = v; min_row = k; }
+        if (mx < v) { mx = v; max_row = k; }
+        lastv = v;
+      }
+      Partial tmp;
+      tmp.v[0].i = gend - i;
+      tmp.v[1].f = sf;
+      tmp.v[2].f = mn;
+      tmp.v[3].f = mx;
+      tmp.v[4].f = firstv;
+      tmp.v[5].f = lastv;
+      tmp.t[0] = t_i;
+      tmp.t[1] = t_i; /* no nils ⇒ valueIndex == row index */
+      tmp.t[2] = t0c + (int64_t)min_row * dtc;
+      tmp.t[3] = t0c + (int64_t)max_row * dtc;
+      tmp.t[4] = t_i;
+      tmp.t[5] = t0c + (int64_t)(gend - 1) * dtc;
+      tmp.first_row_time = t_i;
+      tmp.nilmask = 0;
+      tmp.has_rows = 1;
+      base[ord - sq.w_first] = tmp;
+      i = gend;
+    }
+    /* data records must be uvnan-free and the next record must be the
+     * terminator (batch_float.go:501: decode stops AT uvnan) */
+    GOR_NEXT();
+    if (bad != 1 || g_val != UVNAN) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+#undef GOR_NEXT
+  }
+}
+
 /* cross-field predicate evaluation (config #3: binaryfilterfunc compare
  * kernels, lib/binaryfilterfunc/eval_generator.gen.go:31+, applied as
  * FilterByField over a condition on a DIFFERENT field): one lane per
@@ -3104,8 +3337,8 @@ struct QueryPlan {
   bool clipped = false;
   uint32_t *d_fast_q = nullptr, *d_gen_q = nullptr;
   uint32_t n_fast_q = 0, n_gen_q = 0;
-  uint32_t *d_fastg_q = nullptr, *d_fasts_q = nullptr;
-  uint32_t n_fastg_q = 0, n_fasts_q = 0;
+  uint32_t *d_fastg_q = nullptr, *d_fasts_q = nullptr, *d_fastgor_q = nullptr;
+  uint32_t n_fastg_q = 0, n_fasts_q = 0, n_fastgor_q = 0;
   void *d_gtmp = nullptr; /* GAcc[n_gwins × gsplit] */
   uint32_t gsplit = 1, gper_chunk = 1;
   int64_t W0 = 0;
@@ -3181,9 +3414,10 @@ struct gemx_shard {
   std::vector<uint32_t> fast_ids, general_ids;
   uint32_t *d_fast_ids, *d_general_ids;
   /* fast split by time codec: grid (const-delta, dt>=0) vs streaming */
-  std::vector<uint32_t> fast_grid_ids, fast_stream_ids;
-  uint32_t *d_fast_grid_ids = nullptr, *d_fast_stream_ids = nullptr;
-  std::vector<char> is_grid; /* per segment */
+  std::vector<uint32_t> fast_grid_ids, fast_stream_ids, fast_gor_ids;
+  uint32_t *d_fast_grid_ids = nullptr, *d_fast_stream_ids = nullptr,
+           *d_fast_gor_ids = nullptr;
+  std::vector<char> is_grid, is_gor; /* per segment */
   /* series grouping: ranges in desc order */
   struct SeriesRange {
     uint64_t sid;
@@ -3264,6 +3498,7 @@ static void free_plan(QueryPlan &p) {
   if (p.d_fast_q) (void)hipFree(p.d_fast_q);
   if (p.d_gen_q) (void)hipFree(p.d_gen_q);
   if (p.d_fastg_q) (void)hipFree(p.d_fastg_q);
+  if (p.d_fastgor_q) (void)hipFree(p.d_fastgor_q);
   if (p.d_fasts_q) (void)hipFree(p.d_fasts_q);
   p = QueryPlan();
 }
@@ -3293,12 +3528,13 @@ static int h_uvarint(const uint8_t *p, int64_t len, uint64_t *out) {
  * *grid = const-delta timestamps with non-negative delta — segments the
  * leaner GRIDP=1 instantiation of k_scan_fast can take. */
 static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col_type,
-                            bool *fast, bool *grid) {
+                            bool *fast, bool *grid, bool *gor) {
   if (d.data_size < 1 || d.time_size < 1) return GEMX_E_INVALID;
   const uint8_t *ds = blob + d.data_offset;
   const uint8_t *ts = blob + d.time_offset;
   *fast = true;
   *grid = false;
+  *gor = false;
   uint8_t dt = ds[0];
   /* time: one-value or Full + {const-delta, simple8b, uncompressed} */
   if (ts[0] == 18) {
@@ -3329,6 +3565,7 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
       else if (tag == 6) return GEMX_E_UNSUPPORTED; /* MLF (config-gated off) */
       else if (tag == 1) return GEMX_E_UNSUPPORTED; /* legacy gorilla */
       else if (tag != 0 && tag != 3 && tag != 4 && tag != 5) return GEMX_E_INVALID;
+      *gor = (tag == 3); /* full gorilla block: branchless grid kernel */
     } else {
       if (tag == 3) return GEMX_E_UNSUPPORTED; /* zstd: not on device yet */
       if (tag != 1 && tag != 2 && tag != 4) return GEMX_E_INVALID;
@@ -3510,8 +3747,8 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       delete s;
       return GEMX_E_INVALID;
     }
-    bool fast, grid;
-    int rc = classify_segment(hb, d, col_type, &fast, &grid);
+    bool fast, grid, gor;
+    int rc = classify_segment(hb, d, col_type, &fast, &grid, &gor);
     if (rc != 0) {
       seterr(rc == GEMX_E_UNSUPPORTED
                  ? "segment uses a codec not yet on-device (MLF/legacy gorilla)"
@@ -3521,7 +3758,9 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     }
     if (fast) {
       s->fast_ids.push_back((uint32_t)i);
-      if (grid)
+      if (grid && gor)
+        s->fast_gor_ids.push_back((uint32_t)i);
+      else if (grid)
         s->fast_grid_ids.push_back((uint32_t)i);
       else
         s->fast_stream_ids.push_back((uint32_t)i);
@@ -3529,7 +3768,9 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       s->general_ids.push_back((uint32_t)i);
     }
     if (s->is_grid.size() < i + 1) s->is_grid.resize(nsegs, 0);
+    if (s->is_gor.size() < i + 1) s->is_gor.resize(nsegs, 0);
     s->is_grid[i] = fast && grid;
+    s->is_gor[i] = fast && grid && gor;
     s->total_rows_scanned += d.rows;
     /* series ranges + per-series/shard time bounds (for preagg coverage) */
     if (s->series_ranges.empty() || s->series_ranges.back().sid != d.sid) {
@@ -3562,9 +3803,12 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     HIP_CHECK(hipHostMalloc(&s->h_err2[sl], sizeof(DevErr)));
     HIP_CHECK(hipHostMalloc(&s->h_rerr2[sl], sizeof(DevErr)));
   }
-  HIP_CHECK(hipMalloc(&s->d_blob, blob_bytes ? blob_bytes : 1));
-  HIP_CHECK(hipMemcpyAsync(s->d_blob, blob, blob_bytes, hipMemcpyHostToDevice,
-                           s->stream));
+  /* +16 zeroed pad: k_scan_grid_gor's branchless refill does unchecked
+   * 8-byte loads that may read past the last segment's logical end */
+  HIP_CHECK(hipMalloc(&s->d_blob, blob_bytes + 16));
+  HIP_CHECK(hipMemsetAsync(s->d_blob + blob_bytes, 0, 16, s->stream));
+  HIP_CHECK(hipMemcpyAsync(s->d_blob, blob, blob_bytes ? blob_bytes : 0,
+                           hipMemcpyHostToDevice, s->stream));
   HIP_CHECK(hipMalloc(&s->d_descs, sizeof(gemx_seg_desc) * (nsegs ? nsegs : 1)));
   HIP_CHECK(hipMemcpyAsync(s->d_descs, descs, sizeof(gemx_seg_desc) * nsegs,
                            hipMemcpyHostToDevice, s->stream));
@@ -3587,6 +3831,13 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
   if (!s->fast_stream_ids.empty())
     HIP_CHECK(hipMemcpyAsync(s->d_fast_stream_ids, s->fast_stream_ids.data(),
                              sizeof(uint32_t) * s->fast_stream_ids.size(),
+                             hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&s->d_fast_gor_ids,
+                      sizeof(uint32_t) *
+                          (s->fast_gor_ids.empty() ? 1 : s->fast_gor_ids.size())));
+  if (!s->fast_gor_ids.empty())
+    HIP_CHECK(hipMemcpyAsync(s->d_fast_gor_ids, s->fast_gor_ids.data(),
+                             sizeof(uint32_t) * s->fast_gor_ids.size(),
                              hipMemcpyHostToDevice, s->stream));
   HIP_CHECK(hipMalloc(&s->d_general_ids,
                       sizeof(uint32_t) *
@@ -3617,6 +3868,7 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   (void)hipFree(s->d_fast_ids);
   (void)hipFree(s->d_fast_grid_ids);
   (void)hipFree(s->d_fast_stream_ids);
+  (void)hipFree(s->d_fast_gor_ids);
   (void)hipFree(s->d_general_ids);
   if (s->d_row_base) (void)hipFree(s->d_row_base);
   if (s->d_xbm) (void)hipFree(s->d_xbm);
@@ -3703,7 +3955,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.sq.resize(s->series_ranges.size());
     P.partial_slots = 0;
     P.total_rows = 0;
-    std::vector<uint32_t> fast_q, gen_q, fast_gq, fast_sq;
+    std::vector<uint32_t> fast_q, gen_q, fast_gq, fast_sq, fast_gorq;
     std::vector<char> is_gen(nsegs, 0);
     for (auto id : s->general_ids) is_gen[id] = 1;
     bool any_clip = false;
@@ -3757,7 +4009,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
           gen_q.push_back(i);
         } else {
           fast_q.push_back(i);
-          if (s->is_grid[i])
+          if (s->is_gor[i])
+            fast_gorq.push_back(i);
+          else if (s->is_grid[i])
             fast_gq.push_back(i);
           else
             fast_sq.push_back(i);
@@ -3804,6 +4058,13 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       if (!fast_sq.empty())
         HIP_CHECK(hipMemcpyAsync(P.d_fasts_q, fast_sq.data(),
                                  sizeof(uint32_t) * fast_sq.size(),
+                                 hipMemcpyHostToDevice, s->stream));
+      P.n_fastgor_q = (uint32_t)fast_gorq.size();
+      HIP_CHECK(hipMalloc(&P.d_fastgor_q,
+                          sizeof(uint32_t) * (fast_gorq.empty() ? 1 : fast_gorq.size())));
+      if (!fast_gorq.empty())
+        HIP_CHECK(hipMemcpyAsync(P.d_fastgor_q, fast_gorq.data(),
+                                 sizeof(uint32_t) * fast_gorq.size(),
                                  hipMemcpyHostToDevice, s->stream));
     }
     HIP_CHECK(hipMalloc(&P.d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
@@ -3887,6 +4148,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   uint32_t grid_n = (uint32_t)s->fast_grid_ids.size();
   const uint32_t *strm_list = s->d_fast_stream_ids;
   uint32_t strm_n = (uint32_t)s->fast_stream_ids.size();
+  const uint32_t *gor_list = s->d_fast_gor_ids;
+  uint32_t gor_n = (uint32_t)s->fast_gor_ids.size();
   const uint32_t *gen_list = s->d_general_ids;
   uint32_t gen_n = (uint32_t)s->general_ids.size();
   if (P.clipped) {
@@ -3896,20 +4159,28 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     grid_n = P.n_fastg_q;
     strm_list = P.d_fasts_q;
     strm_n = P.n_fasts_q;
+    gor_list = P.d_fastgor_q;
+    gor_n = P.n_fastgor_q;
     gen_list = P.d_gen_q;
     gen_n = P.n_gen_q;
   }
-  /* the lean GRIDP=1 instantiation takes const-delta-time segments when
-   * the query has an interval and no predicate; everything else fast goes
-   * through the streaming instantiation */
+  /* the branchless gorilla kernel takes const-delta-time full gorilla
+   * float segments; the lean GRIDP=1 instantiation takes the remaining
+   * const-delta-time segments; both only when the query has an interval
+   * and no predicate. Everything else fast goes through the streaming
+   * instantiation */
   const bool use_grid = (interval != 0) && (filter_op == 0);
+  /* A/B escape hatch for profiling: route gorilla segments through the
+   * generic grid kernel instead of k_scan_grid_gor */
+  static const bool no_gork = getenv("GEMX_DISABLE_GORK") != nullptr;
   struct FastLaunch {
     const uint32_t *list;
     uint32_t n;
-    int gridp;
-  } launches[2];
+    int gridp; /* 0 stream, 1 grid, 2 grid+gorilla */
+  } launches[3];
   int n_launches = 0;
   if (use_grid) {
+    if (gor_n) launches[n_launches++] = {gor_list, gor_n, no_gork ? 1 : 2};
     if (grid_n) launches[n_launches++] = {grid_list, grid_n, 1};
     if (strm_n) launches[n_launches++] = {strm_list, strm_n, 0};
   } else if (fast_n) {
@@ -3923,6 +4194,12 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     uint32_t tpb = (n < 64 * 1024) ? 64 : (uint32_t)TPB;
     uint32_t blocks = std::min<uint32_t>((n + tpb - 1) / tpb, 65535);
     const uint32_t *lst = launches[li].list;
+    if (launches[li].gridp == 2) {
+      hipLaunchKernelGGL(k_scan_grid_gor, dim3(blocks), dim3(tpb), 0,
+                         s->stream, s->d_blob, s->blob_bytes, s->d_descs,
+                         d_segq, lst, n, d_part, interval, offset, d_err);
+      continue;
+    }
 #define GEMX_LAUNCH_FAST(CT, FLT, GP)                                          \
     hipLaunchKernelGGL((k_scan_fast<CT, FLT, GP>), dim3(blocks), dim3(tpb), 0, \
                        s->stream, s->d_blob, s->d_descs, d_segq, lst, n,       \
