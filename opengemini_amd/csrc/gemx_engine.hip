@@ -717,6 +717,16 @@ struct SegQ {
   uint32_t series_idx;
 };
 
+/* per-segment gorilla arena descriptor (attach-time precompute for
+ * k_scan_grid_gor): first decoded value, the segment's word-0 index in
+ * the lane-interleaved stream arena, and the const-delta time params —
+ * the kernel never touches the on-disk segment bytes. */
+struct GorDesc {
+  uint64_t first_val;  /* bits of value 0 (batch_float.go:293) */
+  uint64_t arena_base; /* u64 index; word k of this stream at base + 64*k */
+  int64_t t0, dt;      /* const-delta time (timestamp.go:190) */
+};
+
 /* per-series output metadata */
 struct SeriesQ {
   uint64_t sid;
@@ -1090,37 +1100,37 @@ __global__ void __launch_bounds__(256) k_scan_fast(
  *    mid-stream uvnan (corrupt) or a missing terminator raises
  *    GEMX_E_DECODE after the fact instead of branching per record.
  */
-struct GorW {
-  const uint8_t *p;      /* next unloaded word (stream is linear: the load
-                            address chain never depends on decoded bits) */
-  const uint8_t *pclamp; /* last in-bounds 8-byte load (blob + bytes + 8,
-                            inside the 16-byte pad); a corrupt stream can
-                            over-consume far past its segment, so the load
-                            address is clamped — decoded garbage is then
-                            caught by the terminator check */
+struct GorA {
+  const uint64_t *p;      /* arena cursor: word k of this stream lives at
+                             base + 64*k (u64 units, 512-byte stride). No
+                             bounds checks: the arena pad
+                             (GEMX_ARENA_PAD_WORDS) covers the maximum
+                             possible overshoot of a corrupt stream, whose
+                             garbage decode is then caught by the
+                             terminator check. */
   uint64_t w0, w1, w2, w3; /* 256-bit window, bit cursor bp inside w0:
                               any <=77-bit record at bp<=63 needs <=140
-                              bits = always inside w0..w2; w3+L are the
+                              bits = always inside w0..w2; w3+L+M are the
                               refill pipeline */
-  uint64_t L;              /* preloaded word at p: issued ~4 records
-                              before its consumption on walk data */
+  uint64_t L, M;           /* 2-deep load pipeline (an 8-register batch
+                              refill and an LDS-DMA cache warmer both
+                              measured slower — r2 notes in DESIGN.md) */
   int bp;                  /* 0..63 */
 
-  __device__ __forceinline__ uint64_t ld() {
-    uint64_t w = d_u64be(p < pclamp ? p : pclamp);
-    p += 8;
+  __device__ __forceinline__ uint64_t ldw() {
+    uint64_t w = *p;
+    p += 64;
     return w;
   }
-  __device__ __forceinline__ void init(const uint8_t *stream,
-                                       const uint8_t *clamp) {
-    p = stream;
-    pclamp = clamp;
-    w0 = ld();
-    w1 = ld();
-    w2 = ld();
-    w3 = ld();
-    L = ld();
-    p -= 8; /* p tracks the address OF L until L is consumed */
+  __device__ __forceinline__ void init(const uint64_t *arena, uint64_t base) {
+    p = arena + base;
+    w0 = ldw();
+    w1 = ldw();
+    w2 = ldw();
+    w3 = ldw();
+    L = ldw();
+    M = ldw();
+    p -= 2 * 64; /* p tracks the address OF L until L is consumed */
     bp = 0;
   }
   /* 64 bits starting at bit bp of (a,b); bp in [0,63] — the (>>1) split
@@ -1131,9 +1141,14 @@ struct GorW {
   }
 };
 
+/* arena tail pad, u64 units: bounds the worst-case cursor overshoot of a
+ * corrupt stream — (max rows+1) records x <=2 word-advances x 64 u64
+ * stride, plus the prefetch lead */
+#define GEMX_ARENA_PAD_WORDS ((uint64_t)(2 * 4097 + 32) * 64)
+
 __global__ void __launch_bounds__(256) k_scan_grid_gor(
-    const uint8_t *__restrict__ blob, uint64_t blob_bytes,
-    const gemx_seg_desc *__restrict__ descs,
+    const uint64_t *__restrict__ arena, uint64_t arena_words,
+    const GorDesc *__restrict__ gors, const gemx_seg_desc *__restrict__ descs,
     const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
     uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
     int64_t offset, DevErr *err) {
@@ -1144,41 +1159,17 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
     const SegQ sq = segq[si];
     if (sq.n_wins == 0) continue;
 
-    /* const-delta time (routing guarantee; timestamp.go:190) */
-    int64_t t0c, dtc;
-    {
-      const uint8_t *tseg = blob + d.time_offset;
-      if (tseg[0] == 18) {
-        t0c = (int64_t)d_u64le(tseg + 1);
-        dtc = 0;
-      } else {
-        TimeIter ti;
-        if (ti.init(tseg + 5, d.time_size - 5) || ti.kind != 1 ||
-            ti.left < (int64_t)d.rows) {
-          set_err(err, GEMX_E_DECODE);
-          return;
-        }
-        t0c = ti.cur;
-        dtc = ti.delta;
-      }
-    }
-    /* gorilla full data block (routing guarantee) */
-    SegHeader h;
-    if (parse_data_header(blob + d.data_offset, d.data_size, GEMX_TYPE_FLOAT,
-                          &h) ||
-        h.one_value || h.enc_len < 10 || (h.enc[0] >> 4) != 3) {
-      set_err(err, GEMX_E_DECODE);
-      return;
-    }
+    const GorDesc g = gors[si];
+    const int64_t t0c = g.t0, dtc = g.dt;
     const int rows = (int)d.rows;
-    const uint8_t *in = h.enc + 1; /* strip the adaptive tag (float.go:89) */
-    uint64_t g_val = d_u64be(in + 1);
+    uint64_t g_val = g.first_val;
     if (g_val == UVNAN) { /* empty stream but rows > 0 */
       set_err(err, GEMX_E_DECODE);
       return;
     }
-    GorW br;
-    br.init(in + 9, blob + blob_bytes + 8);
+    GorA br;
+    br.init(arena, g.arena_base);
+    (void)arena_words;
     uint32_t g_mean = 64, g_trail = 0;
     uint64_t bad = 0; /* count of uvnan hits; exactly 1 (terminator) is legal */
 
@@ -1193,7 +1184,7 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
  * crossing a word boundary). */
 #define GOR_NEXT()                                                             \
     do {                                                                       \
-      uint64_t A = GorW::fun(br.w0, br.w1, br.bp);                             \
+      uint64_t A = GorA::fun(br.w0, br.w1, br.bp);                             \
       uint32_t p13 = (uint32_t)(A >> 51);                                      \
       uint32_t ctrl1 = p13 >> 12;                                              \
       uint32_t neww = ctrl1 & ((p13 >> 11) & 1);                               \
@@ -1201,7 +1192,7 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
       g_mean = neww ? (mr ? mr : 64u) : g_mean;                                \
       g_trail = neww ? (mr ? (64u - ((p13 >> 6) & 0x1F) - mr) : 0u) : g_trail; \
       uint32_t hdr = 1 + ctrl1 + (neww ? 11u : 0u);                            \
-      uint64_t Cc = GorW::fun(br.w1, br.w2, br.bp);                            \
+      uint64_t Cc = GorA::fun(br.w1, br.w2, br.bp);                            \
       uint64_t B = (A << hdr) | (Cc >> (64 - hdr)); /* hdr >= 1 */             \
       uint64_t sb = (g_mean == 64) ? B : (B >> (64 - g_mean));                 \
       g_val ^= ctrl1 ? (sb << (g_trail & 63)) : 0;                             \
@@ -1213,18 +1204,20 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
         br.w0 = br.w2;                                                         \
         br.w1 = br.w3;                                                         \
         br.w2 = br.L;                                                          \
-        br.p += 8;                                                             \
-        br.w3 = br.ld();                                                       \
-        br.L = br.ld();                                                        \
-        br.p -= 8;                                                             \
+        br.w3 = br.M;                                                          \
+        br.p += 2 * 64;                                                        \
+        br.L = br.ldw();                                                       \
+        br.M = br.ldw();                                                       \
+        br.p -= 2 * 64;                                                        \
       } else {                                                                 \
         const int c1 = (int)adv;                                               \
         br.w0 = c1 ? br.w1 : br.w0;                                            \
         br.w1 = c1 ? br.w2 : br.w1;                                            \
         br.w2 = c1 ? br.w3 : br.w2;                                            \
         br.w3 = c1 ? br.L : br.w3;                                             \
-        br.p += c1 ? 8 : 0;                                                    \
-        br.L = d_u64be(br.p < br.pclamp ? br.p : br.pclamp);                   \
+        br.L = c1 ? br.M : br.L;                                               \
+        br.p += c1 ? 64 : 0;                                                   \
+        br.M = br.p[64];                                                       \
       }                                                                        \
     } while (0)
 
@@ -3296,6 +3289,16 @@ __global__ void __launch_bounds__(256) k_rate_merge(
 static __thread char g_err[512];
 static void seterr(const char *msg) { snprintf(g_err, sizeof(g_err), "%s", msg); }
 
+static inline uint64_t h_u64be(const uint8_t *p) {
+  uint64_t v = 0;
+  for (int i = 0; i < 8; i++) v = (v << 8) | p[i];
+  return v;
+}
+static inline uint64_t h_u64le(const uint8_t *p) {
+  uint64_t v;
+  memcpy(&v, p, 8);
+  return v;
+}
 static inline uint32_t h_u32be(const uint8_t *p) {
   return ((uint32_t)p[0] << 24) | ((uint32_t)p[1] << 16) | ((uint32_t)p[2] << 8) |
          p[3];
@@ -3418,6 +3421,10 @@ struct gemx_shard {
   uint32_t *d_fast_grid_ids = nullptr, *d_fast_stream_ids = nullptr,
            *d_fast_gor_ids = nullptr;
   std::vector<char> is_grid, is_gor; /* per segment */
+  std::vector<GorDesc> h_gor;         /* per segment (zeros for non-gor) */
+  GorDesc *d_gor = nullptr;
+  uint64_t *d_arena = nullptr; /* lane-interleaved gorilla stream arena */
+  uint64_t arena_words = 0;
   /* series grouping: ranges in desc order */
   struct SeriesRange {
     uint64_t sid;
@@ -3584,6 +3591,101 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
   } else
     return GEMX_E_INVALID;
   return 0;
+}
+
+/* Lane-interleaved gorilla stream arena, built once at attach.
+ *
+ * Each gorilla bit stream is consumed serially by ONE lane, so with the
+ * on-disk layout a wave's 64 lanes load from 64 unrelated 2-8 KB regions
+ * and every wave-level load touches 64 distinct cachelines — the r2
+ * profile shows the branchless decode loop ~95% stalled on exactly that.
+ * The arena re-lays the streams for wave-coalesced access: gor segments
+ * are sorted by stream length (minimal padding), grouped 64 to a wave
+ * slot, and word k of slot-lane j lives at group_base + k*512B + j*8B —
+ * a wave whose lanes sit at the same word index loads ONE contiguous
+ * 512-byte line. Words are pre-byteswapped to big-endian-as-u64 so the
+ * kernel issues aligned 8-byte loads with no v_perm. Costs one extra
+ * copy of the compressed float streams in HBM (288 GB per GPU — the
+ * trade the hardware is built for) and a one-time host pass at attach;
+ * the on-disk blob stays authoritative for every other kernel. */
+static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
+  size_t n = s->fast_gor_ids.size();
+  s->h_gor.assign(s->nsegs, GorDesc{0, 0, 0, 0});
+  if (n == 0) return GEMX_OK;
+  struct Item {
+    uint32_t id;
+    const uint8_t *stream;
+    uint64_t bytes, words;
+  };
+  std::vector<Item> items(n);
+  for (size_t j = 0; j < n; j++) {
+    uint32_t i = s->fast_gor_ids[j];
+    const gemx_seg_desc &d = s->h_descs[i];
+    /* full block: [tag][rows u32be] [adaptive 3<<4][tsm1 1<<4][first u64be]
+     * [bitstream]  (column_builder.go:493; float.go:89; batch_float.go:289) */
+    const uint8_t *enc = blob + d.data_offset + 5;
+    uint64_t enc_len = d.data_size - 5;
+    if (enc_len < 10) {
+      seterr("gorilla block too short");
+      return GEMX_E_INVALID;
+    }
+    const uint8_t *in = enc + 1;
+    GorDesc &g = s->h_gor[i];
+    g.first_val = h_u64be(in + 1);
+    uint64_t sb = enc_len - 10;
+    items[j] = {i, in + 9, sb, (sb + 7) / 8};
+    const uint8_t *tseg = blob + d.time_offset;
+    if (tseg[0] == 18) { /* BlockIntegerOne */
+      g.t0 = (int64_t)h_u64le(tseg + 1);
+      g.dt = 0;
+    } else { /* const-delta (timestamp.go:190): [first u64be][delta uv][cnt uv] */
+      const uint8_t *tin = tseg + 6;
+      int64_t tlen = (int64_t)d.time_size - 6;
+      uint64_t dv = 0;
+      if (tlen < 9 || h_uvarint(tin + 8, tlen - 8, &dv) <= 0) {
+        seterr("gorilla segment: bad const-delta time block");
+        return GEMX_E_INVALID;
+      }
+      g.t0 = (int64_t)h_u64be(tin);
+      g.dt = (int64_t)dv;
+    }
+  }
+  std::sort(items.begin(), items.end(),
+            [](const Item &a, const Item &b) { return a.words < b.words; });
+  uint64_t total = 0; /* u64 units */
+  for (size_t g0 = 0; g0 < n; g0 += 64) {
+    size_t ge = std::min(g0 + 64, n);
+    uint64_t maxw = 1;
+    for (size_t j = g0; j < ge; j++) maxw = std::max(maxw, items[j].words);
+    for (size_t j = g0; j < ge; j++)
+      s->h_gor[items[j].id].arena_base = total + (j - g0);
+    total += maxw * 64;
+  }
+  total += GEMX_ARENA_PAD_WORDS; /* see GorA */
+  std::vector<uint64_t> h_arena(total, 0);
+  for (size_t j = 0; j < n; j++) {
+    const Item &it = items[j];
+    uint64_t *dst = h_arena.data() + s->h_gor[it.id].arena_base;
+    uint64_t full = it.bytes / 8;
+    for (uint64_t k = 0; k < full; k++) {
+      uint64_t w;
+      memcpy(&w, it.stream + k * 8, 8);
+      dst[k * 64] = __builtin_bswap64(w);
+    }
+    if (it.bytes & 7) {
+      uint64_t w = 0;
+      for (uint64_t b = full * 8; b < it.bytes; b++) w = (w << 8) | it.stream[b];
+      dst[full * 64] = w << ((8 - (it.bytes & 7)) * 8);
+    }
+  }
+  s->arena_words = total;
+  HIP_CHECK(hipMalloc(&s->d_arena, total * 8));
+  HIP_CHECK(hipMemcpy(s->d_arena, h_arena.data(), total * 8,
+                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMalloc(&s->d_gor, sizeof(GorDesc) * s->nsegs));
+  HIP_CHECK(hipMemcpy(s->d_gor, s->h_gor.data(), sizeof(GorDesc) * s->nsegs,
+                      hipMemcpyHostToDevice));
+  return GEMX_OK;
 }
 
 /* minimal libzstd prototypes (container ships libzstd.so.1 without headers) */
@@ -3791,6 +3893,13 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     }
   }
 
+  {
+    int grc = build_gor_arena(s, hb);
+    if (grc != 0) {
+      delete s;
+      return grc;
+    }
+  }
   HIP_CHECK(hipStreamCreate(&s->stream));
   HIP_CHECK(hipStreamCreate(&s->copy_stream));
   for (int sl = 0; sl < 2; sl++) {
@@ -3869,6 +3978,8 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   (void)hipFree(s->d_fast_grid_ids);
   (void)hipFree(s->d_fast_stream_ids);
   (void)hipFree(s->d_fast_gor_ids);
+  if (s->d_gor) (void)hipFree(s->d_gor);
+  if (s->d_arena) (void)hipFree(s->d_arena);
   (void)hipFree(s->d_general_ids);
   if (s->d_row_base) (void)hipFree(s->d_row_base);
   if (s->d_xbm) (void)hipFree(s->d_xbm);
@@ -4196,8 +4307,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     const uint32_t *lst = launches[li].list;
     if (launches[li].gridp == 2) {
       hipLaunchKernelGGL(k_scan_grid_gor, dim3(blocks), dim3(tpb), 0,
-                         s->stream, s->d_blob, s->blob_bytes, s->d_descs,
-                         d_segq, lst, n, d_part, interval, offset, d_err);
+                         s->stream, s->d_arena, s->arena_words, s->d_gor,
+                         s->d_descs, d_segq, lst, n, d_part, interval,
+                         offset, d_err);
       continue;
     }
 #define GEMX_LAUNCH_FAST(CT, FLT, GP)                                          \
