@@ -1286,6 +1286,143 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
   }
 }
 
+/* ---------------- lane-per-segment RAW float kernel ----------------
+ *
+ * Raw (uncompressed) float blocks — the reference's fallback when gorilla
+ * does not pay, e.g. full-entropy mantissas (lib/compress/float.go:96-99,
+ * decode :139 case 0) — consume exactly one arena word per row, so
+ * unlike gorilla the lanes of a wave stay in PERFECT lockstep and every
+ * wave-level load of the interleaved arena is one contiguous 512-byte
+ * line. Values are processed in 8-row batches, ping-pong double-buffered
+ * so a batch's registers are first read one batch (~8 rows of issue)
+ * after their loads. A wave-per-segment variant with a segmented __shfl
+ * scan measured 3x slower (57% SQ_WAIT_INST_ANY on the serialized
+ * ds_bpermute chain). Reduction semantics are the reference's
+ * (series_agg_func.gen.go): count/sum add; min/max strict
+ * first-occurrence-wins; NaN-holding blocks are routed to the sequential
+ * grid kernel at attach because Go's NaN fall-through is order-dependent.
+ */
+__global__ void __launch_bounds__(256) k_scan_raw_lane(
+    const uint64_t *__restrict__ arena, const GorDesc *__restrict__ gors,
+    const gemx_seg_desc *__restrict__ descs, const SegQ *__restrict__ segq,
+    const uint32_t *__restrict__ seg_ids, uint32_t nseg_ids,
+    Partial *__restrict__ partials, int64_t interval, int64_t offset,
+    DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (uint32_t li = gid; li < nseg_ids; li += gridDim.x * blockDim.x) {
+    uint32_t si = seg_ids[li];
+    const gemx_seg_desc d = descs[si];
+    const SegQ sq = segq[si];
+    if (sq.n_wins == 0) continue;
+    const GorDesc g = gors[si];
+    const int64_t t0c = g.t0, dtc = g.dt;
+    const int rows = (int)d.rows;
+    const uint64_t *w = arena + g.arena_base; /* row k at w[64*k] */
+
+    Partial *base = partials + sq.partial_base;
+    for (uint32_t k = 0; k < sq.n_wins; k++) base[k].has_rows = 0;
+
+    /* running window state */
+    int64_t cur_ord = INT64_MIN;
+    int64_t we_cur = INT64_MIN;
+    int64_t cnt = 0;
+    double sumf = 0, mn = 0, mx = 0, fv0 = 0, lv = 0;
+    int64_t min_t = 0, max_t = 0, grp_t = 0;
+
+#define RAW_FLUSH()                                                            \
+    do {                                                                       \
+      Partial tmp;                                                             \
+      tmp.v[0].i = cnt;                                                        \
+      tmp.v[1].f = sumf;                                                       \
+      tmp.v[2].f = mn;                                                         \
+      tmp.v[3].f = mx;                                                         \
+      tmp.v[4].f = fv0;                                                        \
+      tmp.v[5].f = lv;                                                         \
+      tmp.t[0] = grp_t;                                                        \
+      tmp.t[1] = grp_t; /* no nils: valueIndex == row index */                 \
+      tmp.t[2] = min_t;                                                        \
+      tmp.t[3] = max_t;                                                        \
+      tmp.t[4] = grp_t;                                                        \
+      tmp.t[5] = grp_t + (cnt - 1) * dtc;                                      \
+      tmp.first_row_time = grp_t;                                              \
+      tmp.nilmask = 0;                                                         \
+      tmp.has_rows = 1;                                                        \
+      base[cur_ord - sq.w_first] = tmp;                                        \
+    } while (0)
+
+#define RAW_ROW(I, BITS)                                                       \
+    do {                                                                       \
+      double v;                                                                \
+      uint64_t b_ = (BITS);                                                    \
+      memcpy(&v, &b_, 8);                                                      \
+      int64_t t = t0c + (int64_t)(I)*dtc;                                      \
+      if (t >= we_cur) { /* window change (ascending: dtc >= 0) */             \
+        if (cur_ord != INT64_MIN) RAW_FLUSH();                                 \
+        cur_ord = win_ordinal(t, interval, offset);                            \
+        we_cur = cur_ord * interval + offset + interval;                       \
+        if (cur_ord < sq.w_first ||                                            \
+            cur_ord >= sq.w_first + (int64_t)sq.n_wins) {                      \
+          set_err(err, GEMX_E_INVALID);                                        \
+          return;                                                              \
+        }                                                                      \
+        cnt = 0;                                                               \
+        sumf = 0;                                                              \
+        mn = mx = fv0 = v;                                                     \
+        min_t = max_t = grp_t = t;                                             \
+      }                                                                        \
+      cnt++;                                                                   \
+      sumf += v;                                                               \
+      if (mn > v) {                                                            \
+        mn = v;                                                                \
+        min_t = t;                                                             \
+      }                                                                        \
+      if (mx < v) {                                                            \
+        mx = v;                                                                \
+        max_t = t;                                                             \
+      }                                                                        \
+      lv = v;                                                                  \
+    } while (0)
+
+    uint64_t a0, a1, a2, a3, a4, a5, a6, a7;
+    uint64_t b0, b1, b2, b3, b4, b5, b6, b7;
+#define RAW_LOAD(R, K)                                                         \
+    R##0 = w[(K + 0) * 64];                                                    \
+    R##1 = w[(K + 1) * 64];                                                    \
+    R##2 = w[(K + 2) * 64];                                                    \
+    R##3 = w[(K + 3) * 64];                                                    \
+    R##4 = w[(K + 4) * 64];                                                    \
+    R##5 = w[(K + 5) * 64];                                                    \
+    R##6 = w[(K + 6) * 64];                                                    \
+    R##7 = w[(K + 7) * 64]
+#define RAW_USE(R, K)                                                          \
+    RAW_ROW(K + 0, R##0);                                                      \
+    RAW_ROW(K + 1, R##1);                                                      \
+    RAW_ROW(K + 2, R##2);                                                      \
+    RAW_ROW(K + 3, R##3);                                                      \
+    RAW_ROW(K + 4, R##4);                                                      \
+    RAW_ROW(K + 5, R##5);                                                      \
+    RAW_ROW(K + 6, R##6);                                                      \
+    RAW_ROW(K + 7, R##7)
+
+    int i = 0;
+    if (rows >= 16) {
+      RAW_LOAD(a, 0); /* reads past rows land in the arena pad */
+      for (; i + 16 <= rows; i += 16) {
+        RAW_LOAD(b, i + 8);
+        RAW_USE(a, i);
+        RAW_LOAD(a, i + 16);
+        RAW_USE(b, i + 8);
+      }
+    }
+    for (; i < rows; i++) RAW_ROW(i, w[i * 64]);
+    if (cur_ord != INT64_MIN) RAW_FLUSH();
+#undef RAW_LOAD
+#undef RAW_USE
+#undef RAW_ROW
+#undef RAW_FLUSH
+  }
+}
+
 /* cross-field predicate evaluation (config #3: binaryfilterfunc compare
  * kernels, lib/binaryfilterfunc/eval_generator.gen.go:31+, applied as
  * FilterByField over a condition on a DIFFERENT field): one lane per
@@ -3340,8 +3477,9 @@ struct QueryPlan {
   bool clipped = false;
   uint32_t *d_fast_q = nullptr, *d_gen_q = nullptr;
   uint32_t n_fast_q = 0, n_gen_q = 0;
-  uint32_t *d_fastg_q = nullptr, *d_fasts_q = nullptr, *d_fastgor_q = nullptr;
-  uint32_t n_fastg_q = 0, n_fasts_q = 0, n_fastgor_q = 0;
+  uint32_t *d_fastg_q = nullptr, *d_fasts_q = nullptr, *d_fastgor_q = nullptr,
+           *d_fastraw_q = nullptr;
+  uint32_t n_fastg_q = 0, n_fasts_q = 0, n_fastgor_q = 0, n_fastraw_q = 0;
   void *d_gtmp = nullptr; /* GAcc[n_gwins × gsplit] */
   uint32_t gsplit = 1, gper_chunk = 1;
   int64_t W0 = 0;
@@ -3420,7 +3558,10 @@ struct gemx_shard {
   std::vector<uint32_t> fast_grid_ids, fast_stream_ids, fast_gor_ids;
   uint32_t *d_fast_grid_ids = nullptr, *d_fast_stream_ids = nullptr,
            *d_fast_gor_ids = nullptr;
-  std::vector<char> is_grid, is_gor; /* per segment */
+  std::vector<char> is_grid, is_gor, is_raw; /* per segment */
+  std::vector<uint32_t> fast_raw_ids;
+  uint32_t *d_fast_raw_ids = nullptr;
+  int64_t max_raw_dt = 0;
   std::vector<GorDesc> h_gor;         /* per segment (zeros for non-gor) */
   GorDesc *d_gor = nullptr;
   uint64_t *d_arena = nullptr; /* lane-interleaved gorilla stream arena */
@@ -3506,6 +3647,7 @@ static void free_plan(QueryPlan &p) {
   if (p.d_gen_q) (void)hipFree(p.d_gen_q);
   if (p.d_fastg_q) (void)hipFree(p.d_fastg_q);
   if (p.d_fastgor_q) (void)hipFree(p.d_fastgor_q);
+  if (p.d_fastraw_q) (void)hipFree(p.d_fastraw_q);
   if (p.d_fasts_q) (void)hipFree(p.d_fasts_q);
   p = QueryPlan();
 }
@@ -3535,13 +3677,14 @@ static int h_uvarint(const uint8_t *p, int64_t len, uint64_t *out) {
  * *grid = const-delta timestamps with non-negative delta — segments the
  * leaner GRIDP=1 instantiation of k_scan_fast can take. */
 static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col_type,
-                            bool *fast, bool *grid, bool *gor) {
+                            bool *fast, bool *grid, bool *gor, bool *raw) {
   if (d.data_size < 1 || d.time_size < 1) return GEMX_E_INVALID;
   const uint8_t *ds = blob + d.data_offset;
   const uint8_t *ts = blob + d.time_offset;
   *fast = true;
   *grid = false;
   *gor = false;
+  *raw = false;
   uint8_t dt = ds[0];
   /* time: one-value or Full + {const-delta, simple8b, uncompressed} */
   if (ts[0] == 18) {
@@ -3573,6 +3716,8 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
       else if (tag == 1) return GEMX_E_UNSUPPORTED; /* legacy gorilla */
       else if (tag != 0 && tag != 3 && tag != 4 && tag != 5) return GEMX_E_INVALID;
       *gor = (tag == 3); /* full gorilla block: branchless grid kernel */
+      /* raw block: wave kernel (exact size so value k is at byte 8k) */
+      *raw = (tag == 0 && d.data_size == 6 + 8 * (uint64_t)d.rows);
     } else {
       if (tag == 3) return GEMX_E_UNSUPPORTED; /* zstd: not on device yet */
       if (tag != 1 && tag != 2 && tag != 4) return GEMX_E_INVALID;
@@ -3608,16 +3753,76 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
  * copy of the compressed float streams in HBM (288 GB per GPU — the
  * trade the hardware is built for) and a one-time host pass at attach;
  * the on-disk blob stays authoritative for every other kernel. */
+/* const-delta time params for one grid segment (timestamp.go:190) */
+static int host_grid_time(const uint8_t *blob, const gemx_seg_desc &d,
+                          int64_t *t0, int64_t *dt) {
+  const uint8_t *tseg = blob + d.time_offset;
+  if (tseg[0] == 18) { /* BlockIntegerOne */
+    *t0 = (int64_t)h_u64le(tseg + 1);
+    *dt = 0;
+    return 0;
+  }
+  const uint8_t *tin = tseg + 6;
+  int64_t tlen = (int64_t)d.time_size - 6;
+  uint64_t dv = 0;
+  if (tlen < 9 || h_uvarint(tin + 8, tlen - 8, &dv) <= 0) return -1;
+  *t0 = (int64_t)h_u64be(tin);
+  *dt = (int64_t)dv;
+  return 0;
+}
+
 static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
   size_t n = s->fast_gor_ids.size();
   s->h_gor.assign(s->nsegs, GorDesc{0, 0, 0, 0});
-  if (n == 0) return GEMX_OK;
+  /* raw wave-kernel segments: time params + NaN screen. Go's NaN
+   * comparison fall-through is order-dependent (no associative scan
+   * form), so raw blocks holding any NaN are demoted to the sequential
+   * grid kernel. */
+  {
+    std::vector<uint32_t> keep;
+    keep.reserve(s->fast_raw_ids.size());
+    for (uint32_t i : s->fast_raw_ids) {
+      const gemx_seg_desc &d = s->h_descs[i];
+      GorDesc &g = s->h_gor[i];
+      if (host_grid_time(blob, d, &g.t0, &g.dt) != 0) {
+        seterr("raw segment: bad const-delta time block");
+        return GEMX_E_INVALID;
+      }
+      const uint8_t *vals = blob + d.data_offset + 6;
+      bool has_nan = false;
+      for (uint32_t r = 0; r < d.rows; r++) {
+        uint64_t bits;
+        memcpy(&bits, vals + (size_t)r * 8, 8);
+        if (((bits >> 52) & 0x7FF) == 0x7FF && (bits << 12) != 0) {
+          has_nan = true;
+          break;
+        }
+      }
+      if (has_nan) {
+        s->is_raw[i] = 0;
+        s->fast_grid_ids.push_back(i);
+      } else {
+        keep.push_back(i);
+        s->max_raw_dt = std::max(s->max_raw_dt, g.dt);
+      }
+    }
+    s->fast_raw_ids.swap(keep);
+    std::sort(s->fast_grid_ids.begin(), s->fast_grid_ids.end());
+  }
+  size_t nraw = s->fast_raw_ids.size();
+  if (n + nraw > 0) { /* GorDesc used by gor AND raw */
+    HIP_CHECK(hipMalloc(&s->d_gor, sizeof(GorDesc) * s->nsegs));
+  }
+  if (n + nraw == 0) return GEMX_OK;
   struct Item {
     uint32_t id;
     const uint8_t *stream;
     uint64_t bytes, words;
+    int raw; /* raw items copy little-endian words (values), gorilla items
+                big-endian (bitstream) */
   };
-  std::vector<Item> items(n);
+  std::vector<Item> items;
+  items.reserve(n + nraw);
   for (size_t j = 0; j < n; j++) {
     uint32_t i = s->fast_gor_ids[j];
     const gemx_seg_desc &d = s->h_descs[i];
@@ -3633,28 +3838,37 @@ static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
     GorDesc &g = s->h_gor[i];
     g.first_val = h_u64be(in + 1);
     uint64_t sb = enc_len - 10;
-    items[j] = {i, in + 9, sb, (sb + 7) / 8};
-    const uint8_t *tseg = blob + d.time_offset;
-    if (tseg[0] == 18) { /* BlockIntegerOne */
-      g.t0 = (int64_t)h_u64le(tseg + 1);
-      g.dt = 0;
-    } else { /* const-delta (timestamp.go:190): [first u64be][delta uv][cnt uv] */
-      const uint8_t *tin = tseg + 6;
-      int64_t tlen = (int64_t)d.time_size - 6;
-      uint64_t dv = 0;
-      if (tlen < 9 || h_uvarint(tin + 8, tlen - 8, &dv) <= 0) {
-        seterr("gorilla segment: bad const-delta time block");
-        return GEMX_E_INVALID;
-      }
-      g.t0 = (int64_t)h_u64be(tin);
-      g.dt = (int64_t)dv;
+    items.push_back({i, in + 9, sb, (sb + 7) / 8, 0});
+    if (host_grid_time(blob, d, &g.t0, &g.dt) != 0) {
+      seterr("gorilla segment: bad const-delta time block");
+      return GEMX_E_INVALID;
     }
   }
-  std::sort(items.begin(), items.end(),
-            [](const Item &a, const Item &b) { return a.words < b.words; });
+  for (uint32_t i : s->fast_raw_ids) {
+    const gemx_seg_desc &d = s->h_descs[i];
+    uint64_t sb = (uint64_t)d.rows * 8;
+    items.push_back({i, blob + d.data_offset + 6, sb, sb / 8, 1});
+  }
+  /* sort by (kind, length) and REORDER THE LAUNCH LISTS to match: lane j
+   * of a wave must sit in arena slot j, or the 512-byte interleave never
+   * coalesces (launching in segment-id order scatters slots) */
+  std::sort(items.begin(), items.end(), [](const Item &a, const Item &b) {
+    if (a.raw != b.raw) return a.raw < b.raw;
+    if (a.words != b.words) return a.words < b.words;
+    return a.id < b.id;
+  });
+  {
+    size_t gi = 0, ri = 0;
+    for (const Item &it : items)
+      if (it.raw)
+        s->fast_raw_ids[ri++] = it.id;
+      else
+        s->fast_gor_ids[gi++] = it.id;
+  }
   uint64_t total = 0; /* u64 units */
-  for (size_t g0 = 0; g0 < n; g0 += 64) {
-    size_t ge = std::min(g0 + 64, n);
+  size_t nall = items.size();
+  for (size_t g0 = 0; g0 < nall; g0 += 64) {
+    size_t ge = std::min(g0 + 64, nall);
     uint64_t maxw = 1;
     for (size_t j = g0; j < ge; j++) maxw = std::max(maxw, items[j].words);
     for (size_t j = g0; j < ge; j++)
@@ -3663,26 +3877,34 @@ static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
   }
   total += GEMX_ARENA_PAD_WORDS; /* see GorA */
   std::vector<uint64_t> h_arena(total, 0);
-  for (size_t j = 0; j < n; j++) {
+  for (size_t j = 0; j < nall; j++) {
     const Item &it = items[j];
     uint64_t *dst = h_arena.data() + s->h_gor[it.id].arena_base;
     uint64_t full = it.bytes / 8;
-    for (uint64_t k = 0; k < full; k++) {
-      uint64_t w;
-      memcpy(&w, it.stream + k * 8, 8);
-      dst[k * 64] = __builtin_bswap64(w);
-    }
-    if (it.bytes & 7) {
-      uint64_t w = 0;
-      for (uint64_t b = full * 8; b < it.bytes; b++) w = (w << 8) | it.stream[b];
-      dst[full * 64] = w << ((8 - (it.bytes & 7)) * 8);
+    if (it.raw) {
+      for (uint64_t k = 0; k < full; k++) {
+        uint64_t w;
+        memcpy(&w, it.stream + k * 8, 8);
+        dst[k * 64] = w; /* values stay little-endian */
+      }
+    } else {
+      for (uint64_t k = 0; k < full; k++) {
+        uint64_t w;
+        memcpy(&w, it.stream + k * 8, 8);
+        dst[k * 64] = __builtin_bswap64(w);
+      }
+      if (it.bytes & 7) {
+        uint64_t w = 0;
+        for (uint64_t b = full * 8; b < it.bytes; b++)
+          w = (w << 8) | it.stream[b];
+        dst[full * 64] = w << ((8 - (it.bytes & 7)) * 8);
+      }
     }
   }
   s->arena_words = total;
   HIP_CHECK(hipMalloc(&s->d_arena, total * 8));
   HIP_CHECK(hipMemcpy(s->d_arena, h_arena.data(), total * 8,
                       hipMemcpyHostToDevice));
-  HIP_CHECK(hipMalloc(&s->d_gor, sizeof(GorDesc) * s->nsegs));
   HIP_CHECK(hipMemcpy(s->d_gor, s->h_gor.data(), sizeof(GorDesc) * s->nsegs,
                       hipMemcpyHostToDevice));
   return GEMX_OK;
@@ -3849,8 +4071,8 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       delete s;
       return GEMX_E_INVALID;
     }
-    bool fast, grid, gor;
-    int rc = classify_segment(hb, d, col_type, &fast, &grid, &gor);
+    bool fast, grid, gor, raw;
+    int rc = classify_segment(hb, d, col_type, &fast, &grid, &gor, &raw);
     if (rc != 0) {
       seterr(rc == GEMX_E_UNSUPPORTED
                  ? "segment uses a codec not yet on-device (MLF/legacy gorilla)"
@@ -3862,6 +4084,8 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       s->fast_ids.push_back((uint32_t)i);
       if (grid && gor)
         s->fast_gor_ids.push_back((uint32_t)i);
+      else if (grid && raw)
+        s->fast_raw_ids.push_back((uint32_t)i);
       else if (grid)
         s->fast_grid_ids.push_back((uint32_t)i);
       else
@@ -3871,8 +4095,10 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     }
     if (s->is_grid.size() < i + 1) s->is_grid.resize(nsegs, 0);
     if (s->is_gor.size() < i + 1) s->is_gor.resize(nsegs, 0);
+    if (s->is_raw.size() < i + 1) s->is_raw.resize(nsegs, 0);
     s->is_grid[i] = fast && grid;
     s->is_gor[i] = fast && grid && gor;
+    s->is_raw[i] = fast && grid && raw && !gor;
     s->total_rows_scanned += d.rows;
     /* series ranges + per-series/shard time bounds (for preagg coverage) */
     if (s->series_ranges.empty() || s->series_ranges.back().sid != d.sid) {
@@ -3948,6 +4174,13 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     HIP_CHECK(hipMemcpyAsync(s->d_fast_gor_ids, s->fast_gor_ids.data(),
                              sizeof(uint32_t) * s->fast_gor_ids.size(),
                              hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&s->d_fast_raw_ids,
+                      sizeof(uint32_t) *
+                          (s->fast_raw_ids.empty() ? 1 : s->fast_raw_ids.size())));
+  if (!s->fast_raw_ids.empty())
+    HIP_CHECK(hipMemcpyAsync(s->d_fast_raw_ids, s->fast_raw_ids.data(),
+                             sizeof(uint32_t) * s->fast_raw_ids.size(),
+                             hipMemcpyHostToDevice, s->stream));
   HIP_CHECK(hipMalloc(&s->d_general_ids,
                       sizeof(uint32_t) *
                           (s->general_ids.empty() ? 1 : s->general_ids.size())));
@@ -3978,6 +4211,7 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   (void)hipFree(s->d_fast_grid_ids);
   (void)hipFree(s->d_fast_stream_ids);
   (void)hipFree(s->d_fast_gor_ids);
+  (void)hipFree(s->d_fast_raw_ids);
   if (s->d_gor) (void)hipFree(s->d_gor);
   if (s->d_arena) (void)hipFree(s->d_arena);
   (void)hipFree(s->d_general_ids);
@@ -4066,7 +4300,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.sq.resize(s->series_ranges.size());
     P.partial_slots = 0;
     P.total_rows = 0;
-    std::vector<uint32_t> fast_q, gen_q, fast_gq, fast_sq, fast_gorq;
+    std::vector<uint32_t> fast_q, gen_q, fast_gq, fast_sq, fast_gorq, fast_rawq;
     std::vector<char> is_gen(nsegs, 0);
     for (auto id : s->general_ids) is_gen[id] = 1;
     bool any_clip = false;
@@ -4122,6 +4356,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
           fast_q.push_back(i);
           if (s->is_gor[i])
             fast_gorq.push_back(i);
+          else if (s->is_raw[i] && interval > 0 &&
+                   63 * s->h_gor[i].dt < 8 * interval)
+            fast_rawq.push_back(i); /* wave kernel's compare-chain bound */
           else if (s->is_grid[i])
             fast_gq.push_back(i);
           else
@@ -4142,6 +4379,13 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     }
     P.clipped = any_clip;
     if (any_clip) {
+      /* arena kernels need launch order == arena slot order for the
+       * 512-byte interleave to coalesce (see build_gor_arena) */
+      auto by_slot = [&](uint32_t x, uint32_t y) {
+        return s->h_gor[x].arena_base < s->h_gor[y].arena_base;
+      };
+      std::sort(fast_gorq.begin(), fast_gorq.end(), by_slot);
+      std::sort(fast_rawq.begin(), fast_rawq.end(), by_slot);
       P.n_fast_q = (uint32_t)fast_q.size();
       P.n_gen_q = (uint32_t)gen_q.size();
       HIP_CHECK(hipMalloc(&P.d_fast_q,
@@ -4176,6 +4420,13 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       if (!fast_gorq.empty())
         HIP_CHECK(hipMemcpyAsync(P.d_fastgor_q, fast_gorq.data(),
                                  sizeof(uint32_t) * fast_gorq.size(),
+                                 hipMemcpyHostToDevice, s->stream));
+      P.n_fastraw_q = (uint32_t)fast_rawq.size();
+      HIP_CHECK(hipMalloc(&P.d_fastraw_q,
+                          sizeof(uint32_t) * (fast_rawq.empty() ? 1 : fast_rawq.size())));
+      if (!fast_rawq.empty())
+        HIP_CHECK(hipMemcpyAsync(P.d_fastraw_q, fast_rawq.data(),
+                                 sizeof(uint32_t) * fast_rawq.size(),
                                  hipMemcpyHostToDevice, s->stream));
     }
     HIP_CHECK(hipMalloc(&P.d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
@@ -4261,6 +4512,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   uint32_t strm_n = (uint32_t)s->fast_stream_ids.size();
   const uint32_t *gor_list = s->d_fast_gor_ids;
   uint32_t gor_n = (uint32_t)s->fast_gor_ids.size();
+  const uint32_t *raw_list = s->d_fast_raw_ids;
+  uint32_t raw_n = (uint32_t)s->fast_raw_ids.size();
+  bool raw_ok = (interval > 0) && (63 * s->max_raw_dt < 8 * interval);
   const uint32_t *gen_list = s->d_general_ids;
   uint32_t gen_n = (uint32_t)s->general_ids.size();
   if (P.clipped) {
@@ -4272,6 +4526,9 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     strm_n = P.n_fasts_q;
     gor_list = P.d_fastgor_q;
     gor_n = P.n_fastgor_q;
+    raw_list = P.d_fastraw_q;
+    raw_n = P.n_fastraw_q;
+    raw_ok = true; /* per-segment dt bound applied when queueing */
     gen_list = P.d_gen_q;
     gen_n = P.n_gen_q;
   }
@@ -4288,10 +4545,11 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     const uint32_t *list;
     uint32_t n;
     int gridp; /* 0 stream, 1 grid, 2 grid+gorilla */
-  } launches[3];
+  } launches[4];
   int n_launches = 0;
   if (use_grid) {
     if (gor_n) launches[n_launches++] = {gor_list, gor_n, no_gork ? 1 : 2};
+    if (raw_n) launches[n_launches++] = {raw_list, raw_n, raw_ok ? 3 : 1};
     if (grid_n) launches[n_launches++] = {grid_list, grid_n, 1};
     if (strm_n) launches[n_launches++] = {strm_list, strm_n, 0};
   } else if (fast_n) {
@@ -4310,6 +4568,12 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          s->stream, s->d_arena, s->arena_words, s->d_gor,
                          s->d_descs, d_segq, lst, n, d_part, interval,
                          offset, d_err);
+      continue;
+    }
+    if (launches[li].gridp == 3) { /* lane per raw segment, arena-fed */
+      hipLaunchKernelGGL(k_scan_raw_lane, dim3(blocks), dim3(tpb), 0,
+                         s->stream, s->d_arena, s->d_gor, s->d_descs, d_segq,
+                         lst, n, d_part, interval, offset, d_err);
       continue;
     }
 #define GEMX_LAUNCH_FAST(CT, FLT, GP)                                          \
