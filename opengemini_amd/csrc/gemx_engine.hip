@@ -1385,6 +1385,8 @@ __global__ void __launch_bounds__(256) k_scan_raw_lane(
 
     uint64_t a0, a1, a2, a3, a4, a5, a6, a7;
     uint64_t b0, b1, b2, b3, b4, b5, b6, b7;
+    uint64_t c0, c1, c2, c3, c4, c5, c6, c7;
+    uint64_t e0, e1, e2, e3, e4, e5, e6, e7;
 #define RAW_LOAD(R, K)                                                         \
     R##0 = w[(K + 0) * 64];                                                    \
     R##1 = w[(K + 1) * 64];                                                    \
@@ -1405,13 +1407,23 @@ __global__ void __launch_bounds__(256) k_scan_raw_lane(
     RAW_ROW(K + 7, R##7)
 
     int i = 0;
-    if (rows >= 16) {
-      RAW_LOAD(a, 0); /* reads past rows land in the arena pad */
-      for (; i + 16 <= rows; i += 16) {
-        RAW_LOAD(b, i + 8);
+    if (rows >= 32) {
+      /* quad-buffered: a batch's registers are first read ~24 rows
+       * (three batch-uses of issue) after their loads — enough to cover
+       * an HBM-miss on 1-2 resident waves. Reads past rows land in the
+       * arena pad. */
+      RAW_LOAD(a, 0);
+      RAW_LOAD(b, 8);
+      RAW_LOAD(c, 16);
+      for (; i + 32 <= rows; i += 32) {
+        RAW_LOAD(e, i + 24);
         RAW_USE(a, i);
-        RAW_LOAD(a, i + 16);
+        RAW_LOAD(a, i + 32);
         RAW_USE(b, i + 8);
+        RAW_LOAD(b, i + 40);
+        RAW_USE(c, i + 16);
+        RAW_LOAD(c, i + 48);
+        RAW_USE(e, i + 24);
       }
     }
     for (; i < rows; i++) RAW_ROW(i, w[i * 64]);
