@@ -203,6 +203,49 @@ def gorilla_decode(buf, max_n=1 << 20):
     return out[:n].copy()
 
 
+def snappy_encode(data):
+    lib = get()
+    d = np.ascontiguousarray(data, dtype=np.uint8)
+    cap = int(lib.orc_snappy_max_encoded_len(len(d))) + 16
+    out = np.zeros(cap, dtype=np.uint8)
+    n = lib.orc_snappy_encode(_u8(d), len(d), _u8(out), cap)
+    if n < 0:
+        raise ValueError("snappy encode failed")
+    return out[:n].tobytes()
+
+
+def snappy_decode(buf, out_len):
+    lib = get()
+    b = np.ascontiguousarray(buf, dtype=np.uint8)
+    out = np.zeros(out_len + 16, dtype=np.uint8)
+    n = lib.orc_snappy_decode(_u8(b), len(b), _u8(out), out_len + 16)
+    if n < 0:
+        raise ValueError("snappy decode failed")
+    return out[:n].tobytes()
+
+
+def simple8b_encode(values):
+    lib = get()
+    v = np.ascontiguousarray(values, dtype=np.uint64).copy()  # modified in place
+    out = np.zeros(len(v) + 8, dtype=np.uint64)
+    n = lib.orc_simple8b_encode_all(
+        v.ctypes.data_as(C.POINTER(C.c_uint64)), len(v),
+        out.ctypes.data_as(C.POINTER(C.c_uint64)), len(out))
+    if n < 0:
+        raise ValueError("simple8b encode failed")
+    return out[:n].copy()
+
+
+def simple8b_decode_word(word):
+    lib = get()
+    out = np.zeros(240, dtype=np.uint64)
+    n = lib.orc_simple8b_decode(
+        C.c_uint64(word), out.ctypes.data_as(C.POINTER(C.c_uint64)))
+    if n < 0:
+        raise ValueError("bad selector")
+    return out[:n].copy()
+
+
 def float_encode(values):
     lib = get()
     v = np.ascontiguousarray(values, dtype=np.float64)
