@@ -436,6 +436,35 @@ int gemx_downsample_write(gemx_shard *, int64_t start_time, int64_t end_time,
                           uint64_t descs_cap, uint64_t *n_segs_out,
                           uint64_t *blob_bytes_out);
 
+/* record.ColVal wire view (lib/record/column.go:30-37): dense values
+ * (nils not stored), LSB-first validity bitmap (bit row&7 of byte
+ * row>>3 set = row valid), bitmap offset, row count, nil count. */
+typedef struct {
+  void *val;
+  uint8_t *bitmap;
+  int32_t bitmap_offset;
+  int32_t len;
+  int32_t nil_count;
+} gemx_colval;
+
+/* Assemble the KeyCursor output record from aggregate rows — the
+ * recFromRows step of the cgo cursor: one output row
+ * per input row, one column per op in `ops` order, plus the time
+ * column. Column values pack DENSE into val_bufs[k] (caller-allocated,
+ * >= n_rows * 8 bytes) with validity in bitmaps[k] (>= (n_rows+7)/8
+ * bytes); cols_out[k] receives the ColVal view. Nil semantics follow
+ * series_agg_func.gen.go (count: nil when 0 — countReduce:24; others:
+ * their is-nil flags). The time column (time_out, never nil) follows
+ * aggregate_cursor.go: multi-call queries (n_ops > 1) use the window's
+ * first row time (deriveIntervalIndex :371-374); single-call queries
+ * use the call's own time (series_agg_reducer.gen.go:244,269).
+ * GEMX_OP_COUNT columns are int64; other ops keep col_type. Returns
+ * GEMX_OK or GEMX_E_INVALID. */
+int gemx_rec_from_rows(const gemx_agg_row *rows, uint64_t n_rows,
+                       const int *ops, int n_ops, int col_type,
+                       void *const *val_bufs, uint8_t *const *bitmaps,
+                       int64_t *time_out, gemx_colval *cols_out);
+
 #ifdef __cplusplus
 }
 #endif

@@ -5810,3 +5810,105 @@ extern "C" int gemx_prom_over_time(gemx_shard *s, int64_t start_time,
 
 /* ---------------- write side (downsample output) ---------------- */
 #include "gemx_writer.hpp"
+
+/* recFromRows: agg rows -> record.ColVal wire layout (the cgo cursor's
+ * record assembly; see include/gemx.h for the contract and citations). */
+extern "C" int gemx_rec_from_rows(const gemx_agg_row *rows, uint64_t n_rows,
+                                  const int *ops, int n_ops, int col_type,
+                                  void *const *val_bufs,
+                                  uint8_t *const *bitmaps, int64_t *time_out,
+                                  gemx_colval *cols_out) {
+  if (!rows || !ops || n_ops <= 0 || !val_bufs || !bitmaps || !time_out ||
+      !cols_out) {
+    seterr("rec_from_rows: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  if (col_type != GEMX_TYPE_FLOAT && col_type != GEMX_TYPE_INT) {
+    seterr("rec_from_rows: bad col_type");
+    return GEMX_E_INVALID;
+  }
+  const uint64_t bm_bytes = (n_rows + 7) / 8;
+  for (int k = 0; k < n_ops; k++) {
+    const int op = ops[k];
+    if (op < GEMX_OP_COUNT || op > GEMX_OP_LAST) {
+      seterr("rec_from_rows: bad op");
+      return GEMX_E_INVALID;
+    }
+    uint8_t *bm = bitmaps[k];
+    memset(bm, 0, bm_bytes);
+    gemx_val *dst = (gemx_val *)val_bufs[k];
+    uint64_t dense = 0;
+    for (uint64_t r = 0; r < n_rows; r++) {
+      const gemx_agg_row &a = rows[r];
+      gemx_val v;
+      int isnil;
+      switch (op) {
+      case GEMX_OP_COUNT: /* nil when zero (series_agg_func.gen.go:24-32) */
+        v.i = a.count;
+        isnil = (a.count == 0);
+        break;
+      case GEMX_OP_SUM:
+        v = a.sum;
+        isnil = a.sum_isnil;
+        break;
+      case GEMX_OP_MIN:
+        v = a.minv;
+        isnil = a.min_isnil;
+        break;
+      case GEMX_OP_MAX:
+        v = a.maxv;
+        isnil = a.max_isnil;
+        break;
+      case GEMX_OP_FIRST:
+        v = a.firstv;
+        isnil = a.first_isnil;
+        break;
+      default:
+        v = a.lastv;
+        isnil = a.last_isnil;
+        break;
+      }
+      if (!isnil) {
+        dst[dense++] = v;
+        bm[r >> 3] |= (uint8_t)(1u << (r & 7));
+      }
+    }
+    cols_out[k].val = dst;
+    cols_out[k].bitmap = bm;
+    cols_out[k].bitmap_offset = 0;
+    cols_out[k].len = (int32_t)n_rows;
+    cols_out[k].nil_count = (int32_t)(n_rows - dense);
+  }
+  /* time column: multi-call = window first-row time
+   * (aggregate_cursor.go:371-374); single call = the call's own time */
+  for (uint64_t r = 0; r < n_rows; r++) {
+    const gemx_agg_row &a = rows[r];
+    int64_t t;
+    if (n_ops > 1) {
+      t = a.first_row_time;
+    } else {
+      switch (ops[0]) {
+      case GEMX_OP_COUNT:
+        t = a.count_time;
+        break;
+      case GEMX_OP_SUM:
+        t = a.sum_time;
+        break;
+      case GEMX_OP_MIN:
+        t = a.min_time;
+        break;
+      case GEMX_OP_MAX:
+        t = a.max_time;
+        break;
+      case GEMX_OP_FIRST:
+        t = a.first_time;
+        break;
+      default:
+        t = a.last_time;
+        break;
+      }
+    }
+    time_out[r] = t;
+  }
+  return GEMX_OK;
+}
