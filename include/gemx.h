@@ -413,6 +413,31 @@ int gemx_encode_shard(int col_type, const uint64_t *sids, const int64_t *times,
 int gemx_encode_bound(int col_type, uint64_t n_rows, uint32_t seg_rows,
                       uint64_t *blob_bound, uint64_t *descs_bound);
 
+/* gemx_encode_shard + write-side pre-aggregation metadata: also emits
+ * one whole-range aggregate row per series (the engine's equivalent of
+ * the reference's per-chunk FloatPreAgg/IntegerPreAgg persisted in
+ * ChunkMeta at flush time, engine/immutable/pre_aggregation.go:410,
+ * column_builder.go:233), computed host-side with the same reduce+merge
+ * semantics the scan kernels implement. Feed the rows to
+ * gemx_shard_set_preagg after attaching the written shard and
+ * gemx_scan_preagg covering queries are served with zero scans.
+ * preagg_out/n_preagg_out may be NULL to skip. preagg_cap must be >=
+ * the number of distinct sids. */
+int gemx_encode_shard_pre(int col_type, const uint64_t *sids,
+                          const int64_t *times, const void *values,
+                          const uint8_t *valid, uint64_t n_rows,
+                          uint32_t seg_rows, uint8_t *blob_out,
+                          uint64_t blob_cap, gemx_seg_desc *descs_out,
+                          uint64_t descs_cap, uint64_t *n_segs_out,
+                          uint64_t *blob_bytes_out, gemx_agg_row *preagg_out,
+                          uint64_t preagg_cap, uint64_t *n_preagg_out);
+
+/* Seed an attached shard's pre-aggregation cache from write-side
+ * metadata (one row per series, series order). The first covering
+ * gemx_scan_preagg then runs zero kernels — the write-side equivalent
+ * of the reference serving matchPreAgg straight from ChunkMeta. */
+int gemx_shard_set_preagg(gemx_shard *, const gemx_agg_row *rows, uint64_t n);
+
 /* aggregate column selector for gemx_downsample_write */
 #define GEMX_OP_COUNT 0
 #define GEMX_OP_SUM 1
@@ -435,6 +460,19 @@ int gemx_downsample_write(gemx_shard *, int64_t start_time, int64_t end_time,
                           uint64_t blob_cap, gemx_seg_desc *descs_out,
                           uint64_t descs_cap, uint64_t *n_segs_out,
                           uint64_t *blob_bytes_out);
+
+/* gemx_downsample_write + write-side pre-agg rows for the OUTPUT shard
+ * (feed to gemx_shard_set_preagg after attaching it; out_type_out =
+ * written column type: int for count, else the source type). preagg
+ * args may be NULL to skip. */
+int gemx_downsample_write_pre(gemx_shard *, int64_t start_time,
+                              int64_t end_time, int64_t interval,
+                              int64_t offset, int op, uint32_t seg_rows,
+                              uint8_t *blob_out, uint64_t blob_cap,
+                              gemx_seg_desc *descs_out, uint64_t descs_cap,
+                              uint64_t *n_segs_out, uint64_t *blob_bytes_out,
+                              gemx_agg_row *preagg_out, uint64_t preagg_cap,
+                              uint64_t *n_preagg_out, int *out_type_out);
 
 /* record.ColVal wire view (lib/record/column.go:30-37): dense values
  * (nils not stored), LSB-first validity bitmap (bit row&7 of byte

@@ -595,13 +595,231 @@ extern "C" int gemx_encode_shard(int col_type, const uint64_t *sids,
   return GEMX_OK;
 }
 
-extern "C" int gemx_downsample_write(gemx_shard *s, int64_t start_time,
-                                     int64_t end_time, int64_t interval,
-                                     int64_t offset, int op, uint32_t seg_rows,
-                                     uint8_t *blob_out, uint64_t blob_cap,
-                                     gemx_seg_desc *descs_out,
-                                     uint64_t descs_cap, uint64_t *n_segs_out,
-                                     uint64_t *blob_bytes_out) {
+/* ---- write-side pre-aggregation metadata --------------------------------
+ * The reference persists per-column pre-agg (FloatPreAgg/IntegerPreAgg:
+ * count,min,max,minT,maxT,sum) in ChunkMeta at flush time
+ * (engine/immutable/pre_aggregation.go:410, column_builder.go:233) so
+ * matchPreAgg queries never decode. The engine's equivalent metadata is
+ * one whole-range aggregate row per series (see gemx_preagg_build); this
+ * computes the SAME rows host-side from the writer's inputs — per
+ * segment-group reduce (series_agg_func.gen.go) folded with fv() merge
+ * semantics in segment order — so a written shard can be re-attached and
+ * served with zero scans. Independent restatement; the oracle is not
+ * linked. */
+struct WOpAcc {
+  int active;
+  gemx_val v;
+  int64_t time, niltime;
+};
+
+static int w_preagg_build(int col_type, const uint64_t *sids,
+                          const int64_t *times, const void *values,
+                          const uint8_t *valid, uint64_t n_rows,
+                          uint32_t seg_rows, gemx_agg_row *out, uint64_t cap,
+                          uint64_t *n_out) {
+  using namespace gemxw;
+  uint64_t pos = 0, nser = 0;
+  while (pos < n_rows) {
+    uint64_t sid = sids[pos];
+    /* series state */
+    WOpAcc a[6];
+    memset(a, 0, sizeof(a));
+    /* bug-compatible multiCall time: the reference appends the window's
+     * first-row time PER RECORD (aggregate_cursor.go:371), so a window
+     * spanning records keeps the LAST record's window start — here the
+     * last segment group's first row time (record == segment) */
+    int64_t first_row_time = times[pos];
+    while (pos < n_rows && sids[pos] == sid) {
+      /* one segment group (same chunking rule as the encoder) */
+      uint64_t end = pos;
+      while (end < n_rows && sids[end] == sid && end - pos < seg_rows) end++;
+      int rows = (int)(end - pos);
+      /* per-segment reduce, whole-segment group (start=0, end=rows):
+       * count/sum index = 0; min/max/first/last track the row index
+       * (series_agg_reducer.gen.go:206 semantics via oracle/agg.c's
+       * restatement of the same reference lines) */
+      int64_t cnt = 0;
+      double sf = 0;
+      int64_t si2 = 0;
+      int mn_row = -1, mx_row = -1, fv_row = -1, lv_row = -1;
+      gemx_val mn{}, mx{}, fv{}, lvv{};
+      for (int r = 0; r < rows; r++) {
+        if (valid && !valid[pos + r]) continue;
+        gemx_val v;
+        memcpy(&v, (const uint8_t *)values + (pos + r) * 8, 8);
+        if (cnt == 0) {
+          mn = mx = fv = v;
+          mn_row = mx_row = fv_row = r;
+        } else if (col_type == GEMX_TYPE_FLOAT) {
+          /* strict first-occurrence compares; NaN never replaces */
+          if (mn.f > v.f) {
+            mn = v;
+            mn_row = r;
+          }
+          if (mx.f < v.f) {
+            mx = v;
+            mx_row = r;
+          }
+        } else {
+          if (mn.i > v.i) {
+            mn = v;
+            mn_row = r;
+          }
+          if (mx.i < v.i) {
+            mx = v;
+            mx_row = r;
+          }
+        }
+        if (col_type == GEMX_TYPE_FLOAT)
+          sf += v.f;
+        else
+          si2 += v.i;
+        lvv = v;
+        lv_row = r;
+        cnt++;
+      }
+      const int64_t *st = times + pos;
+      first_row_time = st[0];
+      struct {
+        int isnil;
+        gemx_val v;
+        int64_t rtime;
+      } red[6];
+      red[0].isnil = (cnt == 0);
+      red[0].v.i = cnt;
+      red[0].rtime = st[0]; /* count index = group start */
+      red[1].isnil = (cnt == 0);
+      if (col_type == GEMX_TYPE_FLOAT)
+        red[1].v.f = sf;
+      else
+        red[1].v.i = si2;
+      red[1].rtime = st[0]; /* sum index = value start (bug-compatible) */
+      red[2] = {cnt == 0, mn, st[mn_row < 0 ? 0 : mn_row]};
+      red[3] = {cnt == 0, mx, st[mx_row < 0 ? 0 : mx_row]};
+      red[4] = {cnt == 0, fv, st[fv_row < 0 ? 0 : fv_row]};
+      red[5] = {cnt == 0, lvv, st[lv_row < 0 ? 0 : lv_row]};
+      /* fv() merge into the series accumulators
+       * (series_agg_func.gen.go:44-274) */
+      for (int k = 0; k < 6; k++) {
+        WOpAcc &ac = a[k];
+        if (red[k].isnil) {
+          ac.niltime = red[k].rtime;
+          continue;
+        }
+        if (!ac.active) {
+          ac.active = 1;
+          ac.v = red[k].v;
+          ac.time = red[k].rtime;
+          continue;
+        }
+        switch (k) {
+        case 0:
+          ac.v.i += red[k].v.i;
+          break;
+        case 1:
+          if (col_type == GEMX_TYPE_FLOAT)
+            ac.v.f += red[k].v.f;
+          else
+            ac.v.i += red[k].v.i;
+          break;
+        case 2:
+          if (col_type == GEMX_TYPE_FLOAT ? (red[k].v.f < ac.v.f)
+                                          : (red[k].v.i < ac.v.i)) {
+            ac.v = red[k].v;
+            ac.time = red[k].rtime;
+          }
+          break;
+        case 3:
+          if (col_type == GEMX_TYPE_FLOAT ? (red[k].v.f > ac.v.f)
+                                          : (red[k].v.i > ac.v.i)) {
+            ac.v = red[k].v;
+            ac.time = red[k].rtime;
+          }
+          break;
+        case 4: /* first: keep prev */
+          break;
+        default: /* last: assign curr */
+          ac.v = red[k].v;
+          ac.time = red[k].rtime;
+          break;
+        }
+      }
+      pos = end;
+    }
+    if (nser >= cap) {
+      seterr("preagg capacity too small");
+      return GEMX_E_CAP;
+    }
+    gemx_agg_row &o = out[nser++];
+    memset(&o, 0, sizeof(o));
+    o.sid = sid;
+    o.win_start = first_row_time; /* rewritten at serve (interval==0) */
+    o.first_row_time = first_row_time;
+    o.count = a[0].active ? a[0].v.i : 0;
+    o.count_time = a[0].active ? a[0].time : a[0].niltime;
+    o.sum = a[1].v;
+    o.sum_time = a[1].active ? a[1].time : a[1].niltime;
+    o.sum_isnil = !a[1].active;
+    o.minv = a[2].v;
+    o.min_time = a[2].active ? a[2].time : a[2].niltime;
+    o.min_isnil = !a[2].active;
+    o.maxv = a[3].v;
+    o.max_time = a[3].active ? a[3].time : a[3].niltime;
+    o.max_isnil = !a[3].active;
+    o.firstv = a[4].v;
+    o.first_time = a[4].active ? a[4].time : a[4].niltime;
+    o.first_isnil = !a[4].active;
+    o.lastv = a[5].v;
+    o.last_time = a[5].active ? a[5].time : a[5].niltime;
+    o.last_isnil = !a[5].active;
+  }
+  *n_out = nser;
+  return GEMX_OK;
+}
+
+extern "C" int gemx_encode_shard_pre(
+    int col_type, const uint64_t *sids, const int64_t *times,
+    const void *values, const uint8_t *valid, uint64_t n_rows,
+    uint32_t seg_rows, uint8_t *blob_out, uint64_t blob_cap,
+    gemx_seg_desc *descs_out, uint64_t descs_cap, uint64_t *n_segs_out,
+    uint64_t *blob_bytes_out, gemx_agg_row *preagg_out, uint64_t preagg_cap,
+    uint64_t *n_preagg_out) {
+  int rc = gemx_encode_shard(col_type, sids, times, values, valid, n_rows,
+                             seg_rows, blob_out, blob_cap, descs_out,
+                             descs_cap, n_segs_out, blob_bytes_out);
+  if (rc != 0 || !preagg_out || !n_preagg_out) return rc;
+  return w_preagg_build(col_type, sids, times, values, valid, n_rows,
+                        seg_rows, preagg_out, preagg_cap, n_preagg_out);
+}
+
+extern "C" int gemx_shard_set_preagg(gemx_shard *s, const gemx_agg_row *rows,
+                                     uint64_t n) {
+  if (!s || (!rows && n)) {
+    seterr("set_preagg: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  if (n != s->series_ranges.size()) {
+    seterr("set_preagg: row count != series count");
+    return GEMX_E_INVALID;
+  }
+  for (uint64_t g = 0; g < n; g++) {
+    if (rows[g].sid != s->series_ranges[g].sid) {
+      seterr("set_preagg: sid mismatch (rows must be in series order)");
+      return GEMX_E_INVALID;
+    }
+  }
+  s->preagg.assign(rows, rows + n);
+  s->preagg_valid = true;
+  return GEMX_OK;
+}
+
+static int ds_write_impl(gemx_shard *s, int64_t start_time,
+                         int64_t end_time, int64_t interval, int64_t offset,
+                         int op, uint32_t seg_rows, uint8_t *blob_out,
+                         uint64_t blob_cap, gemx_seg_desc *descs_out,
+                         uint64_t descs_cap, uint64_t *n_segs_out,
+                         uint64_t *blob_bytes_out, gemx_agg_row *preagg_out,
+                         uint64_t preagg_cap, uint64_t *n_preagg_out) {
   if (!s || op < GEMX_OP_COUNT || op > GEMX_OP_LAST) {
     seterr("downsample_write: bad arguments");
     return GEMX_E_INVALID;
@@ -683,7 +901,46 @@ extern "C" int gemx_downsample_write(gemx_shard *s, int64_t start_time,
       break;
     }
   }
-  return gemx_encode_shard(out_type, w_sid.data(), w_t.data(), w_v.data(),
-                           w_ok.data(), n, seg_rows, blob_out, blob_cap,
-                           descs_out, descs_cap, n_segs_out, blob_bytes_out);
+  rc = gemx_encode_shard(out_type, w_sid.data(), w_t.data(), w_v.data(),
+                          w_ok.data(), n, seg_rows, blob_out, blob_cap,
+                          descs_out, descs_cap, n_segs_out, blob_bytes_out);
+  if (rc != 0 || !preagg_out || !n_preagg_out) return rc;
+  return w_preagg_build(out_type, w_sid.data(), w_t.data(), w_v.data(),
+                        w_ok.data(), n, seg_rows, preagg_out, preagg_cap,
+                        n_preagg_out);
+}
+
+extern "C" int gemx_downsample_write(gemx_shard *s, int64_t start_time,
+                                     int64_t end_time, int64_t interval,
+                                     int64_t offset, int op, uint32_t seg_rows,
+                                     uint8_t *blob_out, uint64_t blob_cap,
+                                     gemx_seg_desc *descs_out,
+                                     uint64_t descs_cap, uint64_t *n_segs_out,
+                                     uint64_t *blob_bytes_out) {
+  return ds_write_impl(s, start_time, end_time, interval, offset, op,
+                       seg_rows, blob_out, blob_cap, descs_out, descs_cap,
+                       n_segs_out, blob_bytes_out, nullptr, 0, nullptr);
+}
+
+/* gemx_downsample_write + write-side pre-agg rows for the OUTPUT shard
+ * (see gemx_encode_shard_pre); out_type_out reports the written column
+ * type (int for count, else the source type). */
+extern "C" int gemx_downsample_write_pre(
+    gemx_shard *s, int64_t start_time, int64_t end_time, int64_t interval,
+    int64_t offset, int op, uint32_t seg_rows, uint8_t *blob_out,
+    uint64_t blob_cap, gemx_seg_desc *descs_out, uint64_t descs_cap,
+    uint64_t *n_segs_out, uint64_t *blob_bytes_out, gemx_agg_row *preagg_out,
+    uint64_t preagg_cap, uint64_t *n_preagg_out, int *out_type_out) {
+  if (!s || op < GEMX_OP_COUNT || op > GEMX_OP_LAST) {
+    seterr("downsample_write: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  if (out_type_out)
+    *out_type_out = (op == GEMX_OP_COUNT) ? GEMX_TYPE_INT : s->col_type;
+  int rc = ds_write_impl(s, start_time, end_time, interval, offset, op,
+                         seg_rows, blob_out, blob_cap, descs_out, descs_cap,
+                         n_segs_out, blob_bytes_out, preagg_out, preagg_cap,
+                         n_preagg_out);
+  if (rc == 0 && *n_segs_out == 0 && n_preagg_out) *n_preagg_out = 0;
+  return rc;
 }

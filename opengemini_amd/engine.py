@@ -141,6 +141,9 @@ def _load():
     ]
     lib.gemx_preagg_build.restype = C.c_int
     lib.gemx_preagg_build.argtypes = [C.c_void_p]
+    lib.gemx_shard_set_preagg.restype = C.c_int
+    lib.gemx_shard_set_preagg.argtypes = [C.c_void_p, C.c_void_p, C.c_uint64]
+    lib.gemx_encode_shard_pre.restype = C.c_int
     lib.gemx_scan_preagg.restype = C.c_int
     lib.gemx_scan_preagg.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64,
@@ -212,6 +215,13 @@ def _load():
         C.c_uint32, C.c_void_p, C.c_uint64, C.c_void_p, C.c_uint64,
         C.POINTER(C.c_uint64), C.POINTER(C.c_uint64),
     ]
+    lib.gemx_downsample_write_pre.restype = C.c_int
+    lib.gemx_downsample_write_pre.argtypes = [
+        C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int,
+        C.c_uint32, C.c_void_p, C.c_uint64, C.c_void_p, C.c_uint64,
+        C.POINTER(C.c_uint64), C.POINTER(C.c_uint64), C.c_void_p,
+        C.c_uint64, C.POINTER(C.c_uint64), C.POINTER(C.c_int),
+    ]
     lib.gemx_prom_rate.restype = C.c_int
     lib.gemx_prom_rate.argtypes = [
         C.c_void_p, C.c_int64, C.c_int64, C.c_int64, C.c_int64, C.c_int, C.c_int,
@@ -253,11 +263,15 @@ DOWNSAMPLE_OPS = {"count": 0, "sum": 1, "min": 2, "max": 3, "first": 4,
                   "last": 5}
 
 
-def encode_shard(col_type, sids, times, values, valid=None, seg_rows=1000):
+def encode_shard(col_type, sids, times, values, valid=None, seg_rows=1000,
+                 with_preagg=False):
     """TSSP segment writer (host-side; no GPU needed). Rows must be grouped
     by sid, times ascending within sid. Returns (blob: bytes, descs:
     np.ndarray[SEG_DESC_DTYPE]) attachable by Shard and readable by the
-    reference's segment readers. See include/gemx.h for the codec
+    reference's segment readers — plus, with with_preagg=True, the
+    write-side pre-aggregation rows (one gemx_agg_row per series;
+    pre_aggregation.go:410 role) to seed Shard.set_preagg so covering
+    preagg queries never scan. See include/gemx.h for the codec
     selection (column_builder.go / lib/encoding / lib/compress)."""
     lib = _load()
     sids = np.ascontiguousarray(sids, dtype=np.uint64)
@@ -279,6 +293,22 @@ def encode_shard(col_type, sids, times, values, valid=None, seg_rows=1000):
     descs = np.zeros(db.value, dtype=SEG_DESC_DTYPE)
     nseg = C.c_uint64(0)
     used = C.c_uint64(0)
+    if with_preagg:
+        npre_cap = len(np.unique(sids))
+        pre = np.zeros(max(npre_cap, 1), dtype=AGG_ROW_DTYPE)
+        npre = C.c_uint64(0)
+        rc = lib.gemx_encode_shard_pre(
+            col_type, sids.ctypes.data_as(C.c_void_p),
+            times.ctypes.data_as(C.c_void_p),
+            values.ctypes.data_as(C.c_void_p), vptr, n, seg_rows,
+            blob.ctypes.data_as(C.c_void_p), bb.value,
+            descs.ctypes.data_as(C.c_void_p), db.value, C.byref(nseg),
+            C.byref(used), pre.ctypes.data_as(C.c_void_p),
+            C.c_uint64(npre_cap), C.byref(npre),
+        )
+        _check(rc, lib)
+        return (blob[: used.value].tobytes(), descs[: nseg.value].copy(),
+                pre[: npre.value].copy())
     rc = lib.gemx_encode_shard(
         col_type, sids.ctypes.data_as(C.c_void_p),
         times.ctypes.data_as(C.c_void_p), values.ctypes.data_as(C.c_void_p),
@@ -512,6 +542,15 @@ class Shard:
             n_rows=st.n_rows,
         )
 
+    def set_preagg(self, rows):
+        """Seed the pre-agg cache from write-side metadata
+        (encode_shard(..., with_preagg=True) rows, series order): the
+        first covering scan_preagg then runs zero kernels."""
+        r = np.ascontiguousarray(rows, dtype=AGG_ROW_DTYPE)
+        _check(self._lib.gemx_shard_set_preagg(
+            self._h, r.ctypes.data_as(C.c_void_p), C.c_uint64(len(r))),
+            self._lib)
+
     def preagg_build(self):
         """Compute + cache per-series whole-shard pre-agg rows on the handle
         (the ColumnMeta FloatPreAgg/IntegerPreAgg role,
@@ -702,13 +741,16 @@ class Shard:
         return out[: n.value], stats
 
     def downsample_write(self, start_time, end_time, interval, offset=0,
-                         op="sum", seg_rows=1000):
+                         op="sum", seg_rows=1000, with_preagg=False):
         """Downsample end-to-end (config #4 write side): device scan +
         GROUP BY time aggregate, then re-encode the chosen aggregate
         column as a new TSSP shard (WriteIntoStorageTransform role,
         executor/record_plan.go:494). Returns (blob, descs) — attachable
-        and re-queryable. op: count/sum/min/max/first/last; count yields
-        an int64 column."""
+        and re-queryable — or with with_preagg=True
+        (blob, descs, preagg_rows, out_type): the write-side pre-agg
+        rows (pre_aggregation.go:410 role) seed Shard.set_preagg on the
+        re-attached output so covering preagg queries never scan.
+        op: count/sum/min/max/first/last; count yields an int64 column."""
         lib = self._lib
         opc = DOWNSAMPLE_OPS[op]
         n_rows = self._rows_bound(interval, offset, False)
@@ -720,6 +762,21 @@ class Shard:
         descs = np.zeros(db.value, dtype=SEG_DESC_DTYPE)
         nseg = C.c_uint64(0)
         used = C.c_uint64(0)
+        if with_preagg:
+            n_ser = len(np.unique(self._descs["sid"]))
+            pre = np.zeros(max(n_ser, 1), dtype=AGG_ROW_DTYPE)
+            npre = C.c_uint64(0)
+            out_type = C.c_int(0)
+            rc = lib.gemx_downsample_write_pre(
+                self._h, start_time, end_time, interval, offset, opc,
+                seg_rows, blob.ctypes.data_as(C.c_void_p), bb.value,
+                descs.ctypes.data_as(C.c_void_p), db.value, C.byref(nseg),
+                C.byref(used), pre.ctypes.data_as(C.c_void_p),
+                C.c_uint64(len(pre)), C.byref(npre), C.byref(out_type),
+            )
+            _check(rc, lib)
+            return (blob[: used.value].tobytes(), descs[: nseg.value].copy(),
+                    pre[: npre.value].copy(), out_type.value)
         rc = lib.gemx_downsample_write(
             self._h, start_time, end_time, interval, offset, opc, seg_rows,
             blob.ctypes.data_as(C.c_void_p), bb.value,
