@@ -203,13 +203,15 @@ def main():
                 import torch
                 import torch.distributed as dist
 
+                # one row per (group, window) -> a vectorized scatter,
+                # not np.add.at (add.at was host-bound at
+                # 1M-group shapes)
                 acc = np.zeros((NGROUPS, n_wins, 2))
                 gi = rows["sid"].astype(np.int64)
-                wi = (rows["win_start"] // WINDOW_NS).astype(np.int64)
-                np.add.at(acc, (gi, np.clip(wi, 0, n_wins - 1), 0),
-                          rows["count"])
-                np.add.at(acc, (gi, np.clip(wi, 0, n_wins - 1), 1),
-                          rows["sum"])
+                wi = np.clip((rows["win_start"] // WINDOW_NS).astype(np.int64),
+                             0, n_wins - 1)
+                acc[gi, wi, 0] = rows["count"]
+                acc[gi, wi, 1] = rows["sum"]
                 t = torch.from_numpy(acc.reshape(-1)).to(f"cuda:{local_rank}")
                 dist.all_reduce(t, op=dist.ReduceOp.SUM)
         elif args.query == "preagg":
