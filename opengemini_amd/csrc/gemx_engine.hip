@@ -1286,6 +1286,166 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
   }
 }
 
+/* ---------------- branchless simple8b int grid kernel ----------------
+ *
+ * One lane per const-delta-time full simple8b int64 segment
+ * (lib/encoding/int.go:256-301: zigzag deltas packed 60..240 to a word,
+ * selector in the top nibble; selectors 0/1 decode as runs of ones).
+ * Same two-level window structure as the gorilla kernel; the per-value
+ * decode is a shift+mask+zigzag-add with the selector table packed into
+ * immediate constants (no indexed stack arrays -> no scratch), and the
+ * word refill is a rare (~1 in 15 values) in-blob load: simple8b is
+ * 15-240x lighter on loads than gorilla, so no stream arena is needed.
+ * Semantics identical to k_scan_fast<INT,0,1> over the same segments. */
+/* selector tables packed into immediates, LSB-first per selector
+ * (lib/util/lifted/encoding/simple8b/encoding.go:193):
+ * n    = {240,120,60,30,20,15,12,10, 8,7,6,5,4,3,2,1}
+ * bits = {0,0,1,2,3,4,5,6,7,8, 10,12,15,20,30,60} */
+#define S8B_NS0 0x0A0C0F141E3C78F0ULL /* bytes LSB-first: sel 0..7 */
+#define S8B_NS1 0x0102030405060708ULL /* bytes LSB-first: sel 8..15 */
+#define S8B_BSLO 0x8765432100ULL      /* nibbles LSB-first: sel 0..9 */
+#define S8B_BSHI 0x3C1E140F0C0AULL    /* bytes LSB-first: sel 10..15 */
+
+__device__ __forceinline__ void s8b_sel(uint32_t sel, int *n, int *bits) {
+  uint64_t ns = (sel < 8) ? (S8B_NS0 >> (sel * 8))
+                          : (S8B_NS1 >> ((sel - 8) * 8));
+  *n = (int)(ns & 0xFF);
+  if (sel < 10)
+    *bits = (int)((S8B_BSLO >> (sel * 4)) & 0xF);
+  else
+    *bits = (int)((S8B_BSHI >> ((sel - 10) * 8)) & 0xFF);
+}
+
+__global__ void __launch_bounds__(256) k_scan_grid_s8b(
+    const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
+    const SegQ *__restrict__ segq, const uint32_t *__restrict__ seg_ids,
+    uint32_t nseg_ids, Partial *__restrict__ partials, int64_t interval,
+    int64_t offset, DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (uint32_t li = gid; li < nseg_ids; li += gridDim.x * blockDim.x) {
+    uint32_t si = seg_ids[li];
+    const gemx_seg_desc d = descs[si];
+    const SegQ sq = segq[si];
+    if (sq.n_wins == 0) continue;
+
+    /* const-delta time (routing guarantee; timestamp.go:190) */
+    int64_t t0c, dtc;
+    {
+      const uint8_t *tseg = blob + d.time_offset;
+      if (tseg[0] == 18) {
+        t0c = (int64_t)d_u64le(tseg + 1);
+        dtc = 0;
+      } else {
+        TimeIter ti;
+        if (ti.init(tseg + 5, d.time_size - 5) || ti.kind != 1 ||
+            ti.left < (int64_t)d.rows) {
+          set_err(err, GEMX_E_DECODE);
+          return;
+        }
+        t0c = ti.cur;
+        dtc = ti.delta;
+      }
+    }
+    /* simple8b full data block (routing guarantee):
+     * [tag32][rows u32be][2<<4][encCnt u32be][srcCnt u32be][first zz u64be]
+     * [words u64be x encCnt-1] */
+    const int rows = (int)d.rows;
+    const uint8_t *enc = blob + d.data_offset + 5;
+    if (d.data_size < 5 + 1 + 16 + 8 || (enc[0] >> 4) != 2) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    const uint8_t *in = enc + 1;
+    const int64_t nwords = (int64_t)d_u32be(in) - 1;
+    const int64_t srcCnt = (int64_t)d_u32be(in + 4);
+    if (srcCnt != rows || nwords < 0) {
+      set_err(err, GEMX_E_DECODE);
+      return;
+    }
+    int64_t cur = d_zigzag_dec(d_u64be(in + 8));
+    const uint8_t *words = in + 16;
+    int64_t widx = 0;
+    uint64_t w = 0;
+    int w_n = 0, w_bits = 0, w_i = 0;
+
+#define S8B_NEXT()                                                             \
+    do {                                                                       \
+      if (w_i >= w_n) {                                                        \
+        if (widx >= nwords) {                                                  \
+          set_err(err, GEMX_E_DECODE);                                         \
+          return;                                                              \
+        }                                                                      \
+        w = d_u64be(words + widx * 8);                                         \
+        widx++;                                                                \
+        s8b_sel((uint32_t)(w >> 60), &w_n, &w_bits);                           \
+        w_i = 0;                                                               \
+      }                                                                        \
+      uint64_t zz;                                                             \
+      if (w_bits == 0)                                                         \
+        zz = 1;                                                                \
+      else                                                                     \
+        zz = (w >> (w_i * w_bits)) & ((1ULL << w_bits) - 1);                   \
+      w_i++;                                                                   \
+      cur += d_zigzag_dec(zz);                                                 \
+    } while (0)
+
+    Partial *base = partials + sq.partial_base;
+    for (uint32_t k = 0; k < sq.n_wins; k++) base[k].has_rows = 0;
+
+    int first_pending = 1;
+    int i = 0;
+    while (i < rows) {
+      int64_t t_i = t0c + (int64_t)i * dtc;
+      int64_t ord = win_ordinal(t_i, interval, offset);
+      if (ord < sq.w_first || ord >= sq.w_first + (int64_t)sq.n_wins) {
+        set_err(err, GEMX_E_INVALID);
+        return;
+      }
+      int64_t we = ord * interval + offset + interval;
+      int gend;
+      if (dtc == 0) {
+        gend = rows;
+      } else {
+        int64_t n_in = (we - 1 - t_i) / dtc + 1;
+        gend = (n_in >= (int64_t)(rows - i)) ? rows : i + (int)n_in;
+      }
+      if (!first_pending) S8B_NEXT();
+      first_pending = 0;
+      int64_t sv = cur;
+      int64_t sf = sv, mn = sv, mx = sv, lastv = sv;
+      int min_row = i, max_row = i;
+      for (int k = i + 1; k < gend; k++) {
+        S8B_NEXT();
+        int64_t v = cur;
+        sf += v;
+        /* first-occurrence-wins strict compares */
+        if (mn > v) { mn = v; min_row = k; }
+        if (mx < v) { mx = v; max_row = k; }
+        lastv = v;
+      }
+      Partial tmp;
+      tmp.v[0].i = gend - i;
+      tmp.v[1].i = sf;
+      tmp.v[2].i = mn;
+      tmp.v[3].i = mx;
+      tmp.v[4].i = sv;
+      tmp.v[5].i = lastv;
+      tmp.t[0] = t_i;
+      tmp.t[1] = t_i; /* no nils: valueIndex == row index */
+      tmp.t[2] = t0c + (int64_t)min_row * dtc;
+      tmp.t[3] = t0c + (int64_t)max_row * dtc;
+      tmp.t[4] = t_i;
+      tmp.t[5] = t0c + (int64_t)(gend - 1) * dtc;
+      tmp.first_row_time = t_i;
+      tmp.nilmask = 0;
+      tmp.has_rows = 1;
+      base[ord - sq.w_first] = tmp;
+      i = gend;
+    }
+#undef S8B_NEXT
+  }
+}
+
 /* ---------------- lane-per-segment RAW float kernel ----------------
  *
  * Raw (uncompressed) float blocks — the reference's fallback when gorilla
@@ -3490,8 +3650,9 @@ struct QueryPlan {
   uint32_t *d_fast_q = nullptr, *d_gen_q = nullptr;
   uint32_t n_fast_q = 0, n_gen_q = 0;
   uint32_t *d_fastg_q = nullptr, *d_fasts_q = nullptr, *d_fastgor_q = nullptr,
-           *d_fastraw_q = nullptr;
-  uint32_t n_fastg_q = 0, n_fasts_q = 0, n_fastgor_q = 0, n_fastraw_q = 0;
+           *d_fastraw_q = nullptr, *d_fasts8b_q = nullptr;
+  uint32_t n_fastg_q = 0, n_fasts_q = 0, n_fastgor_q = 0, n_fastraw_q = 0,
+           n_fasts8b_q = 0;
   void *d_gtmp = nullptr; /* GAcc[n_gwins × gsplit] */
   uint32_t gsplit = 1, gper_chunk = 1;
   int64_t W0 = 0;
@@ -3570,9 +3731,9 @@ struct gemx_shard {
   std::vector<uint32_t> fast_grid_ids, fast_stream_ids, fast_gor_ids;
   uint32_t *d_fast_grid_ids = nullptr, *d_fast_stream_ids = nullptr,
            *d_fast_gor_ids = nullptr;
-  std::vector<char> is_grid, is_gor, is_raw; /* per segment */
-  std::vector<uint32_t> fast_raw_ids;
-  uint32_t *d_fast_raw_ids = nullptr;
+  std::vector<char> is_grid, is_gor, is_raw, is_s8b; /* per segment */
+  std::vector<uint32_t> fast_raw_ids, fast_s8b_ids;
+  uint32_t *d_fast_raw_ids = nullptr, *d_fast_s8b_ids = nullptr;
   int64_t max_raw_dt = 0;
   std::vector<GorDesc> h_gor;         /* per segment (zeros for non-gor) */
   GorDesc *d_gor = nullptr;
@@ -3660,6 +3821,7 @@ static void free_plan(QueryPlan &p) {
   if (p.d_fastg_q) (void)hipFree(p.d_fastg_q);
   if (p.d_fastgor_q) (void)hipFree(p.d_fastgor_q);
   if (p.d_fastraw_q) (void)hipFree(p.d_fastraw_q);
+  if (p.d_fasts8b_q) (void)hipFree(p.d_fasts8b_q);
   if (p.d_fasts_q) (void)hipFree(p.d_fasts_q);
   p = QueryPlan();
 }
@@ -3689,7 +3851,8 @@ static int h_uvarint(const uint8_t *p, int64_t len, uint64_t *out) {
  * *grid = const-delta timestamps with non-negative delta — segments the
  * leaner GRIDP=1 instantiation of k_scan_fast can take. */
 static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col_type,
-                            bool *fast, bool *grid, bool *gor, bool *raw) {
+                            bool *fast, bool *grid, bool *gor, bool *raw,
+                            bool *s8b) {
   if (d.data_size < 1 || d.time_size < 1) return GEMX_E_INVALID;
   const uint8_t *ds = blob + d.data_offset;
   const uint8_t *ts = blob + d.time_offset;
@@ -3697,6 +3860,7 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
   *grid = false;
   *gor = false;
   *raw = false;
+  *s8b = false;
   uint8_t dt = ds[0];
   /* time: one-value or Full + {const-delta, simple8b, uncompressed} */
   if (ts[0] == 18) {
@@ -3733,6 +3897,7 @@ static int classify_segment(const uint8_t *blob, const gemx_seg_desc &d, int col
     } else {
       if (tag == 3) return GEMX_E_UNSUPPORTED; /* zstd: not on device yet */
       if (tag != 1 && tag != 2 && tag != 4) return GEMX_E_INVALID;
+      *s8b = (tag == 2); /* full simple8b block: branchless grid kernel */
     }
   } else if (dt == (uint8_t)col_type) {
     *fast = false; /* nil bitmap present */
@@ -4083,8 +4248,8 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       delete s;
       return GEMX_E_INVALID;
     }
-    bool fast, grid, gor, raw;
-    int rc = classify_segment(hb, d, col_type, &fast, &grid, &gor, &raw);
+    bool fast, grid, gor, raw, s8b;
+    int rc = classify_segment(hb, d, col_type, &fast, &grid, &gor, &raw, &s8b);
     if (rc != 0) {
       seterr(rc == GEMX_E_UNSUPPORTED
                  ? "segment uses a codec not yet on-device (MLF/legacy gorilla)"
@@ -4098,6 +4263,8 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
         s->fast_gor_ids.push_back((uint32_t)i);
       else if (grid && raw)
         s->fast_raw_ids.push_back((uint32_t)i);
+      else if (grid && s8b)
+        s->fast_s8b_ids.push_back((uint32_t)i);
       else if (grid)
         s->fast_grid_ids.push_back((uint32_t)i);
       else
@@ -4108,9 +4275,11 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     if (s->is_grid.size() < i + 1) s->is_grid.resize(nsegs, 0);
     if (s->is_gor.size() < i + 1) s->is_gor.resize(nsegs, 0);
     if (s->is_raw.size() < i + 1) s->is_raw.resize(nsegs, 0);
+    if (s->is_s8b.size() < i + 1) s->is_s8b.resize(nsegs, 0);
     s->is_grid[i] = fast && grid;
     s->is_gor[i] = fast && grid && gor;
     s->is_raw[i] = fast && grid && raw && !gor;
+    s->is_s8b[i] = fast && grid && s8b;
     s->total_rows_scanned += d.rows;
     /* series ranges + per-series/shard time bounds (for preagg coverage) */
     if (s->series_ranges.empty() || s->series_ranges.back().sid != d.sid) {
@@ -4193,6 +4362,13 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     HIP_CHECK(hipMemcpyAsync(s->d_fast_raw_ids, s->fast_raw_ids.data(),
                              sizeof(uint32_t) * s->fast_raw_ids.size(),
                              hipMemcpyHostToDevice, s->stream));
+  HIP_CHECK(hipMalloc(&s->d_fast_s8b_ids,
+                      sizeof(uint32_t) *
+                          (s->fast_s8b_ids.empty() ? 1 : s->fast_s8b_ids.size())));
+  if (!s->fast_s8b_ids.empty())
+    HIP_CHECK(hipMemcpyAsync(s->d_fast_s8b_ids, s->fast_s8b_ids.data(),
+                             sizeof(uint32_t) * s->fast_s8b_ids.size(),
+                             hipMemcpyHostToDevice, s->stream));
   HIP_CHECK(hipMalloc(&s->d_general_ids,
                       sizeof(uint32_t) *
                           (s->general_ids.empty() ? 1 : s->general_ids.size())));
@@ -4224,6 +4400,7 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   (void)hipFree(s->d_fast_stream_ids);
   (void)hipFree(s->d_fast_gor_ids);
   (void)hipFree(s->d_fast_raw_ids);
+  (void)hipFree(s->d_fast_s8b_ids);
   if (s->d_gor) (void)hipFree(s->d_gor);
   if (s->d_arena) (void)hipFree(s->d_arena);
   (void)hipFree(s->d_general_ids);
@@ -4312,7 +4489,7 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     P.sq.resize(s->series_ranges.size());
     P.partial_slots = 0;
     P.total_rows = 0;
-    std::vector<uint32_t> fast_q, gen_q, fast_gq, fast_sq, fast_gorq, fast_rawq;
+    std::vector<uint32_t> fast_q, gen_q, fast_gq, fast_sq, fast_gorq, fast_rawq, fast_s8bq;
     std::vector<char> is_gen(nsegs, 0);
     for (auto id : s->general_ids) is_gen[id] = 1;
     bool any_clip = false;
@@ -4371,6 +4548,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
           else if (s->is_raw[i] && interval > 0 &&
                    63 * s->h_gor[i].dt < 8 * interval)
             fast_rawq.push_back(i); /* wave kernel's compare-chain bound */
+          else if (s->is_s8b[i])
+            fast_s8bq.push_back(i);
           else if (s->is_grid[i])
             fast_gq.push_back(i);
           else
@@ -4439,6 +4618,13 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       if (!fast_rawq.empty())
         HIP_CHECK(hipMemcpyAsync(P.d_fastraw_q, fast_rawq.data(),
                                  sizeof(uint32_t) * fast_rawq.size(),
+                                 hipMemcpyHostToDevice, s->stream));
+      P.n_fasts8b_q = (uint32_t)fast_s8bq.size();
+      HIP_CHECK(hipMalloc(&P.d_fasts8b_q,
+                          sizeof(uint32_t) * (fast_s8bq.empty() ? 1 : fast_s8bq.size())));
+      if (!fast_s8bq.empty())
+        HIP_CHECK(hipMemcpyAsync(P.d_fasts8b_q, fast_s8bq.data(),
+                                 sizeof(uint32_t) * fast_s8bq.size(),
                                  hipMemcpyHostToDevice, s->stream));
     }
     HIP_CHECK(hipMalloc(&P.d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
@@ -4527,6 +4713,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
   const uint32_t *raw_list = s->d_fast_raw_ids;
   uint32_t raw_n = (uint32_t)s->fast_raw_ids.size();
   bool raw_ok = (interval > 0) && (63 * s->max_raw_dt < 8 * interval);
+  const uint32_t *s8b_list = s->d_fast_s8b_ids;
+  uint32_t s8b_n = (uint32_t)s->fast_s8b_ids.size();
   const uint32_t *gen_list = s->d_general_ids;
   uint32_t gen_n = (uint32_t)s->general_ids.size();
   if (P.clipped) {
@@ -4541,6 +4729,8 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     raw_list = P.d_fastraw_q;
     raw_n = P.n_fastraw_q;
     raw_ok = true; /* per-segment dt bound applied when queueing */
+    s8b_list = P.d_fasts8b_q;
+    s8b_n = P.n_fasts8b_q;
     gen_list = P.d_gen_q;
     gen_n = P.n_gen_q;
   }
@@ -4557,11 +4747,12 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     const uint32_t *list;
     uint32_t n;
     int gridp; /* 0 stream, 1 grid, 2 grid+gorilla */
-  } launches[4];
+  } launches[5];
   int n_launches = 0;
   if (use_grid) {
     if (gor_n) launches[n_launches++] = {gor_list, gor_n, no_gork ? 1 : 2};
     if (raw_n) launches[n_launches++] = {raw_list, raw_n, raw_ok ? 3 : 1};
+    if (s8b_n) launches[n_launches++] = {s8b_list, s8b_n, no_gork ? 1 : 4};
     if (grid_n) launches[n_launches++] = {grid_list, grid_n, 1};
     if (strm_n) launches[n_launches++] = {strm_list, strm_n, 0};
   } else if (fast_n) {
@@ -4586,6 +4777,12 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       hipLaunchKernelGGL(k_scan_raw_lane, dim3(blocks), dim3(tpb), 0,
                          s->stream, s->d_arena, s->d_gor, s->d_descs, d_segq,
                          lst, n, d_part, interval, offset, d_err);
+      continue;
+    }
+    if (launches[li].gridp == 4) { /* lane per simple8b int segment */
+      hipLaunchKernelGGL(k_scan_grid_s8b, dim3(blocks), dim3(tpb), 0,
+                         s->stream, s->d_blob, s->d_descs, d_segq, lst, n,
+                         d_part, interval, offset, d_err);
       continue;
     }
 #define GEMX_LAUNCH_FAST(CT, FLT, GP)                                          \
