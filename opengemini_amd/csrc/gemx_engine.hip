@@ -727,6 +727,24 @@ struct GorDesc {
   int64_t t0, dt;      /* const-delta time (timestamp.go:190) */
 };
 
+/* gorilla sub-segment resume state (config #1 underfill fix): at attach,
+ * when a shard yields too few gorilla lanes to fill the chip, the host
+ * walks each stream once and records decoder states every R rows; the
+ * scan then runs one lane per SUB-segment into temporary partial slots
+ * and k_submerge folds them back into the original per-(segment,window)
+ * partials — bit-identical except the documented 1e-9 float-sum
+ * reassociation. */
+struct GorSub {
+  uint32_t seg_id; /* original descriptor index */
+  uint32_t row0;   /* first row this sub decodes */
+  uint32_t rows;
+  uint32_t word_idx; /* stream word the window starts at */
+  uint64_t g_val;    /* value of row0 (decoded by the host walk) */
+  uint8_t bp;        /* bit offset inside word_idx */
+  uint8_t mean, trail;
+  uint8_t _pad[5];
+};
+
 /* per-series output metadata */
 struct SeriesQ {
   uint64_t sid;
@@ -1283,6 +1301,191 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
       return;
     }
 #undef GOR_NEXT
+  }
+}
+
+/* ---------------- gorilla sub-segment kernels (config #1 underfill) ----
+ *
+ * One lane per SUB-segment (GorSub resume states recorded by the attach
+ * walk), writing per-(sub,window) partials into a temporary array;
+ * k_submerge then folds each original segment\'s subs back into its main
+ * per-(segment,window) Partial slots in row order, reproducing the
+ * unsplit kernel\'s outputs bit-exactly except the documented 1e-9
+ * float-sum reassociation. The rest of the pipeline (k_merge / k_group)
+ * is unchanged. */
+__global__ void __launch_bounds__(256) k_scan_gor_sub(
+    const uint64_t *__restrict__ arena, const GorDesc *__restrict__ gors,
+    const GorSub *__restrict__ subs, const SegQ *__restrict__ subq,
+    uint32_t nsubs, Partial *__restrict__ partials, int64_t interval,
+    int64_t offset, DevErr *err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (uint32_t li = gid; li < nsubs; li += gridDim.x * blockDim.x) {
+    const GorSub sb = subs[li];
+    const SegQ sq = subq[li];
+    if (sq.n_wins == 0) continue;
+    const GorDesc g = gors[sb.seg_id];
+    const int64_t dtc = g.dt;
+    const int64_t t0c = g.t0 + (int64_t)sb.row0 * dtc;
+    const int rows = (int)sb.rows;
+    uint64_t g_val = sb.g_val;
+    GorA br;
+    br.init(arena, g.arena_base + (uint64_t)sb.word_idx * 64);
+    br.bp = sb.bp;
+    uint32_t g_mean = sb.mean, g_trail = sb.trail;
+    uint64_t bad = 0;
+    (void)bad; /* host walk already validated the stream */
+
+    Partial *base = partials + sq.partial_base;
+    for (uint32_t k = 0; k < sq.n_wins; k++) base[k].has_rows = 0;
+
+#define GOR_NEXT()                                                             \
+    do {                                                                       \
+      uint64_t A = GorA::fun(br.w0, br.w1, br.bp);                             \
+      uint32_t p13 = (uint32_t)(A >> 51);                                      \
+      uint32_t ctrl1 = p13 >> 12;                                              \
+      uint32_t neww = ctrl1 & ((p13 >> 11) & 1);                               \
+      uint32_t mr = p13 & 0x3F;                                                \
+      g_mean = neww ? (mr ? mr : 64u) : g_mean;                                \
+      g_trail = neww ? (mr ? (64u - ((p13 >> 6) & 0x1F) - mr) : 0u) : g_trail; \
+      uint32_t hdr = 1 + ctrl1 + (neww ? 11u : 0u);                            \
+      uint64_t Cc = GorA::fun(br.w1, br.w2, br.bp);                            \
+      uint64_t B = (A << hdr) | (Cc >> (64 - hdr));                            \
+      uint64_t sbv = (g_mean == 64) ? B : (B >> (64 - g_mean));                \
+      g_val ^= ctrl1 ? (sbv << (g_trail & 63)) : 0;                            \
+      uint32_t np = (uint32_t)br.bp + hdr + (ctrl1 ? g_mean : 0);              \
+      uint32_t adv = np >> 6;                                                  \
+      br.bp = (int)(np & 63);                                                  \
+      if (__builtin_expect(adv >= 2, 0)) {                                     \
+        br.w0 = br.w2;                                                         \
+        br.w1 = br.w3;                                                         \
+        br.w2 = br.L;                                                          \
+        br.w3 = br.M;                                                          \
+        br.p += 2 * 64;                                                        \
+        br.L = br.ldw();                                                       \
+        br.M = br.ldw();                                                       \
+        br.p -= 2 * 64;                                                        \
+      } else {                                                                 \
+        const int c1 = (int)adv;                                               \
+        br.w0 = c1 ? br.w1 : br.w0;                                            \
+        br.w1 = c1 ? br.w2 : br.w1;                                            \
+        br.w2 = c1 ? br.w3 : br.w2;                                            \
+        br.w3 = c1 ? br.L : br.w3;                                             \
+        br.L = c1 ? br.M : br.L;                                               \
+        br.p += c1 ? 64 : 0;                                                   \
+        br.M = br.p[64];                                                       \
+      }                                                                        \
+    } while (0)
+
+    int first_pending = 1;
+    int i = 0;
+    while (i < rows) {
+      int64_t t_i = t0c + (int64_t)i * dtc;
+      int64_t ord = win_ordinal(t_i, interval, offset);
+      if (ord < sq.w_first || ord >= sq.w_first + (int64_t)sq.n_wins) {
+        set_err(err, GEMX_E_INVALID);
+        return;
+      }
+      int64_t we = ord * interval + offset + interval;
+      int gend;
+      if (dtc == 0) {
+        gend = rows;
+      } else {
+        int64_t n_in = (we - 1 - t_i) / dtc + 1;
+        gend = (n_in >= (int64_t)(rows - i)) ? rows : i + (int)n_in;
+      }
+      if (!first_pending) GOR_NEXT();
+      first_pending = 0;
+      double fv;
+      memcpy(&fv, &g_val, 8);
+      double sf = fv, mn = fv, mx = fv, lastv = fv;
+      const double firstv = fv;
+      int min_row = i, max_row = i;
+      for (int k = i + 1; k < gend; k++) {
+        GOR_NEXT();
+        double v;
+        memcpy(&v, &g_val, 8);
+        sf += v;
+        if (mn > v) { mn = v; min_row = k; }
+        if (mx < v) { mx = v; max_row = k; }
+        lastv = v;
+      }
+      Partial tmp;
+      tmp.v[0].i = gend - i;
+      tmp.v[1].f = sf;
+      tmp.v[2].f = mn;
+      tmp.v[3].f = mx;
+      tmp.v[4].f = firstv;
+      tmp.v[5].f = lastv;
+      tmp.t[0] = t_i;
+      tmp.t[1] = t_i;
+      tmp.t[2] = t0c + (int64_t)min_row * dtc;
+      tmp.t[3] = t0c + (int64_t)max_row * dtc;
+      tmp.t[4] = t_i;
+      tmp.t[5] = t0c + (int64_t)(gend - 1) * dtc;
+      tmp.first_row_time = t_i;
+      tmp.nilmask = 0;
+      tmp.has_rows = 1;
+      base[ord - sq.w_first] = tmp;
+      i = gend;
+    }
+#undef GOR_NEXT
+  }
+}
+
+/* fold sub partials into the original per-(segment,window) slots.
+ * One thread per (split segment, window): subs contribute in row order,
+ * so fv() merge semantics (count/sum add; min/max strict earlier-wins;
+ * first keeps, last assigns) reproduce the unsplit partial exactly
+ * except float-sum rounding. first_row_time stays the EARLIEST
+ * contributing sub\'s (the unsplit kernel\'s window start). */
+__global__ void __launch_bounds__(256) k_submerge(
+    const uint32_t *__restrict__ seg_ids, uint32_t nsegs_split, uint32_t wmax,
+    const SegQ *__restrict__ segq, Partial *__restrict__ partials,
+    const uint32_t *__restrict__ sub_start,
+    const uint32_t *__restrict__ sub_count, const SegQ *__restrict__ subq,
+    const Partial *__restrict__ subparts) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t total = nsegs_split * wmax;
+  for (uint32_t li = gid; li < total; li += gridDim.x * blockDim.x) {
+    uint32_t seg = seg_ids[li / wmax];
+    uint32_t k = li % wmax;
+    const SegQ sq = segq[seg];
+    if (k >= sq.n_wins) continue;
+    const int64_t W = sq.w_first + (int64_t)k;
+    Partial out;
+    int have = 0;
+    const uint32_t s0 = sub_start[seg], sc = sub_count[seg];
+    for (uint32_t j = s0; j < s0 + sc; j++) {
+      const SegQ q = subq[j];
+      if (W < q.w_first || W >= q.w_first + (int64_t)q.n_wins) continue;
+      const Partial p = subparts[q.partial_base + (W - q.w_first)];
+      if (!p.has_rows) continue;
+      if (!have) {
+        out = p;
+        have = 1;
+        continue;
+      }
+      out.v[0].i += p.v[0].i;
+      out.v[1].f += p.v[1].f;
+      /* strict compares: the earlier sub wins ties (gorilla streams are
+       * NaN-free by construction — the encoder rejects NaN values) */
+      if (p.v[2].f < out.v[2].f) {
+        out.v[2] = p.v[2];
+        out.t[2] = p.t[2];
+      }
+      if (p.v[3].f > out.v[3].f) {
+        out.v[3] = p.v[3];
+        out.t[3] = p.t[3];
+      }
+      out.v[5] = p.v[5]; /* last assigns */
+      out.t[5] = p.t[5];
+    }
+    if (have) {
+      Partial *dst = partials + sq.partial_base + k;
+      *dst = out;
+    } else {
+      partials[sq.partial_base + k].has_rows = 0;
+    }
   }
 }
 
@@ -2503,6 +2706,74 @@ __global__ void __launch_bounds__(256) k_group_p2(
 }
 
 
+/* Degenerate grouped merge for FEW series (config #1: one deep series,
+ * many windows): the p1/p2 pair pays a 256-thread LDS tree per window
+ * with at most `nseries` live inputs — one THREAD per window folding the
+ * series serially is ~100x leaner and massively parallel over windows.
+ * Same semantics (gacc_row order = series order). */
+template <int COLTYPE>
+__global__ void __launch_bounds__(256) k_group_small(
+    const SeriesQ *__restrict__ series, uint32_t nseries,
+    const SegQ *__restrict__ segq, const Partial *__restrict__ partials,
+    gemx_agg_row *__restrict__ out, int64_t W0, uint32_t n_gwins,
+    int64_t interval, int64_t offset, int64_t q_start, int keep_empty,
+    DevErr *__restrict__ err) {
+  uint32_t gid = blockIdx.x * blockDim.x + threadIdx.x;
+  for (uint32_t wb = gid; wb < n_gwins; wb += gridDim.x * blockDim.x) {
+    int64_t w = W0 + (int64_t)wb;
+    GAcc a;
+    memset(&a, 0, sizeof(a));
+    for (uint32_t g = 0; g < nseries; g++) {
+      const SeriesQ sr = series[g];
+      int64_t local = w - sr.w_min;
+      if (local < 0 || local >= (int64_t)sr.n_wins) continue;
+      gemx_agg_row r;
+      memset(&r, 0, sizeof(r));
+      if (!merge_series_window<COLTYPE>(sr, segq, partials, w, &r)) continue;
+      gacc_row<COLTYPE>(&a, &r, g);
+    }
+    gemx_agg_row o;
+    memset(&o, 0, sizeof(o));
+    o.sid = 0;
+    int64_t ws = interval ? win_start_of(w, interval, offset) : q_start;
+    o.win_start = ws;
+    if (!a.used) {
+      if (keep_empty) {
+        o.count = 0;
+        o.first_row_time = ws;
+        o.count_time = ws;
+        o.sum_time = ws;
+        o.min_isnil = o.max_isnil = o.first_isnil = o.last_isnil =
+            o.sum_isnil = 1;
+      } else {
+        o.count = -1; /* gap */
+        __hip_atomic_fetch_add(&err->gaps, 1ull, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_SYSTEM);
+      }
+    } else {
+      o.first_row_time = ws;
+      o.count = a.count;
+      o.count_time = ws;
+      o.sum = a.sum;
+      o.sum_time = ws;
+      o.sum_isnil = !(a.active & 2);
+      o.minv = a.minv;
+      o.min_time = a.min_t;
+      o.min_isnil = !(a.active & 4);
+      o.maxv = a.maxv;
+      o.max_time = a.max_t;
+      o.max_isnil = !(a.active & 8);
+      o.firstv = a.firstv;
+      o.first_time = a.first_t;
+      o.first_isnil = !(a.active & 16);
+      o.lastv = a.lastv;
+      o.last_time = a.last_t;
+      o.last_isnil = !(a.active & 32);
+    }
+    out[wb] = o;
+  }
+}
+
 /* ---------------- hash GROUP BY tag (executor/hash_agg_transform.go) ----
  * The executor's hash-agg step maps each series to a group via the tag-set
  * hash dictionary; here the caller passes that sid→group mapping (one
@@ -3651,6 +3922,10 @@ struct QueryPlan {
   uint32_t n_fast_q = 0, n_gen_q = 0;
   uint32_t *d_fastg_q = nullptr, *d_fasts_q = nullptr, *d_fastgor_q = nullptr,
            *d_fastraw_q = nullptr, *d_fasts8b_q = nullptr;
+  SegQ *d_subq = nullptr;       /* per gorilla sub-segment (underfill split) */
+  Partial *d_subpart = nullptr; /* temp per-(sub,window) partials */
+  uint64_t sub_slots = 0;
+  uint32_t wmax_split = 0;
   uint32_t n_fastg_q = 0, n_fasts_q = 0, n_fastgor_q = 0, n_fastraw_q = 0,
            n_fasts8b_q = 0;
   void *d_gtmp = nullptr; /* GAcc[n_gwins × gsplit] */
@@ -3735,6 +4010,11 @@ struct gemx_shard {
   std::vector<uint32_t> fast_raw_ids, fast_s8b_ids;
   uint32_t *d_fast_raw_ids = nullptr, *d_fast_s8b_ids = nullptr;
   int64_t max_raw_dt = 0;
+  std::vector<GorSub> h_subs; /* empty unless the gor grid is underfilled */
+  std::vector<uint32_t> sub_of_seg_start; /* per segment: first sub index */
+  std::vector<uint32_t> sub_of_seg_count;
+  GorSub *d_subs = nullptr;
+  uint32_t *d_sub_start = nullptr, *d_sub_count = nullptr;
   std::vector<GorDesc> h_gor;         /* per segment (zeros for non-gor) */
   GorDesc *d_gor = nullptr;
   uint64_t *d_arena = nullptr; /* lane-interleaved gorilla stream arena */
@@ -3822,6 +4102,8 @@ static void free_plan(QueryPlan &p) {
   if (p.d_fastgor_q) (void)hipFree(p.d_fastgor_q);
   if (p.d_fastraw_q) (void)hipFree(p.d_fastraw_q);
   if (p.d_fasts8b_q) (void)hipFree(p.d_fasts8b_q);
+  if (p.d_subq) (void)hipFree(p.d_subq);
+  if (p.d_subpart) (void)hipFree(p.d_subpart);
   if (p.d_fasts_q) (void)hipFree(p.d_fasts_q);
   p = QueryPlan();
 }
@@ -3947,6 +4229,88 @@ static int host_grid_time(const uint8_t *blob, const gemx_seg_desc &d,
   *dt = (int64_t)dv;
   return 0;
 }
+
+/* Host gorilla stream walker (batch_float.go:278-514 format): decodes
+ * values[0..rows), recording the decoder state at every sub boundary.
+ * Returns 0, or -1 on a malformed stream. */
+static int host_gor_walk(const uint8_t *stream, uint64_t stream_bytes,
+                         uint64_t first_val, int rows, int sub_rows,
+                         uint32_t seg_id,
+                         std::vector<GorSub> &out) {
+  uint64_t g_val = first_val;
+  uint32_t mean = 64, trail = 0;
+  uint64_t bitpos = 0;
+  const uint64_t nbits = stream_bytes * 8;
+  auto take = [&](int k, uint64_t *v) -> int {
+    uint64_t x = 0;
+    for (int i = 0; i < k; i++) {
+      if (bitpos >= nbits) return -1;
+      x = (x << 1) | ((stream[bitpos >> 3] >> (7 - (bitpos & 7))) & 1);
+      bitpos++;
+    }
+    *v = x;
+    return 0;
+  };
+  for (int r = 0; r < rows; r++) {
+    if (r > 0) { /* rows[1..): decode one record */
+      uint64_t b0;
+      if (take(1, &b0)) return -1;
+      if (b0) {
+        uint64_t b1;
+        if (take(1, &b1)) return -1;
+        if (b1) {
+          uint64_t lm;
+          if (take(11, &lm)) return -1;
+          uint32_t lead = (uint32_t)((lm >> 6) & 0x1F);
+          uint32_t mr = (uint32_t)(lm & 0x3F);
+          mean = mr ? mr : 64;
+          trail = mr ? (64 - lead - mr) : 0;
+        }
+        uint64_t sb;
+        if (take((int)mean, &sb)) return -1;
+        g_val ^= sb << (trail & 63);
+        if (g_val == UVNAN) return -1; /* mid-stream terminator */
+      }
+    }
+    if (r % sub_rows == 0) {
+      GorSub sub;
+      sub.seg_id = seg_id;
+      sub.row0 = (uint32_t)r;
+      sub.rows = (uint32_t)std::min(sub_rows, rows - r);
+      sub.word_idx = (uint32_t)(bitpos >> 6);
+      sub.g_val = g_val;
+      sub.bp = (uint8_t)(bitpos & 63);
+      sub.mean = (uint8_t)mean;
+      sub.trail = (uint8_t)trail;
+      memset(sub._pad, 0, sizeof(sub._pad));
+      out.push_back(sub);
+    }
+  }
+  /* terminator must follow */
+  uint64_t b0;
+  if (take(1, &b0)) return -1;
+  if (b0) {
+    uint64_t b1;
+    if (take(1, &b1)) return -1;
+    if (b1) {
+      uint64_t lm;
+      if (take(11, &lm)) return -1;
+      uint32_t lead = (uint32_t)((lm >> 6) & 0x1F);
+      uint32_t mr = (uint32_t)(lm & 0x3F);
+      mean = mr ? mr : 64;
+      trail = mr ? (64 - lead - mr) : 0;
+    }
+    uint64_t sb;
+    if (take((int)mean, &sb)) return -1;
+    g_val ^= sb << (trail & 63);
+  }
+  return g_val == UVNAN ? 0 : -1;
+}
+
+/* lanes below which the gorilla grid is considered underfilled and worth
+ * the attach-time split walk (256 CU x 4 SIMD x 16 wave slots x 64) */
+#define GEMX_SPLIT_MIN_LANES 65536u
+#define GEMX_SUB_MIN_ROWS 64
 
 static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
   size_t n = s->fast_gor_ids.size();
@@ -4084,6 +4448,67 @@ static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
                       hipMemcpyHostToDevice));
   HIP_CHECK(hipMemcpy(s->d_gor, s->h_gor.data(), sizeof(GorDesc) * s->nsegs,
                       hipMemcpyHostToDevice));
+  /* underfilled gorilla grid (config #1 shape: few deep segments): walk
+   * the streams once on the host and record per-sub resume states so the
+   * scan can run one lane per SUB-segment */
+  if (n > 0 && n < GEMX_SPLIT_MIN_LANES) {
+    uint32_t K = (uint32_t)((GEMX_SPLIT_MIN_LANES + n - 1) / n);
+    if (K > 16) K = 16;
+    s->sub_of_seg_start.assign(s->nsegs, 0);
+    s->sub_of_seg_count.assign(s->nsegs, 0);
+    bool any = false;
+    for (size_t j = 0; j < nall; j++) {
+      const Item &it = items[j];
+      if (it.raw) continue;
+      const gemx_seg_desc &d = s->h_descs[it.id];
+      int sub_rows = ((int)d.rows + (int)K - 1) / (int)K;
+      if (sub_rows < GEMX_SUB_MIN_ROWS) sub_rows = GEMX_SUB_MIN_ROWS;
+      if (sub_rows >= (int)d.rows) { /* not worth splitting */
+        s->sub_of_seg_start[it.id] = (uint32_t)s->h_subs.size();
+        GorSub whole;
+        whole.seg_id = it.id;
+        whole.row0 = 0;
+        whole.rows = d.rows;
+        whole.word_idx = 0;
+        whole.g_val = s->h_gor[it.id].first_val;
+        whole.bp = 0;
+        whole.mean = 64;
+        whole.trail = 0;
+        memset(whole._pad, 0, sizeof(whole._pad));
+        s->h_subs.push_back(whole);
+        s->sub_of_seg_count[it.id] = 1;
+        any = true;
+        continue;
+      }
+      size_t before = s->h_subs.size();
+      s->sub_of_seg_start[it.id] = (uint32_t)before;
+      if (host_gor_walk(it.stream, it.bytes, s->h_gor[it.id].first_val,
+                        (int)d.rows, sub_rows, it.id, s->h_subs) != 0) {
+        seterr("gorilla stream failed host validation walk");
+        return GEMX_E_DECODE;
+      }
+      s->sub_of_seg_count[it.id] = (uint32_t)(s->h_subs.size() - before);
+      any = true;
+    }
+    if (any && !s->h_subs.empty()) {
+      HIP_CHECK(hipMalloc(&s->d_subs, sizeof(GorSub) * s->h_subs.size()));
+      HIP_CHECK(hipMemcpy(s->d_subs, s->h_subs.data(),
+                          sizeof(GorSub) * s->h_subs.size(),
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMalloc(&s->d_sub_start, sizeof(uint32_t) * s->nsegs));
+      HIP_CHECK(hipMemcpy(s->d_sub_start, s->sub_of_seg_start.data(),
+                          sizeof(uint32_t) * s->nsegs,
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMalloc(&s->d_sub_count, sizeof(uint32_t) * s->nsegs));
+      HIP_CHECK(hipMemcpy(s->d_sub_count, s->sub_of_seg_count.data(),
+                          sizeof(uint32_t) * s->nsegs,
+                          hipMemcpyHostToDevice));
+    } else {
+      s->h_subs.clear();
+      s->sub_of_seg_start.clear();
+      s->sub_of_seg_count.clear();
+    }
+  }
   return GEMX_OK;
 }
 
@@ -4401,6 +4826,9 @@ extern "C" int gemx_shard_close(gemx_shard *s) {
   (void)hipFree(s->d_fast_gor_ids);
   (void)hipFree(s->d_fast_raw_ids);
   (void)hipFree(s->d_fast_s8b_ids);
+  if (s->d_subs) (void)hipFree(s->d_subs);
+  if (s->d_sub_start) (void)hipFree(s->d_sub_start);
+  if (s->d_sub_count) (void)hipFree(s->d_sub_count);
   if (s->d_gor) (void)hipFree(s->d_gor);
   if (s->d_arena) (void)hipFree(s->d_arena);
   (void)hipFree(s->d_general_ids);
@@ -4630,6 +5058,44 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     HIP_CHECK(hipMalloc(&P.d_segq, sizeof(SegQ) * (nsegs ? nsegs : 1)));
     HIP_CHECK(hipMemcpyAsync(P.d_segq, P.segq.data(), sizeof(SegQ) * nsegs,
                              hipMemcpyHostToDevice, s->stream));
+    /* gorilla sub-segment plan (underfill split): temp window spans per
+     * sub; only for unclipped interval queries (the gor launch's domain) */
+    if (!s->h_subs.empty() && interval != 0 && !any_clip) {
+      std::vector<SegQ> subq(s->h_subs.size());
+      uint64_t slots = 0;
+      uint32_t wmax = 0;
+      for (uint32_t i : s->fast_gor_ids)
+        wmax = std::max(wmax, P.segq[i].n_wins);
+      for (size_t j = 0; j < s->h_subs.size(); j++) {
+        const GorSub &sb = s->h_subs[j];
+        const SegQ &mq = P.segq[sb.seg_id];
+        subq[j].series_idx = mq.series_idx;
+        if (mq.n_wins == 0) {
+          subq[j].w_first = 0;
+          subq[j].n_wins = 0;
+          subq[j].partial_base = slots;
+          continue;
+        }
+        const GorDesc &g = s->h_gor[sb.seg_id];
+        int64_t t0s = g.t0 + (int64_t)sb.row0 * g.dt;
+        int64_t tes = t0s + (int64_t)(sb.rows - 1) * g.dt;
+        int64_t w0 = win_ordinal(t0s, interval, offset);
+        int64_t w1 = win_ordinal(tes, interval, offset);
+        subq[j].w_first = w0;
+        subq[j].n_wins = (uint32_t)(w1 - w0 + 1);
+        subq[j].partial_base = slots;
+        slots += subq[j].n_wins;
+      }
+      P.sub_slots = slots;
+      P.wmax_split = wmax;
+      HIP_CHECK(hipMalloc(&P.d_subq, sizeof(SegQ) * subq.size()));
+      HIP_CHECK(hipMemcpyAsync(P.d_subq, subq.data(),
+                               sizeof(SegQ) * subq.size(),
+                               hipMemcpyHostToDevice, s->stream));
+      HIP_CHECK(hipMalloc(&P.d_subpart,
+                          sizeof(Partial) * (slots ? slots : 1)));
+      HIP_CHECK(hipStreamSynchronize(s->stream)); /* subq is a local */
+    }
     HIP_CHECK(hipMalloc(&P.d_sq,
                         sizeof(SeriesQ) * (P.sq.empty() ? 1 : P.sq.size())));
     HIP_CHECK(hipMemcpyAsync(P.d_sq, P.sq.data(), sizeof(SeriesQ) * P.sq.size(),
@@ -4767,6 +5233,23 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     uint32_t blocks = std::min<uint32_t>((n + tpb - 1) / tpb, 65535);
     const uint32_t *lst = launches[li].list;
     if (launches[li].gridp == 2) {
+      if (P.d_subq && !P.clipped) { /* underfill split: lane per SUB */
+        uint32_t nsubs = (uint32_t)s->h_subs.size();
+        uint32_t stpb = (nsubs < 64 * 1024) ? 64 : 256;
+        uint32_t sblocks = std::min<uint32_t>((nsubs + stpb - 1) / stpb, 65535);
+        hipLaunchKernelGGL(k_scan_gor_sub, dim3(sblocks), dim3(stpb), 0,
+                           s->stream, s->d_arena, s->d_gor, s->d_subs,
+                           P.d_subq, nsubs, P.d_subpart, interval, offset,
+                           d_err);
+        uint32_t total = n * P.wmax_split;
+        uint32_t mblocks = std::min<uint32_t>((total + 255) / 256, 65535);
+        if (total)
+          hipLaunchKernelGGL(k_submerge, dim3(mblocks), dim3(256), 0,
+                             s->stream, lst, n, P.wmax_split, d_segq, d_part,
+                             s->d_sub_start, s->d_sub_count, P.d_subq,
+                             P.d_subpart);
+        continue;
+      }
       hipLaunchKernelGGL(k_scan_grid_gor, dim3(blocks), dim3(tpb), 0,
                          s->stream, s->d_arena, s->arena_words, s->d_gor,
                          s->d_descs, d_segq, lst, n, d_part, interval,
@@ -4936,7 +5419,22 @@ static int scan_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
                          s->stream, d_sq, (uint32_t)sq.size(), d_segq, d_part, d_rows,
                          total_rows, interval, offset, start_time, d_err);
   }
-  if (!tagq && group_all && P.n_gwins > 0) {
+  if (!tagq && group_all && P.n_gwins > 0 && sq.size() <= 32) {
+    /* few series, many windows (config #1 shape): one thread per window */
+    uint32_t bs = (uint32_t)std::min<uint64_t>((P.n_gwins + TPB - 1) / TPB,
+                                               65535);
+    if (s->col_type == GEMX_TYPE_FLOAT)
+      hipLaunchKernelGGL((k_group_small<GEMX_TYPE_FLOAT>), dim3(bs),
+                         dim3(TPB), 0, s->stream, d_sq, (uint32_t)sq.size(),
+                         d_segq, d_part, P.d_grows2[slot], P.W0,
+                         (uint32_t)P.n_gwins, interval, offset, start_time,
+                         keep_empty, d_err);
+    else
+      hipLaunchKernelGGL((k_group_small<GEMX_TYPE_INT>), dim3(bs), dim3(TPB),
+                         0, s->stream, d_sq, (uint32_t)sq.size(), d_segq,
+                         d_part, P.d_grows2[slot], P.W0, (uint32_t)P.n_gwins,
+                         interval, offset, start_time, keep_empty, d_err);
+  } else if (!tagq && group_all && P.n_gwins > 0) {
     uint32_t b1 = (uint32_t)std::min<uint64_t>(P.n_gwins * P.gsplit, 65535);
     uint32_t b2 = (uint32_t)std::min<uint64_t>(P.n_gwins, 65535);
     if (s->col_type == GEMX_TYPE_FLOAT) {
