@@ -209,6 +209,62 @@ struct BitR {
   }
 };
 
+/* per-segment gorilla arena descriptor (attach-time precompute for
+ * k_scan_grid_gor): first decoded value, the segment's word-0 index in
+ * the lane-interleaved stream arena, and the const-delta time params —
+ * the kernel never touches the on-disk segment bytes. */
+struct GorDesc {
+  uint64_t first_val;  /* bits of value 0 (batch_float.go:293) */
+  uint64_t arena_base; /* u64 index; word k of this stream at base + 64*k */
+  int64_t t0, dt;      /* const-delta time (timestamp.go:190) */
+};
+
+struct GorA {
+  const uint64_t *p;      /* arena cursor: word k of this stream lives at
+                             base + 64*k (u64 units, 512-byte stride). No
+                             bounds checks: the arena pad
+                             (GEMX_ARENA_PAD_WORDS) covers the maximum
+                             possible overshoot of a corrupt stream, whose
+                             garbage decode is then caught by the
+                             terminator check. */
+  uint64_t w0, w1, w2, w3; /* 256-bit window, bit cursor bp inside w0:
+                              any <=77-bit record at bp<=63 needs <=140
+                              bits = always inside w0..w2; w3+L+M are the
+                              refill pipeline */
+  uint64_t L, M;           /* 2-deep load pipeline (an 8-register batch
+                              refill and an LDS-DMA cache warmer both
+                              measured slower — r2 notes in DESIGN.md) */
+  int bp;                  /* 0..63 */
+
+  __device__ __forceinline__ uint64_t ldw() {
+    uint64_t w = *p;
+    p += 64;
+    return w;
+  }
+  __device__ __forceinline__ void init(const uint64_t *arena, uint64_t base) {
+    p = arena + base;
+    w0 = ldw();
+    w1 = ldw();
+    w2 = ldw();
+    w3 = ldw();
+    L = ldw();
+    M = ldw();
+    p -= 2 * 64; /* p tracks the address OF L until L is consumed */
+    bp = 0;
+  }
+  /* 64 bits starting at bit bp of (a,b); bp in [0,63] — the (>>1) split
+   * keeps the shift amount in range without a select */
+  static __device__ __forceinline__ uint64_t fun(uint64_t a, uint64_t b,
+                                                 int bp) {
+    return (a << bp) | ((b >> (63 - bp)) >> 1);
+  }
+};
+
+/* arena tail pad, u64 units: bounds the worst-case cursor overshoot of a
+ * corrupt stream — (max rows+1) records x <=2 word-advances x 64 u64
+ * stride, plus the prefetch lead */
+#define GEMX_ARENA_PAD_WORDS ((uint64_t)(2 * 4097 + 32) * 64)
+
 /* ---------------- device: value iterators ---------------- */
 
 /* float adaptive (lib/compress/float.go:139-161) streaming iterator,
@@ -228,9 +284,26 @@ struct FloatIter {
   double run_v;
   /* gorilla */
   BitR br;
+  GorA ga; /* arena-mode reader (coalesced interleaved stream) */
+  int use_arena;
   uint64_t g_val;
   uint8_t g_trail, g_mean;
   int g_first, g_done;
+
+  /* arena-mode gorilla init: the stream was re-laid at attach
+   * (build_gor_arena) and GorDesc carries the first value — branchless
+   * decode with wave-coalesced loads, same record semantics */
+  __device__ int init_gor_arena(const uint64_t *arena, const GorDesc *g) {
+    kind = 3;
+    use_arena = 1;
+    g_val = g->first_val;
+    g_done = (g_val == UVNAN);
+    g_first = 1;
+    g_trail = 0;
+    g_mean = 64;
+    ga.init(arena, g->arena_base);
+    return 0;
+  }
 
   __device__ int init(const uint8_t *enc, int64_t len) {
     if (len < 1) return -1;
@@ -256,6 +329,7 @@ struct FloatIter {
       run_left = 0;
       return 0;
     case 3: /* gorilla (batch_float.go:278-514); stream has tsm1 tag byte */
+      use_arena = 0;
       if (inlen < 9) { g_done = 1; g_first = 0; return 0; }
       g_val = d_u64be(in + 1);
       g_done = (g_val == UVNAN);
@@ -308,6 +382,53 @@ struct FloatIter {
         return 0;
       }
       if (g_done) return -1;
+      if (use_arena) {
+        /* branchless record decode from the interleaved arena (the
+         * k_scan_grid_gor record path; terminator -> exhausted) */
+        uint64_t A = GorA::fun(ga.w0, ga.w1, ga.bp);
+        uint32_t p13 = (uint32_t)(A >> 51);
+        uint32_t ctrl1 = p13 >> 12;
+        uint32_t neww = ctrl1 & ((p13 >> 11) & 1);
+        uint32_t mr = p13 & 0x3F;
+        g_mean = neww ? (uint8_t)(mr ? mr : 64u) : g_mean;
+        g_trail =
+            neww ? (uint8_t)(mr ? (64u - ((p13 >> 6) & 0x1F) - mr) : 0u)
+                 : g_trail;
+        uint32_t hdr = 1 + ctrl1 + (neww ? 11u : 0u);
+        uint64_t Cc = GorA::fun(ga.w1, ga.w2, ga.bp);
+        uint64_t B = (A << hdr) | (Cc >> (64 - hdr));
+        uint64_t sbv = (g_mean == 64) ? B : (B >> (64 - g_mean));
+        g_val ^= ctrl1 ? (sbv << (g_trail & 63)) : 0;
+        if (g_val == UVNAN) {
+          g_done = 1;
+          return -1;
+        }
+        uint32_t np = (uint32_t)ga.bp + hdr + (ctrl1 ? (uint32_t)g_mean : 0);
+        uint32_t adv = np >> 6;
+        ga.bp = (int)(np & 63);
+        if (__builtin_expect(adv >= 2, 0)) {
+          ga.w0 = ga.w2;
+          ga.w1 = ga.w3;
+          ga.w2 = ga.L;
+          ga.w3 = ga.M;
+          ga.p += 2 * 64;
+          ga.L = ga.ldw();
+          ga.M = ga.ldw();
+          ga.p -= 2 * 64;
+        } else {
+          const int c1 = (int)adv;
+          ga.w0 = c1 ? ga.w1 : ga.w0;
+          ga.w1 = c1 ? ga.w2 : ga.w1;
+          ga.w2 = c1 ? ga.w3 : ga.w2;
+          ga.w3 = c1 ? ga.L : ga.w3;
+          ga.L = c1 ? ga.M : ga.L;
+          ga.p += c1 ? 64 : 0;
+          ga.M = ga.p[64];
+        }
+        uint64_t u = g_val;
+        memcpy(out, &u, 8);
+        return 0;
+      }
       {
         /* branchless control decode: peek 13 bits (ctrl1+ctrl2+5 leading+
          * 6 meaningful), then the significant bits at a known offset —
@@ -717,15 +838,6 @@ struct SegQ {
   uint32_t series_idx;
 };
 
-/* per-segment gorilla arena descriptor (attach-time precompute for
- * k_scan_grid_gor): first decoded value, the segment's word-0 index in
- * the lane-interleaved stream arena, and the const-delta time params —
- * the kernel never touches the on-disk segment bytes. */
-struct GorDesc {
-  uint64_t first_val;  /* bits of value 0 (batch_float.go:293) */
-  uint64_t arena_base; /* u64 index; word k of this stream at base + 64*k */
-  int64_t t0, dt;      /* const-delta time (timestamp.go:190) */
-};
 
 /* gorilla sub-segment resume state (config #1 underfill fix): at attach,
  * when a shard yields too few gorilla lanes to fill the chip, the host
@@ -1118,52 +1230,6 @@ __global__ void __launch_bounds__(256) k_scan_fast(
  *    mid-stream uvnan (corrupt) or a missing terminator raises
  *    GEMX_E_DECODE after the fact instead of branching per record.
  */
-struct GorA {
-  const uint64_t *p;      /* arena cursor: word k of this stream lives at
-                             base + 64*k (u64 units, 512-byte stride). No
-                             bounds checks: the arena pad
-                             (GEMX_ARENA_PAD_WORDS) covers the maximum
-                             possible overshoot of a corrupt stream, whose
-                             garbage decode is then caught by the
-                             terminator check. */
-  uint64_t w0, w1, w2, w3; /* 256-bit window, bit cursor bp inside w0:
-                              any <=77-bit record at bp<=63 needs <=140
-                              bits = always inside w0..w2; w3+L+M are the
-                              refill pipeline */
-  uint64_t L, M;           /* 2-deep load pipeline (an 8-register batch
-                              refill and an LDS-DMA cache warmer both
-                              measured slower — r2 notes in DESIGN.md) */
-  int bp;                  /* 0..63 */
-
-  __device__ __forceinline__ uint64_t ldw() {
-    uint64_t w = *p;
-    p += 64;
-    return w;
-  }
-  __device__ __forceinline__ void init(const uint64_t *arena, uint64_t base) {
-    p = arena + base;
-    w0 = ldw();
-    w1 = ldw();
-    w2 = ldw();
-    w3 = ldw();
-    L = ldw();
-    M = ldw();
-    p -= 2 * 64; /* p tracks the address OF L until L is consumed */
-    bp = 0;
-  }
-  /* 64 bits starting at bit bp of (a,b); bp in [0,63] — the (>>1) split
-   * keeps the shift amount in range without a select */
-  static __device__ __forceinline__ uint64_t fun(uint64_t a, uint64_t b,
-                                                 int bp) {
-    return (a << bp) | ((b >> (63 - bp)) >> 1);
-  }
-};
-
-/* arena tail pad, u64 units: bounds the worst-case cursor overshoot of a
- * corrupt stream — (max rows+1) records x <=2 word-advances x 64 u64
- * stride, plus the prefetch lead */
-#define GEMX_ARENA_PAD_WORDS ((uint64_t)(2 * 4097 + 32) * 64)
-
 __global__ void __launch_bounds__(256) k_scan_grid_gor(
     const uint64_t *__restrict__ arena, uint64_t arena_words,
     const GorDesc *__restrict__ gors, const gemx_seg_desc *__restrict__ descs,
@@ -3089,7 +3155,8 @@ __device__ __forceinline__ void rate_slot_update(RateSlot *s, int64_t t, double 
 
 template <int FAST, int FUNC>
 __global__ void __launch_bounds__(256) k_rate_scan(
-    const uint8_t *__restrict__ blob, const gemx_seg_desc *__restrict__ descs,
+    const uint8_t *__restrict__ blob, const uint64_t *__restrict__ arena,
+    const GorDesc *__restrict__ gors, const gemx_seg_desc *__restrict__ descs,
     const RateSegQ *__restrict__ rsegq, const uint32_t *__restrict__ seg_ids,
     uint32_t nseg_ids, RatePartial *__restrict__ partials, int64_t start_sample,
     int64_t step_ns, int64_t range_ns, uint8_t *__restrict__ scratch,
@@ -3173,7 +3240,11 @@ __global__ void __launch_bounds__(256) k_rate_scan(
         return;
       }
       if (!h.one_value) {
-        int vrc = fit.init(h.enc, h.enc_len);
+        int vrc;
+        if ((h.enc[0] >> 4) == 3 && gors && gors[si].arena_base != ~0ull)
+          vrc = fit.init_gor_arena(arena, &gors[si]);
+        else
+          vrc = fit.init(h.enc, h.enc_len);
         if (vrc) { set_err(err, vrc == -2 ? GEMX_E_UNSUPPORTED : GEMX_E_DECODE); return; }
       }
     } else {
@@ -4314,7 +4385,9 @@ static int host_gor_walk(const uint8_t *stream, uint64_t stream_bytes,
 
 static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
   size_t n = s->fast_gor_ids.size();
-  s->h_gor.assign(s->nsegs, GorDesc{0, 0, 0, 0});
+  /* arena_base sentinel ~0: "this segment has no arena slot" (slot 0 is
+   * a valid base) — consumers fall back to the byte-stream readers */
+  s->h_gor.assign(s->nsegs, GorDesc{0, ~0ull, 0, 0});
   /* raw wave-kernel segments: time params + NaN screen. Go's NaN
    * comparison fall-through is order-dependent (no associative scan
    * form), so raw blocks holding any NaN are demoted to the sequential
@@ -4731,6 +4804,19 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
       delete s;
       return grc;
     }
+    /* slot-order the unified fast list too: k_rate_scan (and the
+     * interval-0 streaming launch) iterate it directly, and arena reads
+     * only coalesce when a wave's lanes sit in adjacent arena slots.
+     * Lane->segment assignment is free to permute (each lane's outputs
+     * go to its own per-segment slots). */
+    std::stable_sort(s->fast_ids.begin(), s->fast_ids.end(),
+                     [&](uint32_t x, uint32_t y) {
+                       uint64_t ax = s->h_gor.empty() ? ~0ull
+                                                      : s->h_gor[x].arena_base;
+                       uint64_t ay = s->h_gor.empty() ? ~0ull
+                                                      : s->h_gor[y].arena_base;
+                       return ax < ay;
+                     });
   }
   HIP_CHECK(hipStreamCreate(&s->stream));
   HIP_CHECK(hipStreamCreate(&s->copy_stream));
@@ -5729,7 +5815,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     HIP_CHECK(hipMemsetAsync(d_qcnt, 0, sizeof(uint32_t) * nb, s->stream));
 #define LAUNCH_QPHASE(FASTV, LIST, NSEG, SCR, SPL, LANES)                      \
     hipLaunchKernelGGL((k_rate_scan<FASTV, GEMX_PF_QUANTILE>), dim3(blocks),   \
-                       dim3(TPB), 0, s->stream, s->d_blob, s->d_descs,         \
+                       dim3(TPB), 0, s->stream, s->d_blob, s->d_arena, s->d_gor, s->d_descs,         \
                        P.d_rsegq, LIST, NSEG, P.d_rpart, start_sample,         \
                        eff_step, range_ns, SCR, SPL, LANES, P.d_rsq, d_qcnt,   \
                        d_qoff, d_qvals, d_qts, d_err)
@@ -5791,7 +5877,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     uint32_t blocks = std::min<uint32_t>((n + TPB - 1) / TPB, 65535);
 #define LAUNCH_RATE_FAST(F)                                                     \
     hipLaunchKernelGGL((k_rate_scan<1, F>), dim3(blocks), dim3(TPB), 0,         \
-                       s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
+                       s->stream, s->d_blob, s->d_arena, s->d_gor, s->d_descs, P.d_rsegq,             \
                        s->d_fast_ids, n, P.d_rpart, start_sample, eff_step,     \
                        range_ns, nullptr, 0, 0, nullptr, nullptr, nullptr,      \
                        nullptr, nullptr, d_err)
@@ -5819,7 +5905,7 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     uint32_t blocks = (P.gen_lanes + TPB - 1) / TPB;
 #define LAUNCH_RATE_GEN(F)                                                      \
     hipLaunchKernelGGL((k_rate_scan<0, F>), dim3(blocks), dim3(TPB), 0,         \
-                       s->stream, s->d_blob, s->d_descs, P.d_rsegq,             \
+                       s->stream, s->d_blob, s->d_arena, s->d_gor, s->d_descs, P.d_rsegq,             \
                        s->d_general_ids, n, P.d_rpart, start_sample, eff_step,  \
                        range_ns, P.d_scratch, scratch_per_lane, P.gen_lanes,    \
                        nullptr, nullptr, nullptr, nullptr, nullptr, d_err)
