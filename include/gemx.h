@@ -432,6 +432,19 @@ int gemx_encode_shard_pre(int col_type, const uint64_t *sids,
                           uint64_t *blob_bytes_out, gemx_agg_row *preagg_out,
                           uint64_t preagg_cap, uint64_t *n_preagg_out);
 
+/* Parse ONE reference ChunkMeta (engine/immutable/tssp_file_meta.go
+ * marshal layout) and emit attach-ready descriptors pairing the NAMED
+ * data column's segments with the chunk's time column — the
+ * column-splitting step for real multi-column TSSP chunks (a cgo caller
+ * loops chunks via *consumed_out and attaches one shard handle per
+ * selected column). blob is the TSSP file bytes (segment offsets are
+ * absolute); rows come from the time segment headers. */
+int gemx_chunkmeta_to_descs(const uint8_t *meta, uint64_t meta_len,
+                            const uint8_t *blob, uint64_t blob_bytes,
+                            const char *column, int col_type,
+                            gemx_seg_desc *descs_out, uint64_t cap,
+                            uint64_t *n_out, uint64_t *consumed_out);
+
 /* Seed an attached shard's pre-aggregation cache from write-side
  * metadata (one row per series, series order). The first covering
  * gemx_scan_preagg then runs zero kernels — the write-side equivalent

@@ -6693,3 +6693,136 @@ extern "C" int gemx_rec_from_rows(const gemx_agg_row *rows, uint64_t n_rows,
   }
   return GEMX_OK;
 }
+
+/* ---- reference ChunkMeta -> attach descriptors -------------------------
+ * Parses ONE ChunkMeta in the reference's marshal layout
+ * (engine/immutable/tssp_file_meta.go:566-581 marshal, :606-621
+ * unmarshal; Segment :92-101, SegmentRange :135-143, ColumnMeta
+ * :248-286; numberenc: u16/u32/u64 big-endian, int64 zigzag big-endian,
+ * lib/numberenc/number.go:53-167) and emits attach-ready descriptors
+ * pairing the NAMED data column's segments with the chunk's time column
+ * (the last colMeta entry, record.TimeField). This is the
+ * column-splitting step a cgo caller would otherwise hand-roll: a real
+ * multi-column TSSP chunk becomes one gemx_seg_desc[] per selected
+ * column. blob (the file bytes) is consulted only for the per-segment
+ * row counts (the time segment header: BlockIntegerOne=18 -> 1 row,
+ * BlockIntegerFull=32 -> count u32be, reader.go:638-646).
+ * Returns GEMX_OK and sets *n_out (= segCount) and *consumed_out (bytes
+ * of one chunk meta, for iterating a packed meta section);
+ * GEMX_E_INVALID on malformed input or when the column is absent. */
+static inline int64_t cm_zz64(const uint8_t *p) {
+  uint64_t u = h_u64be(p);
+  return (int64_t)(u >> 1) ^ -(int64_t)(u & 1);
+}
+
+extern "C" int gemx_chunkmeta_to_descs(const uint8_t *meta, uint64_t meta_len,
+                                       const uint8_t *blob,
+                                       uint64_t blob_bytes, const char *column,
+                                       int col_type,
+                                       gemx_seg_desc *descs_out, uint64_t cap,
+                                       uint64_t *n_out,
+                                       uint64_t *consumed_out) {
+  if (!meta || !blob || !column || !descs_out || !n_out) {
+    seterr("chunkmeta: bad arguments");
+    return GEMX_E_INVALID;
+  }
+  const uint8_t *p = meta, *end = meta + meta_len;
+#define CM_NEED(N)                                                             \
+  do {                                                                         \
+    if ((uint64_t)(end - p) < (uint64_t)(N)) {                                 \
+      seterr("chunkmeta: truncated");                                          \
+      return GEMX_E_INVALID;                                                   \
+    }                                                                          \
+  } while (0)
+  CM_NEED(28);
+  uint64_t sid = h_u64be(p);
+  p += 8;
+  p += 8; /* chunk data offset (zigzag i64) — segments carry absolute offsets */
+  p += 4; /* chunk size */
+  uint32_t ncols = h_u32be(p);
+  p += 4;
+  uint32_t nsegs = h_u32be(p);
+  p += 4;
+  if (ncols == 0 || nsegs == 0 || nsegs > (1u << 24)) {
+    seterr("chunkmeta: bad column/segment count");
+    return GEMX_E_INVALID;
+  }
+  CM_NEED((uint64_t)nsegs * 16);
+  const uint8_t *tr = p; /* timeRange[nsegs]: {min,max} zigzag i64 */
+  p += (uint64_t)nsegs * 16;
+
+  /* walk columns; remember the requested one and the time column (last) */
+  const uint8_t *want_entries = nullptr;
+  uint8_t want_ty = 0;
+  const uint8_t *time_entries = nullptr;
+  for (uint32_t c = 0; c < ncols; c++) {
+    CM_NEED(2);
+    uint32_t nl = ((uint32_t)p[0] << 8) | p[1];
+    p += 2;
+    CM_NEED(nl + 1 + 2);
+    const char *nm = (const char *)p;
+    p += nl;
+    uint8_t ty = *p++;
+    uint32_t pl = ((uint32_t)p[0] << 8) | p[1];
+    p += 2;
+    CM_NEED(pl);
+    p += pl; /* preAgg bytes (count,min,max,minT,maxT,sum) — not needed */
+    CM_NEED((uint64_t)nsegs * 12);
+    bool is_time = (nl == 4 && memcmp(nm, "time", 4) == 0);
+    if (is_time) time_entries = p;
+    if (!is_time && nl == strlen(column) && memcmp(nm, column, nl) == 0) {
+      want_entries = p;
+      want_ty = ty;
+    }
+    p += (uint64_t)nsegs * 12;
+  }
+  if (consumed_out) *consumed_out = (uint64_t)(p - meta);
+  if (!time_entries) {
+    seterr("chunkmeta: no time column");
+    return GEMX_E_INVALID;
+  }
+  if (!want_entries) {
+    seterr("chunkmeta: column not found");
+    return GEMX_E_INVALID;
+  }
+  if ((int)want_ty != col_type) {
+    seterr("chunkmeta: column type mismatch");
+    return GEMX_E_INVALID;
+  }
+  if (cap < nsegs) {
+    seterr("descs capacity too small");
+    return GEMX_E_CAP;
+  }
+  for (uint32_t i = 0; i < nsegs; i++) {
+    gemx_seg_desc &d = descs_out[i];
+    d.sid = sid;
+    int64_t doff = cm_zz64(want_entries + (uint64_t)i * 12);
+    uint32_t dsz = h_u32be(want_entries + (uint64_t)i * 12 + 8);
+    int64_t toff = cm_zz64(time_entries + (uint64_t)i * 12);
+    uint32_t tsz = h_u32be(time_entries + (uint64_t)i * 12 + 8);
+    if (doff < 0 || toff < 0 || (uint64_t)doff + dsz > blob_bytes ||
+        (uint64_t)toff + tsz > blob_bytes || tsz < 1) {
+      seterr("chunkmeta: segment out of file bounds");
+      return GEMX_E_INVALID;
+    }
+    d.data_offset = (uint64_t)doff;
+    d.data_size = dsz;
+    d.time_offset = (uint64_t)toff;
+    d.time_size = tsz;
+    d._pad = 0;
+    d.min_time = cm_zz64(tr + (uint64_t)i * 16);
+    d.max_time = cm_zz64(tr + (uint64_t)i * 16 + 8);
+    const uint8_t *ts = blob + toff;
+    if (ts[0] == 18) { /* BlockIntegerOne */
+      d.rows = 1;
+    } else if (ts[0] == 32 && tsz >= 5) { /* BlockIntegerFull */
+      d.rows = h_u32be(ts + 1);
+    } else {
+      seterr("chunkmeta: unrecognized time segment header");
+      return GEMX_E_INVALID;
+    }
+  }
+#undef CM_NEED
+  *n_out = nsegs;
+  return GEMX_OK;
+}
