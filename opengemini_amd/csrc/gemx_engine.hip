@@ -219,6 +219,71 @@ struct GorDesc {
   int64_t t0, dt;      /* const-delta time (timestamp.go:190) */
 };
 
+#ifndef GEMX_GOR_BATCH_REFILL
+#define GEMX_GOR_BATCH_REFILL 0
+#endif
+#if GEMX_GOR_BATCH_REFILL
+struct GorA {
+  const uint64_t *pR;
+  uint64_t w0, w1, w2, w3;
+  uint64_t R0, R1, R2, R3, R4, R5, R6, R7; /* batch queue: 8 loads per
+                              per-lane refill event, one amortized
+                              latency per batch */
+  int fidx;
+  int bp;
+  __device__ __forceinline__ void refill() {
+    R0 = pR[0 * 64];
+    R1 = pR[1 * 64];
+    R2 = pR[2 * 64];
+    R3 = pR[3 * 64];
+    R4 = pR[4 * 64];
+    R5 = pR[5 * 64];
+    R6 = pR[6 * 64];
+    R7 = pR[7 * 64];
+    pR += 8 * 64;
+    fidx = 0;
+  }
+  __device__ __forceinline__ uint64_t pull() {
+    if (__builtin_expect(fidx >= 8, 0)) refill();
+    uint64_t lo01 = (fidx & 1) ? R1 : R0;
+    uint64_t lo23 = (fidx & 1) ? R3 : R2;
+    uint64_t hi45 = (fidx & 1) ? R5 : R4;
+    uint64_t hi67 = (fidx & 1) ? R7 : R6;
+    uint64_t lo = (fidx & 2) ? lo23 : lo01;
+    uint64_t hi = (fidx & 2) ? hi67 : hi45;
+    uint64_t v = (fidx & 4) ? hi : lo;
+    fidx++;
+    return v;
+  }
+  __device__ __forceinline__ void init(const uint64_t *arena, uint64_t base) {
+    pR = arena + base;
+    w0 = pR[0 * 64];
+    w1 = pR[1 * 64];
+    w2 = pR[2 * 64];
+    w3 = pR[3 * 64];
+    pR += 4 * 64;
+    refill();
+    bp = 0;
+  }
+  __device__ __forceinline__ void step(uint32_t adv) {
+    if (__builtin_expect(adv >= 2, 0)) {
+      w0 = w2;
+      w1 = w3;
+      w2 = pull();
+      w3 = pull();
+    } else if (adv) {
+      w0 = w1;
+      w1 = w2;
+      w2 = w3;
+      w3 = pull();
+    }
+  }
+  static __device__ __forceinline__ uint64_t fun(uint64_t a, uint64_t b,
+                                                 int bp) {
+    return (a << bp) | ((b >> (63 - bp)) >> 1);
+  }
+};
+#else
 struct GorA {
   const uint64_t *p;      /* arena cursor: word k of this stream lives at
                              base + 64*k (u64 units, 512-byte stride). No
@@ -252,6 +317,27 @@ struct GorA {
     p -= 2 * 64; /* p tracks the address OF L until L is consumed */
     bp = 0;
   }
+  __device__ __forceinline__ void step(uint32_t adv) {
+    if (__builtin_expect(adv >= 2, 0)) { /* rare: wide record */
+      w0 = w2;
+      w1 = w3;
+      w2 = L;
+      w3 = M;
+      p += 2 * 64;
+      L = ldw();
+      M = ldw();
+      p -= 2 * 64;
+    } else {
+      const int c1 = (int)adv;
+      w0 = c1 ? w1 : w0;
+      w1 = c1 ? w2 : w1;
+      w2 = c1 ? w3 : w2;
+      w3 = c1 ? L : w3;
+      L = c1 ? M : L;
+      p += c1 ? 64 : 0;
+      M = p[64];
+    }
+  }
   /* 64 bits starting at bit bp of (a,b); bp in [0,63] — the (>>1) split
    * keeps the shift amount in range without a select */
   static __device__ __forceinline__ uint64_t fun(uint64_t a, uint64_t b,
@@ -259,6 +345,8 @@ struct GorA {
     return (a << bp) | ((b >> (63 - bp)) >> 1);
   }
 };
+
+#endif
 
 /* arena tail pad, u64 units: bounds the worst-case cursor overshoot of a
  * corrupt stream — (max rows+1) records x <=2 word-advances x 64 u64
@@ -406,25 +494,7 @@ struct FloatIter {
         uint32_t np = (uint32_t)ga.bp + hdr + (ctrl1 ? (uint32_t)g_mean : 0);
         uint32_t adv = np >> 6;
         ga.bp = (int)(np & 63);
-        if (__builtin_expect(adv >= 2, 0)) {
-          ga.w0 = ga.w2;
-          ga.w1 = ga.w3;
-          ga.w2 = ga.L;
-          ga.w3 = ga.M;
-          ga.p += 2 * 64;
-          ga.L = ga.ldw();
-          ga.M = ga.ldw();
-          ga.p -= 2 * 64;
-        } else {
-          const int c1 = (int)adv;
-          ga.w0 = c1 ? ga.w1 : ga.w0;
-          ga.w1 = c1 ? ga.w2 : ga.w1;
-          ga.w2 = c1 ? ga.w3 : ga.w2;
-          ga.w3 = c1 ? ga.L : ga.w3;
-          ga.L = c1 ? ga.M : ga.L;
-          ga.p += c1 ? 64 : 0;
-          ga.M = ga.p[64];
-        }
+        ga.step(adv);
         uint64_t u = g_val;
         memcpy(out, &u, 8);
         return 0;
@@ -1284,25 +1354,7 @@ __global__ void __launch_bounds__(256) k_scan_grid_gor(
       uint32_t np = (uint32_t)br.bp + hdr + (ctrl1 ? g_mean : 0);              \
       uint32_t adv = np >> 6;                                                  \
       br.bp = (int)(np & 63);                                                  \
-      if (__builtin_expect(adv >= 2, 0)) { /* rare: wide record */             \
-        br.w0 = br.w2;                                                         \
-        br.w1 = br.w3;                                                         \
-        br.w2 = br.L;                                                          \
-        br.w3 = br.M;                                                          \
-        br.p += 2 * 64;                                                        \
-        br.L = br.ldw();                                                       \
-        br.M = br.ldw();                                                       \
-        br.p -= 2 * 64;                                                        \
-      } else {                                                                 \
-        const int c1 = (int)adv;                                               \
-        br.w0 = c1 ? br.w1 : br.w0;                                            \
-        br.w1 = c1 ? br.w2 : br.w1;                                            \
-        br.w2 = c1 ? br.w3 : br.w2;                                            \
-        br.w3 = c1 ? br.L : br.w3;                                             \
-        br.L = c1 ? br.M : br.L;                                               \
-        br.p += c1 ? 64 : 0;                                                   \
-        br.M = br.p[64];                                                       \
-      }                                                                        \
+      br.step(adv);                                                          \
     } while (0)
 
     int first_pending = 1;
@@ -1421,25 +1473,7 @@ __global__ void __launch_bounds__(256) k_scan_gor_sub(
       uint32_t np = (uint32_t)br.bp + hdr + (ctrl1 ? g_mean : 0);              \
       uint32_t adv = np >> 6;                                                  \
       br.bp = (int)(np & 63);                                                  \
-      if (__builtin_expect(adv >= 2, 0)) {                                     \
-        br.w0 = br.w2;                                                         \
-        br.w1 = br.w3;                                                         \
-        br.w2 = br.L;                                                          \
-        br.w3 = br.M;                                                          \
-        br.p += 2 * 64;                                                        \
-        br.L = br.ldw();                                                       \
-        br.M = br.ldw();                                                       \
-        br.p -= 2 * 64;                                                        \
-      } else {                                                                 \
-        const int c1 = (int)adv;                                               \
-        br.w0 = c1 ? br.w1 : br.w0;                                            \
-        br.w1 = c1 ? br.w2 : br.w1;                                            \
-        br.w2 = c1 ? br.w3 : br.w2;                                            \
-        br.w3 = c1 ? br.L : br.w3;                                             \
-        br.L = c1 ? br.M : br.L;                                               \
-        br.p += c1 ? 64 : 0;                                                   \
-        br.M = br.p[64];                                                       \
-      }                                                                        \
+      br.step(adv);                                                          \
     } while (0)
 
     int first_pending = 1;
