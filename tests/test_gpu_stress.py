@@ -83,3 +83,81 @@ def test_mixed_query_stability():
     finally:
         sh.close()
         sh2.close()
+
+
+class TestSubSplitStress:
+    """The gorilla sub-segment split (config #1 underfill fix) triggers on
+    every small shard (< 65536 gorilla lanes), so the whole parity suite
+    exercises it — these cases target its edges specifically."""
+
+    def test_deep_series_many_subs_per_window(self):
+        # 5000-row segments -> ~16 subs/segment; 60-row windows cross sub
+        # boundaries constantly; parity must stay exact
+        import binding as orc
+        from shard_helpers import INT, F, build_shard
+        import numpy as np
+        import opengemini_amd as gx
+
+        rng = np.random.default_rng(71)
+        blob, d, _ = build_shard(rng, F, [1, 2], seg_range=(2, 3),
+                                 row_range=(3000, 4096), null_frac=0.0)
+        sh = gx.Shard(blob, d, F)
+        try:
+            rows, _ = sh.scan_agg(0, 2**62, 60 * 10**9)
+        finally:
+            sh.close()
+        ref = orc.scan_agg(blob, d, F, 0, 2**62, 60 * 10**9)
+        from test_gpu_parity import assert_parity
+        assert_parity(rows, ref, F)
+
+    def test_mixed_split_and_general_segments_one_series(self):
+        # a series whose segments alternate between nil-free gorilla
+        # (sub-split path) and nil-bearing blocks (general kernel): the
+        # per-series window merge folds partials from BOTH pipelines
+        import binding as orc
+        from shard_helpers import F
+        import numpy as np
+        import opengemini_amd as gx
+
+        rng = np.random.default_rng(72)
+        blob = bytearray()
+        descs = []
+        t = 0
+        for k in range(6):
+            rows = 900
+            times = t + np.arange(rows, dtype=np.int64) * 10**9
+            t = int(times[-1]) + 10**9
+            vals = np.round(np.cumsum(rng.normal(0, 1, rows)) * 128) / 128
+            if k % 2 == 0:
+                valid = np.ones(rows, dtype=bool)
+            else:
+                valid = rng.random(rows) > 0.3
+            nil = int((~valid).sum())
+            bm = np.packbits(valid.astype(np.uint8), bitorder="little")
+            dseg = orc.encode_data_segment(F, vals[valid], bm, rows, nil)
+            tseg = orc.encode_time_segment(times)
+            descs.append((5, len(blob), len(dseg), rows,
+                          len(blob) + len(dseg), len(tseg), 0,
+                          int(times[0]), int(times[-1])))
+            blob += dseg + tseg
+        d = np.zeros(len(descs), dtype=orc.SEG_DESC_DTYPE)
+        for i, tup in enumerate(descs):
+            d[i] = tup
+        blob = bytes(blob)
+        sh = gx.Shard(blob, d, F)
+        try:
+            rows_g, _ = sh.scan_agg(0, 2**62, 60 * 10**9)
+            rows_g = rows_g.copy()  # pooled: valid until the next query
+            grows, _ = sh.scan_agg(0, 2**62, 60 * 10**9, group_all=True)
+            grows = grows.copy()
+        finally:
+            sh.close()
+        ref = orc.scan_agg(blob, d, F, 0, 2**62, 60 * 10**9)
+        from test_gpu_parity import assert_parity
+        assert_parity(rows_g, ref, F)
+        gref = orc.group_merge(ref, F, 60 * 10**9)
+        assert len(grows) == len(gref)
+        assert np.array_equal(grows["count"], gref["count"])
+        assert np.array_equal(grows["min"].view(np.uint64),
+                              gref["min"].view(np.uint64))
+        assert np.array_equal(grows["min_time"], gref["min_time"])
