@@ -1780,7 +1780,10 @@ __global__ void __launch_bounds__(256) k_scan_raw_lane(
     const GorDesc g = gors[si];
     const int64_t t0c = g.t0, dtc = g.dt;
     const int rows = (int)d.rows;
-    const uint64_t *w = arena + g.arena_base; /* row k at w[64*k] */
+    /* row k at w[(k>>1)*128 + (k&1)]: 16-byte pairs per lane so batch
+     * loads are dwordx4 (two rows per load, half the load instructions,
+     * 1 KiB contiguous per wave-level pair-load) */
+    const uint64_t *w = arena + g.arena_base;
 
     Partial *base = partials + sq.partial_base;
     for (uint32_t k = 0; k < sq.n_wins; k++) base[k].has_rows = 0;
@@ -1851,14 +1854,22 @@ __global__ void __launch_bounds__(256) k_scan_raw_lane(
     uint64_t c0, c1, c2, c3, c4, c5, c6, c7;
     uint64_t e0, e1, e2, e3, e4, e5, e6, e7;
 #define RAW_LOAD(R, K)                                                         \
-    R##0 = w[(K + 0) * 64];                                                    \
-    R##1 = w[(K + 1) * 64];                                                    \
-    R##2 = w[(K + 2) * 64];                                                    \
-    R##3 = w[(K + 3) * 64];                                                    \
-    R##4 = w[(K + 4) * 64];                                                    \
-    R##5 = w[(K + 5) * 64];                                                    \
-    R##6 = w[(K + 6) * 64];                                                    \
-    R##7 = w[(K + 7) * 64]
+    do {                                                                       \
+      const ulonglong2 *w2_ =                                                  \
+          (const ulonglong2 *)(w + ((uint64_t)(K) >> 1) * 128);                \
+      ulonglong2 x0_ = w2_[0 * 64];                                            \
+      ulonglong2 x1_ = w2_[1 * 64];                                            \
+      ulonglong2 x2_ = w2_[2 * 64];                                            \
+      ulonglong2 x3_ = w2_[3 * 64];                                            \
+      R##0 = x0_.x;                                                            \
+      R##1 = x0_.y;                                                            \
+      R##2 = x1_.x;                                                            \
+      R##3 = x1_.y;                                                            \
+      R##4 = x2_.x;                                                            \
+      R##5 = x2_.y;                                                            \
+      R##6 = x3_.x;                                                            \
+      R##7 = x3_.y;                                                            \
+    } while (0)
 #define RAW_USE(R, K)                                                          \
     RAW_ROW(K + 0, R##0);                                                      \
     RAW_ROW(K + 1, R##1);                                                      \
@@ -1889,7 +1900,7 @@ __global__ void __launch_bounds__(256) k_scan_raw_lane(
         RAW_USE(e, i + 24);
       }
     }
-    for (; i < rows; i++) RAW_ROW(i, w[i * 64]);
+    for (; i < rows; i++) RAW_ROW(i, w[((uint64_t)i >> 1) * 128 + (i & 1)]);
     if (cur_ord != INT64_MIN) RAW_FLUSH();
 #undef RAW_LOAD
 #undef RAW_USE
@@ -4414,7 +4425,9 @@ static int host_gor_walk(const uint8_t *stream, uint64_t stream_bytes,
 
 /* lanes below which the gorilla grid is considered underfilled and worth
  * the attach-time split walk (256 CU x 4 SIMD x 16 wave slots x 64) */
+#ifndef GEMX_SPLIT_MIN_LANES
 #define GEMX_SPLIT_MIN_LANES 65536u
+#endif
 #define GEMX_SUB_MIN_ROWS 64
 
 static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
@@ -4515,13 +4528,29 @@ static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
   }
   uint64_t total = 0; /* u64 units */
   size_t nall = items.size();
-  for (size_t g0 = 0; g0 < nall; g0 += 64) {
-    size_t ge = std::min(g0 + 64, nall);
+  size_t ngor = 0;
+  while (ngor < nall && !items[ngor].raw) ngor++;
+  /* gorilla groups: 1-word (8 B) interleave — word k of slot j at
+   * group_base + k*64 + j */
+  for (size_t g0 = 0; g0 < ngor; g0 += 64) {
+    size_t ge = std::min(g0 + 64, ngor);
     uint64_t maxw = 1;
     for (size_t j = g0; j < ge; j++) maxw = std::max(maxw, items[j].words);
     for (size_t j = g0; j < ge; j++)
       s->h_gor[items[j].id].arena_base = total + (j - g0);
     total += maxw * 64;
+  }
+  /* raw groups: 2-word (16 B) interleave so the kernel loads dwordx4 —
+   * pair p of slot j at group_base + p*128 + j*2, i.e. word k at
+   * base + (k>>1)*128 + (k&1) with base = group_base + j*2 */
+  for (size_t g0 = ngor; g0 < nall; g0 += 64) {
+    size_t ge = std::min(g0 + 64, nall);
+    uint64_t maxp = 1;
+    for (size_t j = g0; j < ge; j++)
+      maxp = std::max(maxp, (items[j].words + 1) / 2);
+    for (size_t j = g0; j < ge; j++)
+      s->h_gor[items[j].id].arena_base = total + (j - g0) * 2;
+    total += maxp * 128;
   }
   total += GEMX_ARENA_PAD_WORDS; /* see GorA */
   std::vector<uint64_t> h_arena(total, 0);
@@ -4533,7 +4562,7 @@ static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
       for (uint64_t k = 0; k < full; k++) {
         uint64_t w;
         memcpy(&w, it.stream + k * 8, 8);
-        dst[k * 64] = w; /* values stay little-endian */
+        dst[(k >> 1) * 128 + (k & 1)] = w; /* little-endian, 16 B pairs */
       }
     } else {
       for (uint64_t k = 0; k < full; k++) {
