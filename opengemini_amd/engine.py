@@ -320,6 +320,30 @@ def encode_shard(col_type, sids, times, values, valid=None, seg_rows=1000,
     return blob[: used.value].tobytes(), descs[: nseg.value].copy()
 
 
+def chunkmeta_to_descs(meta, blob, column, col_type, cap=None):
+    """Parse ONE reference ChunkMeta (tssp_file_meta.go marshal layout)
+    into attach-ready descriptors for the named data column, paired with
+    the chunk's time column — the column-splitting step for real
+    multi-column TSSP chunks. Returns (descs, consumed_bytes); iterate a
+    packed meta section by slicing meta by consumed_bytes."""
+    lib = _load()
+    lib.gemx_chunkmeta_to_descs.restype = C.c_int
+    m = np.frombuffer(bytearray(meta), dtype=np.uint8)
+    b = np.frombuffer(bytearray(blob), dtype=np.uint8)
+    if cap is None:
+        cap = max(16, len(m) // 12)
+    descs = np.zeros(cap, dtype=SEG_DESC_DTYPE)
+    n = C.c_uint64(0)
+    used = C.c_uint64(0)
+    rc = lib.gemx_chunkmeta_to_descs(
+        m.ctypes.data_as(C.c_void_p), len(m), b.ctypes.data_as(C.c_void_p),
+        len(b), column.encode(), col_type,
+        descs.ctypes.data_as(C.c_void_p), C.c_uint64(cap), C.byref(n),
+        C.byref(used))
+    _check(rc, lib)
+    return descs[: n.value].copy(), used.value
+
+
 def abi_version():
     return _load().gemx_abi_version()
 
