@@ -4142,6 +4142,9 @@ struct gemx_shard {
   };
   std::vector<SeriesRange> series_ranges;
   uint64_t total_rows_scanned; /* Σ rows */
+  uint64_t total_compressed_bytes = 0; /* Σ data+time segment bytes (cached:
+      summing 1M descriptors per query in scan_deliver cost ~1 ms/step on
+      the config-#3 shape) */
   hipStream_t stream;
   /* async pipeline: kernels run on `stream`, row D2H on `copy_stream`
    * gated by an event — the copy of query i overlaps the kernels of
@@ -4842,6 +4845,7 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
     s->is_raw[i] = fast && grid && raw && !gor;
     s->is_s8b[i] = fast && grid && s8b;
     s->total_rows_scanned += d.rows;
+    s->total_compressed_bytes += d.data_size + d.time_size;
     /* series ranges + per-series/shard time bounds (for preagg coverage) */
     if (s->series_ranges.empty() || s->series_ranges.back().sid != d.sid) {
       s->series_ranges.push_back({d.sid, (uint32_t)i, 1});
@@ -5688,9 +5692,7 @@ static int scan_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
     stats->merge_ms = ms_merge;
     stats->total_ms = ms_total;
     stats->points = s->total_rows_scanned;
-    uint64_t cb = 0;
-    for (auto &d : s->h_descs) cb += d.data_size + d.time_size;
-    stats->compressed_bytes = cb;
+    stats->compressed_bytes = s->total_compressed_bytes;
     stats->n_rows = n;
     stats->h2d_ms = sync_ms + comp_ms; /* repurposed: host sync+compact ms */
   }
@@ -6090,9 +6092,7 @@ static int rate_deliver(gemx_shard *s, int slot, uint64_t fetch_rows,
     stats->merge_ms = ms_merge;
     stats->total_ms = ms_total;
     stats->points = s->total_rows_scanned;
-    uint64_t cb = 0;
-    for (auto &d : s->h_descs) cb += d.data_size + d.time_size;
-    stats->compressed_bytes = cb;
+    stats->compressed_bytes = s->total_compressed_bytes;
     stats->n_rows = n;
     stats->h2d_ms = 0;
   }
