@@ -249,6 +249,33 @@ class TestGroupedParity:
         blob, descs = orc.gen_shard(46, 2000, 1000)
         self._run(blob, descs, F)
 
+    def test_grouped_bulk_int_tiled(self):
+        # 1 segment/series, >32 series: the tiled p1 staging path, int
+        blob, descs = orc.gen_shard(47, 1500, 1000, mode=orc.GEN_INT_SMALL)
+        self._run(blob, descs, I)
+
+    def test_grouped_tiled_sparse_ranges(self):
+        # series with disjoint time ranges: chunks whose members cover
+        # different window spans (and some series out of query range)
+        rng = np.random.default_rng(48)
+        blob = bytearray()
+        descs = []
+        for sid in range(1, 120):
+            rows = 400
+            t0 = (sid % 7) * 400 * 10**9
+            times = t0 + np.arange(rows, dtype=np.int64) * 10**9
+            vals = np.round(np.cumsum(rng.normal(0, 1, rows)) * 128) / 128
+            dseg = orc.encode_data_segment(F, vals, None, rows, 0)
+            tseg = orc.encode_time_segment(times)
+            descs.append((sid, len(blob), len(dseg), rows,
+                          len(blob) + len(dseg), len(tseg), 0,
+                          int(times[0]), int(times[-1])))
+            blob += dseg + tseg
+        d = np.zeros(len(descs), dtype=orc.SEG_DESC_DTYPE)
+        for i, tup in enumerate(descs):
+            d[i] = tup
+        self._run(bytes(blob), d, F)
+
     def test_grouped_value_ties(self):
         # equal values across series: earliest-time, then first-series wins
         rng = np.random.default_rng(52)
