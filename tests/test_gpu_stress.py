@@ -235,3 +235,63 @@ class TestSeededFuzzParity:
                 sh.close()
             ref = orc.scan_agg(blob, d, I, 0, 2**62, interval)
             assert_parity(rows, ref, I)
+
+
+class TestCorruptionDetection:
+    """Corrupt streams must fail LOUDLY (GEMX_E_DECODE), never decode
+    garbage silently: small shards are caught by the attach-time host
+    walk (sub-split validation), zstd frames by the transcode, and the
+    general/streaming kernels by their own checks."""
+
+    def test_corrupt_gorilla_stream_rejected_at_attach(self):
+        import binding as orc
+        import numpy as np
+        import opengemini_amd as gx
+
+        blob, descs = orc.gen_shard(60, 20, 1000)
+        b = bytearray(blob)
+        # flip bits in the middle of the first data segment's stream
+        off = int(np.asarray(descs)[0]["data_offset"]) + 40
+        for i in range(16):
+            b[off + i] ^= 0xFF
+        with pytest.raises(gx.GemxError):
+            sh = gx.Shard(bytes(b), descs, gx.engine.GEMX_TYPE_FLOAT)
+            try:
+                sh.scan_agg(0, 2**62, 60 * 10**9)
+            finally:
+                sh.close()
+
+    def test_corrupt_zstd_frame_rejected(self):
+        import numpy as np
+        import opengemini_amd as gx
+        from shard_helpers import I, build_shard
+
+        rng = np.random.default_rng(61)
+        blob, d, _ = build_shard(
+            rng, I, [1], null_frac=0.0,
+            value_fn=lambda r, n: np.where(
+                np.arange(n) % 2 == 0, np.int64(2**61), np.int64(7)))
+        # find a zstd segment (tag 3) and trash its frame bytes
+        b = bytearray(blob)
+        hit = False
+        for dd in np.asarray(d):
+            off = int(dd["data_offset"])
+            if b[off + 5] >> 4 == 3:
+                for i in range(10, 26):
+                    b[off + 5 + i] ^= 0xA5
+                hit = True
+                break
+        assert hit, "no zstd segment produced"
+        with pytest.raises(gx.GemxError):
+            gx.Shard(bytes(b), d, gx.engine.GEMX_TYPE_INT).close()
+
+    def test_truncated_descriptor_rejected(self):
+        import binding as orc
+        import numpy as np
+        import opengemini_amd as gx
+
+        blob, descs = orc.gen_shard(62, 5, 500)
+        d = np.asarray(descs).copy()
+        d[0]["data_size"] = 2**31  # out of blob bounds
+        with pytest.raises(gx.GemxError):
+            gx.Shard(blob, d, gx.engine.GEMX_TYPE_FLOAT).close()
