@@ -233,9 +233,10 @@ def main():
                 import torch.distributed as dist
 
                 nsteps = max(1, (args.pts * 10**9 - RANGE_NS) // WINDOW_NS + 1)
-                sums = np.zeros(int(nsteps))
                 idx = ((rows["ts"] - RANGE_NS) // WINDOW_NS).astype(np.int64)
-                np.add.at(sums, np.clip(idx, 0, int(nsteps) - 1), rows["value"])
+                sums = np.bincount(np.clip(idx, 0, int(nsteps) - 1),
+                                   weights=rows["value"],
+                                   minlength=int(nsteps))
                 t = torch.from_numpy(sums).to(f"cuda:{local_rank}")
                 dist.all_reduce(t, op=dist.ReduceOp.SUM)
         return rows, stats
