@@ -338,3 +338,88 @@ class TestSimple8bCrossCheck:
             word |= v << (i * 3)
         got = orc.simple8b_decode_word(word)
         assert list(got[:20]) == vals
+
+
+# ------------------------------------------------------------- time codec
+
+def py_time_decode(buf):
+    """From-spec timestamp block decoder (lib/encoding/timestamp.go):
+    tag nibble in byte 0: 1 = const-delta ([first u64be][delta uvarint]
+    [count uvarint], :190-225), 2 = simple8b x scale ([scale u64be]
+    [encCount u32be][srcCount u32be][first u64be][words u64be...],
+    :227-272), 4 = uncompressed zigzag u64be each (:299-308)."""
+    tag = buf[0] >> 4
+    p = 1
+
+    def uvarint(pos):
+        v, sh = 0, 0
+        while True:
+            b = buf[pos]
+            pos += 1
+            v |= (b & 0x7F) << sh
+            sh += 7
+            if not b & 0x80:
+                return v, pos
+
+    if tag == 1:
+        first = int.from_bytes(buf[p:p + 8], "big", signed=False)
+        p += 8
+        delta, p = uvarint(p)
+        count, p = uvarint(p)
+        out = [first + i * delta for i in range(count + 1)]
+        return [v - 2**64 if v >= 2**63 else v for v in out]
+    if tag == 2:
+        scale = int.from_bytes(buf[p:p + 8], "big")
+        enc_count = int.from_bytes(buf[p + 8:p + 12], "big")
+        src_count = int.from_bytes(buf[p + 12:p + 16], "big")
+        p += 16
+        cur = int.from_bytes(buf[p:p + 8], "big")
+        p += 8
+        out = [cur]
+        for w in range(enc_count - 1):
+            word = int.from_bytes(buf[p:p + 8], "big")
+            p += 8
+            for v in py_simple8b_decode_word(word):
+                cur = (cur + v * scale) & M64
+                out.append(cur)
+        assert len(out) >= src_count
+        return out[:src_count]
+    if tag == 4:
+        n = int.from_bytes(buf[p:p + 4], "big") // 8
+        p += 4
+        out = []
+        for i in range(n):
+            u = int.from_bytes(buf[p + i * 8:p + i * 8 + 8], "big")
+            out.append((u >> 1) ^ -(u & 1))
+        return out
+    raise ValueError(f"tag {tag}")
+
+
+class TestTimeCodecCrossCheck:
+    def test_const_delta(self):
+        times = np.arange(500, dtype=np.int64) * 60 * 10**9 + 17
+        enc = orc.time_encode(times)
+        assert enc[0] >> 4 == 1
+        assert py_time_decode(bytes(enc)) == list(times)
+
+    def test_simple8b_scale(self):
+        rng = np.random.default_rng(19)
+        deltas = rng.integers(1, 10, 400) * 10**9  # common 1e9 scale
+        times = np.cumsum(deltas).astype(np.int64)
+        enc = orc.time_encode(times)
+        assert enc[0] >> 4 == 2
+        assert py_time_decode(bytes(enc)) == list(times)
+
+    def test_uncompressed(self):
+        rng = np.random.default_rng(20)
+        deltas = rng.integers(1, 10**6, 50)
+        deltas[25] = 2**61  # above simple8b's bound -> raw form
+        times = np.cumsum(deltas).astype(np.int64)
+        enc = orc.time_encode(times)
+        assert enc[0] >> 4 == 4
+        assert py_time_decode(bytes(enc)) == list(times)
+
+    def test_negative_first_time(self):
+        times = -300 * 10**9 + np.arange(100, dtype=np.int64) * 10**9
+        enc = orc.time_encode(times)
+        assert py_time_decode(bytes(enc)) == list(times)

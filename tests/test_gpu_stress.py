@@ -295,3 +295,47 @@ class TestCorruptionDetection:
         d[0]["data_size"] = 2**31  # out of blob bounds
         with pytest.raises(gx.GemxError):
             gx.Shard(blob, d, gx.engine.GEMX_TYPE_FLOAT).close()
+
+
+class TestNegativeTimes:
+    """Pre-epoch timestamps: window ordinals come from FLOORED division
+    (aggregate_cursor.go uses Go's math semantics via hybridqp WindowStartTime;
+    truncation-toward-zero would misplace every pre-1970 row)."""
+
+    def test_negative_t0_parity(self):
+        import binding as orc
+        import numpy as np
+        import opengemini_amd as gx
+        from shard_helpers import F
+        from test_gpu_parity import assert_parity
+
+        rng = np.random.default_rng(77)
+        blob = bytearray()
+        descs = []
+        for sid in (1, 2):
+            rows = 700
+            t0 = -350 * 10**9  # spans the epoch
+            times = t0 + np.arange(rows, dtype=np.int64) * 10**9
+            vals = np.round(np.cumsum(rng.normal(0, 1, rows)) * 128) / 128
+            dseg = orc.encode_data_segment(F, vals, None, rows, 0)
+            tseg = orc.encode_time_segment(times)
+            descs.append((sid, len(blob), len(dseg), rows,
+                          len(blob) + len(dseg), len(tseg), 0,
+                          int(times[0]), int(times[-1])))
+            blob += dseg + tseg
+        d = np.zeros(len(descs), dtype=orc.SEG_DESC_DTYPE)
+        for i, tup in enumerate(descs):
+            d[i] = tup
+        blob = bytes(blob)
+        for interval, offset in ((60 * 10**9, 0), (60 * 10**9, 13 * 10**9),
+                                 (7 * 10**9, 0)):
+            sh = gx.Shard(blob, d, gx.engine.GEMX_TYPE_FLOAT)
+            try:
+                rows_g, _ = sh.scan_agg(-2**62, 2**62, interval,
+                                        offset=offset)
+                rows_g = rows_g.copy()
+            finally:
+                sh.close()
+            ref = orc.scan_agg(blob, d, orc.ORC_TYPE_FLOAT, -2**62, 2**62,
+                               interval, offset=offset)
+            assert_parity(rows_g, ref, orc.ORC_TYPE_FLOAT)
