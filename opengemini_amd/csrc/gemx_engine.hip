@@ -4556,35 +4556,67 @@ static int build_gor_arena(gemx_shard *s, const uint8_t *blob) {
     total += maxp * 128;
   }
   total += GEMX_ARENA_PAD_WORDS; /* see GorA */
-  std::vector<uint64_t> h_arena(total, 0);
-  for (size_t j = 0; j < nall; j++) {
-    const Item &it = items[j];
-    uint64_t *dst = h_arena.data() + s->h_gor[it.id].arena_base;
-    uint64_t full = it.bytes / 8;
-    if (it.raw) {
-      for (uint64_t k = 0; k < full; k++) {
-        uint64_t w;
-        memcpy(&w, it.stream + k * 8, 8);
-        dst[(k >> 1) * 128 + (k & 1)] = w; /* little-endian, 16 B pairs */
-      }
-    } else {
-      for (uint64_t k = 0; k < full; k++) {
-        uint64_t w;
-        memcpy(&w, it.stream + k * 8, 8);
-        dst[k * 64] = __builtin_bswap64(w);
-      }
-      if (it.bytes & 7) {
-        uint64_t w = 0;
-        for (uint64_t b = full * 8; b < it.bytes; b++)
-          w = (w << 8) | it.stream[b];
-        dst[full * 64] = w << ((8 - (it.bytes & 7)) * 8);
-      }
-    }
-  }
   s->arena_words = total;
   HIP_CHECK(hipMalloc(&s->d_arena, total * 8));
-  HIP_CHECK(hipMemcpy(s->d_arena, h_arena.data(), total * 8,
-                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemset(s->d_arena, 0, total * 8));
+  /* stage and upload PER 64-SLOT GROUP (each group slab is bounded by
+   * max-stream-words x 512 B, ~2 MB at 4096-row segments) instead of one
+   * whole-arena host copy — a multi-GB shard would otherwise spike host
+   * RAM by the full arena size */
+  {
+    std::vector<uint64_t> slab;
+    auto upload_group = [&](size_t j, size_t ge) {
+      /* j is a group start, so the first item sits in slot 0 and its
+       * arena_base IS the group base */
+      const uint64_t gbase = s->h_gor[items[j].id].arena_base;
+      uint64_t gwords = 0;
+      for (size_t k = j; k < ge; k++) {
+        const Item &it = items[k];
+        uint64_t end_off =
+            (s->h_gor[it.id].arena_base - gbase) +
+            (it.raw ? ((it.words + 1) / 2) * 128 : it.words * 64);
+        gwords = std::max(gwords, end_off);
+      }
+      slab.assign(gwords, 0);
+      for (size_t k = j; k < ge; k++) {
+        const Item &it = items[k];
+        uint64_t *dst = slab.data() + (s->h_gor[it.id].arena_base - gbase);
+        uint64_t full = it.bytes / 8;
+        if (it.raw) {
+          for (uint64_t w = 0; w < full; w++) {
+            uint64_t v;
+            memcpy(&v, it.stream + w * 8, 8);
+            dst[(w >> 1) * 128 + (w & 1)] = v; /* little-endian, 16 B pairs */
+          }
+        } else {
+          for (uint64_t w = 0; w < full; w++) {
+            uint64_t v;
+            memcpy(&v, it.stream + w * 8, 8);
+            dst[w * 64] = __builtin_bswap64(v);
+          }
+          if (it.bytes & 7) {
+            uint64_t v = 0;
+            for (uint64_t b = full * 8; b < it.bytes; b++)
+              v = (v << 8) | it.stream[b];
+            dst[full * 64] = v << ((8 - (it.bytes & 7)) * 8);
+          }
+        }
+      }
+      HIP_CHECK(hipMemcpy(s->d_arena + gbase, slab.data(), gwords * 8,
+                          hipMemcpyHostToDevice));
+      return GEMX_OK;
+    };
+    /* groups were formed per kind (gorilla first, then raw), so iterate
+     * them the same way — a slab must never span the kind boundary */
+    for (size_t g0 = 0; g0 < ngor; g0 += 64) {
+      int rc2 = upload_group(g0, std::min(g0 + 64, ngor));
+      if (rc2 != GEMX_OK) return rc2;
+    }
+    for (size_t g0 = ngor; g0 < nall; g0 += 64) {
+      int rc2 = upload_group(g0, std::min(g0 + 64, nall));
+      if (rc2 != GEMX_OK) return rc2;
+    }
+  }
   HIP_CHECK(hipMemcpy(s->d_gor, s->h_gor.data(), sizeof(GorDesc) * s->nsegs,
                       hipMemcpyHostToDevice));
   /* underfilled gorilla grid (config #1 shape: few deep segments): walk
