@@ -114,6 +114,8 @@ def main():
                          "calls-only no-interval query served from pre-agg "
                          "metadata (SURVEY.md 3d)")
     ap.add_argument("--series", type=int, default=100_000)
+    ap.add_argument("--groups", type=int, default=1000,
+                    help="tag groups for --query tags")
     ap.add_argument("--pts", type=int, default=1000)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
@@ -159,7 +161,7 @@ def main():
     shard = gx.Shard(blob, descs, col_type, device=local_rank)
     log(f"[attach] H2D resident in {time.time()-t0:.1f}s")
 
-    NGROUPS = 1000
+    NGROUPS = args.groups
     sids_u = descs["sid"]
     keep = np.ones(len(sids_u), dtype=bool)
     keep[1:] = sids_u[1:] != sids_u[:-1]
