@@ -205,6 +205,45 @@ class TestSeededFuzzParity:
             ref = orc.scan_agg(blob, d, F, t0, t1, interval, offset=offset)
             assert_parity(rows, ref, F)
 
+    def test_fuzz_grouped(self):
+        # same zoo through the GROUPED merge (k_group_p1/p2 or
+        # k_group_small) vs oracle group_merge
+        import binding as orc
+        from shard_helpers import F, build_shard
+        import numpy as np
+        import opengemini_amd as gx
+
+        for seed in range(600, 610):
+            rng = np.random.default_rng(seed)
+            nsid = int(rng.integers(1, 80))  # crosses the small/large gate
+            sids = sorted(int(x) for x in rng.choice(
+                np.arange(1, 200), size=nsid, replace=False))
+            nf = float(rng.choice([0.0, 0.3]))
+            blob, d, _ = build_shard(rng, F, sids, seg_range=(1, 4),
+                                     row_range=(20, 700), null_frac=nf)
+            interval = int(rng.choice([13, 60])) * 10**9
+            offset = int(rng.integers(0, 2)) * 7 * 10**9
+            sh = gx.Shard(blob, d, F)
+            try:
+                grows, _ = sh.scan_agg(0, 2**62, interval, offset=offset,
+                                       group_all=True)
+                grows = grows.copy()
+            finally:
+                sh.close()
+            per = orc.scan_agg(blob, d, F, 0, 2**62, interval, offset=offset)
+            gref = orc.group_merge(per, F, interval)
+            assert len(grows) == len(gref), seed
+            for f in ("win_start", "count", "min_time", "max_time",
+                      "first_time", "last_time", "min_isnil", "sum_isnil"):
+                assert np.array_equal(grows[f], gref[f]), (seed, f)
+            for f in ("min", "max", "first", "last"):
+                assert np.array_equal(grows[f].view(np.uint64),
+                                      gref[f].view(np.uint64)), (seed, f)
+            tol = 1e-9 * np.maximum(1.0, np.abs(gref["sum"]))
+            both_nan = np.isnan(grows["sum"]) & np.isnan(gref["sum"])
+            assert np.all(both_nan | (np.abs(grows["sum"] - gref["sum"])
+                                      <= tol)), seed
+
     def test_fuzz_int(self):
         import binding as orc
         from shard_helpers import I, build_shard
