@@ -4900,6 +4900,14 @@ extern "C" int gemx_shard_attach(int device, const void *blob, uint64_t blob_byt
   {
     int grc = build_gor_arena(s, hb);
     if (grc != 0) {
+      /* the builder may have device allocations by the time it fails
+       * (corrupt-stream attach is a user-facing path, not just a HIP
+       * catastrophe) — free them before dropping the handle */
+      if (s->d_gor) (void)hipFree(s->d_gor);
+      if (s->d_arena) (void)hipFree(s->d_arena);
+      if (s->d_subs) (void)hipFree(s->d_subs);
+      if (s->d_sub_start) (void)hipFree(s->d_sub_start);
+      if (s->d_sub_count) (void)hipFree(s->d_sub_count);
       delete s;
       return grc;
     }
