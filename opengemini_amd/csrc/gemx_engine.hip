@@ -5952,8 +5952,8 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
       if (hcnt[i] > maxc) maxc = hcnt[i];
     }
     if (maxc > 4096) {
-      hipFree(d_qcnt);
-      hipFree(d_qoff);
+      (void)hipFree(d_qcnt);
+      (void)hipFree(d_qoff);
       seterr("quantile window holds >4096 points — beyond the LDS sort "
              "capacity this round");
       return GEMX_E_UNSUPPORTED;
@@ -6076,10 +6076,10 @@ static int prom_rate_impl(gemx_shard *s, int64_t start_time, int64_t end_time,
     /* synchronous by construction (host prefix-sum already synced the
      * stream once); finish the copy, then free the collect buffers */
     int rcq = rate_deliver(s, slot, P.total_rows, out_host, n_out, stats);
-    hipFree(d_qcnt);
-    hipFree(d_qoff);
-    if (d_qvals) hipFree(d_qvals);
-    if (d_qts) hipFree(d_qts);
+    (void)hipFree(d_qcnt);
+    (void)hipFree(d_qoff);
+    if (d_qvals) (void)hipFree(d_qvals);
+    if (d_qts) (void)hipFree(d_qts);
     return rcq;
   }
   if (async_begin) {
