@@ -1034,7 +1034,6 @@ int64_t orc_prom_rate(const uint8_t *blob, int64_t blob_len,
         memset(out[nout]._pad, 0, sizeof(out[nout]._pad));
         nout++;
       }
-    skip_emit:;
       if (step_ns == 0) break;
     }
   }
