@@ -330,8 +330,9 @@ def main():
                 "downsample": "per-series first/last/sum to 5m buckets "
                               "(downsample pipeline reduce shape)",
                 "rate": "PromQL rate(value[5m]) step 1m over range vectors",
-                "tags": "hash GROUP BY tag, 1000 groups x 1m windows merged "
-                        "on device (config #3 high-cardinality shape)",
+                "tags": f"hash GROUP BY tag, {args.groups} groups x 1m "
+                        "windows merged on device (config #3 "
+                        "high-cardinality shape)",
                 "preagg": "calls-only no-interval query served from pre-agg "
                           "metadata (after first scan caches it)",
             }[args.query] + f" (mode={args.mode}, {args.series}x{args.pts})",
